@@ -1,0 +1,438 @@
+// Paged attention kernels for MI355X (gfx950), bf16 KV cache.
+//
+// Behavioural parity target: the paged-attention op surface the reference
+// stack exercises through its engine images —
+// paged KV, GQA, causal chunked prefill, streaming decode.
+//
+// MI355X-first design (not a port):
+//  * decode is HBM-bandwidth-bound on the KV read -> one workgroup per
+//    (sequence, kv_head) so the GQA query group shares every KV byte read;
+//    16-lane sub-groups each own one KV token (16 lanes x ushort8 = 128 dims)
+//    so all loads are contiguous 16 B per lane, and the score reduction is a
+//    4-step shfl_xor inside the sub-group (wave64-native, no LDS traffic in
+//    the inner loop).
+//  * online softmax (flash-decode) held entirely in registers; wave-level
+//    merge via width-64 shuffles, workgroup merge via a small LDS exchange.
+//  * KV cache layout [num_blocks, kv_heads, block_size, head_dim] keeps one
+//    (block, head) tile contiguous (block_size x head_dim x 2 B = 4 KiB for
+//    head_dim 128, block 16): a whole tile streams as perfectly coalesced
+//    16 B lane loads.
+#include "ps_common.h"
+
+// ---------------------------------------------------------------------------
+// Decode: one query token per sequence.
+//   grid = (num_seqs, num_kv_heads), block = NWAVES * 64
+// ---------------------------------------------------------------------------
+template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES>
+__global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
+    unsigned short* __restrict__ out,            // [S, QH, HEAD_DIM]
+    const unsigned short* __restrict__ q,        // [S, QH, HEAD_DIM]
+    const unsigned short* __restrict__ k_cache,  // [NB, KH, BS, HD]
+    const unsigned short* __restrict__ v_cache,  // [NB, KH, BS, HD]
+    const int* __restrict__ block_tables,        // [S, max_blocks]
+    const int* __restrict__ seq_lens,            // [S]
+    int max_blocks, float scale, int KH) {
+  constexpr int D = HEAD_DIM;
+  constexpr int LPG = D / 8;       // lanes per token sub-group (16 @ D=128)
+  constexpr int TPW = 64 / LPG;    // tokens per wave per iteration (4)
+  static_assert(BLOCK_SIZE % TPW == 0, "block size must divide");
+
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int ctx = seq_lens[seq];
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int sg = lane / LPG;   // sub-group id within wave
+  const int sl = lane % LPG;   // lane within sub-group -> dims [sl*8, sl*8+8)
+
+  // Q for the whole GQA group, pre-scaled, 8 dims per lane per head.
+  float qf[GQ][8];
+  const unsigned short* qbase = q + ((long)seq * KH * GQ + (long)kvh * GQ) * D;
+#pragma unroll
+  for (int g = 0; g < GQ; g++) {
+    ps_bf16x8 qv = *(const ps_bf16x8*)(qbase + g * D + sl * 8);
+#pragma unroll
+    for (int j = 0; j < 8; j++) qf[g][j] = ps_bf16_to_f32(qv[j]) * scale;
+  }
+
+  float m[GQ], l[GQ], acc[GQ][8];
+#pragma unroll
+  for (int g = 0; g < GQ; g++) {
+    m[g] = PS_NEG_INF;
+    l[g] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; j++) acc[g][j] = 0.f;
+  }
+
+  const int nblocks = (ctx + BLOCK_SIZE - 1) / BLOCK_SIZE;
+  const int* bt = block_tables + (long)seq * max_blocks;
+  for (int b = wave; b < nblocks; b += NWAVES) {
+    const long blk = bt[b];
+    const unsigned short* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
+    const unsigned short* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
+    const int valid_tokens = min(BLOCK_SIZE, ctx - b * BLOCK_SIZE);
+#pragma unroll
+    for (int tt = 0; tt < BLOCK_SIZE / TPW; tt++) {
+      const int tok = tt * TPW + sg;
+      const bool valid = tok < valid_tokens;
+      ps_bf16x8 kv = *(const ps_bf16x8*)(kb + tok * D + sl * 8);
+      ps_bf16x8 vv = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
+      float vf[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) vf[j] = ps_bf16_to_f32(vv[j]);
+#pragma unroll
+      for (int g = 0; g < GQ; g++) {
+        float s = 0.f;
+#pragma unroll
+        for (int j = 0; j < 8; j++) s += qf[g][j] * ps_bf16_to_f32(kv[j]);
+        s = ps_group_sum<LPG>(s);
+        const float sv = valid ? s : PS_NEG_INF;
+        const float mnew = fmaxf(m[g], sv);
+        const float corr = __expf(m[g] - mnew);  // <=1; 1 when both -inf
+        const float p = valid ? __expf(sv - mnew) : 0.f;
+        l[g] = l[g] * corr + p;
+#pragma unroll
+        for (int j = 0; j < 8; j++) acc[g][j] = acc[g][j] * corr + p * vf[j];
+        m[g] = mnew;
+      }
+    }
+  }
+
+  // Merge the TPW sub-groups of this wave (lanes sl, sl+LPG, ... hold
+  // partial state for the same dims).
+#pragma unroll
+  for (int off = LPG; off < 64; off <<= 1) {
+#pragma unroll
+    for (int g = 0; g < GQ; g++) {
+      const float mo = __shfl_xor(m[g], off, 64);
+      const float lo = __shfl_xor(l[g], off, 64);
+      const float mnew = fmaxf(m[g], mo);
+      const float c1 = __expf(m[g] - mnew);
+      const float c2 = __expf(mo - mnew);
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        const float ao = __shfl_xor(acc[g][j], off, 64);
+        acc[g][j] = acc[g][j] * c1 + ao * c2;
+      }
+      l[g] = l[g] * c1 + lo * c2;
+      m[g] = mnew;
+    }
+  }
+
+  // Merge the NWAVES waves via LDS.
+  __shared__ float sm[NWAVES][GQ];
+  __shared__ float slse[NWAVES][GQ];
+  __shared__ float sacc[NWAVES][GQ][D];
+  if (sg == 0) {
+#pragma unroll
+    for (int g = 0; g < GQ; g++) {
+      if (sl == 0) {
+        sm[wave][g] = m[g];
+        slse[wave][g] = l[g];
+      }
+#pragma unroll
+      for (int j = 0; j < 8; j++) sacc[wave][g][sl * 8 + j] = acc[g][j];
+    }
+  }
+  __syncthreads();
+  if (wave == 0 && sg == 0) {
+#pragma unroll
+    for (int g = 0; g < GQ; g++) {
+      float M = PS_NEG_INF, L = 0.f, A[8];
+#pragma unroll
+      for (int j = 0; j < 8; j++) A[j] = 0.f;
+#pragma unroll
+      for (int wv = 0; wv < NWAVES; wv++) {
+        const float mw = sm[wv][g];
+        const float mnew = fmaxf(M, mw);
+        const float c1 = __expf(M - mnew);
+        const float c2 = __expf(mw - mnew);
+        L = L * c1 + slse[wv][g] * c2;
+#pragma unroll
+        for (int j = 0; j < 8; j++)
+          A[j] = A[j] * c1 + sacc[wv][g][sl * 8 + j] * c2;
+        M = mnew;
+      }
+      const float inv = L > 0.f ? 1.f / L : 0.f;
+      ps_bf16x8 ov;
+#pragma unroll
+      for (int j = 0; j < 8; j++) ov[j] = ps_f32_to_bf16(A[j] * inv);
+      *(ps_bf16x8*)(out + ((long)seq * KH * GQ + (long)kvh * GQ + g) * D +
+                    sl * 8) = ov;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Chunked prefill: a batch of query tokens (possibly many sequences, variable
+// lengths) attends causally to the paged KV cache, which already contains
+// every query token's K/V (appended before the call).
+//   grid = (num_q_tokens, num_q_heads), block = 64 (one wave per (token, head))
+// Compute-bound VALU v1; MFMA tile version is the planned upgrade.
+// ---------------------------------------------------------------------------
+template <int HEAD_DIM, int BLOCK_SIZE>
+__global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
+    unsigned short* __restrict__ out,            // [T, QH, HD]
+    const unsigned short* __restrict__ q,        // [T, QH, HD]
+    const unsigned short* __restrict__ k_cache,  // [NB, KH, BS, HD]
+    const unsigned short* __restrict__ v_cache,
+    const int* __restrict__ block_tables,  // [S, max_blocks]
+    const int* __restrict__ token_seq,     // [T] sequence index per q token
+    const int* __restrict__ token_pos,     // [T] absolute position per q token
+    int max_blocks, float scale, int KH, int GQ) {
+  constexpr int D = HEAD_DIM;
+  constexpr int LPG = D / 8;
+  constexpr int TPW = 64 / LPG;
+
+  const int tokid = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int kvh = qh / GQ;
+  const int seq = token_seq[tokid];
+  const int pos = token_pos[tokid];
+  const int ctx = pos + 1;  // causal: attend to [0, pos]
+  const int lane = threadIdx.x & 63;
+  const int sg = lane / LPG;
+  const int sl = lane % LPG;
+  const int QH = KH * GQ;
+
+  float qf[8];
+  {
+    ps_bf16x8 qv =
+        *(const ps_bf16x8*)(q + ((long)tokid * QH + qh) * D + sl * 8);
+#pragma unroll
+    for (int j = 0; j < 8; j++) qf[j] = ps_bf16_to_f32(qv[j]) * scale;
+  }
+
+  float m = PS_NEG_INF, l = 0.f, acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++) acc[j] = 0.f;
+
+  const int nblocks = (ctx + BLOCK_SIZE - 1) / BLOCK_SIZE;
+  const int* bt = block_tables + (long)seq * max_blocks;
+  for (int b = 0; b < nblocks; b++) {
+    const long blk = bt[b];
+    const unsigned short* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
+    const unsigned short* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
+    const int valid_tokens = min(BLOCK_SIZE, ctx - b * BLOCK_SIZE);
+#pragma unroll
+    for (int tt = 0; tt < BLOCK_SIZE / TPW; tt++) {
+      const int tok = tt * TPW + sg;
+      const bool valid = tok < valid_tokens;
+      ps_bf16x8 kv = *(const ps_bf16x8*)(kb + tok * D + sl * 8);
+      ps_bf16x8 vv = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; j++) s += qf[j] * ps_bf16_to_f32(kv[j]);
+      s = ps_group_sum<LPG>(s);
+      const float sv = valid ? s : PS_NEG_INF;
+      const float mnew = fmaxf(m, sv);
+      const float corr = __expf(m - mnew);
+      const float p = valid ? __expf(sv - mnew) : 0.f;
+      l = l * corr + p;
+#pragma unroll
+      for (int j = 0; j < 8; j++)
+        acc[j] = acc[j] * corr + p * ps_bf16_to_f32(vv[j]);
+      m = mnew;
+    }
+  }
+
+  // merge sub-groups within the wave
+#pragma unroll
+  for (int off = LPG; off < 64; off <<= 1) {
+    const float mo = __shfl_xor(m, off, 64);
+    const float lo = __shfl_xor(l, off, 64);
+    const float mnew = fmaxf(m, mo);
+    const float c1 = __expf(m - mnew);
+    const float c2 = __expf(mo - mnew);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      const float ao = __shfl_xor(acc[j], off, 64);
+      acc[j] = acc[j] * c1 + ao * c2;
+    }
+    l = l * c1 + lo * c2;
+    m = mnew;
+  }
+  if (sg == 0) {
+    const float inv = l > 0.f ? 1.f / l : 0.f;
+    ps_bf16x8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; j++) ov[j] = ps_f32_to_bf16(acc[j] * inv);
+    *(ps_bf16x8*)(out + ((long)tokid * QH + qh) * D + sl * 8) = ov;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// KV append (reshape_and_cache): scatter the new tokens' K/V into the paged
+// cache according to slot_mapping (slot = block_id * BLOCK_SIZE + offset).
+// ---------------------------------------------------------------------------
+__global__ void reshape_and_cache_kernel(
+    const unsigned short* __restrict__ k,  // [T, KH*HD]
+    const unsigned short* __restrict__ v,  // [T, KH*HD]
+    unsigned short* __restrict__ k_cache,  // [NB, KH, BS, HD]
+    unsigned short* __restrict__ v_cache,
+    const long* __restrict__ slot_mapping,  // [T]
+    int KH, int HD, int BS, long total /* = T * KH * HD / 8 */) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  const int row = KH * HD;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += stride) {
+    const long elem = idx * 8;
+    const long t = elem / row;
+    const int r = (int)(elem % row);
+    const int h = r / HD;
+    const int d = r % HD;
+    const long slot = slot_mapping[t];
+    if (slot < 0) continue;
+    const long blk = slot / BS;
+    const int off = (int)(slot % BS);
+    const long dst = ((blk * KH + h) * BS + off) * (long)HD + d;
+    *(ps_bf16x8*)(k_cache + dst) = *(const ps_bf16x8*)(k + elem);
+    *(ps_bf16x8*)(v_cache + dst) = *(const ps_bf16x8*)(v + elem);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Greedy sampling: argmax over the vocab per row.
+//   grid = (num_rows), block = 256
+// ---------------------------------------------------------------------------
+__global__ void greedy_sample_kernel(long* __restrict__ out,  // [R]
+                                     const unsigned short* __restrict__ logits,
+                                     int V) {
+  const long r = blockIdx.x;
+  const unsigned short* row = logits + r * (long)V;
+  float best = PS_NEG_INF;
+  int besti = 0;
+  for (int i = threadIdx.x * 8; i + 8 <= V; i += blockDim.x * 8) {
+    ps_bf16x8 xv = *(const ps_bf16x8*)(row + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      const float f = ps_bf16_to_f32(xv[j]);
+      if (f > best) {
+        best = f;
+        besti = i + j;
+      }
+    }
+  }
+  // tail (V not multiple of 8*block)
+  const int tail_start = (V / 8) * 8;
+  for (int i = tail_start + threadIdx.x; i < V; i += blockDim.x) {
+    const float f = ps_bf16_to_f32(row[i]);
+    if (f > best) {
+      best = f;
+      besti = i;
+    }
+  }
+  // wave reduce keeping the smallest index among ties
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ob = __shfl_xor(best, off, 64);
+    const int oi = __shfl_xor(besti, off, 64);
+    if (ob > best || (ob == best && oi < besti)) {
+      best = ob;
+      besti = oi;
+    }
+  }
+  __shared__ float wbest[4];
+  __shared__ int wbesti[4];
+  const int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) {
+    wbest[wave] = best;
+    wbesti[wave] = besti;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int wv = 1; wv < (int)(blockDim.x >> 6); wv++) {
+      if (wbest[wv] > wbest[0] ||
+          (wbest[wv] == wbest[0] && wbesti[wv] < wbesti[0])) {
+        wbest[0] = wbest[wv];
+        wbesti[0] = wbesti[wv];
+      }
+    }
+    out[r] = wbesti[0];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C launchers
+// ---------------------------------------------------------------------------
+extern "C" {
+
+int ps_paged_attn_decode(void* out, const void* q, const void* k_cache,
+                         const void* v_cache, const void* block_tables,
+                         const void* seq_lens, int num_seqs, int max_blocks,
+                         float scale, int KH, int GQ, int head_dim,
+                         int block_size, hipStream_t stream) {
+  dim3 grid(num_seqs, KH);
+  constexpr int NW = 4;
+  dim3 block(NW * 64);
+#define PS_DISPATCH_DECODE(HD, G, BS)                                        \
+  paged_attn_decode_kernel<HD, G, BS, NW><<<grid, block, 0, stream>>>(       \
+      (unsigned short*)out, (const unsigned short*)q,                        \
+      (const unsigned short*)k_cache, (const unsigned short*)v_cache,        \
+      (const int*)block_tables, (const int*)seq_lens, max_blocks, scale, KH)
+  if (head_dim == 128 && block_size == 16) {
+    switch (GQ) {
+      case 1: PS_DISPATCH_DECODE(128, 1, 16); return 0;
+      case 2: PS_DISPATCH_DECODE(128, 2, 16); return 0;
+      case 4: PS_DISPATCH_DECODE(128, 4, 16); return 0;
+      case 8: PS_DISPATCH_DECODE(128, 8, 16); return 0;
+      default: return -1;
+    }
+  }
+  if (head_dim == 64 && block_size == 16) {
+    switch (GQ) {
+      case 1: PS_DISPATCH_DECODE(64, 1, 16); return 0;
+      case 4: PS_DISPATCH_DECODE(64, 4, 16); return 0;
+      default: return -1;
+    }
+  }
+  return -1;
+#undef PS_DISPATCH_DECODE
+}
+
+int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
+                          const void* v_cache, const void* block_tables,
+                          const void* token_seq, const void* token_pos,
+                          int num_tokens, int num_q_heads, int max_blocks,
+                          float scale, int KH, int GQ, int head_dim,
+                          int block_size, hipStream_t stream) {
+  dim3 grid(num_tokens, num_q_heads);
+  dim3 block(64);
+#define PS_DISPATCH_PREFILL(HD, BS)                                         \
+  paged_attn_prefill_kernel<HD, BS><<<grid, block, 0, stream>>>(            \
+      (unsigned short*)out, (const unsigned short*)q,                       \
+      (const unsigned short*)k_cache, (const unsigned short*)v_cache,       \
+      (const int*)block_tables, (const int*)token_seq,                      \
+      (const int*)token_pos, max_blocks, scale, KH, GQ)
+  if (head_dim == 128 && block_size == 16) {
+    PS_DISPATCH_PREFILL(128, 16);
+    return 0;
+  }
+  if (head_dim == 64 && block_size == 16) {
+    PS_DISPATCH_PREFILL(64, 16);
+    return 0;
+  }
+  return -1;
+#undef PS_DISPATCH_PREFILL
+}
+
+void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
+                          void* v_cache, const void* slot_mapping, long T,
+                          int KH, int HD, int BS, hipStream_t stream) {
+  const long total = T * (long)KH * HD / 8;
+  const int block = 256;
+  const long want = (total + block - 1) / block;
+  const int grid = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
+  reshape_and_cache_kernel<<<grid, block, 0, stream>>>(
+      (const unsigned short*)k, (const unsigned short*)v,
+      (unsigned short*)k_cache, (unsigned short*)v_cache,
+      (const long*)slot_mapping, KH, HD, BS, total);
+}
+
+void ps_greedy_sample(void* out, const void* logits, long R, int V,
+                      hipStream_t stream) {
+  greedy_sample_kernel<<<dim3((unsigned)R), 256, 0, stream>>>(
+      (long*)out, (const unsigned short*)logits, V);
+}
+
+}  // extern "C"

@@ -1,0 +1,195 @@
+// PyTorch bindings for the MI355X serving kernels (production_stack_amd._C).
+//
+// The kernels themselves live in *.hip translation units behind a plain C
+// ABI (raw pointers + hipStream_t); this file only validates tensors and
+// forwards to them on the current HIP stream.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+extern "C" {
+void ps_rms_norm(void* out, const void* x, const void* w, float eps, long T,
+                 int D, hipStream_t stream);
+void ps_fused_add_rms_norm(void* x, void* residual, const void* w, float eps,
+                           long T, int D, hipStream_t stream);
+void ps_silu_and_mul(void* out, const void* x, long T, int D,
+                     hipStream_t stream);
+void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
+             long T, int QH, int KH, int HD, int ROT, hipStream_t stream);
+int ps_paged_attn_decode(void* out, const void* q, const void* k_cache,
+                         const void* v_cache, const void* block_tables,
+                         const void* seq_lens, int num_seqs, int max_blocks,
+                         float scale, int KH, int GQ, int head_dim,
+                         int block_size, hipStream_t stream);
+int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
+                          const void* v_cache, const void* block_tables,
+                          const void* token_seq, const void* token_pos,
+                          int num_tokens, int num_q_heads, int max_blocks,
+                          float scale, int KH, int GQ, int head_dim,
+                          int block_size, hipStream_t stream);
+void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
+                          void* v_cache, const void* slot_mapping, long T,
+                          int KH, int HD, int BS, hipStream_t stream);
+void ps_greedy_sample(void* out, const void* logits, long R, int V,
+                      hipStream_t stream);
+}
+
+namespace {
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_GPU_BF16(t)                                             \
+  TORCH_CHECK((t).is_cuda(), #t " must be on the GPU");               \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16"); \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+#define CHECK_GPU_DTYPE(t, ty)                                  \
+  TORCH_CHECK((t).is_cuda(), #t " must be on the GPU");         \
+  TORCH_CHECK((t).scalar_type() == (ty), #t " dtype mismatch"); \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+void rms_norm(at::Tensor out, at::Tensor x, at::Tensor w, double eps) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_BF16(x);
+  CHECK_GPU_BF16(w);
+  const long T = x.numel() / x.size(-1);
+  const int D = (int)x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "hidden dim must be a multiple of 8");
+  ps_rms_norm(out.data_ptr(), x.data_ptr(), w.data_ptr(), (float)eps, T, D,
+              current_stream());
+}
+
+void fused_add_rms_norm(at::Tensor x, at::Tensor residual, at::Tensor w,
+                        double eps) {
+  CHECK_GPU_BF16(x);
+  CHECK_GPU_BF16(residual);
+  CHECK_GPU_BF16(w);
+  const long T = x.numel() / x.size(-1);
+  const int D = (int)x.size(-1);
+  TORCH_CHECK(D % 8 == 0, "hidden dim must be a multiple of 8");
+  ps_fused_add_rms_norm(x.data_ptr(), residual.data_ptr(), w.data_ptr(),
+                        (float)eps, T, D, current_stream());
+}
+
+void silu_and_mul(at::Tensor out, at::Tensor x) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_BF16(x);
+  const int D = (int)out.size(-1);
+  TORCH_CHECK(x.size(-1) == 2 * D, "x last dim must be 2*out last dim");
+  TORCH_CHECK(D % 8 == 0, "feature dim must be a multiple of 8");
+  const long T = out.numel() / D;
+  ps_silu_and_mul(out.data_ptr(), x.data_ptr(), T, D, current_stream());
+}
+
+void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
+                      at::Tensor cos_sin, long head_dim) {
+  CHECK_GPU_DTYPE(positions, at::kInt);
+  CHECK_GPU_BF16(q);
+  CHECK_GPU_BF16(k);
+  CHECK_GPU_DTYPE(cos_sin, at::kFloat);
+  const long T = positions.size(0);
+  const int HD = (int)head_dim;
+  const int ROT = (int)cos_sin.size(-1);
+  const int QH = (int)(q.numel() / T / HD);
+  const int KH = (int)(k.numel() / T / HD);
+  ps_rope(positions.data_ptr(), q.data_ptr(), k.data_ptr(),
+          cos_sin.data_ptr(), T, QH, KH, HD, ROT, current_stream());
+}
+
+void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor block_tables,
+                       at::Tensor seq_lens, double scale) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_BF16(q);
+  CHECK_GPU_BF16(k_cache);
+  CHECK_GPU_BF16(v_cache);
+  CHECK_GPU_DTYPE(block_tables, at::kInt);
+  CHECK_GPU_DTYPE(seq_lens, at::kInt);
+  const int S = (int)q.size(0);
+  const int QH = (int)q.size(1);
+  const int HD = (int)q.size(2);
+  const int KH = (int)k_cache.size(1);
+  const int BS = (int)k_cache.size(2);
+  TORCH_CHECK(QH % KH == 0, "GQA group mismatch");
+  const int GQ = QH / KH;
+  const int max_blocks = (int)block_tables.size(1);
+  int rc = ps_paged_attn_decode(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr(), seq_lens.data_ptr(), S, max_blocks,
+      (float)scale, KH, GQ, HD, BS, current_stream());
+  TORCH_CHECK(rc == 0, "unsupported decode config: head_dim=", HD,
+              " block_size=", BS, " gqa=", GQ);
+}
+
+void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
+                        at::Tensor v_cache, at::Tensor block_tables,
+                        at::Tensor token_seq, at::Tensor token_pos,
+                        double scale) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_BF16(q);
+  CHECK_GPU_BF16(k_cache);
+  CHECK_GPU_BF16(v_cache);
+  CHECK_GPU_DTYPE(block_tables, at::kInt);
+  CHECK_GPU_DTYPE(token_seq, at::kInt);
+  CHECK_GPU_DTYPE(token_pos, at::kInt);
+  const int T = (int)q.size(0);
+  const int QH = (int)q.size(1);
+  const int HD = (int)q.size(2);
+  const int KH = (int)k_cache.size(1);
+  const int BS = (int)k_cache.size(2);
+  TORCH_CHECK(QH % KH == 0, "GQA group mismatch");
+  const int GQ = QH / KH;
+  const int max_blocks = (int)block_tables.size(1);
+  int rc = ps_paged_attn_prefill(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr(), token_seq.data_ptr(), token_pos.data_ptr(), T,
+      QH, max_blocks, (float)scale, KH, GQ, HD, BS, current_stream());
+  TORCH_CHECK(rc == 0, "unsupported prefill config: head_dim=", HD,
+              " block_size=", BS);
+}
+
+void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
+                       at::Tensor v_cache, at::Tensor slot_mapping) {
+  CHECK_GPU_BF16(k);
+  CHECK_GPU_BF16(v);
+  CHECK_GPU_BF16(k_cache);
+  CHECK_GPU_BF16(v_cache);
+  CHECK_GPU_DTYPE(slot_mapping, at::kLong);
+  const long T = slot_mapping.size(0);
+  const int KH = (int)k_cache.size(1);
+  const int BS = (int)k_cache.size(2);
+  const int HD = (int)k_cache.size(3);
+  TORCH_CHECK(k.numel() == T * KH * HD, "k shape mismatch");
+  ps_reshape_and_cache(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
+                       v_cache.data_ptr(), slot_mapping.data_ptr(), T, KH, HD,
+                       BS, current_stream());
+}
+
+void greedy_sample(at::Tensor out, at::Tensor logits) {
+  CHECK_GPU_DTYPE(out, at::kLong);
+  CHECK_GPU_BF16(logits);
+  const long R = logits.size(0);
+  const int V = (int)logits.size(1);
+  ps_greedy_sample(out.data_ptr(), logits.data_ptr(), R, V, current_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "RMSNorm (bf16)");
+  m.def("fused_add_rms_norm", &fused_add_rms_norm,
+        "Fused residual add + RMSNorm (bf16, in-place)");
+  m.def("silu_and_mul", &silu_and_mul, "SiLU-and-mul (bf16)");
+  m.def("rotary_embedding", &rotary_embedding,
+        "Neox-style rotary embedding (bf16, in-place)");
+  m.def("paged_attn_decode", &paged_attn_decode,
+        "Paged attention, decode phase (bf16 KV)");
+  m.def("paged_attn_prefill", &paged_attn_prefill,
+        "Paged attention, chunked prefill (bf16 KV)");
+  m.def("reshape_and_cache", &reshape_and_cache,
+        "Append K/V for new tokens into the paged cache");
+  m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");
+}

@@ -1,0 +1,235 @@
+// Fused normalisation / activation / rotary kernels for the MI355X engine.
+//
+// Parity targets (behavioural, not code): the external vLLM engine's
+// rms_norm / fused_add_rms_norm / silu_and_mul / rotary_embedding op surface
+// that /root/reference drives through its helm engine flags
+// (see SURVEY.md section 2.8).  All kernels are memory-bound: the design
+// point is HBM3E bandwidth (~6.3 TB/s achievable), so every bf16 access is a
+// 16-byte ushort8 vector and each tensor element is touched exactly once
+// where possible.
+#include "ps_common.h"
+
+// ---------------------------------------------------------------------------
+// RMSNorm:  out[t, :] = x[t, :] * rsqrt(mean(x^2) + eps) * w
+// One workgroup per token row; row kept in registers between the two passes.
+// ---------------------------------------------------------------------------
+template <int BLOCK, int MAX_VPT>  // MAX_VPT = max ushort8 vectors per thread
+__global__ __launch_bounds__(BLOCK) void rms_norm_kernel(
+    unsigned short* __restrict__ out,      // [T, D]
+    const unsigned short* __restrict__ x,  // [T, D]
+    const unsigned short* __restrict__ w,  // [D]
+    float eps, int D) {
+  const long t = blockIdx.x;
+  const unsigned short* xr = x + t * (long)D;
+  unsigned short* orow = out + t * (long)D;
+
+  float vals[MAX_VPT][8];
+  int nvec = 0;
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 xv = *(const ps_bf16x8*)(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = ps_bf16_to_f32(xv[j]);
+      vals[nvec][j] = f;
+      ss += f * f;
+    }
+    nvec++;
+  }
+  // reduce ss across the block
+  __shared__ float red[BLOCK / 64];
+  ss = ps_group_sum<64>(ss);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tot = 0.f;
+#pragma unroll
+    for (int wv = 0; wv < BLOCK / 64; wv++) tot += red[wv];
+    red[0] = rsqrtf(tot / (float)D + eps);
+  }
+  __syncthreads();
+  const float inv = red[0];
+
+  nvec = 0;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 wv = *(const ps_bf16x8*)(w + i);
+    ps_bf16x8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      ov[j] = ps_f32_to_bf16(vals[nvec][j] * inv * ps_bf16_to_f32(wv[j]));
+    *(ps_bf16x8*)(orow + i) = ov;
+    nvec++;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused residual-add + RMSNorm (the per-layer hot path):
+//   residual[t,:] += x[t,:];  x[t,:] = rmsnorm(residual[t,:]) * w
+// Keeps the summed row in registers: 2 reads + 2 writes per element.
+// ---------------------------------------------------------------------------
+template <int BLOCK, int MAX_VPT>
+__global__ __launch_bounds__(BLOCK) void fused_add_rms_norm_kernel(
+    unsigned short* __restrict__ x,         // [T, D] in: delta, out: normed
+    unsigned short* __restrict__ residual,  // [T, D] in/out: accumulated
+    const unsigned short* __restrict__ w,   // [D]
+    float eps, int D) {
+  const long t = blockIdx.x;
+  unsigned short* xr = x + t * (long)D;
+  unsigned short* rr = residual + t * (long)D;
+
+  float vals[MAX_VPT][8];
+  int nvec = 0;
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 xv = *(const ps_bf16x8*)(xr + i);
+    ps_bf16x8 rv = *(const ps_bf16x8*)(rr + i);
+    ps_bf16x8 sv;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = ps_bf16_to_f32(xv[j]) + ps_bf16_to_f32(rv[j]);
+      vals[nvec][j] = f;
+      ss += f * f;
+      sv[j] = ps_f32_to_bf16(f);
+    }
+    *(ps_bf16x8*)(rr + i) = sv;
+    nvec++;
+  }
+  __shared__ float red[BLOCK / 64];
+  ss = ps_group_sum<64>(ss);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tot = 0.f;
+#pragma unroll
+    for (int wv = 0; wv < BLOCK / 64; wv++) tot += red[wv];
+    red[0] = rsqrtf(tot / (float)D + eps);
+  }
+  __syncthreads();
+  const float inv = red[0];
+
+  nvec = 0;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 wv = *(const ps_bf16x8*)(w + i);
+    ps_bf16x8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      ov[j] = ps_f32_to_bf16(vals[nvec][j] * inv * ps_bf16_to_f32(wv[j]));
+    *(ps_bf16x8*)(xr + i) = ov;
+    nvec++;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// SiLU-and-mul: out[t, d] = silu(x[t, d]) * x[t, D + d], D = out feature dim.
+// ---------------------------------------------------------------------------
+__global__ void silu_and_mul_kernel(unsigned short* __restrict__ out,
+                                    const unsigned short* __restrict__ x,
+                                    int D, long total /* = T * D / 8 */) {
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += stride) {
+    const long elem = idx * 8;
+    const long t = elem / D;
+    const long d = elem % D;
+    ps_bf16x8 g = *(const ps_bf16x8*)(x + t * 2 * (long)D + d);
+    ps_bf16x8 u = *(const ps_bf16x8*)(x + t * 2 * (long)D + D + d);
+    ps_bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float gf = ps_bf16_to_f32(g[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o[j] = ps_f32_to_bf16(s * ps_bf16_to_f32(u[j]));
+    }
+    *(ps_bf16x8*)(out + elem) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Rotary embedding, GPT-NeoX style (Llama): rotate (r, r + D/2) pairs in the
+// leading rot_dim dims of every q / k head.  cos/sin are precomputed on the
+// host (guide App. B: never evaluate trig on-device for RoPE).
+// cos_sin layout: [max_pos, rot_dim]  (cos[rot/2] || sin[rot/2]), fp32.
+// ---------------------------------------------------------------------------
+__global__ void rope_kernel(const int* __restrict__ positions,  // [T]
+                            unsigned short* __restrict__ q,     // [T, QH*HD]
+                            unsigned short* __restrict__ k,     // [T, KH*HD]
+                            const float* __restrict__ cos_sin,
+                            int QH, int KH, int HD, int ROT) {
+  const long t = blockIdx.x;
+  const float* cs = cos_sin + (long)positions[t] * ROT;
+  const int half = ROT / 2;
+  const int total = (QH + KH) * half;
+  for (int idx = threadIdx.x; idx < total; idx += blockDim.x) {
+    const int h = idx / half;
+    const int r = idx % half;
+    unsigned short* base = (h < QH) ? (q + t * (long)QH * HD + (long)h * HD)
+                                    : (k + t * (long)KH * HD + (long)(h - QH) * HD);
+    const float c = cs[r];
+    const float s = cs[half + r];
+    const float x1 = ps_bf16_to_f32(base[r]);
+    const float x2 = ps_bf16_to_f32(base[r + half]);
+    base[r] = ps_f32_to_bf16(x1 * c - x2 * s);
+    base[r + half] = ps_f32_to_bf16(x2 * c + x1 * s);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C launchers
+// ---------------------------------------------------------------------------
+extern "C" {
+
+void ps_rms_norm(void* out, const void* x, const void* w, float eps, long T,
+                 int D, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)T);
+  if (D <= BLOCK * 8 * 2)
+    rms_norm_kernel<BLOCK, 2><<<grid, BLOCK, 0, stream>>>(
+        (unsigned short*)out, (const unsigned short*)x,
+        (const unsigned short*)w, eps, D);
+  else if (D <= BLOCK * 8 * 4)
+    rms_norm_kernel<BLOCK, 4><<<grid, BLOCK, 0, stream>>>(
+        (unsigned short*)out, (const unsigned short*)x,
+        (const unsigned short*)w, eps, D);
+  else
+    rms_norm_kernel<BLOCK, 16><<<grid, BLOCK, 0, stream>>>(
+        (unsigned short*)out, (const unsigned short*)x,
+        (const unsigned short*)w, eps, D);
+}
+
+void ps_fused_add_rms_norm(void* x, void* residual, const void* w, float eps,
+                           long T, int D, hipStream_t stream) {
+  constexpr int BLOCK = 256;
+  dim3 grid((unsigned)T);
+  if (D <= BLOCK * 8 * 2)
+    fused_add_rms_norm_kernel<BLOCK, 2><<<grid, BLOCK, 0, stream>>>(
+        (unsigned short*)x, (unsigned short*)residual,
+        (const unsigned short*)w, eps, D);
+  else if (D <= BLOCK * 8 * 4)
+    fused_add_rms_norm_kernel<BLOCK, 4><<<grid, BLOCK, 0, stream>>>(
+        (unsigned short*)x, (unsigned short*)residual,
+        (const unsigned short*)w, eps, D);
+  else
+    fused_add_rms_norm_kernel<BLOCK, 16><<<grid, BLOCK, 0, stream>>>(
+        (unsigned short*)x, (unsigned short*)residual,
+        (const unsigned short*)w, eps, D);
+}
+
+void ps_silu_and_mul(void* out, const void* x, long T, int D,
+                     hipStream_t stream) {
+  const long total = T * (long)D / 8;
+  const int block = 256;
+  const long want = (total + block - 1) / block;
+  const int grid = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
+  silu_and_mul_kernel<<<grid, block, 0, stream>>>(
+      (unsigned short*)out, (const unsigned short*)x, D, total);
+}
+
+void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
+             long T, int QH, int KH, int HD, int ROT, hipStream_t stream) {
+  const int block = 256;
+  rope_kernel<<<dim3((unsigned)T), block, 0, stream>>>(
+      (const int*)positions, (unsigned short*)q, (unsigned short*)k,
+      (const float*)cos_sin, QH, KH, HD, ROT);
+}
+
+}  // extern "C"

@@ -1,0 +1,144 @@
+"""Op dispatch layer.
+
+On a GPU (any tensor on `cuda`), every op REQUIRES the in-tree HIP extension
+production_stack_amd._C and raises if it is missing -- there is no silent
+eager fallback on MI355X. On CPU tensors the pure-PyTorch references run so
+the engine and its tests work in GPU-less CI.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from production_stack_amd.ops import reference
+
+try:
+    from production_stack_amd import _C  # type: ignore[attr-defined]
+except ImportError:  # pragma: no cover - exercised only when not built
+    _C = None
+
+HAVE_EXT = _C is not None
+
+
+def _require_ext() -> None:
+    if _C is None:
+        raise RuntimeError(
+            "production_stack_amd._C HIP extension is not built; run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace`. "
+            "Refusing to fall back to eager PyTorch on GPU."
+        )
+
+
+def rms_norm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        out = torch.empty_like(x)
+        _C.rms_norm(out, x, w, eps)
+        return out
+    return reference.rms_norm(x, w, eps)
+
+
+def fused_add_rms_norm(
+    x: torch.Tensor, residual: torch.Tensor, w: torch.Tensor, eps: float
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """In-place on GPU: x <- normed(residual + x); residual <- residual + x."""
+    if x.is_cuda:
+        _require_ext()
+        _C.fused_add_rms_norm(x, residual, w, eps)
+        return x, residual
+    normed, summed = reference.fused_add_rms_norm(x, residual, w, eps)
+    return normed, summed
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        _require_ext()
+        d = x.shape[-1] // 2
+        out = torch.empty(x.shape[:-1] + (d,), dtype=x.dtype, device=x.device)
+        _C.silu_and_mul(out, x)
+        return out
+    return reference.silu_and_mul(x)
+
+
+def rotary_embedding(
+    positions: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    cos_sin: torch.Tensor,
+    head_dim: int,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """In-place on GPU. positions int32, q/k bf16 [T, H*HD], cos_sin fp32."""
+    if q.is_cuda:
+        _require_ext()
+        _C.rotary_embedding(positions, q, k, cos_sin, head_dim)
+        return q, k
+    return reference.rotary_embedding(positions, q, k, cos_sin, head_dim)
+
+
+def reshape_and_cache(
+    k: torch.Tensor,
+    v: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if k_cache.is_cuda:
+        _require_ext()
+        _C.reshape_and_cache(
+            k.view(k.shape[0], -1), v.view(v.shape[0], -1), k_cache, v_cache,
+            slot_mapping,
+        )
+        return
+    reference.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def paged_attn_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    seq_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        _C.paged_attn_decode(
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+        )
+        return out
+    return reference.paged_attn_decode(
+        q, k_cache, v_cache, block_tables, seq_lens, scale
+    )
+
+
+def paged_attn_prefill(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    token_seq: torch.Tensor,
+    token_pos: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty_like(q)
+        _C.paged_attn_prefill(
+            out, q, k_cache, v_cache, block_tables, token_seq, token_pos, scale
+        )
+        return out
+    return reference.paged_attn_prefill(
+        q, k_cache, v_cache, block_tables, token_seq, token_pos, scale
+    )
+
+
+def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
+    if logits.is_cuda:
+        _require_ext()
+        out = torch.empty(
+            logits.shape[0], dtype=torch.long, device=logits.device
+        )
+        _C.greedy_sample(out, logits)
+        return out
+    return reference.greedy_sample(logits)

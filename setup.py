@@ -1,0 +1,39 @@
+"""Build the in-tree HIP extension for the MI355X serving stack.
+
+Usage:  PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+The resulting production_stack_amd/_C*.so is git-ignored but ships with the
+repo snapshot to GPU boxes.
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup  # noqa: E402
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+ext = CUDAExtension(
+    name="production_stack_amd._C",
+    sources=[
+        "csrc/bindings.cpp",
+        "csrc/norm_act_rope.hip",
+        "csrc/attention.hip",
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": [
+            "-O3",
+            "-std=c++17",
+            "--offload-arch=gfx950",
+        ],
+    },
+)
+
+setup(
+    name="production-stack-amd-kernels",
+    version="0.1.0",
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

@@ -1,0 +1,155 @@
+"""Numerics tests: every HIP kernel vs the plain-PyTorch fp32 reference.
+
+All tests are gpu-marked; the references themselves are covered by CPU tests
+elsewhere. Tolerances account for bf16 storage (~2^-8 relative).
+"""
+
+import pytest
+import torch
+
+from production_stack_amd import ops
+from production_stack_amd.ops import reference
+
+pytestmark = pytest.mark.gpu
+
+
+def setup_module(module):
+    torch.manual_seed(0)
+    assert ops.HAVE_EXT, "HIP extension must be built on a GPU box"
+
+
+def _close(a, b, atol=2e-2, rtol=2e-2):
+    torch.testing.assert_close(a.float().cpu(), b.float().cpu(), atol=atol, rtol=rtol)
+
+
+@pytest.mark.parametrize("shape", [(7, 4096), (33, 8192), (1, 1024), (256, 4096)])
+def test_rms_norm(shape):
+    x = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device="cuda")
+    got = ops.rms_norm(x, w, 1e-5)
+    want = reference.rms_norm(x.cpu(), w.cpu(), 1e-5)
+    _close(got, want)
+
+
+@pytest.mark.parametrize("shape", [(13, 4096), (256, 8192)])
+def test_fused_add_rms_norm(shape):
+    x = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    res = torch.randn(shape, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(shape[-1], dtype=torch.bfloat16, device="cuda")
+    want_n, want_r = reference.fused_add_rms_norm(x.cpu(), res.cpu(), w.cpu(), 1e-5)
+    got_n, got_r = ops.fused_add_rms_norm(x, res, w, 1e-5)
+    _close(got_r, want_r)
+    _close(got_n, want_n)
+
+
+@pytest.mark.parametrize("shape", [(9, 11008), (128, 28672)])
+def test_silu_and_mul(shape):
+    t, d2 = shape
+    x = torch.randn((t, d2 * 2), dtype=torch.bfloat16, device="cuda")
+    got = ops.silu_and_mul(x)
+    want = reference.silu_and_mul(x.cpu())
+    _close(got, want)
+
+
+@pytest.mark.parametrize("qh,kh,hd,rot", [(32, 8, 128, 128), (8, 1, 64, 64)])
+def test_rope(qh, kh, hd, rot):
+    T = 17
+    max_pos = 512
+    inv = 1.0 / (500000.0 ** (torch.arange(0, rot, 2).float() / rot))
+    t = torch.arange(max_pos).float()
+    freqs = torch.outer(t, inv)
+    cos_sin = torch.cat([freqs.cos(), freqs.sin()], dim=-1).contiguous().cuda()
+    pos = torch.randint(0, max_pos, (T,), dtype=torch.int32, device="cuda")
+    q = torch.randn((T, qh * hd), dtype=torch.bfloat16, device="cuda")
+    k = torch.randn((T, kh * hd), dtype=torch.bfloat16, device="cuda")
+    want_q, want_k = reference.rotary_embedding(
+        pos.cpu(), q.cpu(), k.cpu(), cos_sin.cpu(), hd
+    )
+    got_q, got_k = ops.rotary_embedding(pos, q, k, cos_sin, hd)
+    _close(got_q, want_q)
+    _close(got_k, want_k)
+
+
+def _make_cache(nb, kh, bs, hd, device="cuda"):
+    k_cache = torch.randn((nb, kh, bs, hd), dtype=torch.bfloat16, device=device)
+    v_cache = torch.randn((nb, kh, bs, hd), dtype=torch.bfloat16, device=device)
+    return k_cache, v_cache
+
+
+def test_reshape_and_cache():
+    nb, kh, bs, hd = 32, 8, 16, 128
+    T = 40
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    kc_ref, vc_ref = k_cache.cpu().clone(), v_cache.cpu().clone()
+    k = torch.randn((T, kh, hd), dtype=torch.bfloat16, device="cuda")
+    v = torch.randn((T, kh, hd), dtype=torch.bfloat16, device="cuda")
+    slots = torch.randperm(nb * bs)[:T].to(torch.long)
+    slots[5] = -1  # skipped token
+    reference.reshape_and_cache(k.cpu(), v.cpu(), kc_ref, vc_ref, slots)
+    ops.reshape_and_cache(k, v, k_cache, v_cache, slots.cuda())
+    _close(k_cache, kc_ref, atol=0, rtol=0)
+    _close(v_cache, vc_ref, atol=0, rtol=0)
+
+
+@pytest.mark.parametrize("qh,kh,hd", [(32, 8, 128), (8, 8, 128), (16, 2, 128), (8, 2, 64)])
+def test_paged_attn_decode(qh, kh, hd):
+    torch.manual_seed(1)
+    S, bs = 5, 16
+    seq_lens = torch.tensor([1, 17, 33, 256, 100], dtype=torch.int32)
+    max_blocks = 32
+    nb = S * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    block_tables = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+        S, max_blocks
+    )
+    q = torch.randn((S, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_decode(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), block_tables, seq_lens,
+        1.0 / hd ** 0.5,
+    )
+    got = ops.paged_attn_decode(
+        q, k_cache, v_cache, block_tables.cuda(), seq_lens.cuda(),
+        1.0 / hd ** 0.5,
+    )
+    _close(got, want)
+
+
+@pytest.mark.parametrize("qh,kh,hd", [(32, 8, 128), (4, 4, 64)])
+def test_paged_attn_prefill(qh, kh, hd):
+    torch.manual_seed(2)
+    bs = 16
+    # two sequences: one fresh prefill (pos 0..30), one chunked continuation
+    # (pos 64..95 with 64 tokens of prior context in cache)
+    seqs = [31, 32]
+    start = [0, 64]
+    max_blocks = 16
+    nb = 2 * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    block_tables = torch.arange(1, 2 * max_blocks + 1, dtype=torch.int32).reshape(
+        2, max_blocks
+    )
+    token_seq = torch.cat(
+        [torch.full((n,), i, dtype=torch.int32) for i, n in enumerate(seqs)]
+    )
+    token_pos = torch.cat(
+        [torch.arange(s, s + n, dtype=torch.int32) for s, n in zip(start, seqs)]
+    )
+    T = token_seq.shape[0]
+    q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_prefill(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), block_tables, token_seq,
+        token_pos, 1.0 / hd ** 0.5,
+    )
+    got = ops.paged_attn_prefill(
+        q, k_cache, v_cache, block_tables.cuda(), token_seq.cuda(),
+        token_pos.cuda(), 1.0 / hd ** 0.5,
+    )
+    _close(got, want)
+
+
+def test_greedy_sample():
+    R, V = 9, 128256
+    logits = torch.randn((R, V), dtype=torch.bfloat16, device="cuda")
+    got = ops.greedy_sample(logits)
+    want = reference.greedy_sample(logits.cpu())
+    assert torch.equal(got.cpu(), want)
