@@ -1,0 +1,152 @@
+"""Engine configuration.
+
+Mirrors the engine-flag surface the reference deploys through helm
+(`vllm serve` argv assembled at reference helm/templates/
+deployment-vllm-multi.yaml:127-221 — SURVEY.md section 2.9): model, dtype,
+max-model-len, max-num-seqs, gpu-memory-utilization, enable-prefix-caching,
+enable-chunked-prefill, kv-transfer role. Architecture definitions are local
+(random-init weights by default: no network in the build environment).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class ModelConfig:
+    name: str
+    hidden_size: int
+    num_layers: int
+    num_q_heads: int
+    num_kv_heads: int
+    head_dim: int
+    intermediate_size: int
+    vocab_size: int
+    rope_theta: float = 500000.0
+    max_position: int = 8192
+    rms_norm_eps: float = 1e-5
+    bos_token_id: int = 1
+    eos_token_id: int = 2
+
+    @property
+    def q_size(self) -> int:
+        return self.num_q_heads * self.head_dim
+
+    @property
+    def kv_size(self) -> int:
+        return self.num_kv_heads * self.head_dim
+
+
+# Named architectures (shapes follow the public HF configs for the Llama
+# family; weights are random-init — see BASELINE.md: synthetic data /
+# random-init weights).
+ARCHITECTURES = {
+    "llama-3-8b": ModelConfig(
+        name="llama-3-8b",
+        hidden_size=4096,
+        num_layers=32,
+        num_q_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=128256,
+        rope_theta=500000.0,
+        max_position=8192,
+    ),
+    "llama-3-70b": ModelConfig(
+        name="llama-3-70b",
+        hidden_size=8192,
+        num_layers=80,
+        num_q_heads=64,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=28672,
+        vocab_size=128256,
+        rope_theta=500000.0,
+        max_position=8192,
+    ),
+    # Small config for tests (CPU-runnable).
+    "tiny-llama": ModelConfig(
+        name="tiny-llama",
+        hidden_size=128,
+        num_layers=2,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        intermediate_size=256,
+        vocab_size=1024,
+        rope_theta=10000.0,
+        max_position=2048,
+    ),
+    # GPU-runnable small config with kernel-supported head_dim.
+    "mini-llama": ModelConfig(
+        name="mini-llama",
+        hidden_size=512,
+        num_layers=4,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        intermediate_size=1024,
+        vocab_size=2048,
+        rope_theta=10000.0,
+        max_position=4096,
+    ),
+}
+
+
+@dataclass
+class CacheConfig:
+    block_size: int = 16
+    # Either an explicit block count (tests) or a fraction of free GPU memory.
+    num_gpu_blocks: Optional[int] = None
+    gpu_memory_utilization: float = 0.85
+    enable_prefix_caching: bool = True
+    # Host-DRAM KV offload pool size in GiB (0 disables). LMCache-equivalent
+    # surface: reference LMCACHE_MAX_LOCAL_CPU_SIZE
+    # (deployment-vllm-multi.yaml:336-343).
+    cpu_offload_gb: float = 0.0
+
+
+@dataclass
+class SchedulerConfig:
+    max_num_seqs: int = 256
+    max_num_batched_tokens: int = 8192
+    enable_chunked_prefill: bool = True
+    # Cap on how many prompt tokens a single prefill chunk may carry.
+    max_prefill_chunk: int = 8192
+
+
+@dataclass
+class ParallelConfig:
+    tensor_parallel_size: int = 1
+    rank: int = 0
+    # disaggregated prefill role: None | "prefill" | "decode"
+    kv_role: Optional[str] = None
+
+
+@dataclass
+class EngineConfig:
+    model: str = "llama-3-8b"
+    dtype: str = "bfloat16"
+    max_model_len: int = 4096
+    seed: int = 0
+    weights_path: Optional[str] = None  # safetensors dir (optional)
+    tokenizer: str = "synthetic"  # or a path to a tokenizer.json dir
+    enforce_eager: bool = False  # True disables hipGraph decode capture
+    cache: CacheConfig = field(default_factory=CacheConfig)
+    scheduler: SchedulerConfig = field(default_factory=SchedulerConfig)
+    parallel: ParallelConfig = field(default_factory=ParallelConfig)
+
+    def model_config(self) -> ModelConfig:
+        if self.model not in ARCHITECTURES:
+            raise ValueError(
+                f"unknown model architecture {self.model!r}; "
+                f"known: {sorted(ARCHITECTURES)}"
+            )
+        cfg = ARCHITECTURES[self.model]
+        if self.max_model_len > cfg.max_position:
+            cfg = dataclasses.replace(cfg, max_position=self.max_model_len)
+        return cfg

@@ -1,0 +1,193 @@
+"""LLMEngine: the synchronous core loop (add_request / step / abort).
+
+This is the in-process API used by bench.py and wrapped by the async OpenAI
+HTTP server (engine/server.py). One engine instance == one replica == one
+GPU (TP>1: one engine per rank, rank 0 drives scheduling — see
+parallel/state.py).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional, Union
+
+import torch
+
+from production_stack_amd.engine.block_manager import BlockManager
+from production_stack_amd.engine.config import EngineConfig
+from production_stack_amd.engine.model_runner import ModelRunner
+from production_stack_amd.engine.sampling import SamplingParams
+from production_stack_amd.engine.scheduler import Scheduler
+from production_stack_amd.engine.sequence import (
+    RequestOutput,
+    Sequence,
+    SeqStatus,
+)
+from production_stack_amd.engine.tokenizer import get_tokenizer
+from production_stack_amd.parallel import state as pstate
+
+
+class EngineStats:
+    def __init__(self) -> None:
+        self.prompt_tokens = 0
+        self.generation_tokens = 0
+        self.num_requests = 0
+        self.num_finished = 0
+
+
+class LLMEngine:
+    def __init__(
+        self, config: EngineConfig, device: Optional[str] = None
+    ) -> None:
+        self.config = config
+        self.model_cfg = config.model_config()
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        pstate.init_distributed(config.parallel.tensor_parallel_size)
+        torch.manual_seed(config.seed)
+        self.tokenizer = get_tokenizer(
+            config.tokenizer, self.model_cfg.vocab_size
+        )
+        self.runner = ModelRunner(config, self.model_cfg, self.device)
+        num_blocks = self.runner.profile_num_blocks()
+        self.runner.allocate_kv_cache(num_blocks)
+        self.block_manager = BlockManager(
+            num_blocks,
+            config.cache.block_size,
+            config.cache.enable_prefix_caching,
+        )
+        self.scheduler = Scheduler(
+            config.scheduler, self.block_manager, config.max_model_len
+        )
+        self.stats = EngineStats()
+        self._sleeping = False
+
+    # ------------------------------------------------------------------
+    def add_request(
+        self,
+        request_id: str,
+        prompt: Union[str, List[int]],
+        params: SamplingParams,
+        arrival_time: Optional[float] = None,
+    ) -> None:
+        params.validate(self.config.max_model_len)
+        if isinstance(prompt, str):
+            token_ids = self.tokenizer.encode(prompt)
+        else:
+            token_ids = list(prompt)
+        if not token_ids:
+            token_ids = [self.model_cfg.bos_token_id]
+        seq = Sequence(request_id, token_ids, params, arrival_time)
+        self.scheduler.add(seq)
+        self.stats.num_requests += 1
+        self.stats.prompt_tokens += len(token_ids)
+
+    def abort_request(self, request_id: str) -> None:
+        self.scheduler.abort(request_id)
+
+    def has_unfinished(self) -> bool:
+        return self.scheduler.has_unfinished()
+
+    # ------------------------------------------------------------------
+    def step(self) -> List[RequestOutput]:
+        out = self.scheduler.schedule()
+        if out.is_empty:
+            return []
+        sampled = self.runner.execute(out, self.block_manager)
+        finished = self.scheduler.on_step_done(
+            out, sampled, self.model_cfg.eos_token_id
+        )
+        now = time.time()
+        results: List[RequestOutput] = []
+        for seq in out.capacity_stopped:
+            results.append(
+                RequestOutput(
+                    request_id=seq.request_id,
+                    new_token_ids=[],
+                    text_delta="",
+                    finished=True,
+                    finish_reason="length",
+                    num_prompt_tokens=seq.num_prompt,
+                    num_output_tokens=len(seq.output_token_ids),
+                    num_cached_tokens=seq.num_cached_prompt_tokens,
+                )
+            )
+            self.stats.num_finished += 1
+        for ss in out.scheduled:
+            seq = ss.seq
+            if seq.status is SeqStatus.PREEMPTED:
+                continue
+            new = seq.drain_new_tokens()
+            if not new and not seq.finished:
+                continue
+            first = seq.first_token_time is None and bool(new)
+            if first:
+                seq.first_token_time = now
+            self.stats.generation_tokens += len(new)
+            text_delta = "".join(
+                self.tokenizer.decode_token(t) + " " for t in new
+            ) if new else ""
+            reason = None
+            if seq.status is SeqStatus.FINISHED_STOPPED:
+                reason = "stop"
+            elif seq.status is SeqStatus.FINISHED_LENGTH:
+                reason = "length"
+            elif seq.status is SeqStatus.FINISHED_ABORTED:
+                reason = "abort"
+            results.append(
+                RequestOutput(
+                    request_id=seq.request_id,
+                    new_token_ids=new,
+                    text_delta=text_delta,
+                    finished=seq.finished,
+                    finish_reason=reason,
+                    num_prompt_tokens=seq.num_prompt,
+                    num_output_tokens=len(seq.output_token_ids),
+                    num_cached_tokens=seq.num_cached_prompt_tokens,
+                    first_token=first,
+                )
+            )
+        self.stats.num_finished += len(finished)
+        return results
+
+    # ------------------------------------------------------------------
+    def generate(
+        self,
+        prompts: List[Union[str, List[int]]],
+        params: Union[SamplingParams, List[SamplingParams]],
+    ) -> Dict[str, List[int]]:
+        """Offline batch API: run all prompts to completion."""
+        if isinstance(params, SamplingParams):
+            params = [params] * len(prompts)
+        for i, (p, sp) in enumerate(zip(prompts, params)):
+            self.add_request(f"offline-{i}", p, sp)
+        outputs: Dict[str, List[int]] = {}
+        while self.has_unfinished():
+            for r in self.step():
+                outputs.setdefault(r.request_id, []).extend(r.new_token_ids)
+        return outputs
+
+    # ---- observability (scraped into vllm:* metric names) -------------
+    def engine_metrics(self) -> Dict[str, float]:
+        bm = self.block_manager
+        return {
+            "num_requests_running": float(self.scheduler.num_running),
+            "num_requests_waiting": float(self.scheduler.num_waiting),
+            "gpu_cache_usage_perc": bm.usage,
+            "gpu_prefix_cache_hits_total": float(bm.prefix_hits),
+            "gpu_prefix_cache_queries_total": float(bm.prefix_queries),
+            "prompt_tokens_total": float(self.stats.prompt_tokens),
+            "generation_tokens_total": float(self.stats.generation_tokens),
+        }
+
+    # ---- sleep / wake (reference request.py:1041-1128 parity) ----------
+    def sleep(self, level: int = 1) -> None:
+        self._sleeping = True
+
+    def wake_up(self) -> None:
+        self._sleeping = False
+
+    @property
+    def is_sleeping(self) -> bool:
+        return self._sleeping

@@ -1,0 +1,236 @@
+"""Model runner: batch preparation, forward execution, sampling.
+
+Batch layout per step: [all prefill-chunk tokens..., all decode tokens...].
+Single-token chunks run through the decode paged-attention kernel (its
+attention pattern — last token over full context — is exactly a 1-token
+chunk), multi-token chunks through the chunked-prefill kernel.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from production_stack_amd import ops
+from production_stack_amd.engine.block_manager import BlockManager
+from production_stack_amd.engine.config import EngineConfig, ModelConfig
+from production_stack_amd.engine.models.llama import BatchMeta, LlamaForCausalLM
+from production_stack_amd.engine.scheduler import SchedulerOutput
+from production_stack_amd.engine.sequence import Sequence
+
+
+class ModelRunner:
+    def __init__(
+        self,
+        config: EngineConfig,
+        model_cfg: ModelConfig,
+        device: torch.device,
+    ) -> None:
+        self.config = config
+        self.model_cfg = model_cfg
+        self.device = device
+        tp = config.parallel.tensor_parallel_size
+        self.model = LlamaForCausalLM(model_cfg, tp=tp).to(device)
+        if config.weights_path:
+            from production_stack_amd.engine.weights import load_safetensors
+
+            load_safetensors(self.model, config.weights_path)
+        else:
+            self.model.random_init(config.seed)
+        self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
+        self.num_blocks = 0
+        self._generator = torch.Generator(device="cpu").manual_seed(
+            config.seed + 12345
+        )
+
+    # ------------------------------------------------------------------
+    def profile_num_blocks(self) -> int:
+        """Size the KV cache from free HBM after weights (288 GB per GPU)."""
+        cache_cfg = self.config.cache
+        if cache_cfg.num_gpu_blocks is not None:
+            return cache_cfg.num_gpu_blocks
+        if self.device.type != "cuda":
+            return 512
+        free, total = torch.cuda.mem_get_info(self.device)
+        budget = int(
+            total * cache_cfg.gpu_memory_utilization
+            - (total - free)
+        )
+        per_block = self.model.kv_bytes_per_block(cache_cfg.block_size)
+        n = max(budget // per_block, 16)
+        return int(n)
+
+    def allocate_kv_cache(self, num_blocks: int) -> None:
+        self.num_blocks = num_blocks
+        cfg = self.model_cfg
+        bs = self.config.cache.block_size
+        kh = self.model.kv_heads
+        self.kv_caches = []
+        for _ in range(cfg.num_layers):
+            k = torch.zeros(
+                (num_blocks, kh, bs, cfg.head_dim),
+                dtype=torch.bfloat16,
+                device=self.device,
+            )
+            v = torch.zeros_like(k)
+            self.kv_caches.append((k, v))
+
+    # ------------------------------------------------------------------
+    def prepare(
+        self, out: SchedulerOutput, bm: BlockManager
+    ) -> Tuple[torch.Tensor, BatchMeta, List[Sequence], torch.Tensor]:
+        """Returns (token_ids, meta, seqs_to_sample, sample_row_indices)."""
+        bs = bm.block_size
+        prefills = [s for s in out.scheduled if not s.is_decode]
+        decodes = [s for s in out.scheduled if s.is_decode]
+
+        tokens: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        p_token_seq: List[int] = []
+        p_token_pos: List[int] = []
+        p_tables: List[List[int]] = []
+        sample_rows: List[int] = []
+        sample_seqs: List[Sequence] = []
+
+        for row, ss in enumerate(prefills):
+            seq = ss.seq
+            all_ids = seq.token_ids()
+            start = seq.num_computed
+            end = start + ss.num_tokens
+            p_tables.append(seq.block_table)
+            for pos in range(start, end):
+                tokens.append(all_ids[pos])
+                positions.append(pos)
+                slots.append(seq.block_table[pos // bs] * bs + pos % bs)
+                p_token_seq.append(row)
+                p_token_pos.append(pos)
+            if end == seq.num_tokens:
+                sample_rows.append(len(tokens) - 1)
+                sample_seqs.append(seq)
+
+        d_seq_lens: List[int] = []
+        d_tables: List[List[int]] = []
+        for ss in decodes:
+            seq = ss.seq
+            all_ids = seq.token_ids()
+            pos = seq.num_computed
+            tokens.append(all_ids[pos])
+            positions.append(pos)
+            slots.append(seq.block_table[pos // bs] * bs + pos % bs)
+            d_seq_lens.append(pos + 1)
+            d_tables.append(seq.block_table)
+            sample_rows.append(len(tokens) - 1)
+            sample_seqs.append(seq)
+
+        dev = self.device
+        max_bt = max((len(t) for t in p_tables + d_tables), default=1)
+
+        def pad_tables(tabs: List[List[int]]) -> Optional[torch.Tensor]:
+            if not tabs:
+                return None
+            out_t = torch.zeros((len(tabs), max_bt), dtype=torch.int32)
+            for i, t in enumerate(tabs):
+                if t:
+                    out_t[i, : len(t)] = torch.tensor(t, dtype=torch.int32)
+            return out_t.to(dev, non_blocking=True)
+
+        num_prefill_tokens = len(p_token_seq)
+        meta = BatchMeta(
+            positions=torch.tensor(positions, dtype=torch.int32).to(
+                dev, non_blocking=True
+            ),
+            slot_mapping=torch.tensor(slots, dtype=torch.long).to(
+                dev, non_blocking=True
+            ),
+            num_prefill_tokens=num_prefill_tokens,
+            prefill_token_seq=(
+                torch.tensor(p_token_seq, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                if p_token_seq
+                else None
+            ),
+            prefill_token_pos=(
+                torch.tensor(p_token_pos, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                if p_token_pos
+                else None
+            ),
+            prefill_block_tables=pad_tables(p_tables),
+            num_decode_seqs=len(decodes),
+            decode_seq_lens=(
+                torch.tensor(d_seq_lens, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                if d_seq_lens
+                else None
+            ),
+            decode_block_tables=pad_tables(d_tables),
+        )
+        token_t = torch.tensor(tokens, dtype=torch.long).to(
+            dev, non_blocking=True
+        )
+        rows_t = torch.tensor(sample_rows, dtype=torch.long).to(
+            dev, non_blocking=True
+        )
+        return token_t, meta, sample_seqs, rows_t
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def execute(
+        self, out: SchedulerOutput, bm: BlockManager
+    ) -> Dict[str, int]:
+        """Run one step; returns request_id -> sampled token."""
+        token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
+        if token_t.numel() == 0:
+            return {}
+        hidden = self.model(token_t, meta, self.kv_caches)
+        if not sample_seqs:
+            return {}
+        sel = hidden[rows_t]
+        logits = self.model.compute_logits(sel)
+        tokens = self.sample(logits, sample_seqs)
+        return {
+            seq.request_id: int(tok)
+            for seq, tok in zip(sample_seqs, tokens)
+        }
+
+    def sample(
+        self, logits: torch.Tensor, seqs: List[Sequence]
+    ) -> torch.Tensor:
+        greedy_mask = [s.params.greedy for s in seqs]
+        result = torch.empty(len(seqs), dtype=torch.long)
+        if all(greedy_mask):
+            return ops.greedy_sample(logits).cpu()
+        lf = logits.float()
+        for i, seq in enumerate(seqs):
+            p = seq.params
+            row = lf[i]
+            if p.greedy:
+                result[i] = int(row.argmax())
+                continue
+            row = row / max(p.temperature, 1e-5)
+            if p.top_k > 0 and p.top_k < row.shape[-1]:
+                kth = torch.topk(row, p.top_k).values[-1]
+                row = row.masked_fill(row < kth, float("-inf"))
+            probs = torch.softmax(row, dim=-1)
+            if p.top_p < 1.0:
+                sp, si = torch.sort(probs, descending=True)
+                cum = torch.cumsum(sp, dim=-1)
+                keep = cum - sp < p.top_p
+                keep[0] = True
+                sp = sp * keep
+                sp = sp / sp.sum()
+                idx = torch.multinomial(
+                    sp.cpu(), 1, generator=self._generator
+                )
+                result[i] = int(si[idx])
+            else:
+                idx = torch.multinomial(
+                    probs.cpu(), 1, generator=self._generator
+                )
+                result[i] = int(idx)
+        return result

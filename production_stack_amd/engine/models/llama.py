@@ -1,0 +1,242 @@
+"""Llama-family causal LM built on the MI355X op set.
+
+Compute mapping (MI355X-first):
+  * plain GEMMs (qkv / o / gate-up / down / lm_head) -> torch F.linear, which
+    lowers to hipBLASLt on ROCm;
+  * everything else on the hot path is a hand-written HIP kernel via
+    production_stack_amd.ops: fused residual-add RMSNorm, fused rope over
+    q+k, paged-KV append, paged attention (prefill + decode), SiLU-mul.
+  * weights are plain bf16 Parameters; qkv and gate/up are pre-fused so each
+    layer issues exactly 4 GEMMs.
+
+Supports tensor parallelism: q/kv heads and the MLP intermediate dim are
+sharded across the TP group; o_proj and down_proj outputs are all-reduced
+(RCCL over xGMI on the GPU box, gloo in CPU tests).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from production_stack_amd import ops
+from production_stack_amd.engine.config import ModelConfig
+from production_stack_amd.parallel import state as pstate
+
+
+class BatchMeta:
+    """Per-step metadata consumed by the attention layers."""
+
+    def __init__(
+        self,
+        positions: torch.Tensor,  # [T] int32
+        slot_mapping: torch.Tensor,  # [T] int64
+        num_prefill_tokens: int,
+        prefill_token_seq: Optional[torch.Tensor],  # [Tp] int32
+        prefill_token_pos: Optional[torch.Tensor],  # [Tp] int32
+        prefill_block_tables: Optional[torch.Tensor],  # [Sp, MB] int32
+        num_decode_seqs: int,
+        decode_seq_lens: Optional[torch.Tensor],  # [Sd] int32
+        decode_block_tables: Optional[torch.Tensor],  # [Sd, MB] int32
+    ) -> None:
+        self.positions = positions
+        self.slot_mapping = slot_mapping
+        self.num_prefill_tokens = num_prefill_tokens
+        self.prefill_token_seq = prefill_token_seq
+        self.prefill_token_pos = prefill_token_pos
+        self.prefill_block_tables = prefill_block_tables
+        self.num_decode_seqs = num_decode_seqs
+        self.decode_seq_lens = decode_seq_lens
+        self.decode_block_tables = decode_block_tables
+
+
+def build_cos_sin_cache(
+    head_dim: int, max_position: int, theta: float
+) -> torch.Tensor:
+    """[max_position, head_dim] fp32: cos[half] || sin[half] (host-side)."""
+    half = head_dim // 2
+    inv = 1.0 / (
+        theta ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
+    )
+    t = torch.arange(max_position, dtype=torch.float64)
+    freqs = torch.outer(t, inv)
+    return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().contiguous()
+
+
+class LlamaLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, tp: int) -> None:
+        super().__init__()
+        self.cfg = cfg
+        assert cfg.num_q_heads % tp == 0, "q heads must divide TP"
+        assert cfg.num_kv_heads % tp == 0 or tp % cfg.num_kv_heads == 0
+        self.q_heads = cfg.num_q_heads // tp
+        self.kv_heads = max(cfg.num_kv_heads // tp, 1)
+        self.head_dim = cfg.head_dim
+        self.inter = cfg.intermediate_size // tp
+        h = cfg.hidden_size
+        qs = self.q_heads * self.head_dim
+        kvs = self.kv_heads * self.head_dim
+        self.qkv_proj = nn.Parameter(
+            torch.empty(qs + 2 * kvs, h, dtype=torch.bfloat16)
+        )
+        self.o_proj = nn.Parameter(torch.empty(h, qs, dtype=torch.bfloat16))
+        self.gate_up_proj = nn.Parameter(
+            torch.empty(2 * self.inter, h, dtype=torch.bfloat16)
+        )
+        self.down_proj = nn.Parameter(
+            torch.empty(h, self.inter, dtype=torch.bfloat16)
+        )
+        self.input_norm = nn.Parameter(torch.empty(h, dtype=torch.bfloat16))
+        self.post_attn_norm = nn.Parameter(
+            torch.empty(h, dtype=torch.bfloat16)
+        )
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+
+    def forward(
+        self,
+        hidden: torch.Tensor,  # [T, H]
+        residual: Optional[torch.Tensor],
+        meta: BatchMeta,
+        kv_cache: tuple,  # (k_cache, v_cache) [NB, KH, BS, HD]
+        cos_sin: torch.Tensor,
+    ) -> tuple:
+        cfg = self.cfg
+        if residual is None:
+            residual = hidden
+            hidden = ops.rms_norm(hidden, self.input_norm, cfg.rms_norm_eps)
+        else:
+            hidden, residual = ops.fused_add_rms_norm(
+                hidden, residual, self.input_norm, cfg.rms_norm_eps
+            )
+        qkv = F.linear(hidden, self.qkv_proj)
+        qs = self.q_heads * self.head_dim
+        kvs = self.kv_heads * self.head_dim
+        q = qkv[:, :qs].contiguous()
+        k = qkv[:, qs : qs + kvs].contiguous()
+        v = qkv[:, qs + kvs :].contiguous()
+        q, k = ops.rotary_embedding(meta.positions, q, k, cos_sin, self.head_dim)
+        T = q.shape[0]
+        k_cache, v_cache = kv_cache
+        ops.reshape_and_cache(
+            k.view(T, self.kv_heads, self.head_dim),
+            v.view(T, self.kv_heads, self.head_dim),
+            k_cache,
+            v_cache,
+            meta.slot_mapping,
+        )
+        qh = q.view(T, self.q_heads, self.head_dim)
+        outs: List[torch.Tensor] = []
+        tp = meta.num_prefill_tokens
+        if tp > 0:
+            outs.append(
+                ops.paged_attn_prefill(
+                    qh[:tp].contiguous(),
+                    k_cache,
+                    v_cache,
+                    meta.prefill_block_tables,
+                    meta.prefill_token_seq,
+                    meta.prefill_token_pos,
+                    self.scale,
+                )
+            )
+        if meta.num_decode_seqs > 0:
+            outs.append(
+                ops.paged_attn_decode(
+                    qh[tp:].contiguous(),
+                    k_cache,
+                    v_cache,
+                    meta.decode_block_tables,
+                    meta.decode_seq_lens,
+                    self.scale,
+                )
+            )
+        attn = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
+        attn_out = F.linear(attn.view(T, qs), self.o_proj)
+        attn_out = pstate.tp_all_reduce(attn_out)
+
+        hidden, residual = ops.fused_add_rms_norm(
+            attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
+        )
+        gate_up = F.linear(hidden, self.gate_up_proj)
+        act = ops.silu_and_mul(gate_up)
+        mlp_out = F.linear(act, self.down_proj)
+        mlp_out = pstate.tp_all_reduce(mlp_out)
+        return mlp_out, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: ModelConfig, tp: int = 1) -> None:
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp
+        self.embed = nn.Parameter(
+            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16)
+        )
+        self.layers = nn.ModuleList(
+            LlamaLayer(cfg, tp) for _ in range(cfg.num_layers)
+        )
+        self.final_norm = nn.Parameter(
+            torch.empty(cfg.hidden_size, dtype=torch.bfloat16)
+        )
+        self.lm_head = nn.Parameter(
+            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16)
+        )
+        self.register_buffer(
+            "cos_sin",
+            build_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_theta),
+            persistent=False,
+        )
+
+    @torch.no_grad()
+    def random_init(self, seed: int = 0) -> None:
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        for name, p in self.named_parameters():
+            if "norm" in name:
+                p.fill_(1.0)
+            else:
+                # init on CPU in chunks to bound host memory, then copy
+                t = torch.empty(
+                    p.shape, dtype=torch.float16, device="cpu"
+                ).normal_(0.0, 0.02, generator=gen)
+                p.copy_(t.to(p.dtype))
+
+    @property
+    def kv_heads(self) -> int:
+        return max(self.cfg.num_kv_heads // self.tp, 1)
+
+    def kv_bytes_per_block(self, block_size: int) -> int:
+        return (
+            2  # k and v
+            * self.kv_heads
+            * block_size
+            * self.cfg.head_dim
+            * 2  # bf16
+            * self.cfg.num_layers
+        )
+
+    @torch.no_grad()
+    def forward(
+        self,
+        token_ids: torch.Tensor,  # [T] long
+        meta: BatchMeta,
+        kv_caches: List[tuple],
+    ) -> torch.Tensor:
+        hidden = F.embedding(token_ids, self.embed)
+        residual = None
+        for layer, cache in zip(self.layers, kv_caches):
+            hidden, residual = layer(
+                hidden, residual, meta, cache, self.cos_sin
+            )
+        # final residual add + norm
+        hidden, _ = ops.fused_add_rms_norm(
+            hidden, residual, self.final_norm, self.cfg.rms_norm_eps
+        )
+        return hidden
+
+    @torch.no_grad()
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        return F.linear(hidden, self.lm_head)

@@ -1,0 +1,33 @@
+"""Sampling parameters (OpenAI-compatible subset the router proxies)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+
+@dataclass
+class SamplingParams:
+    max_tokens: int = 16
+    temperature: float = 1.0
+    top_p: float = 1.0
+    top_k: int = -1
+    stop_token_ids: List[int] = field(default_factory=list)
+    stop: List[str] = field(default_factory=list)
+    ignore_eos: bool = False
+    seed: Optional[int] = None
+    logprobs: Optional[int] = None
+
+    @property
+    def greedy(self) -> bool:
+        return self.temperature == 0.0
+
+    def validate(self, max_model_len: int) -> None:
+        if self.max_tokens < 1:
+            raise ValueError("max_tokens must be >= 1")
+        if self.temperature < 0:
+            raise ValueError("temperature must be >= 0")
+        if not (0 < self.top_p <= 1.0):
+            raise ValueError("top_p must be in (0, 1]")
+        if self.top_k == 0 or self.top_k < -1:
+            raise ValueError("top_k must be -1 (off) or >= 1")

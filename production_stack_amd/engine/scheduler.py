@@ -1,0 +1,220 @@
+"""Continuous-batching scheduler with chunked prefill and preemption.
+
+Capability target: the engine behaviour the reference stack configures via
+--max-num-seqs / --enable-chunked-prefill / --enable-prefix-caching
+(reference helm/templates/deployment-vllm-multi.yaml:127-221) and observes
+through vllm:num_requests_running / vllm:num_requests_waiting.
+
+Unified token model: a sequence needs KV for all `num_tokens` tokens; each
+step schedules a chunk of up to `budget` uncomputed tokens; the step whose
+chunk reaches the end samples the next token. Decode is the 1-token case.
+"""
+
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Deque, Dict, List, Optional
+
+from production_stack_amd.engine.block_manager import BlockManager
+from production_stack_amd.engine.config import SchedulerConfig
+from production_stack_amd.engine.sequence import Sequence, SeqStatus
+
+
+@dataclass
+class ScheduledSeq:
+    seq: Sequence
+    num_tokens: int  # chunk size this step
+
+    @property
+    def is_decode(self) -> bool:
+        return self.num_tokens == 1
+
+
+@dataclass
+class SchedulerOutput:
+    scheduled: List[ScheduledSeq] = field(default_factory=list)
+    preempted: List[Sequence] = field(default_factory=list)
+    # sequences force-finished because they can no longer fit in the cache
+    capacity_stopped: List[Sequence] = field(default_factory=list)
+
+    @property
+    def num_tokens(self) -> int:
+        return sum(s.num_tokens for s in self.scheduled)
+
+    @property
+    def is_empty(self) -> bool:
+        return not self.scheduled
+
+
+class Scheduler:
+    def __init__(
+        self,
+        config: SchedulerConfig,
+        block_manager: BlockManager,
+        max_model_len: int,
+    ) -> None:
+        self.config = config
+        self.bm = block_manager
+        self.max_model_len = max_model_len
+        self.waiting: Deque[Sequence] = deque()
+        self.running: List[Sequence] = []
+        self._by_id: Dict[str, Sequence] = {}
+
+    # ------------------------------------------------------------------
+    def add(self, seq: Sequence) -> None:
+        if seq.num_prompt + 1 > self.max_model_len:
+            raise ValueError(
+                f"prompt of {seq.num_prompt} tokens exceeds max_model_len "
+                f"{self.max_model_len}"
+            )
+        self._by_id[seq.request_id] = seq
+        self.waiting.append(seq)
+
+    def abort(self, request_id: str) -> Optional[Sequence]:
+        seq = self._by_id.pop(request_id, None)
+        if seq is None or seq.finished:
+            return None
+        seq.status = SeqStatus.FINISHED_ABORTED
+        if seq in self.running:
+            self.running.remove(seq)
+            self.bm.free_seq(seq)
+        else:
+            try:
+                self.waiting.remove(seq)
+            except ValueError:
+                pass
+        return seq
+
+    @property
+    def num_running(self) -> int:
+        return len(self.running)
+
+    @property
+    def num_waiting(self) -> int:
+        return len(self.waiting)
+
+    def has_unfinished(self) -> bool:
+        return bool(self.running or self.waiting)
+
+    # ------------------------------------------------------------------
+    def _preempt_last(self, out: SchedulerOutput, keep: Sequence) -> bool:
+        """Preempt the most recently admitted running seq (not `keep`)."""
+        for victim in reversed(self.running):
+            if victim is keep:
+                continue
+            self.running.remove(victim)
+            self.bm.free_seq(victim)
+            victim.reset_for_recompute()
+            self.waiting.appendleft(victim)
+            out.preempted.append(victim)
+            # it may already carry a chunk in this step's schedule: drop it
+            out.scheduled = [s for s in out.scheduled if s.seq is not victim]
+            return True
+        return False
+
+    def schedule(self) -> SchedulerOutput:
+        out = SchedulerOutput()
+        budget = self.config.max_num_batched_tokens
+        scheduled_set = set()
+
+        # 1. running sequences, oldest first: decodes (remaining == 1) and
+        #    in-flight chunked prefills.
+        for seq in list(self.running):
+            if budget <= 0:
+                break
+            if seq not in self.running:
+                # preempted as a victim earlier in this same call
+                continue
+            remaining = seq.num_tokens - seq.num_computed
+            if remaining <= 0:
+                continue
+            chunk = min(remaining, budget, self.config.max_prefill_chunk)
+            target = seq.num_computed + chunk
+            while not self.bm.ensure_capacity(seq, target):
+                if not self._preempt_last(out, keep=seq):
+                    # no other sequence to evict: this one alone exceeds the
+                    # whole KV cache — finish it instead of livelocking on
+                    # preempt-recompute.
+                    self.running.remove(seq)
+                    self.bm.free_seq(seq)
+                    seq.status = SeqStatus.FINISHED_LENGTH
+                    self._by_id.pop(seq.request_id, None)
+                    out.capacity_stopped.append(seq)
+                    out.scheduled = [
+                        s for s in out.scheduled if s.seq is not seq
+                    ]
+                    chunk = 0
+                    break
+                if seq in out.preempted:
+                    chunk = 0
+                    break
+            if chunk > 0:
+                out.scheduled.append(ScheduledSeq(seq, chunk))
+                scheduled_set.add(seq.request_id)
+                budget -= chunk
+
+        # 2. admit waiting sequences.
+        while (
+            self.waiting
+            and budget > 0
+            and len(self.running) < self.config.max_num_seqs
+        ):
+            seq = self.waiting[0]
+            if not self.bm.can_allocate_prompt(seq):
+                break
+            remaining_est = seq.num_tokens  # before cache hits
+            if (
+                not self.config.enable_chunked_prefill
+                and remaining_est > budget
+            ):
+                break
+            self.waiting.popleft()
+            self.bm.allocate_prompt(seq)  # sets num_computed via cache hits
+            seq.status = SeqStatus.RUNNING
+            self.running.append(seq)
+            remaining = seq.num_tokens - seq.num_computed
+            chunk = min(remaining, budget, self.config.max_prefill_chunk)
+            out.scheduled.append(ScheduledSeq(seq, chunk))
+            scheduled_set.add(seq.request_id)
+            budget -= chunk
+
+        return out
+
+    # ------------------------------------------------------------------
+    def on_step_done(
+        self,
+        output: SchedulerOutput,
+        sampled: Dict[str, int],
+        eos_token_id: int,
+    ) -> List[Sequence]:
+        """Advance state after the model ran. Returns newly finished seqs."""
+        finished: List[Sequence] = []
+        for ss in output.scheduled:
+            seq = ss.seq
+            if seq.finished or seq.status is SeqStatus.PREEMPTED:
+                continue
+            seq.num_computed += ss.num_tokens
+            tok = sampled.get(seq.request_id)
+            if tok is not None:
+                seq.append_token(tok)
+                seq.num_computed = min(seq.num_computed, seq.num_tokens - 1)
+                p = seq.params
+                if (
+                    (tok == eos_token_id and not p.ignore_eos)
+                    or tok in p.stop_token_ids
+                ):
+                    seq.status = SeqStatus.FINISHED_STOPPED
+                elif len(seq.output_token_ids) >= p.max_tokens:
+                    seq.status = SeqStatus.FINISHED_LENGTH
+                elif seq.num_tokens >= self.max_model_len:
+                    seq.status = SeqStatus.FINISHED_LENGTH
+            if seq.finished:
+                self.running.remove(seq)
+                self.bm.register_computed_blocks(seq)
+                self.bm.free_seq(seq)
+                self._by_id.pop(seq.request_id, None)
+                finished.append(seq)
+            else:
+                self.bm.register_computed_blocks(seq)
+        return finished

@@ -1,0 +1,103 @@
+"""Sequence state for the continuous-batching scheduler."""
+
+from __future__ import annotations
+
+import enum
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+from production_stack_amd.engine.sampling import SamplingParams
+
+
+class SeqStatus(enum.Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    PREEMPTED = "preempted"
+    FINISHED_STOPPED = "stop"
+    FINISHED_LENGTH = "length"
+    FINISHED_ABORTED = "abort"
+
+
+FINISHED = {
+    SeqStatus.FINISHED_STOPPED,
+    SeqStatus.FINISHED_LENGTH,
+    SeqStatus.FINISHED_ABORTED,
+}
+
+
+class Sequence:
+    """One request's generation state.
+
+    token_ids = prompt + generated. num_computed counts tokens whose KV is in
+    the cache AND whose forward pass ran (prefix-cache hits count).
+    """
+
+    def __init__(
+        self,
+        request_id: str,
+        prompt_token_ids: List[int],
+        params: SamplingParams,
+        arrival_time: Optional[float] = None,
+    ) -> None:
+        self.request_id = request_id
+        self.prompt_token_ids = list(prompt_token_ids)
+        self.output_token_ids: List[int] = []
+        self.params = params
+        self.status = SeqStatus.WAITING
+        self.arrival_time = arrival_time or time.time()
+        self.first_token_time: Optional[float] = None
+        self.num_computed = 0  # tokens with KV in cache and attended
+        self.block_table: List[int] = []
+        self.num_cached_prompt_tokens = 0  # prefix-cache hits at admission
+        # streaming cursor: outputs not yet handed to the consumer
+        self._stream_cursor = 0
+
+    # ---- token accounting -------------------------------------------------
+    @property
+    def num_prompt(self) -> int:
+        return len(self.prompt_token_ids)
+
+    @property
+    def num_tokens(self) -> int:
+        return self.num_prompt + len(self.output_token_ids)
+
+    def token_ids(self) -> List[int]:
+        return self.prompt_token_ids + self.output_token_ids
+
+    @property
+    def finished(self) -> bool:
+        return self.status in FINISHED
+
+    @property
+    def in_prefill(self) -> bool:
+        return self.num_computed < self.num_prompt
+
+    def append_token(self, token_id: int) -> None:
+        self.output_token_ids.append(token_id)
+
+    def drain_new_tokens(self) -> List[int]:
+        new = self.output_token_ids[self._stream_cursor :]
+        self._stream_cursor = len(self.output_token_ids)
+        return new
+
+    def reset_for_recompute(self) -> None:
+        """Preemption: KV is dropped; everything recomputes on readmission."""
+        self.num_computed = 0
+        self.block_table = []
+        self.status = SeqStatus.PREEMPTED
+
+
+@dataclass
+class RequestOutput:
+    """Incremental output handed to the server/stream layer after a step."""
+
+    request_id: str
+    new_token_ids: List[int]
+    text_delta: str
+    finished: bool
+    finish_reason: Optional[str] = None
+    num_prompt_tokens: int = 0
+    num_output_tokens: int = 0
+    num_cached_tokens: int = 0
+    first_token: bool = False

@@ -1,0 +1,96 @@
+"""Tokenizers for the engine.
+
+Two implementations behind one interface:
+  * SyntheticTokenizer — deterministic hash-based word tokenizer for the
+    no-network environment (random-init weights make real BPE irrelevant);
+    round-trips text -> ids -> distinct text, stable across processes so
+    prefix-aware routing and prefix caching see consistent ids.
+  * HFTokenizer — loads a real tokenizer.json from a local directory via the
+    `tokenizers` library when one is provided.
+"""
+
+from __future__ import annotations
+
+import hashlib
+from typing import List, Optional
+
+
+class BaseTokenizer:
+    bos_token_id: int = 1
+    eos_token_id: int = 2
+
+    def encode(self, text: str) -> List[int]:
+        raise NotImplementedError
+
+    def decode(self, ids: List[int]) -> str:
+        raise NotImplementedError
+
+    def decode_token(self, token_id: int) -> str:
+        return self.decode([token_id])
+
+
+class SyntheticTokenizer(BaseTokenizer):
+    """Stable word-level tokenizer: each whitespace word maps to an id by
+    BLAKE2 hash into [16, vocab). ids 0-15 are reserved specials."""
+
+    RESERVED = 16
+
+    def __init__(self, vocab_size: int = 128256) -> None:
+        self.vocab_size = vocab_size
+
+    def encode(self, text: str) -> List[int]:
+        ids = []
+        for word in text.split():
+            h = int.from_bytes(
+                hashlib.blake2b(word.encode(), digest_size=4).digest(), "little"
+            )
+            ids.append(self.RESERVED + h % (self.vocab_size - self.RESERVED))
+        return ids
+
+    def decode(self, ids: List[int]) -> str:
+        return " ".join(self.decode_token(i) for i in ids)
+
+    def decode_token(self, token_id: int) -> str:
+        if token_id == self.eos_token_id:
+            return ""
+        return f"w{token_id}"
+
+
+class HFTokenizer(BaseTokenizer):
+    def __init__(self, path: str) -> None:
+        from tokenizers import Tokenizer
+
+        import os
+
+        f = path
+        if os.path.isdir(path):
+            f = os.path.join(path, "tokenizer.json")
+        self.tk = Tokenizer.from_file(f)
+        self.vocab_size = self.tk.get_vocab_size()
+
+    def encode(self, text: str) -> List[int]:
+        return self.tk.encode(text).ids
+
+    def decode(self, ids: List[int]) -> str:
+        return self.tk.decode(ids)
+
+
+def get_tokenizer(spec: str, vocab_size: int) -> BaseTokenizer:
+    if spec == "synthetic":
+        return SyntheticTokenizer(vocab_size)
+    return HFTokenizer(spec)
+
+
+def render_chat(messages: List[dict], add_generation_prompt: bool = True) -> str:
+    """Minimal chat template (role-tagged concatenation)."""
+    parts = []
+    for m in messages:
+        content = m.get("content") or ""
+        if isinstance(content, list):  # OpenAI content-part arrays
+            content = " ".join(
+                p.get("text", "") for p in content if isinstance(p, dict)
+            )
+        parts.append(f"<|{m.get('role', 'user')}|> {content}")
+    if add_generation_prompt:
+        parts.append("<|assistant|>")
+    return "\n".join(parts)

@@ -1,0 +1,63 @@
+"""Load HF-format Llama safetensors into the fused-weight model.
+
+Handles the name mapping from HF checkpoints
+(model.layers.N.self_attn.{q,k,v}_proj.weight etc.) onto the pre-fused
+qkv_proj / gate_up_proj parameters, with TP sharding.
+"""
+
+from __future__ import annotations
+
+import glob
+import os
+
+import torch
+
+from production_stack_amd.parallel import state as pstate
+
+
+def load_safetensors(model, path: str) -> None:
+    from safetensors.torch import load_file
+
+    files = sorted(glob.glob(os.path.join(path, "*.safetensors")))
+    if not files:
+        raise FileNotFoundError(f"no safetensors under {path}")
+    state = {}
+    for f in files:
+        state.update(load_file(f))
+
+    tp = pstate.tp_size()
+    rank = pstate.tp_rank()
+
+    def shard(t: torch.Tensor, dim: int) -> torch.Tensor:
+        if tp == 1:
+            return t
+        n = t.shape[dim] // tp
+        return t.narrow(dim, rank * n, n)
+
+    cfg = model.cfg
+
+    def get(name):
+        return state[name].to(torch.bfloat16)
+
+    with torch.no_grad():
+        model.embed.copy_(get("model.embed_tokens.weight"))
+        model.final_norm.copy_(get("model.norm.weight"))
+        if "lm_head.weight" in state:
+            model.lm_head.copy_(get("lm_head.weight"))
+        else:  # tied embeddings
+            model.lm_head.copy_(get("model.embed_tokens.weight"))
+        for i, layer in enumerate(model.layers):
+            pre = f"model.layers.{i}."
+            q = shard(get(pre + "self_attn.q_proj.weight"), 0)
+            k = shard(get(pre + "self_attn.k_proj.weight"), 0)
+            v = shard(get(pre + "self_attn.v_proj.weight"), 0)
+            layer.qkv_proj.copy_(torch.cat([q, k, v], dim=0))
+            layer.o_proj.copy_(shard(get(pre + "self_attn.o_proj.weight"), 1))
+            g = shard(get(pre + "mlp.gate_proj.weight"), 0)
+            u = shard(get(pre + "mlp.up_proj.weight"), 0)
+            layer.gate_up_proj.copy_(torch.cat([g, u], dim=0))
+            layer.down_proj.copy_(shard(get(pre + "mlp.down_proj.weight"), 1))
+            layer.input_norm.copy_(get(pre + "input_layernorm.weight"))
+            layer.post_attn_norm.copy_(
+                get(pre + "post_attention_layernorm.weight")
+            )
