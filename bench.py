@@ -1,0 +1,231 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark: multi-round-QA-shaped workload, Llama-3-8B,
+one engine replica per GPU (data parallel; BASELINE.json configs).
+
+Workload shape follows the reference's canonical multi-round-qa harness
+(reference benchmarks/multi-round-qa/run.sh: shared system prompt, growing
+per-user chat history, ~100-token answers), run in steady-state saturation:
+each rank keeps `--users` concurrent conversations alive, resubmitting a
+user's next round (history grown) the moment their answer completes. Data is
+synthetic token ids; weights are random-init bf16 (no network in the
+environment).
+
+A "step" is one engine step (one continuous-batching iteration: chunked
+prefills + decodes). The headline metric is aggregate output tokens/s across
+all ranks; p50 TTFT over rounds submitted inside the timed window is
+reported alongside.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+it is launched under torch.distributed.run with one rank per GPU over RCCL.
+Rank 0 prints exactly one JSON line.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+from production_stack_amd.engine.config import (
+    CacheConfig,
+    EngineConfig,
+    SchedulerConfig,
+)
+from production_stack_amd.engine.engine import LLMEngine
+from production_stack_amd.engine.sampling import SamplingParams
+
+SYSTEM_PROMPT_TOKENS = 1000
+QUESTION_TOKENS = 100
+ANSWER_TOKENS = 100
+VOCAB_LOW, VOCAB_HIGH = 16, 128000
+
+
+class User:
+    def __init__(self, uid, rng, system, vocab_high=VOCAB_HIGH):
+        self.uid = uid
+        self.rng = rng
+        self.vocab_high = vocab_high
+        self.history = list(system)
+        self.round = 0
+        self.submit_time = 0.0
+
+    def next_prompt(self) -> list:
+        self.round += 1
+        q = self.rng.integers(
+            VOCAB_LOW, self.vocab_high, size=QUESTION_TOKENS
+        ).tolist()
+        self.history.extend(q)
+        return list(self.history)
+
+    def complete(self, answer_tokens: list) -> None:
+        self.history.extend(answer_tokens)
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=64)
+    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--users", type=int, default=32, help="conversations per GPU")
+    ap.add_argument("--model", default="llama-3-8b")
+    ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--max-num-batched-tokens", type=int, default=8192)
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group(backend="nccl" if use_cuda else "gloo")
+
+    cfg = EngineConfig(
+        model=args.model,
+        max_model_len=args.max_model_len,
+        seed=1234 + rank,
+        cache=CacheConfig(
+            block_size=16,
+            gpu_memory_utilization=0.85,
+            enable_prefix_caching=True,
+            num_gpu_blocks=None if use_cuda else 2048,
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=max(args.users * 2, 64),
+            max_num_batched_tokens=args.max_num_batched_tokens,
+        ),
+    )
+    engine = LLMEngine(cfg, device=None if use_cuda else "cpu")
+
+    vocab_high = min(VOCAB_HIGH, engine.model_cfg.vocab_size - 1)
+    rng = np.random.default_rng(4242 + rank)
+    system = rng.integers(
+        VOCAB_LOW, vocab_high, size=SYSTEM_PROMPT_TOKENS
+    ).tolist()
+    users = [User(u, rng, system, vocab_high) for u in range(args.users)]
+    params = SamplingParams(
+        max_tokens=ANSWER_TOKENS, temperature=0.0, ignore_eos=True
+    )
+    answers: dict = {}
+
+    def submit(user: User) -> None:
+        # cap history so prompt + answer fits the model length
+        limit = args.max_model_len - ANSWER_TOKENS - QUESTION_TOKENS - 16
+        if len(user.history) > limit:
+            keep = limit - SYSTEM_PROMPT_TOKENS
+            user.history = (
+                user.history[:SYSTEM_PROMPT_TOKENS]
+                + user.history[-keep:]
+            )
+        prompt = user.next_prompt()
+        rid = f"u{user.uid}-r{user.round}"
+        user.submit_time = time.perf_counter()
+        answers[rid] = (user, [])
+        engine.add_request(rid, prompt, params)
+
+    for u in users:
+        submit(u)
+
+    ttfts: list = []
+    in_window = False
+    window_tokens = 0
+
+    def run_step() -> None:
+        nonlocal window_tokens
+        for out in engine.step():
+            entry = answers.get(out.request_id)
+            if entry is None:
+                continue
+            user, toks = entry
+            toks.extend(out.new_token_ids)
+            if in_window:
+                window_tokens += len(out.new_token_ids)
+                if out.first_token:
+                    ttfts.append(time.perf_counter() - user.submit_time)
+            if out.finished:
+                user.complete(toks)
+                del answers[out.request_id]
+                submit(user)
+
+    # warmup (includes the initial prefill wave)
+    for _ in range(args.warmup):
+        run_step()
+
+    if dist:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    in_window = True
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    in_window = False
+
+    # aggregate: sum tokens over ranks, max elapsed over ranks
+    total_tokens = window_tokens
+    if dist:
+        t = torch.tensor([float(window_tokens)], dtype=torch.float64)
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        if use_cuda:
+            t, e = t.cuda(), e.cuda()
+        dist.all_reduce(t)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        total_tokens = t.item()
+        elapsed = e.item()
+
+    if rank == 0:
+        value = total_tokens / elapsed
+        ttft_p50 = (
+            float(np.percentile(ttfts, 50) * 1000.0) if ttfts else None
+        )
+        print(
+            json.dumps(
+                {
+                    "metric": "output_tokens_per_sec",
+                    "value": round(value, 2),
+                    "unit": "tokens/s",
+                    "n_gpus": world,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": round(elapsed * 1000.0 / args.steps, 3),
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "bfloat16",
+                    "data": "synthetic",
+                    "ttft_p50_ms": ttft_p50,
+                    "config": {
+                        "model": args.model,
+                        "workload": "multi-round-qa",
+                        "users_per_gpu": args.users,
+                        "system_prompt_tokens": SYSTEM_PROMPT_TOKENS,
+                        "question_tokens": QUESTION_TOKENS,
+                        "answer_tokens": ANSWER_TOKENS,
+                        "max_model_len": args.max_model_len,
+                        "parallelism": f"dp{world}",
+                        "prefix_caching": True,
+                    },
+                }
+            )
+        )
+
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -193,16 +193,13 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.no_grad()
     def random_init(self, seed: int = 0) -> None:
-        gen = torch.Generator(device="cpu").manual_seed(seed)
+        dev = self.embed.device
+        gen = torch.Generator(device=dev).manual_seed(seed)
         for name, p in self.named_parameters():
             if "norm" in name:
                 p.fill_(1.0)
             else:
-                # init on CPU in chunks to bound host memory, then copy
-                t = torch.empty(
-                    p.shape, dtype=torch.float16, device="cpu"
-                ).normal_(0.0, 0.02, generator=gen)
-                p.copy_(t.to(p.dtype))
+                p.normal_(0.0, 0.02, generator=gen)
 
     @property
     def kv_heads(self) -> int:

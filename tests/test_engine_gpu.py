@@ -1,0 +1,113 @@
+"""GPU end-to-end engine tests (mini model with kernel-supported head_dim)."""
+
+import pytest
+import torch
+
+from production_stack_amd.engine.config import (
+    CacheConfig,
+    EngineConfig,
+    SchedulerConfig,
+)
+from production_stack_amd.engine.engine import LLMEngine
+from production_stack_amd.engine.sampling import SamplingParams
+
+pytestmark = pytest.mark.gpu
+
+
+def make_engine(**kw):
+    cfg = EngineConfig(
+        model=kw.pop("model", "mini-llama"),
+        max_model_len=kw.pop("max_model_len", 1024),
+        cache=CacheConfig(
+            num_gpu_blocks=kw.pop("num_gpu_blocks", 256), block_size=16
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=kw.pop("max_num_seqs", 16),
+            max_num_batched_tokens=kw.pop("max_num_batched_tokens", 2048),
+        ),
+        **kw,
+    )
+    return LLMEngine(cfg, device="cuda")
+
+
+def test_extension_loaded():
+    from production_stack_amd import ops
+
+    assert ops.HAVE_EXT, "HIP extension must load on the GPU box"
+
+
+def test_generate_deterministic():
+    eng = make_engine()
+    p = SamplingParams(max_tokens=16, temperature=0.0, ignore_eos=True)
+    prompt = list(range(100, 180))
+    o1 = eng.generate([prompt], p)["offline-0"]
+    o2 = eng.generate([prompt], p)["offline-0"]
+    assert o1 == o2 and len(o1) == 16
+
+
+def test_gpu_matches_cpu_reference_engine():
+    """Same tiny model on GPU (HIP kernels) vs CPU (torch reference):
+    greedy tokens must agree (same random-init seed)."""
+    p = SamplingParams(max_tokens=12, temperature=0.0, ignore_eos=True)
+    prompt = list(range(7, 87))
+    gpu = make_engine()
+    out_gpu = gpu.generate([prompt], p)["offline-0"]
+    cfg = EngineConfig(
+        model="mini-llama",
+        max_model_len=1024,
+        cache=CacheConfig(num_gpu_blocks=256, block_size=16),
+        scheduler=SchedulerConfig(max_num_seqs=16, max_num_batched_tokens=2048),
+    )
+    # CPU random_init uses a CPU generator -> different weights; instead copy
+    # the GPU engine's weights down.
+    cpu = LLMEngine(cfg, device="cpu")
+    sd = {k: v.cpu() for k, v in gpu.runner.model.state_dict().items()}
+    cpu.runner.model.load_state_dict(sd)
+    out_cpu = cpu.generate([prompt], p)["offline-0"]
+    # bf16 kernel-vs-reference drift can flip argmax on random-init logits;
+    # require strong prefix agreement.
+    agree = sum(a == b for a, b in zip(out_gpu, out_cpu))
+    assert agree >= len(out_gpu) // 2, f"{out_gpu} vs {out_cpu}"
+
+
+def test_chunked_prefill_matches_single_shot_gpu():
+    prompt = list(range(10, 500))
+    p = SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True)
+    big = make_engine()
+    out_big = big.generate([prompt], p)["offline-0"]
+    small = make_engine(max_num_batched_tokens=128)
+    out_small = small.generate([prompt], p)["offline-0"]
+    assert out_big == out_small
+
+
+def test_prefix_cache_reuse_gpu():
+    eng = make_engine()
+    p = SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True)
+    prompt = list(range(3, 300))
+    cold = eng.generate([prompt], p)["offline-0"]
+    warm = eng.generate([prompt], p)["offline-0"]
+    assert eng.block_manager.prefix_hits > 0
+    assert cold == warm
+
+
+def test_batched_mixed_phase_gpu():
+    """Batch where prefills and decodes run in one step."""
+    eng = make_engine()
+    p = SamplingParams(max_tokens=6, temperature=0.0, ignore_eos=True)
+    prompts = [list(range(50, 120)), list(range(200, 230)), [9, 8, 7]]
+    batched = eng.generate(prompts, p)
+    for i, prompt in enumerate(prompts):
+        solo = make_engine().generate([prompt], p)["offline-0"]
+        assert batched[f"offline-{i}"] == solo
+
+
+def test_flagship_8b_one_decode():
+    free, _ = torch.cuda.mem_get_info()
+    if free < 30e9:
+        pytest.skip("needs ~30 GB free HBM")
+    eng = make_engine(
+        model="llama-3-8b", num_gpu_blocks=512, max_model_len=2048
+    )
+    p = SamplingParams(max_tokens=2, temperature=0.0, ignore_eos=True)
+    out = eng.generate([list(range(1000, 1128))], p)["offline-0"]
+    assert len(out) == 2
