@@ -1,0 +1,117 @@
+"""Router CLI (flag surface mirrors reference parsers/parser.py:125-495 so
+helm's values->args mapping carries over unchanged)."""
+
+from __future__ import annotations
+
+import argparse
+from typing import Optional
+
+
+def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
+    p = argparse.ArgumentParser(description="MI355X-native vLLM-router-compatible request router")
+    p.add_argument("--host", default="0.0.0.0")
+    p.add_argument("--port", type=int, default=8001)
+
+    # service discovery
+    p.add_argument(
+        "--service-discovery",
+        choices=["static", "k8s", "external"],
+        default="static",
+    )
+    p.add_argument("--static-backends", type=str, default=None,
+                   help="comma-separated backend URLs")
+    p.add_argument("--static-models", type=str, default=None,
+                   help="comma-separated model names aligned with backends")
+    p.add_argument("--static-aliases", type=str, default=None,
+                   help="alias1:model1,alias2:model2")
+    p.add_argument("--static-model-labels", type=str, default=None)
+    p.add_argument("--static-model-types", type=str, default=None)
+    p.add_argument("--static-backend-health-checks", action="store_true")
+    p.add_argument("--health-check-interval", type=float, default=60.0)
+    p.add_argument("--k8s-namespace", default="default")
+    p.add_argument("--k8s-port", type=int, default=8000)
+    p.add_argument("--k8s-label-selector", default=None)
+
+    # routing
+    p.add_argument(
+        "--routing-logic",
+        choices=[
+            "roundrobin",
+            "session",
+            "kvaware",
+            "prefixaware",
+            "disaggregated_prefill",
+            "disaggregated_prefill_orchestrated",
+        ],
+        default="roundrobin",
+    )
+    p.add_argument("--session-key", default=None)
+    p.add_argument("--prefix-min-match-length", type=int, default=128)
+    p.add_argument("--kv-aware-threshold", type=int, default=2000)
+    p.add_argument("--lmcache-controller-port", type=int, default=9000,
+                   help="KV-pool controller port (kvaware routing)")
+    p.add_argument("--kv-controller-host", default="127.0.0.1")
+    p.add_argument("--prefill-model-labels", type=str, default=None)
+    p.add_argument("--decode-model-labels", type=str, default=None)
+    p.add_argument("--max-instance-failover-reroute-attempts", type=int,
+                   default=0)
+
+    # stats / metrics
+    p.add_argument("--engine-stats-interval", type=float, default=10.0)
+    p.add_argument("--request-stats-window", type=float, default=60.0)
+    p.add_argument("--log-stats", action="store_true")
+    p.add_argument("--log-stats-interval", type=float, default=10.0)
+
+    # dynamic config
+    p.add_argument("--dynamic-config-json", default=None)
+
+    # features
+    p.add_argument("--feature-gates", default=None,
+                   help="e.g. SemanticCache=true")
+    p.add_argument("--callbacks", default=None,
+                   help="module path exposing pre/post request hooks")
+    p.add_argument("--request-rewriter", default="noop")
+    p.add_argument("--file-storage-path", default="/tmp/vllm_files")
+    p.add_argument("--enable-batch-api", action="store_true")
+    p.add_argument("--batch-processor-db", default="/tmp/vllm_batches.sqlite")
+
+    # auth / api
+    p.add_argument("--api-key", default=None,
+                   help="bearer token forwarded to engines")
+
+    # logging
+    p.add_argument("--log-level", default="info",
+                   choices=["trace", "debug", "info", "warning", "error"])
+    p.add_argument("--log-format", default="text", choices=["text", "json"])
+
+    args = p.parse_args(argv)
+    validate_args(args)
+    return args
+
+
+def validate_args(args: argparse.Namespace) -> None:
+    if args.service_discovery == "static":
+        if not args.static_backends:
+            raise ValueError(
+                "--static-backends is required with static discovery"
+            )
+        if not args.static_models:
+            raise ValueError(
+                "--static-models is required with static discovery"
+            )
+        n_b = len([u for u in args.static_backends.split(",") if u.strip()])
+        n_m = len([m for m in args.static_models.split(",") if m.strip()])
+        if n_m not in (1, n_b):
+            raise ValueError(
+                "number of --static-models must be 1 or match "
+                "--static-backends"
+            )
+    if args.routing_logic in (
+        "disaggregated_prefill",
+        "disaggregated_prefill_orchestrated",
+    ):
+        if not (args.prefill_model_labels and args.decode_model_labels):
+            raise ValueError(
+                "disaggregated prefill routing requires "
+                "--prefill-model-labels and --decode-model-labels"
+            )

@@ -1,0 +1,352 @@
+"""Routing algorithms — the product's core logic.
+
+Behavioural parity with reference src/vllm_router/routers/routing_logic.py:
+  roundrobin (:139-195), session (:198-249), kvaware (:252-428),
+  prefixaware (:431-522), disaggregated_prefill (:525-565),
+  disaggregated_prefill_orchestrated (:568-673), QPS fallback (:62-84),
+  session-id extraction (:107-115), initialize/get (:677-744).
+"""
+
+from __future__ import annotations
+
+import abc
+import enum
+import logging
+import random
+from typing import Any, Dict, List, Optional
+
+from production_stack_amd.router.hashring import HashRing
+from production_stack_amd.router.hashtrie import HashTrie
+from production_stack_amd.router.service_discovery import EndpointInfo
+from production_stack_amd.router.stats import EngineStats, RequestStats
+
+logger = logging.getLogger("router.routing")
+
+
+class RoutingLogic(str, enum.Enum):
+    ROUND_ROBIN = "roundrobin"
+    SESSION_BASED = "session"
+    KVAWARE = "kvaware"
+    PREFIXAWARE = "prefixaware"
+    DISAGGREGATED_PREFILL = "disaggregated_prefill"
+    DISAGGREGATED_PREFILL_ORCHESTRATED = "disaggregated_prefill_orchestrated"
+
+
+def extract_prompt_text(request_json: Dict[str, Any]) -> str:
+    if "messages" in request_json:
+        parts = []
+        for m in request_json.get("messages") or []:
+            c = m.get("content")
+            if isinstance(c, str):
+                parts.append(c)
+            elif isinstance(c, list):
+                parts.extend(
+                    p.get("text", "") for p in c if isinstance(p, dict)
+                )
+        return "\n".join(parts)
+    prompt = request_json.get("prompt", "")
+    if isinstance(prompt, list):
+        return "\n".join(str(p) for p in prompt)
+    return str(prompt)
+
+
+def _qps_min_endpoint(
+    endpoints: List[EndpointInfo],
+    request_stats: Dict[str, RequestStats],
+) -> str:
+    """Lowest-QPS fallback (reference routing_logic.py:62-84)."""
+    best_url, best_qps = None, float("inf")
+    for ep in endpoints:
+        q = request_stats.get(ep.url)
+        qps = q.qps if q else 0.0
+        if qps < best_qps:
+            best_url, best_qps = ep.url, qps
+    return best_url or endpoints[0].url
+
+
+def get_session_id(request: Any, session_key: Optional[str]) -> Optional[str]:
+    if session_key is None:
+        return None
+    headers = getattr(request, "headers", None) or {}
+    try:
+        return headers.get(session_key)
+    except AttributeError:
+        return None
+
+
+class RoutingInterface(abc.ABC):
+    @abc.abstractmethod
+    async def route_request(
+        self,
+        endpoints: List[EndpointInfo],
+        engine_stats: Dict[str, EngineStats],
+        request_stats: Dict[str, RequestStats],
+        request: Any,
+        request_json: Optional[Dict[str, Any]] = None,
+    ) -> str:
+        ...
+
+    def on_request_done(self, url: str) -> None:
+        pass
+
+
+class RoundRobinRouter(RoutingInterface):
+    _MAX_CACHE_SIZE = 1024
+
+    def __init__(self, **_: Any) -> None:
+        self._counters: Dict[tuple, int] = {}
+
+    async def route_request(
+        self, endpoints, engine_stats, request_stats, request,
+        request_json=None,
+    ) -> str:
+        key = tuple(sorted(ep.url for ep in endpoints))
+        if len(self._counters) > self._MAX_CACHE_SIZE:
+            self._counters.clear()
+        idx = self._counters.get(key, 0)
+        self._counters[key] = idx + 1
+        return key[idx % len(key)]
+
+
+class SessionRouter(RoutingInterface):
+    def __init__(self, session_key: Optional[str] = None, **_: Any) -> None:
+        self.session_key = session_key or "x-user-id"
+        self.ring = HashRing()
+
+    async def route_request(
+        self, endpoints, engine_stats, request_stats, request,
+        request_json=None,
+    ) -> str:
+        urls = [ep.url for ep in endpoints]
+        self.ring.update_nodes(urls)
+        session_id = get_session_id(request, self.session_key)
+        if session_id is None and request_json:
+            session_id = request_json.get("session_id") or request_json.get(
+                "user"
+            )
+        if not session_id:
+            return _qps_min_endpoint(endpoints, request_stats)
+        return self.ring.get_node(str(session_id)) or urls[0]
+
+
+class PrefixAwareRouter(RoutingInterface):
+    def __init__(
+        self,
+        prefix_min_match_length: int = 128,
+        chunk_size: int = 128,
+        **_: Any,
+    ) -> None:
+        self.trie = HashTrie(chunk_size=chunk_size)
+        self.min_match = prefix_min_match_length
+
+    async def route_request(
+        self, endpoints, engine_stats, request_stats, request,
+        request_json=None,
+    ) -> str:
+        if not request_json:
+            return _qps_min_endpoint(endpoints, request_stats)
+        text = extract_prompt_text(request_json)
+        available = {ep.url for ep in endpoints}
+        matched, candidates = await self.trie.longest_prefix_match(
+            text, available
+        )
+        if matched < self.min_match or not candidates:
+            url = _qps_min_endpoint(endpoints, request_stats)
+        else:
+            url = random.choice(sorted(candidates))
+        await self.trie.insert(text, url)
+        return url
+
+
+class KvAwareRouter(RoutingInterface):
+    """Routes to the replica holding the longest KV prefix for the prompt,
+    via the KV-pool controller (production_stack_amd.kvpool.controller).
+
+    Reference parity: kvaware (routing_logic.py:252-428) — tokenize, lookup,
+    threshold, session/QPS fallback.
+    """
+
+    def __init__(
+        self,
+        kv_controller_host: str = "127.0.0.1",
+        kv_controller_port: int = 9000,
+        kv_match_threshold: int = 2000,
+        session_key: Optional[str] = None,
+        **_: Any,
+    ) -> None:
+        self.host = kv_controller_host
+        self.port = kv_controller_port
+        self.threshold = kv_match_threshold
+        self.fallback = SessionRouter(session_key)
+        self._client = None
+
+    async def _lookup(self, token_ids: List[int]) -> Optional[str]:
+        from production_stack_amd.kvpool.client import ControllerClient
+
+        if self._client is None:
+            self._client = ControllerClient(self.host, self.port)
+        try:
+            matches = await self._client.lookup(token_ids)
+        except (ConnectionError, OSError, TimeoutError):
+            return None
+        if not matches:
+            return None
+        # matches: {instance_url: matched_tokens}
+        best_url, best = None, -1
+        for url, n in matches.items():
+            if n > best:
+                best_url, best = url, n
+        if best >= max(len(token_ids) - self.threshold, 0):
+            return best_url
+        return None
+
+    async def _tokenize(
+        self, endpoints: List[EndpointInfo], text: str
+    ) -> Optional[List[int]]:
+        import aiohttp
+
+        for ep in endpoints:
+            try:
+                async with aiohttp.ClientSession() as sess:
+                    async with sess.post(
+                        ep.url + "/tokenize",
+                        json={"prompt": text},
+                        timeout=aiohttp.ClientTimeout(total=5),
+                    ) as r:
+                        if r.status == 200:
+                            data = await r.json()
+                            return data.get("tokens")
+            except Exception:
+                continue
+        return None
+
+    async def route_request(
+        self, endpoints, engine_stats, request_stats, request,
+        request_json=None,
+    ) -> str:
+        if request_json:
+            text = extract_prompt_text(request_json)
+            tokens = await self._tokenize(endpoints, text)
+            if tokens:
+                url = await self._lookup(tokens)
+                if url and any(ep.url == url for ep in endpoints):
+                    return url
+        return await self.fallback.route_request(
+            endpoints, engine_stats, request_stats, request, request_json
+        )
+
+
+class DisaggregatedPrefillRouter(RoutingInterface):
+    """2-request protocol: max_tokens==1 requests go to the prefill pool."""
+
+    def __init__(
+        self,
+        prefill_model_labels: Optional[List[str]] = None,
+        decode_model_labels: Optional[List[str]] = None,
+        **_: Any,
+    ) -> None:
+        self.prefill_labels = prefill_model_labels or ["prefill"]
+        self.decode_labels = decode_model_labels or ["decode"]
+
+    async def route_request(
+        self, endpoints, engine_stats, request_stats, request,
+        request_json=None,
+    ) -> str:
+        is_prefill = bool(request_json) and request_json.get("max_tokens") == 1
+        labels = self.prefill_labels if is_prefill else self.decode_labels
+        for ep in endpoints:
+            if ep.model_label in labels:
+                return ep.url
+        return endpoints[0].url
+
+
+class DisaggregatedPrefillOrchestratedRouter(RoutingInterface):
+    """Single-request P->D chaining; pool selection here, orchestration in
+    request_service.route_disaggregated_request."""
+
+    def __init__(
+        self,
+        prefill_model_labels: Optional[List[str]] = None,
+        decode_model_labels: Optional[List[str]] = None,
+        **_: Any,
+    ) -> None:
+        self.prefill_labels = prefill_model_labels or ["prefill"]
+        self.decode_labels = decode_model_labels or ["decode"]
+        self._p_idx = 0
+        self._d_idx = 0
+
+    def find_pools(self, endpoints: List[EndpointInfo]):
+        prefill = [
+            ep for ep in endpoints if ep.model_label in self.prefill_labels
+        ]
+        decode = [
+            ep for ep in endpoints if ep.model_label in self.decode_labels
+        ]
+        return prefill, decode
+
+    def select_prefill_endpoint(
+        self, endpoints: List[EndpointInfo]
+    ) -> Optional[str]:
+        pool, _ = self.find_pools(endpoints)
+        if not pool:
+            return None
+        url = pool[self._p_idx % len(pool)].url
+        self._p_idx += 1
+        return url
+
+    def select_decode_endpoint(
+        self, endpoints: List[EndpointInfo]
+    ) -> Optional[str]:
+        _, pool = self.find_pools(endpoints)
+        if not pool:
+            return None
+        url = pool[self._d_idx % len(pool)].url
+        self._d_idx += 1
+        return url
+
+    async def route_request(
+        self, endpoints, engine_stats, request_stats, request,
+        request_json=None,
+    ) -> str:
+        url = self.select_decode_endpoint(endpoints)
+        return url or endpoints[0].url
+
+
+_router: Optional[RoutingInterface] = None
+
+_CLASSES = {
+    RoutingLogic.ROUND_ROBIN: RoundRobinRouter,
+    RoutingLogic.SESSION_BASED: SessionRouter,
+    RoutingLogic.KVAWARE: KvAwareRouter,
+    RoutingLogic.PREFIXAWARE: PrefixAwareRouter,
+    RoutingLogic.DISAGGREGATED_PREFILL: DisaggregatedPrefillRouter,
+    RoutingLogic.DISAGGREGATED_PREFILL_ORCHESTRATED:
+        DisaggregatedPrefillOrchestratedRouter,
+}
+
+
+def initialize_routing_logic(
+    routing_logic: str, **kwargs: Any
+) -> RoutingInterface:
+    global _router
+    logic = RoutingLogic(routing_logic)
+    _router = _CLASSES[logic](**kwargs)
+    logger.info("initialized routing logic: %s", logic.value)
+    return _router
+
+
+def reconfigure_routing_logic(
+    routing_logic: str, **kwargs: Any
+) -> RoutingInterface:
+    return initialize_routing_logic(routing_logic, **kwargs)
+
+
+def get_routing_logic() -> RoutingInterface:
+    if _router is None:
+        raise RuntimeError("routing logic not initialized")
+    return _router
+
+
+def reset_routing_logic() -> None:
+    global _router
+    _router = None

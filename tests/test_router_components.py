@@ -1,0 +1,184 @@
+"""Unit tests: trie, ring, stats, engine-stats parsing, parser, log, utils."""
+
+import asyncio
+import time
+
+import pytest
+
+from production_stack_amd.router.hashring import HashRing
+from production_stack_amd.router.hashtrie import HashTrie
+from production_stack_amd.router.log import redact
+from production_stack_amd.router.parser import parse_args
+from production_stack_amd.router.stats import (
+    EngineStats,
+    MovingAverageMonitor,
+    RequestStatsMonitor,
+)
+from production_stack_amd.router.utils import (
+    parse_static_aliases,
+    parse_static_urls,
+)
+
+
+def run(coro):
+    return asyncio.run(coro)
+
+
+# ---- hash trie -----------------------------------------------------------
+def test_trie_longest_prefix():
+    async def go():
+        t = HashTrie(chunk_size=4)
+        await t.insert("abcdefgh", "http://a")
+        await t.insert("abcdxxxx", "http://b")
+        n, eps = await t.longest_prefix_match("abcdefgh")
+        assert n == 8 and eps == {"http://a"}
+        n, eps = await t.longest_prefix_match("abcdzzzz")
+        assert n == 4 and eps == {"http://a", "http://b"}
+        n, eps = await t.longest_prefix_match("zzzz")
+        assert n == 0 and eps == set()
+
+    run(go())
+
+
+def test_trie_available_filter():
+    async def go():
+        t = HashTrie(chunk_size=4)
+        await t.insert("abcdefgh", "http://a")
+        n, eps = await t.longest_prefix_match("abcdefgh", {"http://b"})
+        assert eps == set()
+
+    run(go())
+
+
+def test_trie_remove_endpoint():
+    async def go():
+        t = HashTrie(chunk_size=4)
+        await t.insert("abcdefgh", "http://a")
+        await t.remove_endpoint("http://a")
+        n, eps = await t.longest_prefix_match("abcdefgh")
+        assert eps == set()
+
+    run(go())
+
+
+# ---- hash ring -----------------------------------------------------------
+def test_ring_consistency():
+    ring = HashRing(["a", "b", "c"])
+    assignments = {k: ring.get_node(f"key{k}") for k in range(100)}
+    ring.remove_node("b")
+    moved = 0
+    for k in range(100):
+        new = ring.get_node(f"key{k}")
+        if assignments[k] != "b":
+            if new != assignments[k]:
+                moved += 1
+    # consistent hashing: keys not on the removed node mostly stay put
+    assert moved == 0
+
+
+def test_ring_update_nodes():
+    ring = HashRing(["a", "b"])
+    ring.update_nodes(["b", "c"])
+    nodes = {ring.get_node(f"k{i}") for i in range(50)}
+    assert "a" not in nodes
+
+
+# ---- stats ---------------------------------------------------------------
+def test_moving_average_window_expiry():
+    m = MovingAverageMonitor(window=10.0)
+    now = time.time()
+    m.update(now - 20, 1.0)
+    m.update(now, 5.0)
+    assert m.get_average() == pytest.approx(5.0)
+
+
+def test_request_stats_lifecycle():
+    mon = RequestStatsMonitor(window=60)
+    now = time.time()
+    mon.on_new_request("http://a", "r1", now)
+    stats = mon.get_request_stats(now + 0.1)
+    assert stats["http://a"].in_prefill_requests == 1
+    mon.on_request_response("http://a", "r1", now + 0.5)
+    stats = mon.get_request_stats(now + 0.6)
+    assert stats["http://a"].in_decoding_requests == 1
+    assert stats["http://a"].ttft == pytest.approx(0.5, abs=0.01)
+    mon.on_request_complete("http://a", "r1", now + 1.0)
+    stats = mon.get_request_stats(now + 1.1)
+    assert stats["http://a"].finished_requests == 1
+    assert stats["http://a"].in_decoding_requests == 0
+
+
+def test_request_stats_failure_rolls_back():
+    mon = RequestStatsMonitor()
+    mon.on_new_request("http://a", "r1", time.time())
+    mon.on_request_failed("http://a", "r1")
+    stats = mon.get_request_stats()
+    assert stats["http://a"].in_prefill_requests == 0
+
+
+def test_engine_stats_parsing():
+    text = """# HELP vllm:num_requests_running ...
+vllm:num_requests_running{model_name="m"} 3.0
+vllm:num_requests_waiting{model_name="m"} 2.0
+vllm:gpu_cache_usage_perc{model_name="m"} 0.5
+vllm:gpu_prefix_cache_hits_total{model_name="m"} 30
+vllm:gpu_prefix_cache_queries_total{model_name="m"} 60
+other_metric 1.0
+"""
+    s = EngineStats.from_prometheus_text(text)
+    assert s.num_running_requests == 3
+    assert s.num_queuing_requests == 2
+    assert s.gpu_cache_usage_perc == 0.5
+    assert s.gpu_prefix_cache_hit_rate == pytest.approx(0.5)
+
+
+# ---- parser --------------------------------------------------------------
+def test_parser_static_ok():
+    args = parse_args(
+        [
+            "--service-discovery", "static",
+            "--static-backends", "http://a:8000,http://b:8000",
+            "--static-models", "m1,m2",
+            "--routing-logic", "roundrobin",
+        ]
+    )
+    assert args.static_backends.startswith("http://a")
+
+
+def test_parser_requires_backends():
+    with pytest.raises(ValueError):
+        parse_args(["--service-discovery", "static"])
+
+
+def test_parser_model_count_mismatch():
+    with pytest.raises(ValueError):
+        parse_args(
+            [
+                "--static-backends", "http://a,http://b,http://c",
+                "--static-models", "m1,m2",
+            ]
+        )
+
+
+def test_parser_disagg_requires_labels():
+    with pytest.raises(ValueError):
+        parse_args(
+            [
+                "--static-backends", "http://a",
+                "--static-models", "m",
+                "--routing-logic", "disaggregated_prefill",
+            ]
+        )
+
+
+# ---- log redaction -------------------------------------------------------
+def test_redaction():
+    assert "[REDACTED]" in redact("Authorization: Bearer sk-abc123")
+    assert "sk-abc123" not in redact("api_key=sk-abc123")
+    assert redact("normal message") == "normal message"
+
+
+# ---- utils ---------------------------------------------------------------
+def test_parse_helpers():
+    assert parse_static_urls("http://a/, http://b") == ["http://a", "http://b"]
+    assert parse_static_aliases("al:m1, a2:m2") == {"al": "m1", "a2": "m2"}
