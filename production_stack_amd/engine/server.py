@@ -1,0 +1,427 @@
+"""OpenAI-compatible HTTP front for the MI355X engine.
+
+API surface the router depends on (SURVEY.md section 2.9 / reference
+main_router proxy targets): /v1/completions, /v1/chat/completions (SSE with
+usage in the final chunk), /v1/models, /tokenize, /detokenize, /health,
+/metrics (vllm:* series the router scrapes), /is_sleeping, /sleep,
+/wake_up, /v1/load_lora_adapter, /v1/unload_lora_adapter.
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import contextlib
+import json
+import logging
+import time
+import uuid
+from typing import Optional
+
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
+
+from production_stack_amd.engine.async_engine import AsyncEngine
+from production_stack_amd.engine.config import (
+    CacheConfig,
+    EngineConfig,
+    SchedulerConfig,
+)
+from production_stack_amd.engine.engine import LLMEngine
+from production_stack_amd.engine.sampling import SamplingParams
+from production_stack_amd.engine.tokenizer import render_chat
+
+logger = logging.getLogger("engine.server")
+
+
+def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
+    max_tokens = (
+        body.get("max_tokens")
+        or body.get("max_completion_tokens")
+        or 128
+    )
+    temperature = body.get("temperature")
+    if temperature is None:
+        temperature = 1.0
+    return SamplingParams(
+        max_tokens=int(max_tokens),
+        temperature=float(temperature),
+        top_p=float(body.get("top_p") or 1.0),
+        top_k=int(body.get("top_k") or -1),
+        ignore_eos=bool(body.get("ignore_eos", False)),
+        seed=body.get("seed"),
+    )
+
+
+def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
+    async_engine = AsyncEngine(engine)
+
+    @contextlib.asynccontextmanager
+    async def lifespan(app: FastAPI):
+        async_engine.start(asyncio.get_running_loop())
+        cfg = engine.config
+        if getattr(cfg, "kv_controller_url", None):
+            host, _, port = cfg.kv_controller_url.rpartition(":")
+            from production_stack_amd.kvpool.client import EngineReporter
+
+            app.state.kv_reporter = EngineReporter(
+                engine,
+                url=getattr(cfg, "advertise_url", "http://127.0.0.1:8000"),
+                host=host or "127.0.0.1",
+                port=int(port),
+            )
+            app.state.kv_reporter.start()
+        yield
+        reporter = getattr(app.state, "kv_reporter", None)
+        if reporter is not None:
+            await reporter.stop()
+        async_engine.stop()
+
+    app = FastAPI(
+        title=f"production-stack-amd engine ({served_model})",
+        lifespan=lifespan,
+    )
+    app.state.engine = engine
+    app.state.async_engine = async_engine
+    app.state.served_model = served_model
+    app.state.lora_adapters = {}
+    app.state.start_time = time.time()
+
+    # ------------------------------------------------------------------
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.get("/v1/models")
+    async def models():
+        data = [
+            {
+                "id": served_model,
+                "object": "model",
+                "created": int(app.state.start_time),
+                "owned_by": "production-stack-amd",
+            }
+        ]
+        for name in app.state.lora_adapters:
+            data.append(
+                {
+                    "id": name,
+                    "object": "model",
+                    "created": int(time.time()),
+                    "owned_by": "production-stack-amd",
+                    "parent": served_model,
+                }
+            )
+        return {"object": "list", "data": data}
+
+    @app.get("/metrics")
+    async def metrics():
+        m = engine.engine_metrics()
+        lines = []
+        for k, v in m.items():
+            name = f"vllm:{k}"
+            kind = "counter" if k.endswith("_total") else "gauge"
+            lines.append(f"# TYPE {name} {kind}")
+            lines.append(
+                f'{name}{{model_name="{served_model}"}} {v}'
+            )
+        q = m.get("gpu_prefix_cache_queries_total", 0.0)
+        h = m.get("gpu_prefix_cache_hits_total", 0.0)
+        lines.append("# TYPE vllm:gpu_prefix_cache_hit_rate gauge")
+        lines.append(
+            f'vllm:gpu_prefix_cache_hit_rate{{model_name="{served_model}"}} '
+            f"{h / q if q else 0.0}"
+        )
+        return PlainTextResponse("\n".join(lines) + "\n")
+
+    @app.post("/tokenize")
+    async def tokenize(request: Request):
+        body = await request.json()
+        if "messages" in body:
+            text = render_chat(body["messages"])
+        else:
+            text = str(body.get("prompt", ""))
+        tokens = engine.tokenizer.encode(text)
+        return {"tokens": tokens, "count": len(tokens), "max_model_len":
+                engine.config.max_model_len}
+
+    @app.post("/detokenize")
+    async def detokenize(request: Request):
+        body = await request.json()
+        return {"prompt": engine.tokenizer.decode(body.get("tokens", []))}
+
+    # ---- sleep / wake -------------------------------------------------
+    @app.get("/is_sleeping")
+    async def is_sleeping():
+        return {"is_sleeping": engine.is_sleeping}
+
+    @app.post("/sleep")
+    async def sleep(level: int = 1):
+        engine.sleep(level)
+        return {"status": "ok"}
+
+    @app.post("/wake_up")
+    async def wake_up():
+        engine.wake_up()
+        return {"status": "ok"}
+
+    # ---- LoRA (load/unload surface for the operator's LoraAdapter CRD) --
+    @app.post("/v1/load_lora_adapter")
+    async def load_lora(request: Request):
+        body = await request.json()
+        name = body.get("lora_name")
+        if not name:
+            return JSONResponse(
+                status_code=400, content={"error": "lora_name required"}
+            )
+        app.state.lora_adapters[name] = body.get("lora_path", "")
+        return {"status": "ok"}
+
+    @app.post("/v1/unload_lora_adapter")
+    async def unload_lora(request: Request):
+        body = await request.json()
+        app.state.lora_adapters.pop(body.get("lora_name"), None)
+        return {"status": "ok"}
+
+    # ---- completions ---------------------------------------------------
+    async def _run_completion(request: Request, chat: bool):
+        body = await request.json()
+        params = _params_from_body(body, engine.config.max_model_len)
+        rid = (
+            request.headers.get("x-request-id")
+            or f"cmpl-{uuid.uuid4().hex[:24]}"
+        )
+        if chat:
+            prompt = render_chat(body.get("messages") or [])
+        else:
+            p = body.get("prompt", "")
+            if isinstance(p, list) and p and isinstance(p[0], int):
+                prompt = p
+            elif isinstance(p, list):
+                prompt = "\n".join(str(x) for x in p)
+            else:
+                prompt = str(p)
+        if isinstance(prompt, str):
+            prompt_tokens = engine.tokenizer.encode(prompt)
+        else:
+            prompt_tokens = prompt
+        if len(prompt_tokens) + 1 > engine.config.max_model_len:
+            return JSONResponse(
+                status_code=400,
+                content={
+                    "error": {
+                        "message": (
+                            f"prompt ({len(prompt_tokens)} tokens) exceeds "
+                            f"max_model_len {engine.config.max_model_len}"
+                        ),
+                        "type": "invalid_request_error",
+                    }
+                },
+            )
+        created = int(time.time())
+        model_name = body.get("model", served_model)
+        obj = "chat.completion" if chat else "text_completion"
+
+        if body.get("stream"):
+
+            async def gen():
+                try:
+                    first = True
+                    n_out = 0
+                    async for out in async_engine.generate(
+                        rid, prompt_tokens, params
+                    ):
+                        n_out = out.num_output_tokens
+                        if chat:
+                            delta = (
+                                {"role": "assistant", "content": out.text_delta}
+                                if first
+                                else {"content": out.text_delta}
+                            )
+                            choice = {
+                                "index": 0,
+                                "delta": delta,
+                                "finish_reason": (
+                                    out.finish_reason if out.finished else None
+                                ),
+                            }
+                        else:
+                            choice = {
+                                "index": 0,
+                                "text": out.text_delta,
+                                "finish_reason": (
+                                    out.finish_reason if out.finished else None
+                                ),
+                            }
+                        first = False
+                        chunk = {
+                            "id": rid,
+                            "object": obj + ".chunk" if chat else obj,
+                            "created": created,
+                            "model": model_name,
+                            "choices": [choice],
+                        }
+                        if out.finished:
+                            chunk["usage"] = {
+                                "prompt_tokens": out.num_prompt_tokens,
+                                "completion_tokens": n_out,
+                                "total_tokens": out.num_prompt_tokens + n_out,
+                            }
+                        yield f"data: {json.dumps(chunk)}\n\n".encode()
+                    yield b"data: [DONE]\n\n"
+                except asyncio.CancelledError:
+                    async_engine.abort(rid)
+                    raise
+
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        # non-streaming
+        text = ""
+        tokens = []
+        finish_reason = None
+        n_prompt = len(prompt_tokens)
+        kv_params_out = None
+        async for out in async_engine.generate(rid, prompt_tokens, params):
+            text += out.text_delta
+            tokens.extend(out.new_token_ids)
+            if out.finished:
+                finish_reason = out.finish_reason
+                n_prompt = out.num_prompt_tokens or n_prompt
+        resp = {
+            "id": rid,
+            "object": obj,
+            "created": created,
+            "model": model_name,
+            "choices": [
+                {
+                    "index": 0,
+                    "finish_reason": finish_reason,
+                    **(
+                        {
+                            "message": {
+                                "role": "assistant",
+                                "content": text,
+                            }
+                        }
+                        if chat
+                        else {"text": text}
+                    ),
+                }
+            ],
+            "usage": {
+                "prompt_tokens": n_prompt,
+                "completion_tokens": len(tokens),
+                "total_tokens": n_prompt + len(tokens),
+            },
+        }
+        if body.get("kv_transfer_params"):
+            # disaggregated prefill handshake (PD wiring: parallel/kv_transfer)
+            resp["kv_transfer_params"] = handle_kv_transfer_params(
+                app, body["kv_transfer_params"], rid
+            )
+        return JSONResponse(resp)
+
+    @app.post("/v1/completions")
+    async def completions(request: Request):
+        return await _run_completion(request, chat=False)
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request):
+        return await _run_completion(request, chat=True)
+
+    return app
+
+
+def handle_kv_transfer_params(app, kv_params: dict, request_id: str) -> dict:
+    """Echo the PD handshake shape (reference request.py:793-851). The
+    actual block transfer over RCCL/xGMI lives in parallel/kv_transfer.py;
+    in the HTTP server we advertise this engine as the KV source."""
+    cfg = app.state.engine.config
+    out = dict(kv_params)
+    if kv_params.get("do_remote_decode"):
+        out.update(
+            {
+                "do_remote_decode": False,
+                "do_remote_prefill": True,
+                "remote_engine_id": getattr(cfg, "engine_id", "engine-0"),
+                "remote_block_ids": [],
+                "remote_host": None,
+                "remote_port": getattr(cfg, "kv_transfer_port", 14001),
+            }
+        )
+    return out
+
+
+def main() -> None:
+    import uvicorn
+
+    ap = argparse.ArgumentParser(
+        description="MI355X-native OpenAI-compatible serving engine"
+    )
+    ap.add_argument("model", nargs="?", default="llama-3-8b")
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--served-model-name", default=None)
+    ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--dtype", default="bfloat16")
+    ap.add_argument("--max-num-seqs", type=int, default=256)
+    ap.add_argument("--max-num-batched-tokens", type=int, default=8192)
+    ap.add_argument("--gpu-memory-utilization", type=float, default=0.85)
+    ap.add_argument("--num-gpu-blocks", type=int, default=None)
+    ap.add_argument("--enable-prefix-caching", action="store_true",
+                    default=True)
+    ap.add_argument("--no-enable-prefix-caching", dest="enable_prefix_caching",
+                    action="store_false")
+    ap.add_argument("--enable-chunked-prefill", action="store_true",
+                    default=True)
+    ap.add_argument("--tensor-parallel-size", type=int, default=1)
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--device", default=None, help="cpu forces CPU mode")
+    ap.add_argument("--weights-path", default=None)
+    ap.add_argument("--tokenizer", default="synthetic")
+    ap.add_argument("--kv-controller-url", default=None,
+                    help="host:port of the KV controller for kvaware routing")
+    ap.add_argument("--advertise-url", default=None,
+                    help="this engine's URL as seen by the router")
+    ap.add_argument("--kv-role", default=None,
+                    choices=[None, "kv_producer", "kv_consumer"])
+    ap.add_argument("--cpu-offload-gb", type=float, default=0.0)
+    args = ap.parse_args()
+
+    from production_stack_amd.engine.config import ParallelConfig
+
+    cfg = EngineConfig(
+        model=args.model,
+        max_model_len=args.max_model_len,
+        seed=args.seed,
+        weights_path=args.weights_path,
+        tokenizer=args.tokenizer,
+        cache=CacheConfig(
+            num_gpu_blocks=args.num_gpu_blocks,
+            gpu_memory_utilization=args.gpu_memory_utilization,
+            enable_prefix_caching=args.enable_prefix_caching,
+            cpu_offload_gb=args.cpu_offload_gb,
+        ),
+        scheduler=SchedulerConfig(
+            max_num_seqs=args.max_num_seqs,
+            max_num_batched_tokens=args.max_num_batched_tokens,
+            enable_chunked_prefill=args.enable_chunked_prefill,
+        ),
+        parallel=ParallelConfig(
+            tensor_parallel_size=args.tensor_parallel_size,
+            kv_role=args.kv_role,
+        ),
+    )
+    cfg.kv_controller_url = args.kv_controller_url  # type: ignore[attr-defined]
+    cfg.advertise_url = (  # type: ignore[attr-defined]
+        args.advertise_url or f"http://127.0.0.1:{args.port}"
+    )
+    engine = LLMEngine(cfg, device=args.device)
+    served = args.served_model_name or args.model
+    app = build_server(engine, served)
+    logging.basicConfig(level=logging.INFO)
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

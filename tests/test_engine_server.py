@@ -1,0 +1,188 @@
+"""CPU tests of the engine's OpenAI HTTP server (tiny model, ASGI)."""
+
+import asyncio
+import json
+
+import httpx
+
+from production_stack_amd.engine.config import (
+    CacheConfig,
+    EngineConfig,
+    SchedulerConfig,
+)
+from production_stack_amd.engine.engine import LLMEngine
+from production_stack_amd.engine.server import build_server
+
+
+def make_app():
+    cfg = EngineConfig(
+        model="tiny-llama",
+        max_model_len=512,
+        cache=CacheConfig(num_gpu_blocks=256, block_size=16),
+        scheduler=SchedulerConfig(max_num_seqs=8, max_num_batched_tokens=256),
+    )
+    engine = LLMEngine(cfg, device="cpu")
+    return build_server(engine, served_model="tiny-llama")
+
+
+def with_server(fn):
+    async def go():
+        app = make_app()
+        async with httpx.ASGITransport(app=app) as transport:
+            # run startup/shutdown events manually
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://engine"
+                ) as client:
+                    await fn(client)
+
+    asyncio.run(go())
+
+
+def test_completion_non_streaming():
+    async def go(client):
+        r = await client.post(
+            "/v1/completions",
+            json={
+                "model": "tiny-llama",
+                "prompt": "hello world foo bar",
+                "max_tokens": 5,
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+            timeout=60,
+        )
+        assert r.status_code == 200, r.text
+        data = r.json()
+        assert data["usage"]["completion_tokens"] == 5
+        assert data["usage"]["prompt_tokens"] == 4
+        assert data["choices"][0]["text"]
+
+    with_server(go)
+
+
+def test_chat_streaming_sse_with_usage():
+    async def go(client):
+        chunks = []
+        async with client.stream(
+            "POST",
+            "/v1/chat/completions",
+            json={
+                "model": "tiny-llama",
+                "messages": [{"role": "user", "content": "hi there"}],
+                "max_tokens": 4,
+                "temperature": 0,
+                "ignore_eos": True,
+                "stream": True,
+            },
+            timeout=60,
+        ) as r:
+            assert r.status_code == 200
+            assert "text/event-stream" in r.headers["content-type"]
+            body = ""
+            async for text in r.aiter_text():
+                body += text
+        lines = [
+            line[6:] for line in body.splitlines() if line.startswith("data: ")
+        ]
+        assert lines[-1] == "[DONE]"
+        payloads = [json.loads(x) for x in lines[:-1]]
+        final = payloads[-1]
+        assert final["usage"]["completion_tokens"] == 4
+        assert payloads[0]["choices"][0]["delta"].get("role") == "assistant"
+
+    with_server(go)
+
+
+def test_models_tokenize_health_metrics():
+    async def go(client):
+        r = await client.get("/v1/models")
+        assert r.json()["data"][0]["id"] == "tiny-llama"
+        r = await client.post("/tokenize", json={"prompt": "a b c"})
+        assert r.json()["count"] == 3
+        r = await client.post(
+            "/detokenize", json={"tokens": [100, 200]}
+        )
+        assert "prompt" in r.json()
+        r = await client.get("/health")
+        assert r.status_code == 200
+        r = await client.get("/metrics")
+        assert "vllm:num_requests_running" in r.text
+        assert "vllm:gpu_cache_usage_perc" in r.text
+        assert "vllm:gpu_prefix_cache_hit_rate" in r.text
+
+    with_server(go)
+
+
+def test_sleep_wake_cycle():
+    async def go(client):
+        r = await client.get("/is_sleeping")
+        assert r.json()["is_sleeping"] is False
+        await client.post("/sleep", params={"level": 1})
+        r = await client.get("/is_sleeping")
+        assert r.json()["is_sleeping"] is True
+        await client.post("/wake_up")
+        r = await client.get("/is_sleeping")
+        assert r.json()["is_sleeping"] is False
+
+    with_server(go)
+
+
+def test_lora_load_unload_surface():
+    async def go(client):
+        r = await client.post(
+            "/v1/load_lora_adapter",
+            json={"lora_name": "ad1", "lora_path": "/tmp/x"},
+        )
+        assert r.status_code == 200
+        r = await client.get("/v1/models")
+        ids = {m["id"] for m in r.json()["data"]}
+        assert "ad1" in ids
+        card = [m for m in r.json()["data"] if m["id"] == "ad1"][0]
+        assert card["parent"] == "tiny-llama"
+        r = await client.post(
+            "/v1/unload_lora_adapter", json={"lora_name": "ad1"}
+        )
+        r = await client.get("/v1/models")
+        assert "ad1" not in {m["id"] for m in r.json()["data"]}
+
+    with_server(go)
+
+
+def test_prompt_too_long_400():
+    async def go(client):
+        r = await client.post(
+            "/v1/completions",
+            json={
+                "model": "tiny-llama",
+                "prompt": "x " * 600,
+                "max_tokens": 2,
+            },
+            timeout=60,
+        )
+        assert r.status_code == 400
+
+    with_server(go)
+
+
+def test_concurrent_streaming_requests():
+    async def go(client):
+        async def one(i):
+            r = await client.post(
+                "/v1/completions",
+                json={
+                    "model": "tiny-llama",
+                    "prompt": f"request {i} says hello",
+                    "max_tokens": 6,
+                    "temperature": 0,
+                    "ignore_eos": True,
+                },
+                timeout=60,
+            )
+            assert r.status_code == 200
+            return r.json()["usage"]["completion_tokens"]
+
+        results = await asyncio.gather(*(one(i) for i in range(6)))
+        assert all(n == 6 for n in results)
+
+    with_server(go)
