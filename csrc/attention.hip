@@ -20,12 +20,19 @@
 #include "ps_common.h"
 
 // ---------------------------------------------------------------------------
-// Decode: one query token per sequence.
-//   grid = (num_seqs, num_kv_heads), block = NWAVES * 64
+// Decode: one query token per sequence, flash-decode split-KV.
+//   grid = (num_seqs, num_kv_heads, num_splits), block = NWAVES * 64
+// With num_splits == 1 the result is written directly to `out`; otherwise
+// each split writes a partial (m, l, acc) to the fp32 workspace and a small
+// combine kernel merges the splits. Splitting exists because at decode batch
+// sizes of 16-64 seqs, seqs*kv_heads workgroups cannot fill 256 CUs and the
+// kernel runs at <10% of HBM bandwidth (measured, profiles/ run1).
 // ---------------------------------------------------------------------------
 template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES>
 __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     unsigned short* __restrict__ out,            // [S, QH, HEAD_DIM]
+    float* __restrict__ ws_acc,   // [S, QH, SPLITS, HD] (splits > 1)
+    float* __restrict__ ws_ml,    // [S, QH, SPLITS, 2]
     const unsigned short* __restrict__ q,        // [S, QH, HEAD_DIM]
     const unsigned short* __restrict__ k_cache,  // [NB, KH, BS, HD]
     const unsigned short* __restrict__ v_cache,  // [NB, KH, BS, HD]
@@ -65,8 +72,14 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
   }
 
   const int nblocks = (ctx + BLOCK_SIZE - 1) / BLOCK_SIZE;
+  // split-KV: this workgroup handles blocks [b_begin, b_end)
+  const int nsplit = gridDim.z;
+  const int split = blockIdx.z;
+  const int per_split = (nblocks + nsplit - 1) / nsplit;
+  const int b_begin = split * per_split;
+  const int b_end = min(nblocks, b_begin + per_split);
   const int* bt = block_tables + (long)seq * max_blocks;
-  for (int b = wave; b < nblocks; b += NWAVES) {
+  for (int b = b_begin + wave; b < b_end; b += NWAVES) {
     const long blk = bt[b];
     const unsigned short* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
     const unsigned short* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
@@ -153,14 +166,60 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
           A[j] = A[j] * c1 + sacc[wv][g][sl * 8 + j] * c2;
         M = mnew;
       }
-      const float inv = L > 0.f ? 1.f / L : 0.f;
-      ps_bf16x8 ov;
+      const long head = (long)seq * KH * GQ + (long)kvh * GQ + g;
+      if (nsplit == 1) {
+        const float inv = L > 0.f ? 1.f / L : 0.f;
+        ps_bf16x8 ov;
 #pragma unroll
-      for (int j = 0; j < 8; j++) ov[j] = ps_f32_to_bf16(A[j] * inv);
-      *(ps_bf16x8*)(out + ((long)seq * KH * GQ + (long)kvh * GQ + g) * D +
-                    sl * 8) = ov;
+        for (int j = 0; j < 8; j++) ov[j] = ps_f32_to_bf16(A[j] * inv);
+        *(ps_bf16x8*)(out + head * D + sl * 8) = ov;
+      } else {
+        float* wa = ws_acc + (head * nsplit + split) * D + sl * 8;
+#pragma unroll
+        for (int j = 0; j < 8; j++) wa[j] = A[j];
+        if (sl == 0) {
+          float* wm = ws_ml + (head * nsplit + split) * 2;
+          wm[0] = M;
+          wm[1] = L;
+        }
+      }
     }
   }
+}
+
+// Combine the per-split partials: grid = (S * QH), block = HEAD_DIM/8 lanes
+// padded to a wave.
+template <int HEAD_DIM>
+__global__ __launch_bounds__(64) void paged_attn_combine_kernel(
+    unsigned short* __restrict__ out,   // [S*QH, HD]
+    const float* __restrict__ ws_acc,   // [S*QH, SPLITS, HD]
+    const float* __restrict__ ws_ml,    // [S*QH, SPLITS, 2]
+    int nsplit) {
+  constexpr int D = HEAD_DIM;
+  constexpr int LPG = D / 8;
+  const long head = blockIdx.x;
+  const int sl = threadIdx.x;
+  if (sl >= LPG) return;
+  float M = PS_NEG_INF, L = 0.f, A[8];
+#pragma unroll
+  for (int j = 0; j < 8; j++) A[j] = 0.f;
+  for (int s = 0; s < nsplit; s++) {
+    const float mw = ws_ml[(head * nsplit + s) * 2 + 0];
+    const float lw = ws_ml[(head * nsplit + s) * 2 + 1];
+    const float mnew = fmaxf(M, mw);
+    const float c1 = __expf(M - mnew);
+    const float c2 = __expf(mw - mnew);
+    const float* wa = ws_acc + (head * nsplit + s) * D + sl * 8;
+#pragma unroll
+    for (int j = 0; j < 8; j++) A[j] = A[j] * c1 + wa[j] * c2;
+    L = L * c1 + lw * c2;
+    M = mnew;
+  }
+  const float inv = L > 0.f ? 1.f / L : 0.f;
+  ps_bf16x8 ov;
+#pragma unroll
+  for (int j = 0; j < 8; j++) ov[j] = ps_f32_to_bf16(A[j] * inv);
+  *(ps_bf16x8*)(out + head * D + sl * 8) = ov;
 }
 
 // ---------------------------------------------------------------------------
@@ -357,19 +416,28 @@ __global__ void greedy_sample_kernel(long* __restrict__ out,  // [R]
 // ---------------------------------------------------------------------------
 extern "C" {
 
-int ps_paged_attn_decode(void* out, const void* q, const void* k_cache,
-                         const void* v_cache, const void* block_tables,
-                         const void* seq_lens, int num_seqs, int max_blocks,
-                         float scale, int KH, int GQ, int head_dim,
-                         int block_size, hipStream_t stream) {
-  dim3 grid(num_seqs, KH);
+int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
+                         const void* k_cache, const void* v_cache,
+                         const void* block_tables, const void* seq_lens,
+                         int num_seqs, int max_blocks, float scale, int KH,
+                         int GQ, int head_dim, int block_size, int num_splits,
+                         hipStream_t stream) {
+  dim3 grid(num_seqs, KH, num_splits);
   constexpr int NW = 4;
   dim3 block(NW * 64);
 #define PS_DISPATCH_DECODE(HD, G, BS)                                        \
-  paged_attn_decode_kernel<HD, G, BS, NW><<<grid, block, 0, stream>>>(       \
-      (unsigned short*)out, (const unsigned short*)q,                        \
-      (const unsigned short*)k_cache, (const unsigned short*)v_cache,        \
-      (const int*)block_tables, (const int*)seq_lens, max_blocks, scale, KH)
+  do {                                                                       \
+    paged_attn_decode_kernel<HD, G, BS, NW><<<grid, block, 0, stream>>>(     \
+        (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,                 \
+        (const unsigned short*)q, (const unsigned short*)k_cache,            \
+        (const unsigned short*)v_cache, (const int*)block_tables,            \
+        (const int*)seq_lens, max_blocks, scale, KH);                        \
+    if (num_splits > 1)                                                      \
+      paged_attn_combine_kernel<HD>                                          \
+          <<<dim3(num_seqs * KH * G), 64, 0, stream>>>(                      \
+              (unsigned short*)out, (const float*)ws_acc,                    \
+              (const float*)ws_ml, num_splits);                              \
+  } while (0)
   if (head_dim == 128 && block_size == 16) {
     switch (GQ) {
       case 1: PS_DISPATCH_DECODE(128, 1, 16); return 0;

@@ -8,6 +8,8 @@
 
 #include <c10/hip/HIPStream.h>
 
+#include <algorithm>
+
 extern "C" {
 void ps_rms_norm(void* out, const void* x, const void* w, float eps, long T,
                  int D, hipStream_t stream);
@@ -17,11 +19,12 @@ void ps_silu_and_mul(void* out, const void* x, long T, int D,
                      hipStream_t stream);
 void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
              long T, int QH, int KH, int HD, int ROT, hipStream_t stream);
-int ps_paged_attn_decode(void* out, const void* q, const void* k_cache,
-                         const void* v_cache, const void* block_tables,
-                         const void* seq_lens, int num_seqs, int max_blocks,
-                         float scale, int KH, int GQ, int head_dim,
-                         int block_size, hipStream_t stream);
+int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
+                         const void* k_cache, const void* v_cache,
+                         const void* block_tables, const void* seq_lens,
+                         int num_seqs, int max_blocks, float scale, int KH,
+                         int GQ, int head_dim, int block_size, int num_splits,
+                         hipStream_t stream);
 int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* v_cache, const void* block_tables,
                           const void* token_seq, const void* token_pos,
@@ -101,7 +104,7 @@ void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
 
 void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
-                       at::Tensor seq_lens, double scale) {
+                       at::Tensor seq_lens, double scale, int64_t num_splits) {
   CHECK_GPU_BF16(out);
   CHECK_GPU_BF16(q);
   CHECK_GPU_BF16(k_cache);
@@ -116,10 +119,26 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   TORCH_CHECK(QH % KH == 0, "GQA group mismatch");
   const int GQ = QH / KH;
   const int max_blocks = (int)block_tables.size(1);
+  if (num_splits <= 0) {
+    // fill ~3 workgroups per CU, bounded by useful split granularity
+    const int target = 2048;
+    num_splits = std::min<int64_t>(
+        32, std::max<int64_t>(1, target / std::max(1, S * KH)));
+  }
+  at::Tensor ws_acc, ws_ml;
+  void *acc_p = nullptr, *ml_p = nullptr;
+  if (num_splits > 1) {
+    auto opts = q.options().dtype(at::kFloat);
+    ws_acc = at::empty({(long)S * QH * num_splits * HD}, opts);
+    ws_ml = at::empty({(long)S * QH * num_splits * 2}, opts);
+    acc_p = ws_acc.data_ptr();
+    ml_p = ws_ml.data_ptr();
+  }
   int rc = ps_paged_attn_decode(
-      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-      block_tables.data_ptr(), seq_lens.data_ptr(), S, max_blocks,
-      (float)scale, KH, GQ, HD, BS, current_stream());
+      out.data_ptr(), acc_p, ml_p, q.data_ptr(), k_cache.data_ptr(),
+      v_cache.data_ptr(), block_tables.data_ptr(), seq_lens.data_ptr(), S,
+      max_blocks, (float)scale, KH, GQ, HD, BS, (int)num_splits,
+      current_stream());
   TORCH_CHECK(rc == 0, "unsupported decode config: head_dim=", HD,
               " block_size=", BS, " gqa=", GQ);
 }
@@ -186,7 +205,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rotary_embedding", &rotary_embedding,
         "Neox-style rotary embedding (bf16, in-place)");
   m.def("paged_attn_decode", &paged_attn_decode,
-        "Paged attention, decode phase (bf16 KV)");
+        "Paged attention, decode phase (bf16 KV, split-KV)",
+        pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k_cache"),
+        pybind11::arg("v_cache"), pybind11::arg("block_tables"),
+        pybind11::arg("seq_lens"), pybind11::arg("scale"),
+        pybind11::arg("num_splits") = 0);
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged attention, chunked prefill (bf16 KV)");
   m.def("reshape_and_cache", &reshape_and_cache,

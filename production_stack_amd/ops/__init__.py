@@ -104,7 +104,7 @@ def paged_attn_decode(
         _require_ext()
         out = torch.empty_like(q)
         _C.paged_attn_decode(
-            out, q, k_cache, v_cache, block_tables, seq_lens, scale
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale, 0
         )
         return out
     return reference.paged_attn_decode(

@@ -114,6 +114,35 @@ def test_paged_attn_decode(qh, kh, hd):
     _close(got, want)
 
 
+@pytest.mark.parametrize("num_splits", [1, 2, 4, 8, 32])
+def test_paged_attn_decode_split_kv(num_splits):
+    """All split factors must agree with the reference (incl. splits that
+    exceed the block count of short sequences)."""
+    from production_stack_amd import _C
+
+    torch.manual_seed(3)
+    qh, kh, hd, bs = 32, 8, 128, 16
+    S = 4
+    seq_lens = torch.tensor([5, 700, 64, 333], dtype=torch.int32)
+    max_blocks = 44
+    nb = S * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    block_tables = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+        S, max_blocks
+    )
+    q = torch.randn((S, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_decode(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), block_tables, seq_lens,
+        1.0 / hd ** 0.5,
+    )
+    out = torch.empty_like(q)
+    _C.paged_attn_decode(
+        out, q, k_cache, v_cache, block_tables.cuda(), seq_lens.cuda(),
+        1.0 / hd ** 0.5, num_splits,
+    )
+    _close(out, want)
+
+
 @pytest.mark.parametrize("qh,kh,hd", [(32, 8, 128), (4, 4, 64)])
 def test_paged_attn_prefill(qh, kh, hd):
     torch.manual_seed(2)
