@@ -60,6 +60,9 @@ class LLMEngine:
         self.scheduler = Scheduler(
             config.scheduler, self.block_manager, config.max_model_len
         )
+        self.runner.capture_decode_graphs(
+            min(config.scheduler.max_num_seqs, 256)
+        )
         self.stats = EngineStats()
         self._sleeping = False
 

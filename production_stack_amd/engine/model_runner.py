@@ -40,6 +40,7 @@ class ModelRunner:
             self.model.random_init(config.seed)
         self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
         self.num_blocks = 0
+        self.graphs = None
         self._generator = torch.Generator(device="cpu").manual_seed(
             config.seed + 12345
         )
@@ -75,6 +76,29 @@ class ModelRunner:
             )
             v = torch.zeros_like(k)
             self.kv_caches.append((k, v))
+
+    def capture_decode_graphs(self, max_batch: int) -> None:
+        """hipGraph-capture decode-only steps (see graph_runner.py)."""
+        if self.device.type != "cuda" or self.config.enforce_eager:
+            return
+        from production_stack_amd.engine.graph_runner import DecodeGraphRunner
+
+        max_blocks = (
+            self.config.max_model_len + self.config.cache.block_size - 1
+        ) // self.config.cache.block_size
+        try:
+            runner = DecodeGraphRunner(
+                self.model, self.kv_caches, max_batch, max_blocks, self.device
+            )
+            runner.capture_all()
+            self.graphs = runner
+        except Exception:
+            import logging
+
+            logging.getLogger("engine.runner").exception(
+                "hipGraph capture failed; continuing in eager mode"
+            )
+            self.graphs = None
 
     # ------------------------------------------------------------------
     def prepare(
@@ -180,10 +204,51 @@ class ModelRunner:
 
     # ------------------------------------------------------------------
     @torch.no_grad()
+    def _execute_decode_graph(
+        self, out: SchedulerOutput, bm: BlockManager
+    ) -> Optional[Dict[str, int]]:
+        """Fast path: decode-only step via hipGraph replay."""
+        import numpy as np
+
+        scheduled = out.scheduled
+        n = len(scheduled)
+        if (
+            self.graphs is None
+            or n == 0
+            or self.graphs.bucket_for(n) is None
+            or any(not s.is_decode for s in scheduled)
+        ):
+            return None
+        bs = bm.block_size
+        tokens = np.empty(n, dtype=np.int64)
+        positions = np.empty(n, dtype=np.int32)
+        slots = np.empty(n, dtype=np.int64)
+        seq_lens = np.empty(n, dtype=np.int32)
+        tables: List[List[int]] = []
+        seqs: List[Sequence] = []
+        for i, ss in enumerate(scheduled):
+            seq = ss.seq
+            pos = seq.num_computed
+            tokens[i] = seq.token_ids()[pos]
+            positions[i] = pos
+            slots[i] = seq.block_table[pos // bs] * bs + pos % bs
+            seq_lens[i] = pos + 1
+            tables.append(seq.block_table)
+            seqs.append(seq)
+        logits = self.graphs.run(tokens, positions, slots, seq_lens, tables)
+        sampled = self.sample(logits, seqs)
+        return {
+            seq.request_id: int(tok) for seq, tok in zip(seqs, sampled)
+        }
+
+    @torch.no_grad()
     def execute(
         self, out: SchedulerOutput, bm: BlockManager
     ) -> Dict[str, int]:
         """Run one step; returns request_id -> sampled token."""
+        fast = self._execute_decode_graph(out, bm)
+        if fast is not None:
+            return fast
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
         if token_t.numel() == 0:
             return {}
