@@ -25,6 +25,12 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
                          hipStream_t stream);
+int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
+                               const void* v_cache, const void* block_tables,
+                               const void* tile_info, int num_tiles,
+                               int num_q_heads, int max_blocks, float scale,
+                               int KH, int GQ, int head_dim,
+                               hipStream_t stream);
 int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* v_cache, const void* block_tables,
                           const void* token_seq, const void* token_pos,
@@ -170,6 +176,30 @@ void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
               " block_size=", BS);
 }
 
+void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
+                             at::Tensor k_cache, at::Tensor v_cache,
+                             at::Tensor block_tables, at::Tensor tile_info,
+                             double scale) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_BF16(q);
+  CHECK_GPU_BF16(k_cache);
+  CHECK_GPU_BF16(v_cache);
+  CHECK_GPU_DTYPE(block_tables, at::kInt);
+  CHECK_GPU_DTYPE(tile_info, at::kInt);
+  const int QH = (int)q.size(1);
+  const int HD = (int)q.size(2);
+  const int KH = (int)k_cache.size(1);
+  TORCH_CHECK(QH % KH == 0, "GQA group mismatch");
+  TORCH_CHECK(k_cache.size(2) == 16, "mfma prefill needs block_size 16");
+  const int GQ = QH / KH;
+  const int NT = (int)tile_info.size(0);
+  int rc = ps_paged_attn_prefill_mfma(
+      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+      block_tables.data_ptr(), tile_info.data_ptr(), NT, QH,
+      (int)block_tables.size(1), (float)scale, KH, GQ, HD, current_stream());
+  TORCH_CHECK(rc == 0, "unsupported mfma prefill config: head_dim=", HD);
+}
+
 void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor slot_mapping) {
   CHECK_GPU_BF16(k);
@@ -212,6 +242,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("num_splits") = 0);
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged attention, chunked prefill (bf16 KV)");
+  m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
+        "Paged attention, chunked prefill via MFMA tiles (head_dim 128)");
   m.def("reshape_and_cache", &reshape_and_cache,
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");

@@ -115,15 +115,23 @@ class ModelRunner:
         p_token_seq: List[int] = []
         p_token_pos: List[int] = []
         p_tables: List[List[int]] = []
+        p_tiles: List[List[int]] = []  # (seq_row, q_tok0, q_pos0, n_rows)
         sample_rows: List[int] = []
         sample_seqs: List[Sequence] = []
 
+        TILE = 64
         for row, ss in enumerate(prefills):
             seq = ss.seq
             all_ids = seq.token_ids()
             start = seq.num_computed
             end = start + ss.num_tokens
             p_tables.append(seq.block_table)
+            q_flat0 = len(tokens)
+            for t0 in range(0, ss.num_tokens, TILE):
+                p_tiles.append(
+                    [row, q_flat0 + t0, start + t0,
+                     min(TILE, ss.num_tokens - t0)]
+                )
             for pos in range(start, end):
                 tokens.append(all_ids[pos])
                 positions.append(pos)
@@ -193,6 +201,13 @@ class ModelRunner:
                 else None
             ),
             decode_block_tables=pad_tables(d_tables),
+            prefill_tiles=(
+                torch.tensor(p_tiles, dtype=torch.int32).to(
+                    dev, non_blocking=True
+                )
+                if p_tiles
+                else None
+            ),
         )
         token_t = torch.tensor(tokens, dtype=torch.long).to(
             dev, non_blocking=True

@@ -42,6 +42,7 @@ class BatchMeta:
         num_decode_seqs: int,
         decode_seq_lens: Optional[torch.Tensor],  # [Sd] int32
         decode_block_tables: Optional[torch.Tensor],  # [Sd, MB] int32
+        prefill_tiles: Optional[torch.Tensor] = None,  # [NT, 4] int32
     ) -> None:
         self.positions = positions
         self.slot_mapping = slot_mapping
@@ -52,6 +53,7 @@ class BatchMeta:
         self.num_decode_seqs = num_decode_seqs
         self.decode_seq_lens = decode_seq_lens
         self.decode_block_tables = decode_block_tables
+        self.prefill_tiles = prefill_tiles
 
 
 def build_cos_sin_cache(
@@ -132,17 +134,33 @@ class LlamaLayer(nn.Module):
         outs: List[torch.Tensor] = []
         tp = meta.num_prefill_tokens
         if tp > 0:
-            outs.append(
-                ops.paged_attn_prefill(
-                    qh[:tp].contiguous(),
-                    k_cache,
-                    v_cache,
-                    meta.prefill_block_tables,
-                    meta.prefill_token_seq,
-                    meta.prefill_token_pos,
-                    self.scale,
+            if (
+                qh.is_cuda
+                and self.head_dim == 128
+                and meta.prefill_tiles is not None
+            ):
+                outs.append(
+                    ops.paged_attn_prefill_mfma(
+                        qh[:tp].contiguous(),
+                        k_cache,
+                        v_cache,
+                        meta.prefill_block_tables,
+                        meta.prefill_tiles,
+                        self.scale,
+                    )
                 )
-            )
+            else:
+                outs.append(
+                    ops.paged_attn_prefill(
+                        qh[:tp].contiguous(),
+                        k_cache,
+                        v_cache,
+                        meta.prefill_block_tables,
+                        meta.prefill_token_seq,
+                        meta.prefill_token_pos,
+                        self.scale,
+                    )
+                )
         if meta.num_decode_seqs > 0:
             outs.append(
                 ops.paged_attn_decode(

@@ -133,6 +133,23 @@ def paged_attn_prefill(
     )
 
 
+def paged_attn_prefill_mfma(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    tile_info: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    """MFMA-tiled chunked prefill (GPU, head_dim 128 only)."""
+    _require_ext()
+    out = torch.empty_like(q)
+    _C.paged_attn_prefill_mfma(
+        out, q, k_cache, v_cache, block_tables, tile_info, scale
+    )
+    return out
+
+
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     if logits.is_cuda:
         _require_ext()

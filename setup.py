@@ -20,6 +20,7 @@ ext = CUDAExtension(
         "csrc/bindings.cpp",
         "csrc/norm_act_rope.hip",
         "csrc/attention.hip",
+        "csrc/prefill_mfma.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
