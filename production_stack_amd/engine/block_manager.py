@@ -37,6 +37,8 @@ class BlockManager:
         self._tail_hash: Dict[str, Optional[int]] = {}
         self.prefix_queries = 0
         self.prefix_hits = 0
+        # optional host-DRAM offload tier (kvpool.offload.HostKVPool)
+        self.offload_pool = None
 
     # ------------------------------------------------------------------
     @property
@@ -102,6 +104,23 @@ class BlockManager:
                 self.prefix_queries += 1
                 blk = self.cached.get(h)
                 if blk is None:
+                    # HBM miss: try the host-DRAM offload tier
+                    pool = self.offload_pool
+                    if pool is not None and pool.has(h):
+                        nb = self._pop_block()
+                        if nb is None:
+                            break
+                        if pool.restore(h, nb):
+                            self.ref_count[nb] += 1
+                            self.cached[h] = nb
+                            self.block_hash[nb] = h
+                            table.append(nb)
+                            prev_hash = h
+                            matched += 1
+                            self.prefix_hits += 1
+                            continue
+                        # restore raced an eviction: hand the block back
+                        self.free.appendleft(nb)
                     break
                 self.prefix_hits += 1
                 if self.ref_count[blk] == 0:
@@ -153,6 +172,8 @@ class BlockManager:
             if h not in self.cached and self.block_hash[blk] is None:
                 self.cached[h] = blk
                 self.block_hash[blk] = h
+                if self.offload_pool is not None:
+                    self.offload_pool.offload(h, blk)
             prev = h
         seq._registered_full = n_full  # type: ignore[attr-defined]
         self._tail_hash[seq.request_id] = prev

@@ -60,6 +60,18 @@ class LLMEngine:
         self.scheduler = Scheduler(
             config.scheduler, self.block_manager, config.max_model_len
         )
+        if config.cache.cpu_offload_gb > 0:
+            from production_stack_amd.kvpool.offload import HostKVPool
+
+            self.host_pool = HostKVPool(
+                self.runner.kv_caches,
+                config.cache.block_size,
+                config.cache.cpu_offload_gb,
+                self.device,
+            )
+            self.block_manager.offload_pool = self.host_pool
+        else:
+            self.host_pool = None
         self.runner.capture_decode_graphs(
             min(config.scheduler.max_num_seqs, 256)
         )
@@ -95,8 +107,10 @@ class LLMEngine:
     # ------------------------------------------------------------------
     def step(self) -> List[RequestOutput]:
         out = self.scheduler.schedule()
-        if out.is_empty:
+        if out.is_empty and not out.capacity_stopped:
             return []
+        if self.host_pool is not None:
+            self.host_pool.make_compute_wait()
         sampled = self.runner.execute(out, self.block_manager)
         finished = self.scheduler.on_step_done(
             out, sampled, self.model_cfg.eos_token_id
@@ -182,6 +196,7 @@ class LLMEngine:
             "gpu_prefix_cache_queries_total": float(bm.prefix_queries),
             "prompt_tokens_total": float(self.stats.prompt_tokens),
             "generation_tokens_total": float(self.stats.generation_tokens),
+            **(self.host_pool.metrics() if self.host_pool else {}),
         }
 
     # ---- sleep / wake (reference request.py:1041-1128 parity) ----------
