@@ -61,3 +61,37 @@ def load_safetensors(model, path: str) -> None:
             layer.post_attn_norm.copy_(
                 get(pre + "post_attention_layernorm.weight")
             )
+
+
+def save_hf_safetensors(model, path: str) -> None:
+    """Export the fused-weight model to HF-format safetensors (inverse of
+    load_safetensors; used by tests and for interchange)."""
+    from safetensors.torch import save_file
+
+    os.makedirs(path, exist_ok=True)
+    cfg = model.cfg
+    state = {
+        "model.embed_tokens.weight": model.embed.detach().cpu(),
+        "model.norm.weight": model.final_norm.detach().cpu(),
+        "lm_head.weight": model.lm_head.detach().cpu(),
+    }
+    for i, layer in enumerate(model.layers):
+        pre = f"model.layers.{i}."
+        qs = layer.q_heads * layer.head_dim
+        kvs = layer.kv_heads * layer.head_dim
+        qkv = layer.qkv_proj.detach().cpu()
+        state[pre + "self_attn.q_proj.weight"] = qkv[:qs].clone()
+        state[pre + "self_attn.k_proj.weight"] = qkv[qs : qs + kvs].clone()
+        state[pre + "self_attn.v_proj.weight"] = qkv[qs + kvs :].clone()
+        state[pre + "self_attn.o_proj.weight"] = layer.o_proj.detach().cpu()
+        gu = layer.gate_up_proj.detach().cpu()
+        state[pre + "mlp.gate_proj.weight"] = gu[: layer.inter].clone()
+        state[pre + "mlp.up_proj.weight"] = gu[layer.inter :].clone()
+        state[pre + "mlp.down_proj.weight"] = layer.down_proj.detach().cpu()
+        state[pre + "input_layernorm.weight"] = (
+            layer.input_norm.detach().cpu()
+        )
+        state[pre + "post_attention_layernorm.weight"] = (
+            layer.post_attn_norm.detach().cpu()
+        )
+    save_file(state, os.path.join(path, "model.safetensors"))
