@@ -1,0 +1,263 @@
+"""Disaggregated-prefill KV block transfer (prefill -> decode engines).
+
+Reference behaviour being reimplemented: the NIXL/UCX KV transfer LMCache
+performs between kv_producer and kv_consumer pods, driven by the router's
+kv_transfer_params handshake (reference request.py:733-935).
+
+MI355X-native design:
+  * transport = torch.distributed point-to-point send/recv. On the GPU box
+    backend "nccl" IS RCCL, so block payloads move GPU-to-GPU over xGMI
+    without staging through host memory; CPU tests use gloo.
+  * control = a tiny msgpack-over-TCP side channel per engine (the
+    equivalent of NIXL's side channel / LMCACHE_NIXL_* ports): the decode
+    engine asks the prefill engine to push a request's blocks.
+  * decode-side integration reuses the prefix cache: pulled blocks are
+    chain-hash-registered, so the subsequent add_request() sees them as
+    prefix hits and computes only the prompt tail. No special-case decode
+    path exists anywhere in the scheduler.
+  * prefill-side retention: a finished prefill's blocks get an extra ref
+    until pulled (or a timeout), so the scheduler can free the sequence.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import logging
+import threading
+import time
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from production_stack_amd.kvpool.protocol import recv_msg, send_msg
+
+logger = logging.getLogger("parallel.kv_transfer")
+
+RETAIN_SECONDS = 120.0
+
+
+class KVTransferService:
+    def __init__(
+        self,
+        engine,
+        kv_rank: int,
+        kv_world: int,
+        master_port: int = 14500,
+        side_port: int = 14001,
+        host: str = "127.0.0.1",
+        backend: Optional[str] = None,
+    ) -> None:
+        self.engine = engine
+        self.kv_rank = kv_rank
+        self.kv_world = kv_world
+        self.side_port = side_port
+        self.host = host
+        if backend is None:
+            backend = "nccl" if engine.device.type == "cuda" else "gloo"
+        self.backend = backend
+        self.group = dist.init_process_group(
+            backend=backend,
+            init_method=f"tcp://127.0.0.1:{master_port}",
+            world_size=kv_world,
+            rank=kv_rank,
+            group_name="kv_transfer",
+        ) if not dist.is_initialized() else None
+        # pending prefilled requests: request_id -> (block_ids, n_tokens, ts)
+        self.pending: Dict[str, Tuple[List[int], int, float]] = {}
+        self._lock = threading.Lock()
+        self._send_lock = threading.Lock()
+        self._server: Optional[asyncio.AbstractServer] = None
+
+    # ------------------------------------------------------------------
+    def _caches(self):
+        return self.engine.runner.kv_caches
+
+    def _block_shape(self):
+        k0 = self._caches()[0][0]
+        return len(self._caches()), k0.shape[1] * k0.shape[2] * k0.shape[3]
+
+    def pack_blocks(self, block_ids: List[int]) -> torch.Tensor:
+        layers, elems = self._block_shape()
+        idx = torch.tensor(block_ids, dtype=torch.long,
+                           device=self.engine.device)
+        pack = torch.empty(
+            (len(block_ids), layers, 2, elems),
+            dtype=torch.bfloat16,
+            device=self.engine.device,
+        )
+        for li, (kc, vc) in enumerate(self._caches()):
+            pack[:, li, 0] = kc.index_select(0, idx).flatten(1)
+            pack[:, li, 1] = vc.index_select(0, idx).flatten(1)
+        return pack
+
+    def scatter_blocks(
+        self, pack: torch.Tensor, block_ids: List[int]
+    ) -> None:
+        k0 = self._caches()[0][0]
+        kh, bs, hd = k0.shape[1], k0.shape[2], k0.shape[3]
+        idx = torch.tensor(block_ids, dtype=torch.long,
+                           device=self.engine.device)
+        for li, (kc, vc) in enumerate(self._caches()):
+            kc.index_copy_(
+                0, idx, pack[:, li, 0].view(-1, kh, bs, hd)
+            )
+            vc.index_copy_(
+                0, idx, pack[:, li, 1].view(-1, kh, bs, hd)
+            )
+
+    # ---- prefill (kv_producer) side -----------------------------------
+    def register_prefilled(self, request_id: str, seq) -> Dict:
+        """Pin a finished prefill's blocks and describe them for the
+        kv_transfer_params response."""
+        bm = self.engine.block_manager
+        n_full = seq.num_tokens // bm.block_size
+        block_ids = list(seq.block_table[:n_full])
+        with self._lock:
+            for b in block_ids:
+                bm.ref_count[b] += 1
+            self.pending[request_id] = (block_ids, seq.num_tokens, time.time())
+        return {
+            "do_remote_decode": False,
+            "do_remote_prefill": True,
+            "remote_engine_id": self.kv_rank,
+            "remote_block_ids": block_ids,
+            "remote_host": self.host,
+            "remote_port": self.side_port,
+        }
+
+    def _release(self, request_id: str) -> None:
+        bm = self.engine.block_manager
+        with self._lock:
+            entry = self.pending.pop(request_id, None)
+            if entry is None:
+                return
+            for b in entry[0]:
+                bm._release_block(b)
+
+    def gc_pending(self) -> None:
+        now = time.time()
+        for rid, (_, _, ts) in list(self.pending.items()):
+            if now - ts > RETAIN_SECONDS:
+                self._release(rid)
+
+    async def _serve_client(self, reader, writer) -> None:
+        try:
+            while True:
+                msg = await recv_msg(reader)
+                if msg.get("type") != "pull":
+                    await send_msg(writer, {"ok": False})
+                    continue
+                rid = msg["request_id"]
+                dst = int(msg["dst_rank"])
+                with self._lock:
+                    entry = self.pending.get(rid)
+                if entry is None:
+                    await send_msg(writer, {"ok": False,
+                                            "error": "unknown request"})
+                    continue
+                block_ids, n_tokens, _ = entry
+                await send_msg(
+                    writer,
+                    {"ok": True, "n_blocks": len(block_ids),
+                     "n_tokens": n_tokens},
+                )
+                pack = self.pack_blocks(block_ids)
+                await asyncio.to_thread(self._send_pack, pack, dst)
+                self._release(rid)
+                await send_msg(writer, {"ok": True, "sent": True})
+        except (asyncio.IncompleteReadError, ConnectionError):
+            pass
+        finally:
+            writer.close()
+
+    def _send_pack(self, pack: torch.Tensor, dst: int) -> None:
+        with self._send_lock:
+            dist.send(pack, dst=dst)
+
+    async def start_side_channel(self) -> None:
+        self._server = await asyncio.start_server(
+            self._serve_client, "0.0.0.0", self.side_port
+        )
+        logger.info(
+            "KV transfer side channel on :%d (rank %d, %s)",
+            self.side_port, self.kv_rank, self.backend,
+        )
+
+    async def stop(self) -> None:
+        if self._server is not None:
+            self._server.close()
+            await self._server.wait_closed()
+
+    # ---- decode (kv_consumer) side ------------------------------------
+    async def pull_into_prefix_cache(
+        self,
+        request_id: str,
+        prompt_token_ids: List[int],
+        remote_host: str,
+        remote_port: int,
+        remote_rank: int,
+    ) -> int:
+        """Pull the prompt's full blocks and adopt them into the local
+        prefix cache. Returns the number of adopted tokens."""
+        bm = self.engine.block_manager
+        bs = bm.block_size
+        n_full = len(prompt_token_ids) // bs
+        if n_full * bs == len(prompt_token_ids):
+            n_full -= 1  # last token must be recomputed locally
+        if n_full <= 0:
+            return 0
+        reader, writer = await asyncio.open_connection(
+            remote_host, remote_port
+        )
+        try:
+            await send_msg(
+                writer,
+                {"type": "pull", "request_id": request_id,
+                 "dst_rank": self.kv_rank},
+            )
+            head = await recv_msg(reader)
+            if not head.get("ok"):
+                return 0
+            n_blocks = min(int(head["n_blocks"]), n_full)
+            if n_blocks <= 0:
+                return 0
+            # local destination blocks
+            dst_blocks = []
+            for _ in range(n_blocks):
+                b = bm._pop_block()
+                if b is None:
+                    break
+                dst_blocks.append(b)
+            layers, elems = self._block_shape()
+            pack = torch.empty(
+                (int(head["n_blocks"]), layers, 2, elems),
+                dtype=torch.bfloat16,
+                device=self.engine.device,
+            )
+            await asyncio.to_thread(self._recv_pack, pack, remote_rank)
+            await recv_msg(reader)  # sent ack
+            if not dst_blocks:
+                return 0
+            self.scatter_blocks(pack[: len(dst_blocks)], dst_blocks)
+            # adopt into the prefix cache
+            prev = None
+            adopted = 0
+            for i, blk in enumerate(dst_blocks):
+                h = bm.chain_hash(
+                    prev, tuple(prompt_token_ids[i * bs : (i + 1) * bs])
+                )
+                if h not in bm.cached:
+                    bm.cached[h] = blk
+                    bm.block_hash[blk] = h
+                    bm.evictable[blk] = None  # ref 0, contents valid
+                else:
+                    bm.free.append(blk)
+                prev = h
+                adopted += 1
+            return adopted * bs
+        finally:
+            writer.close()
+
+    def _recv_pack(self, pack: torch.Tensor, src: int) -> None:
+        dist.recv(pack, src=src)
