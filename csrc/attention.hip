@@ -38,7 +38,7 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     const unsigned short* __restrict__ v_cache,  // [NB, KH, BS, HD]
     const int* __restrict__ block_tables,        // [S, max_blocks]
     const int* __restrict__ seq_lens,            // [S]
-    int max_blocks, float scale, int KH) {
+    int max_blocks, float scale, int KH, long q_stride) {
   constexpr int D = HEAD_DIM;
   constexpr int LPG = D / 8;       // lanes per token sub-group (16 @ D=128)
   constexpr int TPW = 64 / LPG;    // tokens per wave per iteration (4)
@@ -54,7 +54,7 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
 
   // Q for the whole GQA group, pre-scaled, 8 dims per lane per head.
   float qf[GQ][8];
-  const unsigned short* qbase = q + ((long)seq * KH * GQ + (long)kvh * GQ) * D;
+  const unsigned short* qbase = q + (long)seq * q_stride + (long)kvh * GQ * D;
 #pragma unroll
   for (int g = 0; g < GQ; g++) {
     ps_bf16x8 qv = *(const ps_bf16x8*)(qbase + g * D + sl * 8);
@@ -238,7 +238,7 @@ __global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ token_seq,     // [T] sequence index per q token
     const int* __restrict__ token_pos,     // [T] absolute position per q token
-    int max_blocks, float scale, int KH, int GQ) {
+    int max_blocks, float scale, int KH, int GQ, long q_stride) {
   constexpr int D = HEAD_DIM;
   constexpr int LPG = D / 8;
   constexpr int TPW = 64 / LPG;
@@ -257,7 +257,7 @@ __global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
   float qf[8];
   {
     ps_bf16x8 qv =
-        *(const ps_bf16x8*)(q + ((long)tokid * QH + qh) * D + sl * 8);
+        *(const ps_bf16x8*)(q + (long)tokid * q_stride + (long)qh * D + sl * 8);
 #pragma unroll
     for (int j = 0; j < 8; j++) qf[j] = ps_bf16_to_f32(qv[j]) * scale;
   }
@@ -421,7 +421,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         hipStream_t stream) {
+                         long q_stride, hipStream_t stream) {
   dim3 grid(num_seqs, KH, num_splits);
   constexpr int NW = 4;
   dim3 block(NW * 64);
@@ -431,7 +431,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
         (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,                 \
         (const unsigned short*)q, (const unsigned short*)k_cache,            \
         (const unsigned short*)v_cache, (const int*)block_tables,            \
-        (const int*)seq_lens, max_blocks, scale, KH);                        \
+        (const int*)seq_lens, max_blocks, scale, KH, q_stride);              \
     if (num_splits > 1)                                                      \
       paged_attn_combine_kernel<HD>                                          \
           <<<dim3(num_seqs * KH * G), 64, 0, stream>>>(                      \
@@ -463,7 +463,7 @@ int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* token_seq, const void* token_pos,
                           int num_tokens, int num_q_heads, int max_blocks,
                           float scale, int KH, int GQ, int head_dim,
-                          int block_size, hipStream_t stream) {
+                          int block_size, long q_stride, hipStream_t stream) {
   dim3 grid(num_tokens, num_q_heads);
   dim3 block(64);
 #define PS_DISPATCH_PREFILL(HD, BS)                                         \
@@ -471,7 +471,7 @@ int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
       (unsigned short*)out, (const unsigned short*)q,                       \
       (const unsigned short*)k_cache, (const unsigned short*)v_cache,       \
       (const int*)block_tables, (const int*)token_seq,                      \
-      (const int*)token_pos, max_blocks, scale, KH, GQ)
+      (const int*)token_pos, max_blocks, scale, KH, GQ, q_stride)
   if (head_dim == 128 && block_size == 16) {
     PS_DISPATCH_PREFILL(128, 16);
     return 0;

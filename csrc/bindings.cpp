@@ -19,24 +19,29 @@ void ps_silu_and_mul(void* out, const void* x, long T, int D,
                      hipStream_t stream);
 void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
              long T, int QH, int KH, int HD, int ROT, hipStream_t stream);
+void ps_fused_rope_cache(void* qkv, const void* positions, const void* cos_sin,
+                         const void* slot_mapping, void* k_cache,
+                         void* v_cache, long T, int QH, int KH, int HD,
+                         int ROT, long qkv_stride, int BS,
+                         hipStream_t stream);
 int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* k_cache, const void* v_cache,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         hipStream_t stream);
+                         long q_stride, hipStream_t stream);
 int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* v_cache, const void* block_tables,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
-                               int KH, int GQ, int head_dim,
+                               int KH, int GQ, int head_dim, long q_stride,
                                hipStream_t stream);
 int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* v_cache, const void* block_tables,
                           const void* token_seq, const void* token_pos,
                           int num_tokens, int num_q_heads, int max_blocks,
                           float scale, int KH, int GQ, int head_dim,
-                          int block_size, hipStream_t stream);
+                          int block_size, long q_stride, hipStream_t stream);
 void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
                           void* v_cache, const void* slot_mapping, long T,
                           int KH, int HD, int BS, hipStream_t stream);
@@ -59,6 +64,15 @@ hipStream_t current_stream() {
   TORCH_CHECK((t).is_cuda(), #t " must be on the GPU");         \
   TORCH_CHECK((t).scalar_type() == (ty), #t " dtype mismatch"); \
   TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+// q may be a row-strided view into the packed qkv tensor
+long q_row_stride(const at::Tensor& q, int HD) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
+              "q must be bf16 on GPU");
+  TORCH_CHECK(q.dim() == 3 && q.stride(2) == 1 && q.stride(1) == HD,
+              "q must be [T, H, HD] with contiguous heads");
+  return q.stride(0);
+}
 
 void rms_norm(at::Tensor out, at::Tensor x, at::Tensor w, double eps) {
   CHECK_GPU_BF16(out);
@@ -112,7 +126,6 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
                        at::Tensor seq_lens, double scale, int64_t num_splits) {
   CHECK_GPU_BF16(out);
-  CHECK_GPU_BF16(q);
   CHECK_GPU_BF16(k_cache);
   CHECK_GPU_BF16(v_cache);
   CHECK_GPU_DTYPE(block_tables, at::kInt);
@@ -144,7 +157,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       out.data_ptr(), acc_p, ml_p, q.data_ptr(), k_cache.data_ptr(),
       v_cache.data_ptr(), block_tables.data_ptr(), seq_lens.data_ptr(), S,
       max_blocks, (float)scale, KH, GQ, HD, BS, (int)num_splits,
-      current_stream());
+      q_row_stride(q, HD), current_stream());
   TORCH_CHECK(rc == 0, "unsupported decode config: head_dim=", HD,
               " block_size=", BS, " gqa=", GQ);
 }
@@ -154,7 +167,6 @@ void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                         at::Tensor token_seq, at::Tensor token_pos,
                         double scale) {
   CHECK_GPU_BF16(out);
-  CHECK_GPU_BF16(q);
   CHECK_GPU_BF16(k_cache);
   CHECK_GPU_BF16(v_cache);
   CHECK_GPU_DTYPE(block_tables, at::kInt);
@@ -171,7 +183,8 @@ void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
   int rc = ps_paged_attn_prefill(
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr(), token_seq.data_ptr(), token_pos.data_ptr(), T,
-      QH, max_blocks, (float)scale, KH, GQ, HD, BS, current_stream());
+      QH, max_blocks, (float)scale, KH, GQ, HD, BS, q_row_stride(q, HD),
+      current_stream());
   TORCH_CHECK(rc == 0, "unsupported prefill config: head_dim=", HD,
               " block_size=", BS);
 }
@@ -181,7 +194,6 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
                              at::Tensor block_tables, at::Tensor tile_info,
                              double scale) {
   CHECK_GPU_BF16(out);
-  CHECK_GPU_BF16(q);
   CHECK_GPU_BF16(k_cache);
   CHECK_GPU_BF16(v_cache);
   CHECK_GPU_DTYPE(block_tables, at::kInt);
@@ -196,8 +208,34 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
   int rc = ps_paged_attn_prefill_mfma(
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr(), tile_info.data_ptr(), NT, QH,
-      (int)block_tables.size(1), (float)scale, KH, GQ, HD, current_stream());
+      (int)block_tables.size(1), (float)scale, KH, GQ, HD,
+      q_row_stride(q, HD), current_stream());
   TORCH_CHECK(rc == 0, "unsupported mfma prefill config: head_dim=", HD);
+}
+
+void fused_rope_cache(at::Tensor qkv, at::Tensor positions,
+                      at::Tensor cos_sin, at::Tensor slot_mapping,
+                      at::Tensor k_cache, at::Tensor v_cache,
+                      int64_t q_heads, int64_t head_dim) {
+  CHECK_GPU_BF16(qkv);
+  CHECK_GPU_DTYPE(positions, at::kInt);
+  CHECK_GPU_DTYPE(cos_sin, at::kFloat);
+  CHECK_GPU_DTYPE(slot_mapping, at::kLong);
+  CHECK_GPU_BF16(k_cache);
+  CHECK_GPU_BF16(v_cache);
+  const long T = positions.size(0);
+  const int KH = (int)k_cache.size(1);
+  const int BS = (int)k_cache.size(2);
+  const int HD = (int)head_dim;
+  const int ROT = (int)cos_sin.size(-1);
+  TORCH_CHECK(
+      qkv.size(-1) == (q_heads + 2 * KH) * HD,
+      "qkv width mismatch");
+  ps_fused_rope_cache(qkv.data_ptr(), positions.data_ptr(),
+                      cos_sin.data_ptr(), slot_mapping.data_ptr(),
+                      k_cache.data_ptr(), v_cache.data_ptr(), T,
+                      (int)q_heads, KH, HD, ROT, qkv.stride(0), BS,
+                      current_stream());
 }
 
 void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
@@ -244,6 +282,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Paged attention, chunked prefill (bf16 KV)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
         "Paged attention, chunked prefill via MFMA tiles (head_dim 128)");
+  m.def("fused_rope_cache", &fused_rope_cache,
+        "Fused RoPE + paged KV append on the packed qkv tensor");
   m.def("reshape_and_cache", &reshape_and_cache,
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");

@@ -233,3 +233,62 @@ void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Fused RoPE + KV append operating IN-PLACE on the packed qkv tensor
+// ([T, (QH+2*KH)*HD] with arbitrary row stride): rotates q and k, then
+// scatters the rotated k and v rows into the paged cache. Replaces
+// rope_kernel + reshape_and_cache_kernel + three .contiguous() copies.
+// ---------------------------------------------------------------------------
+__global__ void fused_rope_cache_kernel(
+    unsigned short* __restrict__ qkv, const int* __restrict__ positions,
+    const float* __restrict__ cos_sin, const long* __restrict__ slot_mapping,
+    unsigned short* __restrict__ k_cache, unsigned short* __restrict__ v_cache,
+    int QH, int KH, int HD, int ROT, long qkv_stride, int BS) {
+  const long t = blockIdx.x;
+  unsigned short* base = qkv + t * qkv_stride;
+  const float* cs = cos_sin + (long)positions[t] * ROT;
+  const int half = ROT / 2;
+  const int total = (QH + KH) * half;
+  for (int idx = threadIdx.x; idx < total; idx += blockDim.x) {
+    const int h = idx / half;
+    const int r = idx % half;
+    unsigned short* hp = base + (long)h * HD;  // q heads then k heads
+    const float c = cs[r];
+    const float s = cs[half + r];
+    const float x1 = ps_bf16_to_f32(hp[r]);
+    const float x2 = ps_bf16_to_f32(hp[r + half]);
+    hp[r] = ps_f32_to_bf16(x1 * c - x2 * s);
+    hp[r + half] = ps_f32_to_bf16(x2 * c + x1 * s);
+  }
+  __syncthreads();
+  const long slot = slot_mapping[t];
+  if (slot < 0) return;
+  const long blk = slot / BS;
+  const int off = (int)(slot % BS);
+  const int kvelems = KH * HD;
+  const unsigned short* ksrc = base + (long)QH * HD;
+  const unsigned short* vsrc = ksrc + kvelems;
+  for (int i = threadIdx.x * 8; i < 2 * kvelems; i += blockDim.x * 8) {
+    const bool isv = i >= kvelems;
+    const int r = isv ? i - kvelems : i;
+    const int h = r / HD;
+    const int d = r % HD;
+    const long dst = ((blk * KH + h) * BS + off) * (long)HD + d;
+    unsigned short* cache = isv ? v_cache : k_cache;
+    const unsigned short* src = (isv ? vsrc : ksrc) + r;
+    *(ps_bf16x8*)(cache + dst) = *(const ps_bf16x8*)src;
+  }
+}
+
+extern "C" void ps_fused_rope_cache(void* qkv, const void* positions,
+                                    const void* cos_sin,
+                                    const void* slot_mapping, void* k_cache,
+                                    void* v_cache, long T, int QH, int KH,
+                                    int HD, int ROT, long qkv_stride, int BS,
+                                    hipStream_t stream) {
+  fused_rope_cache_kernel<<<dim3((unsigned)T), 256, 0, stream>>>(
+      (unsigned short*)qkv, (const int*)positions, (const float*)cos_sin,
+      (const long*)slot_mapping, (unsigned short*)k_cache,
+      (unsigned short*)v_cache, QH, KH, HD, ROT, qkv_stride, BS);
+}

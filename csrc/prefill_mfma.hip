@@ -45,7 +45,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
     const unsigned short* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ tile_info,     // [NT, 4]
-    int max_blocks, float scale, int KH, int GQ) {
+    int max_blocks, float scale, int KH, int GQ, long q_stride) {
   constexpr int D = HEAD_DIM;  // 128
   constexpr int BS = 16;       // page size in tokens
   constexpr int NK = D / 32;   // mfma k-steps over head dim (4)
@@ -78,7 +78,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
   const int my_local_row = wave * 16 + rc;
   const int q_row_clamped = min(my_local_row, n_rows - 1);
   const unsigned short* qrow =
-      q + ((long)(q_tok0 + q_row_clamped) * QH + qh) * D;
+      q + (long)(q_tok0 + q_row_clamped) * q_stride + (long)qh * D;
   ps_mbf16x8 q_frag[NK];
 #pragma unroll
   for (int kk = 0; kk < NK; kk++)
@@ -241,7 +241,7 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* v_cache, const void* block_tables,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
-                               int KH, int GQ, int head_dim,
+                               int KH, int GQ, int head_dim, long q_stride,
                                hipStream_t stream) {
   if (head_dim != 128) return -1;
   dim3 grid(num_tiles, num_q_heads);
@@ -249,7 +249,7 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
       (unsigned short*)out, (const unsigned short*)q,
       (const unsigned short*)k_cache, (const unsigned short*)v_cache,
       (const int*)block_tables, (const int*)tile_info, max_blocks, scale, KH,
-      GQ);
+      GQ, q_stride);
   return 0;
 }
 

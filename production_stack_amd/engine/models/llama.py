@@ -117,20 +117,31 @@ class LlamaLayer(nn.Module):
         qkv = F.linear(hidden, self.qkv_proj)
         qs = self.q_heads * self.head_dim
         kvs = self.kv_heads * self.head_dim
-        q = qkv[:, :qs].contiguous()
-        k = qkv[:, qs : qs + kvs].contiguous()
-        v = qkv[:, qs + kvs :].contiguous()
-        q, k = ops.rotary_embedding(meta.positions, q, k, cos_sin, self.head_dim)
-        T = q.shape[0]
+        T = qkv.shape[0]
         k_cache, v_cache = kv_cache
-        ops.reshape_and_cache(
-            k.view(T, self.kv_heads, self.head_dim),
-            v.view(T, self.kv_heads, self.head_dim),
-            k_cache,
-            v_cache,
-            meta.slot_mapping,
-        )
-        qh = q.view(T, self.q_heads, self.head_dim)
+        if qkv.is_cuda:
+            # fused in-place RoPE + KV append straight on the packed qkv;
+            # attention consumes the q heads as a strided view (no copies)
+            ops.fused_rope_cache(
+                qkv, meta.positions, cos_sin, meta.slot_mapping,
+                k_cache, v_cache, self.q_heads, self.head_dim,
+            )
+            qh = qkv.view(T, -1, self.head_dim)[:, : self.q_heads]
+        else:
+            q = qkv[:, :qs].contiguous()
+            k = qkv[:, qs : qs + kvs].contiguous()
+            v = qkv[:, qs + kvs :].contiguous()
+            q, k = ops.rotary_embedding(
+                meta.positions, q, k, cos_sin, self.head_dim
+            )
+            ops.reshape_and_cache(
+                k.view(T, self.kv_heads, self.head_dim),
+                v.view(T, self.kv_heads, self.head_dim),
+                k_cache,
+                v_cache,
+                meta.slot_mapping,
+            )
+            qh = q.view(T, self.q_heads, self.head_dim)
         outs: List[torch.Tensor] = []
         tp = meta.num_prefill_tokens
         if tp > 0:
@@ -141,7 +152,7 @@ class LlamaLayer(nn.Module):
             ):
                 outs.append(
                     ops.paged_attn_prefill_mfma(
-                        qh[:tp].contiguous(),
+                        qh[:tp],
                         k_cache,
                         v_cache,
                         meta.prefill_block_tables,
@@ -152,7 +163,7 @@ class LlamaLayer(nn.Module):
             else:
                 outs.append(
                     ops.paged_attn_prefill(
-                        qh[:tp].contiguous(),
+                        qh[:tp] if qh.is_cuda else qh[:tp].contiguous(),
                         k_cache,
                         v_cache,
                         meta.prefill_block_tables,
@@ -164,7 +175,7 @@ class LlamaLayer(nn.Module):
         if meta.num_decode_seqs > 0:
             outs.append(
                 ops.paged_attn_decode(
-                    qh[tp:].contiguous(),
+                    qh[tp:] if qh.is_cuda else qh[tp:].contiguous(),
                     k_cache,
                     v_cache,
                     meta.decode_block_tables,

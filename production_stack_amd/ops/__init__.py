@@ -75,6 +75,24 @@ def rotary_embedding(
     return reference.rotary_embedding(positions, q, k, cos_sin, head_dim)
 
 
+def fused_rope_cache(
+    qkv: torch.Tensor,
+    positions: torch.Tensor,
+    cos_sin: torch.Tensor,
+    slot_mapping: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    q_heads: int,
+    head_dim: int,
+) -> None:
+    """Fused in-place RoPE on packed qkv + paged KV append (GPU only)."""
+    _require_ext()
+    _C.fused_rope_cache(
+        qkv, positions, cos_sin, slot_mapping, k_cache, v_cache, q_heads,
+        head_dim,
+    )
+
+
 def reshape_and_cache(
     k: torch.Tensor,
     v: torch.Tensor,
@@ -102,7 +120,7 @@ def paged_attn_decode(
 ) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attn_decode(
             out, q, k_cache, v_cache, block_tables, seq_lens, scale, 0
         )
@@ -123,7 +141,7 @@ def paged_attn_prefill(
 ) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
-        out = torch.empty_like(q)
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attn_prefill(
             out, q, k_cache, v_cache, block_tables, token_seq, token_pos, scale
         )
@@ -143,7 +161,7 @@ def paged_attn_prefill_mfma(
 ) -> torch.Tensor:
     """MFMA-tiled chunked prefill (GPU, head_dim 128 only)."""
     _require_ext()
-    out = torch.empty_like(q)
+    out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
     _C.paged_attn_prefill_mfma(
         out, q, k_cache, v_cache, block_tables, tile_info, scale
     )

@@ -253,3 +253,43 @@ def test_prefill_mfma_matches_valu_kernel():
         token_pos.cuda(), 0.0883883,
     )
     _close(a, b, atol=3e-2, rtol=3e-2)
+
+
+def test_fused_rope_cache_matches_separate_ops():
+    from production_stack_amd import _C
+
+    torch.manual_seed(6)
+    T, qh, kh, hd, bs, nb = 23, 32, 8, 128, 16, 8
+    rot = hd
+    width = (qh + 2 * kh) * hd
+    qkv = torch.randn((T, width), dtype=torch.bfloat16, device="cuda")
+    inv = 1.0 / (500000.0 ** (torch.arange(0, rot, 2).float() / rot))
+    freqs = torch.outer(torch.arange(4096).float(), inv)
+    cos_sin = torch.cat([freqs.cos(), freqs.sin()], -1).contiguous().cuda()
+    pos = torch.randint(0, 4096, (T,), dtype=torch.int32, device="cuda")
+    slots = torch.randperm(nb * bs)[:T].to(torch.long).cuda()
+    slots[3] = -1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    kc2, vc2 = k_cache.clone(), v_cache.clone()
+
+    # reference composition: split -> rope -> reshape_and_cache
+    qs, kvs = qh * hd, kh * hd
+    q = qkv[:, :qs].contiguous()
+    k = qkv[:, qs : qs + kvs].contiguous()
+    v = qkv[:, qs + kvs :].contiguous()
+    q_ref, k_ref = reference.rotary_embedding(
+        pos.cpu(), q.cpu(), k.cpu(), cos_sin.cpu(), hd
+    )
+    kc2, vc2 = kc2.cpu(), vc2.cpu()
+    reference.reshape_and_cache(
+        k_ref.view(T, kh, hd), v.cpu().view(T, kh, hd), kc2, vc2, slots.cpu()
+    )
+
+    qkv_fused = qkv.clone()
+    _C.fused_rope_cache(
+        qkv_fused, pos, cos_sin, slots, k_cache, v_cache, qh, hd
+    )
+    _close(qkv_fused[:, :qs], q_ref)
+    _close(qkv_fused[:, qs : qs + kvs], k_ref)
+    _close(k_cache, kc2, atol=2e-2, rtol=2e-2)
+    _close(v_cache, vc2, atol=0, rtol=0)
