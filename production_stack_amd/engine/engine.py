@@ -8,6 +8,7 @@ parallel/state.py).
 
 from __future__ import annotations
 
+import threading
 import time
 from typing import Dict, List, Optional, Union
 
@@ -77,6 +78,9 @@ class LLMEngine:
         )
         self.stats = EngineStats()
         self._sleeping = False
+        # coarse lock: engine-loop steps vs out-of-band block mutations
+        # (KV transfer registration/adoption)
+        self.lock = threading.RLock()
 
     # ------------------------------------------------------------------
     def add_request(
@@ -106,6 +110,10 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def step(self) -> List[RequestOutput]:
+        with self.lock:
+            return self._step_locked()
+
+    def _step_locked(self) -> List[RequestOutput]:
         out = self.scheduler.schedule()
         if out.is_empty and not out.capacity_stopped:
             return []

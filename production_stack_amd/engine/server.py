@@ -60,6 +60,21 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
     async def lifespan(app: FastAPI):
         async_engine.start(asyncio.get_running_loop())
         cfg = engine.config
+        if getattr(cfg, "kv_transfer", None):
+            from production_stack_amd.parallel.kv_transfer import (
+                KVTransferService,
+            )
+
+            kt = cfg.kv_transfer
+            app.state.kv_service = KVTransferService(
+                engine,
+                kv_rank=kt["rank"],
+                kv_world=kt["world"],
+                master_port=kt.get("master_port", 14500),
+                side_port=kt.get("side_port", 14001),
+                host=kt.get("host", "127.0.0.1"),
+            )
+            await app.state.kv_service.start_side_channel()
         if getattr(cfg, "kv_controller_url", None):
             host, _, port = cfg.kv_controller_url.rpartition(":")
             from production_stack_amd.kvpool.client import EngineReporter
@@ -75,6 +90,9 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         reporter = getattr(app.state, "kv_reporter", None)
         if reporter is not None:
             await reporter.stop()
+        kv_service = getattr(app.state, "kv_service", None)
+        if kv_service is not None:
+            await kv_service.stop()
         async_engine.stop()
 
     app = FastAPI(
@@ -222,6 +240,26 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         model_name = body.get("model", served_model)
         obj = "chat.completion" if chat else "text_completion"
 
+        # disaggregated prefill: decode role pulls the prefiller's KV blocks
+        # into the local prefix cache before scheduling (parallel/kv_transfer)
+        kvp = body.get("kv_transfer_params") or {}
+        kv_service = getattr(app.state, "kv_service", None)
+        if (
+            kv_service is not None
+            and kvp.get("do_remote_prefill")
+            and kvp.get("remote_request_id")
+        ):
+            try:
+                await kv_service.pull_into_prefix_cache(
+                    kvp["remote_request_id"],
+                    list(prompt_tokens),
+                    kvp.get("remote_host") or "127.0.0.1",
+                    int(kvp.get("remote_port") or 14001),
+                    int(kvp.get("remote_engine_id") or 0),
+                )
+            except (ConnectionError, OSError, asyncio.TimeoutError) as e:
+                logger.warning("KV pull failed (%s); recomputing prefill", e)
+
         if body.get("stream"):
 
             async def gen():
@@ -314,11 +352,15 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                 "total_tokens": n_prompt + len(tokens),
             },
         }
-        if body.get("kv_transfer_params"):
-            # disaggregated prefill handshake (PD wiring: parallel/kv_transfer)
-            resp["kv_transfer_params"] = handle_kv_transfer_params(
-                app, body["kv_transfer_params"], rid
-            )
+        if kvp.get("do_remote_decode"):
+            if kv_service is not None:
+                resp["kv_transfer_params"] = kv_service.register_prefilled(
+                    rid, list(prompt_tokens)
+                )
+            else:
+                resp["kv_transfer_params"] = handle_kv_transfer_params(
+                    app, kvp, rid
+                )
         return JSONResponse(resp)
 
     @app.post("/v1/completions")
@@ -385,6 +427,11 @@ def main() -> None:
                     help="this engine's URL as seen by the router")
     ap.add_argument("--kv-role", default=None,
                     choices=[None, "kv_producer", "kv_consumer"])
+    ap.add_argument("--kv-rank", type=int, default=None,
+                    help="rank in the KV-transfer group (disagg prefill)")
+    ap.add_argument("--kv-world", type=int, default=0)
+    ap.add_argument("--kv-master-port", type=int, default=14500)
+    ap.add_argument("--kv-side-port", type=int, default=14001)
     ap.add_argument("--cpu-offload-gb", type=float, default=0.0)
     args = ap.parse_args()
 
@@ -413,6 +460,13 @@ def main() -> None:
         ),
     )
     cfg.kv_controller_url = args.kv_controller_url  # type: ignore[attr-defined]
+    if args.kv_rank is not None and args.kv_world > 1:
+        cfg.kv_transfer = {  # type: ignore[attr-defined]
+            "rank": args.kv_rank,
+            "world": args.kv_world,
+            "master_port": args.kv_master_port,
+            "side_port": args.kv_side_port,
+        }
     cfg.advertise_url = (  # type: ignore[attr-defined]
         args.advertise_url or f"http://127.0.0.1:{args.port}"
     )

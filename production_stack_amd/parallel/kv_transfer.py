@@ -107,20 +107,36 @@ class KVTransferService:
             )
 
     # ---- prefill (kv_producer) side -----------------------------------
-    def register_prefilled(self, request_id: str, seq) -> Dict:
-        """Pin a finished prefill's blocks and describe them for the
-        kv_transfer_params response."""
+    def register_prefilled(self, request_id: str, prompt_token_ids) -> Dict:
+        """Pin the finished prefill's full prompt blocks (found through the
+        prefix cache, which retains them after the 1-token request finishes)
+        and describe them for the kv_transfer_params response."""
         bm = self.engine.block_manager
-        n_full = seq.num_tokens // bm.block_size
-        block_ids = list(seq.block_table[:n_full])
+        bs = bm.block_size
+        n_full = len(prompt_token_ids) // bs
+        block_ids: List[int] = []
+        prev = None
+        with self.engine.lock:
+            for i in range(n_full):
+                h = bm.chain_hash(
+                    prev, tuple(prompt_token_ids[i * bs : (i + 1) * bs])
+                )
+                blk = bm.cached.get(h)
+                if blk is None:
+                    break
+                bm.ref_count[blk] += 1
+                bm.evictable.pop(blk, None)
+                block_ids.append(blk)
+                prev = h
         with self._lock:
-            for b in block_ids:
-                bm.ref_count[b] += 1
-            self.pending[request_id] = (block_ids, seq.num_tokens, time.time())
+            self.pending[request_id] = (
+                block_ids, len(prompt_token_ids), time.time()
+            )
         return {
             "do_remote_decode": False,
             "do_remote_prefill": True,
             "remote_engine_id": self.kv_rank,
+            "remote_request_id": request_id,
             "remote_block_ids": block_ids,
             "remote_host": self.host,
             "remote_port": self.side_port,
@@ -130,8 +146,9 @@ class KVTransferService:
         bm = self.engine.block_manager
         with self._lock:
             entry = self.pending.pop(request_id, None)
-            if entry is None:
-                return
+        if entry is None:
+            return
+        with self.engine.lock:
             for b in entry[0]:
                 bm._release_block(b)
 
@@ -224,11 +241,12 @@ class KVTransferService:
                 return 0
             # local destination blocks
             dst_blocks = []
-            for _ in range(n_blocks):
-                b = bm._pop_block()
-                if b is None:
-                    break
-                dst_blocks.append(b)
+            with self.engine.lock:
+                for _ in range(n_blocks):
+                    b = bm._pop_block()
+                    if b is None:
+                        break
+                    dst_blocks.append(b)
             layers, elems = self._block_shape()
             pack = torch.empty(
                 (int(head["n_blocks"]), layers, 2, elems),
@@ -243,18 +261,19 @@ class KVTransferService:
             # adopt into the prefix cache
             prev = None
             adopted = 0
-            for i, blk in enumerate(dst_blocks):
-                h = bm.chain_hash(
-                    prev, tuple(prompt_token_ids[i * bs : (i + 1) * bs])
-                )
-                if h not in bm.cached:
-                    bm.cached[h] = blk
-                    bm.block_hash[blk] = h
-                    bm.evictable[blk] = None  # ref 0, contents valid
-                else:
-                    bm.free.append(blk)
-                prev = h
-                adopted += 1
+            with self.engine.lock:
+                for i, blk in enumerate(dst_blocks):
+                    h = bm.chain_hash(
+                        prev, tuple(prompt_token_ids[i * bs : (i + 1) * bs])
+                    )
+                    if h not in bm.cached:
+                        bm.cached[h] = blk
+                        bm.block_hash[blk] = h
+                        bm.evictable[blk] = None  # ref 0, contents valid
+                    else:
+                        bm.free.append(blk)
+                    prev = h
+                    adopted += 1
             return adopted * bs
         finally:
             writer.close()
