@@ -133,6 +133,11 @@ def build_app() -> FastAPI:
         for alias, target in aliases.items():
             if alias not in cards:
                 cards[alias] = ModelCard(id=alias, parent=target)
+        ext = getattr(app.state, "external_providers", None)
+        if ext is not None:
+            for name in ext.model_names():
+                if name not in cards:
+                    cards[name] = ModelCard(id=name, owned_by="external")
         return ModelList(data=list(cards.values())).model_dump()
 
     # ---- ops endpoints -------------------------------------------------
@@ -375,7 +380,45 @@ def initialize_all(app: FastAPI, args) -> None:
         initialize_feature_gates,
     )
 
-    initialize_feature_gates(args.feature_gates)
+    gates = initialize_feature_gates(args.feature_gates)
+
+    from production_stack_amd.router import tracing
+
+    tracing.initialize_tracing(
+        getattr(args, "otel_endpoint", None),
+        getattr(args, "otel_service_name", "vllm-router"),
+        getattr(args, "otel_secure", False),
+    )
+
+    if getattr(args, "external_providers_config", None):
+        from production_stack_amd.router.external_providers import (
+            ExternalProviderManager,
+        )
+
+        app.state.external_providers = ExternalProviderManager.from_yaml(
+            args.external_providers_config
+        )
+    else:
+        app.state.external_providers = None
+
+    if gates.is_enabled("PIIDetection"):
+        from production_stack_amd.router.pii import create_analyzer
+
+        app.state.pii_analyzer = create_analyzer(
+            getattr(args, "pii_analyzer", "regex")
+        )
+        app.state.pii_action = getattr(args, "pii_action", "block")
+    else:
+        app.state.pii_analyzer = None
+
+    if gates.is_enabled("SemanticCache"):
+        from production_stack_amd.router.semantic_cache import SemanticCache
+
+        app.state.semantic_cache = SemanticCache(
+            threshold=getattr(args, "semantic_cache_threshold", 0.95)
+        )
+    else:
+        app.state.semantic_cache = None
     app.state.request_rewriter = (
         get_request_rewriter(args.request_rewriter)
         if args.request_rewriter != "noop"

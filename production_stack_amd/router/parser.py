@@ -79,6 +79,18 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     p.add_argument("--api-key", default=None,
                    help="bearer token forwarded to engines")
 
+    # observability / experimental
+    p.add_argument("--otel-endpoint", default=None)
+    p.add_argument("--otel-service-name", default="vllm-router")
+    p.add_argument("--otel-secure", action="store_true")
+    p.add_argument("--external-providers-config", default=None,
+                   help="YAML file of external OpenAI-compatible providers")
+    p.add_argument("--pii-analyzer", default="regex",
+                   choices=["regex", "presidio"])
+    p.add_argument("--pii-action", default="block",
+                   choices=["block", "redact"])
+    p.add_argument("--semantic-cache-threshold", type=float, default=0.95)
+
     # logging
     p.add_argument("--log-level", default="info",
                    choices=["trace", "debug", "info", "warning", "error"])

@@ -45,9 +45,15 @@ HOP_BY_HOP = {
 
 
 def _forward_headers(request: Request) -> Dict[str, str]:
-    return {
+    from production_stack_amd.router import tracing
+
+    headers = {
         k: v for k, v in request.headers.items() if k.lower() not in HOP_BY_HOP
     }
+    if tracing.is_enabled():
+        parent = tracing.extract_context(request.headers)
+        headers = tracing.inject_context(headers, parent)
+    return headers
 
 
 def _session() -> aiohttp.ClientSession:
@@ -122,6 +128,47 @@ async def route_general_request(
 
     app = request.app
     router = app.state.router
+
+    # PII scan (feature-gated)
+    pii = getattr(app.state, "pii_analyzer", None)
+    if pii is not None and request_json:
+        from production_stack_amd.router.pii import scan_request_body
+
+        allowed, request_json, matches = scan_request_body(
+            request_json, pii, getattr(app.state, "pii_action", "block")
+        )
+        if not allowed:
+            return JSONResponse(
+                status_code=400,
+                content={
+                    "error": "request blocked: PII detected",
+                    "entities": sorted({m.entity_type for m in matches}),
+                },
+            )
+        if matches:
+            body = json.dumps(request_json).encode()
+
+    # semantic cache lookup (feature-gated, chat only, non-streaming)
+    sem = getattr(app.state, "semantic_cache", None)
+    if (
+        sem is not None
+        and endpoint == "/v1/chat/completions"
+        and not request_json.get("stream")
+    ):
+        cached = sem.search(request_json)
+        if cached is not None:
+            return JSONResponse(
+                status_code=200,
+                content=cached,
+                headers={"x-semantic-cache": "hit"},
+            )
+
+    # external provider branch
+    ext = getattr(app.state, "external_providers", None)
+    if ext is not None and ext.has_model(requested_model):
+        return await _proxy_external(
+            app, ext, endpoint, request_json, request_id
+        )
     aliases = getattr(app.state, "model_aliases", None)
 
     # PD orchestrated short-circuit
@@ -187,8 +234,18 @@ async def route_general_request(
     tried: set = set()
     attempt = 0
     last_error: Optional[str] = None
+    sem_store = (
+        sem is not None
+        and endpoint == "/v1/chat/completions"
+        and not request_json.get("stream")
+    )
     while True:
         try:
+            if sem_store:
+                return await _proxy_buffered_with_cache(
+                    request, body, server_url, endpoint, request_id,
+                    sem, request_json,
+                )
             return await _proxy_streaming(
                 request, body, server_url, endpoint, request_id
             )
@@ -220,6 +277,76 @@ async def route_general_request(
             "detail": last_error,
         },
     )
+
+
+async def _proxy_buffered_with_cache(
+    request: Request,
+    body: bytes,
+    server_url: str,
+    endpoint: str,
+    request_id: str,
+    sem,
+    request_json: Dict[str, Any],
+) -> Response:
+    """Non-streaming chat proxy that stores the answer in the semantic
+    cache (reference request.py:360-364 behaviour)."""
+    monitor = get_request_stats_monitor()
+    monitor.on_new_request(server_url, request_id, time.time())
+    session = _session()
+    async with session.post(
+        server_url + endpoint,
+        data=body,
+        headers=_forward_headers(request),
+        timeout=aiohttp.ClientTimeout(total=None),
+    ) as resp:
+        monitor.on_request_response(server_url, request_id, time.time())
+        data = await resp.read()
+        status = resp.status
+        ctype = resp.headers.get("content-type", "application/json")
+    monitor.on_request_complete(server_url, request_id, time.time())
+    if status == 200:
+        try:
+            sem.store(request_json, json.loads(data))
+        except (ValueError, TypeError):
+            pass
+    return Response(
+        content=data,
+        status_code=status,
+        media_type=ctype,
+        headers={"x-request-id": request_id},
+    )
+
+
+async def _proxy_external(
+    app, ext, endpoint: str, request_json: Dict[str, Any], request_id: str
+) -> Response:
+    provider = ext.provider_for(request_json.get("model"))
+    session = _session()
+    resp = await provider.forward(session, endpoint, request_json)
+    if request_json.get("stream"):
+
+        async def stream():
+            try:
+                async for chunk in resp.content.iter_any():
+                    yield chunk
+            finally:
+                resp.release()
+
+        return StreamingResponse(
+            stream(),
+            status_code=resp.status,
+            media_type=resp.headers.get("content-type", "application/json"),
+            headers={"x-request-id": request_id},
+        )
+    data = await resp.read()
+    out = Response(
+        content=data,
+        status_code=resp.status,
+        media_type=resp.headers.get("content-type", "application/json"),
+        headers={"x-request-id": request_id},
+    )
+    resp.release()
+    return out
 
 
 async def _proxy_streaming(
