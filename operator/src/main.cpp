@@ -1,0 +1,81 @@
+// psoperator: native reconciler for the production-stack-amd CRDs
+// (VLLMRuntime, VLLMRouter, CacheServer, LoraAdapter).
+//
+// Usage (in cluster):
+//   psoperator --namespace default
+// (reads the service-account token + API server from the pod environment)
+// Usage (tests / out of cluster):
+//   psoperator --api-server http://127.0.0.1:9443 --namespace default --once
+
+#include <unistd.h>
+
+#include <cstdio>
+#include <cstring>
+#include <fstream>
+#include <string>
+
+#include "reconciler.h"
+
+static std::string read_file(const std::string& path) {
+  std::ifstream f(path);
+  if (!f) return "";
+  std::string s((std::istreambuf_iterator<char>(f)),
+                std::istreambuf_iterator<char>());
+  while (!s.empty() && (s.back() == '\n' || s.back() == '\r')) s.pop_back();
+  return s;
+}
+
+int main(int argc, char** argv) {
+  psop::Ctx ctx;
+  ctx.ns = "default";
+  int interval = 10;
+  bool once = false;
+  std::string token_file =
+      "/var/run/secrets/kubernetes.io/serviceaccount/token";
+
+  for (int i = 1; i < argc; i++) {
+    std::string a = argv[i];
+    auto next = [&]() -> std::string {
+      return (i + 1 < argc) ? argv[++i] : "";
+    };
+    if (a == "--api-server") ctx.api_server = next();
+    else if (a == "--namespace") ctx.ns = next();
+    else if (a == "--token-file") token_file = next();
+    else if (a == "--interval") interval = std::stoi(next());
+    else if (a == "--once") once = true;
+    else if (a == "--help") {
+      printf(
+          "psoperator [--api-server URL] [--namespace NS] [--token-file F]\n"
+          "           [--interval SEC] [--once]\n");
+      return 0;
+    }
+  }
+
+  if (ctx.api_server.empty()) {
+    const char* host = getenv("KUBERNETES_SERVICE_HOST");
+    const char* port = getenv("KUBERNETES_SERVICE_PORT");
+    if (host && port)
+      ctx.api_server =
+          "https://" + std::string(host) + ":" + std::string(port);
+    else
+      ctx.api_server = "https://kubernetes.default.svc";
+  }
+  ctx.token = read_file(token_file);
+
+  fprintf(stderr, "[psoperator] api=%s ns=%s interval=%ds once=%d\n",
+          ctx.api_server.c_str(), ctx.ns.c_str(), interval, (int)once);
+
+  while (true) {
+    int actions = 0;
+    try {
+      actions = psop::reconcile_all(ctx);
+    } catch (const std::exception& e) {
+      fprintf(stderr, "[psoperator] reconcile error: %s\n", e.what());
+    }
+    if (actions)
+      fprintf(stderr, "[psoperator] applied %d change(s)\n", actions);
+    if (once) break;
+    sleep(interval);
+  }
+  return 0;
+}

@@ -1,0 +1,33 @@
+#pragma once
+
+#include <string>
+
+#include "psjson.h"
+
+namespace psop {
+
+struct Ctx {
+  std::string api_server;  // e.g. https://kubernetes.default.svc
+  std::string ns;
+  std::string token;
+  std::string group = "production-stack.amd.com";
+  std::string version = "v1alpha1";
+};
+
+// One reconcile pass over every CR kind. Returns the number of actions
+// (creates/updates) performed.
+int reconcile_all(const Ctx& ctx);
+
+// Exposed for tests: builders from CR -> desired child objects.
+psjson::ValuePtr build_engine_deployment(const Ctx& ctx,
+                                         const psjson::ValuePtr& cr);
+psjson::ValuePtr build_engine_service(const Ctx& ctx,
+                                      const psjson::ValuePtr& cr);
+psjson::ValuePtr build_router_deployment(const Ctx& ctx,
+                                         const psjson::ValuePtr& cr);
+psjson::ValuePtr build_cacheserver_deployment(const Ctx& ctx,
+                                              const psjson::ValuePtr& cr);
+
+uint64_t spec_hash(const psjson::ValuePtr& spec);
+
+}  // namespace psop
