@@ -97,30 +97,45 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
 
   const int wave_pos_max = q_pos0 + min(wave * 16 + 15, n_rows - 1);
 
+  // V^T staging is software-pipelined (guide T14 async-STAGE): this
+  // thread's V row for chunk c+1 is issued as global loads during chunk c's
+  // compute phase and written to LDS after the end-of-compute barrier.
+  const int tv = tid & 31;          // staged token (0..31 within chunk)
+  const int d0 = (tid >> 5) * 16;   // staged dim range [d0, d0+16)
+  const int tg = tv >> 3;
+  const int tl = tv & 7;
+  auto v_row_ptr = [&](int chunk) {
+    const int pg_idx = min((chunk * 32 + tv) / BS, n_pages - 1);
+    const long pg = bt[pg_idx];
+    return v_cache +
+           ((pg * KH + kvh) * BS + ((chunk * 32 + tv) & (BS - 1))) * D;
+  };
+  ps_bf16x8 vstage[2];
+  {
+    const unsigned short* vrow = v_row_ptr(0);
+    vstage[0] = *(const ps_bf16x8*)(vrow + d0);
+    vstage[1] = *(const ps_bf16x8*)(vrow + d0 + 8);
+  }
+
   for (int chunk = 0; chunk < n_chunks; chunk++) {
     const int tok0 = chunk * 32;
-    // ---- cooperative V^T staging (all threads) ----
-    // thread t covers token tv = tid&31, dims [ (tid>>5)*16, +16 )
-    {
-      const int tv = tid & 31;          // 0..31
-      const int d0 = (tid >> 5) * 16;   // 0..112
-      const int pg_idx = min((tok0 + tv) / BS, n_pages - 1);
-      const long pg = bt[pg_idx];
-      const unsigned short* vrow =
-          v_cache + ((pg * KH + kvh) * BS + ((tok0 + tv) & (BS - 1))) * D;
-      const int tg = tv >> 3;           // logical token group
-      const int tl = tv & 7;
+    // write the pre-fetched V^T tile for this chunk
 #pragma unroll
-      for (int h = 0; h < 2; h++) {
-        ps_bf16x8 vv = *(const ps_bf16x8*)(vrow + d0 + h * 8);
+    for (int h = 0; h < 2; h++) {
 #pragma unroll
-        for (int j = 0; j < 8; j++) {
-          const int d = d0 + h * 8 + j;
-          v_t[d][(tg ^ (d & 3)) * 8 + tl] = vv[j];
-        }
+      for (int j = 0; j < 8; j++) {
+        const int d = d0 + h * 8 + j;
+        v_t[d][(tg ^ (d & 3)) * 8 + tl] = vstage[h][j];
       }
     }
     __syncthreads();
+    // issue next chunk's V loads now; HBM latency hides under the MFMA
+    // phase below and the ds_write happens after the trailing barrier
+    if (chunk + 1 < n_chunks) {
+      const unsigned short* vrow = v_row_ptr(chunk + 1);
+      vstage[0] = *(const ps_bf16x8*)(vrow + d0);
+      vstage[1] = *(const ps_bf16x8*)(vrow + d0 + 8);
+    }
 
     const bool wave_active = (wave * 16 < n_rows) && (tok0 <= wave_pos_max);
     if (wave_active) {
@@ -142,6 +157,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
         const long pg = bt[pg_idx];
         const unsigned short* krow =
             k_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int kk = 0; kk < NK; kk++) {
           ps_mbf16x8 k_frag =
@@ -149,6 +165,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
           s_frag[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[kk], k_frag, s_frag[ct], 0, 0, 0);
         }
+        __builtin_amdgcn_s_setprio(0);
       }
       // ---- mask + online softmax ----
       // lane holds S[row g*4+r][col ct*16+rc]
@@ -209,6 +226,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
       // A-frag: P[row rc][tokens g*8 .. g*8+8)
       ps_mbf16x8 p_frag =
           ps_as_mbf16(*(const ps_bf16x8*)(&p_lds[wave][rc][g * 8]));
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int s = 0; s < 8; s++) {
         // B-frag: V^T[dim s*16 + rc][tokens g*8 .. +8) (swizzled group)
@@ -218,6 +236,7 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
         o_acc[s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             p_frag, v_frag, o_acc[s], 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();  // protect v_t before next chunk's staging
   }

@@ -1,0 +1,69 @@
+#!/usr/bin/env python3
+"""Standalone prefill/decode attention kernel benchmark on MI355X.
+
+Reports achieved TFLOP/s for the MFMA chunked-prefill kernel and effective
+HBM GB/s for decode, at Llama-3-8B shapes.
+"""
+import sys
+import time
+
+import torch
+
+from production_stack_amd import _C
+
+
+def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20):
+    bs = 16
+    nblocks = ctx_len // bs + 1
+    k = torch.randn(nblocks + 1, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn_like(k)
+    bt = torch.arange(1, nblocks + 1, dtype=torch.int32, device="cuda").reshape(1, -1)
+    T = ctx_len
+    q = torch.randn(T, qh, hd, dtype=torch.bfloat16, device="cuda")
+    tiles = []
+    for t0 in range(0, T, 64):
+        tiles.append([0, t0, t0, min(64, T - t0)])
+    tiles = torch.tensor(tiles, dtype=torch.int32, device="cuda")
+    out = torch.empty_like(q)
+    scale = hd ** -0.5
+    for _ in range(3):
+        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    # causal flops: sum over q of ctx(q) = T*(T+1)/2 per head pair (QK+PV)
+    flops = 4 * hd * qh * (T * (T + 1) / 2)
+    print(f"prefill ctx={ctx_len}: {dt*1e3:.2f} ms  {flops/dt/1e12:.1f} TF")
+
+
+def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50):
+    bs = 16
+    per = ctx // bs
+    nb = batch * per + 1
+    k = torch.randn(nb, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn_like(k)
+    bt = torch.arange(1, batch * per + 1, dtype=torch.int32, device="cuda").reshape(batch, per)
+    sl = torch.full((batch,), ctx, dtype=torch.int32, device="cuda")
+    q = torch.randn(batch, qh, hd, dtype=torch.bfloat16, device="cuda")
+    out = torch.empty_like(q)
+    scale = hd ** -0.5
+    for _ in range(3):
+        _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    bytes_kv = batch * ctx * kh * hd * 2 * 2
+    print(f"decode b={batch} ctx={ctx}: {dt*1e6:.1f} us  {bytes_kv/dt/1e9:.0f} GB/s")
+
+
+if __name__ == "__main__":
+    for ctx in (1024, 2048, 4096):
+        bench_prefill(ctx)
+    for b in (16, 64, 256):
+        bench_decode(batch=b)

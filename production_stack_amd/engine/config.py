@@ -122,6 +122,7 @@ class SchedulerConfig:
 @dataclass
 class ParallelConfig:
     tensor_parallel_size: int = 1
+    pipeline_parallel_size: int = 1
     rank: int = 0
     # disaggregated prefill role: None | "prefill" | "decode"
     kv_role: Optional[str] = None

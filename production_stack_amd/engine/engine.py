@@ -45,7 +45,10 @@ class LLMEngine:
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
-        pstate.init_distributed(config.parallel.tensor_parallel_size)
+        if config.parallel.pipeline_parallel_size > 1:
+            pstate.init_distributed(1)  # join the world group; PP uses p2p
+        else:
+            pstate.init_distributed(config.parallel.tensor_parallel_size)
         torch.manual_seed(config.seed)
         self.tokenizer = get_tokenizer(
             config.tokenizer, self.model_cfg.vocab_size
@@ -206,6 +209,20 @@ class LLMEngine:
             "generation_tokens_total": float(self.stats.generation_tokens),
             **(self.host_pool.metrics() if self.host_pool else {}),
         }
+
+    # ---- pipeline parallelism ------------------------------------------
+    @property
+    def is_pp_worker(self) -> bool:
+        return self.runner.pp_size > 1 and self.runner.pp_rank > 0
+
+    def run_pp_worker(self) -> None:
+        """Blocks serving pipeline stages (ranks > 0)."""
+        assert self.is_pp_worker
+        self.runner.pipeline.worker_loop()
+
+    def stop_pp_workers(self) -> None:
+        if self.runner.pp_size > 1 and self.runner.pp_rank == 0:
+            self.runner.pipeline.stop_workers()
 
     # ---- sleep / wake (reference request.py:1041-1128 parity) ----------
     def sleep(self, level: int = 1) -> None:

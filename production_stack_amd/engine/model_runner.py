@@ -31,7 +31,23 @@ class ModelRunner:
         self.model_cfg = model_cfg
         self.device = device
         tp = config.parallel.tensor_parallel_size
-        self.model = LlamaForCausalLM(model_cfg, tp=tp).to(device)
+        pp = getattr(config.parallel, "pipeline_parallel_size", 1)
+        self.pp_size = pp
+        self.pp_rank = 0
+        if pp > 1:
+            import torch.distributed as dist
+
+            self.pp_rank = dist.get_rank() if dist.is_initialized() else 0
+        self.model = LlamaForCausalLM(
+            model_cfg, tp=tp, pp_rank=self.pp_rank, pp_size=pp
+        ).to(device)
+        self.pipeline = None
+        if pp > 1:
+            from production_stack_amd.engine.pipeline import (
+                PipelineCoordinator,
+            )
+
+            self.pipeline = PipelineCoordinator(self, self.pp_rank, pp)
         if config.weights_path:
             from production_stack_amd.engine.weights import load_safetensors
 
@@ -68,7 +84,7 @@ class ModelRunner:
         bs = self.config.cache.block_size
         kh = self.model.kv_heads
         self.kv_caches = []
-        for _ in range(cfg.num_layers):
+        for _ in range(self.model.num_local_layers):
             k = torch.zeros(
                 (num_blocks, kh, bs, cfg.head_dim),
                 dtype=torch.bfloat16,
@@ -79,7 +95,11 @@ class ModelRunner:
 
     def capture_decode_graphs(self, max_batch: int) -> None:
         """hipGraph-capture decode-only steps (see graph_runner.py)."""
-        if self.device.type != "cuda" or self.config.enforce_eager:
+        if (
+            self.device.type != "cuda"
+            or self.config.enforce_eager
+            or self.pp_size > 1
+        ):
             return
         from production_stack_amd.engine.graph_runner import DecodeGraphRunner
 
@@ -267,6 +287,14 @@ class ModelRunner:
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
         if token_t.numel() == 0:
             return {}
+        if self.pipeline is not None:
+            tokens = self.pipeline.drive(
+                token_t, meta, rows_t, [s.params for s in sample_seqs]
+            )
+            return {
+                seq.request_id: int(tok)
+                for seq, tok in zip(sample_seqs, tokens)
+            }
         hidden = self.model(token_t, meta, self.kv_caches)
         if not sample_seqs:
             return {}
@@ -281,13 +309,17 @@ class ModelRunner:
     def sample(
         self, logits: torch.Tensor, seqs: List[Sequence]
     ) -> torch.Tensor:
-        greedy_mask = [s.params.greedy for s in seqs]
-        result = torch.empty(len(seqs), dtype=torch.long)
+        return self.sample_params(logits, [s.params for s in seqs])
+
+    def sample_params(
+        self, logits: torch.Tensor, params: List
+    ) -> torch.Tensor:
+        greedy_mask = [p.greedy for p in params]
+        result = torch.empty(len(params), dtype=torch.long)
         if all(greedy_mask):
             return ops.greedy_sample(logits).cpu()
         lf = logits.float()
-        for i, seq in enumerate(seqs):
-            p = seq.params
+        for i, p in enumerate(params):
             row = lf[i]
             if p.greedy:
                 result[i] = int(row.argmax())

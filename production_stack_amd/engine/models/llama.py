@@ -198,22 +198,43 @@ class LlamaLayer(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
-    def __init__(self, cfg: ModelConfig, tp: int = 1) -> None:
+    """Optionally a pipeline stage: with pp_size > 1 this rank holds only
+    layers [layer_start, layer_end); embedding lives on the first stage,
+    final norm + lm_head on the last."""
+
+    def __init__(
+        self, cfg: ModelConfig, tp: int = 1, pp_rank: int = 0,
+        pp_size: int = 1,
+    ) -> None:
         super().__init__()
         self.cfg = cfg
         self.tp = tp
-        self.embed = nn.Parameter(
-            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16)
-        )
+        self.pp_rank = pp_rank
+        self.pp_size = pp_size
+        per = (cfg.num_layers + pp_size - 1) // pp_size
+        self.layer_start = pp_rank * per
+        self.layer_end = min(cfg.num_layers, self.layer_start + per)
+        self.is_first = pp_rank == 0
+        self.is_last = pp_rank == pp_size - 1
+        if self.is_first:
+            self.embed = nn.Parameter(
+                torch.empty(
+                    cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16
+                )
+            )
         self.layers = nn.ModuleList(
-            LlamaLayer(cfg, tp) for _ in range(cfg.num_layers)
+            LlamaLayer(cfg, tp)
+            for _ in range(self.layer_end - self.layer_start)
         )
-        self.final_norm = nn.Parameter(
-            torch.empty(cfg.hidden_size, dtype=torch.bfloat16)
-        )
-        self.lm_head = nn.Parameter(
-            torch.empty(cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16)
-        )
+        if self.is_last:
+            self.final_norm = nn.Parameter(
+                torch.empty(cfg.hidden_size, dtype=torch.bfloat16)
+            )
+            self.lm_head = nn.Parameter(
+                torch.empty(
+                    cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16
+                )
+            )
         self.register_buffer(
             "cos_sin",
             build_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_theta),
@@ -222,7 +243,7 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.no_grad()
     def random_init(self, seed: int = 0) -> None:
-        dev = self.embed.device
+        dev = next(self.parameters()).device
         gen = torch.Generator(device=dev).manual_seed(seed)
         for name, p in self.named_parameters():
             if "norm" in name:
@@ -234,6 +255,10 @@ class LlamaForCausalLM(nn.Module):
     def kv_heads(self) -> int:
         return max(self.cfg.num_kv_heads // self.tp, 1)
 
+    @property
+    def num_local_layers(self) -> int:
+        return self.layer_end - self.layer_start
+
     def kv_bytes_per_block(self, block_size: int) -> int:
         return (
             2  # k and v
@@ -241,22 +266,29 @@ class LlamaForCausalLM(nn.Module):
             * block_size
             * self.cfg.head_dim
             * 2  # bf16
-            * self.cfg.num_layers
+            * self.num_local_layers
         )
 
     @torch.no_grad()
     def forward(
         self,
-        token_ids: torch.Tensor,  # [T] long
+        token_ids: torch.Tensor,  # [T] long (first stage) | hidden [T, H]
         meta: BatchMeta,
         kv_caches: List[tuple],
+        residual: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
-        hidden = F.embedding(token_ids, self.embed)
-        residual = None
+        if self.is_first:
+            hidden = F.embedding(token_ids, self.embed)
+            residual = None
+        else:
+            hidden = token_ids  # mid-pipeline: activations from prev stage
         for layer, cache in zip(self.layers, kv_caches):
             hidden, residual = layer(
                 hidden, residual, meta, cache, self.cos_sin
             )
+        if not self.is_last:
+            # caller ships (hidden, residual) to the next stage
+            return torch.stack([hidden, residual])
         # final residual add + norm
         hidden, _ = ops.fused_add_rms_norm(
             hidden, residual, self.final_norm, self.cfg.rms_norm_eps

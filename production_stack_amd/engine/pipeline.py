@@ -1,0 +1,140 @@
+"""Pipeline parallelism: stage-partitioned engine over torch.distributed.
+
+Capability parity: the reference exposes PP through KubeRay +
+`--pipeline-parallel-size` (reference helm/templates/ray-cluster.yaml:716);
+here it is native: rank 0 runs the scheduler and the first stage, per-step
+metadata is broadcast to all stages, activations (hidden + residual) move
+stage-to-stage with send/recv (RCCL p2p over xGMI on GPU, gloo on CPU), and
+the last stage samples and returns tokens to rank 0.
+
+Synchronous single-microbatch pipeline: correctness-first; the continuous
+batching scheduler keeps every stage busy across steps because each step
+carries the whole running batch.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from production_stack_amd.engine.models.llama import BatchMeta
+from production_stack_amd.engine.sampling import SamplingParams
+
+logger = logging.getLogger("engine.pipeline")
+
+
+def _meta_to_payload(meta: BatchMeta) -> dict:
+    def cpu(t):
+        return t.cpu() if t is not None else None
+
+    return {
+        "positions": cpu(meta.positions),
+        "slot_mapping": cpu(meta.slot_mapping),
+        "num_prefill_tokens": meta.num_prefill_tokens,
+        "prefill_token_seq": cpu(meta.prefill_token_seq),
+        "prefill_token_pos": cpu(meta.prefill_token_pos),
+        "prefill_block_tables": cpu(meta.prefill_block_tables),
+        "num_decode_seqs": meta.num_decode_seqs,
+        "decode_seq_lens": cpu(meta.decode_seq_lens),
+        "decode_block_tables": cpu(meta.decode_block_tables),
+        "prefill_tiles": cpu(meta.prefill_tiles),
+    }
+
+
+def _payload_to_meta(p: dict, device: torch.device) -> BatchMeta:
+    def dev(t):
+        return t.to(device) if t is not None else None
+
+    return BatchMeta(
+        positions=dev(p["positions"]),
+        slot_mapping=dev(p["slot_mapping"]),
+        num_prefill_tokens=p["num_prefill_tokens"],
+        prefill_token_seq=dev(p["prefill_token_seq"]),
+        prefill_token_pos=dev(p["prefill_token_pos"]),
+        prefill_block_tables=dev(p["prefill_block_tables"]),
+        num_decode_seqs=p["num_decode_seqs"],
+        decode_seq_lens=dev(p["decode_seq_lens"]),
+        decode_block_tables=dev(p["decode_block_tables"]),
+        prefill_tiles=dev(p["prefill_tiles"]),
+    )
+
+
+class PipelineCoordinator:
+    def __init__(self, runner, pp_rank: int, pp_size: int) -> None:
+        self.runner = runner
+        self.rank = pp_rank
+        self.size = pp_size
+        self.device = runner.device
+        self.hidden = runner.model_cfg.hidden_size
+
+    # ---- rank 0 (driver) ---------------------------------------------
+    @torch.no_grad()
+    def drive(
+        self,
+        token_t: torch.Tensor,
+        meta: BatchMeta,
+        sample_rows: torch.Tensor,
+        params: List[SamplingParams],
+    ) -> torch.Tensor:
+        """Run one step through the pipeline; returns sampled token ids."""
+        payload = {
+            "op": "step",
+            "meta": _meta_to_payload(meta),
+            "tokens": token_t.cpu(),
+            "sample_rows": sample_rows.cpu(),
+            "params": [
+                (p.greedy, p.temperature, p.top_p, p.top_k) for p in params
+            ],
+        }
+        dist.broadcast_object_list([payload], src=0)
+        act = self.runner.model(token_t, meta, self.runner.kv_caches)
+        dist.send(act.contiguous(), dst=1)
+        sampled = torch.empty(len(params), dtype=torch.long)
+        if params:
+            dist.recv(sampled, src=self.size - 1)
+        return sampled
+
+    def stop_workers(self) -> None:
+        try:
+            dist.broadcast_object_list([{"op": "stop"}], src=0)
+        except RuntimeError:
+            pass
+
+    # ---- ranks 1..size-1 ----------------------------------------------
+    @torch.no_grad()
+    def worker_loop(self) -> None:
+        model = self.runner.model
+        while True:
+            box = [None]
+            dist.broadcast_object_list(box, src=0)
+            payload = box[0]
+            if payload is None or payload.get("op") == "stop":
+                logger.info("pp worker rank %d stopping", self.rank)
+                return
+            meta = _payload_to_meta(payload["meta"], self.device)
+            T = payload["tokens"].shape[0]
+            act = torch.empty(
+                (2, T, self.hidden), dtype=torch.bfloat16, device=self.device
+            )
+            dist.recv(act, src=self.rank - 1)
+            hidden, residual = act[0], act[1]
+            out = model(hidden, meta, self.runner.kv_caches, residual)
+            if not model.is_last:
+                dist.send(out.contiguous(), dst=self.rank + 1)
+                continue
+            params_raw = payload["params"]
+            if not params_raw:
+                continue
+            rows = payload["sample_rows"].to(self.device)
+            logits = model.compute_logits(out[rows])
+            params = [
+                SamplingParams(
+                    temperature=t, top_p=tp, top_k=tk, max_tokens=1
+                )
+                for (_, t, tp, tk) in params_raw
+            ]
+            sampled = self.runner.sample_params(logits, params)
+            dist.send(sampled.cpu(), dst=0)

@@ -39,15 +39,19 @@ def load_safetensors(model, path: str) -> None:
     def get(name):
         return state[name].to(torch.bfloat16)
 
+    layer_start = getattr(model, "layer_start", 0)
     with torch.no_grad():
-        model.embed.copy_(get("model.embed_tokens.weight"))
-        model.final_norm.copy_(get("model.norm.weight"))
-        if "lm_head.weight" in state:
-            model.lm_head.copy_(get("lm_head.weight"))
-        else:  # tied embeddings
-            model.lm_head.copy_(get("model.embed_tokens.weight"))
+        if hasattr(model, "embed"):
+            model.embed.copy_(get("model.embed_tokens.weight"))
+        if hasattr(model, "final_norm"):
+            model.final_norm.copy_(get("model.norm.weight"))
+        if hasattr(model, "lm_head"):
+            if "lm_head.weight" in state:
+                model.lm_head.copy_(get("lm_head.weight"))
+            else:  # tied embeddings
+                model.lm_head.copy_(get("model.embed_tokens.weight"))
         for i, layer in enumerate(model.layers):
-            pre = f"model.layers.{i}."
+            pre = f"model.layers.{layer_start + i}."
             q = shard(get(pre + "self_attn.q_proj.weight"), 0)
             k = shard(get(pre + "self_attn.k_proj.weight"), 0)
             v = shard(get(pre + "self_attn.v_proj.weight"), 0)
@@ -75,8 +79,9 @@ def save_hf_safetensors(model, path: str) -> None:
         "model.norm.weight": model.final_norm.detach().cpu(),
         "lm_head.weight": model.lm_head.detach().cpu(),
     }
+    layer_start = getattr(model, "layer_start", 0)
     for i, layer in enumerate(model.layers):
-        pre = f"model.layers.{i}."
+        pre = f"model.layers.{layer_start + i}."
         qs = layer.q_heads * layer.head_dim
         kvs = layer.kv_heads * layer.head_dim
         qkv = layer.qkv_proj.detach().cpu()
