@@ -7,6 +7,9 @@ HBM GB/s for decode, at Llama-3-8B shapes.
 import sys
 import time
 
+import os
+import sys
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", ".."))
 import torch
 
 from production_stack_amd import _C

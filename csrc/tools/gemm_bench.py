@@ -4,6 +4,9 @@ TunableOp when PYTORCH_TUNABLEOP_ENABLED=1)."""
 import os
 import time
 
+import os
+import sys
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", ".."))
 import torch
 
 SHAPES = [  # (out_features, in_features) for Llama-3-8B layer GEMMs

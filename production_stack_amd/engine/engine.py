@@ -124,7 +124,8 @@ class LLMEngine:
             self.host_pool.make_compute_wait()
         sampled = self.runner.execute(out, self.block_manager)
         finished = self.scheduler.on_step_done(
-            out, sampled, self.model_cfg.eos_token_id
+            out, sampled, self.model_cfg.eos_token_id,
+            detok=self.tokenizer.decode,
         )
         now = time.time()
         results: List[RequestOutput] = []

@@ -187,6 +187,7 @@ class Scheduler:
         output: SchedulerOutput,
         sampled: Dict[str, int],
         eos_token_id: int,
+        detok=None,
     ) -> List[Sequence]:
         """Advance state after the model ran. Returns newly finished seqs."""
         finished: List[Sequence] = []
@@ -200,10 +201,14 @@ class Scheduler:
                 seq.append_token(tok)
                 seq.num_computed = min(seq.num_computed, seq.num_tokens - 1)
                 p = seq.params
-                if (
+                stop_hit = (
                     (tok == eos_token_id and not p.ignore_eos)
                     or tok in p.stop_token_ids
-                ):
+                )
+                if not stop_hit and p.stop and detok is not None:
+                    tail = detok(seq.output_token_ids[-16:])
+                    stop_hit = any(st in tail for st in p.stop)
+                if stop_hit:
                     seq.status = SeqStatus.FINISHED_STOPPED
                 elif len(seq.output_token_ids) >= p.max_tokens:
                     seq.status = SeqStatus.FINISHED_LENGTH

@@ -43,12 +43,17 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
     temperature = body.get("temperature")
     if temperature is None:
         temperature = 1.0
+    stop = body.get("stop")
+    if isinstance(stop, str):
+        stop = [stop]
     return SamplingParams(
         max_tokens=int(max_tokens),
         temperature=float(temperature),
         top_p=float(body.get("top_p") or 1.0),
         top_k=int(body.get("top_k") or -1),
         ignore_eos=bool(body.get("ignore_eos", False)),
+        stop=list(stop) if stop else [],
+        stop_token_ids=list(body.get("stop_token_ids") or []),
         seed=body.get("seed"),
     )
 
@@ -200,6 +205,160 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         body = await request.json()
         app.state.lora_adapters.pop(body.get("lora_name"), None)
         return {"status": "ok"}
+
+    # ---- embeddings / rerank / score -----------------------------------
+    def _embed_texts(texts):
+        import torch
+
+        from production_stack_amd.engine.models.llama import BatchMeta
+
+        vecs = []
+        with engine.lock:
+            runner = engine.runner
+            bm = engine.block_manager
+            for text in texts:
+                toks = engine.tokenizer.encode(text)[
+                    : engine.config.max_model_len - 1
+                ] or [engine.model_cfg.bos_token_id]
+                # mean-pool final hidden states over a scratch forward that
+                # borrows cache blocks and returns them immediately
+                n_blocks = (len(toks) + bm.block_size - 1) // bm.block_size
+                blocks = []
+                for _ in range(n_blocks):
+                    b = bm._pop_block()
+                    if b is None:
+                        break
+                    blocks.append(b)
+                if len(blocks) < n_blocks:
+                    for b in blocks:
+                        bm.free.append(b)
+                    raise RuntimeError("no KV blocks free for embedding")
+                dev = runner.device
+                T = len(toks)
+                bs = bm.block_size
+                slots = [
+                    blocks[i // bs] * bs + i % bs for i in range(T)
+                ]
+                meta = BatchMeta(
+                    positions=torch.arange(
+                        T, dtype=torch.int32, device=dev
+                    ),
+                    slot_mapping=torch.tensor(
+                        slots, dtype=torch.long, device=dev
+                    ),
+                    num_prefill_tokens=T,
+                    prefill_token_seq=torch.zeros(
+                        T, dtype=torch.int32, device=dev
+                    ),
+                    prefill_token_pos=torch.arange(
+                        T, dtype=torch.int32, device=dev
+                    ),
+                    prefill_block_tables=torch.tensor(
+                        [blocks], dtype=torch.int32, device=dev
+                    ),
+                    num_decode_seqs=0,
+                    decode_seq_lens=None,
+                    decode_block_tables=None,
+                    prefill_tiles=torch.tensor(
+                        [[0, t0, t0, min(64, T - t0)]
+                         for t0 in range(0, T, 64)],
+                        dtype=torch.int32,
+                        device=dev,
+                    ),
+                )
+                hidden = runner.model(
+                    torch.tensor(toks, dtype=torch.long, device=dev),
+                    meta,
+                    runner.kv_caches,
+                )
+                v = hidden.float().mean(dim=0)
+                v = v / (v.norm() + 1e-6)
+                vecs.append(v.cpu().tolist())
+                for b in blocks:
+                    bm.free.append(b)
+        return vecs
+
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request):
+        body = await request.json()
+        inp = body.get("input")
+        texts = [inp] if isinstance(inp, str) else list(inp or [])
+        try:
+            vecs = await asyncio.to_thread(_embed_texts, texts)
+        except RuntimeError as e:
+            return JSONResponse(status_code=503, content={"error": str(e)})
+        return {
+            "object": "list",
+            "model": body.get("model", served_model),
+            "data": [
+                {"object": "embedding", "index": i, "embedding": v}
+                for i, v in enumerate(vecs)
+            ],
+            "usage": {"prompt_tokens": sum(len(t.split()) for t in texts),
+                      "total_tokens": sum(len(t.split()) for t in texts)},
+        }
+
+    async def _rerank_impl(request: Request):
+        body = await request.json()
+        query = body.get("query", "")
+        docs = body.get("documents") or []
+        texts = [query] + [
+            d if isinstance(d, str) else d.get("text", "") for d in docs
+        ]
+        try:
+            vecs = await asyncio.to_thread(_embed_texts, texts)
+        except RuntimeError as e:
+            return JSONResponse(status_code=503, content={"error": str(e)})
+        import math
+
+        qv = vecs[0]
+        results = []
+        for i, dv in enumerate(vecs[1:]):
+            score = sum(a * b for a, b in zip(qv, dv))
+            results.append(
+                {"index": i, "relevance_score": score,
+                 "document": docs[i] if isinstance(docs[i], dict)
+                 else {"text": docs[i]}}
+            )
+        results.sort(key=lambda r: -r["relevance_score"])
+        return {"model": body.get("model", served_model),
+                "results": results}
+
+    @app.post("/v1/rerank")
+    async def rerank(request: Request):
+        return await _rerank_impl(request)
+
+    @app.post("/rerank")
+    async def rerank_alias(request: Request):
+        return await _rerank_impl(request)
+
+    async def _score_impl(request: Request):
+        body = await request.json()
+        t1 = body.get("text_1", "")
+        t2s = body.get("text_2")
+        t2s = [t2s] if isinstance(t2s, str) else list(t2s or [])
+        try:
+            vecs = await asyncio.to_thread(_embed_texts, [t1] + t2s)
+        except RuntimeError as e:
+            return JSONResponse(status_code=503, content={"error": str(e)})
+        qv = vecs[0]
+        return {
+            "object": "list",
+            "model": body.get("model", served_model),
+            "data": [
+                {"index": i,
+                 "score": sum(a * b for a, b in zip(qv, dv))}
+                for i, dv in enumerate(vecs[1:])
+            ],
+        }
+
+    @app.post("/v1/score")
+    async def score(request: Request):
+        return await _score_impl(request)
+
+    @app.post("/score")
+    async def score_alias(request: Request):
+        return await _score_impl(request)
 
     # ---- completions ---------------------------------------------------
     async def _run_completion(request: Request, chat: bool):

@@ -186,3 +186,80 @@ def test_concurrent_streaming_requests():
         assert all(n == 6 for n in results)
 
     with_server(go)
+
+
+def test_embeddings_endpoint():
+    async def go(client):
+        r = await client.post(
+            "/v1/embeddings",
+            json={"model": "tiny-llama", "input": ["hello world", "bye"]},
+            timeout=60,
+        )
+        assert r.status_code == 200, r.text
+        data = r.json()["data"]
+        assert len(data) == 2
+        assert len(data[0]["embedding"]) == 128  # hidden size
+
+    with_server(go)
+
+
+def test_rerank_and_score_endpoints():
+    async def go(client):
+        r = await client.post(
+            "/v1/rerank",
+            json={
+                "model": "tiny-llama",
+                "query": "apples and oranges",
+                "documents": ["apples and oranges fruit", "quantum physics"],
+            },
+            timeout=60,
+        )
+        assert r.status_code == 200, r.text
+        results = r.json()["results"]
+        assert len(results) == 2
+        assert results[0]["relevance_score"] >= results[1]["relevance_score"]
+        r = await client.post(
+            "/score",
+            json={"model": "tiny-llama", "text_1": "a", "text_2": ["a", "b"]},
+            timeout=60,
+        )
+        assert r.status_code == 200
+        assert len(r.json()["data"]) == 2
+
+    with_server(go)
+
+
+def test_stop_string():
+    async def go(client):
+        # find tokens the model produces, use one of their words as stop
+        r = await client.post(
+            "/v1/completions",
+            json={
+                "model": "tiny-llama",
+                "prompt": "alpha beta gamma",
+                "max_tokens": 8,
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+            timeout=60,
+        )
+        text = r.json()["choices"][0]["text"].split()
+        assert text
+        stop_word = text[1] if len(text) > 1 else text[0]
+        r = await client.post(
+            "/v1/completions",
+            json={
+                "model": "tiny-llama",
+                "prompt": "alpha beta gamma",
+                "max_tokens": 8,
+                "temperature": 0,
+                "ignore_eos": True,
+                "stop": [stop_word],
+            },
+            timeout=60,
+        )
+        out = r.json()
+        assert out["choices"][0]["finish_reason"] == "stop"
+        assert out["usage"]["completion_tokens"] <= len(text)
+
+    with_server(go)
