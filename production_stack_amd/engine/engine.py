@@ -81,6 +81,8 @@ class LLMEngine:
         )
         self.stats = EngineStats()
         self._sleeping = False
+        self.lora_adapters: Dict[str, object] = {}
+        self.runner.lora_registry = self.lora_adapters
         # coarse lock: engine-loop steps vs out-of-band block mutations
         # (KV transfer registration/adoption)
         self.lock = threading.RLock()
@@ -92,15 +94,19 @@ class LLMEngine:
         prompt: Union[str, List[int]],
         params: SamplingParams,
         arrival_time: Optional[float] = None,
+        lora_name: Optional[str] = None,
     ) -> None:
         params.validate(self.config.max_model_len)
+        if lora_name is not None and lora_name not in self.lora_adapters:
+            raise ValueError(f"unknown LoRA adapter {lora_name!r}")
         if isinstance(prompt, str):
             token_ids = self.tokenizer.encode(prompt)
         else:
             token_ids = list(prompt)
         if not token_ids:
             token_ids = [self.model_cfg.bos_token_id]
-        seq = Sequence(request_id, token_ids, params, arrival_time)
+        seq = Sequence(request_id, token_ids, params, arrival_time,
+                       lora_name=lora_name)
         self.scheduler.add(seq)
         self.stats.num_requests += 1
         self.stats.prompt_tokens += len(token_ids)
@@ -210,6 +216,17 @@ class LLMEngine:
             "generation_tokens_total": float(self.stats.generation_tokens),
             **(self.host_pool.metrics() if self.host_pool else {}),
         }
+
+    # ---- LoRA ----------------------------------------------------------
+    def load_lora(self, name: str, path: str) -> None:
+        from production_stack_amd.engine.lora import LoRAAdapter
+
+        self.lora_adapters[name] = LoRAAdapter.load(
+            name, path, self.device
+        )
+
+    def unload_lora(self, name: str) -> None:
+        self.lora_adapters.pop(name, None)
 
     # ---- pipeline parallelism ------------------------------------------
     @property

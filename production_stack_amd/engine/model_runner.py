@@ -57,6 +57,7 @@ class ModelRunner:
         self.kv_caches: List[Tuple[torch.Tensor, torch.Tensor]] = []
         self.num_blocks = 0
         self.graphs = None
+        self.lora_registry = None  # set by the engine (name -> adapter)
         self._generator = torch.Generator(device="cpu").manual_seed(
             config.seed + 12345
         )
@@ -176,7 +177,6 @@ class ModelRunner:
             sample_rows.append(len(tokens) - 1)
             sample_seqs.append(seq)
 
-        dev = self.device
         max_bt = max((len(t) for t in p_tables + d_tables), default=1)
 
         def pad_tables(tabs: List[List[int]]) -> Optional[torch.Tensor]:
@@ -188,7 +188,32 @@ class ModelRunner:
                     out_t[i, : len(t)] = torch.tensor(t, dtype=torch.int32)
             return out_t.to(dev, non_blocking=True)
 
+        dev = self.device
         num_prefill_tokens = len(p_token_seq)
+
+        # per-adapter row groups (LoRA): map flat batch rows to adapters
+        lora_rows: Dict[str, List[int]] = {}
+        row = 0
+        for ss in prefills:
+            if ss.seq.lora_name:
+                lora_rows.setdefault(ss.seq.lora_name, []).extend(
+                    range(row, row + ss.num_tokens)
+                )
+            row += ss.num_tokens
+        for ss in decodes:
+            if ss.seq.lora_name:
+                lora_rows.setdefault(ss.seq.lora_name, []).append(row)
+            row += 1
+        lora_groups = []
+        if lora_rows and self.lora_registry is not None:
+            for name, rows in lora_rows.items():
+                ad = self.lora_registry.get(name)
+                if ad is not None:
+                    lora_groups.append(
+                        (ad, torch.tensor(rows, dtype=torch.long,
+                                          device=dev))
+                    )
+
         meta = BatchMeta(
             positions=torch.tensor(positions, dtype=torch.int32).to(
                 dev, non_blocking=True
@@ -228,6 +253,7 @@ class ModelRunner:
                 if p_tiles
                 else None
             ),
+            lora_groups=lora_groups,
         )
         token_t = torch.tensor(tokens, dtype=torch.long).to(
             dev, non_blocking=True
@@ -252,6 +278,7 @@ class ModelRunner:
             or n == 0
             or self.graphs.bucket_for(n) is None
             or any(not s.is_decode for s in scheduled)
+            or any(s.seq.lora_name for s in scheduled)
         ):
             return None
         bs = bm.block_size

@@ -43,6 +43,7 @@ class BatchMeta:
         decode_seq_lens: Optional[torch.Tensor],  # [Sd] int32
         decode_block_tables: Optional[torch.Tensor],  # [Sd, MB] int32
         prefill_tiles: Optional[torch.Tensor] = None,  # [NT, 4] int32
+        lora_groups=None,  # [(LoRAAdapter, row_idx_tensor)]
     ) -> None:
         self.positions = positions
         self.slot_mapping = slot_mapping
@@ -54,6 +55,7 @@ class BatchMeta:
         self.decode_seq_lens = decode_seq_lens
         self.decode_block_tables = decode_block_tables
         self.prefill_tiles = prefill_tiles
+        self.lora_groups = lora_groups or []
 
 
 def build_cos_sin_cache(
@@ -70,9 +72,10 @@ def build_cos_sin_cache(
 
 
 class LlamaLayer(nn.Module):
-    def __init__(self, cfg: ModelConfig, tp: int) -> None:
+    def __init__(self, cfg: ModelConfig, tp: int, layer_idx: int = 0) -> None:
         super().__init__()
         self.cfg = cfg
+        self.layer_idx = layer_idx
         assert cfg.num_q_heads % tp == 0, "q heads must divide TP"
         assert cfg.num_kv_heads % tp == 0 or tp % cfg.num_kv_heads == 0
         self.q_heads = cfg.num_q_heads // tp
@@ -117,6 +120,15 @@ class LlamaLayer(nn.Module):
         qkv = F.linear(hidden, self.qkv_proj)
         qs = self.q_heads * self.head_dim
         kvs = self.kv_heads * self.head_dim
+        if meta.lora_groups:
+            from production_stack_amd.engine.lora import apply_lora_slice
+
+            li = self.layer_idx
+            apply_lora_slice(qkv, hidden, meta.lora_groups, li, "q", 0, qs)
+            apply_lora_slice(qkv, hidden, meta.lora_groups, li, "k", qs, kvs)
+            apply_lora_slice(
+                qkv, hidden, meta.lora_groups, li, "v", qs + kvs, kvs
+            )
         T = qkv.shape[0]
         k_cache, v_cache = kv_cache
         if qkv.is_cuda:
@@ -185,14 +197,32 @@ class LlamaLayer(nn.Module):
             )
         attn = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
         attn_out = F.linear(attn.view(T, qs), self.o_proj)
+        if meta.lora_groups:
+            apply_lora_slice(
+                attn_out, attn.view(T, qs), meta.lora_groups,
+                self.layer_idx, "o",
+            )
         attn_out = pstate.tp_all_reduce(attn_out)
 
         hidden, residual = ops.fused_add_rms_norm(
             attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
         )
         gate_up = F.linear(hidden, self.gate_up_proj)
+        if meta.lora_groups:
+            apply_lora_slice(
+                gate_up, hidden, meta.lora_groups, self.layer_idx,
+                "gate", 0, self.inter,
+            )
+            apply_lora_slice(
+                gate_up, hidden, meta.lora_groups, self.layer_idx,
+                "up", self.inter, self.inter,
+            )
         act = ops.silu_and_mul(gate_up)
         mlp_out = F.linear(act, self.down_proj)
+        if meta.lora_groups:
+            apply_lora_slice(
+                mlp_out, act, meta.lora_groups, self.layer_idx, "down"
+            )
         mlp_out = pstate.tp_all_reduce(mlp_out)
         return mlp_out, residual
 
@@ -223,8 +253,8 @@ class LlamaForCausalLM(nn.Module):
                 )
             )
         self.layers = nn.ModuleList(
-            LlamaLayer(cfg, tp)
-            for _ in range(self.layer_end - self.layer_start)
+            LlamaLayer(cfg, tp, layer_idx=self.layer_start + i)
+            for i in range(self.layer_end - self.layer_start)
         )
         if self.is_last:
             self.final_norm = nn.Parameter(

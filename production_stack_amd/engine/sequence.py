@@ -39,8 +39,10 @@ class Sequence:
         prompt_token_ids: List[int],
         params: SamplingParams,
         arrival_time: Optional[float] = None,
+        lora_name: Optional[str] = None,
     ) -> None:
         self.request_id = request_id
+        self.lora_name = lora_name
         self.prompt_token_ids = list(prompt_token_ids)
         self.output_token_ids: List[int] = []
         self.params = params

@@ -197,13 +197,24 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
             return JSONResponse(
                 status_code=400, content={"error": "lora_name required"}
             )
-        app.state.lora_adapters[name] = body.get("lora_path", "")
+        path = body.get("lora_path", "")
+        try:
+            if path:
+                engine.load_lora(name, path)
+        except (OSError, ValueError) as e:
+            return JSONResponse(
+                status_code=400,
+                content={"error": f"failed to load adapter: {e}"},
+            )
+        app.state.lora_adapters[name] = path
         return {"status": "ok"}
 
     @app.post("/v1/unload_lora_adapter")
     async def unload_lora(request: Request):
         body = await request.json()
-        app.state.lora_adapters.pop(body.get("lora_name"), None)
+        name = body.get("lora_name")
+        engine.unload_lora(name)
+        app.state.lora_adapters.pop(name, None)
         return {"status": "ok"}
 
     # ---- embeddings / rerank / score -----------------------------------

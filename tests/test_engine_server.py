@@ -128,11 +128,17 @@ def test_sleep_wake_cycle():
     with_server(go)
 
 
-def test_lora_load_unload_surface():
+def test_lora_load_unload_surface(tmp_path):
+    from production_stack_amd.engine.lora import save_synthetic_adapter
+
+    adir = str(tmp_path / "ad1")
+    save_synthetic_adapter(adir, hidden=128, q_size=128, kv_size=64,
+                           num_layers=2)
+
     async def go(client):
         r = await client.post(
             "/v1/load_lora_adapter",
-            json={"lora_name": "ad1", "lora_path": "/tmp/x"},
+            json={"lora_name": "ad1", "lora_path": adir},
         )
         assert r.status_code == 200
         r = await client.get("/v1/models")
