@@ -200,11 +200,15 @@ def test_router_and_cacheserver_reconcile(fake_k8s):
     )
 
 
-def test_lora_adapter_load_call(fake_k8s):
+def test_lora_adapter_load_call(fake_k8s, tmp_path):
     """LoraAdapter CR triggers /v1/load_lora_adapter on the base model's
     pods (served here by a real engine server)."""
+    from production_stack_amd.engine.lora import save_synthetic_adapter
     from tests.test_full_stack_cpu import RealEngineServer
 
+    adir = str(tmp_path / "ad1")
+    save_synthetic_adapter(adir, hidden=128, q_size=128, kv_size=64,
+                           num_layers=2)
     engine = RealEngineServer(18700)
     engine.start()
     try:
@@ -221,7 +225,7 @@ def test_lora_adapter_load_call(fake_k8s):
 
         r = requests.post(
             engine.url + "/v1/load_lora_adapter",
-            json={"lora_name": "ad1", "lora_path": "/tmp/a"},
+            json={"lora_name": "ad1", "lora_path": adir},
             timeout=5,
         )
         assert r.status_code == 200
