@@ -60,6 +60,7 @@ class EndpointInfo:
 class ServiceDiscoveryType(str, enum.Enum):
     static = "static"
     k8s_pod_ip = "k8s"
+    k8s_service_name = "k8s_service_name"
     external = "external"
 
 
@@ -293,6 +294,85 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
         self._stop.set()
 
 
+class K8sServiceNameServiceDiscovery(ServiceDiscovery):
+    """Discovers engine Services (not pods) by label selector; endpoints are
+    the stable in-cluster service DNS names. Parity: reference
+    service_discovery.py:892-1306 (K8sServiceNameServiceDiscovery)."""
+
+    def __init__(
+        self,
+        namespace: str = "default",
+        port: int = 80,
+        label_selector: Optional[str] = None,
+        api_key: Optional[str] = None,
+        refresh_interval: float = 30.0,
+    ) -> None:
+        try:
+            from kubernetes import client, config  # noqa: F401
+        except ImportError as e:  # pragma: no cover
+            raise RuntimeError(
+                "k8s service discovery requires the `kubernetes` package"
+            ) from e
+        self.namespace = namespace
+        self.port = port
+        self.label_selector = label_selector
+        self.api_key = api_key
+        self.refresh_interval = refresh_interval
+        self._endpoints: Dict[str, EndpointInfo] = {}
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        config.load_incluster_config()
+        self._core = client.CoreV1Api()
+        self._thread = threading.Thread(target=self._refresh_loop,
+                                        daemon=True)
+        self._thread.start()
+
+    def _refresh_once(self) -> None:  # pragma: no cover - needs a cluster
+        svcs = self._core.list_namespaced_service(
+            namespace=self.namespace, label_selector=self.label_selector
+        )
+        new: Dict[str, EndpointInfo] = {}
+        for svc in svcs.items:
+            name = svc.metadata.name
+            port = self.port
+            if svc.spec.ports:
+                port = svc.spec.ports[0].port
+            url = f"http://{name}.{self.namespace}.svc:{port}"
+            models: List[str] = []
+            try:
+                headers = {}
+                if self.api_key:
+                    headers["Authorization"] = f"Bearer {self.api_key}"
+                r = requests.get(url + "/v1/models", headers=headers,
+                                 timeout=5)
+                models = [m["id"] for m in r.json().get("data", [])]
+            except requests.RequestException:
+                pass
+            labels = svc.metadata.labels or {}
+            new[name] = EndpointInfo(
+                url=url,
+                model_names=models,
+                model_label=labels.get("model"),
+            )
+        with self._lock:
+            self._endpoints = new
+
+    def _refresh_loop(self) -> None:  # pragma: no cover
+        while not self._stop.is_set():
+            try:
+                self._refresh_once()
+            except Exception as e:
+                logger.warning("k8s service refresh failed: %s", e)
+            self._stop.wait(self.refresh_interval)
+
+    def get_endpoint_info(self) -> List[EndpointInfo]:
+        with self._lock:
+            return list(self._endpoints.values())
+
+    def close(self) -> None:
+        self._stop.set()
+
+
 _instance: Optional[ServiceDiscovery] = None
 
 
@@ -306,6 +386,8 @@ def initialize_service_discovery(
         _instance = StaticServiceDiscovery(**kwargs)
     elif kind in ("k8s", "k8s_pod_ip", ServiceDiscoveryType.k8s_pod_ip):
         _instance = K8sPodIpServiceDiscovery(**kwargs)
+    elif kind in ("k8s_service_name", ServiceDiscoveryType.k8s_service_name):
+        _instance = K8sServiceNameServiceDiscovery(**kwargs)
     elif kind in ("external", ServiceDiscoveryType.external):
         _instance = ExternalOnlyServiceDiscovery()
     else:
