@@ -587,6 +587,7 @@ def main() -> None:
     ap.add_argument("--enable-chunked-prefill", action="store_true",
                     default=True)
     ap.add_argument("--tensor-parallel-size", type=int, default=1)
+    ap.add_argument("--pipeline-parallel-size", type=int, default=1)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--device", default=None, help="cpu forces CPU mode")
     ap.add_argument("--weights-path", default=None)
@@ -626,6 +627,7 @@ def main() -> None:
         ),
         parallel=ParallelConfig(
             tensor_parallel_size=args.tensor_parallel_size,
+            pipeline_parallel_size=args.pipeline_parallel_size,
             kv_role=args.kv_role,
         ),
     )
@@ -641,6 +643,11 @@ def main() -> None:
         args.advertise_url or f"http://127.0.0.1:{args.port}"
     )
     engine = LLMEngine(cfg, device=args.device)
+    if engine.is_pp_worker:
+        # pipeline ranks > 0 serve activations, not HTTP
+        logging.basicConfig(level=logging.INFO)
+        engine.run_pp_worker()
+        return
     served = args.served_model_name or args.model
     app = build_server(engine, served)
     logging.basicConfig(level=logging.INFO)
