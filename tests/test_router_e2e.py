@@ -296,3 +296,60 @@ def test_disaggregated_orchestrated_flow(engines):
         d_reqs[0]["body"]["kv_transfer_params"]["remote_engine_id"]
         == "fake-engine-1"
     )
+
+
+def test_dynamic_config_hot_reload(engines, tmp_path):
+    """Editing the dynamic-config file swaps routing logic without restart
+    (reference dynamic_config.py behaviour)."""
+    import json as _json
+
+    from production_stack_amd.router.dynamic_config import (
+        initialize_dynamic_config_watcher,
+        get_dynamic_config_watcher,
+    )
+    from production_stack_amd.router.routing_logic import (
+        PrefixAwareRouter,
+        RoundRobinRouter,
+        get_routing_logic,
+    )
+
+    cfg_file = tmp_path / "dyn.json"
+    cfg_file.write_text(_json.dumps({"routing_logic": "roundrobin"}))
+    application = make_app(engines)
+    watcher = initialize_dynamic_config_watcher(
+        str(cfg_file), interval=3600, app=application, start=False
+    )
+    assert isinstance(get_routing_logic(), RoundRobinRouter)
+    cfg_file.write_text(
+        _json.dumps(
+            {"routing_logic": "prefixaware", "prefix_min_match_length": 64}
+        )
+    )
+    assert watcher.poll_once()
+    assert isinstance(get_routing_logic(), PrefixAwareRouter)
+    assert isinstance(application.state.router, PrefixAwareRouter)
+    watcher.close()
+
+
+def test_stress_even_distribution(engines):
+    """Stress-test parity (reference tests/e2e/stress-test.sh): many
+    concurrent requests through the round-robin router distribute evenly."""
+    application = make_app(engines)
+    for s in engines:
+        s.seen["requests"].clear()
+
+    async def go(client):
+        async def one(i):
+            r = await client.post(
+                "/v1/completions",
+                json={"model": "m1", "prompt": f"q{i}", "max_tokens": 1},
+            )
+            return r.status_code
+
+        results = await asyncio.gather(*(one(i) for i in range(100)))
+        assert all(c == 200 for c in results)
+
+    with_client(application, go)
+    counts = [len(s.seen["requests"]) for s in engines]
+    assert sum(counts) == 100
+    assert abs(counts[0] - counts[1]) <= 2, counts
