@@ -79,35 +79,58 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
   const int b_begin = split * per_split;
   const int b_end = min(nblocks, b_begin + per_split);
   const int* bt = block_tables + (long)seq * max_blocks;
+  // Per paged block: score all of this sub-group's tokens first
+  // (independent dot products -> independent exps), then ONE online-softmax
+  // rescale per block. The naive per-token update serializes
+  // exp->mul->fma chains 4x deeper and left the kernel latency-bound at
+  // <50% of HBM bandwidth (profiles/ attn_bench).
+  constexpr int TPB = BLOCK_SIZE / TPW;  // tokens per sub-group per block (4)
   for (int b = b_begin + wave; b < b_end; b += NWAVES) {
     const long blk = bt[b];
     const unsigned short* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
     const unsigned short* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
     const int valid_tokens = min(BLOCK_SIZE, ctx - b * BLOCK_SIZE);
+    float sc[GQ][TPB];
+    ps_bf16x8 vv[TPB];
 #pragma unroll
-    for (int tt = 0; tt < BLOCK_SIZE / TPW; tt++) {
+    for (int tt = 0; tt < TPB; tt++) {
       const int tok = tt * TPW + sg;
       const bool valid = tok < valid_tokens;
       ps_bf16x8 kv = *(const ps_bf16x8*)(kb + tok * D + sl * 8);
-      ps_bf16x8 vv = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
-      float vf[8];
-#pragma unroll
-      for (int j = 0; j < 8; j++) vf[j] = ps_bf16_to_f32(vv[j]);
+      vv[tt] = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
 #pragma unroll
       for (int g = 0; g < GQ; g++) {
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < 8; j++) s += qf[g][j] * ps_bf16_to_f32(kv[j]);
         s = ps_group_sum<LPG>(s);
-        const float sv = valid ? s : PS_NEG_INF;
-        const float mnew = fmaxf(m[g], sv);
-        const float corr = __expf(m[g] - mnew);  // <=1; 1 when both -inf
-        const float p = valid ? __expf(sv - mnew) : 0.f;
-        l[g] = l[g] * corr + p;
-#pragma unroll
-        for (int j = 0; j < 8; j++) acc[g][j] = acc[g][j] * corr + p * vf[j];
-        m[g] = mnew;
+        sc[g][tt] = valid ? s : PS_NEG_INF;
       }
+    }
+#pragma unroll
+    for (int g = 0; g < GQ; g++) {
+      float bmax = sc[g][0];
+#pragma unroll
+      for (int tt = 1; tt < TPB; tt++) bmax = fmaxf(bmax, sc[g][tt]);
+      const float mnew = fmaxf(m[g], bmax);
+      const float corr = __expf(m[g] - mnew);  // <=1; 1 when both -inf
+      float p[TPB];
+      float psum = 0.f;
+#pragma unroll
+      for (int tt = 0; tt < TPB; tt++) {
+        p[tt] = sc[g][tt] > PS_NEG_INF ? __expf(sc[g][tt] - mnew) : 0.f;
+        psum += p[tt];
+      }
+      l[g] = l[g] * corr + psum;
+#pragma unroll
+      for (int j = 0; j < 8; j++) {
+        float a = acc[g][j] * corr;
+#pragma unroll
+        for (int tt = 0; tt < TPB; tt++)
+          a += p[tt] * ps_bf16_to_f32(vv[tt][j]);
+        acc[g][j] = a;
+      }
+      m[g] = mnew;
     }
   }
 

@@ -3,16 +3,18 @@
 // Structure (MI355X-first; mfma_f32_16x16x32_bf16, per-wave tiles):
 //   grid = (num_q_tiles, num_q_heads), block = 256 (4 waves)
 //   Each workgroup owns one (q-tile of <=64 rows, q-head); wave w computes
-//   rows [16w, 16w+16). KV is consumed in 32-token chunks (2 paged blocks):
+//   rows [16w, 16w+16). KV is consumed in 64-token chunks (4 paged blocks):
 //     QK^T: A = Q[16 x 128] (registers), B = K^T via direct global 16-B lane
 //           loads (KV pages are L2-resident across the 4 waves x GQ heads
 //           that re-read them -- no LDS staging for K, guide mistake #7).
 //     softmax: online, per-row state in C-frag register layout
-//           (row = (lane>>4)*4 + reg, col = lane&15  [HW-verified]).
-//     PV:   P routed through a small per-wave LDS tile ([16][40] stride
-//           avoids bank conflicts) to convert C-layout -> A-layout;
-//           V staged TRANSPOSED in shared LDS ([128][40]) so B-frags are
-//           single ds_read_b128s.
+//           (row = (lane>>4)*4 + reg, col = lane&15  [HW-verified]); one
+//           rescale per 64-token chunk.
+//     PV:   P routed through a small per-wave LDS tile to convert
+//           C-layout -> A-layout; V staged TRANSPOSED in shared LDS with an
+//           8-token-group XOR swizzle so B-frags are single conflict-light
+//           ds_read_b128s. V staging is software-pipelined (guide T14):
+//           next chunk's global loads issue under this chunk's MFMA phase.
 //   Operand k-pattern: lane l supplies elements k = (l>>4)*8 + i for both A
 //   and B (k-permutation invariance verified on hardware: csrc/tools/
 //   mfma_probe.hip, gpurun_out/mfma_probe.log).
@@ -30,11 +32,8 @@ PS_DEV ps_mbf16x8 ps_as_mbf16(ps_bf16x8 u) {
   return v.bf;
 }
 
-// V^T is stored [128 dims][32 tokens] with the token GROUP (8-token units,
-// 16 B) XOR-swizzled by the dim's low bits: physical group = (tok>>3)^(d&3).
-// This keeps every access a 16-B-aligned ds_read/write unit while spreading
-// banks (2-way worst case on both the staging writes and the B-frag reads).
-#define PS_PL_STRIDE 40  // P row stride in tokens
+#define PS_CHUNK 64      // KV tokens per chunk (4 pages)
+#define PS_PL_STRIDE 72  // P row stride in tokens (multiple of 8)
 
 // tile_info: int4 per tile = (seq_row, q_token_start, q_pos_start, n_rows)
 template <int HEAD_DIM>
@@ -49,6 +48,8 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
   constexpr int D = HEAD_DIM;  // 128
   constexpr int BS = 16;       // page size in tokens
   constexpr int NK = D / 32;   // mfma k-steps over head dim (4)
+  constexpr int NCT = PS_CHUNK / 16;  // QK col-tiles per chunk (4)
+  constexpr int NKC = PS_CHUNK / 32;  // PV k-chunks per chunk (2)
 
   const int tile = blockIdx.x;
   const int qh = blockIdx.y;
@@ -62,16 +63,16 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
-  const int g = lane >> 4;    // 0..3
-  const int rc = lane & 15;   // row (A/C) or col (B) index
+  const int g = lane >> 4;   // 0..3
+  const int rc = lane & 15;  // row (A/C) or col (B) index
 
   const int* bt = block_tables + (long)seq_row * max_blocks;
   const int ctx_limit = q_pos0 + n_rows;  // causal bound for the tile
   const int n_pages = (ctx_limit + BS - 1) / BS;
-  const int n_chunks = (ctx_limit + 31) / 32;
+  const int n_chunks = (ctx_limit + PS_CHUNK - 1) / PS_CHUNK;
 
-  // shared: V^T staging + per-wave P tiles
-  __shared__ __align__(16) unsigned short v_t[D][32];
+  // shared: V^T staging (token-group XOR swizzle) + per-wave P tiles
+  __shared__ __align__(16) unsigned short v_t[D][PS_CHUNK];
   __shared__ __align__(16) unsigned short p_lds[4][16][PS_PL_STRIDE];
 
   // ---- load Q fragments (row rc of this wave's 16) ----
@@ -97,60 +98,52 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
 
   const int wave_pos_max = q_pos0 + min(wave * 16 + 15, n_rows - 1);
 
-  // V^T staging is software-pipelined (guide T14 async-STAGE): this
-  // thread's V row for chunk c+1 is issued as global loads during chunk c's
-  // compute phase and written to LDS after the end-of-compute barrier.
-  const int tv = tid & 31;          // staged token (0..31 within chunk)
-  const int d0 = (tid >> 5) * 16;   // staged dim range [d0, d0+16)
-  const int tg = tv >> 3;
+  // V^T staging pipeline: this thread stages token tv, dims
+  // [wave*32, wave*32+32) (4 x ushort8); loads for chunk c+1 issue during
+  // chunk c's compute phase.
+  const int tv = tid & 63;
+  const int d0 = wave * 32;
+  const int lg = tv >> 3;  // logical 8-token group within the chunk
   const int tl = tv & 7;
   auto v_row_ptr = [&](int chunk) {
-    const int pg_idx = min((chunk * 32 + tv) / BS, n_pages - 1);
+    const int tok = chunk * PS_CHUNK + tv;
+    const int pg_idx = min(tok / BS, n_pages - 1);
     const long pg = bt[pg_idx];
-    return v_cache +
-           ((pg * KH + kvh) * BS + ((chunk * 32 + tv) & (BS - 1))) * D;
+    return v_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
   };
-  ps_bf16x8 vstage[2];
+  ps_bf16x8 vstage[4];
   {
     const unsigned short* vrow = v_row_ptr(0);
-    vstage[0] = *(const ps_bf16x8*)(vrow + d0);
-    vstage[1] = *(const ps_bf16x8*)(vrow + d0 + 8);
+#pragma unroll
+    for (int h = 0; h < 4; h++)
+      vstage[h] = *(const ps_bf16x8*)(vrow + d0 + h * 8);
   }
 
   for (int chunk = 0; chunk < n_chunks; chunk++) {
-    const int tok0 = chunk * 32;
-    // write the pre-fetched V^T tile for this chunk
+    const int tok0 = chunk * PS_CHUNK;
+    // write the pre-fetched V^T tile (swizzled token group per dim)
 #pragma unroll
-    for (int h = 0; h < 2; h++) {
+    for (int h = 0; h < 4; h++) {
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         const int d = d0 + h * 8 + j;
-        v_t[d][(tg ^ (d & 3)) * 8 + tl] = vstage[h][j];
+        v_t[d][((lg ^ (d & 7)) << 3) + tl] = vstage[h][j];
       }
     }
     __syncthreads();
-    // issue next chunk's V loads now; HBM latency hides under the MFMA
-    // phase below and the ds_write happens after the trailing barrier
     if (chunk + 1 < n_chunks) {
       const unsigned short* vrow = v_row_ptr(chunk + 1);
-      vstage[0] = *(const ps_bf16x8*)(vrow + d0);
-      vstage[1] = *(const ps_bf16x8*)(vrow + d0 + 8);
+#pragma unroll
+      for (int h = 0; h < 4; h++)
+        vstage[h] = *(const ps_bf16x8*)(vrow + d0 + h * 8);
     }
 
     const bool wave_active = (wave * 16 < n_rows) && (tok0 <= wave_pos_max);
     if (wave_active) {
-      // ---- QK^T for the two 16-token column tiles ----
-      float p_vals[2][4];  // [col-tile][reg] probability values
-      float row_corr[4];
+      // ---- QK^T over NCT 16-token column tiles ----
+      ps_mf32x4 s_frag[NCT];
 #pragma unroll
-      for (int r = 0; r < 4; r++) row_corr[r] = 1.f;
-      float m_new[4];
-#pragma unroll
-      for (int r = 0; r < 4; r++) m_new[r] = m_run[r];
-
-      ps_mf32x4 s_frag[2];
-#pragma unroll
-      for (int ct = 0; ct < 2; ct++) {
+      for (int ct = 0; ct < NCT; ct++) {
         s_frag[ct] = {0.f, 0.f, 0.f, 0.f};
         const int tok = tok0 + ct * 16 + rc;  // this lane's kv token (col)
         const int pg_idx = min(tok / BS, n_pages - 1);
@@ -167,10 +160,12 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
         }
         __builtin_amdgcn_s_setprio(0);
       }
-      // ---- mask + online softmax ----
-      // lane holds S[row g*4+r][col ct*16+rc]
+      // ---- mask + online softmax (one rescale per chunk) ----
+      float m_new[4];
 #pragma unroll
-      for (int ct = 0; ct < 2; ct++) {
+      for (int r = 0; r < 4; r++) m_new[r] = m_run[r];
+#pragma unroll
+      for (int ct = 0; ct < NCT; ct++) {
         const int kv_pos = tok0 + ct * 16 + rc;
 #pragma unroll
         for (int r = 0; r < 4; r++) {
@@ -182,32 +177,30 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
           m_new[r] = fmaxf(m_new[r], sv);
         }
       }
-      // row max across the 16 lanes of this g-group
 #pragma unroll
       for (int r = 0; r < 4; r++) m_new[r] = ps_group_max<16>(m_new[r]);
+      float row_corr[4];
 #pragma unroll
       for (int r = 0; r < 4; r++) {
         row_corr[r] = __expf(m_run[r] - m_new[r]);
         l_run[r] *= row_corr[r];
         m_run[r] = m_new[r];
       }
+      float psum[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int ct = 0; ct < 2; ct++) {
+      for (int ct = 0; ct < NCT; ct++) {
 #pragma unroll
         for (int r = 0; r < 4; r++) {
-          const int lrow = wave * 16 + g * 4 + r;
-          const int kv_pos = tok0 + ct * 16 + rc;
-          const int q_pos = q_pos0 + lrow;
-          const bool valid = (lrow < n_rows) && (kv_pos <= q_pos);
-          const float p = valid ? __expf(s_frag[ct][r] - m_new[r]) : 0.f;
-          p_vals[ct][r] = p;
+          const float p = s_frag[ct][r] > PS_NEG_INF
+                              ? __expf(s_frag[ct][r] - m_new[r])
+                              : 0.f;
+          s_frag[ct][r] = p;
+          psum[r] += p;
         }
       }
-      // row-sum of p across 16 lanes
 #pragma unroll
       for (int r = 0; r < 4; r++) {
-        float rs = p_vals[0][r] + p_vals[1][r];
-        rs = ps_group_sum<16>(rs);
+        float rs = ps_group_sum<16>(psum[r]);
         l_run[r] += rs;
       }
       // rescale O
@@ -217,26 +210,29 @@ __global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
         for (int r = 0; r < 4; r++) o_acc[s][r] *= row_corr[r];
       // ---- write P to per-wave LDS (C-layout -> A-layout transpose) ----
 #pragma unroll
-      for (int ct = 0; ct < 2; ct++)
+      for (int ct = 0; ct < NCT; ct++)
 #pragma unroll
         for (int r = 0; r < 4; r++)
           p_lds[wave][g * 4 + r][ct * 16 + rc] =
-              ps_f32_to_bf16(p_vals[ct][r]);
-      // ---- PV ----
-      // A-frag: P[row rc][tokens g*8 .. g*8+8)
-      ps_mbf16x8 p_frag =
-          ps_as_mbf16(*(const ps_bf16x8*)(&p_lds[wave][rc][g * 8]));
-      __builtin_amdgcn_s_setprio(1);
+              ps_f32_to_bf16(s_frag[ct][r]);
+      // ---- PV: o[16 x 128] += P[16 x 64] @ V[64 x 128] ----
 #pragma unroll
-      for (int s = 0; s < 8; s++) {
-        // B-frag: V^T[dim s*16 + rc][tokens g*8 .. +8) (swizzled group)
-        const int d = s * 16 + rc;
-        ps_mbf16x8 v_frag = ps_as_mbf16(
-            *(const ps_bf16x8*)(&v_t[d][(g ^ (d & 3)) * 8]));
-        o_acc[s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            p_frag, v_frag, o_acc[s], 0, 0, 0);
+      for (int kc = 0; kc < NKC; kc++) {
+        // A-frag: P[row rc][tokens kc*32 + g*8 ..+8)
+        ps_mbf16x8 p_frag = ps_as_mbf16(
+            *(const ps_bf16x8*)(&p_lds[wave][rc][kc * 32 + g * 8]));
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int s = 0; s < 8; s++) {
+          const int d = s * 16 + rc;
+          const int pg2 = (kc * 4 + g) ^ (d & 7);
+          ps_mbf16x8 v_frag =
+              ps_as_mbf16(*(const ps_bf16x8*)(&v_t[d][pg2 << 3]));
+          o_acc[s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              p_frag, v_frag, o_acc[s], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();  // protect v_t before next chunk's staging
   }
