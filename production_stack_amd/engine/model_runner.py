@@ -341,35 +341,52 @@ class ModelRunner:
     def sample_params(
         self, logits: torch.Tensor, params: List
     ) -> torch.Tensor:
-        greedy_mask = [p.greedy for p in params]
+        """Batched sampling: greedy rows via the HIP argmax kernel, the rest
+        through grouped tensor ops (temperature -> per-group top-k ->
+        sorted-cumsum top-p -> multinomial), one batch per distinct top_k."""
+        greedy_mask = torch.tensor([p.greedy for p in params])
         result = torch.empty(len(params), dtype=torch.long)
-        if all(greedy_mask):
+        if bool(greedy_mask.all()):
             return ops.greedy_sample(logits).cpu()
-        lf = logits.float()
-        for i, p in enumerate(params):
-            row = lf[i]
-            if p.greedy:
-                result[i] = int(row.argmax())
-                continue
-            row = row / max(p.temperature, 1e-5)
-            if p.top_k > 0 and p.top_k < row.shape[-1]:
-                kth = torch.topk(row, p.top_k).values[-1]
-                row = row.masked_fill(row < kth, float("-inf"))
-            probs = torch.softmax(row, dim=-1)
-            if p.top_p < 1.0:
-                sp, si = torch.sort(probs, descending=True)
+        if bool(greedy_mask.any()):
+            g_rows = greedy_mask.nonzero().flatten()
+            g_idx = g_rows.to(logits.device)
+            result[g_rows] = ops.greedy_sample(
+                logits.index_select(0, g_idx)
+            ).cpu()
+        sampled_rows = (~greedy_mask).nonzero().flatten()
+        groups: Dict[int, List[int]] = {}
+        for i in sampled_rows.tolist():
+            groups.setdefault(params[i].top_k, []).append(i)
+        for top_k, rows in groups.items():
+            idx = torch.tensor(rows, dtype=torch.long, device=logits.device)
+            lf = logits.index_select(0, idx).float()
+            temps = torch.tensor(
+                [max(params[i].temperature, 1e-5) for i in rows],
+                device=lf.device,
+            ).unsqueeze(1)
+            lf = lf / temps
+            if 0 < top_k < lf.shape[-1]:
+                kth = torch.topk(lf, top_k, dim=-1).values[:, -1:]
+                lf = lf.masked_fill(lf < kth, float("-inf"))
+            probs = torch.softmax(lf, dim=-1)
+            top_ps = torch.tensor(
+                [params[i].top_p for i in rows], device=lf.device
+            ).unsqueeze(1)
+            if bool((top_ps < 1.0).any()):
+                sp, si = torch.sort(probs, descending=True, dim=-1)
                 cum = torch.cumsum(sp, dim=-1)
-                keep = cum - sp < p.top_p
-                keep[0] = True
+                keep = (cum - sp) < top_ps
+                keep[:, 0] = True
                 sp = sp * keep
-                sp = sp / sp.sum()
-                idx = torch.multinomial(
+                sp = sp / sp.sum(dim=-1, keepdim=True)
+                picks = torch.multinomial(
                     sp.cpu(), 1, generator=self._generator
                 )
-                result[i] = int(si[idx])
+                chosen = torch.gather(si.cpu(), 1, picks).flatten()
             else:
-                idx = torch.multinomial(
+                chosen = torch.multinomial(
                     probs.cpu(), 1, generator=self._generator
-                )
-                result[i] = int(idx)
+                ).flatten()
+            result[torch.tensor(rows)] = chosen
         return result
