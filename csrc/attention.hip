@@ -28,7 +28,7 @@
 // sizes of 16-64 seqs, seqs*kv_heads workgroups cannot fill 256 CUs and the
 // kernel runs at <10% of HBM bandwidth (measured, profiles/ run1).
 // ---------------------------------------------------------------------------
-template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES>
+template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES, int LOWREG>
 __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     unsigned short* __restrict__ out,            // [S, QH, HEAD_DIM]
     float* __restrict__ ws_acc,   // [S, QH, SPLITS, HD] (splits > 1)
@@ -52,14 +52,22 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
   const int sg = lane / LPG;   // sub-group id within wave
   const int sl = lane % LPG;   // lane within sub-group -> dims [sl*8, sl*8+8)
 
-  // Q for the whole GQA group, pre-scaled, 8 dims per lane per head.
-  float qf[GQ][8];
+  // Q for the whole GQA group, kept PACKED bf16 (8 dims per lane per head
+  // = 4 VGPRs/head instead of 8 fp32): at GQ=4 this moves the kernel from
+  // 2 to 3 waves/SIMD occupancy, which matters more than the extra
+  // convert in the dot (the kernel is HBM-latency-bound). The softmax
+  // scale is applied to the reduced score instead of to q.
+  ps_bf16x8 qp[GQ];
   const unsigned short* qbase = q + (long)seq * q_stride + (long)kvh * GQ * D;
 #pragma unroll
-  for (int g = 0; g < GQ; g++) {
-    ps_bf16x8 qv = *(const ps_bf16x8*)(qbase + g * D + sl * 8);
+  for (int g = 0; g < GQ; g++)
+    qp[g] = *(const ps_bf16x8*)(qbase + g * D + sl * 8);
+  float qf[GQ][8];
+  if constexpr (!LOWREG) {
 #pragma unroll
-    for (int j = 0; j < 8; j++) qf[g][j] = ps_bf16_to_f32(qv[j]) * scale;
+    for (int g = 0; g < GQ; g++)
+#pragma unroll
+      for (int j = 0; j < 8; j++) qf[g][j] = ps_bf16_to_f32(qp[g][j]);
   }
 
   float m[GQ], l[GQ], acc[GQ][8];
@@ -92,20 +100,37 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     const int valid_tokens = min(BLOCK_SIZE, ctx - b * BLOCK_SIZE);
     float sc[GQ][TPB];
     ps_bf16x8 vv[TPB];
+    if constexpr (LOWREG) {
+      // keep q packed: block the compiler from hoisting converted copies
+#pragma unroll
+      for (int g = 0; g < GQ; g++)
+        asm volatile("" : "+v"(qp[g]));
+    }
 #pragma unroll
     for (int tt = 0; tt < TPB; tt++) {
       const int tok = tt * TPW + sg;
       const bool valid = tok < valid_tokens;
       ps_bf16x8 kv = *(const ps_bf16x8*)(kb + tok * D + sl * 8);
-      vv[tt] = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
+      if constexpr (!LOWREG)
+        vv[tt] = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
 #pragma unroll
       for (int g = 0; g < GQ; g++) {
         float s = 0.f;
 #pragma unroll
-        for (int j = 0; j < 8; j++) s += qf[g][j] * ps_bf16_to_f32(kv[j]);
-        s = ps_group_sum<LPG>(s);
+        for (int j = 0; j < 8; j++) {
+          if constexpr (LOWREG)
+            s += ps_bf16_to_f32(qp[g][j]) * ps_bf16_to_f32(kv[j]);
+          else
+            s += qf[g][j] * ps_bf16_to_f32(kv[j]);
+        }
+        s = ps_group_sum<LPG>(s) * scale;
         sc[g][tt] = valid ? s : PS_NEG_INF;
       }
+    }
+    if constexpr (LOWREG) {
+#pragma unroll
+      for (int tt = 0; tt < TPB; tt++)
+        vv[tt] = *(const ps_bf16x8*)(vb + (tt * TPW + sg) * D + sl * 8);
     }
 #pragma unroll
     for (int g = 0; g < GQ; g++) {
@@ -444,13 +469,22 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         long q_stride, hipStream_t stream) {
+                         long q_stride, int variant, hipStream_t stream) {
   dim3 grid(num_seqs, KH, num_splits);
   constexpr int NW = 4;
   dim3 block(NW * 64);
 #define PS_DISPATCH_DECODE(HD, G, BS)                                        \
   do {                                                                       \
-    paged_attn_decode_kernel<HD, G, BS, NW><<<grid, block, 0, stream>>>(     \
+    if (variant == 1)                                                        \
+      paged_attn_decode_kernel<HD, G, BS, NW, 1><<<grid, block, 0,           \
+                                                   stream>>>(                \
+          (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,               \
+          (const unsigned short*)q, (const unsigned short*)k_cache,          \
+          (const unsigned short*)v_cache, (const int*)block_tables,          \
+          (const int*)seq_lens, max_blocks, scale, KH, q_stride);            \
+    else                                                                     \
+      paged_attn_decode_kernel<HD, G, BS, NW, 0><<<grid, block, 0,           \
+                                                   stream>>>(                \
         (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,                 \
         (const unsigned short*)q, (const unsigned short*)k_cache,            \
         (const unsigned short*)v_cache, (const int*)block_tables,            \

@@ -29,7 +29,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         long q_stride, hipStream_t stream);
+                         long q_stride, int variant, hipStream_t stream);
 int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* v_cache, const void* block_tables,
                                const void* tile_info, int num_tiles,
@@ -124,7 +124,8 @@ void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
 
 void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
-                       at::Tensor seq_lens, double scale, int64_t num_splits) {
+                       at::Tensor seq_lens, double scale, int64_t num_splits,
+                       int64_t variant) {
   CHECK_GPU_BF16(out);
   CHECK_GPU_BF16(k_cache);
   CHECK_GPU_BF16(v_cache);
@@ -157,7 +158,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       out.data_ptr(), acc_p, ml_p, q.data_ptr(), k_cache.data_ptr(),
       v_cache.data_ptr(), block_tables.data_ptr(), seq_lens.data_ptr(), S,
       max_blocks, (float)scale, KH, GQ, HD, BS, (int)num_splits,
-      q_row_stride(q, HD), current_stream());
+      q_row_stride(q, HD), (int)variant, current_stream());
   TORCH_CHECK(rc == 0, "unsupported decode config: head_dim=", HD,
               " block_size=", BS, " gqa=", GQ);
 }
@@ -277,7 +278,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k_cache"),
         pybind11::arg("v_cache"), pybind11::arg("block_tables"),
         pybind11::arg("seq_lens"), pybind11::arg("scale"),
-        pybind11::arg("num_splits") = 0);
+        pybind11::arg("num_splits") = 0, pybind11::arg("variant") = 0);
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged attention, chunked prefill (bf16 KV)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,

@@ -42,7 +42,8 @@ def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20):
     print(f"prefill ctx={ctx_len}: {dt*1e3:.2f} ms  {flops/dt/1e12:.1f} TF")
 
 
-def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50):
+def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50,
+                 variant=0):
     bs = 16
     per = ctx // bs
     nb = batch * per + 1
@@ -54,19 +55,23 @@ def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50):
     out = torch.empty_like(q)
     scale = hd ** -0.5
     for _ in range(3):
-        _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0)
+        _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0, variant)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0)
+        _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0, variant)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
     bytes_kv = batch * ctx * kh * hd * 2 * 2
-    print(f"decode b={batch} ctx={ctx}: {dt*1e6:.1f} us  {bytes_kv/dt/1e9:.0f} GB/s")
+    print(f"decode v{variant} b={batch} ctx={ctx}: {dt*1e6:.1f} us  "
+          f"{bytes_kv/dt/1e9:.0f} GB/s")
 
 
 if __name__ == "__main__":
     for ctx in (1024, 2048, 4096):
         bench_prefill(ctx)
-    for b in (16, 64, 256):
-        bench_decode(batch=b)
+    for v in (0, 1):
+        for b in (16, 64, 256):
+            bench_decode(batch=b, variant=v)
+        for b in (64, 256):
+            bench_decode(batch=b, ctx=2048, variant=v)

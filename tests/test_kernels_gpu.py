@@ -293,3 +293,29 @@ def test_fused_rope_cache_matches_separate_ops():
     _close(qkv_fused[:, qs : qs + kvs], k_ref)
     _close(k_cache, kc2, atol=2e-2, rtol=2e-2)
     _close(v_cache, vc2, atol=0, rtol=0)
+
+
+def test_paged_attn_decode_lowreg_variant():
+    from production_stack_amd import _C
+
+    torch.manual_seed(7)
+    qh, kh, hd, bs = 32, 8, 128, 16
+    S = 3
+    seq_lens = torch.tensor([5, 700, 64], dtype=torch.int32)
+    max_blocks = 44
+    nb = S * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    block_tables = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+        S, max_blocks
+    )
+    q = torch.randn((S, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_decode(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), block_tables, seq_lens,
+        1.0 / hd ** 0.5,
+    )
+    out = torch.empty_like(q)
+    _C.paged_attn_decode(
+        out, q, k_cache, v_cache, block_tables.cuda(), seq_lens.cuda(),
+        1.0 / hd ** 0.5, 0, 1,
+    )
+    _close(out, want)
