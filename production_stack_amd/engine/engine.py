@@ -228,6 +228,20 @@ class LLMEngine:
     def unload_lora(self, name: str) -> None:
         self.lora_adapters.pop(name, None)
 
+    # ---- tensor parallelism (HTTP-served mode) -------------------------
+    @property
+    def is_tp_worker(self) -> bool:
+        from production_stack_amd.parallel import state as ps
+
+        return (
+            self.config.parallel.tensor_parallel_size > 1
+            and ps.tp_rank() > 0
+        )
+
+    def run_tp_worker(self) -> None:
+        assert self.is_tp_worker
+        self.runner.tp_coord.worker_loop()
+
     # ---- pipeline parallelism ------------------------------------------
     @property
     def is_pp_worker(self) -> bool:

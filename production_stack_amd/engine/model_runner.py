@@ -48,6 +48,17 @@ class ModelRunner:
             )
 
             self.pipeline = PipelineCoordinator(self, self.pp_rank, pp)
+        self.tp_coord = None
+        if tp > 1:
+            import torch.distributed as dist
+
+            from production_stack_amd.engine.tp_worker import TPCoordinator
+            from production_stack_amd.parallel import state as ps
+
+            # only needed when driven over HTTP (rank 0 schedules); tests
+            # that run lockstep generate() on every rank leave this unused
+            if dist.is_initialized():
+                self.tp_coord = TPCoordinator(self, ps.tp_rank(), tp)
         if config.weights_path:
             from production_stack_amd.engine.weights import load_safetensors
 
@@ -100,6 +111,7 @@ class ModelRunner:
             self.device.type != "cuda"
             or self.config.enforce_eager
             or self.pp_size > 1
+            or self.config.parallel.tensor_parallel_size > 1
         ):
             return
         from production_stack_amd.engine.graph_runner import DecodeGraphRunner
@@ -322,6 +334,8 @@ class ModelRunner:
                 seq.request_id: int(tok)
                 for seq, tok in zip(sample_seqs, tokens)
             }
+        if self.tp_coord is not None and self.tp_coord.rank == 0 and                 getattr(self, "tp_serving", False):
+            self.tp_coord.broadcast_step(token_t, meta)
         hidden = self.model(token_t, meta, self.kv_caches)
         if not sample_seqs:
             return {}

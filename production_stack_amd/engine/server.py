@@ -643,6 +643,10 @@ def main() -> None:
         args.advertise_url or f"http://127.0.0.1:{args.port}"
     )
     engine = LLMEngine(cfg, device=args.device)
+    if engine.is_tp_worker:
+        logging.basicConfig(level=logging.INFO)
+        engine.run_tp_worker()
+        return
     if engine.is_pp_worker:
         # pipeline ranks > 0 serve activations, not HTTP
         logging.basicConfig(level=logging.INFO)
