@@ -35,7 +35,7 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
                                int KH, int GQ, int head_dim, long q_stride,
-                               hipStream_t stream);
+                               int variant, hipStream_t stream);
 int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* v_cache, const void* block_tables,
                           const void* token_seq, const void* token_pos,
@@ -47,6 +47,10 @@ void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
                           int KH, int HD, int BS, hipStream_t stream);
 void ps_greedy_sample(void* out, const void* logits, long R, int V,
                       hipStream_t stream);
+void ps_kv_quant(void* out, void* scales, const void* in, long rows, int hd,
+                 hipStream_t stream);
+void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
+                   int hd, hipStream_t stream);
 }
 
 namespace {
@@ -193,7 +197,7 @@ void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
                              at::Tensor k_cache, at::Tensor v_cache,
                              at::Tensor block_tables, at::Tensor tile_info,
-                             double scale) {
+                             double scale, int64_t variant) {
   CHECK_GPU_BF16(out);
   CHECK_GPU_BF16(k_cache);
   CHECK_GPU_BF16(v_cache);
@@ -210,7 +214,7 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr(), tile_info.data_ptr(), NT, QH,
       (int)block_tables.size(1), (float)scale, KH, GQ, HD,
-      q_row_stride(q, HD), current_stream());
+      q_row_stride(q, HD), (int)variant, current_stream());
   TORCH_CHECK(rc == 0, "unsupported mfma prefill config: head_dim=", HD);
 }
 
@@ -256,6 +260,27 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        BS, current_stream());
 }
 
+void kv_quant(at::Tensor out, at::Tensor scales, at::Tensor in) {
+  CHECK_GPU_DTYPE(out, at::kChar);
+  CHECK_GPU_DTYPE(scales, at::kFloat);
+  CHECK_GPU_BF16(in);
+  const int hd = (int)in.size(-1);
+  TORCH_CHECK(hd % 8 == 0, "row width must be multiple of 8");
+  const long rows = in.numel() / hd;
+  ps_kv_quant(out.data_ptr(), scales.data_ptr(), in.data_ptr(), rows, hd,
+              current_stream());
+}
+
+void kv_dequant(at::Tensor out, at::Tensor in, at::Tensor scales) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_DTYPE(in, at::kChar);
+  CHECK_GPU_DTYPE(scales, at::kFloat);
+  const int hd = (int)out.size(-1);
+  const long rows = out.numel() / hd;
+  ps_kv_dequant(out.data_ptr(), in.data_ptr(), scales.data_ptr(), rows, hd,
+                current_stream());
+}
+
 void greedy_sample(at::Tensor out, at::Tensor logits) {
   CHECK_GPU_DTYPE(out, at::kLong);
   CHECK_GPU_BF16(logits);
@@ -282,10 +307,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged attention, chunked prefill (bf16 KV)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
-        "Paged attention, chunked prefill via MFMA tiles (head_dim 128)");
+        "Paged attention, chunked prefill via MFMA tiles (head_dim 128)",
+        pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k_cache"),
+        pybind11::arg("v_cache"), pybind11::arg("block_tables"),
+        pybind11::arg("tile_info"), pybind11::arg("scale"),
+        pybind11::arg("variant") = 4);
   m.def("fused_rope_cache", &fused_rope_cache,
         "Fused RoPE + paged KV append on the packed qkv tensor");
   m.def("reshape_and_cache", &reshape_and_cache,
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");
+  m.def("kv_quant", &kv_quant, "Row-wise int8 KV quantization");
+  m.def("kv_dequant", &kv_dequant, "Row-wise int8 KV dequantization");
 }

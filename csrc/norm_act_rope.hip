@@ -292,3 +292,73 @@ extern "C" void ps_fused_rope_cache(void* qkv, const void* positions,
       (const long*)slot_mapping, (unsigned short*)k_cache,
       (unsigned short*)v_cache, QH, KH, HD, ROT, qkv_stride, BS);
 }
+
+// ---------------------------------------------------------------------------
+// Row-wise int8 KV quantization (CacheGen-style serde for the offload /
+// transfer tiers): each row of `hd` elements gets one fp32 scale
+// (max|x| / 127). Halves PCIe/host-pool traffic vs bf16 at ~0.4% RMS error.
+// ---------------------------------------------------------------------------
+__global__ void kv_quant_kernel(signed char* __restrict__ out,
+                                float* __restrict__ scales,
+                                const unsigned short* __restrict__ in,
+                                int hd, long rows) {
+  const long r = (long)blockIdx.x * blockDim.y + threadIdx.y;
+  if (r >= rows) return;
+  const unsigned short* row = in + r * hd;
+  signed char* orow = out + r * hd;
+  const int lane = threadIdx.x;  // 64 lanes per row
+  float amax = 0.f;
+  for (int i = lane * 8; i < hd; i += 64 * 8) {
+    ps_bf16x8 v = *(const ps_bf16x8*)(row + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) amax = fmaxf(amax, fabsf(ps_bf16_to_f32(v[j])));
+  }
+  amax = ps_group_max<64>(amax);
+  const float scale = amax > 0.f ? amax / 127.f : 1.f;
+  const float inv = 1.f / scale;
+  if (lane == 0) scales[r] = scale;
+  for (int i = lane * 8; i < hd; i += 64 * 8) {
+    ps_bf16x8 v = *(const ps_bf16x8*)(row + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = ps_bf16_to_f32(v[j]) * inv;
+      orow[i + j] = (signed char)lrintf(fminf(fmaxf(f, -127.f), 127.f));
+    }
+  }
+}
+
+__global__ void kv_dequant_kernel(unsigned short* __restrict__ out,
+                                  const signed char* __restrict__ in,
+                                  const float* __restrict__ scales,
+                                  int hd, long rows) {
+  const long r = (long)blockIdx.x * blockDim.y + threadIdx.y;
+  if (r >= rows) return;
+  const signed char* row = in + r * hd;
+  unsigned short* orow = out + r * hd;
+  const float scale = scales[r];
+  for (int i = threadIdx.x * 8; i < hd; i += 64 * 8) {
+    ps_bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      v[j] = ps_f32_to_bf16((float)row[i + j] * scale);
+    *(ps_bf16x8*)(orow + i) = v;
+  }
+}
+
+extern "C" {
+void ps_kv_quant(void* out, void* scales, const void* in, long rows, int hd,
+                 hipStream_t stream) {
+  dim3 block(64, 4);
+  dim3 grid((unsigned)((rows + 3) / 4));
+  kv_quant_kernel<<<grid, block, 0, stream>>>(
+      (signed char*)out, (float*)scales, (const unsigned short*)in, hd, rows);
+}
+void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
+                   int hd, hipStream_t stream) {
+  dim3 block(64, 4);
+  dim3 grid((unsigned)((rows + 3) / 4));
+  kv_dequant_kernel<<<grid, block, 0, stream>>>(
+      (unsigned short*)out, (const signed char*)in, (const float*)scales, hd,
+      rows);
+}
+}

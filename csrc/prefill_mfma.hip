@@ -36,8 +36,8 @@ PS_DEV ps_mbf16x8 ps_as_mbf16(ps_bf16x8 u) {
 #define PS_PL_STRIDE 72  // P row stride in tokens (multiple of 8)
 
 // tile_info: int4 per tile = (seq_row, q_token_start, q_pos_start, n_rows)
-template <int HEAD_DIM>
-__global__ __launch_bounds__(256) void paged_attn_prefill_mfma_kernel(
+template <int HEAD_DIM, int WPS>
+__global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     unsigned short* __restrict__ out,            // [T, QH, HD]
     const unsigned short* __restrict__ q,        // [T, QH, HD]
     const unsigned short* __restrict__ k_cache,  // [NB, KH, 16, HD]
@@ -257,14 +257,21 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
                                int KH, int GQ, int head_dim, long q_stride,
-                               hipStream_t stream) {
+                               int variant, hipStream_t stream) {
   if (head_dim != 128) return -1;
   dim3 grid(num_tiles, num_q_heads);
-  paged_attn_prefill_mfma_kernel<128><<<grid, 256, 0, stream>>>(
-      (unsigned short*)out, (const unsigned short*)q,
-      (const unsigned short*)k_cache, (const unsigned short*)v_cache,
-      (const int*)block_tables, (const int*)tile_info, max_blocks, scale, KH,
-      GQ, q_stride);
+  if (variant == 3)
+    paged_attn_prefill_mfma_kernel<128, 3><<<grid, 256, 0, stream>>>(
+        (unsigned short*)out, (const unsigned short*)q,
+        (const unsigned short*)k_cache, (const unsigned short*)v_cache,
+        (const int*)block_tables, (const int*)tile_info, max_blocks, scale,
+        KH, GQ, q_stride);
+  else
+    paged_attn_prefill_mfma_kernel<128, 4><<<grid, 256, 0, stream>>>(
+        (unsigned short*)out, (const unsigned short*)q,
+        (const unsigned short*)k_cache, (const unsigned short*)v_cache,
+        (const int*)block_tables, (const int*)tile_info, max_blocks, scale,
+        KH, GQ, q_stride);
   return 0;
 }
 

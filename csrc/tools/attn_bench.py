@@ -15,7 +15,7 @@ import torch
 from production_stack_amd import _C
 
 
-def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20):
+def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20, variant=4):
     bs = 16
     nblocks = ctx_len // bs + 1
     k = torch.randn(nblocks + 1, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
@@ -30,16 +30,17 @@ def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20):
     out = torch.empty_like(q)
     scale = hd ** -0.5
     for _ in range(3):
-        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale)
+        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale, variant)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale)
+        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale, variant)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
     # causal flops: sum over q of ctx(q) = T*(T+1)/2 per head pair (QK+PV)
     flops = 4 * hd * qh * (T * (T + 1) / 2)
-    print(f"prefill ctx={ctx_len}: {dt*1e3:.2f} ms  {flops/dt/1e12:.1f} TF")
+    print(f"prefill v{variant} ctx={ctx_len}: {dt*1e3:.2f} ms  "
+          f"{flops/dt/1e12:.1f} TF")
 
 
 def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50,
@@ -68,8 +69,9 @@ def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50,
 
 
 if __name__ == "__main__":
-    for ctx in (1024, 2048, 4096):
-        bench_prefill(ctx)
+    for v in (3, 4):
+        for ctx in (1024, 2048, 4096):
+            bench_prefill(ctx, variant=v)
     for v in (0, 1):
         for b in (16, 64, 256):
             bench_decode(batch=b, variant=v)

@@ -165,7 +165,7 @@ def paged_attn_prefill_mfma(
     _require_ext()
     out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
     _C.paged_attn_prefill_mfma(
-        out, q, k_cache, v_cache, block_tables, tile_info, scale
+        out, q, k_cache, v_cache, block_tables, tile_info, scale, 4
     )
     return out
 
@@ -179,3 +179,28 @@ def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
         _C.greedy_sample(out, logits)
         return out
     return reference.greedy_sample(logits)
+
+
+def kv_quant(x: torch.Tensor) -> tuple:
+    """Row-wise int8 KV quantization (CacheGen-style serde)."""
+    if x.is_cuda:
+        _require_ext()
+        hd = x.shape[-1]
+        out = torch.empty(x.shape, dtype=torch.int8, device=x.device)
+        scales = torch.empty(
+            x.numel() // hd, dtype=torch.float32, device=x.device
+        )
+        _C.kv_quant(out, scales, x)
+        return out, scales
+    return reference.kv_quant(x)
+
+
+def kv_dequant(
+    q: torch.Tensor, scales: torch.Tensor, dtype=torch.bfloat16
+) -> torch.Tensor:
+    if q.is_cuda:
+        _require_ext()
+        out = torch.empty(q.shape, dtype=dtype, device=q.device)
+        _C.kv_dequant(out, q, scales)
+        return out
+    return reference.kv_dequant(q, scales, dtype)

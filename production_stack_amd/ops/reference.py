@@ -157,3 +157,22 @@ def paged_attn_prefill(
 
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     return logits.float().argmax(dim=-1)
+
+
+def kv_quant(x: torch.Tensor) -> tuple:
+    """Row-wise int8 quantization over the last dim. Returns (int8, scales)."""
+    hd = x.shape[-1]
+    rows = x.reshape(-1, hd).float()
+    amax = rows.abs().amax(dim=-1)
+    scales = torch.where(amax > 0, amax / 127.0, torch.ones_like(amax))
+    q = torch.clamp(
+        torch.round(rows / scales[:, None]), -127, 127
+    ).to(torch.int8)
+    return q.view(x.shape), scales
+
+
+def kv_dequant(q: torch.Tensor, scales: torch.Tensor,
+               dtype=torch.bfloat16) -> torch.Tensor:
+    hd = q.shape[-1]
+    rows = q.reshape(-1, hd).float() * scales[:, None]
+    return rows.to(dtype).view(q.shape)
