@@ -72,6 +72,7 @@ class LLMEngine:
                 config.cache.block_size,
                 config.cache.cpu_offload_gb,
                 self.device,
+                offload_dtype=config.cache.offload_dtype,
             )
             self.block_manager.offload_pool = self.host_pool
         else:

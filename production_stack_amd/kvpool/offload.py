@@ -31,6 +31,7 @@ class HostKVPool:
         block_size: int,
         capacity_gb: float,
         device: torch.device,
+        offload_dtype: str = "bf16",
     ) -> None:
         self.kv_caches = kv_caches
         self.device = device
@@ -39,14 +40,34 @@ class HostKVPool:
         self.kh, self.bs, self.hd = k0.shape[1], k0.shape[2], k0.shape[3]
         assert self.bs == block_size
         self.block_elems = self.kh * self.bs * self.hd
-        block_bytes = self.layers * 2 * self.block_elems * 2
+        self.quantized = offload_dtype == "int8"
+        self.rows = self.layers * 2 * self.kh * self.bs  # scale rows/block
+        elem_bytes = 1 if self.quantized else 2
+        block_bytes = self.layers * 2 * self.block_elems * elem_bytes
+        if self.quantized:
+            block_bytes += self.rows * 4
         self.capacity = max(int(capacity_gb * (1 << 30)) // block_bytes, 1)
         pin = device.type == "cuda"
         self.store = torch.empty(
             (self.capacity, self.layers, 2, self.block_elems),
-            dtype=torch.bfloat16,
+            dtype=torch.int8 if self.quantized else torch.bfloat16,
             pin_memory=pin,
         )
+        self.scale_store = (
+            torch.empty(
+                (self.capacity, self.rows), dtype=torch.float32,
+                pin_memory=pin,
+            )
+            if self.quantized
+            else None
+        )
+        # GPU staging for the quantized path
+        if self.quantized:
+            self._stage = torch.empty(
+                (self.layers, 2, self.block_elems),
+                dtype=torch.bfloat16,
+                device=device,
+            )
         self.free_slots: Deque[int] = deque(range(self.capacity))
         self.slot_of: "OrderedDict[int, int]" = OrderedDict()  # hash -> slot
         self.stream = (
@@ -68,6 +89,17 @@ class HostKVPool:
         self.evicted += 1
         return slot
 
+    def _gather_quant(self, block_id: int):
+        """D2D gather + row-wise int8 quantization on the side stream."""
+        from production_stack_amd import ops
+
+        for li, (kc, vc) in enumerate(self.kv_caches):
+            self._stage[li, 0].copy_(kc[block_id].flatten(),
+                                     non_blocking=True)
+            self._stage[li, 1].copy_(vc[block_id].flatten(),
+                                     non_blocking=True)
+        return ops.kv_quant(self._stage.view(-1, self.hd))
+
     def offload(self, h: int, block_id: int) -> None:
         """Async D2H of a (now immutable) full block."""
         if h in self.slot_of:
@@ -78,17 +110,37 @@ class HostKVPool:
         if self.stream is not None:
             self.stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self.stream):
-                for li, (kc, vc) in enumerate(self.kv_caches):
-                    dst[li, 0].copy_(
-                        kc[block_id].flatten(), non_blocking=True
-                    )
-                    dst[li, 1].copy_(
-                        vc[block_id].flatten(), non_blocking=True
-                    )
+                if self.quantized:
+                    q, scales = self._gather_quant(block_id)
+                    dst.view(-1, self.hd).copy_(q, non_blocking=True)
+                    self.scale_store[slot].copy_(scales, non_blocking=True)
+                else:
+                    for li, (kc, vc) in enumerate(self.kv_caches):
+                        dst[li, 0].copy_(
+                            kc[block_id].flatten(), non_blocking=True
+                        )
+                        dst[li, 1].copy_(
+                            vc[block_id].flatten(), non_blocking=True
+                        )
         else:
-            for li, (kc, vc) in enumerate(self.kv_caches):
-                dst[li, 0].copy_(kc[block_id].flatten())
-                dst[li, 1].copy_(vc[block_id].flatten())
+            if self.quantized:
+                from production_stack_amd.ops import reference
+
+                rows = torch.cat(
+                    [
+                        torch.stack(
+                            [kc[block_id].flatten(), vc[block_id].flatten()]
+                        )
+                        for kc, vc in self.kv_caches
+                    ]
+                ).view(-1, self.hd)
+                q, scales = reference.kv_quant(rows)
+                dst.view(-1, self.hd).copy_(q)
+                self.scale_store[slot].copy_(scales)
+            else:
+                for li, (kc, vc) in enumerate(self.kv_caches):
+                    dst[li, 0].copy_(kc[block_id].flatten())
+                    dst[li, 1].copy_(vc[block_id].flatten())
         self.slot_of[h] = slot
         self.offloaded += 1
 
@@ -102,20 +154,48 @@ class HostKVPool:
         if self.stream is not None:
             self.stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self.stream):
-                for li, (kc, vc) in enumerate(self.kv_caches):
-                    kc[block_id].flatten().copy_(
-                        src[li, 0], non_blocking=True
+                if self.quantized:
+                    from production_stack_amd import ops
+
+                    qdev = src.to(self.device, non_blocking=True)
+                    sdev = self.scale_store[slot].to(
+                        self.device, non_blocking=True
                     )
-                    vc[block_id].flatten().copy_(
-                        src[li, 1], non_blocking=True
-                    )
+                    deq = ops.kv_dequant(
+                        qdev.view(-1, self.hd), sdev
+                    ).view(self.layers, 2, self.block_elems)
+                    for li, (kc, vc) in enumerate(self.kv_caches):
+                        kc[block_id].flatten().copy_(
+                            deq[li, 0], non_blocking=True
+                        )
+                        vc[block_id].flatten().copy_(
+                            deq[li, 1], non_blocking=True
+                        )
+                else:
+                    for li, (kc, vc) in enumerate(self.kv_caches):
+                        kc[block_id].flatten().copy_(
+                            src[li, 0], non_blocking=True
+                        )
+                        vc[block_id].flatten().copy_(
+                            src[li, 1], non_blocking=True
+                        )
                 ev = torch.cuda.Event()
                 ev.record(self.stream)
                 self._restore_events.append(ev)
         else:
-            for li, (kc, vc) in enumerate(self.kv_caches):
-                kc[block_id].flatten().copy_(src[li, 0])
-                vc[block_id].flatten().copy_(src[li, 1])
+            if self.quantized:
+                from production_stack_amd.ops import reference
+
+                deq = reference.kv_dequant(
+                    src.view(-1, self.hd), self.scale_store[slot]
+                ).view(self.layers, 2, self.block_elems)
+                for li, (kc, vc) in enumerate(self.kv_caches):
+                    kc[block_id].flatten().copy_(deq[li, 0])
+                    vc[block_id].flatten().copy_(deq[li, 1])
+            else:
+                for li, (kc, vc) in enumerate(self.kv_caches):
+                    kc[block_id].flatten().copy_(src[li, 0])
+                    vc[block_id].flatten().copy_(src[li, 1])
         self.restored += 1
         return True
 

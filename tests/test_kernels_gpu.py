@@ -319,3 +319,19 @@ def test_paged_attn_decode_lowreg_variant():
         1.0 / hd ** 0.5, 0, 1,
     )
     _close(out, want)
+
+
+def test_kv_quant_gpu_matches_reference():
+    x = torch.randn(200, 128, dtype=torch.bfloat16, device="cuda") * 2
+    q_gpu, s_gpu = ops.kv_quant(x)
+    q_ref, s_ref = reference.kv_quant(x.cpu())
+    torch.testing.assert_close(
+        s_gpu.cpu(), s_ref, atol=1e-5, rtol=1e-4
+    )
+    # quantized codes may differ by 1 ulp at rounding boundaries
+    diff = (q_gpu.cpu().int() - q_ref.int()).abs()
+    assert diff.max() <= 1
+    y = ops.kv_dequant(q_gpu, s_gpu)
+    torch.testing.assert_close(
+        y.float().cpu(), x.float().cpu(), atol=0.05, rtol=0.05
+    )

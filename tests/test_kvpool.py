@@ -206,3 +206,54 @@ def test_host_pool_lru_eviction():
     pool.offload(222, 1)
     assert not pool.has(111) and pool.has(222)
     assert pool.evicted == 1
+
+
+def test_kv_quant_reference_roundtrip():
+    import torch
+
+    from production_stack_amd.ops import reference
+
+    x = torch.randn(64, 128, dtype=torch.bfloat16) * 3
+    q, s = reference.kv_quant(x)
+    y = reference.kv_dequant(q, s)
+    err = (x.float() - y.float()).abs().max() / x.float().abs().max()
+    assert err < 0.02
+    # zero rows survive
+    x[0] = 0
+    q, s = reference.kv_quant(x)
+    y = reference.kv_dequant(q, s)
+    assert torch.all(y[0] == 0)
+
+
+def test_host_offload_int8_roundtrip_cpu():
+    import torch
+
+    from production_stack_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        SchedulerConfig,
+    )
+    from production_stack_amd.engine.engine import LLMEngine
+    from production_stack_amd.engine.sampling import SamplingParams
+
+    cfg = EngineConfig(
+        model="tiny-llama",
+        max_model_len=256,
+        cache=CacheConfig(
+            num_gpu_blocks=16, block_size=16, cpu_offload_gb=0.01,
+            offload_dtype="int8",
+        ),
+        scheduler=SchedulerConfig(max_num_seqs=4, max_num_batched_tokens=256),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    assert eng.host_pool.quantized
+    p = SamplingParams(max_tokens=4, temperature=0.0, ignore_eos=True)
+    prompt = list(range(10, 90))
+    out1 = eng.generate([prompt], p)["offline-0"]
+    assert eng.host_pool.offloaded > 0
+    eng.block_manager.reset_prefix_cache()
+    out2 = eng.generate([prompt], p)["offline-0"]
+    assert eng.host_pool.restored > 0
+    # int8 KV is lossy: require strong (not exact) agreement
+    agree = sum(a == b for a, b in zip(out1, out2))
+    assert agree >= len(out1) - 1, (out1, out2)
