@@ -71,7 +71,12 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
   const int n_pages = (ctx_limit + BS - 1) / BS;
   const int n_chunks = (ctx_limit + PS_CHUNK - 1) / PS_CHUNK;
 
-  // shared: V^T staging (token-group XOR swizzle) + per-wave P tiles
+  // shared: K rows + V^T staging (both XOR-swizzled) + per-wave P tiles.
+  // K is staged cooperatively ONCE per workgroup: the 4 waves all consume
+  // the same 64-token K tile, and per-wave direct global loads both
+  // quadruplicated HBM traffic and exposed load latency (the 128-VGPR
+  // budget can't keep 16 loads in flight per wave).
+  __shared__ __align__(16) unsigned short k_lds[PS_CHUNK][D];
   __shared__ __align__(16) unsigned short v_t[D][PS_CHUNK];
   __shared__ __align__(16) unsigned short p_lds[4][16][PS_PL_STRIDE];
 
@@ -111,19 +116,33 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     const long pg = bt[pg_idx];
     return v_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
   };
+  auto k_row_ptr = [&](int chunk) {
+    const int tok = chunk * PS_CHUNK + tv;
+    const int pg_idx = min(tok / BS, n_pages - 1);
+    const long pg = bt[pg_idx];
+    return k_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
+  };
   ps_bf16x8 vstage[4];
+  ps_bf16x8 kstage[4];
   {
     const unsigned short* vrow = v_row_ptr(0);
+    const unsigned short* krow = k_row_ptr(0);
 #pragma unroll
-    for (int h = 0; h < 4; h++)
+    for (int h = 0; h < 4; h++) {
       vstage[h] = *(const ps_bf16x8*)(vrow + d0 + h * 8);
+      kstage[h] = *(const ps_bf16x8*)(krow + d0 + h * 8);
+    }
   }
 
   for (int chunk = 0; chunk < n_chunks; chunk++) {
     const int tok0 = chunk * PS_CHUNK;
-    // write the pre-fetched V^T tile (swizzled token group per dim)
+    // write the pre-fetched K rows (16-B slot index XOR-swizzled by the
+    // token's low bits: row-major [64][128] would put all 16 lanes of a
+    // B-frag read in one bank — guide Guideline 4) and V^T tile
 #pragma unroll
     for (int h = 0; h < 4; h++) {
+      const int slot = ((wave * 4 + h) ^ (tv & 7));
+      *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = kstage[h];
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         const int d = d0 + h * 8 + j;
@@ -133,9 +152,12 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     __syncthreads();
     if (chunk + 1 < n_chunks) {
       const unsigned short* vrow = v_row_ptr(chunk + 1);
+      const unsigned short* krow = k_row_ptr(chunk + 1);
 #pragma unroll
-      for (int h = 0; h < 4; h++)
+      for (int h = 0; h < 4; h++) {
         vstage[h] = *(const ps_bf16x8*)(vrow + d0 + h * 8);
+        kstage[h] = *(const ps_bf16x8*)(krow + d0 + h * 8);
+      }
     }
 
     const bool wave_active = (wave * 16 < n_rows) && (tok0 <= wave_pos_max);
@@ -145,16 +167,13 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 #pragma unroll
       for (int ct = 0; ct < NCT; ct++) {
         s_frag[ct] = {0.f, 0.f, 0.f, 0.f};
-        const int tok = tok0 + ct * 16 + rc;  // this lane's kv token (col)
-        const int pg_idx = min(tok / BS, n_pages - 1);
-        const long pg = bt[pg_idx];
-        const unsigned short* krow =
-            k_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
+        const int trow = ct * 16 + rc;  // this lane's kv token row (col)
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int kk = 0; kk < NK; kk++) {
-          ps_mbf16x8 k_frag =
-              ps_as_mbf16(*(const ps_bf16x8*)(krow + kk * 32 + g * 8));
+          const int slot = ((kk * 4 + g) ^ (trow & 7));
+          ps_mbf16x8 k_frag = ps_as_mbf16(
+              *(const ps_bf16x8*)(&k_lds[trow][slot * 8]));
           s_frag[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[kk], k_frag, s_frag[ct], 0, 0, 0);
         }
