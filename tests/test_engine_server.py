@@ -269,3 +269,45 @@ def test_stop_string():
         assert out["usage"]["completion_tokens"] <= len(text)
 
     with_server(go)
+
+
+def test_client_disconnect_aborts_request():
+    """Dropping an SSE stream mid-generation aborts the engine request
+    (the router holds connections open indefinitely; reference request.py
+    :312 — our engine must reclaim the slot)."""
+
+    async def go():
+        app = make_app()
+        engine = app.state.engine
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://engine"
+                ) as client:
+                    async with client.stream(
+                        "POST",
+                        "/v1/completions",
+                        json={
+                            "model": "tiny-llama",
+                            "prompt": "long generation",
+                            "max_tokens": 500,
+                            "temperature": 0,
+                            "ignore_eos": True,
+                            "stream": True,
+                        },
+                        timeout=60,
+                    ) as r:
+                        assert r.status_code == 200
+                        count = 0
+                        async for _ in r.aiter_text():
+                            count += 1
+                            if count >= 3:
+                                break  # drop the stream
+                # the abort must drain the scheduler
+                for _ in range(100):
+                    if not engine.has_unfinished():
+                        break
+                    await asyncio.sleep(0.05)
+                assert not engine.has_unfinished()
+
+    asyncio.run(go())
