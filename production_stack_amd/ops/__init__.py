@@ -164,8 +164,10 @@ def paged_attn_prefill_mfma(
     """MFMA-tiled chunked prefill (GPU, head_dim 128 only)."""
     _require_ext()
     out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+    # variant 3 = 3-waves/SIMD bound: best measured (222 TF @ctx4096,
+    # profiles/attn_bench_prefill)
     _C.paged_attn_prefill_mfma(
-        out, q, k_cache, v_cache, block_tables, tile_info, scale, 4
+        out, q, k_cache, v_cache, block_tables, tile_info, scale, 3
     )
     return out
 
