@@ -51,6 +51,8 @@ void ps_kv_quant(void* out, void* scales, const void* in, long rows, int hd,
                  hipStream_t stream);
 void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
                    int hd, hipStream_t stream);
+int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
+                   int K, long x_stride, hipStream_t stream);
 }
 
 namespace {
@@ -260,6 +262,22 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        BS, current_stream());
 }
 
+void skinny_gemm(at::Tensor out_f32, at::Tensor x, at::Tensor w) {
+  CHECK_GPU_DTYPE(out_f32, at::kFloat);
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "x must be bf16");
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1, "x must be 2D row-major");
+  CHECK_GPU_BF16(w);
+  const int M = (int)x.size(0);
+  const int K = (int)x.size(1);
+  const int N = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  int rc = ps_skinny_gemm(out_f32.data_ptr(), x.data_ptr(), w.data_ptr(), M,
+                          N, K, x.stride(0), current_stream());
+  TORCH_CHECK(rc == 0, "unsupported skinny gemm shape M=", M, " N=", N,
+              " K=", K);
+}
+
 void kv_quant(at::Tensor out, at::Tensor scales, at::Tensor in) {
   CHECK_GPU_DTYPE(out, at::kChar);
   CHECK_GPU_DTYPE(scales, at::kFloat);
@@ -318,5 +336,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");
   m.def("kv_quant", &kv_quant, "Row-wise int8 KV quantization");
+  m.def("skinny_gemm", &skinny_gemm,
+        "Split-K MFMA GEMM for decode-shaped (M<=128) projections");
   m.def("kv_dequant", &kv_dequant, "Row-wise int8 KV dequantization");
 }

@@ -37,7 +37,30 @@ def bench(M, iters=200):
           f"(weights roofline ~{(sum(n*k for n,k,_ in SHAPES)*2/6.3e12)*1e6:.1f} us)")
 
 
+def bench_skinny(M, iters=200):
+    from production_stack_amd import _C
+
+    for (N, K, name) in SHAPES:
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+        x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+        out = torch.zeros(M, N, dtype=torch.float32, device="cuda")
+        for _ in range(10):
+            out.zero_()
+            _C.skinny_gemm(out, x, w)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            out.zero_()
+            _C.skinny_gemm(out, x, w)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / iters
+        gb = N * K * 2 / 1e9
+        print(f"skinny M={M:4d} {name:8s}: {dt*1e6:7.1f} us  {gb/dt:6.0f} GB/s")
+
+
 if __name__ == "__main__":
     print("TunableOp:", os.environ.get("PYTORCH_TUNABLEOP_ENABLED", "0"))
     for M in (16, 32, 64, 128, 256):
         bench(M)
+    for M in (16, 32, 64, 128):
+        bench_skinny(M)

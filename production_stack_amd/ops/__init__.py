@@ -206,3 +206,14 @@ def kv_dequant(
         _C.kv_dequant(out, q, scales)
         return out
     return reference.kv_dequant(q, scales, dtype)
+
+
+def skinny_gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """out = x @ w.T for decode-shaped M<=128 (weight-stream-bound GEMMs
+    where hipBLASLt tiles poorly — see csrc/skinny_gemm.hip). GPU only."""
+    _require_ext()
+    out = torch.zeros(
+        (x.shape[0], w.shape[0]), dtype=torch.float32, device=x.device
+    )
+    _C.skinny_gemm(out, x, w)
+    return out.to(x.dtype)

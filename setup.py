@@ -21,6 +21,7 @@ ext = CUDAExtension(
         "csrc/norm_act_rope.hip",
         "csrc/attention.hip",
         "csrc/prefill_mfma.hip",
+        "csrc/skinny_gemm.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -335,3 +335,27 @@ def test_kv_quant_gpu_matches_reference():
     torch.testing.assert_close(
         y.float().cpu(), x.float().cpu(), atol=0.05, rtol=0.05
     )
+
+
+@pytest.mark.parametrize("M,N,K", [
+    (1, 6144, 4096), (16, 6144, 4096), (17, 4096, 4096),
+    (64, 4096, 14336), (128, 28672, 4096), (100, 4096, 4096),
+])
+def test_skinny_gemm(M, N, K):
+    torch.manual_seed(M)
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") / 8
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / 8
+    got = ops.skinny_gemm(x, w)
+    want = (x.float() @ w.float().t()).to(torch.bfloat16)
+    _close(got, want, atol=3e-2, rtol=3e-2)
+
+
+def test_skinny_gemm_strided_x():
+    """x as a strided row view (the qkv input is a slice of a wider buf)."""
+    M, K, N = 32, 4096, 4096
+    buf = torch.randn(M, K + 512, dtype=torch.bfloat16, device="cuda") / 8
+    x = buf[:, :K]
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / 8
+    got = ops.skinny_gemm(x, w)
+    want = (x.float() @ w.float().t()).to(torch.bfloat16)
+    _close(got, want, atol=3e-2, rtol=3e-2)
