@@ -104,6 +104,21 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         title=f"production-stack-amd engine ({served_model})",
         lifespan=lifespan,
     )
+
+    import os
+
+    api_key = os.environ.get("VLLM_API_KEY")
+    if api_key:
+        @app.middleware("http")
+        async def _auth(request: Request, call_next):
+            if request.url.path not in ("/health", "/metrics"):
+                auth = request.headers.get("authorization", "")
+                if auth != f"Bearer {api_key}":
+                    return JSONResponse(
+                        status_code=401,
+                        content={"error": "invalid or missing API key"},
+                    )
+            return await call_next(request)
     app.state.engine = engine
     app.state.async_engine = async_engine
     app.state.served_model = served_model

@@ -382,6 +382,17 @@ def initialize_all(app: FastAPI, args) -> None:
 
     gates = initialize_feature_gates(args.feature_gates)
 
+    if getattr(args, "sentry_dsn", None):
+        try:  # sentry-sdk is optional (absent in the offline image)
+            import sentry_sdk
+
+            sentry_sdk.init(
+                dsn=args.sentry_dsn,
+                traces_sample_rate=args.sentry_traces_sample_rate,
+            )
+        except ImportError:
+            logger.warning("--sentry-dsn set but sentry_sdk not installed")
+
     from production_stack_amd.router import tracing
 
     tracing.initialize_tracing(

@@ -80,6 +80,8 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
                    help="bearer token forwarded to engines")
 
     # observability / experimental
+    p.add_argument("--sentry-dsn", default=None)
+    p.add_argument("--sentry-traces-sample-rate", type=float, default=0.0)
     p.add_argument("--otel-endpoint", default=None)
     p.add_argument("--otel-service-name", default="vllm-router")
     p.add_argument("--otel-secure", action="store_true")

@@ -311,3 +311,29 @@ def test_client_disconnect_aborts_request():
                 assert not engine.has_unfinished()
 
     asyncio.run(go())
+
+
+def test_api_key_auth(monkeypatch):
+    monkeypatch.setenv("VLLM_API_KEY", "sk-secret")
+
+    async def go():
+        app = make_app()
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://engine"
+                ) as client:
+                    r = await client.post(
+                        "/tokenize", json={"prompt": "a"}
+                    )
+                    assert r.status_code == 401
+                    r = await client.get("/health")
+                    assert r.status_code == 200  # probes stay open
+                    r = await client.post(
+                        "/tokenize",
+                        json={"prompt": "a"},
+                        headers={"Authorization": "Bearer sk-secret"},
+                    )
+                    assert r.status_code == 200
+
+    asyncio.run(go())
