@@ -24,6 +24,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from production_stack_amd import ops
+from production_stack_amd.ops import gemm_policy
 from production_stack_amd.engine.config import ModelConfig
 from production_stack_amd.parallel import state as pstate
 
@@ -117,7 +118,7 @@ class LlamaLayer(nn.Module):
             hidden, residual = ops.fused_add_rms_norm(
                 hidden, residual, self.input_norm, cfg.rms_norm_eps
             )
-        qkv = F.linear(hidden, self.qkv_proj)
+        qkv = gemm_policy.linear(hidden, self.qkv_proj)
         qs = self.q_heads * self.head_dim
         kvs = self.kv_heads * self.head_dim
         if meta.lora_groups:
@@ -196,7 +197,7 @@ class LlamaLayer(nn.Module):
                 )
             )
         attn = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
-        attn_out = F.linear(attn.view(T, qs), self.o_proj)
+        attn_out = gemm_policy.linear(attn.view(T, qs), self.o_proj)
         if meta.lora_groups:
             apply_lora_slice(
                 attn_out, attn.view(T, qs), meta.lora_groups,
@@ -207,7 +208,7 @@ class LlamaLayer(nn.Module):
         hidden, residual = ops.fused_add_rms_norm(
             attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
         )
-        gate_up = F.linear(hidden, self.gate_up_proj)
+        gate_up = gemm_policy.linear(hidden, self.gate_up_proj)
         if meta.lora_groups:
             apply_lora_slice(
                 gate_up, hidden, meta.lora_groups, self.layer_idx,
@@ -218,7 +219,7 @@ class LlamaLayer(nn.Module):
                 "up", self.inter, self.inter,
             )
         act = ops.silu_and_mul(gate_up)
-        mlp_out = F.linear(act, self.down_proj)
+        mlp_out = gemm_policy.linear(act, self.down_proj)
         if meta.lora_groups:
             apply_lora_slice(
                 mlp_out, act, meta.lora_groups, self.layer_idx, "down"
