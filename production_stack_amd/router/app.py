@@ -78,8 +78,13 @@ async def lifespan(app: FastAPI):
         await _http_session.close()
 
 
+_last_app = None
+
+
 def build_app() -> FastAPI:
+    global _last_app
     app = FastAPI(title="production-stack-amd router", lifespan=lifespan)
+    _last_app = app
 
     # ---- proxied OpenAI API -------------------------------------------
     PROXIED = [

@@ -111,6 +111,16 @@ router_memory_usage = Gauge(
 router_disk_usage = Gauge(
     "router_disk_usage_percent", "Router disk usage", registry=REGISTRY
 )
+semantic_cache_hits = Gauge(
+    "vllm:semantic_cache_hits", "Semantic cache hits", registry=REGISTRY
+)
+semantic_cache_misses = Gauge(
+    "vllm:semantic_cache_misses", "Semantic cache misses", registry=REGISTRY
+)
+semantic_cache_hit_rate = Gauge(
+    "vllm:semantic_cache_hit_ratio", "Semantic cache hit ratio",
+    registry=REGISTRY,
+)
 
 
 def fill_and_render() -> bytes:
@@ -149,6 +159,19 @@ def fill_and_render() -> bytes:
         in_prefill_requests.labels(server=url).set(rs.in_prefill_requests)
         in_decoding_requests.labels(server=url).set(rs.in_decoding_requests)
         finished_requests.labels(server=url).set(rs.finished_requests)
+
+    from production_stack_amd.router.utils import get_singleton  # noqa
+    try:
+        import production_stack_amd.router.app as _app_mod  # noqa
+    except ImportError:
+        _app_mod = None
+    sem = getattr(getattr(_app_mod, "_last_app", None), "state", None)
+    sem = getattr(sem, "semantic_cache", None) if sem else None
+    if sem is not None:
+        m = sem.metrics()
+        semantic_cache_hits.set(m["semantic_cache_hits"])
+        semantic_cache_misses.set(m["semantic_cache_misses"])
+        semantic_cache_hit_rate.set(m["semantic_cache_hit_rate"])
 
     router_cpu_usage.set(psutil.cpu_percent())
     router_memory_usage.set(psutil.virtual_memory().percent)

@@ -34,8 +34,28 @@ ext = CUDAExtension(
 )
 
 setup(
-    name="production-stack-amd-kernels",
+    name="production-stack-amd",
     version="0.1.0",
+    packages=[
+        "production_stack_amd",
+        "production_stack_amd.engine",
+        "production_stack_amd.engine.models",
+        "production_stack_amd.ops",
+        "production_stack_amd.router",
+        "production_stack_amd.router.examples",
+        "production_stack_amd.kvpool",
+        "production_stack_amd.parallel",
+        "production_stack_amd.gateway",
+    ],
+    entry_points={
+        "console_scripts": [
+            "vllm-router=production_stack_amd.router.app:main",
+            "ps-engine=production_stack_amd.engine.server:main",
+            "ps-kv-controller=production_stack_amd.kvpool.controller:main",
+            "ps-endpoint-picker="
+            "production_stack_amd.gateway.picker_service:main",
+        ]
+    },
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
