@@ -69,11 +69,15 @@ async def process_request(
     endpoint: str,
     request_id: str,
     debug_request: Optional[Any] = None,
+    model: Optional[str] = None,
 ):
     """Stream a request to a backend, yielding (first) status+headers then
     content chunks. Drives the request-stats lifecycle hooks."""
+    from production_stack_amd.router import metrics as rm
+
     monitor = get_request_stats_monitor()
-    monitor.on_new_request(backend_url, request_id, time.time())
+    start = time.time()
+    monitor.on_new_request(backend_url, request_id, start)
     session = _session()
     first = True
     async with session.request(
@@ -91,7 +95,11 @@ async def process_request(
                 )
                 first = False
             yield chunk
-    monitor.on_request_complete(backend_url, request_id, time.time())
+    end = time.time()
+    monitor.on_request_complete(backend_url, request_id, end)
+    rm.request_latency_hist.labels(model=model or "unknown").observe(
+        end - start
+    )
 
 
 def filter_endpoints(
@@ -125,6 +133,10 @@ async def route_general_request(
             content={"error": "invalid JSON body"},
         )
     requested_model = request_json.get("model")
+    try:
+        request.state.model_name = requested_model
+    except AttributeError:
+        pass
 
     app = request.app
     router = app.state.router
@@ -250,9 +262,15 @@ async def route_general_request(
                 request, body, server_url, endpoint, request_id
             )
         except (aiohttp.ClientError, OSError, TimeoutError) as e:
+            from production_stack_amd.router import metrics as rm
+
             get_request_stats_monitor().on_request_failed(
                 server_url, request_id
             )
+            rm.request_errors.labels(
+                model=requested_model or "unknown",
+                reason=type(e).__name__,
+            ).inc()
             last_error = f"{type(e).__name__}: {e}"
             tried.add(server_url)
             attempt += 1
@@ -277,6 +295,26 @@ async def route_general_request(
             "detail": last_error,
         },
     )
+
+
+def request_json_model_for_metrics(request) -> Optional[str]:
+    try:
+        return getattr(request.state, "model_name", None)
+    except AttributeError:
+        return None
+
+
+def _record_usage(model, payload) -> None:
+    from production_stack_amd.router import metrics as rm
+
+    usage = payload.get("usage") or {}
+    m = model or "unknown"
+    if usage.get("prompt_tokens"):
+        rm.model_input_tokens.labels(model=m).inc(usage["prompt_tokens"])
+    if usage.get("completion_tokens"):
+        rm.model_output_tokens.labels(model=m).inc(
+            usage["completion_tokens"]
+        )
 
 
 async def _proxy_buffered_with_cache(
@@ -306,7 +344,9 @@ async def _proxy_buffered_with_cache(
     monitor.on_request_complete(server_url, request_id, time.time())
     if status == 200:
         try:
-            sem.store(request_json, json.loads(data))
+            payload = json.loads(data)
+            sem.store(request_json, payload)
+            _record_usage(request_json.get("model"), payload)
         except (ValueError, TypeError):
             pass
     return Response(
@@ -356,7 +396,10 @@ async def _proxy_streaming(
     endpoint: str,
     request_id: str,
 ) -> Response:
-    gen = process_request(request, body, server_url, endpoint, request_id)
+    gen = process_request(
+        request, body, server_url, endpoint, request_id,
+        model=request_json_model_for_metrics(request),
+    )
     status, headers = await gen.__anext__()
     media_type = headers.get("content-type", "application/json")
     out_headers = {

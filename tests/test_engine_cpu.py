@@ -119,3 +119,50 @@ def test_metrics_surface():
         "gpu_prefix_cache_queries_total",
     ):
         assert key in m
+
+
+def test_concurrent_add_abort_step_thread_safety():
+    """Engine lock: concurrent add/abort from another thread while the step
+    loop runs must not corrupt block accounting."""
+    import threading
+
+    eng = make_engine(max_num_seqs=8, max_num_batched_tokens=256)
+    stop = threading.Event()
+    errors = []
+
+    def chaos():
+        i = 0
+        try:
+            while not stop.is_set():
+                rid = f"c{i}"
+                with eng.lock:
+                    eng.add_request(
+                        rid, list(range(10, 42)),
+                        SamplingParams(max_tokens=6, temperature=0.0,
+                                       ignore_eos=True),
+                    )
+                if i % 3 == 0:
+                    with eng.lock:
+                        eng.abort_request(rid)
+                i += 1
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    t = threading.Thread(target=chaos)
+    t.start()
+    import time as _t
+
+    deadline = _t.time() + 3
+    while _t.time() < deadline:
+        eng.step()
+    stop.set()
+    t.join()
+    assert not errors
+    # drain
+    for _ in range(500):
+        if not eng.has_unfinished():
+            break
+        eng.step()
+    assert not eng.has_unfinished()
+    bm = eng.block_manager
+    assert bm.num_free == bm.num_blocks
