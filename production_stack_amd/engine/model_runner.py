@@ -170,6 +170,8 @@ class ModelRunner:
         sample_rows: List[int] = []
         sample_seqs: List[Sequence] = []
 
+        import numpy as np
+
         TILE = 64
         for row, ss in enumerate(prefills):
             seq = ss.seq
@@ -183,12 +185,14 @@ class ModelRunner:
                     [row, q_flat0 + t0, start + t0,
                      min(TILE, ss.num_tokens - t0)]
                 )
-            for pos in range(start, end):
-                tokens.append(all_ids[pos])
-                positions.append(pos)
-                slots.append(seq.block_table[pos // bs] * bs + pos % bs)
-                p_token_seq.append(row)
-                p_token_pos.append(pos)
+            pos_arr = np.arange(start, end, dtype=np.int64)
+            bt_arr = np.asarray(seq.block_table, dtype=np.int64)
+            slot_arr = bt_arr[pos_arr // bs] * bs + pos_arr % bs
+            tokens.extend(all_ids[start:end])
+            positions.extend(pos_arr.tolist())
+            slots.extend(slot_arr.tolist())
+            p_token_seq.extend([row] * ss.num_tokens)
+            p_token_pos.extend(pos_arr.tolist())
             if end == seq.num_tokens:
                 sample_rows.append(len(tokens) - 1)
                 sample_seqs.append(seq)
