@@ -71,7 +71,7 @@ def main() -> None:
     ap.add_argument("--users", type=int, default=128, help="conversations per GPU")
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--max-model-len", type=int, default=4096)
-    ap.add_argument("--max-num-batched-tokens", type=int, default=8192)
+    ap.add_argument("--max-num-batched-tokens", type=int, default=2048)
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 

@@ -592,7 +592,7 @@ def main() -> None:
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--dtype", default="bfloat16")
     ap.add_argument("--max-num-seqs", type=int, default=256)
-    ap.add_argument("--max-num-batched-tokens", type=int, default=8192)
+    ap.add_argument("--max-num-batched-tokens", type=int, default=2048)
     ap.add_argument("--gpu-memory-utilization", type=float, default=0.85)
     ap.add_argument("--num-gpu-blocks", type=int, default=None)
     ap.add_argument("--enable-prefix-caching", action="store_true",
