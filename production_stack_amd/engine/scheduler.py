@@ -116,7 +116,6 @@ class Scheduler:
     def schedule(self) -> SchedulerOutput:
         out = SchedulerOutput()
         budget = self.config.max_num_batched_tokens
-        scheduled_set = set()
 
         # 1. running sequences, oldest first: decodes (remaining == 1) and
         #    in-flight chunked prefills.
@@ -151,7 +150,6 @@ class Scheduler:
                     break
             if chunk > 0:
                 out.scheduled.append(ScheduledSeq(seq, chunk))
-                scheduled_set.add(seq.request_id)
                 budget -= chunk
 
         # 2. admit waiting sequences.
@@ -176,7 +174,6 @@ class Scheduler:
             remaining = seq.num_tokens - seq.num_computed
             chunk = min(remaining, budget, self.config.max_prefill_chunk)
             out.scheduled.append(ScheduledSeq(seq, chunk))
-            scheduled_set.add(seq.request_id)
             budget -= chunk
 
         return out

@@ -160,7 +160,6 @@ def fill_and_render() -> bytes:
         in_decoding_requests.labels(server=url).set(rs.in_decoding_requests)
         finished_requests.labels(server=url).set(rs.finished_requests)
 
-    from production_stack_amd.router.utils import get_singleton  # noqa
     try:
         import production_stack_amd.router.app as _app_mod  # noqa
     except ImportError:

@@ -68,7 +68,6 @@ async def process_request(
     backend_url: str,
     endpoint: str,
     request_id: str,
-    debug_request: Optional[Any] = None,
     model: Optional[str] = None,
 ):
     """Stream a request to a backend, yielding (first) status+headers then
@@ -198,7 +197,7 @@ async def route_general_request(
     if callbacks and hasattr(callbacks, "pre_request"):
         maybe = callbacks.pre_request(request, request_json, requested_model)
         if maybe is not None:
-            request_json, body_changed = maybe, True
+            request_json = maybe
             body = json.dumps(request_json).encode()
     rewriter = getattr(app.state, "request_rewriter", None)
     if rewriter is not None:
