@@ -72,6 +72,8 @@ def main() -> None:
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--max-num-batched-tokens", type=int, default=2048)
+    ap.add_argument("--kv-cache-dtype", default="auto",
+                    choices=["auto", "bf16", "fp8", "fp8_e4m3"])
     ap.add_argument("--device", default=None)
     args = ap.parse_args()
 
@@ -98,6 +100,7 @@ def main() -> None:
             gpu_memory_utilization=0.85,
             enable_prefix_caching=True,
             num_gpu_blocks=None if use_cuda else 2048,
+            kv_cache_dtype=args.kv_cache_dtype,
         ),
         scheduler=SchedulerConfig(
             max_num_seqs=max(args.users * 2, 64),
@@ -218,6 +221,7 @@ def main() -> None:
                         "max_model_len": args.max_model_len,
                         "parallelism": f"dp{world}",
                         "prefix_caching": True,
+                        "kv_cache_dtype": args.kv_cache_dtype,
                     },
                 }
             )

@@ -28,17 +28,20 @@
 // sizes of 16-64 seqs, seqs*kv_heads workgroups cannot fill 256 CUs and the
 // kernel runs at <10% of HBM bandwidth (measured, profiles/ run1).
 // ---------------------------------------------------------------------------
-template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES, int LOWREG>
+template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES, int LOWREG,
+          typename KVT>
 __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     unsigned short* __restrict__ out,            // [S, QH, HEAD_DIM]
     float* __restrict__ ws_acc,   // [S, QH, SPLITS, HD] (splits > 1)
     float* __restrict__ ws_ml,    // [S, QH, SPLITS, 2]
     const unsigned short* __restrict__ q,        // [S, QH, HEAD_DIM]
-    const unsigned short* __restrict__ k_cache,  // [NB, KH, BS, HD]
-    const unsigned short* __restrict__ v_cache,  // [NB, KH, BS, HD]
+    const KVT* __restrict__ k_cache,             // [NB, KH, BS, HD]
+    const KVT* __restrict__ v_cache,             // [NB, KH, BS, HD]
     const int* __restrict__ block_tables,        // [S, max_blocks]
     const int* __restrict__ seq_lens,            // [S]
     int max_blocks, float scale, int KH, long q_stride) {
+  using KVTr = ps_kv_traits<KVT>;
+  using kvec8 = typename KVTr::vec8;
   constexpr int D = HEAD_DIM;
   constexpr int LPG = D / 8;       // lanes per token sub-group (16 @ D=128)
   constexpr int TPW = 64 / LPG;    // tokens per wave per iteration (4)
@@ -95,11 +98,11 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
   constexpr int TPB = BLOCK_SIZE / TPW;  // tokens per sub-group per block (4)
   for (int b = b_begin + wave; b < b_end; b += NWAVES) {
     const long blk = bt[b];
-    const unsigned short* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
-    const unsigned short* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
+    const KVT* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
+    const KVT* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
     const int valid_tokens = min(BLOCK_SIZE, ctx - b * BLOCK_SIZE);
     float sc[GQ][TPB];
-    ps_bf16x8 vv[TPB];
+    kvec8 vv[TPB];
     if constexpr (LOWREG) {
       // keep q packed: block the compiler from hoisting converted copies
 #pragma unroll
@@ -110,18 +113,18 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     for (int tt = 0; tt < TPB; tt++) {
       const int tok = tt * TPW + sg;
       const bool valid = tok < valid_tokens;
-      ps_bf16x8 kv = *(const ps_bf16x8*)(kb + tok * D + sl * 8);
+      kvec8 kv = *(const kvec8*)(kb + tok * D + sl * 8);
       if constexpr (!LOWREG)
-        vv[tt] = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
+        vv[tt] = *(const kvec8*)(vb + tok * D + sl * 8);
 #pragma unroll
       for (int g = 0; g < GQ; g++) {
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < 8; j++) {
           if constexpr (LOWREG)
-            s += ps_bf16_to_f32(qp[g][j]) * ps_bf16_to_f32(kv[j]);
+            s += ps_bf16_to_f32(qp[g][j]) * KVTr::to_f32(kv[j]);
           else
-            s += qf[g][j] * ps_bf16_to_f32(kv[j]);
+            s += qf[g][j] * KVTr::to_f32(kv[j]);
         }
         s = ps_group_sum<LPG>(s) * scale;
         sc[g][tt] = valid ? s : PS_NEG_INF;
@@ -130,7 +133,7 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
     if constexpr (LOWREG) {
 #pragma unroll
       for (int tt = 0; tt < TPB; tt++)
-        vv[tt] = *(const ps_bf16x8*)(vb + (tt * TPW + sg) * D + sl * 8);
+        vv[tt] = *(const kvec8*)(vb + (tt * TPW + sg) * D + sl * 8);
     }
 #pragma unroll
     for (int g = 0; g < GQ; g++) {
@@ -152,7 +155,7 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
         float a = acc[g][j] * corr;
 #pragma unroll
         for (int tt = 0; tt < TPB; tt++)
-          a += p[tt] * ps_bf16_to_f32(vv[tt][j]);
+          a += p[tt] * KVTr::to_f32(vv[tt][j]);
         acc[g][j] = a;
       }
       m[g] = mnew;
@@ -372,13 +375,16 @@ __global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
 // KV append (reshape_and_cache): scatter the new tokens' K/V into the paged
 // cache according to slot_mapping (slot = block_id * BLOCK_SIZE + offset).
 // ---------------------------------------------------------------------------
+template <typename KVT>
 __global__ void reshape_and_cache_kernel(
     const unsigned short* __restrict__ k,  // [T, KH*HD]
     const unsigned short* __restrict__ v,  // [T, KH*HD]
-    unsigned short* __restrict__ k_cache,  // [NB, KH, BS, HD]
-    unsigned short* __restrict__ v_cache,
+    KVT* __restrict__ k_cache,             // [NB, KH, BS, HD]
+    KVT* __restrict__ v_cache,
     const long* __restrict__ slot_mapping,  // [T]
     int KH, int HD, int BS, long total /* = T * KH * HD / 8 */) {
+  using KVTr = ps_kv_traits<KVT>;
+  using kvec8 = typename KVTr::vec8;
   const long stride = (long)gridDim.x * blockDim.x;
   const int row = KH * HD;
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
@@ -393,8 +399,16 @@ __global__ void reshape_and_cache_kernel(
     const long blk = slot / BS;
     const int off = (int)(slot % BS);
     const long dst = ((blk * KH + h) * BS + off) * (long)HD + d;
-    *(ps_bf16x8*)(k_cache + dst) = *(const ps_bf16x8*)(k + elem);
-    *(ps_bf16x8*)(v_cache + dst) = *(const ps_bf16x8*)(v + elem);
+    ps_bf16x8 kv = *(const ps_bf16x8*)(k + elem);
+    ps_bf16x8 vv = *(const ps_bf16x8*)(v + elem);
+    kvec8 ko, vo;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      ko[j] = KVTr::from_f32(ps_bf16_to_f32(kv[j]));
+      vo[j] = KVTr::from_f32(ps_bf16_to_f32(vv[j]));
+    }
+    *(kvec8*)(k_cache + dst) = ko;
+    *(kvec8*)(v_cache + dst) = vo;
   }
 }
 
@@ -543,15 +557,22 @@ int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
 
 void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
                           void* v_cache, const void* slot_mapping, long T,
-                          int KH, int HD, int BS, hipStream_t stream) {
+                          int KH, int HD, int BS, int kv_fp8,
+                          hipStream_t stream) {
   const long total = T * (long)KH * HD / 8;
   const int block = 256;
   const long want = (total + block - 1) / block;
   const int grid = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
-  reshape_and_cache_kernel<<<grid, block, 0, stream>>>(
-      (const unsigned short*)k, (const unsigned short*)v,
-      (unsigned short*)k_cache, (unsigned short*)v_cache,
-      (const long*)slot_mapping, KH, HD, BS, total);
+  if (kv_fp8)
+    reshape_and_cache_kernel<unsigned char><<<grid, block, 0, stream>>>(
+        (const unsigned short*)k, (const unsigned short*)v,
+        (unsigned char*)k_cache, (unsigned char*)v_cache,
+        (const long*)slot_mapping, KH, HD, BS, total);
+  else
+    reshape_and_cache_kernel<unsigned short><<<grid, block, 0, stream>>>(
+        (const unsigned short*)k, (const unsigned short*)v,
+        (unsigned short*)k_cache, (unsigned short*)v_cache,
+        (const long*)slot_mapping, KH, HD, BS, total);
 }
 
 void ps_greedy_sample(void* out, const void* logits, long R, int V,

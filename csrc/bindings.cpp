@@ -22,20 +22,21 @@ void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
 void ps_fused_rope_cache(void* qkv, const void* positions, const void* cos_sin,
                          const void* slot_mapping, void* k_cache,
                          void* v_cache, long T, int QH, int KH, int HD,
-                         int ROT, long qkv_stride, int BS,
+                         int ROT, long qkv_stride, int BS, int kv_fp8,
                          hipStream_t stream);
 int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* k_cache, const void* v_cache,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         long q_stride, int variant, hipStream_t stream);
+                         long q_stride, int variant, int kv_fp8,
+                         hipStream_t stream);
 int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* v_cache, const void* block_tables,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
                                int KH, int GQ, int head_dim, long q_stride,
-                               int variant, hipStream_t stream);
+                               int variant, int kv_fp8, hipStream_t stream);
 int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* v_cache, const void* block_tables,
                           const void* token_seq, const void* token_pos,
@@ -44,7 +45,8 @@ int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           int block_size, long q_stride, hipStream_t stream);
 void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
                           void* v_cache, const void* slot_mapping, long T,
-                          int KH, int HD, int BS, hipStream_t stream);
+                          int KH, int HD, int BS, int kv_fp8,
+                          hipStream_t stream);
 void ps_greedy_sample(void* out, const void* logits, long R, int V,
                       hipStream_t stream);
 void ps_kv_quant(void* out, void* scales, const void* in, long rows, int hd,
@@ -70,6 +72,15 @@ hipStream_t current_stream() {
   TORCH_CHECK((t).is_cuda(), #t " must be on the GPU");         \
   TORCH_CHECK((t).scalar_type() == (ty), #t " dtype mismatch"); \
   TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+// KV caches are bf16 or OCP fp8 e4m3; returns 1 for fp8
+int cache_fp8(const at::Tensor& c) {
+  TORCH_CHECK(c.is_cuda() && c.is_contiguous(), "cache must be GPU-contig");
+  if (c.scalar_type() == at::kBFloat16) return 0;
+  if (c.scalar_type() == at::kFloat8_e4m3fn) return 1;
+  TORCH_CHECK(false, "KV cache must be bf16 or float8_e4m3fn");
+  return 0;
+}
 
 // q may be a row-strided view into the packed qkv tensor
 long q_row_stride(const at::Tensor& q, int HD) {
@@ -133,8 +144,8 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor seq_lens, double scale, int64_t num_splits,
                        int64_t variant) {
   CHECK_GPU_BF16(out);
-  CHECK_GPU_BF16(k_cache);
-  CHECK_GPU_BF16(v_cache);
+  const int kv_fp8 = cache_fp8(k_cache);
+  cache_fp8(v_cache);
   CHECK_GPU_DTYPE(block_tables, at::kInt);
   CHECK_GPU_DTYPE(seq_lens, at::kInt);
   const int S = (int)q.size(0);
@@ -164,7 +175,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       out.data_ptr(), acc_p, ml_p, q.data_ptr(), k_cache.data_ptr(),
       v_cache.data_ptr(), block_tables.data_ptr(), seq_lens.data_ptr(), S,
       max_blocks, (float)scale, KH, GQ, HD, BS, (int)num_splits,
-      q_row_stride(q, HD), (int)variant, current_stream());
+      q_row_stride(q, HD), (int)variant, kv_fp8, current_stream());
   TORCH_CHECK(rc == 0, "unsupported decode config: head_dim=", HD,
               " block_size=", BS, " gqa=", GQ);
 }
@@ -201,8 +212,8 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
                              at::Tensor block_tables, at::Tensor tile_info,
                              double scale, int64_t variant) {
   CHECK_GPU_BF16(out);
-  CHECK_GPU_BF16(k_cache);
-  CHECK_GPU_BF16(v_cache);
+  const int kv_fp8 = cache_fp8(k_cache);
+  cache_fp8(v_cache);
   CHECK_GPU_DTYPE(block_tables, at::kInt);
   CHECK_GPU_DTYPE(tile_info, at::kInt);
   const int QH = (int)q.size(1);
@@ -216,7 +227,7 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr(), tile_info.data_ptr(), NT, QH,
       (int)block_tables.size(1), (float)scale, KH, GQ, HD,
-      q_row_stride(q, HD), (int)variant, current_stream());
+      q_row_stride(q, HD), (int)variant, kv_fp8, current_stream());
   TORCH_CHECK(rc == 0, "unsupported mfma prefill config: head_dim=", HD);
 }
 
@@ -228,8 +239,8 @@ void fused_rope_cache(at::Tensor qkv, at::Tensor positions,
   CHECK_GPU_DTYPE(positions, at::kInt);
   CHECK_GPU_DTYPE(cos_sin, at::kFloat);
   CHECK_GPU_DTYPE(slot_mapping, at::kLong);
-  CHECK_GPU_BF16(k_cache);
-  CHECK_GPU_BF16(v_cache);
+  const int kv_fp8 = cache_fp8(k_cache);
+  cache_fp8(v_cache);
   const long T = positions.size(0);
   const int KH = (int)k_cache.size(1);
   const int BS = (int)k_cache.size(2);
@@ -241,7 +252,7 @@ void fused_rope_cache(at::Tensor qkv, at::Tensor positions,
   ps_fused_rope_cache(qkv.data_ptr(), positions.data_ptr(),
                       cos_sin.data_ptr(), slot_mapping.data_ptr(),
                       k_cache.data_ptr(), v_cache.data_ptr(), T,
-                      (int)q_heads, KH, HD, ROT, qkv.stride(0), BS,
+                      (int)q_heads, KH, HD, ROT, qkv.stride(0), BS, kv_fp8,
                       current_stream());
 }
 
@@ -249,8 +260,8 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor slot_mapping) {
   CHECK_GPU_BF16(k);
   CHECK_GPU_BF16(v);
-  CHECK_GPU_BF16(k_cache);
-  CHECK_GPU_BF16(v_cache);
+  const int kv_fp8 = cache_fp8(k_cache);
+  cache_fp8(v_cache);
   CHECK_GPU_DTYPE(slot_mapping, at::kLong);
   const long T = slot_mapping.size(0);
   const int KH = (int)k_cache.size(1);
@@ -259,7 +270,7 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
   TORCH_CHECK(k.numel() == T * KH * HD, "k shape mismatch");
   ps_reshape_and_cache(k.data_ptr(), v.data_ptr(), k_cache.data_ptr(),
                        v_cache.data_ptr(), slot_mapping.data_ptr(), T, KH, HD,
-                       BS, current_stream());
+                       BS, kv_fp8, current_stream());
 }
 
 void skinny_gemm(at::Tensor out_f32, at::Tensor x, at::Tensor w) {

@@ -240,11 +240,14 @@ void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
 // scatters the rotated k and v rows into the paged cache. Replaces
 // rope_kernel + reshape_and_cache_kernel + three .contiguous() copies.
 // ---------------------------------------------------------------------------
+template <typename KVT>
 __global__ void fused_rope_cache_kernel(
     unsigned short* __restrict__ qkv, const int* __restrict__ positions,
     const float* __restrict__ cos_sin, const long* __restrict__ slot_mapping,
-    unsigned short* __restrict__ k_cache, unsigned short* __restrict__ v_cache,
+    KVT* __restrict__ k_cache, KVT* __restrict__ v_cache,
     int QH, int KH, int HD, int ROT, long qkv_stride, int BS) {
+  using KVTr = ps_kv_traits<KVT>;
+  using kvec8 = typename KVTr::vec8;
   const long t = blockIdx.x;
   unsigned short* base = qkv + t * qkv_stride;
   const float* cs = cos_sin + (long)positions[t] * ROT;
@@ -275,9 +278,14 @@ __global__ void fused_rope_cache_kernel(
     const int h = r / HD;
     const int d = r % HD;
     const long dst = ((blk * KH + h) * BS + off) * (long)HD + d;
-    unsigned short* cache = isv ? v_cache : k_cache;
+    KVT* cache = isv ? v_cache : k_cache;
     const unsigned short* src = (isv ? vsrc : ksrc) + r;
-    *(ps_bf16x8*)(cache + dst) = *(const ps_bf16x8*)src;
+    ps_bf16x8 sv = *(const ps_bf16x8*)src;
+    kvec8 ov;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      ov[j] = KVTr::from_f32(ps_bf16_to_f32(sv[j]));
+    *(kvec8*)(cache + dst) = ov;
   }
 }
 
@@ -286,11 +294,21 @@ extern "C" void ps_fused_rope_cache(void* qkv, const void* positions,
                                     const void* slot_mapping, void* k_cache,
                                     void* v_cache, long T, int QH, int KH,
                                     int HD, int ROT, long qkv_stride, int BS,
-                                    hipStream_t stream) {
-  fused_rope_cache_kernel<<<dim3((unsigned)T), 256, 0, stream>>>(
-      (unsigned short*)qkv, (const int*)positions, (const float*)cos_sin,
-      (const long*)slot_mapping, (unsigned short*)k_cache,
-      (unsigned short*)v_cache, QH, KH, HD, ROT, qkv_stride, BS);
+                                    int kv_fp8, hipStream_t stream) {
+  if (kv_fp8)
+    fused_rope_cache_kernel<unsigned char>
+        <<<dim3((unsigned)T), 256, 0, stream>>>(
+            (unsigned short*)qkv, (const int*)positions,
+            (const float*)cos_sin, (const long*)slot_mapping,
+            (unsigned char*)k_cache, (unsigned char*)v_cache, QH, KH, HD,
+            ROT, qkv_stride, BS);
+  else
+    fused_rope_cache_kernel<unsigned short>
+        <<<dim3((unsigned)T), 256, 0, stream>>>(
+            (unsigned short*)qkv, (const int*)positions,
+            (const float*)cos_sin, (const long*)slot_mapping,
+            (unsigned short*)k_cache, (unsigned short*)v_cache, QH, KH, HD,
+            ROT, qkv_stride, BS);
 }
 
 // ---------------------------------------------------------------------------

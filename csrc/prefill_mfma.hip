@@ -36,15 +36,17 @@ PS_DEV ps_mbf16x8 ps_as_mbf16(ps_bf16x8 u) {
 #define PS_PL_STRIDE 72  // P row stride in tokens (multiple of 8)
 
 // tile_info: int4 per tile = (seq_row, q_token_start, q_pos_start, n_rows)
-template <int HEAD_DIM, int WPS>
+template <int HEAD_DIM, int WPS, typename KVT>
 __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     unsigned short* __restrict__ out,            // [T, QH, HD]
     const unsigned short* __restrict__ q,        // [T, QH, HD]
-    const unsigned short* __restrict__ k_cache,  // [NB, KH, 16, HD]
-    const unsigned short* __restrict__ v_cache,
+    const KVT* __restrict__ k_cache,             // [NB, KH, 16, HD]
+    const KVT* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ tile_info,     // [NT, 4]
     int max_blocks, float scale, int KH, int GQ, long q_stride) {
+  using KVTr = ps_kv_traits<KVT>;
+  using kvec8 = typename KVTr::vec8;
   constexpr int D = HEAD_DIM;  // 128
   constexpr int BS = 16;       // page size in tokens
   constexpr int NK = D / 32;   // mfma k-steps over head dim (4)
@@ -122,15 +124,15 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     const long pg = bt[pg_idx];
     return k_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
   };
-  ps_bf16x8 vstage[4];
-  ps_bf16x8 kstage[4];
+  kvec8 vstage[4];
+  kvec8 kstage[4];
   {
-    const unsigned short* vrow = v_row_ptr(0);
-    const unsigned short* krow = k_row_ptr(0);
+    const KVT* vrow = v_row_ptr(0);
+    const KVT* krow = k_row_ptr(0);
 #pragma unroll
     for (int h = 0; h < 4; h++) {
-      vstage[h] = *(const ps_bf16x8*)(vrow + d0 + h * 8);
-      kstage[h] = *(const ps_bf16x8*)(krow + d0 + h * 8);
+      vstage[h] = *(const kvec8*)(vrow + d0 + h * 8);
+      kstage[h] = *(const kvec8*)(krow + d0 + h * 8);
     }
   }
 
@@ -142,21 +144,28 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 #pragma unroll
     for (int h = 0; h < 4; h++) {
       const int slot = ((wave * 4 + h) ^ (tv & 7));
-      *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = kstage[h];
+      if constexpr (sizeof(KVT) == 2) {
+        *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = kstage[h];
+      } else {
+        ps_bf16x8 kb16;
+#pragma unroll
+        for (int j = 0; j < 8; j++) kb16[j] = KVTr::to_bf16(kstage[h][j]);
+        *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = kb16;
+      }
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         const int d = d0 + h * 8 + j;
-        v_t[d][((lg ^ (d & 7)) << 3) + tl] = vstage[h][j];
+        v_t[d][((lg ^ (d & 7)) << 3) + tl] = KVTr::to_bf16(vstage[h][j]);
       }
     }
     __syncthreads();
     if (chunk + 1 < n_chunks) {
-      const unsigned short* vrow = v_row_ptr(chunk + 1);
-      const unsigned short* krow = k_row_ptr(chunk + 1);
+      const KVT* vrow = v_row_ptr(chunk + 1);
+      const KVT* krow = k_row_ptr(chunk + 1);
 #pragma unroll
       for (int h = 0; h < 4; h++) {
-        vstage[h] = *(const ps_bf16x8*)(vrow + d0 + h * 8);
-        kstage[h] = *(const ps_bf16x8*)(krow + d0 + h * 8);
+        vstage[h] = *(const kvec8*)(vrow + d0 + h * 8);
+        kstage[h] = *(const kvec8*)(krow + d0 + h * 8);
       }
     }
 
@@ -276,21 +285,22 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
                                int KH, int GQ, int head_dim, long q_stride,
-                               int variant, hipStream_t stream) {
+                               int variant, int kv_fp8, hipStream_t stream) {
   if (head_dim != 128) return -1;
   dim3 grid(num_tiles, num_q_heads);
-  if (variant == 3)
-    paged_attn_prefill_mfma_kernel<128, 3><<<grid, 256, 0, stream>>>(
-        (unsigned short*)out, (const unsigned short*)q,
-        (const unsigned short*)k_cache, (const unsigned short*)v_cache,
-        (const int*)block_tables, (const int*)tile_info, max_blocks, scale,
-        KH, GQ, q_stride);
-  else
-    paged_attn_prefill_mfma_kernel<128, 4><<<grid, 256, 0, stream>>>(
-        (unsigned short*)out, (const unsigned short*)q,
-        (const unsigned short*)k_cache, (const unsigned short*)v_cache,
-        (const int*)block_tables, (const int*)tile_info, max_blocks, scale,
-        KH, GQ, q_stride);
+#define PS_PREFILL_T(WPS, KVT)                                              \
+  paged_attn_prefill_mfma_kernel<128, WPS, KVT><<<grid, 256, 0, stream>>>(  \
+      (unsigned short*)out, (const unsigned short*)q,                       \
+      (const KVT*)k_cache, (const KVT*)v_cache, (const int*)block_tables,   \
+      (const int*)tile_info, max_blocks, scale, KH, GQ, q_stride)
+  if (kv_fp8) {
+    if (variant == 3) PS_PREFILL_T(3, unsigned char);
+    else PS_PREFILL_T(4, unsigned char);
+  } else {
+    if (variant == 3) PS_PREFILL_T(3, unsigned short);
+    else PS_PREFILL_T(4, unsigned short);
+  }
+#undef PS_PREFILL_T
   return 0;
 }
 

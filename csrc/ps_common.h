@@ -7,6 +7,7 @@
 //    the coalescing sweet spot on CDNA4 (Guideline 13).
 #pragma once
 
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 
 #define PS_DEV __device__ __forceinline__
@@ -55,5 +56,38 @@ PS_DEV float ps_group_max(float v) {
 }
 
 #define PS_NEG_INF (-3.0e38f)
+
+// ---- KV-cache element traits: bf16 (ushort) or OCP fp8 e4m3 (uchar) ----
+typedef __attribute__((ext_vector_type(8))) unsigned char ps_fp8x8;  // 8 B
+
+PS_DEV float ps_fp8_to_f32(unsigned char u) {
+  __hip_fp8_e4m3 v;
+  v.__x = u;
+  return (float)v;
+}
+PS_DEV unsigned char ps_f32_to_fp8(float f) {
+  return __hip_fp8_e4m3(f).__x;
+}
+
+template <typename KVT>
+struct ps_kv_traits;
+
+template <>
+struct ps_kv_traits<unsigned short> {  // bf16 cache
+  using vec8 = ps_bf16x8;
+  static PS_DEV float to_f32(unsigned short u) { return ps_bf16_to_f32(u); }
+  static PS_DEV unsigned short from_f32(float f) { return ps_f32_to_bf16(f); }
+  static PS_DEV unsigned short to_bf16(unsigned short u) { return u; }
+};
+
+template <>
+struct ps_kv_traits<unsigned char> {  // fp8 e4m3 cache
+  using vec8 = ps_fp8x8;
+  static PS_DEV float to_f32(unsigned char u) { return ps_fp8_to_f32(u); }
+  static PS_DEV unsigned char from_f32(float f) { return ps_f32_to_fp8(f); }
+  static PS_DEV unsigned short to_bf16(unsigned char u) {
+    return ps_f32_to_bf16(ps_fp8_to_f32(u));
+  }
+};
 
 static inline int ps_cdiv(int a, int b) { return (a + b - 1) / b; }

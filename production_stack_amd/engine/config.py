@@ -111,6 +111,9 @@ class CacheConfig:
     # "bf16" (raw) or "int8" (CacheGen-style row-quantized serde, halves
     # host-pool bytes; reference LMCACHE remote serde surface)
     offload_dtype: str = "bf16"
+    # KV cache storage dtype: "auto"/"bf16" or "fp8"/"fp8_e4m3" (OCP e4m3;
+    # halves decode KV bandwidth; vLLM --kv-cache-dtype parity)
+    kv_cache_dtype: str = "auto"
 
 
 @dataclass

@@ -74,6 +74,14 @@ class ModelRunner:
         )
 
     # ------------------------------------------------------------------
+    @property
+    def kv_elem_bytes(self) -> int:
+        return (
+            1
+            if self.config.cache.kv_cache_dtype in ("fp8", "fp8_e4m3")
+            else 2
+        )
+
     def profile_num_blocks(self) -> int:
         """Size the KV cache from free HBM after weights (288 GB per GPU)."""
         cache_cfg = self.config.cache
@@ -87,6 +95,8 @@ class ModelRunner:
             - (total - free)
         )
         per_block = self.model.kv_bytes_per_block(cache_cfg.block_size)
+        if self.kv_elem_bytes == 1:
+            per_block //= 2
         n = max(budget // per_block, 16)
         return int(n)
 
@@ -95,11 +105,17 @@ class ModelRunner:
         cfg = self.model_cfg
         bs = self.config.cache.block_size
         kh = self.model.kv_heads
+        kv_dt = self.config.cache.kv_cache_dtype
+        dtype = (
+            torch.float8_e4m3fn
+            if kv_dt in ("fp8", "fp8_e4m3")
+            else torch.bfloat16
+        )
         self.kv_caches = []
         for _ in range(self.model.num_local_layers):
             k = torch.zeros(
                 (num_blocks, kh, bs, cfg.head_dim),
-                dtype=torch.bfloat16,
+                dtype=dtype,
                 device=self.device,
             )
             v = torch.zeros_like(k)

@@ -85,8 +85,8 @@ def reshape_and_cache(
     slots = slot_mapping[mask]
     blk = torch.div(slots, bs, rounding_mode="floor").long()
     off = (slots % bs).long()
-    k_cache[blk, :, off] = k.view(-1, kh, hd)[mask]
-    v_cache[blk, :, off] = v.view(-1, kh, hd)[mask]
+    k_cache[blk, :, off] = k.view(-1, kh, hd)[mask].to(k_cache.dtype)
+    v_cache[blk, :, off] = v.view(-1, kh, hd)[mask].to(v_cache.dtype)
 
 
 def _gather_kv(

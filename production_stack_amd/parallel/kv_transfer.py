@@ -87,8 +87,12 @@ class KVTransferService:
             device=self.engine.device,
         )
         for li, (kc, vc) in enumerate(self._caches()):
-            pack[:, li, 0] = kc.index_select(0, idx).flatten(1)
-            pack[:, li, 1] = vc.index_select(0, idx).flatten(1)
+            pack[:, li, 0] = (
+                kc.index_select(0, idx).flatten(1).to(pack.dtype)
+            )
+            pack[:, li, 1] = (
+                vc.index_select(0, idx).flatten(1).to(pack.dtype)
+            )
         return pack
 
     def scatter_blocks(
@@ -100,10 +104,10 @@ class KVTransferService:
                            device=self.engine.device)
         for li, (kc, vc) in enumerate(self._caches()):
             kc.index_copy_(
-                0, idx, pack[:, li, 0].view(-1, kh, bs, hd)
+                0, idx, pack[:, li, 0].view(-1, kh, bs, hd).to(kc.dtype)
             )
             vc.index_copy_(
-                0, idx, pack[:, li, 1].view(-1, kh, bs, hd)
+                0, idx, pack[:, li, 1].view(-1, kh, bs, hd).to(vc.dtype)
             )
 
     # ---- prefill (kv_producer) side -----------------------------------

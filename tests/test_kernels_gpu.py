@@ -359,3 +359,114 @@ def test_skinny_gemm_strided_x():
     got = ops.skinny_gemm(x, w)
     want = (x.float() @ w.float().t()).to(torch.bfloat16)
     _close(got, want, atol=3e-2, rtol=3e-2)
+
+
+def test_paged_attn_decode_fp8_kv():
+    from production_stack_amd import _C
+
+    torch.manual_seed(9)
+    qh, kh, hd, bs = 32, 8, 128, 16
+    S = 3
+    seq_lens = torch.tensor([5, 300, 64], dtype=torch.int32)
+    max_blocks = 20
+    nb = S * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    k8 = k_cache.to(torch.float8_e4m3fn)
+    v8 = v_cache.to(torch.float8_e4m3fn)
+    block_tables = torch.arange(1, S * max_blocks + 1, dtype=torch.int32).reshape(
+        S, max_blocks
+    )
+    q = torch.randn((S, qh, hd), dtype=torch.bfloat16, device="cuda")
+    # reference on the DEQUANTIZED cache (fp8 is lossy by design)
+    want = reference.paged_attn_decode(
+        q.cpu(), k8.cpu().to(torch.bfloat16), v8.cpu().to(torch.bfloat16),
+        block_tables, seq_lens, 1.0 / hd ** 0.5,
+    )
+    out = torch.empty_like(q)
+    _C.paged_attn_decode(
+        out, q, k8, v8, block_tables.cuda(), seq_lens.cuda(),
+        1.0 / hd ** 0.5, 0, 1,
+    )
+    _close(out, want, atol=3e-2, rtol=3e-2)
+
+
+def test_paged_attn_prefill_mfma_fp8_kv():
+    from production_stack_amd import _C
+
+    torch.manual_seed(10)
+    qh, kh, hd, bs = 32, 8, 128, 16
+    chunks = [(0, 0, 150), (1, 64, 80)]
+    max_blocks = 16
+    nb = 2 * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    k8 = k_cache.to(torch.float8_e4m3fn)
+    v8 = v_cache.to(torch.float8_e4m3fn)
+    block_tables = torch.arange(1, 2 * max_blocks + 1, dtype=torch.int32).reshape(
+        2, max_blocks
+    )
+    tiles, token_seq, token_pos = _build_tiles(chunks)
+    T = token_seq.shape[0]
+    q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_prefill(
+        q.cpu(), k8.cpu().to(torch.bfloat16), v8.cpu().to(torch.bfloat16),
+        block_tables, token_seq, token_pos, 1.0 / hd ** 0.5,
+    )
+    out = torch.empty_like(q)
+    _C.paged_attn_prefill_mfma(
+        out, q, k8, v8, block_tables.cuda(), tiles.cuda(),
+        1.0 / hd ** 0.5, 3,
+    )
+    _close(out, want, atol=3e-2, rtol=3e-2)
+
+
+def test_fused_rope_cache_fp8():
+    from production_stack_amd import _C
+
+    torch.manual_seed(11)
+    T, qh, kh, hd, bs, nb = 9, 8, 4, 128, 16, 4
+    width = (qh + 2 * kh) * hd
+    qkv = torch.randn((T, width), dtype=torch.bfloat16, device="cuda")
+    inv = 1.0 / (500000.0 ** (torch.arange(0, hd, 2).float() / hd))
+    freqs = torch.outer(torch.arange(64).float(), inv)
+    cos_sin = torch.cat([freqs.cos(), freqs.sin()], -1).contiguous().cuda()
+    pos = torch.randint(0, 64, (T,), dtype=torch.int32, device="cuda")
+    slots = torch.randperm(nb * bs)[:T].to(torch.long).cuda()
+    k8 = torch.zeros(nb, kh, bs, hd, dtype=torch.float8_e4m3fn, device="cuda")
+    v8 = torch.zeros_like(k8)
+    kb = torch.zeros(nb, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
+    vb = torch.zeros_like(kb)
+    qkv2 = qkv.clone()
+    _C.fused_rope_cache(qkv2, pos, cos_sin, slots, kb, vb, qh, hd)
+    _C.fused_rope_cache(qkv.clone(), pos, cos_sin, slots, k8, v8, qh, hd)
+    _close(k8.to(torch.bfloat16), kb, atol=4e-2, rtol=6e-2)
+    _close(v8.to(torch.bfloat16), vb, atol=4e-2, rtol=6e-2)
+
+
+def test_engine_fp8_kv_e2e():
+    from production_stack_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        SchedulerConfig,
+    )
+    from production_stack_amd.engine.engine import LLMEngine
+    from production_stack_amd.engine.sampling import SamplingParams
+
+    def run(kv_dtype):
+        cfg = EngineConfig(
+            model="mini-llama",
+            max_model_len=1024,
+            seed=7,
+            cache=CacheConfig(num_gpu_blocks=128, block_size=16,
+                              kv_cache_dtype=kv_dtype),
+            scheduler=SchedulerConfig(max_num_seqs=8,
+                                      max_num_batched_tokens=1024),
+        )
+        eng = LLMEngine(cfg, device="cuda")
+        p = SamplingParams(max_tokens=12, temperature=0.0, ignore_eos=True)
+        return eng.generate([list(range(40, 140))], p)["offline-0"]
+
+    bf = run("auto")
+    f8 = run("fp8_e4m3")
+    assert len(f8) == 12
+    agree = sum(a == b for a, b in zip(bf, f8))
+    assert agree >= len(bf) // 2, (bf, f8)
