@@ -483,26 +483,31 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         long q_stride, int variant, hipStream_t stream) {
+                         long q_stride, int variant, int kv_fp8,
+                         hipStream_t stream) {
   dim3 grid(num_seqs, KH, num_splits);
   constexpr int NW = 4;
   dim3 block(NW * 64);
+#define PS_DECODE_T(HD, G, BS, LR, KVT)                                      \
+  paged_attn_decode_kernel<HD, G, BS, NW, LR, KVT>                           \
+      <<<grid, block, 0, stream>>>(                                          \
+          (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,               \
+          (const unsigned short*)q, (const KVT*)k_cache,                     \
+          (const KVT*)v_cache, (const int*)block_tables,                     \
+          (const int*)seq_lens, max_blocks, scale, KH, q_stride)
 #define PS_DISPATCH_DECODE(HD, G, BS)                                        \
   do {                                                                       \
-    if (variant == 1)                                                        \
-      paged_attn_decode_kernel<HD, G, BS, NW, 1><<<grid, block, 0,           \
-                                                   stream>>>(                \
-          (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,               \
-          (const unsigned short*)q, (const unsigned short*)k_cache,          \
-          (const unsigned short*)v_cache, (const int*)block_tables,          \
-          (const int*)seq_lens, max_blocks, scale, KH, q_stride);            \
-    else                                                                     \
-      paged_attn_decode_kernel<HD, G, BS, NW, 0><<<grid, block, 0,           \
-                                                   stream>>>(                \
-        (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,                 \
-        (const unsigned short*)q, (const unsigned short*)k_cache,            \
-        (const unsigned short*)v_cache, (const int*)block_tables,            \
-        (const int*)seq_lens, max_blocks, scale, KH, q_stride);              \
+    if (kv_fp8) {                                                            \
+      if (variant == 1)                                                      \
+        PS_DECODE_T(HD, G, BS, 1, unsigned char);                            \
+      else                                                                   \
+        PS_DECODE_T(HD, G, BS, 0, unsigned char);                            \
+    } else {                                                                 \
+      if (variant == 1)                                                      \
+        PS_DECODE_T(HD, G, BS, 1, unsigned short);                           \
+      else                                                                   \
+        PS_DECODE_T(HD, G, BS, 0, unsigned short);                           \
+    }                                                                        \
     if (num_splits > 1)                                                      \
       paged_attn_combine_kernel<HD>                                          \
           <<<dim3(num_seqs * KH * G), 64, 0, stream>>>(                      \
@@ -518,7 +523,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
       default: return -1;
     }
   }
-  if (head_dim == 64 && block_size == 16) {
+  if (head_dim == 64 && block_size == 16 && !kv_fp8) {
     switch (GQ) {
       case 1: PS_DISPATCH_DECODE(64, 1, 16); return 0;
       case 4: PS_DISPATCH_DECODE(64, 4, 16); return 0;
@@ -526,6 +531,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
     }
   }
   return -1;
+#undef PS_DECODE_T
 #undef PS_DISPATCH_DECODE
 }
 
