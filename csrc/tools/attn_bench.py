@@ -44,12 +44,15 @@ def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20, variant=4):
 
 
 def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50,
-                 variant=0):
+                 variant=0, fp8=False):
     bs = 16
     per = ctx // bs
     nb = batch * per + 1
     k = torch.randn(nb, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
     v = torch.randn_like(k)
+    if fp8:
+        k = k.to(torch.float8_e4m3fn)
+        v = v.to(torch.float8_e4m3fn)
     bt = torch.arange(1, batch * per + 1, dtype=torch.int32, device="cuda").reshape(batch, per)
     sl = torch.full((batch,), ctx, dtype=torch.int32, device="cuda")
     q = torch.randn(batch, qh, hd, dtype=torch.bfloat16, device="cuda")
@@ -63,9 +66,10 @@ def bench_decode(batch=64, ctx=1024, qh=32, kh=8, hd=128, iters=50,
         _C.paged_attn_decode(out, q, k, v, bt, sl, scale, 0, variant)
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
-    bytes_kv = batch * ctx * kh * hd * 2 * 2
-    print(f"decode v{variant} b={batch} ctx={ctx}: {dt*1e6:.1f} us  "
-          f"{bytes_kv/dt/1e9:.0f} GB/s")
+    eb = 1 if fp8 else 2
+    bytes_kv = batch * ctx * kh * hd * 2 * eb
+    print(f"decode v{variant}{' fp8' if fp8 else ''} b={batch} ctx={ctx}: "
+          f"{dt*1e6:.1f} us  {bytes_kv/dt/1e9:.0f} GB/s")
 
 
 if __name__ == "__main__":
@@ -77,3 +81,5 @@ if __name__ == "__main__":
             bench_decode(batch=b, variant=v)
         for b in (64, 256):
             bench_decode(batch=b, ctx=2048, variant=v)
+    for b in (64, 256):
+        bench_decode(batch=b, ctx=2048, variant=1, fp8=True)
