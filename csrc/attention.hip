@@ -30,7 +30,10 @@
 // ---------------------------------------------------------------------------
 template <int HEAD_DIM, int GQ, int BLOCK_SIZE, int NWAVES, int LOWREG,
           typename KVT>
-__global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
+// LOWREG: cap registers so 3 workgroups co-reside per SIMD (168-VGPR
+// budget; the fp8 instantiation otherwise lands on 170 -> 176 alloc ->
+// 2 waves and runs 36% slower despite half the KV bytes).
+__global__ __launch_bounds__(NWAVES * 64, LOWREG ? 3 : 1) void paged_attn_decode_kernel(
     unsigned short* __restrict__ out,            // [S, QH, HEAD_DIM]
     float* __restrict__ ws_acc,   // [S, QH, SPLITS, HD] (splits > 1)
     float* __restrict__ ws_ml,    // [S, QH, SPLITS, 2]
@@ -116,15 +119,17 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
       kvec8 kv = *(const kvec8*)(kb + tok * D + sl * 8);
       if constexpr (!LOWREG)
         vv[tt] = *(const kvec8*)(vb + tok * D + sl * 8);
+      float kf[8];
+      KVTr::to_f32x8(kv, kf);  // packed v_cvt_pk_f32_fp8 on fp8 caches
 #pragma unroll
       for (int g = 0; g < GQ; g++) {
         float s = 0.f;
 #pragma unroll
         for (int j = 0; j < 8; j++) {
           if constexpr (LOWREG)
-            s += ps_bf16_to_f32(qp[g][j]) * KVTr::to_f32(kv[j]);
+            s += ps_bf16_to_f32(qp[g][j]) * kf[j];
           else
-            s += qf[g][j] * KVTr::to_f32(kv[j]);
+            s += qf[g][j] * kf[j];
         }
         s = ps_group_sum<LPG>(s) * scale;
         sc[g][tt] = valid ? s : PS_NEG_INF;
@@ -151,12 +156,13 @@ __global__ __launch_bounds__(NWAVES * 64) void paged_attn_decode_kernel(
       }
       l[g] = l[g] * corr + psum;
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        float a = acc[g][j] * corr;
+      for (int j = 0; j < 8; j++) acc[g][j] *= corr;
 #pragma unroll
-        for (int tt = 0; tt < TPB; tt++)
-          a += p[tt] * KVTr::to_f32(vv[tt][j]);
-        acc[g][j] = a;
+      for (int tt = 0; tt < TPB; tt++) {
+        float vf[8];
+        KVTr::to_f32x8(vv[tt], vf);
+#pragma unroll
+        for (int j = 0; j < 8; j++) acc[g][j] += p[tt] * vf[j];
       }
       m[g] = mnew;
     }

@@ -144,18 +144,12 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 #pragma unroll
     for (int h = 0; h < 4; h++) {
       const int slot = ((wave * 4 + h) ^ (tv & 7));
-      if constexpr (sizeof(KVT) == 2) {
-        *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = kstage[h];
-      } else {
-        ps_bf16x8 kb16;
-#pragma unroll
-        for (int j = 0; j < 8; j++) kb16[j] = KVTr::to_bf16(kstage[h][j]);
-        *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = kb16;
-      }
+      *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = KVTr::to_bf16x8(kstage[h]);
+      const ps_bf16x8 vb16 = KVTr::to_bf16x8(vstage[h]);
 #pragma unroll
       for (int j = 0; j < 8; j++) {
         const int d = d0 + h * 8 + j;
-        v_t[d][((lg ^ (d & 7)) << 3) + tl] = KVTr::to_bf16(vstage[h][j]);
+        v_t[d][((lg ^ (d & 7)) << 3) + tl] = vb16[j];
       }
     }
     __syncthreads();

@@ -69,6 +69,47 @@ PS_DEV unsigned char ps_f32_to_fp8(float f) {
   return __hip_fp8_e4m3(f).__x;
 }
 
+typedef __attribute__((ext_vector_type(2))) float ps_f32x2;
+
+// Packed fp8->f32: v_cvt_pk_f32_fp8 converts a byte pair per instruction
+// (half the VALU of scalar v_cvt_f32_fp8 per element).
+PS_DEV void ps_fp8x8_to_f32(ps_fp8x8 v, float* out) {
+  union {
+    ps_fp8x8 v8;
+    int i2[2];
+  } u;
+  u.v8 = v;
+#pragma unroll
+  for (int w = 0; w < 2; w++) {
+    ps_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(u.i2[w], false);
+    ps_f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(u.i2[w], true);
+    out[w * 4 + 0] = lo.x;
+    out[w * 4 + 1] = lo.y;
+    out[w * 4 + 2] = hi.x;
+    out[w * 4 + 3] = hi.y;
+  }
+}
+
+typedef __attribute__((ext_vector_type(2))) __bf16 ps_rawbf16x2;
+
+// Packed fp8x8 -> bf16x8 (for LDS staging in the MFMA prefill kernel):
+// 4x v_cvt_pk_f32_fp8 + 4x v_cvt_pk_bf16_f32 instead of ~8 scalar chains.
+PS_DEV ps_bf16x8 ps_fp8x8_to_bf16x8(ps_fp8x8 v) {
+  float f[8];
+  ps_fp8x8_to_f32(v, f);
+  union {
+    ps_bf16x8 v8;
+    unsigned int i4[4];
+  } out;
+#pragma unroll
+  for (int w = 0; w < 4; w++) {
+    ps_f32x2 p = {f[2 * w], f[2 * w + 1]};
+    ps_rawbf16x2 b = __builtin_convertvector(p, ps_rawbf16x2);
+    out.i4[w] = *(unsigned int*)&b;
+  }
+  return out.v8;
+}
+
 template <typename KVT>
 struct ps_kv_traits;
 
@@ -78,6 +119,11 @@ struct ps_kv_traits<unsigned short> {  // bf16 cache
   static PS_DEV float to_f32(unsigned short u) { return ps_bf16_to_f32(u); }
   static PS_DEV unsigned short from_f32(float f) { return ps_f32_to_bf16(f); }
   static PS_DEV unsigned short to_bf16(unsigned short u) { return u; }
+  static PS_DEV void to_f32x8(ps_bf16x8 v, float* out) {
+#pragma unroll
+    for (int j = 0; j < 8; j++) out[j] = ps_bf16_to_f32(v[j]);
+  }
+  static PS_DEV ps_bf16x8 to_bf16x8(ps_bf16x8 v) { return v; }
 };
 
 template <>
@@ -87,6 +133,12 @@ struct ps_kv_traits<unsigned char> {  // fp8 e4m3 cache
   static PS_DEV unsigned char from_f32(float f) { return ps_f32_to_fp8(f); }
   static PS_DEV unsigned short to_bf16(unsigned char u) {
     return ps_f32_to_bf16(ps_fp8_to_f32(u));
+  }
+  static PS_DEV void to_f32x8(ps_fp8x8 v, float* out) {
+    ps_fp8x8_to_f32(v, out);
+  }
+  static PS_DEV ps_bf16x8 to_bf16x8(ps_fp8x8 v) {
+    return ps_fp8x8_to_bf16x8(v);
   }
 };
 
