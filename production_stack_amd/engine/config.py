@@ -129,6 +129,10 @@ class SchedulerConfig:
 class ParallelConfig:
     tensor_parallel_size: int = 1
     pipeline_parallel_size: int = 1
+    # >1 splits each step's batch into M microbatches and issues them
+    # back-to-back through the PP stages so stage k computes microbatch i
+    # while stage k-1 computes i+1 (in-flight pipelining; see pipeline.py)
+    pp_microbatches: int = 1
     rank: int = 0
     # disaggregated prefill role: None | "prefill" | "decode"
     kv_role: Optional[str] = None

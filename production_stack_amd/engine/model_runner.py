@@ -361,6 +361,9 @@ class ModelRunner:
         fast = self._execute_decode_graph(out, bm)
         if fast is not None:
             return fast
+        n_mb = self.config.parallel.pp_microbatches
+        if self.pipeline is not None and n_mb > 1 and len(out.scheduled) > 1:
+            return self._execute_pp_microbatched(out, bm, n_mb)
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
         if token_t.numel() == 0:
             return {}
@@ -383,6 +386,44 @@ class ModelRunner:
         return {
             seq.request_id: int(tok)
             for seq, tok in zip(sample_seqs, tokens)
+        }
+
+    @torch.no_grad()
+    def _execute_pp_microbatched(
+        self, out: SchedulerOutput, bm: BlockManager, n_mb: int
+    ) -> Dict[str, int]:
+        """Split the step at sequence boundaries into up to n_mb
+        microbatches (token-count balanced, greedy) and pipeline them
+        through the stages in flight (pipeline.drive_many)."""
+        scheduled = out.scheduled
+        n_mb = min(n_mb, len(scheduled))
+        # greedy balance by token count: big (prefill) seqs first
+        order = sorted(range(len(scheduled)),
+                       key=lambda i: -scheduled[i].num_tokens)
+        groups: List[List[int]] = [[] for _ in range(n_mb)]
+        loads = [0] * n_mb
+        for i in order:
+            g = loads.index(min(loads))
+            groups[g].append(i)
+            loads[g] += scheduled[i].num_tokens
+        mbs = []
+        all_seqs: List[Sequence] = []
+        for g in groups:
+            if not g:
+                continue
+            sub = SchedulerOutput(scheduled=[scheduled[i] for i in sorted(g)])
+            token_t, meta, sample_seqs, rows_t = self.prepare(sub, bm)
+            if token_t.numel() == 0:
+                continue
+            mbs.append((token_t, meta, rows_t,
+                        [s.params for s in sample_seqs]))
+            all_seqs.extend(sample_seqs)
+        if not mbs:
+            return {}
+        tokens = self.pipeline.drive_many(mbs)
+        return {
+            seq.request_id: int(tok)
+            for seq, tok in zip(all_seqs, tokens)
         }
 
     def sample(

@@ -7,9 +7,13 @@ metadata is broadcast to all stages, activations (hidden + residual) move
 stage-to-stage with send/recv (RCCL p2p over xGMI on GPU, gloo on CPU), and
 the last stage samples and returns tokens to rank 0.
 
-Synchronous single-microbatch pipeline: correctness-first; the continuous
-batching scheduler keeps every stage busy across steps because each step
-carries the whole running batch.
+With `ParallelConfig.pp_microbatches > 1` each step's batch is split at
+sequence boundaries into M microbatches that are issued back-to-back:
+rank 0 computes and sends microbatch i's stage-0 activations before
+waiting on anything, so stage k runs microbatch i while stage k-1 runs
+i+1 (classic in-flight pipelining — GPU work overlaps because RCCL
+send/recv enqueue on streams without host-blocking on completion).
+Sampled tokens are collected per microbatch after the last send.
 """
 
 from __future__ import annotations
@@ -80,22 +84,38 @@ class PipelineCoordinator:
         params: List[SamplingParams],
     ) -> torch.Tensor:
         """Run one step through the pipeline; returns sampled token ids."""
-        payload = {
-            "op": "step",
-            "meta": _meta_to_payload(meta),
-            "tokens": token_t.cpu(),
-            "sample_rows": sample_rows.cpu(),
-            "params": [
-                (p.greedy, p.temperature, p.top_p, p.top_k) for p in params
-            ],
-        }
-        dist.broadcast_object_list([payload], src=0)
-        act = self.runner.model(token_t, meta, self.runner.kv_caches)
-        dist.send(act.contiguous(), dst=1)
-        sampled = torch.empty(len(params), dtype=torch.long)
-        if params:
+        return self.drive_many([(token_t, meta, sample_rows, params)])
+
+    @torch.no_grad()
+    def drive_many(self, microbatches) -> torch.Tensor:
+        """Issue all microbatches through the pipeline before collecting
+        any sampled tokens; returns token ids concatenated in mb order."""
+        mb_payloads = []
+        for token_t, meta, sample_rows, params in microbatches:
+            mb_payloads.append({
+                "meta": _meta_to_payload(meta),
+                "tokens": token_t.cpu(),
+                "sample_rows": sample_rows.cpu(),
+                "params": [
+                    (p.greedy, p.temperature, p.top_p, p.top_k)
+                    for p in params
+                ],
+            })
+        dist.broadcast_object_list([{"op": "step", "mbs": mb_payloads}],
+                                   src=0)
+        for token_t, meta, _, _ in microbatches:
+            act = self.runner.model(token_t, meta, self.runner.kv_caches)
+            dist.send(act.contiguous(), dst=1)
+        outs = []
+        for _, _, _, params in microbatches:
+            if not params:
+                continue
+            sampled = torch.empty(len(params), dtype=torch.long)
             dist.recv(sampled, src=self.size - 1)
-        return sampled
+            outs.append(sampled)
+        if not outs:
+            return torch.empty(0, dtype=torch.long)
+        return torch.cat(outs)
 
     def stop_workers(self) -> None:
         try:
@@ -114,27 +134,29 @@ class PipelineCoordinator:
             if payload is None or payload.get("op") == "stop":
                 logger.info("pp worker rank %d stopping", self.rank)
                 return
-            meta = _payload_to_meta(payload["meta"], self.device)
-            T = payload["tokens"].shape[0]
-            act = torch.empty(
-                (2, T, self.hidden), dtype=torch.bfloat16, device=self.device
-            )
-            dist.recv(act, src=self.rank - 1)
-            hidden, residual = act[0], act[1]
-            out = model(hidden, meta, self.runner.kv_caches, residual)
-            if not model.is_last:
-                dist.send(out.contiguous(), dst=self.rank + 1)
-                continue
-            params_raw = payload["params"]
-            if not params_raw:
-                continue
-            rows = payload["sample_rows"].to(self.device)
-            logits = model.compute_logits(out[rows])
-            params = [
-                SamplingParams(
-                    temperature=t, top_p=tp, top_k=tk, max_tokens=1
+            for mb in payload["mbs"]:
+                meta = _payload_to_meta(mb["meta"], self.device)
+                T = mb["tokens"].shape[0]
+                act = torch.empty(
+                    (2, T, self.hidden), dtype=torch.bfloat16,
+                    device=self.device
                 )
-                for (_, t, tp, tk) in params_raw
-            ]
-            sampled = self.runner.sample_params(logits, params)
-            dist.send(sampled.cpu(), dst=0)
+                dist.recv(act, src=self.rank - 1)
+                hidden, residual = act[0], act[1]
+                out = model(hidden, meta, self.runner.kv_caches, residual)
+                if not model.is_last:
+                    dist.send(out.contiguous(), dst=self.rank + 1)
+                    continue
+                params_raw = mb["params"]
+                if not params_raw:
+                    continue
+                rows = mb["sample_rows"].to(self.device)
+                logits = model.compute_logits(out[rows])
+                params = [
+                    SamplingParams(
+                        temperature=t, top_p=tp, top_k=tk, max_tokens=1
+                    )
+                    for (_, t, tp, tk) in params_raw
+                ]
+                sampled = self.runner.sample_params(logits, params)
+                dist.send(sampled.cpu(), dst=0)
