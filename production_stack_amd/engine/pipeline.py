@@ -103,9 +103,17 @@ class PipelineCoordinator:
             })
         dist.broadcast_object_list([{"op": "step", "mbs": mb_payloads}],
                                    src=0)
+        # isend: a blocking send of mb i+1 would deadlock against the last
+        # stage blocking on its sampled-token send for mb i (gloo
+        # rendezvous); isend also lets RCCL enqueue all stage-0 work so the
+        # stages genuinely overlap.
+        works, acts = [], []
         for token_t, meta, _, _ in microbatches:
-            act = self.runner.model(token_t, meta, self.runner.kv_caches)
-            dist.send(act.contiguous(), dst=1)
+            act = self.runner.model(
+                token_t, meta, self.runner.kv_caches
+            ).contiguous()
+            works.append(dist.isend(act, dst=1))
+            acts.append(act)  # keep alive until matched
         outs = []
         for _, _, _, params in microbatches:
             if not params:
@@ -113,6 +121,9 @@ class PipelineCoordinator:
             sampled = torch.empty(len(params), dtype=torch.long)
             dist.recv(sampled, src=self.size - 1)
             outs.append(sampled)
+        for w in works:
+            w.wait()
+        del acts
         if not outs:
             return torch.empty(0, dtype=torch.long)
         return torch.cat(outs)
