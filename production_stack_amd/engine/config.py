@@ -108,6 +108,9 @@ class CacheConfig:
     # surface: reference LMCACHE_MAX_LOCAL_CPU_SIZE
     # (deployment-vllm-multi.yaml:336-343).
     cpu_offload_gb: float = 0.0
+    # remote cacheserver data plane ("host:port"), shared across instances
+    # (reference cacheserverSpec / LMCACHE_REMOTE_URL)
+    remote_kv_url: Optional[str] = None
     # "bf16" (raw) or "int8" (CacheGen-style row-quantized serde, halves
     # host-pool bytes; reference LMCACHE remote serde surface)
     offload_dtype: str = "bf16"

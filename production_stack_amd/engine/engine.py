@@ -64,15 +64,16 @@ class LLMEngine:
         self.scheduler = Scheduler(
             config.scheduler, self.block_manager, config.max_model_len
         )
-        if config.cache.cpu_offload_gb > 0:
+        if config.cache.cpu_offload_gb > 0 or config.cache.remote_kv_url:
             from production_stack_amd.kvpool.offload import HostKVPool
 
             self.host_pool = HostKVPool(
                 self.runner.kv_caches,
                 config.cache.block_size,
-                config.cache.cpu_offload_gb,
+                max(config.cache.cpu_offload_gb, 0.25),
                 self.device,
                 offload_dtype=config.cache.offload_dtype,
+                remote_url=config.cache.remote_kv_url,
             )
             self.block_manager.offload_pool = self.host_pool
         else:

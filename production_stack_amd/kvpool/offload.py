@@ -16,6 +16,8 @@ ordering is enforced with events, never a device-wide sync.
 from __future__ import annotations
 
 import logging
+import queue
+import threading
 from collections import OrderedDict, deque
 from typing import Deque, Dict, List, Optional, Tuple
 
@@ -32,6 +34,7 @@ class HostKVPool:
         capacity_gb: float,
         device: torch.device,
         offload_dtype: str = "bf16",
+        remote_url: Optional[str] = None,
     ) -> None:
         self.kv_caches = kv_caches
         self.device = device
@@ -77,10 +80,27 @@ class HostKVPool:
         self.offloaded = 0
         self.restored = 0
         self.evicted = 0
+        # ---- optional remote tier (kvpool.cacheserver data plane) ----
+        self.remote = None
+        self._push_q: Optional[queue.Queue] = None
+        self._push_thread: Optional[threading.Thread] = None
+        self.remote_pushed = 0
+        self.remote_restored = 0
+        if remote_url:
+            from production_stack_amd.kvpool.cacheserver import RemoteKVClient
+
+            self.remote = RemoteKVClient(remote_url)
+            self._push_q = queue.Queue(maxsize=1024)
+            self._push_thread = threading.Thread(
+                target=self._push_loop, daemon=True
+            )
+            self._push_thread.start()
 
     # ------------------------------------------------------------------
     def has(self, h: int) -> bool:
-        return h in self.slot_of
+        if h in self.slot_of:
+            return True
+        return self.remote is not None and self.remote.exists(h)
 
     def _take_slot(self) -> int:
         if self.free_slots:
@@ -143,10 +163,21 @@ class HostKVPool:
                     dst[li, 1].copy_(vc[block_id].flatten())
         self.slot_of[h] = slot
         self.offloaded += 1
+        if self.remote is not None:
+            ev = None
+            if self.stream is not None:
+                ev = torch.cuda.Event()
+                ev.record(self.stream)
+            try:
+                self._push_q.put_nowait((h, slot, ev))
+            except queue.Full:
+                pass  # lossy best-effort: the local tier still has it
 
     def restore(self, h: int, block_id: int) -> bool:
         """Async H2D into a freshly allocated GPU block."""
         slot = self.slot_of.get(h)
+        if slot is None and self.remote is not None:
+            slot = self._fetch_remote(h)
         if slot is None:
             return False
         self.slot_of.move_to_end(h)
@@ -199,6 +230,58 @@ class HostKVPool:
         self.restored += 1
         return True
 
+    # ---- remote tier ---------------------------------------------------
+    def _record_bytes(self, slot: int):
+        data = self.store[slot].contiguous().view(torch.uint8)
+        scales = None
+        if self.quantized:
+            scales = (
+                self.scale_store[slot].contiguous().view(torch.uint8)
+                .numpy().tobytes()
+            )
+        return data.numpy().tobytes(), scales
+
+    def _push_loop(self) -> None:
+        while True:
+            item = self._push_q.get()
+            if item is None:
+                return
+            h, slot, ev = item
+            try:
+                if ev is not None:
+                    ev.synchronize()  # D2H of this record has landed
+                if self.slot_of.get(h) != slot:
+                    continue  # slot was LRU-recycled before we shipped it
+                data, scales = self._record_bytes(slot)
+                if self.remote.put(h, data, scales):
+                    self.remote_pushed += 1
+            except Exception:  # network best-effort; never kill the engine
+                logger.exception("remote KV push failed")
+
+    def _fetch_remote(self, h: int) -> Optional[int]:
+        """Blocking fetch into a local slot; returns the slot or None."""
+        rec = self.remote.get(h)
+        if rec is None:
+            return None
+        data, scales = rec
+        slot = self._take_slot()
+        flat = torch.frombuffer(bytearray(data), dtype=torch.uint8)
+        dst = self.store[slot].view(torch.uint8)
+        dst.copy_(flat.view(dst.shape))
+        if self.quantized and scales is not None:
+            sflat = torch.frombuffer(bytearray(scales), dtype=torch.uint8)
+            self.scale_store[slot].view(torch.uint8).copy_(sflat)
+        self.slot_of[h] = slot
+        self.remote_restored += 1
+        return slot
+
+    def stop(self) -> None:
+        if self._push_q is not None:
+            self._push_q.put(None)
+            self._push_thread.join(timeout=5)
+        if self.remote is not None:
+            self.remote.close()
+
     def make_compute_wait(self) -> None:
         """Compute stream must not read restored blocks before H2D lands."""
         if self.stream is None:
@@ -215,4 +298,6 @@ class HostKVPool:
             "cpu_offloaded_total": float(self.offloaded),
             "cpu_restored_total": float(self.restored),
             "cpu_evicted_total": float(self.evicted),
+            "remote_pushed_total": float(self.remote_pushed),
+            "remote_restored_total": float(self.remote_restored),
         }

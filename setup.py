@@ -52,6 +52,7 @@ setup(
             "vllm-router=production_stack_amd.router.app:main",
             "ps-engine=production_stack_amd.engine.server:main",
             "ps-kv-controller=production_stack_amd.kvpool.controller:main",
+            "ps-cacheserver=production_stack_amd.kvpool.cacheserver:main",
             "ps-endpoint-picker="
             "production_stack_amd.gateway.picker_service:main",
         ]

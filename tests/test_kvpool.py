@@ -257,3 +257,119 @@ def test_host_offload_int8_roundtrip_cpu():
     # int8 KV is lossy: require strong (not exact) agreement
     agree = sum(a == b for a, b in zip(out1, out2))
     assert agree >= len(out1) - 1, (out1, out2)
+
+
+def _start_cacheserver():
+    """Run a CacheServer on an ephemeral port in a daemon thread."""
+    import threading
+
+    from production_stack_amd.kvpool.cacheserver import CacheServer
+
+    srv = CacheServer(host="127.0.0.1", port=0, capacity_gb=0.1)
+    loop = asyncio.new_event_loop()
+    started = threading.Event()
+
+    def run():
+        asyncio.set_event_loop(loop)
+        loop.run_until_complete(srv.start())
+        started.set()
+        loop.run_forever()
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    started.wait(timeout=10)
+    return srv, loop
+
+
+def test_cacheserver_store_and_client_roundtrip():
+    from production_stack_amd.kvpool.cacheserver import RemoteKVClient
+
+    srv, loop = _start_cacheserver()
+    try:
+        c = RemoteKVClient(f"127.0.0.1:{srv.port}")
+        assert not c.exists(42)
+        assert c.put(42, b"kv-bytes", None)
+        assert c.exists(42)
+        assert c.get(42) == (b"kv-bytes", None)
+        assert c.get(7) is None
+        assert c.put(43, b"x" * 100, b"scales")
+        data, scales = c.get(43)
+        assert data == b"x" * 100 and scales == b"scales"
+        st = c.stats()
+        assert st["records"] == 2 and st["hits"] == 2
+        c.close()
+    finally:
+        loop.call_soon_threadsafe(loop.stop)
+
+
+def test_cacheserver_lru_eviction_by_bytes():
+    from production_stack_amd.kvpool.cacheserver import CacheStore
+
+    st = CacheStore(capacity_gb=1e-6)  # ~1073 bytes
+    st.put(1, b"a" * 400, None)
+    st.put(2, b"b" * 400, None)
+    st.put(3, b"c" * 400, None)  # evicts key 1
+    assert st.get(1) is None
+    assert st.get(2) is not None
+    assert st.evictions == 1
+
+
+def test_remote_kv_shared_across_engines():
+    """Engine A prefills and pushes KV to the cacheserver; a *fresh*
+    engine B with the same weights restores A's blocks from the remote
+    tier instead of recomputing (cross-instance KV reuse — the
+    reference's cacheserver capability)."""
+    import torch
+
+    from production_stack_amd.engine.config import (
+        CacheConfig,
+        EngineConfig,
+        SchedulerConfig,
+    )
+    from production_stack_amd.engine.engine import LLMEngine
+    from production_stack_amd.engine.sampling import SamplingParams
+
+    srv, loop = _start_cacheserver()
+    try:
+        def mk(seed_engine=None):
+            cfg = EngineConfig(
+                model="tiny-llama",
+                max_model_len=256,
+                seed=11,
+                cache=CacheConfig(
+                    num_gpu_blocks=32, block_size=16, cpu_offload_gb=0.01,
+                    remote_kv_url=f"127.0.0.1:{srv.port}",
+                ),
+                scheduler=SchedulerConfig(
+                    max_num_seqs=4, max_num_batched_tokens=256
+                ),
+            )
+            eng = LLMEngine(cfg, device="cpu")
+            if seed_engine is not None:  # identical weights
+                eng.runner.model.load_state_dict(
+                    seed_engine.runner.model.state_dict()
+                )
+            return eng
+
+        p = SamplingParams(max_tokens=4, temperature=0.0, ignore_eos=True)
+        prompt = list(range(20, 100))  # 5 blocks
+        a = mk()
+        out_a = a.generate([prompt], p)["offline-0"]
+        import time
+
+        deadline = time.time() + 10
+        while a.host_pool.remote_pushed == 0 and time.time() < deadline:
+            time.sleep(0.05)
+        assert a.host_pool.remote_pushed > 0
+
+        b = mk(seed_engine=a)
+        out_b = b.generate([prompt], p)["offline-0"]
+        assert b.host_pool.remote_restored > 0, "B must hit the remote tier"
+        assert b.block_manager.prefix_hits > 0
+        assert out_a == out_b
+        m = b.engine_metrics()
+        assert m["remote_restored_total"] > 0
+        a.host_pool.stop()
+        b.host_pool.stop()
+    finally:
+        loop.call_soon_threadsafe(loop.stop)
