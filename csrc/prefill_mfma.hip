@@ -44,7 +44,8 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     const KVT* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ tile_info,     // [NT, 4]
-    int max_blocks, float scale, int KH, int GQ, long q_stride) {
+    int max_blocks, float scale, int KH, int GQ, long q_stride,
+    int QH, int n_work) {
   using KVTr = ps_kv_traits<KVT>;
   using kvec8 = typename KVTr::vec8;
   constexpr int D = HEAD_DIM;  // 128
@@ -53,10 +54,19 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
   constexpr int NCT = PS_CHUNK / 16;  // QK col-tiles per chunk (4)
   constexpr int NKC = PS_CHUNK / 32;  // PV k-chunks per chunk (2)
 
-  const int tile = blockIdx.x;
-  const int qh = blockIdx.y;
-  const int kvh = qh / GQ;
-  const int QH = gridDim.y;
+  // T1 XCD swizzle: works ordered (kvh, gq, tile) and sliced into 8
+  // contiguous chunks, one per XCD (dispatch round-robins linear id mod
+  // NXCD) -> the GQ=4 heads of one kv-head plus neighboring tiles of the
+  // same sequence hit the same XCD's L2 with the same KV pages.
+  const int W = n_work;  // num_tiles * QH
+  const int cpx = (W + 7) >> 3;
+  const int w = (blockIdx.x & 7) * cpx + (blockIdx.x >> 3);
+  if (w >= W) return;
+  const int n_tiles = W / QH;
+  const int tile = w % n_tiles;
+  const int gq = (w / n_tiles) % GQ;
+  const int kvh = w / (n_tiles * GQ);
+  const int qh = kvh * GQ + gq;
   const int seq_row = tile_info[tile * 4 + 0];
   const int q_tok0 = tile_info[tile * 4 + 1];
   const int q_pos0 = tile_info[tile * 4 + 2];
@@ -201,12 +211,24 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
       }
 #pragma unroll
       for (int r = 0; r < 4; r++) m_new[r] = ps_group_max<16>(m_new[r]);
-      float row_corr[4];
+      // T13 defer-max: if no row's max grew by more than THR, keep the
+      // old m (P is then bounded by exp(THR), fine in f32 accum) and skip
+      // the whole O rescale pass. Wave-uniform so the branch is free.
+      constexpr float PS_RESCALE_THR = 8.f;
+      bool grew = false;
 #pragma unroll
-      for (int r = 0; r < 4; r++) {
-        row_corr[r] = __expf(m_run[r] - m_new[r]);
-        l_run[r] *= row_corr[r];
-        m_run[r] = m_new[r];
+      for (int r = 0; r < 4; r++)
+        grew |= (m_new[r] > m_run[r] + PS_RESCALE_THR) ||
+                (m_run[r] == PS_NEG_INF && m_new[r] > PS_NEG_INF);
+      if (__any(grew)) {
+#pragma unroll
+        for (int r = 0; r < 4; r++) {
+          const float corr = __expf(m_run[r] - m_new[r]);
+          l_run[r] *= corr;
+#pragma unroll
+          for (int s = 0; s < 8; s++) o_acc[s][r] *= corr;
+          m_run[r] = m_new[r];
+        }
       }
       float psum[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -214,7 +236,7 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 #pragma unroll
         for (int r = 0; r < 4; r++) {
           const float p = s_frag[ct][r] > PS_NEG_INF
-                              ? __expf(s_frag[ct][r] - m_new[r])
+                              ? __expf(s_frag[ct][r] - m_run[r])
                               : 0.f;
           s_frag[ct][r] = p;
           psum[r] += p;
@@ -225,11 +247,6 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
         float rs = ps_group_sum<16>(psum[r]);
         l_run[r] += rs;
       }
-      // rescale O
-#pragma unroll
-      for (int s = 0; s < 8; s++)
-#pragma unroll
-        for (int r = 0; r < 4; r++) o_acc[s][r] *= row_corr[r];
       // ---- write P to per-wave LDS (C-layout -> A-layout transpose) ----
 #pragma unroll
       for (int ct = 0; ct < NCT; ct++)
@@ -281,12 +298,14 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                int KH, int GQ, int head_dim, long q_stride,
                                int variant, int kv_fp8, hipStream_t stream) {
   if (head_dim != 128) return -1;
-  dim3 grid(num_tiles, num_q_heads);
+  const int n_work = num_tiles * num_q_heads;
+  dim3 grid(((n_work + 7) / 8) * 8);
 #define PS_PREFILL_T(WPS, KVT)                                              \
   paged_attn_prefill_mfma_kernel<128, WPS, KVT><<<grid, 256, 0, stream>>>(  \
       (unsigned short*)out, (const unsigned short*)q,                       \
       (const KVT*)k_cache, (const KVT*)v_cache, (const int*)block_tables,   \
-      (const int*)tile_info, max_blocks, scale, KH, GQ, q_stride)
+      (const int*)tile_info, max_blocks, scale, KH, GQ, q_stride,           \
+      num_q_heads, n_work)
   if (kv_fp8) {
     if (variant == 3) PS_PREFILL_T(3, unsigned char);
     else PS_PREFILL_T(4, unsigned char);
