@@ -55,6 +55,10 @@ void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
                    int hd, hipStream_t stream);
 int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
                    int K, long x_stride, hipStream_t stream);
+int ps_lora_bgmv(void* out, const void* x, const void* A, const void* B,
+                 const void* scale, const void* idx, int T, int IN, int W,
+                 int R, long out_stride, long x_stride, int col_off,
+                 hipStream_t stream);
 }
 
 namespace {
@@ -289,6 +293,33 @@ void skinny_gemm(at::Tensor out_f32, at::Tensor x, at::Tensor w) {
               " K=", K);
 }
 
+void lora_bgmv(at::Tensor out, at::Tensor x, at::Tensor A, at::Tensor B,
+               at::Tensor scale, at::Tensor idx, long col_off) {
+  CHECK_GPU_BF16(out);
+  CHECK_GPU_BF16(x);
+  CHECK_GPU_BF16(A);
+  CHECK_GPU_BF16(B);
+  CHECK_GPU_DTYPE(scale, at::kFloat);
+  CHECK_GPU_DTYPE(idx, at::kInt);
+  TORCH_CHECK(out.dim() == 2 && out.stride(1) == 1, "out 2D row-major");
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1, "x 2D row-major");
+  TORCH_CHECK(A.dim() == 3 && A.is_contiguous(), "A [S,R,IN] contiguous");
+  TORCH_CHECK(B.dim() == 3 && B.is_contiguous(), "B [S,W,R] contiguous");
+  const int T = (int)x.size(0);
+  const int IN = (int)x.size(1);
+  const int R = (int)A.size(1);
+  const int W = (int)B.size(1);
+  TORCH_CHECK(A.size(2) == IN, "A IN mismatch");
+  TORCH_CHECK(B.size(2) == R, "B R mismatch");
+  TORCH_CHECK(out.size(0) == T && idx.numel() == T, "T mismatch");
+  TORCH_CHECK(col_off + W <= out.size(1), "col range out of bounds");
+  int rc = ps_lora_bgmv(out.data_ptr(), x.data_ptr(), A.data_ptr(),
+                        B.data_ptr(), scale.data_ptr(), idx.data_ptr(), T,
+                        IN, W, R, out.stride(0), x.stride(0), (int)col_off,
+                        current_stream());
+  TORCH_CHECK(rc == 0, "lora_bgmv unsupported (R>64?) R=", R);
+}
+
 void kv_quant(at::Tensor out, at::Tensor scales, at::Tensor in) {
   CHECK_GPU_DTYPE(out, at::kChar);
   CHECK_GPU_DTYPE(scales, at::kFloat);
@@ -335,6 +366,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("num_splits") = 0, pybind11::arg("variant") = 0);
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged attention, chunked prefill (bf16 KV)");
+  m.def("lora_bgmv", &lora_bgmv,
+        "batched-grouped LoRA matvec (per-token adapter slots)");
   m.def("paged_attn_prefill_mfma", &paged_attn_prefill_mfma,
         "Paged attention, chunked prefill via MFMA tiles (head_dim 128)",
         pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k_cache"),

@@ -97,10 +97,19 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
   const int q_row_clamped = min(my_local_row, n_rows - 1);
   const unsigned short* qrow =
       q + (long)(q_tok0 + q_row_clamped) * q_stride + (long)qh * D;
+  // scale*log2(e) folded into q so the softmax runs in the exp2 domain:
+  // saves two v_mul per score element (scale + ln->log2 conversion)
+  const float qmul = scale * 1.44269504f;
   ps_mbf16x8 q_frag[NK];
 #pragma unroll
-  for (int kk = 0; kk < NK; kk++)
-    q_frag[kk] = ps_as_mbf16(*(const ps_bf16x8*)(qrow + kk * 32 + g * 8));
+  for (int kk = 0; kk < NK; kk++) {
+    ps_bf16x8 qv = *(const ps_bf16x8*)(qrow + kk * 32 + g * 8);
+    ps_bf16x8 qs;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      qs[j] = ps_f32_to_bf16(ps_bf16_to_f32(qv[j]) * qmul);
+    q_frag[kk] = ps_as_mbf16(qs);
+  }
 
   // online-softmax state: rows wave*16 + g*4 + r
   float m_run[4], l_run[4];
@@ -115,18 +124,24 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 
   const int wave_pos_max = q_pos0 + min(wave * 16 + 15, n_rows - 1);
 
-  // V^T staging pipeline: this thread stages token tv, dims
-  // [wave*32, wave*32+32) (4 x ushort8); loads for chunk c+1 issue during
-  // chunk c's compute phase.
+  // Staging assignments:
+  //   K: thread stages token tv, dims [wave*32, +32) as 4 x b128.
+  //   V: thread (octet vo = tv>>3, slice vds = tv&7) stages tokens
+  //      [vo*8, +8) x dims [d0 + vds*4, +4): an octet never crosses a
+  //      16-token page, and the register-transposed write is 4 x b128
+  //      rows of v_t instead of 32 scalar b16 stores (the phase probe
+  //      put the scalar staging at 24% of kernel time).
+  // Loads for chunk c+1 issue during chunk c's compute phase (T14).
   const int tv = tid & 63;
   const int d0 = wave * 32;
-  const int lg = tv >> 3;  // logical 8-token group within the chunk
-  const int tl = tv & 7;
-  auto v_row_ptr = [&](int chunk) {
-    const int tok = chunk * PS_CHUNK + tv;
+  const int vo = tv >> 3;
+  const int vd0 = d0 + (tv & 7) * 4;
+  using kvec4 = typename KVTr::vec4;
+  auto v_oct_ptr = [&](int chunk) {
+    const int tok = chunk * PS_CHUNK + vo * 8;
     const int pg_idx = min(tok / BS, n_pages - 1);
     const long pg = bt[pg_idx];
-    return v_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
+    return v_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D + vd0;
   };
   auto k_row_ptr = [&](int chunk) {
     const int tok = chunk * PS_CHUNK + tv;
@@ -134,16 +149,16 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     const long pg = bt[pg_idx];
     return k_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D;
   };
-  kvec8 vstage[4];
+  kvec4 vstage[8];
   kvec8 kstage[4];
   {
-    const KVT* vrow = v_row_ptr(0);
+    const KVT* vrow = v_oct_ptr(0);
     const KVT* krow = k_row_ptr(0);
 #pragma unroll
-    for (int h = 0; h < 4; h++) {
-      vstage[h] = *(const kvec8*)(vrow + d0 + h * 8);
+    for (int i = 0; i < 8; i++) vstage[i] = *(const kvec4*)(vrow + i * D);
+#pragma unroll
+    for (int h = 0; h < 4; h++)
       kstage[h] = *(const kvec8*)(krow + d0 + h * 8);
-    }
   }
 
   for (int chunk = 0; chunk < n_chunks; chunk++) {
@@ -155,22 +170,25 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     for (int h = 0; h < 4; h++) {
       const int slot = ((wave * 4 + h) ^ (tv & 7));
       *(ps_bf16x8*)(&k_lds[tv][slot * 8]) = KVTr::to_bf16x8(kstage[h]);
-      const ps_bf16x8 vb16 = KVTr::to_bf16x8(vstage[h]);
+    }
 #pragma unroll
-      for (int j = 0; j < 8; j++) {
-        const int d = d0 + h * 8 + j;
-        v_t[d][((lg ^ (d & 7)) << 3) + tl] = vb16[j];
-      }
+    for (int dd = 0; dd < 4; dd++) {
+      const int d = vd0 + dd;
+      ps_bf16x8 row;
+#pragma unroll
+      for (int i = 0; i < 8; i++) row[i] = KVTr::to_bf16(vstage[i][dd]);
+      *(ps_bf16x8*)(&v_t[d][(vo ^ (d & 7)) << 3]) = row;
     }
     __syncthreads();
     if (chunk + 1 < n_chunks) {
-      const KVT* vrow = v_row_ptr(chunk + 1);
+      const KVT* vrow = v_oct_ptr(chunk + 1);
       const KVT* krow = k_row_ptr(chunk + 1);
 #pragma unroll
-      for (int h = 0; h < 4; h++) {
-        vstage[h] = *(const kvec8*)(vrow + d0 + h * 8);
+      for (int i = 0; i < 8; i++)
+        vstage[i] = *(const kvec4*)(vrow + i * D);
+#pragma unroll
+      for (int h = 0; h < 4; h++)
         kstage[h] = *(const kvec8*)(krow + d0 + h * 8);
-      }
     }
 
     const bool wave_active = (wave * 16 < n_rows) && (tok0 <= wave_pos_max);
@@ -193,20 +211,32 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
       // ---- mask + online softmax (one rescale per chunk) ----
+      // interior chunks (entirely below every row's diagonal, all rows
+      // real) skip the per-element mask compare
+      const bool full_chunk = (tok0 + PS_CHUNK - 1 <= q_pos0 + wave * 16) &&
+                              (wave * 16 + 15 < n_rows);
       float m_new[4];
 #pragma unroll
       for (int r = 0; r < 4; r++) m_new[r] = m_run[r];
+      if (full_chunk) {
 #pragma unroll
-      for (int ct = 0; ct < NCT; ct++) {
-        const int kv_pos = tok0 + ct * 16 + rc;
+        for (int ct = 0; ct < NCT; ct++)
 #pragma unroll
-        for (int r = 0; r < 4; r++) {
-          const int lrow = wave * 16 + g * 4 + r;
-          const int q_pos = q_pos0 + lrow;
-          const bool valid = (lrow < n_rows) && (kv_pos <= q_pos);
-          const float sv = valid ? s_frag[ct][r] * scale : PS_NEG_INF;
-          s_frag[ct][r] = sv;
-          m_new[r] = fmaxf(m_new[r], sv);
+          for (int r = 0; r < 4; r++)
+            m_new[r] = fmaxf(m_new[r], s_frag[ct][r]);
+      } else {
+#pragma unroll
+        for (int ct = 0; ct < NCT; ct++) {
+          const int kv_pos = tok0 + ct * 16 + rc;
+#pragma unroll
+          for (int r = 0; r < 4; r++) {
+            const int lrow = wave * 16 + g * 4 + r;
+            const int q_pos = q_pos0 + lrow;
+            const bool valid = (lrow < n_rows) && (kv_pos <= q_pos);
+            const float sv = valid ? s_frag[ct][r] : PS_NEG_INF;
+            s_frag[ct][r] = sv;
+            m_new[r] = fmaxf(m_new[r], sv);
+          }
         }
       }
 #pragma unroll
@@ -214,7 +244,7 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
       // T13 defer-max: if no row's max grew by more than THR, keep the
       // old m (P is then bounded by exp(THR), fine in f32 accum) and skip
       // the whole O rescale pass. Wave-uniform so the branch is free.
-      constexpr float PS_RESCALE_THR = 8.f;
+      constexpr float PS_RESCALE_THR = 11.54f;  // 8 nats in log2 units
       bool grew = false;
 #pragma unroll
       for (int r = 0; r < 4; r++)
@@ -223,7 +253,7 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
       if (__any(grew)) {
 #pragma unroll
         for (int r = 0; r < 4; r++) {
-          const float corr = __expf(m_run[r] - m_new[r]);
+          const float corr = exp2f(m_run[r] - m_new[r]);
           l_run[r] *= corr;
 #pragma unroll
           for (int s = 0; s < 8; s++) o_acc[s][r] *= corr;
@@ -236,7 +266,7 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 #pragma unroll
         for (int r = 0; r < 4; r++) {
           const float p = s_frag[ct][r] > PS_NEG_INF
-                              ? __expf(s_frag[ct][r] - m_run[r])
+                              ? exp2f(s_frag[ct][r] - m_run[r])
                               : 0.f;
           s_frag[ct][r] = p;
           psum[r] += p;

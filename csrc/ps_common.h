@@ -113,9 +113,12 @@ PS_DEV ps_bf16x8 ps_fp8x8_to_bf16x8(ps_fp8x8 v) {
 template <typename KVT>
 struct ps_kv_traits;
 
+typedef __attribute__((ext_vector_type(4))) unsigned char ps_fp8x4;
+
 template <>
 struct ps_kv_traits<unsigned short> {  // bf16 cache
   using vec8 = ps_bf16x8;
+  using vec4 = ps_bf16x4;
   static PS_DEV float to_f32(unsigned short u) { return ps_bf16_to_f32(u); }
   static PS_DEV unsigned short from_f32(float f) { return ps_f32_to_bf16(f); }
   static PS_DEV unsigned short to_bf16(unsigned short u) { return u; }
@@ -129,6 +132,7 @@ struct ps_kv_traits<unsigned short> {  // bf16 cache
 template <>
 struct ps_kv_traits<unsigned char> {  // fp8 e4m3 cache
   using vec8 = ps_fp8x8;
+  using vec4 = ps_fp8x4;
   static PS_DEV float to_f32(unsigned char u) { return ps_fp8_to_f32(u); }
   static PS_DEV unsigned char from_f32(float f) { return ps_f32_to_fp8(f); }
   static PS_DEV unsigned short to_bf16(unsigned char u) {

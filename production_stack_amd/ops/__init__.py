@@ -217,3 +217,21 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     )
     _C.skinny_gemm(out, x, w)
     return out.to(x.dtype)
+
+
+def lora_bgmv(
+    out: torch.Tensor,
+    x: torch.Tensor,
+    A: torch.Tensor,
+    B: torch.Tensor,
+    scale: torch.Tensor,
+    idx: torch.Tensor,
+    col_off: int = 0,
+) -> None:
+    """out[t, col_off:+W] += scale[idx[t]] * B[idx[t]] @ (A[idx[t]] @ x[t]);
+    idx[t] < 0 skips the row. hipGraph-safe (static shapes, no host sync)."""
+    if out.is_cuda:
+        _require_ext()
+        _C.lora_bgmv(out, x, A, B, scale, idx, col_off)
+        return
+    reference.lora_bgmv(out, x, A, B, scale, idx, col_off)

@@ -176,3 +176,17 @@ def kv_dequant(q: torch.Tensor, scales: torch.Tensor,
     hd = q.shape[-1]
     rows = q.reshape(-1, hd).float() * scales[:, None]
     return rows.to(dtype).view(q.shape)
+
+
+def lora_bgmv(out, x, A, B, scale, idx, col_off):
+    """Reference BGMV: out[t, col_off:+W] += scale[s] * B[s] @ (A[s] @ x[t])
+    for s = idx[t] >= 0."""
+    import torch
+
+    W = B.shape[1]
+    for t in range(x.shape[0]):
+        s = int(idx[t])
+        if s < 0:
+            continue
+        y = (A[s].float() @ x[t].float()) * float(scale[s])
+        out[t, col_off : col_off + W] += (B[s].float() @ y).to(out.dtype)

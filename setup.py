@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "csrc/attention.hip",
         "csrc/prefill_mfma.hip",
         "csrc/skinny_gemm.hip",
+        "csrc/lora_bgmv.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
