@@ -287,6 +287,10 @@ void skinny_gemm(at::Tensor out_f32, at::Tensor x, at::Tensor w) {
   const int K = (int)x.size(1);
   const int N = (int)w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
+  // split-K atomicAdd accumulation needs a zeroed output; memset on the
+  // stream is ~100x cheaper than a torch elementwise fill kernel
+  hipMemsetAsync(out_f32.data_ptr(), 0, (size_t)M * N * sizeof(float),
+                 current_stream());
   int rc = ps_skinny_gemm(out_f32.data_ptr(), x.data_ptr(), w.data_ptr(), M,
                           N, K, x.stride(0), current_stream());
   TORCH_CHECK(rc == 0, "unsupported skinny gemm shape M=", M, " N=", N,

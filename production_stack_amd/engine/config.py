@@ -150,6 +150,11 @@ class EngineConfig:
     weights_path: Optional[str] = None  # safetensors dir (optional)
     tokenizer: str = "synthetic"  # or a path to a tokenizer.json dir
     enforce_eager: bool = False  # True disables hipGraph decode capture
+    # graph-safe batched LoRA (BGMV slots); when False, adapters still run
+    # through the eager grouped path
+    enable_lora: bool = False
+    max_loras: int = 4
+    max_lora_rank: int = 16
     cache: CacheConfig = field(default_factory=CacheConfig)
     scheduler: SchedulerConfig = field(default_factory=SchedulerConfig)
     parallel: ParallelConfig = field(default_factory=ParallelConfig)

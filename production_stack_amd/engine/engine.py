@@ -223,12 +223,17 @@ class LLMEngine:
     def load_lora(self, name: str, path: str) -> None:
         from production_stack_amd.engine.lora import LoRAAdapter
 
-        self.lora_adapters[name] = LoRAAdapter.load(
-            name, path, self.device
+        ad = LoRAAdapter.load(
+            name, path, self.device, torch.bfloat16
         )
+        self.lora_adapters[name] = ad
+        if self.runner.lora_slots is not None:
+            self.runner.lora_slots.register(ad)
 
     def unload_lora(self, name: str) -> None:
         self.lora_adapters.pop(name, None)
+        if self.runner.lora_slots is not None:
+            self.runner.lora_slots.unregister(name)
 
     # ---- tensor parallelism (HTTP-served mode) -------------------------
     @property

@@ -53,6 +53,9 @@ class DecodeGraphRunner:
         self.block_tables = torch.zeros(
             (B, max_blocks_per_seq), dtype=torch.int32, device=dev
         )
+        # per-row LoRA slot (-1 = none); captured into the graphs when the
+        # model carries BGMV slot stacks
+        self.lora_idx = torch.full((B,), -1, dtype=torch.int32, device=dev)
         # pinned host staging
         self.h_tokens = torch.zeros(B, dtype=torch.long, pin_memory=True)
         self.h_positions = torch.zeros(B, dtype=torch.int32, pin_memory=True)
@@ -61,6 +64,9 @@ class DecodeGraphRunner:
         self.h_block_tables = torch.zeros(
             (B, max_blocks_per_seq), dtype=torch.int32, pin_memory=True
         )
+        self.h_lora_idx = torch.full((B,), -1, dtype=torch.int32,
+                                     pin_memory=True)
+        self.use_lora = getattr(model, "lora_slots", None) is not None
 
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self.logits: Dict[int, torch.Tensor] = {}
@@ -77,6 +83,7 @@ class DecodeGraphRunner:
             num_decode_seqs=b,
             decode_seq_lens=self.seq_lens[:b],
             decode_block_tables=self.block_tables[:b],
+            lora_idx=self.lora_idx[:b] if self.use_lora else None,
         )
 
     @torch.no_grad()
@@ -126,6 +133,7 @@ class DecodeGraphRunner:
         slots: np.ndarray,
         seq_lens: np.ndarray,
         block_tables: List[List[int]],
+        lora_idx: Optional[np.ndarray] = None,
     ) -> torch.Tensor:
         n = len(tokens)
         b = self.bucket_for(n)
@@ -143,6 +151,13 @@ class DecodeGraphRunner:
         for i, bt in enumerate(block_tables):
             hbt[i, : len(bt)] = bt
         hbt[n:b, 0] = 0
+        if self.use_lora:
+            if lora_idx is None:
+                self.h_lora_idx[:b] = -1
+            else:
+                self.h_lora_idx[:n] = torch.from_numpy(lora_idx)
+                self.h_lora_idx[n:b] = -1
+            self.lora_idx[:b].copy_(self.h_lora_idx[:b], non_blocking=True)
         # H2D into the static buffers
         self.tokens[:b].copy_(self.h_tokens[:b], non_blocking=True)
         self.positions[:b].copy_(self.h_positions[:b], non_blocking=True)

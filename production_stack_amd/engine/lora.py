@@ -146,3 +146,90 @@ def save_synthetic_adapter(
     with open(os.path.join(path, "adapter_config.json"), "w") as f:
         json.dump({"r": rank, "lora_alpha": 16,
                    "target_modules": list(targets)}, f)
+
+
+class LoRASlotManager:
+    """Stacked per-slot adapter weights for graph-safe batched LoRA.
+
+    One pair of stacks per logical projection:
+      A[proj]: [L, S, R, IN]  bf16   B[proj]: [L, S, W, R]  bf16
+    plus a shared per-slot scale vector. The BGMV kernel (csrc/
+    lora_bgmv.hip) indexes slots per token, so the decode hipGraph can stay
+    captured while adapters load/unload (tensor storage is fixed; only
+    contents and the per-step idx buffer change) — the same design point as
+    vLLM's punica stacks, built MI355X-native.
+    """
+
+    def __init__(self, model_cfg, max_loras: int, max_rank: int,
+                 device, dtype=torch.bfloat16) -> None:
+        h = model_cfg.hidden_size
+        q = model_cfg.num_q_heads * model_cfg.head_dim
+        kv = model_cfg.num_kv_heads * model_cfg.head_dim
+        inter = model_cfg.intermediate_size
+        L = model_cfg.num_layers
+        self.S = max_loras
+        self.R = max_rank
+        self.device = device
+        # proj -> (IN, W, target tensor name, column offset in target)
+        self.spec = {
+            "q": (h, q, "qkv", 0),
+            "k": (h, kv, "qkv", q),
+            "v": (h, kv, "qkv", q + kv),
+            "o": (q, h, "o", 0),
+            "gate": (h, inter, "gate_up", 0),
+            "up": (h, inter, "gate_up", inter),
+            "down": (inter, h, "down", 0),
+        }
+        self.A: Dict[str, torch.Tensor] = {}
+        self.B: Dict[str, torch.Tensor] = {}
+        for p, (IN, W, _, _) in self.spec.items():
+            self.A[p] = torch.zeros(
+                (L, max_loras, max_rank, IN), dtype=dtype, device=device
+            )
+            self.B[p] = torch.zeros(
+                (L, max_loras, W, max_rank), dtype=dtype, device=device
+            )
+        self.scale = torch.zeros(max_loras, dtype=torch.float32,
+                                 device=device)
+        self.slot_by_name: Dict[str, int] = {}
+        self._free = list(range(max_loras))
+
+    def register(self, adapter: "LoRAAdapter") -> int:
+        if adapter.name in self.slot_by_name:
+            return self.slot_by_name[adapter.name]
+        if not self._free:
+            raise RuntimeError(
+                f"all {self.S} LoRA slots in use (max_loras)"
+            )
+        if adapter.rank > self.R:
+            raise ValueError(
+                f"adapter rank {adapter.rank} > max_lora_rank {self.R}"
+            )
+        slot = self._free.pop(0)
+        for li, projs in adapter.layers.items():
+            for p, (A, B) in projs.items():
+                r = A.shape[0]
+                self.A[p][li, slot, :r].copy_(A)
+                self.B[p][li, slot, :, :r].copy_(B)
+        self.scale[slot] = adapter.scaling
+        self.slot_by_name[adapter.name] = slot
+        return slot
+
+    def unregister(self, name: str) -> None:
+        slot = self.slot_by_name.pop(name, None)
+        if slot is None:
+            return
+        for p in self.spec:
+            self.A[p][:, slot].zero_()
+            self.B[p][:, slot].zero_()
+        self.scale[slot] = 0.0
+        self._free.append(slot)
+
+    def apply(self, proj: str, layer: int, out: torch.Tensor,
+              x: torch.Tensor, idx: torch.Tensor) -> None:
+        """out[:, off:+W] += per-token adapter delta (BGMV kernel)."""
+        from production_stack_amd import ops
+
+        _, _, _, off = self.spec[proj]
+        ops.lora_bgmv(out, x, self.A[proj][layer], self.B[proj][layer],
+                      self.scale, idx, off)

@@ -45,6 +45,7 @@ class BatchMeta:
         decode_block_tables: Optional[torch.Tensor],  # [Sd, MB] int32
         prefill_tiles: Optional[torch.Tensor] = None,  # [NT, 4] int32
         lora_groups=None,  # [(LoRAAdapter, row_idx_tensor)]
+        lora_idx: Optional[torch.Tensor] = None,  # [T] int32 slot per row
     ) -> None:
         self.positions = positions
         self.slot_mapping = slot_mapping
@@ -57,6 +58,7 @@ class BatchMeta:
         self.decode_block_tables = decode_block_tables
         self.prefill_tiles = prefill_tiles
         self.lora_groups = lora_groups or []
+        self.lora_idx = lora_idx
 
 
 def build_cos_sin_cache(
@@ -121,7 +123,13 @@ class LlamaLayer(nn.Module):
         qkv = gemm_policy.linear(hidden, self.qkv_proj)
         qs = self.q_heads * self.head_dim
         kvs = self.kv_heads * self.head_dim
-        if meta.lora_groups:
+        slots = getattr(self, "lora_slots", None)
+        if slots is not None and meta.lora_idx is not None:
+            li = self.layer_idx
+            slots.apply("q", li, qkv, hidden, meta.lora_idx)
+            slots.apply("k", li, qkv, hidden, meta.lora_idx)
+            slots.apply("v", li, qkv, hidden, meta.lora_idx)
+        elif meta.lora_groups:
             from production_stack_amd.engine.lora import apply_lora_slice
 
             li = self.layer_idx
@@ -198,7 +206,10 @@ class LlamaLayer(nn.Module):
             )
         attn = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
         attn_out = gemm_policy.linear(attn.view(T, qs), self.o_proj)
-        if meta.lora_groups:
+        if slots is not None and meta.lora_idx is not None:
+            slots.apply("o", self.layer_idx, attn_out,
+                        attn.reshape(T, qs), meta.lora_idx)
+        elif meta.lora_groups:
             apply_lora_slice(
                 attn_out, attn.view(T, qs), meta.lora_groups,
                 self.layer_idx, "o",
@@ -209,7 +220,12 @@ class LlamaLayer(nn.Module):
             attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
         )
         gate_up = gemm_policy.linear(hidden, self.gate_up_proj)
-        if meta.lora_groups:
+        if slots is not None and meta.lora_idx is not None:
+            slots.apply("gate", self.layer_idx, gate_up, hidden,
+                        meta.lora_idx)
+            slots.apply("up", self.layer_idx, gate_up, hidden,
+                        meta.lora_idx)
+        elif meta.lora_groups:
             apply_lora_slice(
                 gate_up, hidden, meta.lora_groups, self.layer_idx,
                 "gate", 0, self.inter,
@@ -220,7 +236,9 @@ class LlamaLayer(nn.Module):
             )
         act = ops.silu_and_mul(gate_up)
         mlp_out = gemm_policy.linear(act, self.down_proj)
-        if meta.lora_groups:
+        if slots is not None and meta.lora_idx is not None:
+            slots.apply("down", self.layer_idx, mlp_out, act, meta.lora_idx)
+        elif meta.lora_groups:
             apply_lora_slice(
                 mlp_out, act, meta.lora_groups, self.layer_idx, "down"
             )

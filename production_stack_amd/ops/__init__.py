@@ -212,10 +212,10 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """out = x @ w.T for decode-shaped M<=128 (weight-stream-bound GEMMs
     where hipBLASLt tiles poorly — see csrc/skinny_gemm.hip). GPU only."""
     _require_ext()
-    out = torch.zeros(
+    out = torch.empty(
         (x.shape[0], w.shape[0]), dtype=torch.float32, device=x.device
     )
-    _C.skinny_gemm(out, x, w)
+    _C.skinny_gemm(out, x, w)  # zeroes out via hipMemsetAsync internally
     return out.to(x.dtype)
 
 

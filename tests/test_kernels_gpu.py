@@ -475,3 +475,24 @@ def test_engine_fp8_kv_e2e():
     # design, so only require early-token agreement
     agree = sum(a == b for a, b in zip(bf[:4], f8[:4]))
     assert agree >= 2, (bf, f8)
+
+
+def test_lora_bgmv_kernel_matches_reference():
+    from production_stack_amd import ops
+    from production_stack_amd.ops import reference
+
+    torch.manual_seed(1)
+    T, IN, W, R, S = 33, 1024, 768, 16, 4
+    x = torch.randn(T, IN, dtype=torch.bfloat16, device="cuda")
+    out = torch.randn(T, W + 32, dtype=torch.bfloat16, device="cuda")
+    A = (torch.randn(S, R, IN, dtype=torch.bfloat16, device="cuda") * 0.05)
+    B = (torch.randn(S, W, R, dtype=torch.bfloat16, device="cuda") * 0.05)
+    scale = torch.rand(S, device="cuda") + 0.5
+    idx = torch.randint(-1, S, (T,), dtype=torch.int32, device="cuda")
+    want = out.clone().cpu()
+    reference.lora_bgmv(want, x.cpu(), A.cpu(), B.cpu(), scale.cpu(),
+                        idx.cpu(), 16)
+    ops.lora_bgmv(out, x, A, B, scale, idx, col_off=16)
+    torch.cuda.synchronize()
+    diff = (out.cpu().float() - want.float()).abs().max()
+    assert diff < 0.25, f"max diff {diff}"
