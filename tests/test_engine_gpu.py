@@ -127,7 +127,7 @@ def test_lora_served_through_decode_graphs(tmp_path):
     from production_stack_amd.engine.sampling import SamplingParams
 
     adir = str(tmp_path / "ad")
-    save_synthetic_adapter(adir, hidden=1024, q_size=1024, kv_size=512,
+    save_synthetic_adapter(adir, hidden=512, q_size=512, kv_size=256,
                            num_layers=4, seed=9)
     eng = make_engine(enable_lora=True, max_loras=2, max_lora_rank=8)
     assert eng.runner.graphs is not None
