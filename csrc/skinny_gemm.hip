@@ -1,5 +1,5 @@
 // Skinny-batch GEMM for decode steps: out[M, N] = x[M, K] @ W[N, K]^T,
-// bf16 inputs, fp32 accumulation, M <= 128.
+// bf16 inputs, fp32 accumulation, M <= 256.
 //
 // Why it exists: hipBLASLt's tile selection for the decode-shaped GEMMs
 // (qkv [M,4096]x[6144,4096], o, down) streams weights at only 1.4-2 TB/s at
@@ -110,7 +110,7 @@ extern "C" {
 
 int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
                    int K, long x_stride, hipStream_t stream) {
-  if (N % 64 != 0 || K % 64 != 0 || M > 128) return -1;
+  if (N % 64 != 0 || K % 64 != 0 || M > 256) return -1;
   const int kchunks = K / 64;
   // pick splits to land near ~1024 workgroups
   int splits = 1024 / (N / 64);
@@ -125,7 +125,8 @@ int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
   if (M <= 16) PS_SG(1);
   else if (M <= 32) PS_SG(2);
   else if (M <= 64) PS_SG(4);
-  else PS_SG(8);
+  else if (M <= 128) PS_SG(8);
+  else PS_SG(16);
 #undef PS_SG
   return 0;
 }

@@ -62,5 +62,5 @@ if __name__ == "__main__":
     print("TunableOp:", os.environ.get("PYTORCH_TUNABLEOP_ENABLED", "0"))
     for M in (16, 32, 64, 128, 256):
         bench(M)
-    for M in (16, 32, 64, 128):
+    for M in (16, 32, 64, 128, 256):
         bench_skinny(M)

@@ -399,6 +399,28 @@ class ModelRunner:
         n_mb = self.config.parallel.pp_microbatches
         if self.pipeline is not None and n_mb > 1 and len(out.scheduled) > 1:
             return self._execute_pp_microbatched(out, bm, n_mb)
+        # mixed step: replay the decode rows through their captured graph
+        # and run only the prefill chunks eagerly (per-row outputs are
+        # independent; both enqueue on the same stream so KV ordering
+        # holds). Without this, one prefill chunk forced the entire
+        # 256-row step into eager mode.
+        if (
+            self.graphs is not None
+            and self.pipeline is None
+            and not getattr(self, "tp_serving", False)
+        ):
+            dec = [s for s in out.scheduled if s.is_decode]
+            pre = [s for s in out.scheduled if not s.is_decode]
+            if dec and pre:
+                fast = self._execute_decode_graph(
+                    SchedulerOutput(scheduled=dec), bm
+                )
+                if fast is not None:
+                    eager = self.execute(
+                        SchedulerOutput(scheduled=pre), bm
+                    )
+                    fast.update(eager)
+                    return fast
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
         if token_t.numel() == 0:
             return {}

@@ -25,7 +25,7 @@ _POLICY: Dict[Tuple[int, int, int], bool] = {}
 
 
 def _eligible(M: int, N: int, K: int) -> bool:
-    return M <= 128 and N % 64 == 0 and K % 64 == 0
+    return M <= 256 and N % 64 == 0 and K % 64 == 0
 
 
 def _time_fn(fn, iters: int = 20) -> float:
