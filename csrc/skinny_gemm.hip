@@ -106,6 +106,98 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
   }
 }
 
+
+// Wave-M-split variant for M in (64, 256]: the single-wave-per-16-N-rows
+// kernel above reads one A-frag from LDS per MFMA (1:1), which makes it
+// LDS-read-bound once M_TILES grows (measured: 667 GB/s W-stream at
+// M=256 vs 1869 at M=64). Here each wave owns MT m-tiles x 64 N-rows
+// (4 n-tiles held as register B-frags, shared W loads L2-coalesced
+// across the 4 waves), so one LDS A-frag pair feeds 8 MFMAs and the
+// kernel returns to W-streaming-bound.
+template <int MT>  // m-tiles per wave; WG covers 4*MT*16 rows of M
+__global__ __launch_bounds__(256, 4) void
+skinny_gemm_wide_kernel(
+    float* __restrict__ out,               // [M, N] fp32, pre-zeroed
+    const unsigned short* __restrict__ x,  // [M, K]
+    const unsigned short* __restrict__ w,  // [N, K]
+    int M, int N, int K, long x_stride) {
+  constexpr int MP = 4 * MT * 16;
+  const int n0 = blockIdx.x * 64;
+  const int splits = gridDim.y;
+  const int kchunks = K / 64;
+  const int per_split = (kchunks + splits - 1) / splits;
+  const int kc_begin = blockIdx.y * per_split;
+  const int kc_end = min(kchunks, kc_begin + per_split);
+  if (kc_begin >= kc_end) return;
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int g = lane >> 4;
+  const int rc = lane & 15;
+
+  __shared__ __align__(16) unsigned short x_lds[MP][64];
+
+  ps_gf32x4 acc[MT][4];
+#pragma unroll
+  for (int mt = 0; mt < MT; mt++)
+#pragma unroll
+    for (int nt = 0; nt < 4; nt++) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int kc = kc_begin; kc < kc_end; kc++) {
+    for (int u = tid; u < MP * 8; u += 256) {
+      const int row = u >> 3;
+      const int slot = u & 7;
+      ps_bf16x8 v = {};
+      if (row < M)
+        v = *(const ps_bf16x8*)(x + (long)row * x_stride + kc * 64 +
+                                slot * 8);
+      *(ps_bf16x8*)(&x_lds[row][(slot ^ (row & 7)) * 8]) = v;
+    }
+    __syncthreads();
+    // B-frags: 4 n-tiles x 2 k-halves; all 4 waves issue the same
+    // addresses (L2-served after the first)
+    ps_gbf16x8 b[4][2];
+#pragma unroll
+    for (int nt = 0; nt < 4; nt++) {
+      const unsigned short* wr = w + (long)(n0 + nt * 16 + rc) * K + kc * 64;
+      b[nt][0] = ps_as_gbf16(*(const ps_bf16x8*)(wr + g * 8));
+      b[nt][1] = ps_as_gbf16(*(const ps_bf16x8*)(wr + 32 + g * 8));
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < MT; mt++) {
+      const int row = (wave * MT + mt) * 16 + rc;
+      ps_gbf16x8 a0 = ps_as_gbf16(
+          *(const ps_bf16x8*)(&x_lds[row][(g ^ (row & 7)) * 8]));
+      ps_gbf16x8 a1 = ps_as_gbf16(
+          *(const ps_bf16x8*)(&x_lds[row][((4 + g) ^ (row & 7)) * 8]));
+#pragma unroll
+      for (int nt = 0; nt < 4; nt++) {
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a0, b[nt][0], acc[mt][nt], 0, 0, 0);
+        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a1, b[nt][1], acc[mt][nt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mt = 0; mt < MT; mt++) {
+#pragma unroll
+    for (int r = 0; r < 4; r++) {
+      const int m = (wave * MT + mt) * 16 + g * 4 + r;
+      if (m < M) {
+#pragma unroll
+        for (int nt = 0; nt < 4; nt++)
+          atomicAdd(&out[(long)m * N + n0 + nt * 16 + rc], acc[mt][nt][r]);
+      }
+    }
+  }
+}
+
 extern "C" {
 
 int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
@@ -122,12 +214,17 @@ int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
   skinny_gemm_kernel<MT><<<grid, block, 0, stream>>>(                       \
       (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
       M, N, K, x_stride)
+#define PS_SGW(MT)                                                          \
+  skinny_gemm_wide_kernel<MT><<<grid, block, 0, stream>>>(                  \
+      (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
+      M, N, K, x_stride)
   if (M <= 16) PS_SG(1);
   else if (M <= 32) PS_SG(2);
   else if (M <= 64) PS_SG(4);
-  else if (M <= 128) PS_SG(8);
-  else PS_SG(16);
+  else if (M <= 128) PS_SGW(2);
+  else PS_SGW(4);
 #undef PS_SG
+#undef PS_SGW
   return 0;
 }
 

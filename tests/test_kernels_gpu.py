@@ -340,6 +340,8 @@ def test_kv_quant_gpu_matches_reference():
 @pytest.mark.parametrize("M,N,K", [
     (1, 6144, 4096), (16, 6144, 4096), (17, 4096, 4096),
     (64, 4096, 14336), (128, 28672, 4096), (100, 4096, 4096),
+    (192, 6144, 4096), (256, 4096, 4096), (256, 28672, 4096),
+    (250, 4096, 14336),
 ])
 def test_skinny_gemm(M, N, K):
     torch.manual_seed(M)
