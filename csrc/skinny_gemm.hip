@@ -115,7 +115,7 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
 // across the 4 waves), so one LDS A-frag pair feeds 8 MFMAs and the
 // kernel returns to W-streaming-bound.
 template <int MT>  // m-tiles per wave; WG covers 4*MT*16 rows of M
-__global__ __launch_bounds__(256, 4) void
+__global__ __launch_bounds__(256, 2) void
 skinny_gemm_wide_kernel(
     float* __restrict__ out,               // [M, N] fp32, pre-zeroed
     const unsigned short* __restrict__ x,  // [M, K]
@@ -144,26 +144,52 @@ skinny_gemm_wide_kernel(
 #pragma unroll
     for (int nt = 0; nt < 4; nt++) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int kc = kc_begin; kc < kc_end; kc++) {
-    for (int u = tid; u < MP * 8; u += 256) {
+  // T14 software pipeline: x-chunk and B-frags for chunk c+1 load into
+  // registers while chunk c's MFMAs run, so the per-chunk critical path
+  // is max(MFMA, load) instead of their sum.
+  constexpr int XS = MP * 8 / 256;  // staging slots per thread
+  ps_bf16x8 xstage[XS];
+  ps_gbf16x8 b[4][2];
+  auto load_x = [&](int kc) {
+#pragma unroll
+    for (int i = 0; i < XS; i++) {
+      const int u = tid + i * 256;
       const int row = u >> 3;
       const int slot = u & 7;
-      ps_bf16x8 v = {};
-      if (row < M)
-        v = *(const ps_bf16x8*)(x + (long)row * x_stride + kc * 64 +
-                                slot * 8);
-      *(ps_bf16x8*)(&x_lds[row][(slot ^ (row & 7)) * 8]) = v;
+      xstage[i] = (row < M)
+                      ? *(const ps_bf16x8*)(x + (long)row * x_stride +
+                                            kc * 64 + slot * 8)
+                      : ps_bf16x8{};
     }
-    __syncthreads();
-    // B-frags: 4 n-tiles x 2 k-halves; all 4 waves issue the same
-    // addresses (L2-served after the first)
-    ps_gbf16x8 b[4][2];
+  };
+  auto load_b = [&](int kc) {
 #pragma unroll
     for (int nt = 0; nt < 4; nt++) {
-      const unsigned short* wr = w + (long)(n0 + nt * 16 + rc) * K + kc * 64;
+      const unsigned short* wr =
+          w + (long)(n0 + nt * 16 + rc) * K + kc * 64;
       b[nt][0] = ps_as_gbf16(*(const ps_bf16x8*)(wr + g * 8));
       b[nt][1] = ps_as_gbf16(*(const ps_bf16x8*)(wr + 32 + g * 8));
     }
+  };
+  load_x(kc_begin);
+  load_b(kc_begin);
+  for (int kc = kc_begin; kc < kc_end; kc++) {
+#pragma unroll
+    for (int i = 0; i < XS; i++) {
+      const int u = tid + i * 256;
+      const int row = u >> 3;
+      const int slot = u & 7;
+      *(ps_bf16x8*)(&x_lds[row][(slot ^ (row & 7)) * 8]) = xstage[i];
+    }
+    __syncthreads();
+    if (kc + 1 < kc_end) load_x(kc + 1);
+    ps_gbf16x8 bc[4][2];
+#pragma unroll
+    for (int nt = 0; nt < 4; nt++) {
+      bc[nt][0] = b[nt][0];
+      bc[nt][1] = b[nt][1];
+    }
+    if (kc + 1 < kc_end) load_b(kc + 1);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mt = 0; mt < MT; mt++) {
@@ -175,9 +201,9 @@ skinny_gemm_wide_kernel(
 #pragma unroll
       for (int nt = 0; nt < 4; nt++) {
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a0, b[nt][0], acc[mt][nt], 0, 0, 0);
+            a0, bc[nt][0], acc[mt][nt], 0, 0, 0);
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a1, b[nt][1], acc[mt][nt], 0, 0, 0);
+            a1, bc[nt][1], acc[mt][nt], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
