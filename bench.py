@@ -50,6 +50,15 @@ class User:
         self.history = list(system)
         self.round = 0
         self.submit_time = 0.0
+        # randomized FIRST answer length desynchronizes the user rounds:
+        # without it every conversation finishes on the same step and the
+        # measured window is an artificial all-decode phase with no TTFT
+        # samples; with it the steady state is the real serving mix of
+        # per-round prefills over running decodes.
+        self.first_len = int(rng.integers(10, 2 * ANSWER_TOKENS))
+
+    def answer_len(self) -> int:
+        return self.first_len if self.round <= 1 else ANSWER_TOKENS
 
     def next_prompt(self) -> list:
         self.round += 1
@@ -68,7 +77,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=64)
     ap.add_argument("--warmup", type=int, default=16)
-    ap.add_argument("--users", type=int, default=128, help="conversations per GPU")
+    ap.add_argument("--users", type=int, default=256, help="conversations per GPU")
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--max-num-batched-tokens", type=int, default=2048)
@@ -115,9 +124,6 @@ def main() -> None:
         VOCAB_LOW, vocab_high, size=SYSTEM_PROMPT_TOKENS
     ).tolist()
     users = [User(u, rng, system, vocab_high) for u in range(args.users)]
-    params = SamplingParams(
-        max_tokens=ANSWER_TOKENS, temperature=0.0, ignore_eos=True
-    )
     answers: dict = {}
 
     def submit(user: User) -> None:
@@ -133,7 +139,11 @@ def main() -> None:
         rid = f"u{user.uid}-r{user.round}"
         user.submit_time = time.perf_counter()
         answers[rid] = (user, [])
-        engine.add_request(rid, prompt, params)
+        engine.add_request(
+            rid, prompt,
+            SamplingParams(max_tokens=user.answer_len(), temperature=0.0,
+                           ignore_eos=True),
+        )
 
     for u in users:
         submit(u)
