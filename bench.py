@@ -76,7 +76,7 @@ def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=64)
-    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--warmup", type=int, default=48)
     ap.add_argument("--users", type=int, default=256, help="conversations per GPU")
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--max-model-len", type=int, default=4096)
