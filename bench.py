@@ -23,6 +23,7 @@ Rank 0 prints exactly one JSON line.
 import argparse
 import json
 import os
+import sys
 import time
 
 import numpy as np
@@ -84,6 +85,8 @@ def main() -> None:
     ap.add_argument("--kv-cache-dtype", default="auto",
                     choices=["auto", "bf16", "fp8", "fp8_e4m3"])
     ap.add_argument("--device", default=None)
+    ap.add_argument("--profile-host", action="store_true",
+                    help="cProfile the measured loop; report to stderr")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -178,6 +181,12 @@ def main() -> None:
     if use_cuda:
         torch.cuda.synchronize()
     in_window = True
+    prof = None
+    if args.profile_host:
+        import cProfile
+
+        prof = cProfile.Profile()
+        prof.enable()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         run_step()
@@ -187,6 +196,14 @@ def main() -> None:
         dist.barrier()
     elapsed = time.perf_counter() - t0
     in_window = False
+    if prof is not None:
+        import io
+        import pstats
+
+        prof.disable()
+        buf = io.StringIO()
+        pstats.Stats(prof, stream=buf).sort_stats("cumulative").print_stats(25)
+        print(buf.getvalue(), file=sys.stderr)
 
     # aggregate: sum tokens over ranks, max elapsed over ranks
     total_tokens = window_tokens
