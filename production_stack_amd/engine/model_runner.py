@@ -342,9 +342,11 @@ class ModelRunner:
     # ------------------------------------------------------------------
     @torch.no_grad()
     def _execute_decode_graph(
-        self, out: SchedulerOutput, bm: BlockManager
-    ) -> Optional[Dict[str, int]]:
-        """Fast path: decode-only step via hipGraph replay."""
+        self, out: SchedulerOutput, bm: BlockManager, defer_sample=False
+    ):
+        """Fast path: decode-only step via hipGraph replay. With
+        defer_sample the (logits, seqs) pair is returned un-sampled so a
+        caller can enqueue more GPU work before the first D2H sync."""
         import numpy as np
 
         scheduled = out.scheduled
@@ -383,6 +385,8 @@ class ModelRunner:
                 )
         logits = self.graphs.run(tokens, positions, slots, seq_lens, tables,
                                  lora_idx)
+        if defer_sample:
+            return logits, seqs
         sampled = self.sample(logits, seqs)
         return {
             seq.request_id: int(tok) for seq, tok in zip(seqs, sampled)
@@ -412,15 +416,40 @@ class ModelRunner:
             dec = [s for s in out.scheduled if s.is_decode]
             pre = [s for s in out.scheduled if not s.is_decode]
             if dec and pre:
-                fast = self._execute_decode_graph(
-                    SchedulerOutput(scheduled=dec), bm
+                got = self._execute_decode_graph(
+                    SchedulerOutput(scheduled=dec), bm, defer_sample=True
                 )
-                if fast is not None:
-                    eager = self.execute(
+                if got is not None:
+                    # decode replay is in flight; enqueue the prefill
+                    # forward BEFORE the first D2H sync so the two halves
+                    # pipeline on the stream (host profile showed two
+                    # serialized GPU waits per mixed step without this)
+                    logits_d, seqs_d = got
+                    token_t, meta, sample_seqs, rows_t = self.prepare(
                         SchedulerOutput(scheduled=pre), bm
                     )
-                    fast.update(eager)
-                    return fast
+                    logits_p = None
+                    if token_t.numel():
+                        hidden = self.model(token_t, meta, self.kv_caches)
+                        if sample_seqs:
+                            logits_p = self.model.compute_logits(
+                                hidden[rows_t]
+                            )
+                    result = {
+                        seq.request_id: int(tok)
+                        for seq, tok in zip(
+                            seqs_d, self.sample(logits_d, seqs_d)
+                        )
+                    }
+                    if logits_p is not None:
+                        result.update({
+                            seq.request_id: int(tok)
+                            for seq, tok in zip(
+                                sample_seqs,
+                                self.sample(logits_p, sample_seqs),
+                            )
+                        })
+                    return result
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
         if token_t.numel() == 0:
             return {}
