@@ -458,6 +458,9 @@ def initialize_all(app: FastAPI, args) -> None:
         app.state.file_storage = FileStorage(args.file_storage_path)
         app.state.batch_processor = None
 
+    if getattr(args, "dynamic_config_yaml", None) and not \
+            args.dynamic_config_json:
+        args.dynamic_config_json = args.dynamic_config_yaml
     if args.dynamic_config_json:
         from production_stack_amd.router.dynamic_config import (
             initialize_dynamic_config_watcher,

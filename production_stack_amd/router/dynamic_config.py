@@ -68,8 +68,10 @@ class DynamicConfigWatcher:
         self._last = raw
         try:
             if self.path.endswith((".yaml", ".yml")):
-                return yaml.safe_load(raw)
-            return json.loads(raw)
+                d = yaml.safe_load(raw)
+            else:
+                d = json.loads(raw)
+            return _flatten_structured(d)
         except (ValueError, yaml.YAMLError) as e:
             logger.error("invalid dynamic config: %s", e)
             return None
@@ -129,6 +131,35 @@ class DynamicConfigWatcher:
 
 
 _watcher: Optional[DynamicConfigWatcher] = None
+
+
+def _flatten_structured(d):
+    """Accept the reference's structured YAML schema
+    (`models: {name: {static_backends: [...], static_model_type: t}}`,
+    `aliases: {alias: model}`) by flattening it to the comma-separated
+    static_* strings the router config uses
+    (reference parsers/yaml_utils.py:10-38)."""
+    if not isinstance(d, dict) or "models" not in d:
+        return d
+    out = dict(d)
+    models = out.pop("models") or {}
+    backends, names, types = [], [], []
+    for name, det in models.items():
+        for b in det.get("static_backends", []):
+            backends.append(b)
+            names.append(name)
+            types.append(det.get("static_model_type", "chat"))
+    out.setdefault("static_backends", ",".join(backends))
+    out.setdefault("static_models", ",".join(names))
+    if any(t != "chat" for t in types):
+        out.setdefault("static_model_types", ",".join(types))
+    aliases = out.pop("aliases", None)
+    if aliases:
+        out.setdefault(
+            "static_aliases",
+            ",".join(f"{a}:{m}" for a, m in aliases.items()),
+        )
+    return out
 
 
 def initialize_dynamic_config_watcher(

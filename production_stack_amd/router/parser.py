@@ -64,6 +64,9 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
 
     # dynamic config
     p.add_argument("--dynamic-config-json", default=None)
+    p.add_argument("--dynamic-config-yaml", default=None,
+                   help="YAML form of --dynamic-config-json; supports the "
+                        "structured models:/aliases: schema")
 
     # features
     p.add_argument("--feature-gates", default=None,
