@@ -28,6 +28,14 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     p.add_argument("--static-model-types", type=str, default=None)
     p.add_argument("--static-backend-health-checks", action="store_true")
     p.add_argument("--health-check-interval", type=float, default=60.0)
+    p.add_argument("--static-backend-health-check-interval", type=float,
+                   default=None, dest="health_check_interval_alias",
+                   help="alias for --health-check-interval")
+    p.add_argument("--static-backend-health-check-timeout-seconds",
+                   type=float, default=10.0)
+    p.add_argument("--backend-health-check-timeout-seconds", type=float,
+                   default=None,
+                   help="alias for --static-backend-health-check-timeout-seconds")
     p.add_argument("--k8s-namespace", default="default")
     p.add_argument("--k8s-port", type=int, default=8000)
     p.add_argument("--k8s-label-selector", default=None)
@@ -64,6 +72,10 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
 
     # dynamic config
     p.add_argument("--dynamic-config-json", default=None)
+    p.add_argument("--root-path", default=None,
+                   help="ASGI root_path when served behind a path prefix")
+    p.add_argument("--version", action="version",
+                   version="production-stack-amd router 0.1")
     p.add_argument("--dynamic-config-yaml", default=None,
                    help="YAML form of --dynamic-config-json; supports the "
                         "structured models:/aliases: schema")
@@ -75,6 +87,10 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
                    help="module path exposing pre/post request hooks")
     p.add_argument("--request-rewriter", default="noop")
     p.add_argument("--file-storage-path", default="/tmp/vllm_files")
+    p.add_argument("--file-storage-class", default="local_file",
+                   choices=["local_file"])
+    p.add_argument("--batch-processor", default="local",
+                   choices=["local"])
     p.add_argument("--enable-batch-api", action="store_true")
     p.add_argument("--batch-processor-db", default="/tmp/vllm_batches.sqlite")
 
@@ -102,6 +118,12 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     p.add_argument("--log-format", default="text", choices=["text", "json"])
 
     args = p.parse_args(argv)
+    # flag aliases (reference parser compatibility)
+    if getattr(args, "health_check_interval_alias", None):
+        args.health_check_interval = args.health_check_interval_alias
+    if getattr(args, "backend_health_check_timeout_seconds", None):
+        args.static_backend_health_check_timeout_seconds = \
+            args.backend_health_check_timeout_seconds
     validate_args(args)
     return args
 
