@@ -517,7 +517,8 @@ def main() -> None:
     app = build_app()
     initialize_all(app, args)
     set_ulimit()
-    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    uvicorn.run(app, host=args.host, port=args.port, log_level="warning",
+                root_path=args.root_path or "")
 
 
 if __name__ == "__main__":
