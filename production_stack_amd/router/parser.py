@@ -59,6 +59,15 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     p.add_argument("--lmcache-controller-port", type=int, default=9000,
                    help="KV-pool controller port (kvaware routing)")
     p.add_argument("--kv-controller-host", default="127.0.0.1")
+    # accepted for reference CLI compatibility; the in-house msgpack-TCP
+    # controller uses a single port (no separate ZMQ reply/heartbeat
+    # sockets), and health/timeout knobs map onto its client timeouts
+    p.add_argument("--lmcache-controller-reply-port", type=int, default=None)
+    p.add_argument("--lmcache-controller-heartbeat-port", type=int,
+                   default=None)
+    p.add_argument("--lmcache-health-check-interval", type=float,
+                   default=30.0)
+    p.add_argument("--lmcache-worker-timeout", type=float, default=10.0)
     p.add_argument("--prefill-model-labels", type=str, default=None)
     p.add_argument("--decode-model-labels", type=str, default=None)
     p.add_argument("--max-instance-failover-reroute-attempts", type=int,
