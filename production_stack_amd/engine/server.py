@@ -623,7 +623,30 @@ def main() -> None:
                     help="cacheserver data plane host:port (shared KV tier)")
     ap.add_argument("--kv-cache-dtype", default="auto",
                     choices=["auto", "bf16", "fp8", "fp8_e4m3"])
+    ap.add_argument("--enable-lora", action="store_true",
+                    help="enable graph-safe BGMV adapter slots")
+    ap.add_argument("--max-loras", type=int, default=4)
+    ap.add_argument("--max-lora-rank", type=int, default=16)
+    ap.add_argument("--kv-transfer-config", default=None,
+                    help="vLLM-style JSON; kv_role/kv_rank/kv_parallel_size "
+                         "are mapped onto the native RCCL KV transfer")
+    ap.add_argument("--chat-template", default=None,
+                    help="path to a chat template file (tokenizer)")
+    # accepted for `vllm serve` CLI compatibility (no-ops here)
+    ap.add_argument("--revision", default=None)
+    ap.add_argument("--runner", default=None)
+    ap.add_argument("--convert", default=None)
+    ap.add_argument("--trust-remote-code", action="store_true")
     args = ap.parse_args()
+    if args.kv_transfer_config:
+        import json as _json
+
+        ktc = _json.loads(args.kv_transfer_config)
+        args.kv_role = ktc.get("kv_role", args.kv_role)
+        if ktc.get("kv_rank") is not None:
+            args.kv_rank = int(ktc["kv_rank"])
+        if ktc.get("kv_parallel_size"):
+            args.kv_world = int(ktc["kv_parallel_size"])
 
     from production_stack_amd.engine.config import ParallelConfig
 
@@ -646,6 +669,9 @@ def main() -> None:
             max_num_batched_tokens=args.max_num_batched_tokens,
             enable_chunked_prefill=args.enable_chunked_prefill,
         ),
+        enable_lora=args.enable_lora,
+        max_loras=args.max_loras,
+        max_lora_rank=args.max_lora_rank,
         parallel=ParallelConfig(
             tensor_parallel_size=args.tensor_parallel_size,
             pipeline_parallel_size=args.pipeline_parallel_size,
