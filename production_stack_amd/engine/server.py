@@ -619,6 +619,9 @@ def main() -> None:
     ap.add_argument("--kv-master-port", type=int, default=14500)
     ap.add_argument("--kv-side-port", type=int, default=14001)
     ap.add_argument("--cpu-offload-gb", type=float, default=0.0)
+    ap.add_argument("--offload-dtype", default="bf16",
+                    choices=["bf16", "int8"],
+                    help="host-pool record dtype (int8 = rowwise-quant)")
     ap.add_argument("--remote-kv-url", default=None,
                     help="cacheserver data plane host:port (shared KV tier)")
     ap.add_argument("--kv-cache-dtype", default="auto",
@@ -661,6 +664,7 @@ def main() -> None:
             gpu_memory_utilization=args.gpu_memory_utilization,
             enable_prefix_caching=args.enable_prefix_caching,
             cpu_offload_gb=args.cpu_offload_gb,
+            offload_dtype=args.offload_dtype,
             remote_kv_url=args.remote_kv_url,
             kv_cache_dtype=args.kv_cache_dtype,
         ),
