@@ -42,7 +42,7 @@ __global__ __launch_bounds__(NWAVES * 64, LOWREG ? 3 : 1) void paged_attn_decode
     const KVT* __restrict__ v_cache,             // [NB, KH, BS, HD]
     const int* __restrict__ block_tables,        // [S, max_blocks]
     const int* __restrict__ seq_lens,            // [S]
-    int max_blocks, float scale, int KH, long q_stride) {
+    int max_blocks, float scale, int KH, long q_stride, int window) {
   using KVTr = ps_kv_traits<KVT>;
   using kvec8 = typename KVTr::vec8;
   constexpr int D = HEAD_DIM;
@@ -86,11 +86,15 @@ __global__ __launch_bounds__(NWAVES * 64, LOWREG ? 3 : 1) void paged_attn_decode
   }
 
   const int nblocks = (ctx + BLOCK_SIZE - 1) / BLOCK_SIZE;
+  // sliding window (Mistral-style): only keys with pos >= win_lo attend
+  const int win_lo = (window > 0 && ctx > window) ? ctx - window : 0;
+  const int first_block = win_lo / BLOCK_SIZE;
   // split-KV: this workgroup handles blocks [b_begin, b_end)
   const int nsplit = gridDim.z;
   const int split = blockIdx.z;
-  const int per_split = (nblocks + nsplit - 1) / nsplit;
-  const int b_begin = split * per_split;
+  const int live_blocks = nblocks - first_block;
+  const int per_split = (live_blocks + nsplit - 1) / nsplit;
+  const int b_begin = first_block + split * per_split;
   const int b_end = min(nblocks, b_begin + per_split);
   const int* bt = block_tables + (long)seq * max_blocks;
   // Per paged block: score all of this sub-group's tokens first
@@ -115,7 +119,8 @@ __global__ __launch_bounds__(NWAVES * 64, LOWREG ? 3 : 1) void paged_attn_decode
 #pragma unroll
     for (int tt = 0; tt < TPB; tt++) {
       const int tok = tt * TPW + sg;
-      const bool valid = tok < valid_tokens;
+      const bool valid =
+          tok < valid_tokens && b * BLOCK_SIZE + tok >= win_lo;
       kvec8 kv = *(const kvec8*)(kb + tok * D + sl * 8);
       if constexpr (!LOWREG)
         vv[tt] = *(const kvec8*)(vb + tok * D + sl * 8);
@@ -295,7 +300,8 @@ __global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ token_seq,     // [T] sequence index per q token
     const int* __restrict__ token_pos,     // [T] absolute position per q token
-    int max_blocks, float scale, int KH, int GQ, long q_stride) {
+    int max_blocks, float scale, int KH, int GQ, long q_stride,
+    int window) {
   constexpr int D = HEAD_DIM;
   constexpr int LPG = D / 8;
   constexpr int TPW = 64 / LPG;
@@ -324,8 +330,9 @@ __global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
   for (int j = 0; j < 8; j++) acc[j] = 0.f;
 
   const int nblocks = (ctx + BLOCK_SIZE - 1) / BLOCK_SIZE;
+  const int win_lo = (window > 0 && ctx > window) ? ctx - window : 0;
   const int* bt = block_tables + (long)seq * max_blocks;
-  for (int b = 0; b < nblocks; b++) {
+  for (int b = win_lo / BLOCK_SIZE; b < nblocks; b++) {
     const long blk = bt[b];
     const unsigned short* kb = k_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
     const unsigned short* vb = v_cache + ((blk * KH + kvh) * BLOCK_SIZE) * D;
@@ -333,7 +340,8 @@ __global__ __launch_bounds__(64) void paged_attn_prefill_kernel(
 #pragma unroll
     for (int tt = 0; tt < BLOCK_SIZE / TPW; tt++) {
       const int tok = tt * TPW + sg;
-      const bool valid = tok < valid_tokens;
+      const bool valid =
+          tok < valid_tokens && b * BLOCK_SIZE + tok >= win_lo;
       ps_bf16x8 kv = *(const ps_bf16x8*)(kb + tok * D + sl * 8);
       ps_bf16x8 vv = *(const ps_bf16x8*)(vb + tok * D + sl * 8);
       float s = 0.f;
@@ -490,7 +498,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
                          long q_stride, int variant, int kv_fp8,
-                         hipStream_t stream) {
+                         int window, hipStream_t stream) {
   dim3 grid(num_seqs, KH, num_splits);
   constexpr int NW = 4;
   dim3 block(NW * 64);
@@ -500,7 +508,7 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
           (unsigned short*)out, (float*)ws_acc, (float*)ws_ml,               \
           (const unsigned short*)q, (const KVT*)k_cache,                     \
           (const KVT*)v_cache, (const int*)block_tables,                     \
-          (const int*)seq_lens, max_blocks, scale, KH, q_stride)
+          (const int*)seq_lens, max_blocks, scale, KH, q_stride, window)
 #define PS_DISPATCH_DECODE(HD, G, BS)                                        \
   do {                                                                       \
     if (kv_fp8) {                                                            \
@@ -546,7 +554,8 @@ int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* token_seq, const void* token_pos,
                           int num_tokens, int num_q_heads, int max_blocks,
                           float scale, int KH, int GQ, int head_dim,
-                          int block_size, long q_stride, hipStream_t stream) {
+                          int block_size, long q_stride, int window,
+                          hipStream_t stream) {
   dim3 grid(num_tokens, num_q_heads);
   dim3 block(64);
 #define PS_DISPATCH_PREFILL(HD, BS)                                         \
@@ -554,7 +563,7 @@ int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
       (unsigned short*)out, (const unsigned short*)q,                       \
       (const unsigned short*)k_cache, (const unsigned short*)v_cache,       \
       (const int*)block_tables, (const int*)token_seq,                      \
-      (const int*)token_pos, max_blocks, scale, KH, GQ, q_stride)
+      (const int*)token_pos, max_blocks, scale, KH, GQ, q_stride, window)
   if (head_dim == 128 && block_size == 16) {
     PS_DISPATCH_PREFILL(128, 16);
     return 0;

@@ -29,20 +29,21 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, float scale, int KH,
                          int GQ, int head_dim, int block_size, int num_splits,
-                         long q_stride, int variant, int kv_fp8,
+                         long q_stride, int variant, int kv_fp8, int window,
                          hipStream_t stream);
 int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* v_cache, const void* block_tables,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
                                int KH, int GQ, int head_dim, long q_stride,
-                               int variant, int kv_fp8, hipStream_t stream);
+                               int variant, int kv_fp8, int window, hipStream_t stream);
 int ps_paged_attn_prefill(void* out, const void* q, const void* k_cache,
                           const void* v_cache, const void* block_tables,
                           const void* token_seq, const void* token_pos,
                           int num_tokens, int num_q_heads, int max_blocks,
                           float scale, int KH, int GQ, int head_dim,
-                          int block_size, long q_stride, hipStream_t stream);
+                          int block_size, long q_stride, int window,
+                          hipStream_t stream);
 void ps_reshape_and_cache(const void* k, const void* v, void* k_cache,
                           void* v_cache, const void* slot_mapping, long T,
                           int KH, int HD, int BS, int kv_fp8,
@@ -146,7 +147,7 @@ void rotary_embedding(at::Tensor positions, at::Tensor q, at::Tensor k,
 void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                        at::Tensor v_cache, at::Tensor block_tables,
                        at::Tensor seq_lens, double scale, int64_t num_splits,
-                       int64_t variant) {
+                       int64_t variant, int64_t window) {
   CHECK_GPU_BF16(out);
   const int kv_fp8 = cache_fp8(k_cache);
   cache_fp8(v_cache);
@@ -179,7 +180,8 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       out.data_ptr(), acc_p, ml_p, q.data_ptr(), k_cache.data_ptr(),
       v_cache.data_ptr(), block_tables.data_ptr(), seq_lens.data_ptr(), S,
       max_blocks, (float)scale, KH, GQ, HD, BS, (int)num_splits,
-      q_row_stride(q, HD), (int)variant, kv_fp8, current_stream());
+      q_row_stride(q, HD), (int)variant, kv_fp8, (int)window,
+      current_stream());
   TORCH_CHECK(rc == 0, "unsupported decode config: head_dim=", HD,
               " block_size=", BS, " gqa=", GQ);
 }
@@ -187,7 +189,7 @@ void paged_attn_decode(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
                         at::Tensor v_cache, at::Tensor block_tables,
                         at::Tensor token_seq, at::Tensor token_pos,
-                        double scale) {
+                        double scale, int64_t window) {
   CHECK_GPU_BF16(out);
   CHECK_GPU_BF16(k_cache);
   CHECK_GPU_BF16(v_cache);
@@ -206,7 +208,7 @@ void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr(), token_seq.data_ptr(), token_pos.data_ptr(), T,
       QH, max_blocks, (float)scale, KH, GQ, HD, BS, q_row_stride(q, HD),
-      current_stream());
+      (int)window, current_stream());
   TORCH_CHECK(rc == 0, "unsupported prefill config: head_dim=", HD,
               " block_size=", BS);
 }
@@ -214,7 +216,7 @@ void paged_attn_prefill(at::Tensor out, at::Tensor q, at::Tensor k_cache,
 void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
                              at::Tensor k_cache, at::Tensor v_cache,
                              at::Tensor block_tables, at::Tensor tile_info,
-                             double scale, int64_t variant) {
+                             double scale, int64_t variant, int64_t window) {
   CHECK_GPU_BF16(out);
   const int kv_fp8 = cache_fp8(k_cache);
   cache_fp8(v_cache);
@@ -231,7 +233,8 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
       out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
       block_tables.data_ptr(), tile_info.data_ptr(), NT, QH,
       (int)block_tables.size(1), (float)scale, KH, GQ, HD,
-      q_row_stride(q, HD), (int)variant, kv_fp8, current_stream());
+      q_row_stride(q, HD), (int)variant, kv_fp8, (int)window,
+      current_stream());
   TORCH_CHECK(rc == 0, "unsupported mfma prefill config: head_dim=", HD);
 }
 
@@ -367,7 +370,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k_cache"),
         pybind11::arg("v_cache"), pybind11::arg("block_tables"),
         pybind11::arg("seq_lens"), pybind11::arg("scale"),
-        pybind11::arg("num_splits") = 0, pybind11::arg("variant") = 0);
+        pybind11::arg("num_splits") = 0, pybind11::arg("variant") = 0,
+        pybind11::arg("window") = 0);
   m.def("paged_attn_prefill", &paged_attn_prefill,
         "Paged attention, chunked prefill (bf16 KV)");
   m.def("lora_bgmv", &lora_bgmv,
@@ -377,7 +381,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("out"), pybind11::arg("q"), pybind11::arg("k_cache"),
         pybind11::arg("v_cache"), pybind11::arg("block_tables"),
         pybind11::arg("tile_info"), pybind11::arg("scale"),
-        pybind11::arg("variant") = 4);
+        pybind11::arg("variant") = 4, pybind11::arg("window") = 0);
   m.def("fused_rope_cache", &fused_rope_cache,
         "Fused RoPE + paged KV append on the packed qkv tensor");
   m.def("reshape_and_cache", &reshape_and_cache,

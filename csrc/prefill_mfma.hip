@@ -45,7 +45,7 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
     const int* __restrict__ block_tables,  // [S, max_blocks]
     const int* __restrict__ tile_info,     // [NT, 4]
     int max_blocks, float scale, int KH, int GQ, long q_stride,
-    int QH, int n_work) {
+    int QH, int n_work, int window) {
   using KVTr = ps_kv_traits<KVT>;
   using kvec8 = typename KVTr::vec8;
   constexpr int D = HEAD_DIM;  // 128
@@ -123,6 +123,10 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
   for (int s = 0; s < 8; s++) o_acc[s] = {0.f, 0.f, 0.f, 0.f};
 
   const int wave_pos_max = q_pos0 + min(wave * 16 + 15, n_rows - 1);
+  // sliding window: the wave's largest-window row sets which chunks can
+  // contribute at all; per-element masking handles the ragged edge
+  const int wave_win_lo =
+      window > 0 ? max(0, (q_pos0 + wave * 16) - window + 1) : 0;
 
   // Staging assignments:
   //   K: thread stages token tv, dims [wave*32, +32) as 4 x b128.
@@ -191,7 +195,9 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
         kstage[h] = *(const kvec8*)(krow + d0 + h * 8);
     }
 
-    const bool wave_active = (wave * 16 < n_rows) && (tok0 <= wave_pos_max);
+    const bool wave_active = (wave * 16 < n_rows) &&
+                             (tok0 <= wave_pos_max) &&
+                             (tok0 + PS_CHUNK > wave_win_lo);
     if (wave_active) {
       // ---- QK^T over NCT 16-token column tiles ----
       ps_mf32x4 s_frag[NCT];
@@ -213,8 +219,10 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
       // ---- mask + online softmax (one rescale per chunk) ----
       // interior chunks (entirely below every row's diagonal, all rows
       // real) skip the per-element mask compare
-      const bool full_chunk = (tok0 + PS_CHUNK - 1 <= q_pos0 + wave * 16) &&
-                              (wave * 16 + 15 < n_rows);
+      bool full_chunk = (tok0 + PS_CHUNK - 1 <= q_pos0 + wave * 16) &&
+                        (wave * 16 + 15 < n_rows);
+      if (window > 0)
+        full_chunk = full_chunk && (tok0 >= wave_pos_max - window + 1);
       float m_new[4];
 #pragma unroll
       for (int r = 0; r < 4; r++) m_new[r] = m_run[r];
@@ -232,7 +240,8 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
           for (int r = 0; r < 4; r++) {
             const int lrow = wave * 16 + g * 4 + r;
             const int q_pos = q_pos0 + lrow;
-            const bool valid = (lrow < n_rows) && (kv_pos <= q_pos);
+            bool valid = (lrow < n_rows) && (kv_pos <= q_pos);
+            if (window > 0) valid = valid && (kv_pos > q_pos - window);
             const float sv = valid ? s_frag[ct][r] : PS_NEG_INF;
             s_frag[ct][r] = sv;
             m_new[r] = fmaxf(m_new[r], sv);
@@ -326,7 +335,8 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* tile_info, int num_tiles,
                                int num_q_heads, int max_blocks, float scale,
                                int KH, int GQ, int head_dim, long q_stride,
-                               int variant, int kv_fp8, hipStream_t stream) {
+                               int variant, int kv_fp8, int window,
+                               hipStream_t stream) {
   if (head_dim != 128) return -1;
   const int n_work = num_tiles * num_q_heads;
   dim3 grid(((n_work + 7) / 8) * 8);
@@ -335,7 +345,7 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
       (unsigned short*)out, (const unsigned short*)q,                       \
       (const KVT*)k_cache, (const KVT*)v_cache, (const int*)block_tables,   \
       (const int*)tile_info, max_blocks, scale, KH, GQ, q_stride,           \
-      num_q_heads, n_work)
+      num_q_heads, n_work, window)
   if (kv_fp8) {
     if (variant == 3) PS_PREFILL_T(3, unsigned char);
     else PS_PREFILL_T(4, unsigned char);

@@ -30,6 +30,11 @@ class ModelConfig:
     rms_norm_eps: float = 1e-5
     bos_token_id: int = 1
     eos_token_id: int = 2
+    # family knobs: Qwen2 uses qkv biases; Mistral attends over a sliding
+    # window; small models often tie lm_head to the embedding
+    qkv_bias: bool = False
+    tie_word_embeddings: bool = False
+    sliding_window: Optional[int] = None
 
     @property
     def q_size(self) -> int:
@@ -68,6 +73,32 @@ ARCHITECTURES = {
         rope_theta=500000.0,
         max_position=8192,
     ),
+    "mistral-7b": ModelConfig(
+        name="mistral-7b",
+        hidden_size=4096,
+        num_layers=32,
+        num_q_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=32000,
+        rope_theta=10000.0,
+        max_position=32768,
+        sliding_window=4096,
+    ),
+    "qwen2-7b": ModelConfig(
+        name="qwen2-7b",
+        hidden_size=3584,
+        num_layers=28,
+        num_q_heads=28,
+        num_kv_heads=4,
+        head_dim=128,
+        intermediate_size=18944,
+        vocab_size=152064,
+        rope_theta=1000000.0,
+        max_position=32768,
+        qkv_bias=True,
+    ),
     # Small config for tests (CPU-runnable).
     "tiny-llama": ModelConfig(
         name="tiny-llama",
@@ -80,6 +111,46 @@ ARCHITECTURES = {
         vocab_size=1024,
         rope_theta=10000.0,
         max_position=2048,
+    ),
+    "tiny-mistral": ModelConfig(
+        name="tiny-mistral",
+        hidden_size=128,
+        num_layers=2,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        intermediate_size=256,
+        vocab_size=1024,
+        rope_theta=10000.0,
+        max_position=2048,
+        sliding_window=48,
+    ),
+    "tiny-qwen2": ModelConfig(
+        name="tiny-qwen2",
+        hidden_size=128,
+        num_layers=2,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=32,
+        intermediate_size=256,
+        vocab_size=1024,
+        rope_theta=10000.0,
+        max_position=2048,
+        qkv_bias=True,
+        tie_word_embeddings=True,
+    ),
+    "mini-mistral": ModelConfig(
+        name="mini-mistral",
+        hidden_size=512,
+        num_layers=4,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        intermediate_size=1024,
+        vocab_size=2048,
+        rope_theta=10000.0,
+        max_position=4096,
+        sliding_window=64,
     ),
     # GPU-runnable small config with kernel-supported head_dim.
     "mini-llama": ModelConfig(

@@ -103,6 +103,11 @@ class LlamaLayer(nn.Module):
             torch.empty(h, dtype=torch.bfloat16)
         )
         self.scale = 1.0 / math.sqrt(self.head_dim)
+        self.window = cfg.sliding_window or 0
+        self.qkv_bias = (
+            nn.Parameter(torch.empty(qs + 2 * kvs, dtype=torch.bfloat16))
+            if cfg.qkv_bias else None
+        )
 
     def forward(
         self,
@@ -121,6 +126,8 @@ class LlamaLayer(nn.Module):
                 hidden, residual, self.input_norm, cfg.rms_norm_eps
             )
         qkv = gemm_policy.linear(hidden, self.qkv_proj)
+        if self.qkv_bias is not None:
+            qkv = qkv + self.qkv_bias
         qs = self.q_heads * self.head_dim
         kvs = self.kv_heads * self.head_dim
         slots = getattr(self, "lora_slots", None)
@@ -179,6 +186,7 @@ class LlamaLayer(nn.Module):
                         meta.prefill_block_tables,
                         meta.prefill_tiles,
                         self.scale,
+                        self.window,
                     )
                 )
             else:
@@ -191,6 +199,7 @@ class LlamaLayer(nn.Module):
                         meta.prefill_token_seq,
                         meta.prefill_token_pos,
                         self.scale,
+                        self.window,
                     )
                 )
         if meta.num_decode_seqs > 0:
@@ -202,6 +211,7 @@ class LlamaLayer(nn.Module):
                     meta.decode_block_tables,
                     meta.decode_seq_lens,
                     self.scale,
+                    self.window,
                 )
             )
         attn = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
@@ -279,11 +289,14 @@ class LlamaForCausalLM(nn.Module):
             self.final_norm = nn.Parameter(
                 torch.empty(cfg.hidden_size, dtype=torch.bfloat16)
             )
-            self.lm_head = nn.Parameter(
-                torch.empty(
-                    cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16
+            if cfg.tie_word_embeddings and self.is_first:
+                self.lm_head = self.embed  # tied (Qwen2-small style)
+            else:
+                self.lm_head = nn.Parameter(
+                    torch.empty(
+                        cfg.vocab_size, cfg.hidden_size, dtype=torch.bfloat16
+                    )
                 )
-            )
         self.register_buffer(
             "cos_sin",
             build_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_theta),
@@ -297,6 +310,8 @@ class LlamaForCausalLM(nn.Module):
         for name, p in self.named_parameters():
             if "norm" in name:
                 p.fill_(1.0)
+            elif "bias" in name:
+                p.normal_(0.0, 0.002, generator=gen)
             else:
                 p.normal_(0.0, 0.02, generator=gen)
 

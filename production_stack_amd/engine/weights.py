@@ -56,6 +56,11 @@ def load_safetensors(model, path: str) -> None:
             k = shard(get(pre + "self_attn.k_proj.weight"), 0)
             v = shard(get(pre + "self_attn.v_proj.weight"), 0)
             layer.qkv_proj.copy_(torch.cat([q, k, v], dim=0))
+            if layer.qkv_bias is not None:
+                qb = shard(get(pre + "self_attn.q_proj.bias"), 0)
+                kb = shard(get(pre + "self_attn.k_proj.bias"), 0)
+                vb = shard(get(pre + "self_attn.v_proj.bias"), 0)
+                layer.qkv_bias.copy_(torch.cat([qb, kb, vb], dim=0))
             layer.o_proj.copy_(shard(get(pre + "self_attn.o_proj.weight"), 1))
             g = shard(get(pre + "mlp.gate_proj.weight"), 0)
             u = shard(get(pre + "mlp.up_proj.weight"), 0)
@@ -77,8 +82,9 @@ def save_hf_safetensors(model, path: str) -> None:
     state = {
         "model.embed_tokens.weight": model.embed.detach().cpu(),
         "model.norm.weight": model.final_norm.detach().cpu(),
-        "lm_head.weight": model.lm_head.detach().cpu(),
     }
+    if model.lm_head is not model.embed:  # tied weights saved once
+        state["lm_head.weight"] = model.lm_head.detach().cpu()
     layer_start = getattr(model, "layer_start", 0)
     for i, layer in enumerate(model.layers):
         pre = f"model.layers.{layer_start + i}."
@@ -88,6 +94,13 @@ def save_hf_safetensors(model, path: str) -> None:
         state[pre + "self_attn.q_proj.weight"] = qkv[:qs].clone()
         state[pre + "self_attn.k_proj.weight"] = qkv[qs : qs + kvs].clone()
         state[pre + "self_attn.v_proj.weight"] = qkv[qs + kvs :].clone()
+        if layer.qkv_bias is not None:
+            qb = layer.qkv_bias.detach().cpu()
+            state[pre + "self_attn.q_proj.bias"] = qb[:qs].clone()
+            state[pre + "self_attn.k_proj.bias"] = qb[qs : qs + kvs].clone()
+            state[pre + "self_attn.v_proj.bias"] = (
+                qb[qs + kvs :].clone()
+            )
         state[pre + "self_attn.o_proj.weight"] = layer.o_proj.detach().cpu()
         gu = layer.gate_up_proj.detach().cpu()
         state[pre + "mlp.gate_proj.weight"] = gu[: layer.inter].clone()

@@ -117,6 +117,7 @@ def paged_attn_decode(
     block_tables: torch.Tensor,
     seq_lens: torch.Tensor,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
@@ -124,11 +125,12 @@ def paged_attn_decode(
         # variant 1 = low-register packed-q kernel (3 waves/SIMD): measured
         # +17-26% over variant 0 across batch/ctx (profiles/attn_bench).
         _C.paged_attn_decode(
-            out, q, k_cache, v_cache, block_tables, seq_lens, scale, 0, 1
+            out, q, k_cache, v_cache, block_tables, seq_lens, scale, 0, 1,
+            window
         )
         return out
     return reference.paged_attn_decode(
-        q, k_cache, v_cache, block_tables, seq_lens, scale
+        q, k_cache, v_cache, block_tables, seq_lens, scale, window
     )
 
 
@@ -140,16 +142,19 @@ def paged_attn_prefill(
     token_seq: torch.Tensor,
     token_pos: torch.Tensor,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     if q.is_cuda:
         _require_ext()
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         _C.paged_attn_prefill(
-            out, q, k_cache, v_cache, block_tables, token_seq, token_pos, scale
+            out, q, k_cache, v_cache, block_tables, token_seq, token_pos,
+            scale, window
         )
         return out
     return reference.paged_attn_prefill(
-        q, k_cache, v_cache, block_tables, token_seq, token_pos, scale
+        q, k_cache, v_cache, block_tables, token_seq, token_pos, scale,
+        window
     )
 
 
@@ -160,6 +165,7 @@ def paged_attn_prefill_mfma(
     block_tables: torch.Tensor,
     tile_info: torch.Tensor,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
     """MFMA-tiled chunked prefill (GPU, head_dim 128 only)."""
     _require_ext()
@@ -167,7 +173,7 @@ def paged_attn_prefill_mfma(
     # variant 3 = 3-waves/SIMD bound: best measured (222 TF @ctx4096,
     # profiles/attn_bench_prefill)
     _C.paged_attn_prefill_mfma(
-        out, q, k_cache, v_cache, block_tables, tile_info, scale, 3
+        out, q, k_cache, v_cache, block_tables, tile_info, scale, 3, window
     )
     return out
 

@@ -108,8 +108,10 @@ def paged_attn_decode(
     block_tables: torch.Tensor,
     seq_lens: torch.Tensor,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
-    """q: [S, QH, HD] -> out [S, QH, HD] (fp32 math, cast back)."""
+    """q: [S, QH, HD] -> out [S, QH, HD] (fp32 math, cast back).
+    window > 0 = sliding-window attention over the last `window` keys."""
     S, QH, HD = q.shape
     KH = k_cache.shape[1]
     GQ = QH // KH
@@ -118,6 +120,9 @@ def paged_attn_decode(
         ctx = int(seq_lens[s])
         k = _gather_kv(k_cache, block_tables[s], ctx).float()  # [ctx, KH, HD]
         v = _gather_kv(v_cache, block_tables[s], ctx).float()
+        if window and ctx > window:
+            k = k[ctx - window:]
+            v = v[ctx - window:]
         for h in range(QH):
             kvh = h // GQ
             qs = q[s, h].float()
@@ -135,8 +140,10 @@ def paged_attn_prefill(
     token_seq: torch.Tensor,
     token_pos: torch.Tensor,
     scale: float,
+    window: int = 0,
 ) -> torch.Tensor:
-    """q: [T, QH, HD]; each token t attends to cache[token_seq[t]][0..pos]."""
+    """q: [T, QH, HD]; each token t attends to cache[token_seq[t]][0..pos]
+    (or the last `window` positions when window > 0)."""
     T, QH, HD = q.shape
     KH = k_cache.shape[1]
     GQ = QH // KH
@@ -146,6 +153,9 @@ def paged_attn_prefill(
         ctx = int(token_pos[t]) + 1
         k = _gather_kv(k_cache, block_tables[s], ctx).float()
         v = _gather_kv(v_cache, block_tables[s], ctx).float()
+        if window and ctx > window:
+            k = k[ctx - window:]
+            v = v[ctx - window:]
         for h in range(QH):
             kvh = h // GQ
             qs = q[t, h].float()

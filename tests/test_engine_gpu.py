@@ -157,3 +157,35 @@ def test_lora_served_through_decode_graphs(tmp_path):
             ref_out.extend(r.new_token_ids)
     agree = sum(a == b for a, b in zip(outs["lora"], ref_out))
     assert agree >= 4, f"{outs['lora']} vs {ref_out}"
+
+
+def test_mistral_sliding_window_gpu_matches_cpu():
+    """mini-mistral (window=64): windowed HIP kernels (decode + MFMA
+    prefill) must agree with the CPU reference implementation."""
+    from production_stack_amd.engine.config import (
+        CacheConfig as CC,
+        EngineConfig as EC,
+        SchedulerConfig as SC,
+    )
+    from production_stack_amd.ops import gemm_policy
+
+    def cfg():
+        return EC(
+            model="mini-mistral",
+            max_model_len=1024,
+            cache=CC(num_gpu_blocks=256, block_size=16),
+            scheduler=SC(max_num_seqs=8, max_num_batched_tokens=2048),
+        )
+
+    p = SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True)
+    prompt = list(range(9, 209))  # 200 tokens >> window 64
+    gemm_policy.reset()
+    gpu = LLMEngine(cfg(), device="cuda")
+    out_gpu = gpu.generate([prompt], p)["offline-0"]
+    cpu = LLMEngine(cfg(), device="cpu")
+    cpu.runner.model.load_state_dict(
+        {k: v.cpu() for k, v in gpu.runner.model.state_dict().items()}
+    )
+    out_cpu = cpu.generate([prompt], p)["offline-0"]
+    agree = sum(a == b for a, b in zip(out_gpu, out_cpu))
+    assert agree >= 4, f"{out_gpu} vs {out_cpu}"

@@ -498,3 +498,52 @@ def test_lora_bgmv_kernel_matches_reference():
     torch.cuda.synchronize()
     diff = (out.cpu().float() - want.float()).abs().max()
     assert diff < 0.25, f"max diff {diff}"
+
+
+def test_windowed_decode_matches_reference():
+    from production_stack_amd.ops import reference
+
+    torch.manual_seed(21)
+    qh, kh, hd, bs = 8, 2, 128, 16
+    S, max_blocks, W = 3, 24, 80
+    seq_lens = torch.tensor([30, 300, 150], dtype=torch.int32)
+    nb = S * max_blocks + 1
+    k_cache = torch.randn(nb, kh, bs, hd, dtype=torch.bfloat16,
+                          device="cuda") / 4
+    v_cache = torch.randn_like(k_cache) / 4
+    bt = torch.arange(1, S * max_blocks + 1, dtype=torch.int32,
+                      device="cuda").view(S, max_blocks)
+    q = torch.randn(S, qh, hd, dtype=torch.bfloat16, device="cuda") / 4
+    got = ops.paged_attn_decode(q, k_cache, v_cache, bt,
+                                seq_lens.cuda(), 0.0883, window=W)
+    want = reference.paged_attn_decode(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), seq_lens,
+        0.0883, window=W)
+    _close(got.cpu(), want, atol=2e-2, rtol=2e-2)
+
+
+def test_windowed_prefill_mfma_matches_reference():
+    from production_stack_amd.ops import reference
+
+    torch.manual_seed(22)
+    qh, kh, hd, bs = 8, 2, 128, 16
+    W = 64
+    ctx = 256
+    nb = ctx // bs + 1
+    k_cache = torch.randn(nb, kh, bs, hd, dtype=torch.bfloat16,
+                          device="cuda") / 4
+    v_cache = torch.randn_like(k_cache) / 4
+    bt = torch.arange(1, nb, dtype=torch.int32, device="cuda").view(1, -1)
+    q = torch.randn(ctx, qh, hd, dtype=torch.bfloat16, device="cuda") / 4
+    tiles = []
+    for t0 in range(0, ctx, 64):
+        tiles.append([0, t0, t0, min(64, ctx - t0)])
+    tile_info = torch.tensor(tiles, dtype=torch.int32, device="cuda")
+    got = ops.paged_attn_prefill_mfma(q, k_cache, v_cache, bt, tile_info,
+                                      0.0883, window=W)
+    token_seq = torch.zeros(ctx, dtype=torch.int32)
+    token_pos = torch.arange(ctx, dtype=torch.int32)
+    want = reference.paged_attn_prefill(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), token_seq,
+        token_pos, 0.0883, window=W)
+    _close(got.cpu(), want, atol=3e-2, rtol=3e-2)
