@@ -532,7 +532,10 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
     switch (GQ) {
       case 1: PS_DISPATCH_DECODE(128, 1, 16); return 0;
       case 2: PS_DISPATCH_DECODE(128, 2, 16); return 0;
+      case 3: PS_DISPATCH_DECODE(128, 3, 16); return 0;
       case 4: PS_DISPATCH_DECODE(128, 4, 16); return 0;
+      case 6: PS_DISPATCH_DECODE(128, 6, 16); return 0;
+      case 7: PS_DISPATCH_DECODE(128, 7, 16); return 0;
       case 8: PS_DISPATCH_DECODE(128, 8, 16); return 0;
       default: return -1;
     }
@@ -540,7 +543,9 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
   if (head_dim == 64 && block_size == 16 && !kv_fp8) {
     switch (GQ) {
       case 1: PS_DISPATCH_DECODE(64, 1, 16); return 0;
+      case 2: PS_DISPATCH_DECODE(64, 2, 16); return 0;
       case 4: PS_DISPATCH_DECODE(64, 4, 16); return 0;
+      case 8: PS_DISPATCH_DECODE(64, 8, 16); return 0;
       default: return -1;
     }
   }

@@ -73,6 +73,69 @@ ARCHITECTURES = {
         rope_theta=500000.0,
         max_position=8192,
     ),
+    "llama-2-7b": ModelConfig(
+        name="llama-2-7b",
+        hidden_size=4096,
+        num_layers=32,
+        num_q_heads=32,
+        num_kv_heads=32,
+        head_dim=128,
+        intermediate_size=11008,
+        vocab_size=32000,
+        rope_theta=10000.0,
+        max_position=4096,
+    ),
+    "llama-2-13b": ModelConfig(
+        name="llama-2-13b",
+        hidden_size=5120,
+        num_layers=40,
+        num_q_heads=40,
+        num_kv_heads=40,
+        head_dim=128,
+        intermediate_size=13824,
+        vocab_size=32000,
+        rope_theta=10000.0,
+        max_position=4096,
+    ),
+    "llama-3.2-1b": ModelConfig(
+        name="llama-3.2-1b",
+        hidden_size=2048,
+        num_layers=16,
+        num_q_heads=32,
+        num_kv_heads=8,
+        head_dim=64,
+        intermediate_size=8192,
+        vocab_size=128256,
+        rope_theta=500000.0,
+        max_position=8192,
+        tie_word_embeddings=True,
+    ),
+    "llama-3.2-3b": ModelConfig(
+        name="llama-3.2-3b",
+        hidden_size=3072,
+        num_layers=28,
+        num_q_heads=24,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=8192,
+        vocab_size=128256,
+        rope_theta=500000.0,
+        max_position=8192,
+        tie_word_embeddings=True,
+    ),
+    "qwen2-72b": ModelConfig(
+        name="qwen2-72b",
+        hidden_size=8192,
+        num_layers=80,
+        num_q_heads=64,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=29568,
+        vocab_size=152064,
+        rope_theta=1000000.0,
+        max_position=32768,
+        qkv_bias=True,
+    ),
     "mistral-7b": ModelConfig(
         name="mistral-7b",
         hidden_size=4096,

@@ -91,7 +91,10 @@ def test_reshape_and_cache():
     _close(v_cache, vc_ref, atol=0, rtol=0)
 
 
-@pytest.mark.parametrize("qh,kh,hd", [(32, 8, 128), (8, 8, 128), (16, 2, 128), (8, 2, 64)])
+@pytest.mark.parametrize("qh,kh,hd", [(32, 8, 128), (8, 8, 128),
+                                      (16, 2, 128), (8, 2, 64),
+                                      (24, 8, 128), (28, 4, 128),
+                                      (16, 8, 64)])
 def test_paged_attn_decode(qh, kh, hd):
     torch.manual_seed(1)
     S, bs = 5, 16
