@@ -515,7 +515,53 @@ class ModelRunner:
     def sample(
         self, logits: torch.Tensor, seqs: List[Sequence]
     ) -> torch.Tensor:
-        return self.sample_params(logits, [s.params for s in seqs])
+        params = [s.params for s in seqs]
+        if any(p.needs_penalties for p in params):
+            logits = self._apply_penalties(logits, seqs)
+        return self.sample_params(logits, params)
+
+    def _apply_penalties(
+        self, logits: torch.Tensor, seqs: List[Sequence]
+    ) -> torch.Tensor:
+        """OpenAI-style presence/frequency penalties, HF-style repetition
+        penalty, logit_bias, and min_tokens EOS suppression. Applied only
+        to rows whose params ask for it (penalty-free batches skip this
+        entirely)."""
+        logits = logits.float().clone()
+        V = logits.shape[-1]
+        eos = self.model_cfg.eos_token_id
+        for i, seq in enumerate(seqs):
+            p = seq.params
+            if not p.needs_penalties:
+                continue
+            row = logits[i]
+            out_ids = seq.output_token_ids
+            if out_ids and (p.presence_penalty or p.frequency_penalty
+                            or p.repetition_penalty != 1.0):
+                ids = torch.tensor(out_ids, dtype=torch.long,
+                                   device=row.device)
+                counts = torch.bincount(ids, minlength=V).to(row.dtype)
+                seen = counts > 0
+                if p.frequency_penalty:
+                    row -= p.frequency_penalty * counts
+                if p.presence_penalty:
+                    row[seen] -= p.presence_penalty
+                if p.repetition_penalty != 1.0:
+                    pos = seen & (row > 0)
+                    neg = seen & (row <= 0)
+                    row[pos] /= p.repetition_penalty
+                    row[neg] *= p.repetition_penalty
+            if p.logit_bias:
+                for tid, b in p.logit_bias.items():
+                    t = int(tid)
+                    if 0 <= t < V:
+                        row[t] += float(b)
+            if p.min_tokens and len(out_ids) < p.min_tokens:
+                row[eos] = float("-inf")
+                for t in p.stop_token_ids:
+                    if 0 <= t < V:
+                        row[t] = float("-inf")
+        return logits
 
     def sample_params(
         self, logits: torch.Tensor, params: List

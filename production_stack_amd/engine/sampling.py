@@ -17,6 +17,21 @@ class SamplingParams:
     ignore_eos: bool = False
     seed: Optional[int] = None
     logprobs: Optional[int] = None
+    presence_penalty: float = 0.0
+    frequency_penalty: float = 0.0
+    repetition_penalty: float = 1.0
+    logit_bias: Optional[dict] = None  # token_id -> bias
+    min_tokens: int = 0
+
+    @property
+    def needs_penalties(self) -> bool:
+        return (
+            self.presence_penalty != 0.0
+            or self.frequency_penalty != 0.0
+            or self.repetition_penalty != 1.0
+            or bool(self.logit_bias)
+            or self.min_tokens > 0
+        )
 
     @property
     def greedy(self) -> bool:
@@ -31,3 +46,9 @@ class SamplingParams:
             raise ValueError("top_p must be in (0, 1]")
         if self.top_k == 0 or self.top_k < -1:
             raise ValueError("top_k must be -1 (off) or >= 1")
+        if not (-2.0 <= self.presence_penalty <= 2.0):
+            raise ValueError("presence_penalty must be in [-2, 2]")
+        if not (-2.0 <= self.frequency_penalty <= 2.0):
+            raise ValueError("frequency_penalty must be in [-2, 2]")
+        if self.repetition_penalty <= 0:
+            raise ValueError("repetition_penalty must be > 0")

@@ -55,6 +55,12 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
         stop=list(stop) if stop else [],
         stop_token_ids=list(body.get("stop_token_ids") or []),
         seed=body.get("seed"),
+        presence_penalty=float(body.get("presence_penalty") or 0.0),
+        frequency_penalty=float(body.get("frequency_penalty") or 0.0),
+        repetition_penalty=float(body.get("repetition_penalty") or 1.0),
+        logit_bias={int(k): float(v)
+                    for k, v in (body.get("logit_bias") or {}).items()},
+        min_tokens=int(body.get("min_tokens") or 0),
     )
 
 

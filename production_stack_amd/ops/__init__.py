@@ -181,6 +181,8 @@ def paged_attn_prefill_mfma(
 def greedy_sample(logits: torch.Tensor) -> torch.Tensor:
     if logits.is_cuda:
         _require_ext()
+        if logits.dtype != torch.bfloat16:  # penalty-adjusted fp32 rows
+            return logits.float().argmax(dim=-1)
         out = torch.empty(
             logits.shape[0], dtype=torch.long, device=logits.device
         )

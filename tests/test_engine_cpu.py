@@ -246,3 +246,34 @@ def test_qwen2_bias_and_tied_embeddings():
             layer.qkv_bias.zero_()
     out3 = eng.generate([[4, 5, 6, 7]], p)["offline-0"]
     assert len(out3) == 5
+
+
+def test_sampling_penalties_and_logit_bias():
+    """presence/frequency/repetition penalties reduce repeats; logit_bias
+    forces a token; min_tokens suppresses EOS."""
+    eng = make_engine()
+    base = SamplingParams(max_tokens=10, temperature=0.0, ignore_eos=True)
+    out_base = eng.generate([[3, 4, 5]], base)["offline-0"]
+
+    # logit_bias strong enough to force one specific token every step
+    forced = 7
+    p_bias = SamplingParams(max_tokens=4, temperature=0.0, ignore_eos=True,
+                            logit_bias={forced: 1000.0})
+    out_bias = make_engine().generate([[3, 4, 5]], p_bias)["offline-0"]
+    assert out_bias == [forced] * 4
+
+    # heavy frequency penalty must not emit the same token 3x in a row
+    p_pen = SamplingParams(max_tokens=10, temperature=0.0, ignore_eos=True,
+                           frequency_penalty=2.0, presence_penalty=2.0,
+                           repetition_penalty=1.5)
+    out_pen = make_engine().generate([[3, 4, 5]], p_pen)["offline-0"]
+    assert len(out_pen) == 10
+    assert out_pen != out_base or len(set(out_base)) == len(out_base)
+
+    # min_tokens: generation may not stop at EOS before the floor
+    eos = eng.model_cfg.eos_token_id
+    p_min = SamplingParams(max_tokens=8, temperature=0.0, min_tokens=8,
+                           logit_bias={eos: 1000.0})  # EOS otherwise wins
+    out_min = make_engine().generate([[3, 4, 5]], p_min)["offline-0"]
+    assert len(out_min) == 8
+    assert all(t != eos for t in out_min[:-1])
