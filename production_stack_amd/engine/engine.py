@@ -158,6 +158,11 @@ class LLMEngine:
             new = seq.drain_new_tokens()
             if not new and not seq.finished:
                 continue
+            lps = None
+            if seq.params.logprobs is not None and new:
+                lp = self.runner.last_logprobs.pop(seq.request_id, None)
+                # one token sampled per step -> at most one value per drain
+                lps = [lp] * len(new) if lp is not None else None
             first = seq.first_token_time is None and bool(new)
             if first:
                 seq.first_token_time = now
@@ -183,6 +188,7 @@ class LLMEngine:
                     num_output_tokens=len(seq.output_token_ids),
                     num_cached_tokens=seq.num_cached_prompt_tokens,
                     first_token=first,
+                    new_logprobs=lps,
                 )
             )
         self.stats.num_finished += len(finished)

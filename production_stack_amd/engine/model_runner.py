@@ -84,6 +84,9 @@ class ModelRunner:
         self._generator = torch.Generator(device="cpu").manual_seed(
             config.seed + 12345
         )
+        # request_id -> logprob of the most recent sampled token (only for
+        # requests that asked for logprobs; read by engine.step)
+        self.last_logprobs: Dict[str, float] = {}
 
     # ------------------------------------------------------------------
     @property
@@ -518,7 +521,19 @@ class ModelRunner:
         params = [s.params for s in seqs]
         if any(p.needs_penalties for p in params):
             logits = self._apply_penalties(logits, seqs)
-        return self.sample_params(logits, params)
+        sampled = self.sample_params(logits, params)
+        want_lp = [i for i, p in enumerate(params)
+                   if p.logprobs is not None]
+        if want_lp:
+            idx = torch.tensor(want_lp, dtype=torch.long,
+                               device=logits.device)
+            lf = logits.index_select(0, idx).float()
+            lp = torch.log_softmax(lf, dim=-1)
+            chosen = sampled[idx.cpu()].to(lp.device)
+            vals = lp.gather(1, chosen.unsqueeze(1)).squeeze(1).cpu()
+            for j, i in enumerate(want_lp):
+                self.last_logprobs[seqs[i].request_id] = float(vals[j])
+        return sampled
 
     def _apply_penalties(
         self, logits: torch.Tensor, seqs: List[Sequence]

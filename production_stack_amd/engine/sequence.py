@@ -103,3 +103,5 @@ class RequestOutput:
     num_output_tokens: int = 0
     num_cached_tokens: int = 0
     first_token: bool = False
+    # per-token logprob of each entry in new_token_ids (params.logprobs)
+    new_logprobs: Optional[List[float]] = None

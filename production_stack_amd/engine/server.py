@@ -61,7 +61,18 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
         logit_bias={int(k): float(v)
                     for k, v in (body.get("logit_bias") or {}).items()},
         min_tokens=int(body.get("min_tokens") or 0),
+        logprobs=_parse_logprobs(body),
     )
+
+
+def _parse_logprobs(body: dict):
+    """completions: logprobs=<int>; chat: logprobs=true [+ top_logprobs]."""
+    lp = body.get("logprobs")
+    if lp is None or lp is False:
+        return None
+    if lp is True:
+        return int(body.get("top_logprobs") or 1)
+    return int(lp)
 
 
 def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
@@ -504,43 +515,76 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
 
             return StreamingResponse(gen(), media_type="text/event-stream")
 
-        # non-streaming
-        text = ""
-        tokens = []
-        finish_reason = None
-        n_prompt = len(prompt_tokens)
-        kv_params_out = None
-        async for out in async_engine.generate(rid, prompt_tokens, params):
-            text += out.text_delta
-            tokens.extend(out.new_token_ids)
-            if out.finished:
-                finish_reason = out.finish_reason
-                n_prompt = out.num_prompt_tokens or n_prompt
+        # non-streaming (n parallel choices share the prompt's KV via the
+        # prefix cache; each choice is its own engine request)
+        n_choices = max(int(body.get("n") or 1), 1)
+
+        async def run_one(idx: int):
+            import dataclasses as _dc
+
+            p_i = params
+            if idx > 0:
+                p_i = _dc.replace(
+                    params,
+                    seed=(params.seed + idx) if params.seed is not None
+                    else None,
+                )
+            r_i = rid if idx == 0 else f"{rid}-{idx}"
+            text, toks, lps = "", [], []
+            reason, npr = None, len(prompt_tokens)
+            async for out in async_engine.generate(
+                r_i, prompt_tokens, p_i
+            ):
+                text += out.text_delta
+                toks.extend(out.new_token_ids)
+                if out.new_logprobs:
+                    lps.extend(out.new_logprobs)
+                if out.finished:
+                    reason = out.finish_reason
+                    npr = out.num_prompt_tokens or npr
+            return text, toks, lps, reason, npr
+
+        results = await asyncio.gather(
+            *[run_one(i) for i in range(n_choices)]
+        )
+        text, tokens, _, finish_reason, n_prompt = results[0]
+        choices = []
+        for i, (txt, toks, lps, reason, _) in enumerate(results):
+            choice = {
+                "index": i,
+                "finish_reason": reason,
+                **(
+                    {"message": {"role": "assistant", "content": txt}}
+                    if chat
+                    else {"text": txt}
+                ),
+            }
+            if params.logprobs is not None and lps:
+                tok_strs = [engine.tokenizer.decode_token(t) for t in toks]
+                if chat:
+                    choice["logprobs"] = {
+                        "content": [
+                            {"token": ts, "logprob": lp}
+                            for ts, lp in zip(tok_strs, lps)
+                        ]
+                    }
+                else:
+                    choice["logprobs"] = {
+                        "tokens": tok_strs,
+                        "token_logprobs": lps,
+                    }
+            choices.append(choice)
+        total_out = sum(len(r[1]) for r in results)
         resp = {
             "id": rid,
             "object": obj,
             "created": created,
             "model": model_name,
-            "choices": [
-                {
-                    "index": 0,
-                    "finish_reason": finish_reason,
-                    **(
-                        {
-                            "message": {
-                                "role": "assistant",
-                                "content": text,
-                            }
-                        }
-                        if chat
-                        else {"text": text}
-                    ),
-                }
-            ],
+            "choices": choices,
             "usage": {
                 "prompt_tokens": n_prompt,
-                "completion_tokens": len(tokens),
-                "total_tokens": n_prompt + len(tokens),
+                "completion_tokens": total_out,
+                "total_tokens": n_prompt + total_out,
             },
         }
         if kvp.get("do_remote_decode"):

@@ -337,3 +337,33 @@ def test_api_key_auth(monkeypatch):
                     assert r.status_code == 200
 
     asyncio.run(go())
+
+
+def test_n_choices_and_logprobs():
+    async def go(client):
+        r = await client.post(
+            "/v1/completions",
+            json={
+                "model": "tiny-llama",
+                "prompt": "a b c",
+                "max_tokens": 4,
+                "temperature": 0,
+                "ignore_eos": True,
+                "n": 2,
+                "logprobs": 1,
+            },
+            timeout=120,
+        )
+        assert r.status_code == 200
+        d = r.json()
+        assert len(d["choices"]) == 2
+        assert d["choices"][0]["index"] == 0
+        assert d["choices"][1]["index"] == 1
+        # greedy: both choices identical
+        assert d["choices"][0]["text"] == d["choices"][1]["text"]
+        lp = d["choices"][0]["logprobs"]
+        assert len(lp["token_logprobs"]) == 4
+        assert all(v <= 0.0 for v in lp["token_logprobs"])
+        assert d["usage"]["completion_tokens"] == 8
+
+    with_server(go)
