@@ -260,6 +260,13 @@ class SchedulerConfig:
     enable_chunked_prefill: bool = True
     # Cap on how many prompt tokens a single prefill chunk may carry.
     max_prefill_chunk: int = 8192
+    # n-gram (prompt-lookup) speculative decoding: propose up to K draft
+    # tokens per greedy decode by matching the sequence's trailing n-gram
+    # against its own history; the drafts are verified in one multi-token
+    # step through the prefill path. 0 = off.
+    num_speculative_tokens: int = 0
+    ngram_min: int = 2
+    ngram_max: int = 4
 
 
 @dataclass

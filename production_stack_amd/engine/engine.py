@@ -222,6 +222,8 @@ class LLMEngine:
             "gpu_prefix_cache_queries_total": float(bm.prefix_queries),
             "prompt_tokens_total": float(self.stats.prompt_tokens),
             "generation_tokens_total": float(self.stats.generation_tokens),
+            "spec_decode_proposed_total": float(self.runner.spec_proposed),
+            "spec_decode_accepted_total": float(self.runner.spec_accepted),
             **(self.host_pool.metrics() if self.host_pool else {}),
         }
 

@@ -87,6 +87,8 @@ class ModelRunner:
         # request_id -> logprob of the most recent sampled token (only for
         # requests that asked for logprobs; read by engine.step)
         self.last_logprobs: Dict[str, float] = {}
+        self.spec_proposed = 0
+        self.spec_accepted = 0
 
     # ------------------------------------------------------------------
     @property
@@ -219,14 +221,22 @@ class ModelRunner:
             pos_arr = np.arange(start, end, dtype=np.int64)
             bt_arr = np.asarray(seq.block_table, dtype=np.int64)
             slot_arr = bt_arr[pos_arr // bs] * bs + pos_arr % bs
-            tokens.extend(all_ids[start:end])
+            if ss.draft_tokens:
+                # speculative verification chunk: 1 real token + drafts,
+                # logits needed at every position
+                tokens.extend([all_ids[start]] + ss.draft_tokens)
+                base = len(tokens) - ss.num_tokens
+                sample_rows.extend(range(base, len(tokens)))
+                sample_seqs.extend([seq] * ss.num_tokens)
+            else:
+                tokens.extend(all_ids[start:end])
+                if end == seq.num_tokens:
+                    sample_rows.append(len(tokens) - 1)
+                    sample_seqs.append(seq)
             positions.extend(pos_arr.tolist())
             slots.extend(slot_arr.tolist())
             p_token_seq.extend([row] * ss.num_tokens)
             p_token_pos.extend(pos_arr.tolist())
-            if end == seq.num_tokens:
-                sample_rows.append(len(tokens) - 1)
-                sample_seqs.append(seq)
 
         d_seq_lens: List[int] = []
         d_tables: List[List[int]] = []
@@ -445,13 +455,10 @@ class ModelRunner:
                         )
                     }
                     if logits_p is not None:
-                        result.update({
-                            seq.request_id: int(tok)
-                            for seq, tok in zip(
-                                sample_seqs,
-                                self.sample(logits_p, sample_seqs),
-                            )
-                        })
+                        result.update(self._collect_sampled(
+                            SchedulerOutput(scheduled=pre), sample_seqs,
+                            self.sample(logits_p, sample_seqs),
+                        ))
                     return result
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
         if token_t.numel() == 0:
@@ -472,10 +479,7 @@ class ModelRunner:
         sel = hidden[rows_t]
         logits = self.model.compute_logits(sel)
         tokens = self.sample(logits, sample_seqs)
-        return {
-            seq.request_id: int(tok)
-            for seq, tok in zip(sample_seqs, tokens)
-        }
+        return self._collect_sampled(out, sample_seqs, tokens)
 
     @torch.no_grad()
     def _execute_pp_microbatched(
@@ -514,6 +518,39 @@ class ModelRunner:
             seq.request_id: int(tok)
             for seq, tok in zip(all_seqs, tokens)
         }
+
+    def _collect_sampled(self, out, sample_seqs, tokens) -> Dict[str, object]:
+        """Map sampled rows back to requests; speculative chunks contribute
+        their accepted-prefix token list (draft j is accepted when it
+        equals the model's own prediction at the previous position)."""
+        draft_map = {
+            ss.seq.request_id: ss.draft_tokens
+            for ss in out.scheduled
+            if getattr(ss, "draft_tokens", None)
+        }
+        result: Dict[str, object] = {}
+        i = 0
+        n = len(sample_seqs)
+        while i < n:
+            seq = sample_seqs[i]
+            rid = seq.request_id
+            drafts = draft_map.get(rid)
+            if not drafts:
+                result[rid] = int(tokens[i])
+                i += 1
+                continue
+            k1 = len(drafts) + 1
+            row_toks = [int(t) for t in tokens[i : i + k1]]
+            accepted = [row_toks[0]]
+            for j, d in enumerate(drafts):
+                if row_toks[j] != d:
+                    break
+                accepted.append(row_toks[j + 1])
+            self.spec_proposed += len(drafts)
+            self.spec_accepted += len(accepted) - 1
+            result[rid] = accepted
+            i += k1
+        return result
 
     def sample(
         self, logits: torch.Tensor, seqs: List[Sequence]

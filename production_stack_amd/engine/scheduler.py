@@ -25,6 +25,9 @@ from production_stack_amd.engine.sequence import Sequence, SeqStatus
 class ScheduledSeq:
     seq: Sequence
     num_tokens: int  # chunk size this step
+    # speculative drafts verified within this chunk (chunk = 1 real token
+    # + the drafts); empty for normal chunks
+    draft_tokens: List[int] = field(default_factory=list)
 
     @property
     def is_decode(self) -> bool:
@@ -129,6 +132,22 @@ class Scheduler:
             if remaining <= 0:
                 continue
             chunk = min(remaining, budget, self.config.max_prefill_chunk)
+            drafts: List[int] = []
+            if (
+                remaining == 1
+                and self.config.num_speculative_tokens > 0
+                and seq.params.greedy
+                and seq.output_token_ids
+            ):
+                drafts = self._propose_drafts(seq)
+                if drafts and not self.bm.ensure_capacity(
+                    seq, seq.num_computed + 1 + len(drafts)
+                ):
+                    drafts = []  # fall back to plain decode
+                chunk = min(1 + len(drafts), budget)
+                if chunk <= len(drafts):
+                    drafts = drafts[: max(chunk - 1, 0)]
+                    chunk = 1 + len(drafts)
             target = seq.num_computed + chunk
             while not self.bm.ensure_capacity(seq, target):
                 if not self._preempt_last(out, keep=seq):
@@ -149,7 +168,9 @@ class Scheduler:
                     chunk = 0
                     break
             if chunk > 0:
-                out.scheduled.append(ScheduledSeq(seq, chunk))
+                out.scheduled.append(
+                    ScheduledSeq(seq, chunk, draft_tokens=drafts)
+                )
                 budget -= chunk
 
         # 2. admit waiting sequences.
@@ -179,14 +200,46 @@ class Scheduler:
         return out
 
     # ------------------------------------------------------------------
+    def _propose_drafts(self, seq: Sequence) -> List[int]:
+        """Prompt-lookup drafts: the longest trailing n-gram that recurs in
+        the sequence's own history proposes the tokens that followed it.
+        Byte-encoded rfind so the scan is C-speed."""
+        import numpy as np
+
+        k = self.config.num_speculative_tokens
+        k = min(k, self.max_model_len - seq.num_tokens - 1)
+        if k <= 0:
+            return []
+        ids = seq.token_ids()
+        win = ids[-1024:]
+        arr = np.asarray(win, dtype=np.int32).tobytes()
+        for n in range(self.config.ngram_max, self.config.ngram_min - 1, -1):
+            if len(win) <= n:
+                continue
+            pat = np.asarray(win[-n:], dtype=np.int32).tobytes()
+            # search excluding the trailing n-gram itself
+            hay = arr[: -4]  # allow overlap up to the final token
+            idx = hay.rfind(pat)
+            while idx >= 0 and idx % 4 != 0:
+                idx = hay.rfind(pat, 0, idx + len(pat) - 1)
+            if idx < 0:
+                continue
+            start = idx // 4 + n
+            drafts = win[start : start + k]
+            if drafts:
+                return list(drafts)
+        return []
+
     def on_step_done(
         self,
         output: SchedulerOutput,
-        sampled: Dict[str, int],
+        sampled: Dict[str, object],
         eos_token_id: int,
         detok=None,
     ) -> List[Sequence]:
-        """Advance state after the model ran. Returns newly finished seqs."""
+        """Advance state after the model ran. Returns newly finished seqs.
+        A sampled value may be a single token or (speculative decoding) a
+        list of accepted tokens."""
         finished: List[Sequence] = []
         for ss in output.scheduled:
             seq = ss.seq
@@ -195,22 +248,26 @@ class Scheduler:
             seq.num_computed += ss.num_tokens
             tok = sampled.get(seq.request_id)
             if tok is not None:
-                seq.append_token(tok)
-                seq.num_computed = min(seq.num_computed, seq.num_tokens - 1)
+                toks = tok if isinstance(tok, list) else [tok]
                 p = seq.params
-                stop_hit = (
-                    (tok == eos_token_id and not p.ignore_eos)
-                    or tok in p.stop_token_ids
-                )
-                if not stop_hit and p.stop and detok is not None:
-                    tail = detok(seq.output_token_ids[-16:])
-                    stop_hit = any(st in tail for st in p.stop)
-                if stop_hit:
-                    seq.status = SeqStatus.FINISHED_STOPPED
-                elif len(seq.output_token_ids) >= p.max_tokens:
-                    seq.status = SeqStatus.FINISHED_LENGTH
-                elif seq.num_tokens >= self.max_model_len:
-                    seq.status = SeqStatus.FINISHED_LENGTH
+                for t in toks:
+                    seq.append_token(t)
+                    stop_hit = (
+                        (t == eos_token_id and not p.ignore_eos)
+                        or t in p.stop_token_ids
+                    )
+                    if not stop_hit and p.stop and detok is not None:
+                        tail = detok(seq.output_token_ids[-16:])
+                        stop_hit = any(st in tail for st in p.stop)
+                    if stop_hit:
+                        seq.status = SeqStatus.FINISHED_STOPPED
+                    elif len(seq.output_token_ids) >= p.max_tokens:
+                        seq.status = SeqStatus.FINISHED_LENGTH
+                    elif seq.num_tokens >= self.max_model_len:
+                        seq.status = SeqStatus.FINISHED_LENGTH
+                    if seq.finished:
+                        break
+                seq.num_computed = min(seq.num_computed, seq.num_tokens - 1)
             if seq.finished:
                 self.running.remove(seq)
                 self.bm.register_computed_blocks(seq)

@@ -676,6 +676,8 @@ def main() -> None:
                     help="cacheserver data plane host:port (shared KV tier)")
     ap.add_argument("--kv-cache-dtype", default="auto",
                     choices=["auto", "bf16", "fp8", "fp8_e4m3"])
+    ap.add_argument("--num-speculative-tokens", type=int, default=0,
+                    help="n-gram (prompt-lookup) speculative decoding")
     ap.add_argument("--enable-lora", action="store_true",
                     help="enable graph-safe BGMV adapter slots")
     ap.add_argument("--max-loras", type=int, default=4)
@@ -722,6 +724,7 @@ def main() -> None:
             max_num_seqs=args.max_num_seqs,
             max_num_batched_tokens=args.max_num_batched_tokens,
             enable_chunked_prefill=args.enable_chunked_prefill,
+            num_speculative_tokens=args.num_speculative_tokens,
         ),
         enable_lora=args.enable_lora,
         max_loras=args.max_loras,

@@ -189,3 +189,30 @@ def test_mistral_sliding_window_gpu_matches_cpu():
     out_cpu = cpu.generate([prompt], p)["offline-0"]
     agree = sum(a == b for a, b in zip(out_gpu, out_cpu))
     assert agree >= 4, f"{out_gpu} vs {out_cpu}"
+
+
+def test_speculative_ngram_gpu_matches_plain():
+    """Speculative chunks run through the MFMA prefill path on GPU and
+    must reproduce the plain greedy output exactly."""
+    from production_stack_amd.engine.config import SchedulerConfig as SC
+
+    def mk(spec):
+        cfg = EngineConfig(
+            model="mini-llama",
+            max_model_len=1024,
+            seed=2,
+            cache=CacheConfig(num_gpu_blocks=256, block_size=16),
+            scheduler=SC(max_num_seqs=8, max_num_batched_tokens=2048,
+                         num_speculative_tokens=4 if spec else 0),
+        )
+        return LLMEngine(cfg, device="cuda")
+
+    p = SamplingParams(max_tokens=24, temperature=0.0, ignore_eos=True)
+    prompt = [11, 12, 13, 14, 15] * 10
+    plain = mk(False)
+    want = plain.generate([prompt], p)["offline-0"]
+    spec = mk(True)
+    spec.runner.model.load_state_dict(plain.runner.model.state_dict())
+    got = spec.generate([prompt], p)["offline-0"]
+    assert got == want, f"{got} != {want}"
+    assert spec.runner.spec_proposed > 0
