@@ -135,3 +135,71 @@ def test_capacity_stop_when_single_seq_exceeds_cache():
         run_step(sched)
     assert s.finished and s.status is SeqStatus.FINISHED_LENGTH
     assert len(s.output_token_ids) > 0
+
+
+def test_speculative_drafts_respect_capacity_and_model_len():
+    """Draft proposal must clamp at max_model_len and fall back cleanly
+    when KV blocks for the drafts cannot be allocated."""
+    from production_stack_amd.engine.block_manager import BlockManager
+    from production_stack_amd.engine.config import SchedulerConfig
+    from production_stack_amd.engine.scheduler import Scheduler
+    from production_stack_amd.engine.sampling import SamplingParams
+    from production_stack_amd.engine.sequence import Sequence, SeqStatus
+
+    bm = BlockManager(num_blocks=8, block_size=16,
+                      enable_prefix_caching=False)
+    cfg = SchedulerConfig(max_num_seqs=4, max_num_batched_tokens=64,
+                          num_speculative_tokens=4)
+    # model_len barely above the sequence: k must clamp to fit
+    sched = Scheduler(cfg, bm, max_model_len=68)
+    seq = Sequence("s1", [5, 6] * 31,  # 62-token periodic prompt
+                   SamplingParams(max_tokens=8, temperature=0.0,
+                                  ignore_eos=True))
+    sched.add(seq)
+    out = sched.schedule()  # prefill
+    sched.on_step_done(out, {"s1": 5}, eos_token_id=99999)
+    out2 = sched.schedule()
+    ss = out2.scheduled[0]
+    # 63 tokens now; max_model_len 68 -> k <= 68 - 63 - 1 = 4 but the
+    # proposal is also bounded by what fits
+    assert ss.num_tokens - 1 == len(ss.draft_tokens)
+    assert seq.num_computed + ss.num_tokens <= 68
+    sched.on_step_done(out2, {"s1": [6] * min(2, ss.num_tokens)},
+                       eos_token_id=99999)
+    assert seq.num_computed == seq.num_tokens - 1  # decode-ready again
+
+
+def test_speculative_acceptance_burst_bookkeeping():
+    """A fully-accepted draft burst must leave num_computed = num_tokens-1
+    and block accounting consistent."""
+    from production_stack_amd.engine.block_manager import BlockManager
+    from production_stack_amd.engine.config import SchedulerConfig
+    from production_stack_amd.engine.scheduler import Scheduler
+    from production_stack_amd.engine.sampling import SamplingParams
+    from production_stack_amd.engine.sequence import Sequence
+
+    bm = BlockManager(num_blocks=64, block_size=16,
+                      enable_prefix_caching=False)
+    cfg = SchedulerConfig(max_num_seqs=4, max_num_batched_tokens=256,
+                          num_speculative_tokens=3)
+    sched = Scheduler(cfg, bm, max_model_len=512)
+    seq = Sequence("s1", [1, 2, 3, 4] * 10,
+                   SamplingParams(max_tokens=32, temperature=0.0,
+                                  ignore_eos=True))
+    sched.add(seq)
+    out = sched.schedule()
+    sched.on_step_done(out, {"s1": 1}, eos_token_id=99999)
+    out2 = sched.schedule()
+    ss = out2.scheduled[0]
+    assert len(ss.draft_tokens) == 3
+    accepted = [int(t) for t in (2, 3, 4, 1)]  # all drafts accepted
+    sched.on_step_done(out2, {"s1": accepted}, eos_token_id=99999)
+    assert len(seq.output_token_ids) == 1 + 4
+    assert seq.num_computed == seq.num_tokens - 1
+    # drain to completion without accounting drift
+    while sched.has_unfinished():
+        o = sched.schedule()
+        sched.on_step_done(
+            o, {"s1": 7}, eos_token_id=99999
+        )
+    assert bm.num_free == bm.num_blocks
