@@ -107,22 +107,21 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
 }
 
 
-// Wave-M-split variant for M in (64, 256]: the single-wave-per-16-N-rows
-// kernel above reads one A-frag from LDS per MFMA (1:1), which makes it
-// LDS-read-bound once M_TILES grows (measured: 667 GB/s W-stream at
-// M=256 vs 1869 at M=64). Here each wave owns MT m-tiles x 64 N-rows
-// (4 n-tiles held as register B-frags, shared W loads L2-coalesced
-// across the 4 waves), so one LDS A-frag pair feeds 8 MFMAs and the
-// kernel returns to W-streaming-bound.
-template <int MT>  // m-tiles per wave; WG covers 4*MT*16 rows of M
-__global__ __launch_bounds__(256, 2) void
-skinny_gemm_wide_kernel(
+// Wide-N variant for M <= 64: one workgroup covers 256 N-cols (wave w
+// owns cols [n0 + 64w, +64) as 4 register B-frag tiles) and ALL m-tiles.
+// Why: the x chunk is re-staged from L2 by every (n-WG, k-chunk) pair, so
+// x traffic scales as M x N/COLS_PER_WG x K -- at M=256 with 64-col WGs
+// that is 200 MB against 50 MB of weights and the kernel collapses to
+// ~0.6 TB/s. 256-col WGs cut x traffic 4x; each LDS A-frag pair feeds
+// 8 MFMAs (vs 2 in the 64-col kernel), so the W stream is the bound
+// again at decode batch sizes.
+template <int MT>  // m-tiles (= ceil(M/16), M <= 64)
+__global__ __launch_bounds__(256, 4) void skinny_gemm_wide_kernel(
     float* __restrict__ out,               // [M, N] fp32, pre-zeroed
     const unsigned short* __restrict__ x,  // [M, K]
     const unsigned short* __restrict__ w,  // [N, K]
     int M, int N, int K, long x_stride) {
-  constexpr int MP = 4 * MT * 16;
-  const int n0 = blockIdx.x * 64;
+  constexpr int MP = MT * 16;
   const int splits = gridDim.y;
   const int kchunks = K / 64;
   const int per_split = (kchunks + splits - 1) / splits;
@@ -135,6 +134,7 @@ skinny_gemm_wide_kernel(
   const int lane = tid & 63;
   const int g = lane >> 4;
   const int rc = lane & 15;
+  const int n0 = blockIdx.x * 256 + wave * 64;
 
   __shared__ __align__(16) unsigned short x_lds[MP][64];
 
@@ -144,25 +144,18 @@ skinny_gemm_wide_kernel(
 #pragma unroll
     for (int nt = 0; nt < 4; nt++) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
 
-  // T14 software pipeline: x-chunk and B-frags for chunk c+1 load into
-  // registers while chunk c's MFMAs run, so the per-chunk critical path
-  // is max(MFMA, load) instead of their sum.
-  constexpr int XS = MP * 8 / 256;  // staging slots per thread
-  ps_bf16x8 xstage[XS];
-  ps_gbf16x8 b[4][2];
-  auto load_x = [&](int kc) {
-#pragma unroll
-    for (int i = 0; i < XS; i++) {
-      const int u = tid + i * 256;
+  for (int kc = kc_begin; kc < kc_end; kc++) {
+    for (int u = tid; u < MP * 8; u += 256) {
       const int row = u >> 3;
       const int slot = u & 7;
-      xstage[i] = (row < M)
-                      ? *(const ps_bf16x8*)(x + (long)row * x_stride +
-                                            kc * 64 + slot * 8)
-                      : ps_bf16x8{};
+      ps_bf16x8 v = {};
+      if (row < M)
+        v = *(const ps_bf16x8*)(x + (long)row * x_stride + kc * 64 +
+                                slot * 8);
+      *(ps_bf16x8*)(&x_lds[row][(slot ^ (row & 7)) * 8]) = v;
     }
-  };
-  auto load_b = [&](int kc) {
+    __syncthreads();
+    ps_gbf16x8 b[4][2];
 #pragma unroll
     for (int nt = 0; nt < 4; nt++) {
       const unsigned short* wr =
@@ -170,30 +163,10 @@ skinny_gemm_wide_kernel(
       b[nt][0] = ps_as_gbf16(*(const ps_bf16x8*)(wr + g * 8));
       b[nt][1] = ps_as_gbf16(*(const ps_bf16x8*)(wr + 32 + g * 8));
     }
-  };
-  load_x(kc_begin);
-  load_b(kc_begin);
-  for (int kc = kc_begin; kc < kc_end; kc++) {
-#pragma unroll
-    for (int i = 0; i < XS; i++) {
-      const int u = tid + i * 256;
-      const int row = u >> 3;
-      const int slot = u & 7;
-      *(ps_bf16x8*)(&x_lds[row][(slot ^ (row & 7)) * 8]) = xstage[i];
-    }
-    __syncthreads();
-    if (kc + 1 < kc_end) load_x(kc + 1);
-    ps_gbf16x8 bc[4][2];
-#pragma unroll
-    for (int nt = 0; nt < 4; nt++) {
-      bc[nt][0] = b[nt][0];
-      bc[nt][1] = b[nt][1];
-    }
-    if (kc + 1 < kc_end) load_b(kc + 1);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int mt = 0; mt < MT; mt++) {
-      const int row = (wave * MT + mt) * 16 + rc;
+      const int row = mt * 16 + rc;
       ps_gbf16x8 a0 = ps_as_gbf16(
           *(const ps_bf16x8*)(&x_lds[row][(g ^ (row & 7)) * 8]));
       ps_gbf16x8 a1 = ps_as_gbf16(
@@ -201,9 +174,9 @@ skinny_gemm_wide_kernel(
 #pragma unroll
       for (int nt = 0; nt < 4; nt++) {
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a0, bc[nt][0], acc[mt][nt], 0, 0, 0);
+            a0, b[nt][0], acc[mt][nt], 0, 0, 0);
         acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a1, bc[nt][1], acc[mt][nt], 0, 0, 0);
+            a1, b[nt][1], acc[mt][nt], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
@@ -214,7 +187,7 @@ skinny_gemm_wide_kernel(
   for (int mt = 0; mt < MT; mt++) {
 #pragma unroll
     for (int r = 0; r < 4; r++) {
-      const int m = (wave * MT + mt) * 16 + g * 4 + r;
+      const int m = mt * 16 + g * 4 + r;
       if (m < M) {
 #pragma unroll
         for (int nt = 0; nt < 4; nt++)
@@ -240,15 +213,27 @@ int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
   skinny_gemm_kernel<MT><<<grid, block, 0, stream>>>(                       \
       (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
       M, N, K, x_stride)
+  // wide-N path needs N % 256; fall back to the per-64-col kernel else.
+  // Its own split count targets ~512 WGs: atomic-combine traffic scales
+  // with splits, so do not inherit the 64-col grid's deeper split.
+  int n_wg = N / 256;
+  int splits_w = 512 / (n_wg > 0 ? n_wg : 1);
+  if (splits_w < 1) splits_w = 1;
+  if (splits_w > kchunks) splits_w = kchunks;
+  dim3 gridw(N / 256, splits_w);
 #define PS_SGW(MT)                                                          \
-  skinny_gemm_wide_kernel<MT><<<grid, block, 0, stream>>>(                  \
+  skinny_gemm_wide_kernel<MT><<<gridw, block, 0, stream>>>(                 \
       (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
       M, N, K, x_stride)
-  if (M <= 16) PS_SG(1);
-  else if (M <= 32) PS_SG(2);
-  else if (M <= 64) PS_SG(4);
-  else if (M <= 128) PS_SGW(2);
-  else PS_SGW(4);
+  const bool wide_ok = (N % 256 == 0);
+  if (M <= 16) {
+    if (wide_ok) PS_SGW(1); else PS_SG(1);
+  } else if (M <= 32) {
+    if (wide_ok) PS_SGW(2); else PS_SG(2);
+  } else if (M <= 64) {
+    if (wide_ok) PS_SGW(4); else PS_SG(4);
+  } else if (M <= 128) PS_SG(8);
+  else PS_SG(16);
 #undef PS_SG
 #undef PS_SGW
   return 0;
