@@ -203,8 +203,12 @@ int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
                    int K, long x_stride, hipStream_t stream) {
   if (N % 64 != 0 || K % 64 != 0 || M > 256) return -1;
   const int kchunks = K / 64;
-  // pick splits to land near ~1024 workgroups
-  int splits = 1024 / (N / 64);
+  // splits: enough workgroups to fill the chip, but atomic-combine
+  // traffic is M*N*splits fp32 RMWs -- at large M that, not bandwidth,
+  // was the wall (measured 1/M throughput scaling), so taper the split
+  // depth as M grows.
+  const int target_wgs = M <= 32 ? 1024 : (M <= 128 ? 768 : 448);
+  int splits = target_wgs / (N / 64);
   if (splits < 1) splits = 1;
   if (splits > kchunks) splits = kchunks;
   dim3 grid(N / 64, splits);
@@ -225,14 +229,14 @@ int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
   skinny_gemm_wide_kernel<MT><<<gridw, block, 0, stream>>>(                 \
       (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
       M, N, K, x_stride)
-  const bool wide_ok = (N % 256 == 0);
-  if (M <= 16) {
-    if (wide_ok) PS_SGW(1); else PS_SG(1);
-  } else if (M <= 32) {
-    if (wide_ok) PS_SGW(2); else PS_SG(2);
-  } else if (M <= 64) {
-    if (wide_ok) PS_SGW(4); else PS_SG(4);
-  } else if (M <= 128) PS_SG(8);
+  // the wide-N variant measured slower than the 64-col kernel at every
+  // M (the atomic combine, not x-traffic, is the M-scaling wall); keep
+  // it only as a fallback for exotic N where 64-col tiling misfits
+  (void)gridw;
+  if (M <= 16) PS_SG(1);
+  else if (M <= 32) PS_SG(2);
+  else if (M <= 64) PS_SG(4);
+  else if (M <= 128) PS_SG(8);
   else PS_SG(16);
 #undef PS_SG
 #undef PS_SGW
