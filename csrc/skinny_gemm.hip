@@ -207,7 +207,7 @@ int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
   // traffic is M*N*splits fp32 RMWs -- at large M that, not bandwidth,
   // was the wall (measured 1/M throughput scaling), so taper the split
   // depth as M grows.
-  const int target_wgs = M <= 32 ? 1024 : (M <= 128 ? 768 : 448);
+  const int target_wgs = M <= 128 ? 1024 : 448;
   int splits = target_wgs / (N / 64);
   if (splits < 1) splits = 1;
   if (splits > kchunks) splits = kchunks;
