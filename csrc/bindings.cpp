@@ -54,8 +54,9 @@ void ps_kv_quant(void* out, void* scales, const void* in, long rows, int hd,
                  hipStream_t stream);
 void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
                    int hd, hipStream_t stream);
-int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
-                   int K, long x_stride, hipStream_t stream);
+int ps_skinny_gemm_splits(int M, int N, int K);
+int ps_skinny_gemm(void* out_bf16, void* ws, const void* x, const void* w,
+                   int M, int N, int K, long x_stride, hipStream_t stream);
 int ps_lora_bgmv(void* out, const void* x, const void* A, const void* B,
                  const void* scale, const void* idx, int T, int IN, int W,
                  int R, long out_stride, long x_stride, int col_off,
@@ -280,8 +281,8 @@ void reshape_and_cache(at::Tensor k, at::Tensor v, at::Tensor k_cache,
                        BS, kv_fp8, current_stream());
 }
 
-void skinny_gemm(at::Tensor out_f32, at::Tensor x, at::Tensor w) {
-  CHECK_GPU_DTYPE(out_f32, at::kFloat);
+void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w) {
+  CHECK_GPU_BF16(out);
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
               "x must be bf16");
   TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1, "x must be 2D row-major");
@@ -290,14 +291,19 @@ void skinny_gemm(at::Tensor out_f32, at::Tensor x, at::Tensor w) {
   const int K = (int)x.size(1);
   const int N = (int)w.size(0);
   TORCH_CHECK(w.size(1) == K, "K mismatch");
-  // split-K atomicAdd accumulation needs a zeroed output; memset on the
-  // stream is ~100x cheaper than a torch elementwise fill kernel
-  hipMemsetAsync(out_f32.data_ptr(), 0, (size_t)M * N * sizeof(float),
-                 current_stream());
-  int rc = ps_skinny_gemm(out_f32.data_ptr(), x.data_ptr(), w.data_ptr(), M,
-                          N, K, x.stride(0), current_stream());
-  TORCH_CHECK(rc == 0, "unsupported skinny gemm shape M=", M, " N=", N,
+  const int splits = ps_skinny_gemm_splits(M, N, K);
+  TORCH_CHECK(splits > 0, "unsupported skinny gemm shape M=", M, " N=", N,
               " K=", K);
+  void* ws_p = nullptr;
+  at::Tensor ws;
+  if (splits > 1) {
+    ws = at::empty({(long)splits * M * N},
+                   x.options().dtype(at::kFloat));
+    ws_p = ws.data_ptr();
+  }
+  int rc = ps_skinny_gemm(out.data_ptr(), ws_p, x.data_ptr(), w.data_ptr(),
+                          M, N, K, x.stride(0), current_stream());
+  TORCH_CHECK(rc == 0, "skinny gemm launch failed");
 }
 
 void lora_bgmv(at::Tensor out, at::Tensor x, at::Tensor A, at::Tensor B,

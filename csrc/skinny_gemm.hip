@@ -33,9 +33,10 @@ PS_DEV ps_gbf16x8 ps_as_gbf16(ps_bf16x8 u) {
 
 template <int M_TILES>
 __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
-    float* __restrict__ out,             // [M, N] fp32, pre-zeroed
-    const unsigned short* __restrict__ x,  // [M, K]
-    const unsigned short* __restrict__ w,  // [N, K]
+    unsigned short* __restrict__ out_bf16,  // [M, N] (splits == 1)
+    float* __restrict__ ws,                 // [S, M, N] (splits > 1)
+    const unsigned short* __restrict__ x,   // [M, K]
+    const unsigned short* __restrict__ w,   // [N, K]
     int M, int N, int K, long x_stride /* row stride of x, elems */) {
   constexpr int MP = M_TILES * 16;  // padded M
   const int n0 = blockIdx.x * 64;
@@ -93,153 +94,75 @@ __global__ __launch_bounds__(256, 4) void skinny_gemm_kernel(
     __syncthreads();
   }
 
-  // ---- combine: atomicAdd partials into out (C layout: row=(g*4+r),
-  // col=rc of each 16x16 tile) ----
+  // ---- epilogue (C layout: row=(g*4+r), col=rc of each 16x16 tile).
+  // splits==1 writes bf16 output directly; otherwise each split streams
+  // its fp32 partial to ws[split][M][N] and a combine kernel reduces —
+  // device atomicAdd RMW traffic (M*N*splits) was the measured 1/M
+  // throughput wall of the earlier design.
 #pragma unroll
   for (int mt = 0; mt < M_TILES; mt++) {
 #pragma unroll
     for (int r = 0; r < 4; r++) {
       const int m = mt * 16 + g * 4 + r;
-      if (m < M)
-        atomicAdd(&out[(long)m * N + nrow], acc[mt][r]);
-    }
-  }
-}
-
-
-// Wide-N variant for M <= 64: one workgroup covers 256 N-cols (wave w
-// owns cols [n0 + 64w, +64) as 4 register B-frag tiles) and ALL m-tiles.
-// Why: the x chunk is re-staged from L2 by every (n-WG, k-chunk) pair, so
-// x traffic scales as M x N/COLS_PER_WG x K -- at M=256 with 64-col WGs
-// that is 200 MB against 50 MB of weights and the kernel collapses to
-// ~0.6 TB/s. 256-col WGs cut x traffic 4x; each LDS A-frag pair feeds
-// 8 MFMAs (vs 2 in the 64-col kernel), so the W stream is the bound
-// again at decode batch sizes.
-template <int MT>  // m-tiles (= ceil(M/16), M <= 64)
-__global__ __launch_bounds__(256, 4) void skinny_gemm_wide_kernel(
-    float* __restrict__ out,               // [M, N] fp32, pre-zeroed
-    const unsigned short* __restrict__ x,  // [M, K]
-    const unsigned short* __restrict__ w,  // [N, K]
-    int M, int N, int K, long x_stride) {
-  constexpr int MP = MT * 16;
-  const int splits = gridDim.y;
-  const int kchunks = K / 64;
-  const int per_split = (kchunks + splits - 1) / splits;
-  const int kc_begin = blockIdx.y * per_split;
-  const int kc_end = min(kchunks, kc_begin + per_split);
-  if (kc_begin >= kc_end) return;
-
-  const int tid = threadIdx.x;
-  const int wave = tid >> 6;
-  const int lane = tid & 63;
-  const int g = lane >> 4;
-  const int rc = lane & 15;
-  const int n0 = blockIdx.x * 256 + wave * 64;
-
-  __shared__ __align__(16) unsigned short x_lds[MP][64];
-
-  ps_gf32x4 acc[MT][4];
-#pragma unroll
-  for (int mt = 0; mt < MT; mt++)
-#pragma unroll
-    for (int nt = 0; nt < 4; nt++) acc[mt][nt] = {0.f, 0.f, 0.f, 0.f};
-
-  for (int kc = kc_begin; kc < kc_end; kc++) {
-    for (int u = tid; u < MP * 8; u += 256) {
-      const int row = u >> 3;
-      const int slot = u & 7;
-      ps_bf16x8 v = {};
-      if (row < M)
-        v = *(const ps_bf16x8*)(x + (long)row * x_stride + kc * 64 +
-                                slot * 8);
-      *(ps_bf16x8*)(&x_lds[row][(slot ^ (row & 7)) * 8]) = v;
-    }
-    __syncthreads();
-    ps_gbf16x8 b[4][2];
-#pragma unroll
-    for (int nt = 0; nt < 4; nt++) {
-      const unsigned short* wr =
-          w + (long)(n0 + nt * 16 + rc) * K + kc * 64;
-      b[nt][0] = ps_as_gbf16(*(const ps_bf16x8*)(wr + g * 8));
-      b[nt][1] = ps_as_gbf16(*(const ps_bf16x8*)(wr + 32 + g * 8));
-    }
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int mt = 0; mt < MT; mt++) {
-      const int row = mt * 16 + rc;
-      ps_gbf16x8 a0 = ps_as_gbf16(
-          *(const ps_bf16x8*)(&x_lds[row][(g ^ (row & 7)) * 8]));
-      ps_gbf16x8 a1 = ps_as_gbf16(
-          *(const ps_bf16x8*)(&x_lds[row][((4 + g) ^ (row & 7)) * 8]));
-#pragma unroll
-      for (int nt = 0; nt < 4; nt++) {
-        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a0, b[nt][0], acc[mt][nt], 0, 0, 0);
-        acc[mt][nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a1, b[nt][1], acc[mt][nt], 0, 0, 0);
-      }
-    }
-    __builtin_amdgcn_s_setprio(0);
-    __syncthreads();
-  }
-
-#pragma unroll
-  for (int mt = 0; mt < MT; mt++) {
-#pragma unroll
-    for (int r = 0; r < 4; r++) {
-      const int m = mt * 16 + g * 4 + r;
-      if (m < M) {
-#pragma unroll
-        for (int nt = 0; nt < 4; nt++)
-          atomicAdd(&out[(long)m * N + n0 + nt * 16 + rc], acc[mt][nt][r]);
+      if (m >= M) continue;
+      if (gridDim.y == 1) {
+        out_bf16[(long)m * N + nrow] = ps_f32_to_bf16(acc[mt][r]);
+      } else {
+        ws[((long)blockIdx.y * M + m) * N + nrow] = acc[mt][r];
       }
     }
   }
 }
+
+__global__ __launch_bounds__(256) void skinny_combine_kernel(
+    unsigned short* __restrict__ out,  // [M, N] bf16
+    const float* __restrict__ ws,      // [S, M, N] fp32
+    long MN, int splits) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i >= MN) return;
+  float acc = 0.f;
+  for (int s = 0; s < splits; s++) acc += ws[(long)s * MN + i];
+  out[i] = ps_f32_to_bf16(acc);
+}
+
 
 extern "C" {
 
-int ps_skinny_gemm(void* out_f32, const void* x, const void* w, int M, int N,
-                   int K, long x_stride, hipStream_t stream) {
+// ws == nullptr forces splits = 1. ps_skinny_gemm_splits reports the
+// split depth the heuristic will pick for a shape so the caller can size
+// the fp32 workspace.
+int ps_skinny_gemm_splits(int M, int N, int K) {
   if (N % 64 != 0 || K % 64 != 0 || M > 256) return -1;
   const int kchunks = K / 64;
-  // splits: enough workgroups to fill the chip, but atomic-combine
-  // traffic is M*N*splits fp32 RMWs -- at large M that, not bandwidth,
-  // was the wall (measured 1/M throughput scaling), so taper the split
-  // depth as M grows.
-  const int target_wgs = M <= 128 ? 1024 : 448;
+  const int target_wgs = 1024;
   int splits = target_wgs / (N / 64);
   if (splits < 1) splits = 1;
   if (splits > kchunks) splits = kchunks;
+  return splits;
+}
+
+int ps_skinny_gemm(void* out_bf16, void* ws, const void* x, const void* w,
+                   int M, int N, int K, long x_stride, hipStream_t stream) {
+  int splits = ps_skinny_gemm_splits(M, N, K);
+  if (splits < 0) return -1;
+  if (ws == nullptr) splits = 1;
   dim3 grid(N / 64, splits);
   dim3 block(256);
 #define PS_SG(MT)                                                           \
   skinny_gemm_kernel<MT><<<grid, block, 0, stream>>>(                       \
-      (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
-      M, N, K, x_stride)
-  // wide-N path needs N % 256; fall back to the per-64-col kernel else.
-  // Its own split count targets ~512 WGs: atomic-combine traffic scales
-  // with splits, so do not inherit the 64-col grid's deeper split.
-  int n_wg = N / 256;
-  int splits_w = 512 / (n_wg > 0 ? n_wg : 1);
-  if (splits_w < 1) splits_w = 1;
-  if (splits_w > kchunks) splits_w = kchunks;
-  dim3 gridw(N / 256, splits_w);
-#define PS_SGW(MT)                                                          \
-  skinny_gemm_wide_kernel<MT><<<gridw, block, 0, stream>>>(                 \
-      (float*)out_f32, (const unsigned short*)x, (const unsigned short*)w,  \
-      M, N, K, x_stride)
-  // the wide-N variant measured slower than the 64-col kernel at every
-  // M (the atomic combine, not x-traffic, is the M-scaling wall); keep
-  // it only as a fallback for exotic N where 64-col tiling misfits
-  (void)gridw;
+      (unsigned short*)out_bf16, (float*)ws, (const unsigned short*)x,      \
+      (const unsigned short*)w, M, N, K, x_stride)
   if (M <= 16) PS_SG(1);
   else if (M <= 32) PS_SG(2);
   else if (M <= 64) PS_SG(4);
   else if (M <= 128) PS_SG(8);
   else PS_SG(16);
 #undef PS_SG
-#undef PS_SGW
+  if (splits > 1) {
+    const long MN = (long)M * N;
+    skinny_combine_kernel<<<dim3((MN + 255) / 256), 256, 0, stream>>>(
+        (unsigned short*)out_bf16, (const float*)ws, MN, splits);
+  }
   return 0;
 }
 

@@ -221,10 +221,10 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     where hipBLASLt tiles poorly — see csrc/skinny_gemm.hip). GPU only."""
     _require_ext()
     out = torch.empty(
-        (x.shape[0], w.shape[0]), dtype=torch.float32, device=x.device
+        (x.shape[0], w.shape[0]), dtype=torch.bfloat16, device=x.device
     )
-    _C.skinny_gemm(out, x, w)  # zeroes out via hipMemsetAsync internally
-    return out.to(x.dtype)
+    _C.skinny_gemm(out, x, w)  # split-K partials + combine (no atomics)
+    return out
 
 
 def lora_bgmv(
