@@ -87,6 +87,8 @@ def main() -> None:
     ap.add_argument("--device", default=None)
     ap.add_argument("--profile-host", action="store_true",
                     help="cProfile the measured loop; report to stderr")
+    ap.add_argument("--no-async-scheduling", action="store_true",
+                    help="disable one-step-lagged sampling")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -106,6 +108,7 @@ def main() -> None:
     cfg = EngineConfig(
         model=args.model,
         max_model_len=args.max_model_len,
+        async_scheduling=not args.no_async_scheduling,
         seed=1234 + rank,
         cache=CacheConfig(
             block_size=16,

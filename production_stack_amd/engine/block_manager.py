@@ -168,7 +168,12 @@ class BlockManager:
             done = seq.num_cached_prompt_tokens // bs
         for i in range(done, n_full):
             blk = seq.block_table[i]
-            h = self.chain_hash(prev, tuple(tokens[i * bs : (i + 1) * bs]))
+            chunk = tokens[i * bs : (i + 1) * bs]
+            if any(t < 0 for t in chunk):
+                # async placeholder not yet resolved: register next step
+                n_full = i
+                break
+            h = self.chain_hash(prev, tuple(chunk))
             if h not in self.cached and self.block_hash[blk] is None:
                 self.cached[h] = blk
                 self.block_hash[blk] = h

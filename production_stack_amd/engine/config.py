@@ -293,6 +293,12 @@ class EngineConfig:
     enforce_eager: bool = False  # True disables hipGraph decode capture
     # graph-safe batched LoRA (BGMV slots); when False, adapters still run
     # through the eager grouped path
+    # one-step-lagged sampling: the host schedules/launches step N+1 while
+    # step N's sampled tokens are still in flight (decode inputs gather
+    # from the previous step's device tensor). Greedy-exact; steps with
+    # non-greedy sampling, penalties, logprobs or speculative chunks fall
+    # back to the synchronous path transparently.
+    async_scheduling: bool = False
     enable_lora: bool = False
     max_loras: int = 4
     max_lora_rank: int = 16

@@ -83,6 +83,7 @@ class LLMEngine:
         )
         self.stats = EngineStats()
         self._sleeping = False
+        self._pending = None  # async-scheduling in-flight step
         self.lora_adapters: Dict[str, object] = {}
         self.runner.lora_registry = self.lora_adapters
         # coarse lock: engine-loop steps vs out-of-band block mutations
@@ -117,14 +118,57 @@ class LLMEngine:
         self.scheduler.abort(request_id)
 
     def has_unfinished(self) -> bool:
-        return self.scheduler.has_unfinished()
+        return self.scheduler.has_unfinished() or self._pending is not None
 
     # ------------------------------------------------------------------
     def step(self) -> List[RequestOutput]:
         with self.lock:
             return self._step_locked()
 
+    # ---- async scheduling (one-step-lagged sampling) -----------------
+    def _step_locked_async(self) -> List[RequestOutput]:
+        out = self.scheduler.schedule()
+        if out.is_empty and not out.capacity_stopped:
+            return self._finalize_pending()
+        if self.host_pool is not None:
+            self.host_pool.make_compute_wait()
+        handle = self.runner.execute_async(out, self.block_manager)
+        if handle is None:
+            # resolve in-flight placeholders, then take the sync path for
+            # this step (prepare must see real token values)
+            prev = self._finalize_pending()
+            sampled = self.runner.execute(out, self.block_manager)
+            fin = self.scheduler.on_step_done(
+                out, sampled, self.model_cfg.eos_token_id,
+                detok=self.tokenizer.decode,
+            )
+            self.stats.num_finished += len(fin)
+            return prev + self._build_outputs(out)
+        ph_idx = self.scheduler.advance_async(
+            out, {s.request_id for s in handle["sample_seqs"]}
+        )
+        prev = self._finalize_pending()
+        self._pending = (handle, out, ph_idx)
+        return prev + self._build_outputs(
+            out, include_scheduled=False
+        )
+
+    def _finalize_pending(self) -> List[RequestOutput]:
+        if self._pending is None:
+            return []
+        handle, p_out, ph_idx = self._pending
+        self._pending = None
+        sampled = self.runner.finalize_async(handle)
+        finished = self.scheduler.finalize_async(
+            sampled, ph_idx, self.model_cfg.eos_token_id,
+            detok=self.tokenizer.decode,
+        )
+        self.stats.num_finished += len(finished)
+        return self._build_outputs(p_out, count_finished=False)
+
     def _step_locked(self) -> List[RequestOutput]:
+        if self.config.async_scheduling:
+            return self._step_locked_async()
         out = self.scheduler.schedule()
         if out.is_empty and not out.capacity_stopped:
             return []
@@ -135,6 +179,13 @@ class LLMEngine:
             out, sampled, self.model_cfg.eos_token_id,
             detok=self.tokenizer.decode,
         )
+        self.stats.num_finished += len(finished)
+        return self._build_outputs(out)
+
+    def _build_outputs(
+        self, out, include_scheduled: bool = True,
+        count_finished: bool = True,
+    ) -> List[RequestOutput]:
         now = time.time()
         results: List[RequestOutput] = []
         for seq in out.capacity_stopped:
@@ -151,6 +202,9 @@ class LLMEngine:
                 )
             )
             self.stats.num_finished += 1
+        if not include_scheduled:
+            out.capacity_stopped = []  # don't re-emit on a later build
+            return results
         for ss in out.scheduled:
             seq = ss.seq
             if seq.status is SeqStatus.PREEMPTED:
@@ -191,7 +245,6 @@ class LLMEngine:
                     new_logprobs=lps,
                 )
             )
-        self.stats.num_finished += len(finished)
         return results
 
     # ------------------------------------------------------------------

@@ -134,6 +134,7 @@ class DecodeGraphRunner:
         seq_lens: np.ndarray,
         block_tables: List[List[int]],
         lora_idx: Optional[np.ndarray] = None,
+        prev_fill=None,  # (rows, src_rows, prev_sampled_dev) async gather
     ) -> torch.Tensor:
         n = len(tokens)
         b = self.bucket_for(n)
@@ -166,5 +167,10 @@ class DecodeGraphRunner:
         self.block_tables[:b].copy_(
             self.h_block_tables[:b], non_blocking=True
         )
+        if prev_fill is not None:
+            rows, src, prev_dev = prev_fill
+            ridx = torch.tensor(rows, dtype=torch.long, device=self.device)
+            sidx = torch.tensor(src, dtype=torch.long, device=self.device)
+            self.tokens[ridx] = prev_dev[sidx]
         self.graphs[b].replay()
         return self.logits[b][:n]

@@ -89,6 +89,11 @@ class ModelRunner:
         self.last_logprobs: Dict[str, float] = {}
         self.spec_proposed = 0
         self.spec_accepted = 0
+        # async-scheduling state (execute_async/finalize_async)
+        self._prev_tokens_dev = None
+        self._prev_rows: Dict[str, int] = {}
+        self._async_host = [None, None]
+        self._async_host_i = 0
 
     # ------------------------------------------------------------------
     @property
@@ -383,10 +388,20 @@ class ModelRunner:
         lora_idx = None
         if self.lora_slots is not None:
             lora_idx = np.full(n, -1, dtype=np.int32)
+        ph_rows: List[int] = []
+        ph_src: List[int] = []
         for i, ss in enumerate(scheduled):
             seq = ss.seq
             pos = seq.num_computed
-            tokens[i] = seq.token_ids()[pos]
+            tok = seq.token_ids()[pos]
+            if tok < 0:  # async placeholder: gather from prev device toks
+                row = self._prev_rows.get(seq.request_id)
+                if row is None or self._prev_tokens_dev is None:
+                    return None
+                ph_rows.append(i)
+                ph_src.append(row)
+                tok = 0
+            tokens[i] = tok
             positions[i] = pos
             slots[i] = seq.block_table[pos // bs] * bs + pos % bs
             seq_lens[i] = pos + 1
@@ -396,8 +411,11 @@ class ModelRunner:
                 lora_idx[i] = self.lora_slots.slot_by_name.get(
                     seq.lora_name, -1
                 )
+        prev_fill = None
+        if ph_rows:
+            prev_fill = (ph_rows, ph_src, self._prev_tokens_dev)
         logits = self.graphs.run(tokens, positions, slots, seq_lens, tables,
-                                 lora_idx)
+                                 lora_idx, prev_fill)
         if defer_sample:
             return logits, seqs
         sampled = self.sample(logits, seqs)
@@ -519,6 +537,122 @@ class ModelRunner:
             for seq, tok in zip(all_seqs, tokens)
         }
 
+    @torch.no_grad()
+    def execute_async(self, out: SchedulerOutput, bm: BlockManager):
+        """Async-scheduling fast path: launch this step's GPU work and
+        device-side greedy sampling WITHOUT any host sync; returns a
+        handle the engine finalizes on the next step (one-step-lagged
+        pipeline), or None when the step needs the sync path (non-greedy
+        sampling, speculative chunks, logprobs, TP/PP, no graphs).
+
+        Decode rows whose input token is still the -1 placeholder gather
+        it on-device from the previous step's sampled tensor."""
+        if (
+            self.pipeline is not None
+            or getattr(self, "tp_serving", False)
+            or self.lora_slots is not None and any(
+                s.seq.lora_name for s in out.scheduled)
+        ):
+            return None
+        for ss in out.scheduled:
+            p = ss.seq.params
+            if not p.greedy or p.logprobs is not None or p.needs_penalties:
+                return None
+            if ss.draft_tokens:
+                return None
+        dec = [ss for ss in out.scheduled if ss.is_decode]
+        pre = [ss for ss in out.scheduled if not ss.is_decode]
+        logits_d, seqs_d = None, []
+        use_graph = (
+            self.graphs is not None and dec
+            and self.graphs.bucket_for(len(dec)) is not None
+        )
+        if use_graph:
+            got = self._execute_decode_graph(
+                SchedulerOutput(scheduled=dec), bm, defer_sample=True
+            )
+            if got is None:
+                use_graph = False
+            else:
+                logits_d, seqs_d = got
+        eager = pre if use_graph else out.scheduled
+        logits_p, seqs_p = None, []
+        if eager:
+            sub = SchedulerOutput(scheduled=list(eager))
+            token_t, meta, seqs_p, rows_t = self.prepare(sub, bm)
+            if token_t.numel():
+                # resolve -1 placeholders (decode rows sit after the
+                # prefill-chunk rows) from the prev step's device tokens
+                prefill_rows = sum(
+                    ss.num_tokens for ss in eager if not ss.is_decode
+                )
+                decode_rids = [
+                    ss.seq.request_id for ss in eager if ss.is_decode
+                ]
+                ph = (token_t < 0).nonzero().flatten()
+                if ph.numel():
+                    if self._prev_tokens_dev is None:
+                        return None
+                    src = []
+                    for i in ph.tolist():
+                        row = self._prev_rows.get(
+                            decode_rids[i - prefill_rows]
+                        )
+                        if row is None:
+                            return None
+                        src.append(row)
+                    gather = torch.tensor(src, dtype=torch.long,
+                                          device=token_t.device)
+                    token_t[ph] = self._prev_tokens_dev[gather]
+                hidden = self.model(token_t, meta, self.kv_caches)
+                if seqs_p:
+                    logits_p = self.model.compute_logits(hidden[rows_t])
+        sample_seqs = list(seqs_d) + list(seqs_p)
+        if not sample_seqs:
+            return None
+        parts = []
+        if logits_d is not None and len(seqs_d):
+            parts.append(ops.greedy_sample(logits_d))
+        if logits_p is not None and len(seqs_p):
+            parts.append(ops.greedy_sample(logits_p))
+        sampled_dev = parts[0] if len(parts) == 1 else torch.cat(parts)
+        # async D2H into pinned staging; two buffers alternate so the
+        # in-flight previous step's values survive until its finalize
+        self._async_host_i ^= 1
+        host = self._async_host[self._async_host_i]
+        if host is None or host.numel() < sampled_dev.numel():
+            host = torch.empty(
+                max(sampled_dev.numel(), 512), dtype=torch.long,
+                pin_memory=(self.device.type == "cuda"),
+            )
+            self._async_host[self._async_host_i] = host
+        host[: sampled_dev.numel()].copy_(sampled_dev, non_blocking=True)
+        ev = None
+        if self.device.type == "cuda":
+            ev = torch.cuda.Event()
+            ev.record()
+        self._prev_tokens_dev = sampled_dev
+        self._prev_rows = {
+            seq.request_id: i for i, seq in enumerate(sample_seqs)
+        }
+        return {
+            "out": out,
+            "sample_seqs": sample_seqs,
+            "host": host,
+            "n": sampled_dev.numel(),
+            "event": ev,
+        }
+
+    def finalize_async(self, handle) -> Dict[str, int]:
+        """Wait for the handle's D2H and return request_id -> token."""
+        if handle["event"] is not None:
+            handle["event"].synchronize()
+        vals = handle["host"][: handle["n"]].tolist()
+        return {
+            seq.request_id: int(t)
+            for seq, t in zip(handle["sample_seqs"], vals)
+        }
+
     def _collect_sampled(self, out, sample_seqs, tokens) -> Dict[str, object]:
         """Map sampled rows back to requests; speculative chunks contribute
         their accepted-prefix token list (draft j is accepted when it
@@ -587,7 +721,7 @@ class ModelRunner:
             if not p.needs_penalties:
                 continue
             row = logits[i]
-            out_ids = seq.output_token_ids
+            out_ids = [t for t in seq.output_token_ids if t >= 0]
             if out_ids and (p.presence_penalty or p.frequency_penalty
                             or p.repetition_penalty != 1.0):
                 ids = torch.tensor(out_ids, dtype=torch.long,

@@ -138,6 +138,7 @@ class Scheduler:
                 and self.config.num_speculative_tokens > 0
                 and seq.params.greedy
                 and seq.output_token_ids
+                and seq.output_token_ids[-1] >= 0
             ):
                 drafts = self._propose_drafts(seq)
                 if drafts and not self.bm.ensure_capacity(
@@ -276,4 +277,72 @@ class Scheduler:
                 finished.append(seq)
             else:
                 self.bm.register_computed_blocks(seq)
+        return finished
+
+    # ---- async scheduling (one-step-lagged sampling) -----------------
+    def advance_async(self, output: SchedulerOutput,
+                      sampled_rids) -> Dict[str, int]:
+        """Bookkeep a launched-but-unsampled step: advance num_computed
+        and append a -1 placeholder for every row that will receive a
+        token. Returns request_id -> placeholder index (into
+        output_token_ids) for finalize_async."""
+        ph_idx: Dict[str, int] = {}
+        for ss in output.scheduled:
+            seq = ss.seq
+            if seq.finished or seq.status is SeqStatus.PREEMPTED:
+                continue
+            seq.num_computed += ss.num_tokens
+            if seq.request_id in sampled_rids:
+                seq.append_token(-1)
+                ph_idx[seq.request_id] = len(seq.output_token_ids) - 1
+                seq.num_computed = min(seq.num_computed,
+                                       seq.num_tokens - 1)
+            self.bm.register_computed_blocks(seq)
+        return ph_idx
+
+    def finalize_async(self, sampled: Dict[str, int],
+                       ph_idx: Dict[str, int], eos_token_id: int,
+                       detok=None) -> List[Sequence]:
+        """Resolve placeholders with the now-arrived tokens; run the stop
+        checks that were deferred. A stop truncates any newer in-flight
+        placeholder (its step's work is simply discarded)."""
+        finished: List[Sequence] = []
+        for rid, tok in sampled.items():
+            seq = self._by_id.get(rid)
+            if seq is None:
+                continue  # aborted while in flight
+            idx = ph_idx.get(rid)
+            if idx is None or idx >= len(seq.output_token_ids):
+                continue
+            seq.output_token_ids[idx] = tok
+            if seq.finished:
+                continue
+            p = seq.params
+            stop_hit = (
+                (tok == eos_token_id and not p.ignore_eos)
+                or tok in p.stop_token_ids
+            )
+            if not stop_hit and p.stop and detok is not None:
+                real = [t for t in seq.output_token_ids[: idx + 1]][-16:]
+                stop_hit = any(st in detok(real) for st in p.stop)
+            n_real = idx + 1
+            if stop_hit:
+                seq.status = SeqStatus.FINISHED_STOPPED
+            elif n_real >= p.max_tokens:
+                seq.status = SeqStatus.FINISHED_LENGTH
+            elif seq.num_prompt + n_real >= self.max_model_len:
+                seq.status = SeqStatus.FINISHED_LENGTH
+            if seq.finished:
+                # drop any newer in-flight placeholder tokens
+                if len(seq.output_token_ids) > n_real:
+                    del seq.output_token_ids[n_real:]
+                    seq.num_computed = min(
+                        seq.num_computed, seq.num_tokens - 1
+                    )
+                if seq in self.running:
+                    self.running.remove(seq)
+                self.bm.register_computed_blocks(seq)
+                self.bm.free_seq(seq)
+                self._by_id.pop(seq.request_id, None)
+                finished.append(seq)
         return finished

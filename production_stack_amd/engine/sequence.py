@@ -80,7 +80,12 @@ class Sequence:
 
     def drain_new_tokens(self) -> List[int]:
         new = self.output_token_ids[self._stream_cursor :]
-        self._stream_cursor = len(self.output_token_ids)
+        # async scheduling: never stream an unresolved -1 placeholder
+        for j, t in enumerate(new):
+            if t < 0:
+                new = new[:j]
+                break
+        self._stream_cursor += len(new)
         return new
 
     def reset_for_recompute(self) -> None:

@@ -676,6 +676,8 @@ def main() -> None:
                     help="cacheserver data plane host:port (shared KV tier)")
     ap.add_argument("--kv-cache-dtype", default="auto",
                     choices=["auto", "bf16", "fp8", "fp8_e4m3"])
+    ap.add_argument("--async-scheduling", action="store_true",
+                    help="one-step-lagged sampling (greedy-exact overlap)")
     ap.add_argument("--num-speculative-tokens", type=int, default=0,
                     help="n-gram (prompt-lookup) speculative decoding")
     ap.add_argument("--enable-lora", action="store_true",
@@ -726,6 +728,7 @@ def main() -> None:
             enable_chunked_prefill=args.enable_chunked_prefill,
             num_speculative_tokens=args.num_speculative_tokens,
         ),
+        async_scheduling=args.async_scheduling,
         enable_lora=args.enable_lora,
         max_loras=args.max_loras,
         max_lora_rank=args.max_lora_rank,

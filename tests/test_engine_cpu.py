@@ -335,3 +335,80 @@ def test_speculative_respects_stops_and_length():
     p = SamplingParams(max_tokens=7, temperature=0.0, ignore_eos=True)
     out = eng.generate([[1, 2, 3] * 6], p)["offline-0"]
     assert len(out) == 7  # acceptance bursts must not overshoot max_tokens
+
+
+def test_async_scheduling_matches_sync_exactly():
+    """One-step-lagged sampling must be invisible: identical greedy
+    outputs (including EOS/stop behaviour) vs the synchronous engine."""
+    def mk(async_on):
+        cfg = EngineConfig(
+            model="tiny-llama",
+            max_model_len=256,
+            seed=5,
+            async_scheduling=async_on,
+            cache=CacheConfig(num_gpu_blocks=128, block_size=16),
+            scheduler=SchedulerConfig(max_num_seqs=8,
+                                      max_num_batched_tokens=128),
+        )
+        return LLMEngine(cfg, device="cpu")
+
+    prompts = [list(range(10, 60)), list(range(200, 230)), [7, 8, 9]]
+    for params in (
+        SamplingParams(max_tokens=10, temperature=0.0, ignore_eos=True),
+        SamplingParams(max_tokens=40, temperature=0.0),  # EOS may fire
+    ):
+        sync = mk(False)
+        want = sync.generate(prompts, params)
+        a = mk(True)
+        a.runner.model.load_state_dict(sync.runner.model.state_dict())
+        got = a.generate(prompts, params)
+        assert got == want, f"{got} != {want}"
+
+
+def test_async_scheduling_nongreedy_fallback():
+    """Non-greedy requests silently take the synchronous path."""
+    cfg = EngineConfig(
+        model="tiny-llama",
+        max_model_len=256,
+        seed=5,
+        async_scheduling=True,
+        cache=CacheConfig(num_gpu_blocks=128, block_size=16),
+        scheduler=SchedulerConfig(max_num_seqs=8,
+                                  max_num_batched_tokens=128),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    p = SamplingParams(max_tokens=6, temperature=0.8, top_p=0.9,
+                       ignore_eos=True)
+    out = eng.generate([[1, 2, 3, 4]], p)["offline-0"]
+    assert len(out) == 6
+    bm = eng.block_manager
+    assert bm.num_free == bm.num_blocks
+
+
+def test_async_scheduling_preemption_consistency():
+    """Tight KV cache forces preemption while a step is in flight; async
+    must still match sync exactly."""
+    def mk(async_on):
+        cfg = EngineConfig(
+            model="tiny-llama",
+            max_model_len=256,
+            seed=9,
+            async_scheduling=async_on,
+            cache=CacheConfig(num_gpu_blocks=24, block_size=16,
+                              enable_prefix_caching=False),
+            scheduler=SchedulerConfig(max_num_seqs=4,
+                                      max_num_batched_tokens=64),
+        )
+        return LLMEngine(cfg, device="cpu")
+
+    prompts = [list(range(5, 85)), list(range(100, 170)),
+               list(range(300, 350))]
+    p = SamplingParams(max_tokens=12, temperature=0.0, ignore_eos=True)
+    sync = mk(False)
+    want = sync.generate(prompts, p)
+    a = mk(True)
+    a.runner.model.load_state_dict(sync.runner.model.state_dict())
+    got = a.generate(prompts, p)
+    assert got == want, f"{got} != {want}"
+    bm = a.block_manager
+    assert bm.num_free == bm.num_blocks

@@ -216,3 +216,16 @@ def test_speculative_ngram_gpu_matches_plain():
     got = spec.generate([prompt], p)["offline-0"]
     assert got == want, f"{got} != {want}"
     assert spec.runner.spec_proposed > 0
+
+
+def test_async_scheduling_gpu_matches_sync():
+    """Async (one-step-lagged) scheduling on GPU: decode graphs + device
+    token gather must reproduce the synchronous outputs exactly."""
+    p = SamplingParams(max_tokens=16, temperature=0.0, ignore_eos=True)
+    prompts = [list(range(100, 180)), list(range(7, 40)), [5, 6, 7]]
+    sync = make_engine()
+    want = sync.generate(prompts, p)
+    a = make_engine(async_scheduling=True)
+    a.runner.model.load_state_dict(sync.runner.model.state_dict())
+    got = a.generate(prompts, p)
+    assert got == want, f"{got} != {want}"
