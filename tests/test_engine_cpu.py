@@ -412,3 +412,58 @@ def test_async_scheduling_preemption_consistency():
     assert got == want, f"{got} != {want}"
     bm = a.block_manager
     assert bm.num_free == bm.num_blocks
+
+
+def test_async_scheduling_abort_and_chaos():
+    """Aborting requests while a step is in flight must not corrupt
+    accounting; bounded add/abort chaos drains clean."""
+    import threading
+    import time as _t
+
+    cfg = EngineConfig(
+        model="tiny-llama",
+        max_model_len=256,
+        seed=3,
+        async_scheduling=True,
+        cache=CacheConfig(num_gpu_blocks=128, block_size=16),
+        scheduler=SchedulerConfig(max_num_seqs=8,
+                                  max_num_batched_tokens=256),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    p = SamplingParams(max_tokens=6, temperature=0.0, ignore_eos=True)
+    # abort while pending
+    eng.add_request("a1", list(range(30)), p)
+    eng.step()  # launches, pending in flight
+    eng.abort_request("a1")
+    for _ in range(5):
+        eng.step()
+    assert not eng.has_unfinished()
+
+    errors = []
+
+    def chaos():
+        try:
+            for i in range(100):
+                rid = f"c{i}"
+                with eng.lock:
+                    eng.add_request(rid, list(range(10, 42)), p)
+                if i % 3 == 0:
+                    with eng.lock:
+                        eng.abort_request(rid)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    t = threading.Thread(target=chaos)
+    t.start()
+    deadline = _t.time() + 3
+    while _t.time() < deadline:
+        eng.step()
+    t.join()
+    assert not errors
+    for _ in range(2000):
+        if not eng.has_unfinished():
+            break
+        eng.step()
+    assert not eng.has_unfinished()
+    bm = eng.block_manager
+    assert bm.num_free == bm.num_blocks
