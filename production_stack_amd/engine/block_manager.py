@@ -159,13 +159,18 @@ class BlockManager:
         if not self.enable_prefix_caching:
             return
         bs = self.block_size
-        tokens = seq.token_ids()
         n_full = seq.num_computed // bs
-        prev = self._tail_hash.get(seq.request_id)
         # count already registered full blocks for this seq
         done = getattr(seq, "_registered_full", None)
         if done is None:
             done = seq.num_cached_prompt_tokens // bs
+        if n_full <= done:
+            # no new full block (15 of every 16 decode steps): skip the
+            # O(len) token_ids() materialisation entirely
+            seq._registered_full = done  # type: ignore[attr-defined]
+            return
+        tokens = seq.token_ids()
+        prev = self._tail_hash.get(seq.request_id)
         for i in range(done, n_full):
             blk = seq.block_table[i]
             chunk = tokens[i * bs : (i + 1) * bs]
