@@ -127,69 +127,6 @@ __global__ __launch_bounds__(256) void skinny_combine_kernel(
 
 
 
-// Barrier-free variant for M <= 64: A-frags load DIRECTLY from global x
-// (same scattered-row pattern as the W loads; x is L2-resident at these
-// sizes), so there is no LDS staging and no __syncthreads at all — waves
-// of co-resident workgroups hide each other's memory latency freely.
-template <int MT>  // m-tiles (M <= MT*16, MT <= 4)
-__global__ __launch_bounds__(256, 4) void skinny_gemm_direct_kernel(
-    unsigned short* __restrict__ out_bf16,  // [M, N] (splits == 1)
-    float* __restrict__ ws,                 // [S, M, N] (splits > 1)
-    const unsigned short* __restrict__ x,   // [M, K]
-    const unsigned short* __restrict__ w,   // [N, K]
-    int M, int N, int K, long x_stride) {
-  const int splits = gridDim.y;
-  const int kchunks = K / 64;
-  const int per_split = (kchunks + splits - 1) / splits;
-  const int kc_begin = blockIdx.y * per_split;
-  const int kc_end = min(kchunks, kc_begin + per_split);
-  if (kc_begin >= kc_end) return;
-
-  const int tid = threadIdx.x;
-  const int wave = tid >> 6;
-  const int lane = tid & 63;
-  const int g = lane >> 4;
-  const int rc = lane & 15;
-  const int nrow = blockIdx.x * 64 + wave * 16 + rc;
-
-  ps_gf32x4 acc[MT];
-#pragma unroll
-  for (int mt = 0; mt < MT; mt++) acc[mt] = {0.f, 0.f, 0.f, 0.f};
-
-  const unsigned short* wr = w + (long)nrow * K;
-  for (int kc = kc_begin; kc < kc_end; kc++) {
-    const int k0 = kc * 64;
-    ps_gbf16x8 b0 = ps_as_gbf16(*(const ps_bf16x8*)(wr + k0 + g * 8));
-    ps_gbf16x8 b1 = ps_as_gbf16(*(const ps_bf16x8*)(wr + k0 + 32 + g * 8));
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int mt = 0; mt < MT; mt++) {
-      const int row = min(mt * 16 + rc, M - 1);
-      const unsigned short* xr = x + (long)row * x_stride + k0;
-      ps_gbf16x8 a0 = ps_as_gbf16(*(const ps_bf16x8*)(xr + g * 8));
-      ps_gbf16x8 a1 = ps_as_gbf16(*(const ps_bf16x8*)(xr + 32 + g * 8));
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[mt],
-                                                        0, 0, 0);
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc[mt],
-                                                        0, 0, 0);
-    }
-    __builtin_amdgcn_s_setprio(0);
-  }
-
-#pragma unroll
-  for (int mt = 0; mt < MT; mt++) {
-#pragma unroll
-    for (int r = 0; r < 4; r++) {
-      const int m = mt * 16 + g * 4 + r;
-      if (m >= M) continue;
-      if (gridDim.y == 1)
-        out_bf16[(long)m * N + nrow] = ps_f32_to_bf16(acc[mt][r]);
-      else
-        ws[((long)blockIdx.y * M + m) * N + nrow] = acc[mt][r];
-    }
-  }
-}
-
 extern "C" {
 
 // ws == nullptr forces splits = 1. ps_skinny_gemm_splits reports the
@@ -216,17 +153,12 @@ int ps_skinny_gemm(void* out_bf16, void* ws, const void* x, const void* w,
   skinny_gemm_kernel<MT><<<grid, block, 0, stream>>>(                       \
       (unsigned short*)out_bf16, (float*)ws, (const unsigned short*)x,      \
       (const unsigned short*)w, M, N, K, x_stride)
-#define PS_SGD(MT)                                                          \
-  skinny_gemm_direct_kernel<MT><<<grid, block, 0, stream>>>(                \
-      (unsigned short*)out_bf16, (float*)ws, (const unsigned short*)x,      \
-      (const unsigned short*)w, M, N, K, x_stride)
-  if (M <= 16) PS_SGD(1);
-  else if (M <= 32) PS_SGD(2);
-  else if (M <= 64) PS_SGD(4);
+  if (M <= 16) PS_SG(1);
+  else if (M <= 32) PS_SG(2);
+  else if (M <= 64) PS_SG(4);
   else if (M <= 128) PS_SG(8);
   else PS_SG(16);
 #undef PS_SG
-#undef PS_SGD
   if (splits > 1) {
     const long MN = (long)M * N;
     skinny_combine_kernel<<<dim3((MN + 255) / 256), 256, 0, stream>>>(
