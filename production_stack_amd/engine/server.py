@@ -598,6 +598,45 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                 )
         return JSONResponse(resp)
 
+    @app.get("/version")
+    async def version():
+        return {"version": "production-stack-amd 0.1 (MI355X)"}
+
+    @app.post("/v1/responses")
+    async def responses(request: Request):
+        """Minimal OpenAI Responses API: `input` string or message list
+        runs through the chat path; output in Responses format."""
+        body = await request.json()
+        inp = body.get("input", "")
+        if isinstance(inp, str):
+            messages = [{"role": "user", "content": inp}]
+        else:
+            messages = inp
+        chat_body = dict(body)
+        chat_body.pop("input", None)
+        chat_body["messages"] = messages
+        prompt_tokens = render_chat(
+            engine.tokenizer, messages, engine.model_cfg
+        )
+        params = _params_from_body(chat_body, engine.config.max_model_len)
+        rid = f"resp-{uuid.uuid4().hex[:12]}"
+        text = ""
+        async for out in async_engine.generate(rid, prompt_tokens, params):
+            text += out.text_delta
+        return JSONResponse({
+            "id": rid,
+            "object": "response",
+            "created_at": int(time.time()),
+            "model": body.get("model", served_model),
+            "status": "completed",
+            "output": [{
+                "type": "message",
+                "role": "assistant",
+                "content": [{"type": "output_text", "text": text}],
+            }],
+            "output_text": text,
+        })
+
     @app.post("/v1/completions")
     async def completions(request: Request):
         return await _run_completion(request, chat=False)

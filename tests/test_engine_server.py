@@ -367,3 +367,22 @@ def test_n_choices_and_logprobs():
         assert d["usage"]["completion_tokens"] == 8
 
     with_server(go)
+
+
+def test_responses_api_and_version():
+    async def go(client):
+        r = await client.get("/version")
+        assert r.status_code == 200 and "version" in r.json()
+        r = await client.post(
+            "/v1/responses",
+            json={"model": "tiny-llama", "input": "hello there",
+                  "max_tokens": 4, "temperature": 0, "ignore_eos": True},
+            timeout=120,
+        )
+        assert r.status_code == 200
+        d = r.json()
+        assert d["object"] == "response" and d["status"] == "completed"
+        assert d["output"][0]["content"][0]["type"] == "output_text"
+        assert d["output_text"]
+
+    with_server(go)
