@@ -615,9 +615,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         chat_body = dict(body)
         chat_body.pop("input", None)
         chat_body["messages"] = messages
-        prompt_tokens = render_chat(
-            engine.tokenizer, messages, engine.model_cfg
-        )
+        prompt_tokens = engine.tokenizer.encode(render_chat(messages))
         params = _params_from_body(chat_body, engine.config.max_model_len)
         rid = f"resp-{uuid.uuid4().hex[:12]}"
         text = ""
