@@ -89,6 +89,7 @@ def main() -> None:
                     help="cProfile the measured loop; report to stderr")
     ap.add_argument("--no-async-scheduling", action="store_true",
                     help="disable one-step-lagged sampling")
+    ap.add_argument("--quantization", default=None, choices=[None, "fp8"])
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -109,6 +110,7 @@ def main() -> None:
         model=args.model,
         max_model_len=args.max_model_len,
         async_scheduling=not args.no_async_scheduling,
+        quantization=args.quantization,
         seed=1234 + rank,
         cache=CacheConfig(
             block_size=16,
@@ -252,6 +254,7 @@ def main() -> None:
                         "parallelism": f"dp{world}",
                         "prefix_caching": True,
                         "kv_cache_dtype": args.kv_cache_dtype,
+                "quantization": args.quantization or "none",
                     },
                 }
             )

@@ -299,6 +299,9 @@ class EngineConfig:
     # non-greedy sampling, penalties, logprobs or speculative chunks fall
     # back to the synchronous path transparently.
     async_scheduling: bool = False
+    # weight quantization: None (bf16) or "fp8" (OCP e4m3 weights +
+    # dynamic per-tensor activation quant through the fp8 MFMA pipe)
+    quantization: Optional[str] = None
     enable_lora: bool = False
     max_loras: int = 4
     max_lora_rank: int = 16

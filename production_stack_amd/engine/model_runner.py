@@ -81,6 +81,8 @@ class ModelRunner:
             self.model.lora_slots = self.lora_slots
             for layer in self.model.layers:
                 layer.lora_slots = self.lora_slots
+        if config.quantization == "fp8":
+            self.model.quantize_fp8()
         self._generator = torch.Generator(device="cpu").manual_seed(
             config.seed + 12345
         )
@@ -156,6 +158,8 @@ class ModelRunner:
         from production_stack_amd.ops import gemm_policy
 
         try:
+            if self.config.quantization == "fp8":
+                raise RuntimeError("skip autotune: fp8-quantized weights")
             layer0 = self.model.layers[0]
             weights = [
                 layer0.qkv_proj, layer0.o_proj, layer0.gate_up_proj,

@@ -125,7 +125,10 @@ class LlamaLayer(nn.Module):
             hidden, residual = ops.fused_add_rms_norm(
                 hidden, residual, self.input_norm, cfg.rms_norm_eps
             )
-        qkv = gemm_policy.linear(hidden, self.qkv_proj)
+        if getattr(self, "fp8_w", None) is not None:
+            qkv = ops.fp8_linear(hidden, *self.fp8_w["qkv"])
+        else:
+            qkv = gemm_policy.linear(hidden, self.qkv_proj)
         if self.qkv_bias is not None:
             qkv = qkv + self.qkv_bias
         qs = self.q_heads * self.head_dim
@@ -215,7 +218,11 @@ class LlamaLayer(nn.Module):
                 )
             )
         attn = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
-        attn_out = gemm_policy.linear(attn.view(T, qs), self.o_proj)
+        if getattr(self, "fp8_w", None) is not None:
+            attn_out = ops.fp8_linear(attn.reshape(T, qs),
+                                      *self.fp8_w["o"])
+        else:
+            attn_out = gemm_policy.linear(attn.view(T, qs), self.o_proj)
         if slots is not None and meta.lora_idx is not None:
             slots.apply("o", self.layer_idx, attn_out,
                         attn.reshape(T, qs), meta.lora_idx)
@@ -229,7 +236,10 @@ class LlamaLayer(nn.Module):
         hidden, residual = ops.fused_add_rms_norm(
             attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
         )
-        gate_up = gemm_policy.linear(hidden, self.gate_up_proj)
+        if getattr(self, "fp8_w", None) is not None:
+            gate_up = ops.fp8_linear(hidden, *self.fp8_w["gate_up"])
+        else:
+            gate_up = gemm_policy.linear(hidden, self.gate_up_proj)
         if slots is not None and meta.lora_idx is not None:
             slots.apply("gate", self.layer_idx, gate_up, hidden,
                         meta.lora_idx)
@@ -245,7 +255,10 @@ class LlamaLayer(nn.Module):
                 "up", self.inter, self.inter,
             )
         act = ops.silu_and_mul(gate_up)
-        mlp_out = gemm_policy.linear(act, self.down_proj)
+        if getattr(self, "fp8_w", None) is not None:
+            mlp_out = ops.fp8_linear(act, *self.fp8_w["down"])
+        else:
+            mlp_out = gemm_policy.linear(act, self.down_proj)
         if slots is not None and meta.lora_idx is not None:
             slots.apply("down", self.layer_idx, mlp_out, act, meta.lora_idx)
         elif meta.lora_groups:
@@ -314,6 +327,26 @@ class LlamaForCausalLM(nn.Module):
                 p.normal_(0.0, 0.002, generator=gen)
             else:
                 p.normal_(0.0, 0.02, generator=gen)
+
+    @torch.no_grad()
+    def quantize_fp8(self) -> None:
+        """Convert the four per-layer projections to OCP fp8-e4m3 with
+        per-tensor scales (vLLM --quantization fp8 analogue); halves
+        weight HBM and runs GEMMs through the fp8 MFMA pipe. Embedding,
+        lm_head and norms stay bf16."""
+        for layer in self.layers:
+            fp8_w = {}
+            for key, pname in (("qkv", "qkv_proj"), ("o", "o_proj"),
+                               ("gate_up", "gate_up_proj"),
+                               ("down", "down_proj")):
+                w = getattr(layer, pname)
+                w_q, scale = ops.fp8_quantize_weight(w.data)
+                fp8_w[key] = (w_q, scale.to(w.device))
+                # free the bf16 copy (replace with a tiny stub so
+                # state_dict/save paths still see the attribute)
+                w.data = torch.empty(0, dtype=torch.bfloat16,
+                                     device=w.device)
+            layer.fp8_w = fp8_w
 
     @property
     def kv_heads(self) -> int:

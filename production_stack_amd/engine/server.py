@@ -713,6 +713,9 @@ def main() -> None:
                     help="cacheserver data plane host:port (shared KV tier)")
     ap.add_argument("--kv-cache-dtype", default="auto",
                     choices=["auto", "bf16", "fp8", "fp8_e4m3"])
+    ap.add_argument("--quantization", default=None,
+                    choices=[None, "fp8"],
+                    help="fp8 = OCP e4m3 weights through the fp8 MFMA pipe")
     ap.add_argument("--async-scheduling", action="store_true",
                     help="one-step-lagged sampling (greedy-exact overlap)")
     ap.add_argument("--num-speculative-tokens", type=int, default=0,
@@ -766,6 +769,7 @@ def main() -> None:
             num_speculative_tokens=args.num_speculative_tokens,
         ),
         async_scheduling=args.async_scheduling,
+        quantization=args.quantization,
         enable_lora=args.enable_lora,
         max_loras=args.max_loras,
         max_lora_rank=args.max_lora_rank,

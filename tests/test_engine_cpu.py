@@ -467,3 +467,26 @@ def test_async_scheduling_abort_and_chaos():
     assert not eng.has_unfinished()
     bm = eng.block_manager
     assert bm.num_free == bm.num_blocks
+
+
+def test_fp8_weight_quantization_cpu():
+    """--quantization fp8: engine runs with fp8 weights (dequant matmul
+    reference on CPU); early greedy tokens agree with bf16."""
+    def mk(q):
+        cfg = EngineConfig(
+            model="tiny-llama",
+            max_model_len=256,
+            seed=5,
+            quantization=q,
+            cache=CacheConfig(num_gpu_blocks=128, block_size=16),
+            scheduler=SchedulerConfig(max_num_seqs=8,
+                                      max_num_batched_tokens=128),
+        )
+        return LLMEngine(cfg, device="cpu")
+
+    p = SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True)
+    want = mk(None).generate([[5, 6, 7, 8, 9]], p)["offline-0"]
+    got = mk("fp8").generate([[5, 6, 7, 8, 9]], p)["offline-0"]
+    assert len(got) == 8
+    agree = sum(a == b for a, b in zip(want[:4], got[:4]))
+    assert agree >= 2, (want, got)
