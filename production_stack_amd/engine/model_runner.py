@@ -157,24 +157,24 @@ class ModelRunner:
         from production_stack_amd.engine.graph_runner import DecodeGraphRunner
         from production_stack_amd.ops import gemm_policy
 
-        try:
-            if self.config.quantization == "fp8":
-                raise RuntimeError("skip autotune: fp8-quantized weights")
-            layer0 = self.model.layers[0]
-            weights = [
-                layer0.qkv_proj, layer0.o_proj, layer0.gate_up_proj,
-                layer0.down_proj,
-            ]
-            from production_stack_amd.engine.graph_runner import BUCKETS
+        # fp8-quantized GEMMs run through scaled_mm: nothing to autotune
+        if self.config.quantization != "fp8":
+            try:
+                layer0 = self.model.layers[0]
+                weights = [
+                    layer0.qkv_proj, layer0.o_proj, layer0.gate_up_proj,
+                    layer0.down_proj,
+                ]
+                from production_stack_amd.engine.graph_runner import BUCKETS
 
-            buckets = [b for b in BUCKETS if b <= max_batch]
-            gemm_policy.tune(weights, buckets, self.device)
-        except Exception:
-            import logging
+                buckets = [b for b in BUCKETS if b <= max_batch]
+                gemm_policy.tune(weights, buckets, self.device)
+            except Exception:
+                import logging
 
-            logging.getLogger("engine.runner").exception(
-                "gemm autotune failed; using hipBLASLt everywhere"
-            )
+                logging.getLogger("engine.runner").exception(
+                    "gemm autotune failed; using hipBLASLt everywhere"
+                )
 
         max_blocks = (
             self.config.max_model_len + self.config.cache.block_size - 1
