@@ -22,18 +22,19 @@ from production_stack_amd.engine.config import (
 PROMPT = list(range(20, 84))  # 64 tokens
 
 
-def _config(weights_path, tp):
+def _config(weights_path, tp, async_sched=False):
     return EngineConfig(
         model="tiny-llama",
         max_model_len=256,
         weights_path=weights_path,
+        async_scheduling=async_sched,
         cache=CacheConfig(num_gpu_blocks=64, block_size=16),
         scheduler=SchedulerConfig(max_num_seqs=4, max_num_batched_tokens=256),
         parallel=ParallelConfig(tensor_parallel_size=tp),
     )
 
 
-def _tp_worker(rank, world, weights_path, port, q):
+def _tp_worker(rank, world, weights_path, port, q, async_sched=False):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     os.environ["RANK"] = str(rank)
@@ -45,7 +46,8 @@ def _tp_worker(rank, world, weights_path, port, q):
     from production_stack_amd.parallel import state as pstate
 
     try:
-        eng = LLMEngine(_config(weights_path, world), device="cpu")
+        eng = LLMEngine(_config(weights_path, world, async_sched),
+                        device="cpu")
         out = eng.generate(
             [PROMPT],
             SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True),
@@ -97,3 +99,36 @@ def test_tp2_matches_tp1(tmp_path):
         p.join(timeout=60)
     assert status == "ok", out
     assert out == want, f"TP2 {out} != TP1 {want}"
+
+
+@pytest.mark.timeout(180)
+def test_tp2_lockstep_with_async_scheduling(tmp_path):
+    """TP=2 lockstep generate with async scheduling: every rank runs the
+    same one-step-lagged pipeline (device sampling is deterministic, so
+    the ranks stay in lockstep) and must match the TP=1 sync output."""
+    from production_stack_amd.engine.engine import LLMEngine
+    from production_stack_amd.engine.sampling import SamplingParams
+    from production_stack_amd.engine.weights import save_hf_safetensors
+    import torch.multiprocessing as mp
+
+    ref = LLMEngine(_config(None, 1), device="cpu")
+    wdir = str(tmp_path / "w")
+    save_hf_safetensors(ref.runner.model, wdir)
+    want = ref.generate(
+        [PROMPT],
+        SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True),
+    )["offline-0"]
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=_tp_worker, args=(r, 2, wdir, 29791, q, True))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    status, out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", out
+    assert out == want, f"TP2-async {out} != TP1 {want}"
