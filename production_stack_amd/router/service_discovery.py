@@ -211,9 +211,16 @@ class StaticServiceDiscovery(ServiceDiscovery):
 class K8sPodIpServiceDiscovery(ServiceDiscovery):
     """Watches pods matching a label selector; endpoint per ready pod IP.
 
-    Requires the `kubernetes` package (present in cluster images; absent in
-    the offline build env, in which case construction raises).
+    Implemented over the raw Kubernetes REST API (streaming watch with
+    resourceVersion resume) rather than the `kubernetes` client package —
+    same approach as the C++ operator — so it runs (and is testable)
+    without the client library. In-cluster credentials default to the
+    mounted service-account token/CA; `api_base`/`sa_token` override for
+    tests or out-of-cluster use. Parity: reference
+    service_discovery.py:411-889 (K8sPodIpServiceDiscovery).
     """
+
+    SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
 
     def __init__(
         self,
@@ -221,68 +228,137 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
         port: int = 8000,
         label_selector: Optional[str] = None,
         api_key: Optional[str] = None,
+        api_base: Optional[str] = None,
+        sa_token: Optional[str] = None,
+        probe_models: bool = True,
     ) -> None:
-        try:
-            from kubernetes import client, config, watch  # noqa: F401
-        except ImportError as e:  # pragma: no cover
-            raise RuntimeError(
-                "k8s service discovery requires the `kubernetes` package"
-            ) from e
         self.namespace = namespace
         self.port = port
         self.label_selector = label_selector
         self.api_key = api_key
+        self.probe_models = probe_models
+        self.api_base = api_base or "https://kubernetes.default.svc"
+        self._token = sa_token
+        if self._token is None:
+            try:
+                with open(f"{self.SA_DIR}/token") as f:
+                    self._token = f.read().strip()
+            except OSError:
+                self._token = ""
+        import os
+
+        self._verify = (
+            f"{self.SA_DIR}/ca.crt"
+            if os.path.exists(f"{self.SA_DIR}/ca.crt")
+            else False
+        )
         self._endpoints: Dict[str, EndpointInfo] = {}
         self._lock = threading.Lock()
         self._stop = threading.Event()
-        config.load_incluster_config()
-        self._core = client.CoreV1Api()
-        self._watch = watch.Watch()
+        self._resource_version = ""
         self._thread = threading.Thread(target=self._watch_pods, daemon=True)
         self._thread.start()
 
+    # ---- REST helpers -------------------------------------------------
+    def _headers(self) -> Dict[str, str]:
+        h = {"Accept": "application/json"}
+        if self._token:
+            h["Authorization"] = f"Bearer {self._token}"
+        return h
+
+    def _pods_url(self) -> str:
+        url = f"{self.api_base}/api/v1/namespaces/{self.namespace}/pods"
+        if self.label_selector:
+            from urllib.parse import quote
+
+            url += f"?labelSelector={quote(self.label_selector)}"
+        return url
+
     def _get_model_names(self, ip: str) -> List[str]:
+        if not self.probe_models:
+            return []
         try:
             headers = {}
             if self.api_key:
                 headers["Authorization"] = f"Bearer {self.api_key}"
             r = requests.get(
-                f"http://{ip}:{self.port}/v1/models", headers=headers, timeout=5
+                f"http://{ip}:{self.port}/v1/models", headers=headers,
+                timeout=5
             )
             return [m["id"] for m in r.json().get("data", [])]
         except requests.RequestException:
             return []
 
-    def _watch_pods(self) -> None:  # pragma: no cover - needs a cluster
+    def _apply_pod(self, ev_type: str, pod: dict) -> None:
+        meta = pod.get("metadata", {})
+        status = pod.get("status", {})
+        name = meta.get("name", "")
+        ip = status.get("podIP")
+        statuses = status.get("containerStatuses") or []
+        ready = (
+            bool(statuses)
+            and all(c.get("ready") for c in statuses)
+            and meta.get("deletionTimestamp") is None
+        )
+        url = f"http://{ip}:{self.port}" if ip else None
+        with self._lock:
+            if ev_type == "DELETED" or not ready or not url:
+                self._endpoints.pop(name, None)
+            else:
+                labels = meta.get("labels") or {}
+                self._endpoints[name] = EndpointInfo(
+                    url=url,
+                    model_names=self._get_model_names(ip),
+                    model_label=labels.get("model"),
+                    pod_name=name,
+                    sleep=labels.get("sleeping") == "true",
+                )
+
+    def _list_once(self) -> None:
+        r = requests.get(self._pods_url(), headers=self._headers(),
+                         verify=self._verify, timeout=15)
+        r.raise_for_status()
+        body = r.json()
+        self._resource_version = body.get("metadata", {}).get(
+            "resourceVersion", ""
+        )
+        with self._lock:
+            self._endpoints.clear()
+        for pod in body.get("items", []):
+            self._apply_pod("ADDED", pod)
+
+    def _watch_pods(self) -> None:
+        import json as _json
+
         while not self._stop.is_set():
             try:
-                for event in self._watch.stream(
-                    self._core.list_namespaced_pod,
-                    namespace=self.namespace,
-                    label_selector=self.label_selector,
-                    timeout_seconds=30,
-                ):
-                    pod = event["object"]
-                    name = pod.metadata.name
-                    ip = pod.status.pod_ip
-                    ready = (
-                        pod.status.container_statuses is not None
-                        and all(c.ready for c in pod.status.container_statuses)
-                        and pod.metadata.deletion_timestamp is None
-                    )
-                    url = f"http://{ip}:{self.port}" if ip else None
-                    with self._lock:
-                        if event["type"] == "DELETED" or not ready or not url:
-                            self._endpoints.pop(name, None)
-                        else:
-                            labels = pod.metadata.labels or {}
-                            self._endpoints[name] = EndpointInfo(
-                                url=url,
-                                model_names=self._get_model_names(ip),
-                                model_label=labels.get("model"),
-                                pod_name=name,
-                            )
+                self._list_once()
+                sep = "&" if "?" in self._pods_url() else "?"
+                url = (
+                    f"{self._pods_url()}{sep}watch=true"
+                    f"&resourceVersion={self._resource_version}"
+                    f"&timeoutSeconds=30"
+                )
+                with requests.get(
+                    url, headers=self._headers(), verify=self._verify,
+                    stream=True, timeout=40,
+                ) as r:
+                    r.raise_for_status()
+                    # chunk_size=1: watch events are sparse; the default
+                    # 512-byte buffering would sit on an event until more
+                    # bytes arrive or the stream closes
+                    for line in r.iter_lines(chunk_size=1):
+                        if self._stop.is_set():
+                            return
+                        if not line:
+                            continue
+                        ev = _json.loads(line)
+                        self._apply_pod(
+                            ev.get("type", ""), ev.get("object", {})
+                        )
             except Exception as e:
+                if self._stop.is_set():
+                    return
                 logger.warning("k8s watch error: %s", e)
                 time.sleep(1)
 
@@ -299,6 +375,8 @@ class K8sServiceNameServiceDiscovery(ServiceDiscovery):
     the stable in-cluster service DNS names. Parity: reference
     service_discovery.py:892-1306 (K8sServiceNameServiceDiscovery)."""
 
+    SA_DIR = K8sPodIpServiceDiscovery.SA_DIR
+
     def __init__(
         self,
         namespace: str = "default",
@@ -306,58 +384,82 @@ class K8sServiceNameServiceDiscovery(ServiceDiscovery):
         label_selector: Optional[str] = None,
         api_key: Optional[str] = None,
         refresh_interval: float = 30.0,
+        api_base: Optional[str] = None,
+        sa_token: Optional[str] = None,
+        probe_models: bool = True,
     ) -> None:
-        try:
-            from kubernetes import client, config  # noqa: F401
-        except ImportError as e:  # pragma: no cover
-            raise RuntimeError(
-                "k8s service discovery requires the `kubernetes` package"
-            ) from e
         self.namespace = namespace
         self.port = port
         self.label_selector = label_selector
         self.api_key = api_key
         self.refresh_interval = refresh_interval
+        self.probe_models = probe_models
+        self.api_base = api_base or "https://kubernetes.default.svc"
+        self._token = sa_token
+        if self._token is None:
+            try:
+                with open(f"{self.SA_DIR}/token") as f:
+                    self._token = f.read().strip()
+            except OSError:
+                self._token = ""
+        import os
+
+        self._verify = (
+            f"{self.SA_DIR}/ca.crt"
+            if os.path.exists(f"{self.SA_DIR}/ca.crt")
+            else False
+        )
         self._endpoints: Dict[str, EndpointInfo] = {}
         self._lock = threading.Lock()
         self._stop = threading.Event()
-        config.load_incluster_config()
-        self._core = client.CoreV1Api()
         self._thread = threading.Thread(target=self._refresh_loop,
                                         daemon=True)
         self._thread.start()
 
-    def _refresh_once(self) -> None:  # pragma: no cover - needs a cluster
-        svcs = self._core.list_namespaced_service(
-            namespace=self.namespace, label_selector=self.label_selector
+    def _refresh_once(self) -> None:
+        url = (
+            f"{self.api_base}/api/v1/namespaces/{self.namespace}/services"
         )
+        if self.label_selector:
+            from urllib.parse import quote
+
+            url += f"?labelSelector={quote(self.label_selector)}"
+        headers = {"Accept": "application/json"}
+        if self._token:
+            headers["Authorization"] = f"Bearer {self._token}"
+        r = requests.get(url, headers=headers, verify=self._verify,
+                         timeout=15)
+        r.raise_for_status()
         new: Dict[str, EndpointInfo] = {}
-        for svc in svcs.items:
-            name = svc.metadata.name
+        for svc in r.json().get("items", []):
+            meta = svc.get("metadata", {})
+            spec = svc.get("spec", {})
+            name = meta.get("name", "")
             port = self.port
-            if svc.spec.ports:
-                port = svc.spec.ports[0].port
-            url = f"http://{name}.{self.namespace}.svc:{port}"
+            if spec.get("ports"):
+                port = spec["ports"][0].get("port", port)
+            svc_url = f"http://{name}.{self.namespace}.svc:{port}"
             models: List[str] = []
-            try:
-                headers = {}
-                if self.api_key:
-                    headers["Authorization"] = f"Bearer {self.api_key}"
-                r = requests.get(url + "/v1/models", headers=headers,
-                                 timeout=5)
-                models = [m["id"] for m in r.json().get("data", [])]
-            except requests.RequestException:
-                pass
-            labels = svc.metadata.labels or {}
+            if self.probe_models:
+                try:
+                    h = {}
+                    if self.api_key:
+                        h["Authorization"] = f"Bearer {self.api_key}"
+                    rr = requests.get(svc_url + "/v1/models", headers=h,
+                                      timeout=5)
+                    models = [m["id"] for m in rr.json().get("data", [])]
+                except requests.RequestException:
+                    pass
+            labels = meta.get("labels") or {}
             new[name] = EndpointInfo(
-                url=url,
+                url=svc_url,
                 model_names=models,
                 model_label=labels.get("model"),
             )
         with self._lock:
             self._endpoints = new
 
-    def _refresh_loop(self) -> None:  # pragma: no cover
+    def _refresh_loop(self) -> None:
         while not self._stop.is_set():
             try:
                 self._refresh_once()

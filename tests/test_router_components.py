@@ -182,3 +182,143 @@ def test_redaction():
 def test_parse_helpers():
     assert parse_static_urls("http://a/, http://b") == ["http://a", "http://b"]
     assert parse_static_aliases("al:m1, a2:m2") == {"al": "m1", "a2": "m2"}
+
+
+def test_k8s_pod_ip_discovery_over_fake_api():
+    """K8sPodIpServiceDiscovery over the raw REST API against a fake
+    apiserver: initial list + watch events add/remove ready pods."""
+    import http.server
+    import json
+    import threading
+    import time
+
+    from production_stack_amd.router.service_discovery import (
+        K8sPodIpServiceDiscovery,
+    )
+
+    def pod(name, ip, ready=True, labels=None, deleted=False):
+        return {
+            "metadata": {
+                "name": name,
+                "labels": labels or {"model": "m"},
+                **({"deletionTimestamp": "now"} if deleted else {}),
+            },
+            "status": {
+                "podIP": ip,
+                "containerStatuses": [{"ready": ready}],
+            },
+        }
+
+    events = [
+        {"type": "ADDED", "object": pod("p2", "10.0.0.2")},
+        {"type": "MODIFIED", "object": pod("p1", "10.0.0.1", ready=False)},
+    ]
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def log_message(self, *a):  # quiet
+            pass
+
+        def do_GET(self):
+            if "watch=true" in self.path:
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.end_headers()
+                for ev in events:
+                    self.wfile.write(
+                        (json.dumps(ev) + "\n").encode()
+                    )
+                    self.wfile.flush()
+                time.sleep(3)  # hold the stream open
+            else:
+                body = {
+                    "metadata": {"resourceVersion": "1"},
+                    "items": [pod("p1", "10.0.0.1")],
+                }
+                data = json.dumps(body).encode()
+                self.send_response(200)
+                self.send_header("Content-Type", "application/json")
+                self.send_header("Content-Length", str(len(data)))
+                self.end_headers()
+                self.wfile.write(data)
+
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        disc = K8sPodIpServiceDiscovery(
+            namespace="ns",
+            port=9000,
+            label_selector="app=engine",
+            api_base=f"http://127.0.0.1:{srv.server_port}",
+            sa_token="test-token",
+            probe_models=False,
+        )
+        deadline = time.time() + 10
+        urls = []
+        while time.time() < deadline:
+            urls = sorted(e.url for e in disc.get_endpoint_info())
+            if urls == ["http://10.0.0.2:9000"]:
+                break
+            time.sleep(0.1)
+        # p1 listed then marked not-ready by the watch; p2 added by watch
+        assert urls == ["http://10.0.0.2:9000"], urls
+        disc.close()
+    finally:
+        srv.shutdown()
+
+
+def test_k8s_service_name_discovery_over_fake_api():
+    import http.server
+    import json
+    import threading
+    import time
+
+    from production_stack_amd.router.service_discovery import (
+        K8sServiceNameServiceDiscovery,
+    )
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def log_message(self, *a):
+            pass
+
+        def do_GET(self):
+            body = {"items": [
+                {"metadata": {"name": "engine-a",
+                              "labels": {"model": "decode"}},
+                 "spec": {"ports": [{"port": 8000}]}},
+                {"metadata": {"name": "engine-b", "labels": {}},
+                 "spec": {"ports": []}},
+            ]}
+            data = json.dumps(body).encode()
+            self.send_response(200)
+            self.send_header("Content-Length", str(len(data)))
+            self.end_headers()
+            self.wfile.write(data)
+
+    srv = http.server.ThreadingHTTPServer(("127.0.0.1", 0), Handler)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        disc = K8sServiceNameServiceDiscovery(
+            namespace="ns",
+            port=9000,
+            api_base=f"http://127.0.0.1:{srv.server_port}",
+            sa_token="t",
+            probe_models=False,
+            refresh_interval=1.0,
+        )
+        deadline = time.time() + 10
+        urls = []
+        while time.time() < deadline:
+            urls = sorted(e.url for e in disc.get_endpoint_info())
+            if len(urls) == 2:
+                break
+            time.sleep(0.05)
+        assert urls == [
+            "http://engine-a.ns.svc:8000",
+            "http://engine-b.ns.svc:9000",
+        ], urls
+        labels = {e.url: e.model_label for e in disc.get_endpoint_info()}
+        assert labels["http://engine-a.ns.svc:8000"] == "decode"
+        disc.close()
+    finally:
+        srv.shutdown()
