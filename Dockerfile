@@ -12,5 +12,5 @@ RUN python3 setup.py build_ext --inplace && \
 # router:  python3 -m production_stack_amd.router.app --help
 # kv ctrl: python3 -m production_stack_amd.kvpool.controller
 # operator: /workspace/operator/psoperator
-EXPOSE 8000 8001 9000 14001
+EXPOSE 8000 8001 9000 9400 14001
 CMD ["python3", "-m", "production_stack_amd.engine.server", "llama-3-8b"]
