@@ -79,7 +79,7 @@ class LLMEngine:
         else:
             self.host_pool = None
         self.runner.capture_decode_graphs(
-            min(config.scheduler.max_num_seqs, 256)
+            min(config.scheduler.max_num_seqs, 512)
         )
         self.stats = EngineStats()
         self._sleeping = False

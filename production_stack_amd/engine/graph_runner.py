@@ -23,7 +23,7 @@ from production_stack_amd.engine.models.llama import BatchMeta
 
 logger = logging.getLogger("engine.graphs")
 
-BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256]
+BUCKETS = [1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128, 192, 256, 384, 512]
 
 
 class DecodeGraphRunner:
