@@ -78,7 +78,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=64)
     ap.add_argument("--warmup", type=int, default=48)
-    ap.add_argument("--users", type=int, default=256, help="conversations per GPU")
+    ap.add_argument("--users", type=int, default=320, help="conversations per GPU (reference run.sh canonical: 320)")
     ap.add_argument("--model", default="llama-3-8b")
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--max-num-batched-tokens", type=int, default=2048)
