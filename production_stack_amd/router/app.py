@@ -345,9 +345,18 @@ def initialize_all(app: FastAPI, args) -> None:
             api_key=args.api_key,
         )
         app.state.model_aliases = parse_static_aliases(args.static_aliases)
-    elif args.service_discovery == "k8s":
+    elif args.service_discovery in ("k8s", "k8s_pod_ip"):
         sd.initialize_service_discovery(
             "k8s",
+            namespace=args.k8s_namespace,
+            port=args.k8s_port,
+            label_selector=args.k8s_label_selector,
+            api_key=args.api_key,
+        )
+        app.state.model_aliases = {}
+    elif args.service_discovery == "k8s_service_name":
+        sd.initialize_service_discovery(
+            "k8s_service_name",
             namespace=args.k8s_namespace,
             port=args.k8s_port,
             label_selector=args.k8s_label_selector,

@@ -15,7 +15,8 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     # service discovery
     p.add_argument(
         "--service-discovery",
-        choices=["static", "k8s", "external"],
+        choices=["static", "k8s", "k8s_pod_ip", "k8s_service_name",
+                 "external", "external-only"],
         default="static",
     )
     p.add_argument("--static-backends", type=str, default=None,
