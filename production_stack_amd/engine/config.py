@@ -35,6 +35,9 @@ class ModelConfig:
     qkv_bias: bool = False
     tie_word_embeddings: bool = False
     sliding_window: Optional[int] = None
+    # Llama-3.1-style RoPE frequency scaling (None = off):
+    # (factor, low_freq_factor, high_freq_factor, original_max_position)
+    rope_scaling: Optional[tuple] = None
 
     @property
     def q_size(self) -> int:
@@ -72,6 +75,32 @@ ARCHITECTURES = {
         vocab_size=128256,
         rope_theta=500000.0,
         max_position=8192,
+    ),
+    "llama-3.1-8b": ModelConfig(
+        name="llama-3.1-8b",
+        hidden_size=4096,
+        num_layers=32,
+        num_q_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=128256,
+        rope_theta=500000.0,
+        max_position=131072,
+        rope_scaling=(8.0, 1.0, 4.0, 8192),
+    ),
+    "llama-3.1-70b": ModelConfig(
+        name="llama-3.1-70b",
+        hidden_size=8192,
+        num_layers=80,
+        num_q_heads=64,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=28672,
+        vocab_size=128256,
+        rope_theta=500000.0,
+        max_position=131072,
+        rope_scaling=(8.0, 1.0, 4.0, 8192),
     ),
     "llama-2-7b": ModelConfig(
         name="llama-2-7b",

@@ -62,13 +62,29 @@ class BatchMeta:
 
 
 def build_cos_sin_cache(
-    head_dim: int, max_position: int, theta: float
+    head_dim: int, max_position: int, theta: float,
+    rope_scaling=None,
 ) -> torch.Tensor:
-    """[max_position, head_dim] fp32: cos[half] || sin[half] (host-side)."""
+    """[max_position, head_dim] fp32: cos[half] || sin[half] (host-side).
+
+    rope_scaling = (factor, low_freq_factor, high_freq_factor,
+    original_max_position) applies the Llama-3.1 frequency remap: long
+    wavelengths divide by `factor`, short ones pass through, the band in
+    between interpolates smoothly (HF modeling_rope_utils llama3 rule)."""
     half = head_dim // 2
     inv = 1.0 / (
         theta ** (torch.arange(0, head_dim, 2, dtype=torch.float64) / head_dim)
     )
+    if rope_scaling is not None:
+        factor, low_f, high_f, orig_max = rope_scaling
+        low_wl = orig_max / low_f
+        high_wl = orig_max / high_f
+        wavelen = 2 * math.pi / inv
+        scaled = torch.where(wavelen > low_wl, inv / factor, inv)
+        smooth = (orig_max / wavelen - low_f) / (high_f - low_f)
+        mid = (1 - smooth) * (inv / factor) + smooth * inv
+        in_band = (wavelen <= low_wl) & (wavelen >= high_wl)
+        inv = torch.where(in_band, mid, scaled)
     t = torch.arange(max_position, dtype=torch.float64)
     freqs = torch.outer(t, inv)
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1).float().contiguous()
@@ -312,7 +328,8 @@ class LlamaForCausalLM(nn.Module):
                 )
         self.register_buffer(
             "cos_sin",
-            build_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_theta),
+            build_cos_sin_cache(cfg.head_dim, cfg.max_position,
+                                cfg.rope_theta, cfg.rope_scaling),
             persistent=False,
         )
 

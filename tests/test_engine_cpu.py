@@ -490,3 +490,22 @@ def test_fp8_weight_quantization_cpu():
     assert len(got) == 8
     agree = sum(a == b for a, b in zip(want[:4], got[:4]))
     assert agree >= 2, (want, got)
+
+
+def test_llama31_rope_scaling():
+    """Llama-3.1 frequency remap: high-frequency dims untouched,
+    low-frequency wavelengths divided by the scale factor."""
+    import torch
+
+    from production_stack_amd.engine.models.llama import build_cos_sin_cache
+
+    N = 65536
+    base = build_cos_sin_cache(128, N, 500000.0)
+    scaled = build_cos_sin_cache(128, N, 500000.0, (8.0, 1.0, 4.0, 8192))
+    diff = (base - scaled).abs().view(N, 2, 64).amax(dim=(0, 1))
+    assert diff[:4].max() < 1e-5, f"high-freq dims changed: {diff[:4]}"
+    assert diff[-4:].min() > 0.1, f"low-freq dims unscaled: {diff[-4:]}"
+
+    # engine-level: llama-3.1 config constructs and generates
+    from production_stack_amd.engine.config import ARCHITECTURES
+    assert ARCHITECTURES["llama-3.1-8b"].rope_scaling is not None
