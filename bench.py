@@ -133,6 +133,11 @@ def main() -> None:
     ).tolist()
     users = [User(u, rng, system, vocab_high) for u in range(args.users)]
     answers: dict = {}
+    ttfts: list = []
+    in_window = False
+    in_steady = False
+    window_tokens = 0
+    done_once: set = set()
 
     def submit(user: User) -> None:
         # cap history so prompt + answer fits the model length
@@ -146,6 +151,7 @@ def main() -> None:
         prompt = user.next_prompt()
         rid = f"u{user.uid}-r{user.round}"
         user.submit_time = time.perf_counter()
+        user.steady_submit = in_steady
         answers[rid] = (user, [])
         engine.add_request(
             rid, prompt,
@@ -155,10 +161,6 @@ def main() -> None:
 
     for u in users:
         submit(u)
-
-    ttfts: list = []
-    in_window = False
-    window_tokens = 0
 
     def run_step() -> None:
         nonlocal window_tokens
@@ -170,14 +172,36 @@ def main() -> None:
             toks.extend(out.new_token_ids)
             if in_window:
                 window_tokens += len(out.new_token_ids)
-                if out.first_token:
-                    ttfts.append(time.perf_counter() - user.submit_time)
+            # TTFT is only meaningful in steady state: rounds submitted
+            # while the initial 320-user prefill wave is still draining
+            # measure queueing of a cold start, not serving latency.
+            if out.first_token and in_steady and user.steady_submit:
+                ttfts.append(time.perf_counter() - user.submit_time)
             if out.finished:
                 user.complete(toks)
                 del answers[out.request_id]
+                done_once.add(user.uid)
                 submit(user)
 
-    # warmup (includes the initial prefill wave)
+    # Steady-state guard: the timed window must measure the serving steady
+    # state no matter what --steps/--warmup the driver passes.  The initial
+    # prefill wave of `users` x ~1.1k-token prompts needs
+    # ~users*1100/max_num_batched_tokens steps to drain, far more than a
+    # typical --warmup.  So before honoring --warmup we keep stepping until
+    # >=90% of users have completed their first round (mixed
+    # prefill-over-decode steady state), with a hard time cap as a
+    # safety net.  Only then do the driver's warmup steps run.
+    steady_target = max(1, int(0.9 * len(users)))
+    t_guard = time.perf_counter()
+    guard_cap_s = 300.0 if use_cuda else 120.0
+    guard_steps = 0
+    while (
+        len(done_once) < steady_target
+        and time.perf_counter() - t_guard < guard_cap_s
+    ):
+        run_step()
+        guard_steps += 1
+    in_steady = True
     for _ in range(args.warmup):
         run_step()
 

@@ -9,6 +9,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <cstdlib>
 #include <cstring>
 #include <sstream>
 #include <stdexcept>
@@ -116,12 +117,33 @@ Response request(const std::string& method, const std::string& url,
     }();
     (void)init;
     c.ctx = SSL_CTX_new(TLS_client_method());
-    // in-cluster: the pod's service-account CA should be loaded here;
-    // verification is relaxed for this build (no cluster to test against)
-    SSL_CTX_set_verify(c.ctx, SSL_VERIFY_NONE, nullptr);
+    // Verify the peer against the mounted service-account CA by default —
+    // a Bearer token travels on this connection, so SSL_VERIFY_NONE would
+    // hand credentials to any MITM. Opt-out only via the explicit
+    // PS_OPERATOR_TLS_INSECURE=1 escape hatch.
+    const char* insecure_env = getenv("PS_OPERATOR_TLS_INSECURE");
+    bool insecure = insecure_env && std::string(insecure_env) == "1";
+    if (!insecure) {
+      const char* ca = getenv("PS_OPERATOR_CA_FILE");
+      std::string ca_file = ca && *ca
+          ? ca
+          : "/var/run/secrets/kubernetes.io/serviceaccount/ca.crt";
+      int loaded = SSL_CTX_load_verify_locations(c.ctx, ca_file.c_str(),
+                                                 nullptr);
+      if (!loaded) loaded = SSL_CTX_set_default_verify_paths(c.ctx);
+      if (!loaded)
+        throw std::runtime_error(
+            "no CA bundle loadable (" + ca_file +
+            "); refusing unverified TLS with a Bearer token. "
+            "Set PS_OPERATOR_TLS_INSECURE=1 to override.");
+      SSL_CTX_set_verify(c.ctx, SSL_VERIFY_PEER, nullptr);
+    } else {
+      SSL_CTX_set_verify(c.ctx, SSL_VERIFY_NONE, nullptr);
+    }
     c.ssl = SSL_new(c.ctx);
     SSL_set_fd(c.ssl, c.fd);
     SSL_set_tlsext_host_name(c.ssl, u.host.c_str());
+    if (!insecure) SSL_set1_host(c.ssl, u.host.c_str());
     if (SSL_connect(c.ssl) != 1)
       throw std::runtime_error("TLS handshake failed: " + u.host);
   }

@@ -23,9 +23,9 @@ struct Url {
 Url parse_url(const std::string& url);
 
 // Performs one request. `token` (if non-empty) is sent as a Bearer header.
-// `ca_insecure` skips certificate verification (in-cluster CA handling is
-// out of scope for this build; production deployments mount the CA and use
-// verify mode).
+// TLS peers are verified against the mounted service-account CA
+// (PS_OPERATOR_CA_FILE overrides the path); set PS_OPERATOR_TLS_INSECURE=1
+// to explicitly skip verification (test clusters only).
 Response request(const std::string& method, const std::string& url,
                  const std::string& body, const std::string& token,
                  const std::string& content_type = "application/json",

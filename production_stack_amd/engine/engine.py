@@ -221,8 +221,11 @@ class LLMEngine:
             if first:
                 seq.first_token_time = now
             self.stats.generation_tokens += len(new)
-            text_delta = "".join(
-                self.tokenizer.decode_token(t) + " " for t in new
+            # decode only up to the streaming cursor: under async
+            # scheduling the tail may still hold -1 placeholders
+            text_delta = self.tokenizer.stream_decode(
+                seq.output_token_ids[: seq._stream_cursor],
+                seq.detok_state,
             ) if new else ""
             reason = None
             if seq.status is SeqStatus.FINISHED_STOPPED:

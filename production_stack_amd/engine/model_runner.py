@@ -795,13 +795,35 @@ class ModelRunner:
                 keep[:, 0] = True
                 sp = sp * keep
                 sp = sp / sp.sum(dim=-1, keepdim=True)
+                sp_cpu = sp.cpu()
                 picks = torch.multinomial(
-                    sp.cpu(), 1, generator=self._generator
+                    sp_cpu, 1, generator=self._generator
                 )
+                self._redraw_seeded(sp_cpu, picks, rows, params)
                 chosen = torch.gather(si.cpu(), 1, picks).flatten()
             else:
-                chosen = torch.multinomial(
-                    probs.cpu(), 1, generator=self._generator
-                ).flatten()
+                probs_cpu = probs.cpu()
+                picks = torch.multinomial(
+                    probs_cpu, 1, generator=self._generator
+                )
+                self._redraw_seeded(probs_cpu, picks, rows, params)
+                chosen = picks.flatten()
             result[torch.tensor(rows)] = chosen
         return result
+
+    def _redraw_seeded(
+        self, probs_cpu: torch.Tensor, picks: torch.Tensor,
+        rows: List[int], params: List,
+    ) -> None:
+        """OpenAI-parity per-request determinism: rows carrying
+        SamplingParams.seed draw from their own persistent generator
+        (seeded once per request) instead of the shared engine one."""
+        for j, i in enumerate(rows):
+            p = params[i]
+            if p.seed is None:
+                continue
+            g = getattr(p, "_seed_generator", None)
+            if g is None:
+                g = torch.Generator(device="cpu").manual_seed(int(p.seed))
+                p._seed_generator = g
+            picks[j] = torch.multinomial(probs_cpu[j], 1, generator=g)

@@ -54,6 +54,8 @@ class Sequence:
         self.num_cached_prompt_tokens = 0  # prefix-cache hits at admission
         # streaming cursor: outputs not yet handed to the consumer
         self._stream_cursor = 0
+        # incremental-detokenization offsets (tokenizer.stream_decode)
+        self.detok_state: dict = {}
 
     # ---- token accounting -------------------------------------------------
     @property

@@ -28,6 +28,25 @@ class BaseTokenizer:
     def decode_token(self, token_id: int) -> str:
         return self.decode([token_id])
 
+    def stream_decode(self, output_ids: List[int], state: dict) -> str:
+        """Incremental detokenization: given the full output-id list so far
+        and a per-sequence mutable state dict, return the newly decodable
+        text suffix.  Single-token decode loses BPE word-boundary markers,
+        so we decode a sliding window [prefix_offset:] and diff against the
+        previously decoded [prefix_offset:read_offset] text (the vLLM
+        scheme); a trailing U+FFFD means a multi-token UTF-8 sequence is
+        still incomplete and we hold the delta until the next token.
+        """
+        p = state.get("p", 0)
+        r = state.get("r", 0)
+        new_text = self.decode(output_ids[p:])
+        if new_text.endswith("�"):
+            return ""
+        prefix_text = self.decode(output_ids[p:r])
+        state["p"] = r
+        state["r"] = len(output_ids)
+        return new_text[len(prefix_text):]
+
 
 class SyntheticTokenizer(BaseTokenizer):
     """Stable word-level tokenizer: each whitespace word maps to an id by
@@ -54,6 +73,16 @@ class SyntheticTokenizer(BaseTokenizer):
         if token_id == self.eos_token_id:
             return ""
         return f"w{token_id}"
+
+    def stream_decode(self, output_ids: List[int], state: dict) -> str:
+        # word-per-id: every id is independently decodable; keep the
+        # historical trailing-space framing so accumulated deltas render
+        # as space-separated words.
+        r = state.get("r", 0)
+        state["r"] = len(output_ids)
+        return "".join(
+            self.decode_token(t) + " " for t in output_ids[r:]
+        )
 
 
 class HFTokenizer(BaseTokenizer):

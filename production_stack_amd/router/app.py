@@ -352,6 +352,8 @@ def initialize_all(app: FastAPI, args) -> None:
             port=args.k8s_port,
             label_selector=args.k8s_label_selector,
             api_key=args.api_key,
+            insecure_skip_tls_verify=getattr(
+                args, "k8s_insecure_skip_tls_verify", False),
         )
         app.state.model_aliases = {}
     elif args.service_discovery == "k8s_service_name":
@@ -361,6 +363,8 @@ def initialize_all(app: FastAPI, args) -> None:
             port=args.k8s_port,
             label_selector=args.k8s_label_selector,
             api_key=args.api_key,
+            insecure_skip_tls_verify=getattr(
+                args, "k8s_insecure_skip_tls_verify", False),
         )
         app.state.model_aliases = {}
     else:

@@ -40,6 +40,9 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     p.add_argument("--k8s-namespace", default="default")
     p.add_argument("--k8s-port", type=int, default=8000)
     p.add_argument("--k8s-label-selector", default=None)
+    p.add_argument("--k8s-insecure-skip-tls-verify", action="store_true",
+                   help="explicitly disable K8s API TLS verification "
+                        "(test clusters only; never silently implied)")
 
     # routing
     p.add_argument(

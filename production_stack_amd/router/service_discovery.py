@@ -208,6 +208,35 @@ class StaticServiceDiscovery(ServiceDiscovery):
             self._thread.join(timeout=2)
 
 
+def _resolve_tls_verify(
+    sa_dir: str, token: str, insecure: bool
+):
+    """TLS verify policy for K8s API calls. Never silently downgrade to
+    verify=False while a Bearer token is being sent (credential exposure
+    to a MITM): use the mounted SA CA when present, else the system trust
+    store (fails closed on unknown CAs). verify=False only on the explicit
+    insecure flag, with a loud warning."""
+    import os
+
+    ca = f"{sa_dir}/ca.crt"
+    if insecure:
+        logger.warning(
+            "TLS verification DISABLED for the K8s API "
+            "(insecure_skip_tls_verify=true); Bearer credentials are "
+            "exposed to any on-path attacker"
+        )
+        return False
+    if os.path.exists(ca):
+        return ca
+    if token:
+        logger.warning(
+            "service-account CA bundle %s missing; falling back to the "
+            "system trust store (set insecure_skip_tls_verify=true to "
+            "skip verification explicitly)", ca,
+        )
+    return True
+
+
 class K8sPodIpServiceDiscovery(ServiceDiscovery):
     """Watches pods matching a label selector; endpoint per ready pod IP.
 
@@ -231,6 +260,7 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
         api_base: Optional[str] = None,
         sa_token: Optional[str] = None,
         probe_models: bool = True,
+        insecure_skip_tls_verify: bool = False,
     ) -> None:
         self.namespace = namespace
         self.port = port
@@ -245,12 +275,8 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
                     self._token = f.read().strip()
             except OSError:
                 self._token = ""
-        import os
-
-        self._verify = (
-            f"{self.SA_DIR}/ca.crt"
-            if os.path.exists(f"{self.SA_DIR}/ca.crt")
-            else False
+        self._verify = _resolve_tls_verify(
+            self.SA_DIR, self._token, insecure_skip_tls_verify
         )
         self._endpoints: Dict[str, EndpointInfo] = {}
         self._lock = threading.Lock()
@@ -289,7 +315,11 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
         except requests.RequestException:
             return []
 
-    def _apply_pod(self, ev_type: str, pod: dict) -> None:
+    def _build_endpoint(self, pod: dict):
+        """Build (name, EndpointInfo|None) for a pod — including the
+        blocking /v1/models probe — WITHOUT holding self._lock, so the
+        routing hot path (get_endpoint_info) is never stalled behind a
+        5 s HTTP probe."""
         meta = pod.get("metadata", {})
         status = pod.get("status", {})
         name = meta.get("name", "")
@@ -300,19 +330,24 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
             and all(c.get("ready") for c in statuses)
             and meta.get("deletionTimestamp") is None
         )
-        url = f"http://{ip}:{self.port}" if ip else None
+        if not ready or not ip:
+            return name, None
+        labels = meta.get("labels") or {}
+        return name, EndpointInfo(
+            url=f"http://{ip}:{self.port}",
+            model_names=self._get_model_names(ip),
+            model_label=labels.get("model"),
+            pod_name=name,
+            sleep=labels.get("sleeping") == "true",
+        )
+
+    def _apply_pod(self, ev_type: str, pod: dict) -> None:
+        name, info = self._build_endpoint(pod)  # probe outside the lock
         with self._lock:
-            if ev_type == "DELETED" or not ready or not url:
+            if ev_type == "DELETED" or info is None:
                 self._endpoints.pop(name, None)
             else:
-                labels = meta.get("labels") or {}
-                self._endpoints[name] = EndpointInfo(
-                    url=url,
-                    model_names=self._get_model_names(ip),
-                    model_label=labels.get("model"),
-                    pod_name=name,
-                    sleep=labels.get("sleeping") == "true",
-                )
+                self._endpoints[name] = info
 
     def _list_once(self) -> None:
         r = requests.get(self._pods_url(), headers=self._headers(),
@@ -322,17 +357,26 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
         self._resource_version = body.get("metadata", {}).get(
             "resourceVersion", ""
         )
-        with self._lock:
-            self._endpoints.clear()
+        # build the refreshed map off to the side (probes included) and
+        # swap atomically — the router must never observe a cleared or
+        # half-rebuilt endpoint set mid-refresh
+        new: Dict[str, EndpointInfo] = {}
         for pod in body.get("items", []):
-            self._apply_pod("ADDED", pod)
+            name, info = self._build_endpoint(pod)
+            if info is not None:
+                new[name] = info
+        with self._lock:
+            self._endpoints = new
 
     def _watch_pods(self) -> None:
         import json as _json
 
+        need_list = True
         while not self._stop.is_set():
             try:
-                self._list_once()
+                if need_list:
+                    self._list_once()
+                    need_list = False
                 sep = "&" if "?" in self._pods_url() else "?"
                 url = (
                     f"{self._pods_url()}{sep}watch=true"
@@ -343,6 +387,9 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
                     url, headers=self._headers(), verify=self._verify,
                     stream=True, timeout=40,
                 ) as r:
+                    if r.status_code == 410:  # resourceVersion expired
+                        need_list = True
+                        continue
                     r.raise_for_status()
                     # chunk_size=1: watch events are sparse; the default
                     # 512-byte buffering would sit on an event until more
@@ -353,13 +400,22 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
                         if not line:
                             continue
                         ev = _json.loads(line)
-                        self._apply_pod(
-                            ev.get("type", ""), ev.get("object", {})
-                        )
+                        if ev.get("type") == "ERROR":
+                            # typically 410 Gone inside the stream
+                            need_list = True
+                            break
+                        obj = ev.get("object", {})
+                        rv = obj.get("metadata", {}).get("resourceVersion")
+                        if rv:
+                            # continue the watch from here on reconnect
+                            # instead of re-listing every ~30 s cycle
+                            self._resource_version = rv
+                        self._apply_pod(ev.get("type", ""), obj)
             except Exception as e:
                 if self._stop.is_set():
                     return
                 logger.warning("k8s watch error: %s", e)
+                need_list = True
                 time.sleep(1)
 
     def get_endpoint_info(self) -> List[EndpointInfo]:
@@ -387,6 +443,7 @@ class K8sServiceNameServiceDiscovery(ServiceDiscovery):
         api_base: Optional[str] = None,
         sa_token: Optional[str] = None,
         probe_models: bool = True,
+        insecure_skip_tls_verify: bool = False,
     ) -> None:
         self.namespace = namespace
         self.port = port
@@ -402,12 +459,8 @@ class K8sServiceNameServiceDiscovery(ServiceDiscovery):
                     self._token = f.read().strip()
             except OSError:
                 self._token = ""
-        import os
-
-        self._verify = (
-            f"{self.SA_DIR}/ca.crt"
-            if os.path.exists(f"{self.SA_DIR}/ca.crt")
-            else False
+        self._verify = _resolve_tls_verify(
+            self.SA_DIR, self._token, insecure_skip_tls_verify
         )
         self._endpoints: Dict[str, EndpointInfo] = {}
         self._lock = threading.Lock()
