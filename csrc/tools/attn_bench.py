@@ -17,6 +17,7 @@ from production_stack_amd import _C
 
 def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20, variant=4):
     bs = 16
+    tile = 256 if variant == 5 else 64
     nblocks = ctx_len // bs + 1
     k = torch.randn(nblocks + 1, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
     v = torch.randn_like(k)
@@ -24,8 +25,8 @@ def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20, variant=4):
     T = ctx_len
     q = torch.randn(T, qh, hd, dtype=torch.bfloat16, device="cuda")
     tiles = []
-    for t0 in range(0, T, 64):
-        tiles.append([0, t0, t0, min(64, T - t0)])
+    for t0 in range(0, T, tile):
+        tiles.append([0, t0, t0, min(tile, T - t0)])
     tiles = torch.tensor(tiles, dtype=torch.int32, device="cuda")
     out = torch.empty_like(q)
     scale = hd ** -0.5

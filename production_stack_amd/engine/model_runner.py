@@ -214,7 +214,7 @@ class ModelRunner:
 
         import numpy as np
 
-        TILE = 64
+        TILE = ops.PREFILL_TILE  # 256 for the 8-wave 32x32 kernel (v5)
         for row, ss in enumerate(prefills):
             seq = ss.seq
             all_ids = seq.token_ids()

@@ -8,6 +8,8 @@ the engine and its tests work in GPU-less CI.
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from production_stack_amd.ops import reference
@@ -158,6 +160,13 @@ def paged_attn_prefill(
     )
 
 
+# Default MFMA prefill variant: 5 = 8-wave 32x32 swapped-QK^T kernel
+# (csrc/prefill_mfma32.hip, 256-row tiles); 3/4 = the 4-wave 16x16 kernel
+# (64-row tiles). Overridable for A/B via PS_PREFILL_VARIANT.
+PREFILL_VARIANT = int(os.environ.get("PS_PREFILL_VARIANT", "5"))
+PREFILL_TILE = 256 if PREFILL_VARIANT == 5 else 64
+
+
 def paged_attn_prefill_mfma(
     q: torch.Tensor,
     k_cache: torch.Tensor,
@@ -166,14 +175,15 @@ def paged_attn_prefill_mfma(
     tile_info: torch.Tensor,
     scale: float,
     window: int = 0,
+    variant: int = None,
 ) -> torch.Tensor:
-    """MFMA-tiled chunked prefill (GPU, head_dim 128 only)."""
+    """MFMA-tiled chunked prefill (GPU, head_dim 128 only). tile_info rows
+    must be built with n_rows <= PREFILL_TILE for the active variant."""
     _require_ext()
     out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-    # variant 3 = 3-waves/SIMD bound: best measured (222 TF @ctx4096,
-    # profiles/attn_bench_prefill)
     _C.paged_attn_prefill_mfma(
-        out, q, k_cache, v_cache, block_tables, tile_info, scale, 3, window
+        out, q, k_cache, v_cache, block_tables, tile_info, scale,
+        PREFILL_VARIANT if variant is None else variant, window
     )
     return out
 

@@ -21,6 +21,7 @@ ext = CUDAExtension(
         "csrc/norm_act_rope.hip",
         "csrc/attention.hip",
         "csrc/prefill_mfma.hip",
+        "csrc/prefill_mfma32.hip",
         "csrc/skinny_gemm.hip",
         "csrc/lora_bgmv.hip",
     ],

@@ -550,3 +550,111 @@ def test_windowed_prefill_mfma_matches_reference():
         q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), token_seq,
         token_pos, 0.0883, window=W)
     _close(got.cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("qh,kh", [(32, 8), (16, 2)])
+def test_prefill_mfma32_v5_vs_reference(qh, kh):
+    """v5 (8-wave 32x32 swapped-QK^T, 256-row tiles) vs fp32 reference:
+    fresh prefill, chunked continuation, ragged tails, 1-token chunk."""
+    torch.manual_seed(40)
+    hd, bs = 128, 16
+    chunks = [(0, 0, 600), (1, 256, 300), (2, 0, 1), (3, 100, 130)]
+    max_blocks = 64
+    nb = 4 * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    block_tables = torch.arange(1, 4 * max_blocks + 1, dtype=torch.int32).reshape(
+        4, max_blocks
+    )
+    tiles, token_seq, token_pos = _build_tiles(chunks, tile=256)
+    T = token_seq.shape[0]
+    q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_prefill(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), block_tables, token_seq,
+        token_pos, 1.0 / hd ** 0.5,
+    )
+    got = ops.paged_attn_prefill_mfma(
+        q, k_cache, v_cache, block_tables.cuda(), tiles.cuda(),
+        1.0 / hd ** 0.5, variant=5,
+    )
+    _close(got, want)
+
+
+def test_prefill_mfma32_v5_matches_v3():
+    torch.manual_seed(41)
+    qh, kh, hd = 32, 8, 128
+    chunks = [(0, 2048, 1024)]
+    max_blocks = 192
+    nb = max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, 16, hd)
+    block_tables = torch.arange(1, max_blocks + 1, dtype=torch.int32).reshape(
+        1, max_blocks
+    )
+    t5, _, _ = _build_tiles(chunks, tile=256)
+    t3, _, _ = _build_tiles(chunks, tile=64)
+    T = 1024
+    q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
+    a = ops.paged_attn_prefill_mfma(
+        q, k_cache, v_cache, block_tables.cuda(), t5.cuda(), 0.0883883,
+        variant=5,
+    )
+    b = ops.paged_attn_prefill_mfma(
+        q, k_cache, v_cache, block_tables.cuda(), t3.cuda(), 0.0883883,
+        variant=3,
+    )
+    _close(a, b, atol=3e-2, rtol=3e-2)
+
+
+def test_prefill_mfma32_v5_fp8_kv():
+    from production_stack_amd import _C
+
+    torch.manual_seed(42)
+    qh, kh, hd, bs = 32, 8, 128, 16
+    chunks = [(0, 0, 300), (1, 64, 80)]
+    max_blocks = 32
+    nb = 2 * max_blocks + 1
+    k_cache, v_cache = _make_cache(nb, kh, bs, hd)
+    k8 = k_cache.to(torch.float8_e4m3fn)
+    v8 = v_cache.to(torch.float8_e4m3fn)
+    block_tables = torch.arange(1, 2 * max_blocks + 1, dtype=torch.int32).reshape(
+        2, max_blocks
+    )
+    tiles, token_seq, token_pos = _build_tiles(chunks, tile=256)
+    T = token_seq.shape[0]
+    q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
+    want = reference.paged_attn_prefill(
+        q.cpu(), k8.cpu().to(torch.bfloat16), v8.cpu().to(torch.bfloat16),
+        block_tables, token_seq, token_pos, 1.0 / hd ** 0.5,
+    )
+    out = torch.empty_like(q)
+    _C.paged_attn_prefill_mfma(
+        out, q, k8, v8, block_tables.cuda(), tiles.cuda(),
+        1.0 / hd ** 0.5, 5,
+    )
+    _close(out, want, atol=3e-2, rtol=3e-2)
+
+
+def test_prefill_mfma32_v5_windowed():
+    from production_stack_amd.ops import reference
+
+    torch.manual_seed(43)
+    qh, kh, hd, bs = 8, 2, 128, 16
+    W = 160
+    ctx = 640
+    nb = ctx // bs + 1
+    k_cache = torch.randn(nb, kh, bs, hd, dtype=torch.bfloat16,
+                          device="cuda") / 4
+    v_cache = torch.randn_like(k_cache) / 4
+    bt = torch.arange(1, nb, dtype=torch.int32, device="cuda").view(1, -1)
+    q = torch.randn(ctx, qh, hd, dtype=torch.bfloat16, device="cuda") / 4
+    tiles = []
+    for t0 in range(0, ctx, 256):
+        tiles.append([0, t0, t0, min(256, ctx - t0)])
+    tile_info = torch.tensor(tiles, dtype=torch.int32, device="cuda")
+    got = ops.paged_attn_prefill_mfma(q, k_cache, v_cache, bt, tile_info,
+                                      0.0883, window=W, variant=5)
+    token_seq = torch.zeros(ctx, dtype=torch.int32)
+    token_pos = torch.arange(ctx, dtype=torch.int32)
+    want = reference.paged_attn_prefill(
+        q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), token_seq,
+        token_pos, 0.0883, window=W)
+    _close(got.cpu(), want, atol=3e-2, rtol=3e-2)
