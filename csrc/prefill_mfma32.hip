@@ -34,7 +34,6 @@ PS_DEV ps_mbf16x8 ps_as_mbf16_32(ps_bf16x8 u) {
 
 #define PS_CHUNK32 64    // KV tokens per chunk (4 pages)
 #define PS_TILE32 256    // q rows per workgroup (8 waves x 32)
-#define PS_PSTRIDE 64    // P row stride in tokens
 
 // tile_info: int4 per tile = (seq_row, q_token_start, q_pos_start, n_rows)
 template <int HEAD_DIM, typename KVT>
@@ -83,7 +82,7 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
 
   __shared__ __align__(16) unsigned short k_lds[PS_CHUNK32][D];
   __shared__ __align__(16) unsigned short v_t[D][PS_CHUNK32];
-  __shared__ __align__(16) unsigned short p_lds[8][32][PS_PSTRIDE];
+  __shared__ __align__(16) unsigned short p_lds[8][32][64];
 
   // ---- Q fragments: B-operand, lane supplies Q[q=rc][d=ks*16+hi*8+i] ----
   const int q_row_clamped = min(wq0 + rc, n_rows - 1);
@@ -115,7 +114,8 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
       window > 0 ? max(0, (q_pos0 + wq0) - window + 1) : 0;
 
   // ---- staging assignments (512 threads) ----
-  // K: thread t stages 16B: token t>>3, dim slot t&7 (XOR-swizzled).
+  // K: 64 tokens x 16 slots of 8 dims = 1024 16-B segments -> each thread
+  //    stages TWO segments (slots t&7 and (t&7)+8) of token t>>3.
   // V: thread t stages 4 tokens x 4 dims register-transposed:
   //    token quad tq = t&15, dim slice ds = t>>4.
   const int ktok = tid >> 3, kslot = tid & 7;
@@ -131,21 +131,25 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
     const long pg = bt[min(tok / BS, n_pages - 1)];
     return v_cache + ((pg * KH + kvh) * BS + (tok & (BS - 1))) * D + vds * 4;
   };
-  kvec8 kstage;
+  kvec8 kstage[2];
   kvec4 vstage[4];
   {
     const KVT* ks = k_seg_ptr(0);
     const KVT* vq = v_quad_ptr(0);
-    kstage = *(const kvec8*)ks;
+    kstage[0] = *(const kvec8*)ks;
+    kstage[1] = *(const kvec8*)(ks + 64);
 #pragma unroll
     for (int i = 0; i < 4; i++) vstage[i] = *(const kvec4*)(vq + i * D);
   }
 
   for (int chunk = 0; chunk < n_chunks; chunk++) {
     const int tok0 = chunk * PS_CHUNK32;
-    // write prefetched K segment (16B-slot XOR swizzle by token low bits)
+    // write prefetched K segments (16B-slot XOR swizzle by token low bits;
+    // the XOR only permutes the low 3 slot bits, so slot+8 stays in 8..15)
     *(ps_bf16x8*)(&k_lds[ktok][(kslot ^ (ktok & 7)) * 8]) =
-        KVTr::to_bf16x8(kstage);
+        KVTr::to_bf16x8(kstage[0]);
+    *(ps_bf16x8*)(&k_lds[ktok][((kslot ^ (ktok & 7)) + 8) * 8]) =
+        KVTr::to_bf16x8(kstage[1]);
     // write prefetched V quad transposed (octet swizzle by dim low bits)
     {
       const int vo = vtq >> 1;  // token octet
@@ -164,7 +168,8 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
     if (chunk + 1 < n_chunks) {  // T14: next chunk's loads under compute
       const KVT* ks = k_seg_ptr(chunk + 1);
       const KVT* vq = v_quad_ptr(chunk + 1);
-      kstage = *(const kvec8*)ks;
+      kstage[0] = *(const kvec8*)ks;
+      kstage[1] = *(const kvec8*)(ks + 64);
 #pragma unroll
       for (int i = 0; i < 4; i++) vstage[i] = *(const kvec4*)(vq + i * D);
     }
@@ -251,7 +256,9 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
       l_run += psum + __shfl_xor(psum, 32);
       // ---- P to per-wave LDS (S-layout -> A-frag slots, swizzled) ----
       // reg quad r=4qd..4qd+3 holds 4 consecutive kv tokens -> one b64
-      // store per (kt, qd): 8 vector stores/lane instead of 32 scalar
+      // store per (kt, qd): 8 vector stores/lane. (An in-register
+      // T12-style shfl_xor exchange measured 7% SLOWER than this path --
+      // the vectorized LDS round-trip is cheap at this occupancy.)
 #pragma unroll
       for (int kt = 0; kt < 2; kt++)
 #pragma unroll

@@ -84,3 +84,42 @@ if __name__ == "__main__":
             bench_decode(batch=b, ctx=2048, variant=v)
     for b in (64, 256):
         bench_decode(batch=b, ctx=2048, variant=1, fp8=True)
+
+
+def bench_prefill_serving(nseq=10, hist=2000, rows=200, qh=32, kh=8, hd=128,
+                          iters=20, variant=5):
+    """Serving-shaped chunked continuation: nseq sequences each prefilling
+    `rows` new tokens on top of `hist` cached tokens (multi-round-QA shape)."""
+    bs = 16
+    tile = 256 if variant == 5 else 64
+    ctx = hist + rows
+    per = (ctx + bs - 1) // bs
+    nb = nseq * per + 1
+    k = torch.randn(nb, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn_like(k)
+    bt = torch.arange(1, nseq * per + 1, dtype=torch.int32,
+                      device="cuda").reshape(nseq, per)
+    T = nseq * rows
+    q = torch.randn(T, qh, hd, dtype=torch.bfloat16, device="cuda")
+    tiles = []
+    flat = 0
+    for s in range(nseq):
+        for t0 in range(0, rows, tile):
+            tiles.append([s, flat + t0, hist + t0, min(tile, rows - t0)])
+        flat += rows
+    tiles = torch.tensor(tiles, dtype=torch.int32, device="cuda")
+    out = torch.empty_like(q)
+    scale = hd ** -0.5
+    for _ in range(3):
+        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale, variant)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        _C.paged_attn_prefill_mfma(out, q, k, v, bt, tiles, scale, variant)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    # flops: per row, attended kv = hist + row_idx + 1
+    att = sum(hist + r + 1 for r in range(rows)) * nseq
+    flops = 4 * hd * qh * att
+    print(f"serving v{variant} nseq={nseq} hist={hist} rows={rows}: "
+          f"{dt*1e3:.3f} ms  {flops/dt/1e12:.1f} TF")
