@@ -57,6 +57,9 @@ void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
 int ps_skinny_gemm_splits(int M, int N, int K);
 int ps_skinny_gemm(void* out_bf16, void* ws, const void* x, const void* w,
                    int M, int N, int K, long x_stride, hipStream_t stream);
+int ps_gemm8p_splits(int M, int N, int K);
+int ps_gemm8p(void* out_bf16, void* ws, const void* x, const void* w, int M,
+              int N, int K, long x_stride, hipStream_t stream);
 int ps_lora_bgmv(void* out, const void* x, const void* A, const void* B,
                  const void* scale, const void* idx, int T, int IN, int W,
                  int R, long out_stride, long x_stride, int col_off,
@@ -306,6 +309,30 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w) {
   TORCH_CHECK(rc == 0, "skinny gemm launch failed");
 }
 
+void gemm8p(at::Tensor out, at::Tensor x, at::Tensor w) {
+  CHECK_GPU_BF16(out);
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
+              "x must be bf16");
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1, "x must be 2D row-major");
+  CHECK_GPU_BF16(w);
+  const int M = (int)x.size(0);
+  const int K = (int)x.size(1);
+  const int N = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  const int splits = ps_gemm8p_splits(M, N, K);
+  TORCH_CHECK(splits > 0, "unsupported gemm8p shape M=", M, " N=", N,
+              " K=", K);
+  void* ws_p = nullptr;
+  at::Tensor ws;
+  if (splits > 1) {
+    ws = at::empty({(long)splits * M * N}, x.options().dtype(at::kFloat));
+    ws_p = ws.data_ptr();
+  }
+  int rc = ps_gemm8p(out.data_ptr(), ws_p, x.data_ptr(), w.data_ptr(), M, N,
+                     K, x.stride(0), current_stream());
+  TORCH_CHECK(rc == 0, "gemm8p launch failed");
+}
+
 void lora_bgmv(at::Tensor out, at::Tensor x, at::Tensor A, at::Tensor B,
                at::Tensor scale, at::Tensor idx, long col_off) {
   CHECK_GPU_BF16(out);
@@ -394,6 +421,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");
   m.def("kv_quant", &kv_quant, "Row-wise int8 KV quantization");
+  m.def("gemm8p", &gemm8p,
+        "8-phase deep-pipelined bf16 GEMM (out = x @ w.T)");
   m.def("skinny_gemm", &skinny_gemm,
         "Split-K MFMA GEMM for decode-shaped (M<=128) projections");
   m.def("kv_dequant", &kv_dequant, "Row-wise int8 KV dequantization");

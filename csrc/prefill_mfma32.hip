@@ -36,7 +36,12 @@ PS_DEV ps_mbf16x8 ps_as_mbf16_32(ps_bf16x8 u) {
 #define PS_TILE32 256    // q rows per workgroup (8 waves x 32)
 
 // tile_info: int4 per tile = (seq_row, q_token_start, q_pos_start, n_rows)
-template <int HEAD_DIM, typename KVT>
+// GQW = GQA q-heads processed per workgroup: the 8 waves split into
+// GQW head-columns x (8/GQW) row-halves, so all GQW heads of a kv-head
+// share ONE K/V staging pass (serving-shape prefill chunks are
+// staging-bound: a 200-row continuation re-reads a 2000+-token context).
+// Tile rows = 32 * (8/GQW).
+template <int HEAD_DIM, int GQW, typename KVT>
 __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
     unsigned short* __restrict__ out,            // [T, QH, HD]
     const unsigned short* __restrict__ q,        // [T, QH, HD]
@@ -49,20 +54,21 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
   using KVTr = ps_kv_traits<KVT>;
   using kvec8 = typename KVTr::vec8;
   using kvec4 = typename KVTr::vec4;
-  constexpr int D = HEAD_DIM;  // 128
-  constexpr int BS = 16;       // page size in tokens
-  constexpr int NKS = D / 16;  // QK k-steps over head dim (8)
+  constexpr int D = HEAD_DIM;   // 128
+  constexpr int BS = 16;        // page size in tokens
+  constexpr int NKS = D / 16;   // QK k-steps over head dim (8)
+  constexpr int ROWH = 8 / GQW; // row-halves (32 rows each) per tile
 
   // T1 XCD swizzle (same scheme as v3/v4): contiguous work chunks per XCD
-  const int W = n_work;
+  const int W = n_work;  // n_tiles * KH * (GQ/GQW)
   const int cpx = (W + 7) >> 3;
   const int w = (blockIdx.x & 7) * cpx + (blockIdx.x >> 3);
   if (w >= W) return;
-  const int n_tiles = W / QH;
+  const int n_grp = GQ / GQW;
+  const int n_tiles = W / (n_grp * KH);
   const int tile = w % n_tiles;
-  const int gq = (w / n_tiles) % GQ;
-  const int kvh = w / (n_tiles * GQ);
-  const int qh = kvh * GQ + gq;
+  const int g_grp = (w / n_tiles) % n_grp;
+  const int kvh = w / (n_tiles * n_grp);
   const int seq_row = tile_info[tile * 4 + 0];
   const int q_tok0 = tile_info[tile * 4 + 1];
   const int q_pos0 = tile_info[tile * 4 + 2];
@@ -73,7 +79,9 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
   const int lane = tid & 63;
   const int hi = lane >> 5;   // half-wave
   const int rc = lane & 31;   // q row (B/D col) or kv row (A row)
-  const int wq0 = wave * 32;  // wave's first local q row
+  // wave -> (head column, row half)
+  const int qh = kvh * GQ + g_grp * GQW + (wave / ROWH);
+  const int wq0 = (wave % ROWH) * 32;  // wave's first local q row
 
   const int* bt = block_tables + (long)seq_row * max_blocks;
   const int ctx_limit = q_pos0 + n_rows;
@@ -319,23 +327,30 @@ int ps_paged_attn_prefill_mfma32(void* out, const void* q,
                                  long q_stride, int kv_fp8, int window,
                                  hipStream_t stream) {
   if (head_dim != 128) return -1;
-  const int n_work = num_tiles * num_q_heads;
+  // GQW = largest power-of-2 divisor of GQ, capped at 4 so the tile row
+  // capacity 32*(8/GQW) never drops below 64 (the engine and tests build
+  // 64..256-row tiles via ops.prefill_tile_rows).
+  int gqw = 1;
+  while (gqw < 4 && GQ % (gqw * 2) == 0) gqw *= 2;
+  const int n_work = num_tiles * KH * (GQ / gqw);
   dim3 grid(((n_work + 7) / 8) * 8);
+#define PS_PF32_LAUNCH(GQW, KVT)                                          \
+  paged_attn_prefill_mfma32_kernel<128, GQW, KVT>                         \
+      <<<grid, 512, 0, stream>>>(                                         \
+          (unsigned short*)out, (const unsigned short*)q,                 \
+          (const KVT*)k_cache, (const KVT*)v_cache,                       \
+          (const int*)block_tables, (const int*)tile_info, max_blocks,    \
+          scale, KH, GQ, q_stride, num_q_heads, n_work, window)
   if (kv_fp8) {
-    paged_attn_prefill_mfma32_kernel<128, unsigned char>
-        <<<grid, 512, 0, stream>>>(
-            (unsigned short*)out, (const unsigned short*)q,
-            (const unsigned char*)k_cache, (const unsigned char*)v_cache,
-            (const int*)block_tables, (const int*)tile_info, max_blocks,
-            scale, KH, GQ, q_stride, num_q_heads, n_work, window);
+    if (gqw == 4) PS_PF32_LAUNCH(4, unsigned char);
+    else if (gqw == 2) PS_PF32_LAUNCH(2, unsigned char);
+    else PS_PF32_LAUNCH(1, unsigned char);
   } else {
-    paged_attn_prefill_mfma32_kernel<128, unsigned short>
-        <<<grid, 512, 0, stream>>>(
-            (unsigned short*)out, (const unsigned short*)q,
-            (const unsigned short*)k_cache, (const unsigned short*)v_cache,
-            (const int*)block_tables, (const int*)tile_info, max_blocks,
-            scale, KH, GQ, q_stride, num_q_heads, n_work, window);
+    if (gqw == 4) PS_PF32_LAUNCH(4, unsigned short);
+    else if (gqw == 2) PS_PF32_LAUNCH(2, unsigned short);
+    else PS_PF32_LAUNCH(1, unsigned short);
   }
+#undef PS_PF32_LAUNCH
   return 0;
 }
 

@@ -16,8 +16,9 @@ from production_stack_amd import _C
 
 
 def bench_prefill(ctx_len=4096, qh=32, kh=8, hd=128, iters=20, variant=4):
+    from production_stack_amd import ops
     bs = 16
-    tile = 256 if variant == 5 else 64
+    tile = ops.prefill_tile_rows(qh, kh) if variant == 5 else 64
     nblocks = ctx_len // bs + 1
     k = torch.randn(nblocks + 1, kh, bs, hd, dtype=torch.bfloat16, device="cuda")
     v = torch.randn_like(k)
@@ -90,8 +91,9 @@ def bench_prefill_serving(nseq=10, hist=2000, rows=200, qh=32, kh=8, hd=128,
                           iters=20, variant=5):
     """Serving-shaped chunked continuation: nseq sequences each prefilling
     `rows` new tokens on top of `hist` cached tokens (multi-round-QA shape)."""
+    from production_stack_amd import ops
     bs = 16
-    tile = 256 if variant == 5 else 64
+    tile = ops.prefill_tile_rows(qh, kh) if variant == 5 else 64
     ctx = hist + rows
     per = (ctx + bs - 1) // bs
     nb = nseq * per + 1

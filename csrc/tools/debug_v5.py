@@ -27,8 +27,9 @@ def run_case(name, T, qh=1, kh=1, hd=128, kzero=False, vones=False,
     bt = torch.arange(1, nb, dtype=torch.int32, device="cuda").view(1, -1)
     q = torch.randn(T, qh, hd, dtype=torch.bfloat16, device="cuda") / 4
     tiles = []
-    for t0 in range(0, T, 256):
-        tiles.append([0, t0, start + t0, min(256, T - t0)])
+    TR = ops.prefill_tile_rows(qh, kh)
+    for t0 in range(0, T, TR):
+        tiles.append([0, t0, start + t0, min(TR, T - t0)])
     tiles = torch.tensor(tiles, dtype=torch.int32, device="cuda")
     token_seq = torch.zeros(T, dtype=torch.int32)
     token_pos = torch.arange(start, start + T, dtype=torch.int32)

@@ -53,7 +53,7 @@ chunks = [(0, 2048, 1024)]
 max_blocks = 192
 k_cache, v_cache = make_cache(max_blocks + 1, kh, 16, hd)
 bt = torch.arange(1, max_blocks + 1, dtype=torch.int32).reshape(1, -1)
-t5, _, tpos = build_tiles(chunks, 256)
+t5, _, tpos = build_tiles(chunks, 64)
 t3, _, _ = build_tiles(chunks, 64)
 q = torch.randn((1024, qh, hd), dtype=torch.bfloat16, device="cuda")
 a = ops.paged_attn_prefill_mfma(q, k_cache, v_cache, bt.cuda(), t5.cuda(),
@@ -76,7 +76,7 @@ k2 = torch.randn(nb, 2, 16, hd, dtype=torch.bfloat16, device="cuda") / 4
 v2 = torch.randn_like(k2) / 4
 bt2 = torch.arange(1, nb, dtype=torch.int32, device="cuda").view(1, -1)
 q2 = torch.randn(ctx, 8, hd, dtype=torch.bfloat16, device="cuda") / 4
-t5w, _, _ = build_tiles([(0, 0, ctx)], 256)
+t5w, _, _ = build_tiles([(0, 0, ctx)], 64)
 gw = ops.paged_attn_prefill_mfma(q2, k2, v2, bt2, t5w.cuda(), 0.0883,
                                  window=W, variant=5)
 ww = reference.paged_attn_prefill(

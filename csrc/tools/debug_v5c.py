@@ -64,7 +64,7 @@ def case(name, chunks, qh, kh, tile, nseq, scale_div=1, seed=4):
 
 FULL = [(0, 0, 200), (1, 128, 100), (2, 0, 1), (3, 31, 64)]
 case("exact failing", FULL, 32, 8, 64, 4)
-case("tile=256", FULL, 32, 8, 256, 4)
+case("tile=128gq1", FULL, 8, 8, 128, 4)
 case("qh=kh=8", FULL, 8, 8, 64, 4)
 case("qh=kh=1", FULL, 1, 1, 64, 4)
 case("seq0 only", [(0, 0, 200)], 32, 8, 64, 1)
@@ -72,4 +72,4 @@ case("seq3 only", [(0, 31, 64)], 32, 8, 64, 1)
 case("seq1 only", [(0, 128, 100)], 32, 8, 64, 1)
 case("seq0 qh1", [(0, 0, 200)], 1, 1, 64, 1)
 case("seq0 div4", [(0, 0, 200)], 32, 8, 64, 1, scale_div=4)
-case("seq0 t256", [(0, 0, 200)], 32, 8, 256, 1)
+case("seq0 t256gq1", [(0, 0, 200)], 8, 8, 256, 1)

@@ -214,7 +214,9 @@ class ModelRunner:
 
         import numpy as np
 
-        TILE = ops.PREFILL_TILE  # 256 for the 8-wave 32x32 kernel (v5)
+        TILE = ops.prefill_tile_rows(
+            self.model_cfg.num_q_heads, self.model_cfg.num_kv_heads
+        )
         for row, ss in enumerate(prefills):
             seq = ss.seq
             all_ids = seq.token_ids()

@@ -161,10 +161,23 @@ def paged_attn_prefill(
 
 
 # Default MFMA prefill variant: 5 = 8-wave 32x32 swapped-QK^T kernel
-# (csrc/prefill_mfma32.hip, 256-row tiles); 3/4 = the 4-wave 16x16 kernel
-# (64-row tiles). Overridable for A/B via PS_PREFILL_VARIANT.
+# (csrc/prefill_mfma32.hip); 3/4 = the 4-wave 16x16 kernel (64-row
+# tiles). Overridable for A/B via PS_PREFILL_VARIANT.
 PREFILL_VARIANT = int(os.environ.get("PS_PREFILL_VARIANT", "5"))
 PREFILL_TILE = 256 if PREFILL_VARIANT == 5 else 64
+
+
+def prefill_tile_rows(num_q_heads: int, num_kv_heads: int) -> int:
+    """Prefill tile row count for the active kernel. v5 splits its 8 waves
+    into GQW head-columns x row-halves (GQW = largest pow2 divisor of the
+    GQA ratio, <=4), so tiles carry 32*(8/GQW) rows (64..256)."""
+    if PREFILL_VARIANT != 5:
+        return 64
+    gq = max(1, num_q_heads // max(1, num_kv_heads))
+    gqw = 1
+    while gqw < 4 and gq % (gqw * 2) == 0:
+        gqw *= 2
+    return 32 * (8 // gqw)
 
 
 def paged_attn_prefill_mfma(
@@ -234,6 +247,17 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         (x.shape[0], w.shape[0]), dtype=torch.bfloat16, device=x.device
     )
     _C.skinny_gemm(out, x, w)  # split-K partials + combine (no atomics)
+    return out
+
+
+def gemm8p(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """out = x @ w.T via the 8-phase deep-pipelined MFMA GEMM
+    (csrc/gemm8p.hip; 256x256 tiles + split-K for skinny M). GPU only."""
+    _require_ext()
+    out = torch.empty(
+        (x.shape[0], w.shape[0]), dtype=torch.bfloat16, device=x.device
+    )
+    _C.gemm8p(out, x, w)
     return out
 
 

@@ -20,12 +20,16 @@ import torch.nn.functional as F
 
 logger = logging.getLogger("ops.gemm_policy")
 
-# (M, N, K) -> True if the skinny kernel won
-_POLICY: Dict[Tuple[int, int, int], bool] = {}
+# (M, N, K) -> winning impl: 0 = hipBLASLt, 1 = skinny split-K, 2 = 8-phase
+_POLICY: Dict[Tuple[int, int, int], int] = {}
 
 
 def _eligible(M: int, N: int, K: int) -> bool:
     return M <= 256 and N % 64 == 0 and K % 64 == 0
+
+
+def _eligible_8p(M: int, N: int, K: int) -> bool:
+    return N % 256 == 0 and K % 128 == 0
 
 
 def _time_fn(fn, iters: int = 20) -> float:
@@ -50,22 +54,23 @@ def tune(weights, buckets, device) -> None:
         w_by_shape.setdefault((int(w.shape[0]), int(w.shape[1])), w)
     for M in buckets:
         for (N, K) in shapes:
-            if not _eligible(M, N, K):
-                _POLICY[(M, N, K)] = False
-                continue
             w = w_by_shape[(N, K)]
             x = torch.randn(M, K, dtype=torch.bfloat16, device=device)
-            t_blas = _time_fn(lambda: F.linear(x, w))
-            t_skinny = _time_fn(lambda: ops.skinny_gemm(x, w))
-            _POLICY[(M, N, K)] = t_skinny < t_blas
-            if t_skinny < t_blas:
+            cands = [(0, _time_fn(lambda: F.linear(x, w)))]
+            if _eligible(M, N, K):
+                cands.append((1, _time_fn(lambda: ops.skinny_gemm(x, w))))
+            if _eligible_8p(M, N, K):
+                cands.append((2, _time_fn(lambda: ops.gemm8p(x, w))))
+            impl, t = min(cands, key=lambda c: c[1])
+            _POLICY[(M, N, K)] = impl
+            if impl:
                 logger.info(
-                    "gemm M=%d N=%d K=%d: skinny %.1fus < blas %.1fus",
-                    M, N, K, t_skinny * 1e6, t_blas * 1e6,
+                    "gemm M=%d N=%d K=%d: impl %s wins at %.1fus",
+                    M, N, K, ("blas", "skinny", "8p")[impl], t * 1e6,
                 )
     logger.info(
-        "gemm autotune done: %d/%d cells use the skinny kernel",
-        sum(_POLICY.values()), len(_POLICY),
+        "gemm autotune done: %d/%d cells use an in-house kernel",
+        sum(1 for v in _POLICY.values() if v), len(_POLICY),
     )
 
 
@@ -73,10 +78,11 @@ def linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """F.linear with autotuned dispatch (no bias; Llama has none)."""
     if x.is_cuda and _POLICY:
         key = (x.shape[0], w.shape[0], w.shape[1])
-        if _POLICY.get(key):
+        impl = _POLICY.get(key, 0)
+        if impl:
             from production_stack_amd import ops
 
-            return ops.skinny_gemm(x, w)
+            return (ops.skinny_gemm if impl == 1 else ops.gemm8p)(x, w)
     return F.linear(x, w)
 
 

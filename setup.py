@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "csrc/prefill_mfma.hip",
         "csrc/prefill_mfma32.hip",
         "csrc/skinny_gemm.hip",
+        "csrc/gemm8p.hip",
         "csrc/lora_bgmv.hip",
     ],
     extra_compile_args={

@@ -565,7 +565,8 @@ def test_prefill_mfma32_v5_vs_reference(qh, kh):
     block_tables = torch.arange(1, 4 * max_blocks + 1, dtype=torch.int32).reshape(
         4, max_blocks
     )
-    tiles, token_seq, token_pos = _build_tiles(chunks, tile=256)
+    tiles, token_seq, token_pos = _build_tiles(
+        chunks, tile=ops.prefill_tile_rows(qh, kh))
     T = token_seq.shape[0]
     q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
     want = reference.paged_attn_prefill(
@@ -589,7 +590,7 @@ def test_prefill_mfma32_v5_matches_v3():
     block_tables = torch.arange(1, max_blocks + 1, dtype=torch.int32).reshape(
         1, max_blocks
     )
-    t5, _, _ = _build_tiles(chunks, tile=256)
+    t5, _, _ = _build_tiles(chunks, tile=ops.prefill_tile_rows(qh, kh))
     t3, _, _ = _build_tiles(chunks, tile=64)
     T = 1024
     q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
@@ -618,7 +619,8 @@ def test_prefill_mfma32_v5_fp8_kv():
     block_tables = torch.arange(1, 2 * max_blocks + 1, dtype=torch.int32).reshape(
         2, max_blocks
     )
-    tiles, token_seq, token_pos = _build_tiles(chunks, tile=256)
+    tiles, token_seq, token_pos = _build_tiles(
+        chunks, tile=ops.prefill_tile_rows(qh, kh))
     T = token_seq.shape[0]
     q = torch.randn((T, qh, hd), dtype=torch.bfloat16, device="cuda")
     want = reference.paged_attn_prefill(
@@ -646,9 +648,10 @@ def test_prefill_mfma32_v5_windowed():
     v_cache = torch.randn_like(k_cache) / 4
     bt = torch.arange(1, nb, dtype=torch.int32, device="cuda").view(1, -1)
     q = torch.randn(ctx, qh, hd, dtype=torch.bfloat16, device="cuda") / 4
+    tr = ops.prefill_tile_rows(qh, kh)
     tiles = []
-    for t0 in range(0, ctx, 256):
-        tiles.append([0, t0, t0, min(256, ctx - t0)])
+    for t0 in range(0, ctx, tr):
+        tiles.append([0, t0, t0, min(tr, ctx - t0)])
     tile_info = torch.tensor(tiles, dtype=torch.int32, device="cuda")
     got = ops.paged_attn_prefill_mfma(q, k_cache, v_cache, bt, tile_info,
                                       0.0883, window=W, variant=5)
