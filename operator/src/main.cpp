@@ -14,6 +14,7 @@
 #include <fstream>
 #include <string>
 
+#include "leader.h"
 #include "reconciler.h"
 
 static std::string read_file(const std::string& path) {
@@ -33,6 +34,8 @@ int main(int argc, char** argv) {
   std::string token_file =
       "/var/run/secrets/kubernetes.io/serviceaccount/token";
 
+  bool leader_elect = false;
+  int health_port = 0;
   for (int i = 1; i < argc; i++) {
     std::string a = argv[i];
     auto next = [&]() -> std::string {
@@ -43,10 +46,13 @@ int main(int argc, char** argv) {
     else if (a == "--token-file") token_file = next();
     else if (a == "--interval") interval = std::stoi(next());
     else if (a == "--once") once = true;
+    else if (a == "--leader-elect") leader_elect = true;
+    else if (a == "--health-port") health_port = std::stoi(next());
     else if (a == "--help") {
       printf(
           "psoperator [--api-server URL] [--namespace NS] [--token-file F]\n"
-          "           [--interval SEC] [--once]\n");
+          "           [--interval SEC] [--once] [--leader-elect]\n"
+          "           [--health-port P]\n");
       return 0;
     }
   }
@@ -62,15 +68,44 @@ int main(int argc, char** argv) {
   }
   ctx.token = read_file(token_file);
 
-  fprintf(stderr, "[psoperator] api=%s ns=%s interval=%ds once=%d\n",
-          ctx.api_server.c_str(), ctx.ns.c_str(), interval, (int)once);
+  fprintf(stderr,
+          "[psoperator] api=%s ns=%s interval=%ds once=%d leader-elect=%d\n",
+          ctx.api_server.c_str(), ctx.ns.c_str(), interval, (int)once,
+          (int)leader_elect);
+
+  psop::HealthServer health;
+  if (health_port > 0) {
+    if (health.start(health_port))
+      fprintf(stderr, "[psoperator] health/metrics on :%d\n", health_port);
+    else
+      fprintf(stderr, "[psoperator] health server failed on :%d\n",
+              health_port);
+  }
+
+  char host[256] = "psoperator";
+  gethostname(host, sizeof(host));
+  std::string identity =
+      std::string(host) + "_" + std::to_string((long)getpid());
 
   while (true) {
+    bool lead = true;
+    if (leader_elect) {
+      lead = psop::acquire_lease(ctx, identity, 2 * interval + 10);
+      health.is_leader = lead;
+      if (!lead)
+        fprintf(stderr, "[psoperator] standing by (lease held)\n");
+    } else {
+      health.is_leader = true;
+    }
     int actions = 0;
-    try {
-      actions = psop::reconcile_all(ctx);
-    } catch (const std::exception& e) {
-      fprintf(stderr, "[psoperator] reconcile error: %s\n", e.what());
+    if (lead) {
+      try {
+        actions = psop::reconcile_all(ctx);
+      } catch (const std::exception& e) {
+        fprintf(stderr, "[psoperator] reconcile error: %s\n", e.what());
+      }
+      health.reconcile_total++;
+      health.reconcile_actions_total += actions;
     }
     if (actions)
       fprintf(stderr, "[psoperator] applied %d change(s)\n", actions);

@@ -11,8 +11,11 @@
 
 #include "reconciler.h"
 
+#include <algorithm>
 #include <cstdio>
 #include <functional>
+#include <map>
+#include <vector>
 
 #include "http_client.h"
 
@@ -191,9 +194,138 @@ ValuePtr build_engine_deployment(const Ctx& ctx, const ValuePtr& cr) {
   res->set("requests", reqs);
   c->set("resources", res);
 
+  // storage PVC mount (/data, HF_HOME) + configData mount (/config)
+  auto storage = spec ? spec->get("storage") : nullptr;
+  auto cdata = spec ? spec->get("configData") : nullptr;
+  auto vmounts = Value::array();
+  auto vols = Value::array();
+  if (storage && storage->get_bool("enabled")) {
+    auto vm = Value::object();
+    vm->set("name", "model-storage");
+    vm->set("mountPath", "/data");
+    vmounts->push(vm);
+    auto v = Value::object();
+    v->set("name", "model-storage");
+    auto pvc = Value::object();
+    pvc->set("claimName", name + "-storage");
+    v->set("persistentVolumeClaim", pvc);
+    vols->push(v);
+    auto env = Value::array();
+    auto e = Value::object();
+    e->set("name", "HF_HOME");
+    e->set("value", "/data");
+    env->push(e);
+    c->set("env", env);
+  }
+  if (cdata) {
+    auto vm = Value::object();
+    vm->set("name", "engine-config");
+    vm->set("mountPath", "/config");
+    vmounts->push(vm);
+    auto v = Value::object();
+    v->set("name", "engine-config");
+    auto cm = Value::object();
+    cm->set("name", name + "-config");
+    v->set("configMap", cm);
+    vols->push(v);
+  }
+  if (!vmounts->arr.empty()) c->set("volumeMounts", vmounts);
+
   std::string app = name + "-engine";
   auto meta = metadata(name + "-engine", ctx.ns, app, hash_str(spec));
-  return deployment_skeleton(meta, app, replicas, c);
+  auto d = deployment_skeleton(meta, app, replicas, c);
+  if (!vols->arr.empty())
+    d->get("spec")->get("template")->get("spec")->set("volumes", vols);
+  return d;
+}
+
+// PVC for model weights (reference vllmruntime_controller.go:148-200):
+// spec.storage { enabled, size, storageClassName } -> <name>-storage PVC,
+// mounted at /data with HF_HOME pointing at it.
+ValuePtr build_engine_pvc(const Ctx& ctx, const ValuePtr& cr) {
+  auto spec = cr->get("spec");
+  auto storage = spec ? spec->get("storage") : nullptr;
+  if (!storage || !storage->get_bool("enabled")) return nullptr;
+  std::string name = cr->get("metadata")->get_str("name");
+  auto pvc = Value::object();
+  pvc->set("apiVersion", "v1");
+  pvc->set("kind", "PersistentVolumeClaim");
+  pvc->set("metadata", metadata(name + "-storage", ctx.ns, name + "-engine",
+                                hash_str(spec)));
+  auto pspec = Value::object();
+  auto modes = Value::array();
+  modes->push(Value::of("ReadWriteOnce"));
+  pspec->set("accessModes", modes);
+  auto res = Value::object();
+  auto req = Value::object();
+  req->set("storage", storage->get_str("size", "50Gi"));
+  res->set("requests", req);
+  pspec->set("resources", res);
+  if (!storage->get_str("storageClassName").empty())
+    pspec->set("storageClassName", storage->get_str("storageClassName"));
+  pvc->set("spec", pspec);
+  return pvc;
+}
+
+// ConfigMap carrying spec.configData (reference :1435); mounted at /config.
+ValuePtr build_engine_configmap(const Ctx& ctx, const ValuePtr& cr) {
+  auto spec = cr->get("spec");
+  auto data = spec ? spec->get("configData") : nullptr;
+  if (!data) return nullptr;
+  std::string name = cr->get("metadata")->get_str("name");
+  auto cm = Value::object();
+  cm->set("apiVersion", "v1");
+  cm->set("kind", "ConfigMap");
+  cm->set("metadata", metadata(name + "-config", ctx.ns, name + "-engine",
+                               hash_str(spec)));
+  cm->set("data", data);
+  return cm;
+}
+
+// KEDA ScaledObject on the engine deployment (reference :1201-1326):
+// spec.autoscaling { enabled, minReplicas, maxReplicas, idleReplicaCount,
+// prometheusAddress, threshold } -> prometheus trigger on
+// vllm:num_requests_waiting, matching the Helm chart's KEDA block.
+ValuePtr build_keda_scaledobject(const Ctx& ctx, const ValuePtr& cr) {
+  auto spec = cr->get("spec");
+  auto as = spec ? spec->get("autoscaling") : nullptr;
+  if (!as || !as->get_bool("enabled")) return nullptr;
+  std::string name = cr->get("metadata")->get_str("name");
+  auto so = Value::object();
+  so->set("apiVersion", "keda.sh/v1alpha1");
+  so->set("kind", "ScaledObject");
+  so->set("metadata", metadata(name + "-scaler", ctx.ns, name + "-engine",
+                               hash_str(spec)));
+  auto sspec = Value::object();
+  auto target = Value::object();
+  target->set("apiVersion", "apps/v1");
+  target->set("kind", "Deployment");
+  target->set("name", name + "-engine");
+  sspec->set("scaleTargetRef", target);
+  sspec->set("minReplicaCount", (int)as->get_num("minReplicas", 1));
+  sspec->set("maxReplicaCount", (int)as->get_num("maxReplicas", 8));
+  if (as->get("idleReplicaCount"))
+    sspec->set("idleReplicaCount", (int)as->get_num("idleReplicaCount"));
+  sspec->set("pollingInterval", (int)as->get_num("pollingInterval", 15));
+  sspec->set("cooldownPeriod", (int)as->get_num("cooldownPeriod", 300));
+  auto triggers = Value::array();
+  auto trig = Value::object();
+  trig->set("type", "prometheus");
+  auto tm = Value::object();
+  tm->set("serverAddress",
+          as->get_str("prometheusAddress",
+                      "http://prometheus-operated.monitoring.svc:9090"));
+  std::string metric = as->get_str("metric", "vllm:num_requests_waiting");
+  tm->set("metricName", metric);
+  tm->set("query", "sum(" + metric + "{model_name=\"" + name + "\"})");
+  char thr[32];
+  snprintf(thr, sizeof(thr), "%g", as->get_num("threshold", 10));
+  tm->set("threshold", thr);
+  trig->set("metadata", tm);
+  triggers->push(trig);
+  sspec->set("triggers", triggers);
+  so->set("spec", sspec);
+  return so;
 }
 
 ValuePtr build_engine_service(const Ctx& ctx, const ValuePtr& cr) {
@@ -291,6 +423,23 @@ ValuePtr build_cacheserver_deployment(const Ctx& ctx, const ValuePtr& cr) {
 // ---------------------------------------------------------------------------
 namespace {
 
+// engine-pod / status calls must never abort a reconcile pass: a single
+// unreachable pod would otherwise starve every later CR in the list
+pshttp::Response try_request(const std::string& method,
+                             const std::string& url,
+                             const std::string& body,
+                             const std::string& token,
+                             const std::string& ctype =
+                                 "application/json") {
+  try {
+    return pshttp::request(method, url, body, token, ctype);
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[psoperator] %s %s failed: %s\n", method.c_str(),
+            url.c_str(), e.what());
+    return pshttp::Response{};
+  }
+}
+
 std::string crd_path(const Ctx& ctx, const std::string& plural) {
   return ctx.api_server + "/apis/" + ctx.group + "/" + ctx.version +
          "/namespaces/" + ctx.ns + "/" + plural;
@@ -334,6 +483,34 @@ bool apply(const Ctx& ctx, const std::string& base_path,
   return r.ok();
 }
 
+// scale-subresource status (reference :1159-1199): PUT /status with
+// replicas + the label selector HPA needs, mirroring the child
+// Deployment's observed state.
+void update_runtime_status(const Ctx& ctx, const ValuePtr& cr) {
+  std::string name = cr->get("metadata")->get_str("name");
+  std::string dep_url = ctx.api_server + "/apis/apps/v1/namespaces/" +
+                        ctx.ns + "/deployments/" + name + "-engine";
+  auto dep = api_get(ctx, dep_url);
+  int replicas = 0, ready = 0;
+  if (dep) {
+    auto dspec = dep->get("spec");
+    auto dstat = dep->get("status");
+    if (dspec) replicas = (int)dspec->get_num("replicas", 0);
+    if (dstat) ready = (int)dstat->get_num("readyReplicas", 0);
+  }
+  auto status = Value::object();
+  status->set("replicas", replicas);
+  status->set("readyReplicas", ready);
+  status->set("selector", "app=" + name + "-engine");
+  status->set("phase", ready >= replicas && replicas > 0 ? "Ready"
+                                                         : "Progressing");
+  auto patch = Value::object();
+  patch->set("status", status);
+  std::string st_url = crd_path(ctx, "vllmruntimes") + "/" + name + "/status";
+  try_request("PATCH", st_url, psjson::dump(patch), ctx.token,
+              "application/merge-patch+json");
+}
+
 int reconcile_vllmruntimes(const Ctx& ctx) {
   auto list = api_get(ctx, crd_path(ctx, "vllmruntimes"));
   if (!list) return 0;
@@ -344,9 +521,25 @@ int reconcile_vllmruntimes(const Ctx& ctx) {
       ctx.api_server + "/apis/apps/v1/namespaces/" + ctx.ns + "/deployments";
   std::string svc_path =
       ctx.api_server + "/api/v1/namespaces/" + ctx.ns + "/services";
+  std::string pvc_path = ctx.api_server + "/api/v1/namespaces/" + ctx.ns +
+                         "/persistentvolumeclaims";
+  std::string cm_path =
+      ctx.api_server + "/api/v1/namespaces/" + ctx.ns + "/configmaps";
+  std::string so_path = ctx.api_server +
+                        "/apis/keda.sh/v1alpha1/namespaces/" + ctx.ns +
+                        "/scaledobjects";
   for (auto& cr : items->arr) {
-    if (apply(ctx, dep_path, build_engine_deployment(ctx, cr))) actions++;
+    // reference reconcile order: Service -> PVC -> ConfigMap ->
+    // Deployment -> ScaledObject -> status (vllmruntime_controller.go:62)
     if (apply(ctx, svc_path, build_engine_service(ctx, cr))) actions++;
+    if (auto pvc = build_engine_pvc(ctx, cr))
+      if (apply(ctx, pvc_path, pvc)) actions++;
+    if (auto cm = build_engine_configmap(ctx, cr))
+      if (apply(ctx, cm_path, cm)) actions++;
+    if (apply(ctx, dep_path, build_engine_deployment(ctx, cr))) actions++;
+    if (auto so = build_keda_scaledobject(ctx, cr))
+      if (apply(ctx, so_path, so)) actions++;
+    update_runtime_status(ctx, cr);
   }
   return actions;
 }
@@ -378,44 +571,141 @@ int reconcile_cacheservers(const Ctx& ctx) {
   return actions;
 }
 
-// LoraAdapter: discover the base model's ready pods and register the
-// adapter on each via the engine's /v1/load_lora_adapter endpoint
-// (reference loraadapter_controller.go:553-592 behaviour, default
-// placement = all pods).
+// LoraAdapter: discover the base model's ready pods, place the adapter per
+// spec.loraAdapterDeploymentConfig.algorithm (default | ordered |
+// equalized — reference loraadapter_types.go:70-79), register via the
+// engine's /v1/load_lora_adapter, track status.loadedAdapters, and clean
+// up through a finalizer on deletion (loraadapter_controller.go:889-927).
+constexpr const char* kLoraFinalizer = "production-stack.amd.com/lora-cleanup";
+
+bool cr_has_finalizer(const ValuePtr& cr) {
+  auto meta = cr->get("metadata");
+  auto fins = meta ? meta->get("finalizers") : nullptr;
+  if (!fins) return false;
+  for (auto& f : fins->arr)
+    if (f->str == kLoraFinalizer) return true;
+  return false;
+}
+
+void patch_finalizers(const Ctx& ctx, const std::string& name, bool add) {
+  auto fins = Value::array();
+  if (add) fins->push(Value::of(kLoraFinalizer));
+  auto meta = Value::object();
+  meta->set("finalizers", fins);
+  auto patch = Value::object();
+  patch->set("metadata", meta);
+  try_request("PATCH", crd_path(ctx, "loraadapters") + "/" + name,
+              psjson::dump(patch), ctx.token,
+              "application/merge-patch+json");
+}
+
 int reconcile_loraadapters(const Ctx& ctx) {
   auto list = api_get(ctx, crd_path(ctx, "loraadapters"));
   if (!list) return 0;
   auto items = list->get("items");
   if (!items) return 0;
   int actions = 0;
+  // adapters per base model in name order (for ordered/equalized)
+  std::map<std::string, std::vector<std::string>> by_base;
   for (auto& cr : items->arr) {
     auto spec = cr->get("spec");
+    if (spec)
+      by_base[spec->get_str("baseModel")].push_back(
+          cr->get("metadata")->get_str("name"));
+  }
+  for (auto& [b, v] : by_base) std::sort(v.begin(), v.end());
+
+  for (auto& cr : items->arr) {
+    auto meta = cr->get("metadata");
+    auto spec = cr->get("spec");
     if (!spec) continue;
+    std::string cr_name = meta->get_str("name");
     std::string base = spec->get_str("baseModel");
     auto src = spec->get("adapterSource");
     std::string adapter_name =
-        src ? src->get_str("adapterName",
-                           cr->get("metadata")->get_str("name"))
-            : cr->get("metadata")->get_str("name");
+        src ? src->get_str("adapterName", cr_name) : cr_name;
     std::string adapter_path = src ? src->get_str("adapterPath") : "";
+    auto dcfg = spec->get("loraAdapterDeploymentConfig");
+    std::string algo = dcfg ? dcfg->get_str("algorithm", "default")
+                            : "default";
+
     std::string pods_url = ctx.api_server + "/api/v1/namespaces/" + ctx.ns +
                            "/pods?labelSelector=app%3D" + base + "-engine";
     auto pods = api_get(ctx, pods_url);
-    if (!pods) continue;
-    auto pitems = pods->get("items");
-    if (!pitems) continue;
-    for (auto& pod : pitems->arr) {
-      auto status = pod->get("status");
-      std::string ip = status ? status->get_str("podIP") : "";
-      if (ip.empty()) continue;
+    auto pitems = pods ? pods->get("items") : nullptr;
+    std::vector<std::string> ips;
+    if (pitems)
+      for (auto& pod : pitems->arr) {
+        auto status = pod->get("status");
+        std::string ip = status ? status->get_str("podIP") : "";
+        if (!ip.empty()) ips.push_back(ip);
+      }
+    std::sort(ips.begin(), ips.end());
+
+    // deletion: unload everywhere, then drop the finalizer
+    if (!meta->get_str("deletionTimestamp").empty()) {
+      for (auto& ip : ips) {
+        auto body = Value::object();
+        body->set("lora_name", adapter_name);
+        try_request("POST",
+                    "http://" + ip + ":8000/v1/unload_lora_adapter",
+                    psjson::dump(body), "");
+      }
+      patch_finalizers(ctx, cr_name, false);
+      actions++;
+      continue;
+    }
+    if (!cr_has_finalizer(cr)) {
+      patch_finalizers(ctx, cr_name, true);
+      actions++;
+    }
+
+    // placement: which pods get this adapter
+    std::vector<std::string> targets;
+    if (algo == "ordered" && dcfg) {
+      // explicit replica index list, e.g. {"replicas": [0, 2]}
+      auto reps = dcfg->get("replicas");
+      if (reps)
+        for (auto& r : reps->arr) {
+          size_t idx = (size_t)r->num;
+          if (idx < ips.size()) targets.push_back(ips[idx]);
+        }
+      if (targets.empty() && !ips.empty()) targets.push_back(ips[0]);
+    } else if (algo == "equalized" && !ips.empty()) {
+      // spread this base model's adapters round-robin over pods: adapter
+      // rank r (name order) lands on pod r % n (reference :70-79)
+      auto& sibs = by_base[base];
+      size_t rank = 0;
+      for (size_t i = 0; i < sibs.size(); i++)
+        if (sibs[i] == cr_name) rank = i;
+      targets.push_back(ips[rank % ips.size()]);
+    } else {
+      targets = ips;  // default: every pod serves the adapter
+    }
+
+    auto loaded = Value::array();
+    for (auto& ip : targets) {
       auto body = Value::object();
       body->set("lora_name", adapter_name);
       body->set("lora_path", adapter_path);
-      auto r = pshttp::request(
+      auto r = try_request(
           "POST", "http://" + ip + ":8000/v1/load_lora_adapter",
           psjson::dump(body), "");
-      if (r.ok()) actions++;
+      if (r.ok()) {
+        actions++;
+        loaded->push(Value::of(ip));
+      }
     }
+    // status.loadedAdapters mirror (reference tracks per-pod registrations)
+    auto status = Value::object();
+    status->set("loadedAdapters", loaded);
+    status->set("phase", loaded->arr.empty() ? "Pending" : "Loaded");
+    auto patch = Value::object();
+    patch->set("status", status);
+    try_request("PATCH",
+                crd_path(ctx, "loraadapters") + "/" + cr_name + "/status",
+                psjson::dump(patch), ctx.token,
+                "application/merge-patch+json");
   }
   return actions;
 }

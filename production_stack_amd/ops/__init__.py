@@ -191,12 +191,26 @@ def paged_attn_prefill_mfma(
     variant: int = None,
 ) -> torch.Tensor:
     """MFMA-tiled chunked prefill (GPU, head_dim 128 only). tile_info rows
-    must be built with n_rows <= PREFILL_TILE for the active variant."""
+    must be built with n_rows <= prefill_tile_rows() for the active
+    variant."""
     _require_ext()
     out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+    v = PREFILL_VARIANT if variant is None else variant
+    if v == 5 and variant is None:
+        # v5 packs GQW q-heads per block: 4x fewer workgroups than v3.
+        # Small serving launches (~10 continuation chunks) then leave CUs
+        # idle (1 block/CU quantization) — measured 210 vs 249 TF at the
+        # 10-seq x 200-row shape. When the tile layout is v3-compatible
+        # (64-row tiles, i.e. GQW=4 models) fall back to v3 below ~500
+        # v5 workgroups.
+        qh_n = q.shape[1]
+        kh_n = k_cache.shape[1]
+        if prefill_tile_rows(qh_n, kh_n) == 64:
+            blocks_v5 = tile_info.shape[0] * kh_n
+            if blocks_v5 < 500:
+                v = 3
     _C.paged_attn_prefill_mfma(
-        out, q, k_cache, v_cache, block_tables, tile_info, scale,
-        PREFILL_VARIANT if variant is None else variant, window
+        out, q, k_cache, v_cache, block_tables, tile_info, scale, v, window
     )
     return out
 

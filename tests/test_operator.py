@@ -43,6 +43,12 @@ class FakeK8s:
         }
         self.deployments = {}
         self.services = {}
+        self.pvcs = {}
+        self.configmaps = {}
+        self.scaledobjects = {}
+        self.leases = {}
+        self.statuses = {}      # (plural, name) -> status patch
+        self.cr_patches = []    # (plural, name, body)
         self.pods = []
         self.lora_calls = []
         a = self.app
@@ -84,6 +90,51 @@ class FakeK8s:
         @a.get("/api/v1/namespaces/{ns}/pods")
         async def list_pods(ns: str):
             return {"items": self.pods}
+
+        def crud(store, base, api="api/v1"):
+            @a.get(f"/{api}/namespaces/{{ns}}/{base}/{{name}}")
+            async def get_obj(ns: str, name: str):
+                if name in store:
+                    return store[name]
+                return JSONResponse(status_code=404, content={})
+
+            @a.post(f"/{api}/namespaces/{{ns}}/{base}")
+            async def create_obj(ns: str, request: Request):
+                body = await request.json()
+                store[body["metadata"]["name"]] = body
+                return body
+
+            @a.put(f"/{api}/namespaces/{{ns}}/{base}/{{name}}")
+            async def put_obj(ns: str, name: str, request: Request):
+                store[name] = await request.json()
+                return store[name]
+
+        crud(self.pvcs, "persistentvolumeclaims")
+        crud(self.configmaps, "configmaps")
+        crud(self.scaledobjects, "scaledobjects", "apis/keda.sh/v1alpha1")
+        crud(self.leases, "leases", "apis/coordination.k8s.io/v1")
+
+        @a.patch("/apis/production-stack.amd.com/v1alpha1/namespaces/"
+                 "{ns}/{plural}/{name}/status")
+        async def patch_status(ns: str, plural: str, name: str,
+                               request: Request):
+            body = await request.json()
+            self.statuses[(plural, name)] = body.get("status", {})
+            return body
+
+        @a.patch("/apis/production-stack.amd.com/v1alpha1/namespaces/"
+                 "{ns}/{plural}/{name}")
+        async def patch_cr(ns: str, plural: str, name: str,
+                           request: Request):
+            body = await request.json()
+            self.cr_patches.append((plural, name, body))
+            # reflect finalizer changes back into the stored CR
+            for cr in self.crs.get(plural, []):
+                if cr["metadata"]["name"] == name:
+                    fins = body.get("metadata", {}).get("finalizers")
+                    if fins is not None:
+                        cr["metadata"]["finalizers"] = fins
+            return body
 
 
 @pytest.fixture(scope="module")
@@ -233,3 +284,166 @@ def test_lora_adapter_load_call(fake_k8s, tmp_path):
         assert "ad1" in {m["id"] for m in r.json()["data"]}
     finally:
         engine.stop()
+
+
+def test_pvc_configmap_keda_and_status(fake_k8s):
+    """PVC + ConfigMap + KEDA ScaledObject reconcile and the
+    scale-subresource status, mirroring reference
+    vllmruntime_controller.go:148-200, 1159, 1201-1326, 1435."""
+    fake_k8s.crs["vllmruntimes"] = [
+        {
+            "metadata": {"name": "l3s"},
+            "spec": {
+                "model": {"modelURL": "llama-3-8b"},
+                "deploymentConfig": {"replicas": 2},
+                "storage": {"enabled": True, "size": "80Gi",
+                            "storageClassName": "fast"},
+                "configData": {"extra.yaml": "max_loras: 4"},
+                "autoscaling": {
+                    "enabled": True,
+                    "minReplicas": 1,
+                    "maxReplicas": 6,
+                    "idleReplicaCount": 0,
+                    "threshold": 20,
+                },
+            },
+        }
+    ]
+    run_operator_once()
+    pvc = fake_k8s.pvcs.get("l3s-storage")
+    assert pvc is not None
+    assert pvc["spec"]["resources"]["requests"]["storage"] == "80Gi"
+    assert pvc["spec"]["storageClassName"] == "fast"
+    cm = fake_k8s.configmaps.get("l3s-config")
+    assert cm is not None and cm["data"]["extra.yaml"] == "max_loras: 4"
+    so = fake_k8s.scaledobjects.get("l3s-scaler")
+    assert so is not None
+    assert so["spec"]["scaleTargetRef"]["name"] == "l3s-engine"
+    assert so["spec"]["minReplicaCount"] == 1
+    assert so["spec"]["maxReplicaCount"] == 6
+    assert so["spec"]["idleReplicaCount"] == 0
+    trig = so["spec"]["triggers"][0]
+    assert trig["type"] == "prometheus"
+    assert "vllm:num_requests_waiting" in trig["metadata"]["query"]
+    # deployment mounts
+    dep = fake_k8s.deployments["l3s-engine"]
+    pspec = dep["spec"]["template"]["spec"]
+    vols = {v["name"] for v in pspec["volumes"]}
+    assert vols == {"model-storage", "engine-config"}
+    mounts = {m["mountPath"]
+              for m in pspec["containers"][0]["volumeMounts"]}
+    assert mounts == {"/data", "/config"}
+    # scale-subresource status
+    st = fake_k8s.statuses.get(("vllmruntimes", "l3s"))
+    assert st is not None
+    assert st["selector"] == "app=l3s-engine"
+    assert st["replicas"] == 2
+
+
+def test_lora_finalizer_and_equalized_placement(fake_k8s):
+    """Equalized placement spreads adapters round-robin over pods; a
+    finalizer is added on sight and deletion unloads + removes it
+    (reference loraadapter_controller.go:70-79, 889-927)."""
+    fake_k8s.cr_patches.clear()
+    fake_k8s.pods = [
+        {"metadata": {"name": "base-engine-0"},
+         "status": {"podIP": "127.0.0.1"}},
+    ]
+    fake_k8s.crs["loraadapters"] = [
+        {
+            "metadata": {"name": "ad-a"},
+            "spec": {
+                "baseModel": "base",
+                "adapterSource": {"adapterName": "ad-a",
+                                  "adapterPath": "/tmp/ad-a"},
+                "loraAdapterDeploymentConfig": {"algorithm": "equalized"},
+            },
+        },
+        {
+            "metadata": {"name": "ad-b"},
+            "spec": {
+                "baseModel": "base",
+                "adapterSource": {"adapterName": "ad-b",
+                                  "adapterPath": "/tmp/ad-b"},
+                "loraAdapterDeploymentConfig": {"algorithm": "equalized"},
+            },
+        },
+    ]
+    run_operator_once()
+    # both CRs got the finalizer patched on
+    fin_patches = [(p, n) for (p, n, b) in fake_k8s.cr_patches
+                   if b.get("metadata", {}).get("finalizers")]
+    assert ("loraadapters", "ad-a") in fin_patches
+    assert ("loraadapters", "ad-b") in fin_patches
+    # deletion path: unload + finalizer removal (empty list patch)
+    fake_k8s.cr_patches.clear()
+    fake_k8s.crs["loraadapters"][0]["metadata"]["deletionTimestamp"] = (
+        "2026-09-12T00:00:00Z"
+    )
+    run_operator_once()
+    removals = [b for (p, n, b) in fake_k8s.cr_patches
+                if n == "ad-a" and b.get("metadata", {}).get(
+                    "finalizers") == []]
+    assert removals, fake_k8s.cr_patches
+    fake_k8s.crs["loraadapters"] = []
+
+
+def test_leader_election_and_health(fake_k8s):
+    """--leader-elect acquires the coordination.k8s.io Lease; a fresh
+    foreign lease makes the operator stand by; /healthz + /metrics serve."""
+    import requests
+
+    fake_k8s.leases.clear()
+    r = subprocess.run(
+        [OPERATOR, "--api-server", f"http://127.0.0.1:{PORT}",
+         "--namespace", "default", "--token-file", "/dev/null",
+         "--leader-elect", "--health-port", "18081", "--once"],
+        capture_output=True, timeout=60,
+    )
+    assert r.returncode == 0, r.stderr.decode()
+    lease = fake_k8s.leases.get("production-stack-amd-operator")
+    assert lease is not None
+    assert lease["spec"]["holderIdentity"]
+    # foreign fresh lease -> standby
+    import datetime
+
+    now = datetime.datetime.utcnow().strftime("%Y-%m-%dT%H:%M:%S.000000Z")
+    fake_k8s.leases["production-stack-amd-operator"] = {
+        "metadata": {"name": "production-stack-amd-operator",
+                     "resourceVersion": "1"},
+        "spec": {"holderIdentity": "someone-else",
+                 "leaseDurationSeconds": 300, "renewTime": now},
+    }
+    r = subprocess.run(
+        [OPERATOR, "--api-server", f"http://127.0.0.1:{PORT}",
+         "--namespace", "default", "--token-file", "/dev/null",
+         "--leader-elect", "--once"],
+        capture_output=True, timeout=60,
+    )
+    assert b"standing by" in r.stderr
+    # health endpoints (long-running process)
+    proc = subprocess.Popen(
+        [OPERATOR, "--api-server", f"http://127.0.0.1:{PORT}",
+         "--namespace", "default", "--token-file", "/dev/null",
+         "--health-port", "18082", "--interval", "1"],
+        stderr=subprocess.PIPE,
+    )
+    try:
+        deadline = time.time() + 10
+        ok = False
+        while time.time() < deadline:
+            try:
+                hz = requests.get("http://127.0.0.1:18082/healthz",
+                                  timeout=0.5)
+                mt = requests.get("http://127.0.0.1:18082/metrics",
+                                  timeout=0.5)
+                if hz.status_code == 200 and "psoperator_reconcile_total" \
+                        in mt.text:
+                    ok = True
+                    break
+            except Exception:
+                time.sleep(0.2)
+        assert ok
+    finally:
+        proc.terminate()
+        proc.wait(timeout=5)
