@@ -48,11 +48,13 @@ class AsyncEngine:
     def _drain_control(self) -> None:
         while True:
             try:
-                rid, prompt, params, arrival = self._pending.get_nowait()
+                (rid, prompt, params, arrival,
+                 mm) = self._pending.get_nowait()
             except queue.Empty:
                 break
             try:
-                self.engine.add_request(rid, prompt, params, arrival)
+                self.engine.add_request(rid, prompt, params, arrival,
+                                        mm_embeds=mm)
             except ValueError as e:
                 self._emit(
                     RequestOutput(
@@ -106,10 +108,12 @@ class AsyncEngine:
         request_id: str,
         prompt: Union[str, List[int]],
         params: SamplingParams,
+        mm_embeds=None,
     ) -> AsyncIterator[RequestOutput]:
         q: asyncio.Queue = asyncio.Queue()
         self._streams[request_id] = q
-        self._pending.put((request_id, prompt, params, time.time()))
+        self._pending.put(
+            (request_id, prompt, params, time.time(), mm_embeds))
         self._wake.set()
         try:
             while True:

@@ -98,6 +98,7 @@ class LLMEngine:
         params: SamplingParams,
         arrival_time: Optional[float] = None,
         lora_name: Optional[str] = None,
+        mm_embeds=None,
     ) -> None:
         params.validate(self.config.max_model_len)
         if lora_name is not None and lora_name not in self.lora_adapters:
@@ -108,8 +109,11 @@ class LLMEngine:
             token_ids = list(prompt)
         if not token_ids:
             token_ids = [self.model_cfg.bos_token_id]
+        if mm_embeds and self.config.parallel.tensor_parallel_size > 1:
+            raise ValueError(
+                "multimodal embeddings are not supported with TP > 1")
         seq = Sequence(request_id, token_ids, params, arrival_time,
-                       lora_name=lora_name)
+                       lora_name=lora_name, mm_embeds=mm_embeds)
         self.scheduler.add(seq)
         self.stats.num_requests += 1
         self.stats.prompt_tokens += len(token_ids)
@@ -326,6 +330,29 @@ class LLMEngine:
     def stop_pp_workers(self) -> None:
         if self.runner.pp_size > 1 and self.runner.pp_rank == 0:
             self.runner.pipeline.stop_workers()
+
+    # ---- multimodal encoders (vision/audio adapters; lazy) -------------
+    def get_vision_encoder(self):
+        if getattr(self, "_vision_encoder", None) is None:
+            from production_stack_amd.engine.models.multimodal import (
+                VisionEncoder,
+            )
+
+            self._vision_encoder = VisionEncoder(
+                self.model_cfg.hidden_size, seed=self.config.seed + 7
+            ).to(self.device)
+        return self._vision_encoder
+
+    def get_audio_encoder(self):
+        if getattr(self, "_audio_encoder", None) is None:
+            from production_stack_amd.engine.models.multimodal import (
+                AudioEncoder,
+            )
+
+            self._audio_encoder = AudioEncoder(
+                self.model_cfg.hidden_size, seed=self.config.seed + 9
+            ).to(self.device)
+        return self._audio_encoder
 
     # ---- sleep / wake (reference request.py:1041-1128 parity) ----------
     def sleep(self, level: int = 1) -> None:

@@ -205,6 +205,8 @@ class ModelRunner:
         tokens: List[int] = []
         positions: List[int] = []
         slots: List[int] = []
+        mm_rows: List[int] = []
+        mm_parts: List = []
         p_token_seq: List[int] = []
         p_token_pos: List[int] = []
         p_tables: List[List[int]] = []
@@ -248,6 +250,14 @@ class ModelRunner:
             slots.extend(slot_arr.tolist())
             p_token_seq.extend([row] * ss.num_tokens)
             p_token_pos.extend(pos_arr.tolist())
+            # multimodal: slice each media embed span into this chunk
+            for off, emb in getattr(seq, "mm_embeds", []):
+                lo = max(off, start)
+                hi = min(off + emb.shape[0], end)
+                if lo < hi:
+                    mm_rows.extend(range(q_flat0 + (lo - start),
+                                         q_flat0 + (hi - start)))
+                    mm_parts.append(emb[lo - off:hi - off])
 
         d_seq_lens: List[int] = []
         d_tables: List[List[int]] = []
@@ -354,6 +364,18 @@ class ModelRunner:
             ),
             lora_groups=lora_groups,
             lora_idx=lora_idx_t,
+            mm_rows=(
+                torch.tensor(mm_rows, dtype=torch.long).to(
+                    dev, non_blocking=True
+                )
+                if mm_rows
+                else None
+            ),
+            mm_embeds=(
+                torch.cat(mm_parts).to(dev, non_blocking=True)
+                if mm_parts
+                else None
+            ),
         )
         token_t = torch.tensor(tokens, dtype=torch.long).to(
             dev, non_blocking=True

@@ -46,6 +46,8 @@ class BatchMeta:
         prefill_tiles: Optional[torch.Tensor] = None,  # [NT, 4] int32
         lora_groups=None,  # [(LoRAAdapter, row_idx_tensor)]
         lora_idx: Optional[torch.Tensor] = None,  # [T] int32 slot per row
+        mm_rows: Optional[torch.Tensor] = None,   # [M] long batch rows
+        mm_embeds: Optional[torch.Tensor] = None,  # [M, H] bf16
     ) -> None:
         self.positions = positions
         self.slot_mapping = slot_mapping
@@ -59,6 +61,8 @@ class BatchMeta:
         self.prefill_tiles = prefill_tiles
         self.lora_groups = lora_groups or []
         self.lora_idx = lora_idx
+        self.mm_rows = mm_rows
+        self.mm_embeds = mm_embeds
 
 
 def build_cos_sin_cache(
@@ -393,6 +397,12 @@ class LlamaForCausalLM(nn.Module):
     ) -> torch.Tensor:
         if self.is_first:
             hidden = F.embedding(token_ids, self.embed)
+            if meta.mm_rows is not None:
+                # multimodal injection: encoder embeddings overwrite the
+                # placeholder positions (vision/audio adapter pattern)
+                hidden = hidden.index_copy(
+                    0, meta.mm_rows, meta.mm_embeds.to(hidden.dtype)
+                )
             residual = None
         else:
             hidden = token_ids  # mid-pipeline: activations from prev stage

@@ -40,9 +40,11 @@ class Sequence:
         params: SamplingParams,
         arrival_time: Optional[float] = None,
         lora_name: Optional[str] = None,
+        mm_embeds=None,  # [(prompt_offset, embeds [n, hidden])]
     ) -> None:
         self.request_id = request_id
         self.lora_name = lora_name
+        self.mm_embeds = mm_embeds or []
         self.prompt_token_ids = list(prompt_token_ids)
         self.output_token_ids: List[int] = []
         self.params = params

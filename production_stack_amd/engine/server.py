@@ -411,8 +411,13 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
             request.headers.get("x-request-id")
             or f"cmpl-{uuid.uuid4().hex[:24]}"
         )
+        mm_embeds = None
         if chat:
-            prompt = render_chat(body.get("messages") or [])
+            messages = body.get("messages") or []
+            if _has_media(messages):
+                prompt, mm_embeds = _assemble_multimodal(messages, engine)
+            else:
+                prompt = render_chat(messages)
         else:
             p = body.get("prompt", "")
             if isinstance(p, list) and p and isinstance(p[0], int):
@@ -469,7 +474,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                     first = True
                     n_out = 0
                     async for out in async_engine.generate(
-                        rid, prompt_tokens, params
+                        rid, prompt_tokens, params, mm_embeds=mm_embeds
                     ):
                         n_out = out.num_output_tokens
                         if chat:
@@ -533,7 +538,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
             text, toks, lps = "", [], []
             reason, npr = None, len(prompt_tokens)
             async for out in async_engine.generate(
-                r_i, prompt_tokens, p_i
+                r_i, prompt_tokens, p_i, mm_embeds=mm_embeds
             ):
                 text += out.text_delta
                 toks.extend(out.new_token_ids)
@@ -597,6 +602,142 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                     app, kvp, rid
                 )
         return JSONResponse(resp)
+
+    # ---- multimodal: vision content parts + audio endpoints -----------
+    def _has_media(messages) -> bool:
+        for m in messages:
+            c = m.get("content")
+            if isinstance(c, list):
+                for part in c:
+                    if isinstance(part, dict) and part.get("type") in (
+                            "image_url", "input_audio"):
+                        return True
+        return False
+
+    def _assemble_multimodal(messages, eng):
+        """Token stream with per-media placeholder ids + (offset, embeds)
+        pairs (vision/audio adapter injection; engine/models/multimodal)."""
+        import base64 as b64
+
+        from production_stack_amd.engine.models.multimodal import (
+            decode_image,
+            decode_wav,
+            media_placeholder_tokens,
+        )
+
+        toks: list = []
+        mm: list = []
+        vocab = eng.model_cfg.vocab_size
+
+        def add_text(t: str) -> None:
+            toks.extend(eng.tokenizer.encode(t))
+
+        for m in messages:
+            add_text(f"<|{m.get('role', 'user')}|> ")
+            c = m.get("content")
+            parts = c if isinstance(c, list) else [
+                {"type": "text", "text": c or ""}]
+            for part in parts:
+                if not isinstance(part, dict):
+                    continue
+                kind = part.get("type")
+                if kind == "text":
+                    add_text(part.get("text", ""))
+                elif kind == "image_url":
+                    url = (part.get("image_url") or {}).get("url", "")
+                    if not url.startswith("data:"):
+                        raise ValueError(
+                            "only data: image URLs are supported offline")
+                    data = b64.b64decode(url.split(",", 1)[1])
+                    emb = eng.get_vision_encoder()(
+                        decode_image(data).to(eng.device))
+                    mm.append((len(toks), emb))
+                    toks.extend(media_placeholder_tokens(
+                        data, emb.shape[0], vocab))
+                elif kind == "input_audio":
+                    data = b64.b64decode(
+                        (part.get("input_audio") or {}).get("data", ""))
+                    emb = eng.get_audio_encoder()(
+                        decode_wav(data).to(eng.device))
+                    mm.append((len(toks), emb))
+                    toks.extend(media_placeholder_tokens(
+                        data, emb.shape[0], vocab))
+        add_text("<|assistant|>")
+        return toks, mm
+
+    def _parse_multipart(body: bytes, content_type: str):
+        """Minimal multipart/form-data parser (python-multipart is not in
+        the image). Returns {field: bytes}."""
+        import re as _re
+
+        m = _re.search(r'boundary="?([^";,]+)"?', content_type)
+        if not m:
+            return {}
+        sep = b"--" + m.group(1).encode()
+        fields = {}
+        for chunk in body.split(sep):
+            if b"\r\n\r\n" not in chunk:
+                continue
+            head, _, payload = chunk.partition(b"\r\n\r\n")
+            nm = _re.search(rb'name="([^"]+)"', head)
+            if nm:
+                fields[nm.group(1).decode()] = payload.rstrip(b"\r\n-")
+        return fields
+
+    async def _transcribe_impl(request: Request, translate: bool):
+        ctype = request.headers.get("content-type", "")
+        raw = await request.body()
+        language = None
+        fmt = "json"
+        if ctype.startswith("multipart/form-data"):
+            fields = _parse_multipart(raw, ctype)
+            wav = fields.get("file", b"")
+            language = (fields.get("language") or b"").decode() or None
+            fmt = (fields.get("response_format") or b"json").decode()
+        elif ctype.startswith("application/json"):
+            body = json.loads(raw or b"{}")
+            import base64 as b64
+
+            wav = b64.b64decode(body.get("file", ""))
+            language = body.get("language")
+            fmt = body.get("response_format", "json")
+        else:  # raw audio body
+            wav = raw
+        if not wav:
+            return JSONResponse(status_code=400, content={
+                "error": {"message": "no audio payload",
+                          "type": "invalid_request_error"}})
+        from production_stack_amd.engine.models.multimodal import (
+            decode_wav,
+            media_placeholder_tokens,
+        )
+
+        emb = engine.get_audio_encoder()(
+            decode_wav(wav).to(engine.device))
+        task = "translate" if translate else "transcribe"
+        head = engine.tokenizer.encode(f"<|audio|> {task} ")
+        toks = list(head) + media_placeholder_tokens(
+            wav, emb.shape[0], engine.model_cfg.vocab_size)
+        mm = [(len(head), emb)]
+        rid = f"transcribe-{uuid.uuid4().hex[:16]}"
+        text = ""
+        params = SamplingParams(max_tokens=48, temperature=0.0,
+                                ignore_eos=False)
+        async for out in async_engine.generate(rid, toks, params,
+                                               mm_embeds=mm):
+            text += out.text_delta
+        if fmt == "text":
+            return PlainTextResponse(text.strip())
+        return {"text": text.strip(),
+                **({"language": language} if language else {})}
+
+    @app.post("/v1/audio/transcriptions")
+    async def audio_transcriptions(request: Request):
+        return await _transcribe_impl(request, translate=False)
+
+    @app.post("/v1/audio/translations")
+    async def audio_translations(request: Request):
+        return await _transcribe_impl(request, translate=True)
 
     @app.get("/version")
     async def version():

@@ -266,3 +266,47 @@ def test_fp8_weight_quantization_gpu():
     want = cpu.generate([prompt], p)["offline-0"]
     agree = sum(a == b for a, b in zip(out[:4], want[:4]))
     assert agree >= 2, (out, want)
+
+
+def test_multimodal_vision_injection_gpu():
+    """Vision embeddings injected over placeholder tokens on the GPU path
+    (eager prefill + decode graphs): deterministic per image, different
+    across images."""
+    import io
+
+    import numpy as np
+    from PIL import Image
+
+    from production_stack_amd.engine.models.multimodal import (
+        decode_image,
+        media_placeholder_tokens,
+    )
+
+    eng = make_engine()
+    enc = eng.get_vision_encoder()
+
+    def png(color):
+        img = Image.new("RGB", (64, 64), color)
+        buf = io.BytesIO()
+        img.save(buf, format="PNG")
+        return buf.getvalue()
+
+    def run(data, tag):
+        emb = enc(decode_image(data).to(eng.device))
+        toks = [5, 6, 7] + media_placeholder_tokens(
+            data, emb.shape[0], eng.model_cfg.vocab_size) + [9, 10]
+        eng.add_request(tag, toks, SamplingParams(
+            max_tokens=8, temperature=0.0, ignore_eos=True),
+            mm_embeds=[(3, emb)])
+        out = []
+        while eng.has_unfinished():
+            for o in eng.step():
+                if o.request_id == tag:
+                    out.extend(o.new_token_ids)
+        return out
+
+    a1 = run(png((255, 0, 0)), "a1")
+    a2 = run(png((255, 0, 0)), "a2")
+    b = run(png((0, 0, 255)), "b1")
+    assert a1 == a2
+    assert a1 != b
