@@ -66,6 +66,9 @@ int ps_lora_bgmv(void* out, const void* x, const void* A, const void* B,
                  hipStream_t stream);
 }
 
+at::Tensor cachegen_encode(at::Tensor q);
+at::Tensor cachegen_decode(at::Tensor blob, int64_t hd);
+
 namespace {
 
 hipStream_t current_stream() {
@@ -421,6 +424,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");
   m.def("kv_quant", &kv_quant, "Row-wise int8 KV quantization");
+  m.def("cachegen_encode", &cachegen_encode,
+        "CacheGen-style adaptive range encode of int8 KV (CPU)");
+  m.def("cachegen_decode", &cachegen_decode,
+        "CacheGen-style range decode -> int8 (CPU)");
   m.def("gemm8p", &gemm8p,
         "8-phase deep-pipelined bf16 GEMM (out = x @ w.T)");
   m.def("skinny_gemm", &skinny_gemm,

@@ -274,6 +274,7 @@ class CacheConfig:
     # remote cacheserver data plane ("host:port"), shared across instances
     # (reference cacheserverSpec / LMCACHE_REMOTE_URL)
     remote_kv_url: Optional[str] = None
+    remote_kv_serde: str = "cachegen"  # raw | cachegen | cachegen4
     # "bf16" (raw) or "int8" (CacheGen-style row-quantized serde, halves
     # host-pool bytes; reference LMCACHE remote serde surface)
     offload_dtype: str = "bf16"

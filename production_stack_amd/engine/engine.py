@@ -74,6 +74,7 @@ class LLMEngine:
                 self.device,
                 offload_dtype=config.cache.offload_dtype,
                 remote_url=config.cache.remote_kv_url,
+                remote_serde=config.cache.remote_kv_serde,
             )
             self.block_manager.offload_pool = self.host_pool
         else:

@@ -35,7 +35,9 @@ class HostKVPool:
         device: torch.device,
         offload_dtype: str = "bf16",
         remote_url: Optional[str] = None,
+        remote_serde: str = "cachegen",  # raw | cachegen | cachegen4
     ) -> None:
+        self.remote_serde = remote_serde
         self.kv_caches = kv_caches
         self.device = device
         self.layers = len(kv_caches)
@@ -232,13 +234,26 @@ class HostKVPool:
 
     # ---- remote tier ---------------------------------------------------
     def _record_bytes(self, slot: int):
-        data = self.store[slot].contiguous().view(torch.uint8)
         scales = None
         if self.quantized:
             scales = (
                 self.scale_store[slot].contiguous().view(torch.uint8)
                 .numpy().tobytes()
             )
+            if self.remote_serde in ("cachegen", "cachegen4"):
+                # CacheGen-style serde: entropy-encode the int8 record
+                # (optionally re-binned to 4-bit levels) before it goes
+                # over the wire; _fetch_remote auto-detects the PSKV blob
+                from production_stack_amd import ops
+
+                q = self.store[slot].contiguous().view(torch.int8)
+                if self.remote_serde == "cachegen4":
+                    q = (
+                        (q.float() / 16.0).round().clamp(-8, 7) * 16
+                    ).to(torch.int8)
+                blob = ops.cachegen_encode(q.view(-1, self.hd))
+                return blob.numpy().tobytes(), scales
+        data = self.store[slot].contiguous().view(torch.uint8)
         return data.numpy().tobytes(), scales
 
     def _push_loop(self) -> None:
@@ -267,7 +282,13 @@ class HostKVPool:
         slot = self._take_slot()
         flat = torch.frombuffer(bytearray(data), dtype=torch.uint8)
         dst = self.store[slot].view(torch.uint8)
-        dst.copy_(flat.view(dst.shape))
+        if flat.numel() >= 4 and bytes(flat[:4].tolist()) == b"VKSP":
+            from production_stack_amd import ops
+
+            dec = ops.cachegen_decode(flat, self.hd)
+            dst.copy_(dec.view(torch.uint8).view(dst.shape))
+        else:
+            dst.copy_(flat.view(dst.shape))
         if self.quantized and scales is not None:
             sflat = torch.frombuffer(bytearray(scales), dtype=torch.uint8)
             self.scale_store[slot].view(torch.uint8).copy_(sflat)

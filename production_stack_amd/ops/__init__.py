@@ -264,6 +264,21 @@ def skinny_gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return out
 
 
+def cachegen_encode(q: torch.Tensor) -> torch.Tensor:
+    """Entropy-encode int8 KV (CPU adaptive range coder, per-channel
+    contexts). Falls back to a pure-Python reference when the extension
+    is unavailable (CPU CI)."""
+    if _C is not None and hasattr(_C, "cachegen_encode"):
+        return _C.cachegen_encode(q.contiguous().cpu())
+    return reference.cachegen_encode(q)
+
+
+def cachegen_decode(blob: torch.Tensor, hd: int) -> torch.Tensor:
+    if _C is not None and hasattr(_C, "cachegen_decode"):
+        return _C.cachegen_decode(blob.contiguous().cpu(), hd)
+    return reference.cachegen_decode(blob, hd)
+
+
 def gemm8p(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """out = x @ w.T via the 8-phase deep-pipelined MFMA GEMM
     (csrc/gemm8p.hip; 256x256 tiles + split-K for skinny M). GPU only."""

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         "csrc/prefill_mfma32.hip",
         "csrc/skinny_gemm.hip",
         "csrc/gemm8p.hip",
+        "csrc/cachegen.cpp",
         "csrc/lora_bgmv.hip",
     ],
     extra_compile_args={

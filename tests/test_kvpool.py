@@ -432,3 +432,84 @@ def test_remote_kv_int8_records_roundtrip():
         b.host_pool.stop()
     finally:
         loop.call_soon_threadsafe(loop.stop)
+
+
+def test_cachegen_codec_roundtrip_and_ratio():
+    """CacheGen-style serde (csrc/cachegen.cpp): adaptive range coder with
+    per-channel contexts + token-axis deltas. Exact roundtrip always;
+    compression where the data has structure (sparse/slowly-varying KV and
+    the 4-bit re-binned tier); near-1x on entropy-saturated rowwise-int8
+    (honest bound: 8-bit full-range Gaussian has ~6.6 bits/symbol)."""
+    import torch
+
+    from production_stack_amd import ops
+
+    torch.manual_seed(5)
+    # structured (slowly varying across tokens)
+    base = (torch.randn(1, 128) * 40).clamp(-127, 127)
+    walk = base + torch.randn(256, 128).cumsum(0) * 1.5
+    q = walk.clamp(-127, 127).to(torch.int8)
+    blob = ops.cachegen_encode(q)
+    back = ops.cachegen_decode(blob, 128).view(q.shape)
+    assert torch.equal(q, back)
+    assert q.numel() / blob.numel() > 1.3, q.numel() / blob.numel()
+
+    # 4-bit tier (cachegen4): re-binned levels compress well
+    q4 = ((q.float() / 16).round().clamp(-8, 7) * 16).to(torch.int8)
+    blob4 = ops.cachegen_encode(q4)
+    assert torch.equal(ops.cachegen_decode(blob4, 128).view(q4.shape), q4)
+    assert q4.numel() / blob4.numel() > 1.8, q4.numel() / blob4.numel()
+
+    # zeros (padding / unwritten tails)
+    z = torch.zeros(4096, dtype=torch.int8).view(32, 128)
+    bz = ops.cachegen_encode(z)
+    assert z.numel() / bz.numel() > 15
+
+
+def test_remote_tier_cachegen_serde_roundtrip():
+    """HostKVPool remote tier with serde=cachegen: pushed blobs carry the
+    PSKV magic and fetch restores bit-exact int8 records."""
+    import torch
+
+    from production_stack_amd.kvpool.offload import HostKVPool
+
+    torch.manual_seed(9)
+    layers = 2
+    kv = [
+        (torch.randn(8, 2, 16, 64, dtype=torch.bfloat16),
+         torch.randn(8, 2, 16, 64, dtype=torch.bfloat16))
+        for _ in range(layers)
+    ]
+    pool = HostKVPool(kv, 16, 0.01, torch.device("cpu"),
+                      offload_dtype="int8", remote_serde="cachegen")
+
+    class FakeRemote:
+        def __init__(self):
+            self.store = {}
+
+        def put(self, h, data, scales):
+            assert data[:4] == b"VKSP", data[:8]
+            self.store[h] = (data, scales)
+            return True
+
+        def get(self, h):
+            return self.store.get(h)
+
+        def exists(self, h):
+            return h in self.store
+
+    pool.offload(h=111, block_id=3)  # local offload (no remote yet)
+    pool.remote = FakeRemote()
+    if pool.stream is None:
+        pass  # CPU: synchronous
+    slot = pool.slot_of[111]
+    data, scales = pool._record_bytes(slot)
+    pool.remote.put(111, data, scales)
+    want = pool.store[slot].clone()
+    # wipe local and fetch back through the serde
+    del pool.slot_of[111]
+    pool.free_slots.append(slot)
+    got_slot = pool._fetch_remote(111)
+    assert got_slot is not None
+    assert torch.equal(pool.store[got_slot].view(torch.int8),
+                       want.view(torch.int8))
