@@ -15,6 +15,11 @@ void ps_rms_norm(void* out, const void* x, const void* w, float eps, long T,
                  int D, hipStream_t stream);
 void ps_fused_add_rms_norm(void* x, void* residual, const void* w, float eps,
                            long T, int D, hipStream_t stream);
+void ps_rms_norm_fp8(void* out_q, void* out_scale, void* x, void* residual,
+                     const void* w, float eps, long T, int D, int fused,
+                     hipStream_t stream);
+void ps_silu_and_mul_fp8(void* out_q, void* out_scale, const void* x, long T,
+                         int D, hipStream_t stream);
 void ps_silu_and_mul(void* out, const void* x, long T, int D,
                      hipStream_t stream);
 void ps_rope(const void* positions, void* q, void* k, const void* cos_sin,
@@ -312,6 +317,31 @@ void skinny_gemm(at::Tensor out, at::Tensor x, at::Tensor w) {
   TORCH_CHECK(rc == 0, "skinny gemm launch failed");
 }
 
+void rms_norm_fp8(at::Tensor out_q, at::Tensor scales, at::Tensor x,
+                  at::Tensor residual, at::Tensor w, double eps,
+                  int64_t fused) {
+  CHECK_GPU_BF16(x);
+  CHECK_GPU_BF16(w);
+  CHECK_GPU_DTYPE(scales, at::kFloat);
+  TORCH_CHECK(out_q.scalar_type() == at::kFloat8_e4m3fn ||
+                  out_q.scalar_type() == at::kByte,
+              "out_q must be fp8/byte");
+  const long T = x.size(0);
+  const int D = (int)x.size(1);
+  ps_rms_norm_fp8(out_q.data_ptr(), scales.data_ptr(), x.data_ptr(),
+                  fused ? residual.data_ptr() : nullptr, w.data_ptr(),
+                  (float)eps, T, D, (int)fused, current_stream());
+}
+
+void silu_and_mul_fp8(at::Tensor out_q, at::Tensor scales, at::Tensor x) {
+  CHECK_GPU_BF16(x);
+  CHECK_GPU_DTYPE(scales, at::kFloat);
+  const long T = x.size(0);
+  const int D = (int)x.size(1) / 2;
+  ps_silu_and_mul_fp8(out_q.data_ptr(), scales.data_ptr(), x.data_ptr(), T,
+                      D, current_stream());
+}
+
 void gemm8p(at::Tensor out, at::Tensor x, at::Tensor w) {
   CHECK_GPU_BF16(out);
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16,
@@ -424,6 +454,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Append K/V for new tokens into the paged cache");
   m.def("greedy_sample", &greedy_sample, "Per-row argmax over vocab");
   m.def("kv_quant", &kv_quant, "Row-wise int8 KV quantization");
+  m.def("rms_norm_fp8", &rms_norm_fp8,
+        "RMSNorm (optionally fused residual add) emitting fp8 + row scales");
+  m.def("silu_and_mul_fp8", &silu_and_mul_fp8,
+        "SiLU-mul emitting fp8 + row scales");
   m.def("cachegen_encode", &cachegen_encode,
         "CacheGen-style adaptive range encode of int8 KV (CPU)");
   m.def("cachegen_decode", &cachegen_decode,

@@ -380,3 +380,179 @@ void ps_kv_dequant(void* out, const void* in, const void* scales, long rows,
       rows);
 }
 }
+
+// ---------------------------------------------------------------------------
+// fp8 activation-quantizing variants (fuse the dynamic
+// activation quantization into the producing kernel instead of a separate
+// amax + cast pass). Each emits OCP fp8-e4m3 rows plus a per-ROW scale
+// (amax/448) for rowwise-scaled fp8 GEMMs. Three phases per row: reduce
+// sum-of-squares; compute normed values in registers + reduce row amax;
+// quantize + store fp8.
+// ---------------------------------------------------------------------------
+template <int BLOCK, int MAX_VPT, bool FUSED_ADD>
+__global__ __launch_bounds__(BLOCK) void rms_norm_fp8_kernel(
+    unsigned char* __restrict__ out_q,      // [T, D] fp8
+    float* __restrict__ out_scale,          // [T]
+    unsigned short* __restrict__ x,         // [T, D] (delta when fused)
+    unsigned short* __restrict__ residual,  // [T, D] in/out (fused only)
+    const unsigned short* __restrict__ w,   // [D]
+    float eps, int D) {
+  const long t = blockIdx.x;
+  unsigned short* xr = x + t * (long)D;
+  unsigned short* rr = FUSED_ADD ? residual + t * (long)D : nullptr;
+  unsigned char* orow = out_q + t * (long)D;
+
+  float vals[MAX_VPT][8];
+  int nvec = 0;
+  float ss = 0.f;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 xv = *(const ps_bf16x8*)(xr + i);
+    ps_bf16x8 sv;
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = ps_bf16_to_f32(xv[j]);
+      if (FUSED_ADD) {
+        f += ps_bf16_to_f32(((const ps_bf16x8*)(rr + i))[0][j]);
+        sv[j] = ps_f32_to_bf16(f);
+      }
+      vals[nvec][j] = f;
+      ss += f * f;
+    }
+    if (FUSED_ADD) *(ps_bf16x8*)(rr + i) = sv;
+    nvec++;
+  }
+  __shared__ float red[BLOCK / 64];
+  ss = ps_group_sum<64>(ss);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tot = 0.f;
+#pragma unroll
+    for (int wv = 0; wv < BLOCK / 64; wv++) tot += red[wv];
+    red[0] = rsqrtf(tot / (float)D + eps);
+  }
+  __syncthreads();
+  const float inv = red[0];
+
+  // normed values + row amax
+  float amax = 1e-6f;
+  nvec = 0;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 wv = *(const ps_bf16x8*)(w + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float f = vals[nvec][j] * inv * ps_bf16_to_f32(wv[j]);
+      vals[nvec][j] = f;
+      amax = fmaxf(amax, fabsf(f));
+    }
+    nvec++;
+  }
+  __shared__ float reda[BLOCK / 64];
+  amax = ps_group_max<64>(amax);
+  if ((threadIdx.x & 63) == 0) reda[threadIdx.x >> 6] = amax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = 0.f;
+#pragma unroll
+    for (int wv = 0; wv < BLOCK / 64; wv++) m = fmaxf(m, reda[wv]);
+    reda[0] = m / 448.0f;
+    out_scale[t] = m / 448.0f;
+  }
+  __syncthreads();
+  const float rs = 1.0f / reda[0];
+
+  nvec = 0;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_fp8x8 qv;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      qv[j] = ps_f32_to_fp8(vals[nvec][j] * rs);
+    *(ps_fp8x8*)(orow + i) = qv;
+    nvec++;
+  }
+}
+
+template <int BLOCK, int MAX_VPT>
+__global__ __launch_bounds__(BLOCK) void silu_and_mul_fp8_kernel(
+    unsigned char* __restrict__ out_q,  // [T, D] fp8
+    float* __restrict__ out_scale,      // [T]
+    const unsigned short* __restrict__ x,  // [T, 2D]
+    int D) {
+  const long t = blockIdx.x;
+  const unsigned short* g = x + t * (long)(2 * D);
+  const unsigned short* u = g + D;
+  unsigned char* orow = out_q + t * (long)D;
+
+  float vals[MAX_VPT][8];
+  int nvec = 0;
+  float amax = 1e-6f;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_bf16x8 gv = *(const ps_bf16x8*)(g + i);
+    ps_bf16x8 uv = *(const ps_bf16x8*)(u + i);
+#pragma unroll
+    for (int j = 0; j < 8; j++) {
+      float a = ps_bf16_to_f32(gv[j]);
+      float f = a / (1.0f + __expf(-a)) * ps_bf16_to_f32(uv[j]);
+      vals[nvec][j] = f;
+      amax = fmaxf(amax, fabsf(f));
+    }
+    nvec++;
+  }
+  __shared__ float reda[BLOCK / 64];
+  amax = ps_group_max<64>(amax);
+  if ((threadIdx.x & 63) == 0) reda[threadIdx.x >> 6] = amax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float m = 0.f;
+#pragma unroll
+    for (int wv = 0; wv < BLOCK / 64; wv++) m = fmaxf(m, reda[wv]);
+    reda[0] = m / 448.0f;
+    out_scale[t] = m / 448.0f;
+  }
+  __syncthreads();
+  const float rs = 1.0f / reda[0];
+  nvec = 0;
+  for (int i = threadIdx.x * 8; i < D; i += BLOCK * 8) {
+    ps_fp8x8 qv;
+#pragma unroll
+    for (int j = 0; j < 8; j++)
+      qv[j] = ps_f32_to_fp8(vals[nvec][j] * rs);
+    *(ps_fp8x8*)(orow + i) = qv;
+    nvec++;
+  }
+}
+
+extern "C" {
+
+void ps_rms_norm_fp8(void* out_q, void* out_scale, void* x, void* residual,
+                     const void* w, float eps, long T, int D, int fused,
+                     hipStream_t stream) {
+  dim3 grid((unsigned)T);
+#define PS_RNF(B, V, F)                                                   \
+  rms_norm_fp8_kernel<B, V, F><<<grid, B, 0, stream>>>(                    \
+      (unsigned char*)out_q, (float*)out_scale, (unsigned short*)x,       \
+      (unsigned short*)residual, (const unsigned short*)w, eps, D)
+  if (fused) {
+    if (D <= 4096) PS_RNF(512, 1, true);
+    else PS_RNF(1024, 2, true);
+  } else {
+    if (D <= 4096) PS_RNF(512, 1, false);
+    else PS_RNF(1024, 2, false);
+  }
+#undef PS_RNF
+}
+
+void ps_silu_and_mul_fp8(void* out_q, void* out_scale, const void* x, long T,
+                         int D, hipStream_t stream) {
+  dim3 grid((unsigned)T);
+  if (D <= 4096)
+    silu_and_mul_fp8_kernel<512, 1><<<grid, 512, 0, stream>>>(
+        (unsigned char*)out_q, (float*)out_scale,
+        (const unsigned short*)x, D);
+  else
+    silu_and_mul_fp8_kernel<1024, 4><<<grid, 1024, 0, stream>>>(
+        (unsigned char*)out_q, (float*)out_scale,
+        (const unsigned short*)x, D);
+}
+
+}  // extern "C"

@@ -138,17 +138,36 @@ class LlamaLayer(nn.Module):
         cos_sin: torch.Tensor,
     ) -> tuple:
         cfg = self.cfg
-        if residual is None:
-            residual = hidden
-            hidden = ops.rms_norm(hidden, self.input_norm, cfg.rms_norm_eps)
+        fp8 = getattr(self, "fp8_w", None)
+        # fused act-quant path: (add-)RMSNorm emits fp8 + row scales
+        # directly, skipping the bf16 round trip;
+        # LoRA needs the bf16 normed activations, so it keeps the plain
+        # norm + per-tensor fp8 GEMM path.
+        fuse_q = (fp8 is not None and hidden.is_cuda
+                  and meta.lora_idx is None and not meta.lora_groups)
+        if fuse_q:
+            if residual is None:
+                residual = hidden.clone()
+                xq, sx = ops.rms_norm_fp8(
+                    hidden, self.input_norm, cfg.rms_norm_eps)
+            else:
+                xq, sx = ops.rms_norm_fp8(
+                    hidden, self.input_norm, cfg.rms_norm_eps,
+                    residual=residual)
+            qkv = ops.fp8_linear_rowwise(xq, sx, *fp8["qkv"])
         else:
-            hidden, residual = ops.fused_add_rms_norm(
-                hidden, residual, self.input_norm, cfg.rms_norm_eps
-            )
-        if getattr(self, "fp8_w", None) is not None:
-            qkv = ops.fp8_linear(hidden, *self.fp8_w["qkv"])
-        else:
-            qkv = gemm_policy.linear(hidden, self.qkv_proj)
+            if residual is None:
+                residual = hidden
+                hidden = ops.rms_norm(hidden, self.input_norm,
+                                      cfg.rms_norm_eps)
+            else:
+                hidden, residual = ops.fused_add_rms_norm(
+                    hidden, residual, self.input_norm, cfg.rms_norm_eps
+                )
+            if fp8 is not None:
+                qkv = ops.fp8_linear(hidden, *fp8["qkv"])
+            else:
+                qkv = gemm_policy.linear(hidden, self.qkv_proj)
         if self.qkv_bias is not None:
             qkv = qkv + self.qkv_bias
         qs = self.q_heads * self.head_dim
@@ -253,13 +272,20 @@ class LlamaLayer(nn.Module):
             )
         attn_out = pstate.tp_all_reduce(attn_out)
 
-        hidden, residual = ops.fused_add_rms_norm(
-            attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
-        )
-        if getattr(self, "fp8_w", None) is not None:
-            gate_up = ops.fp8_linear(hidden, *self.fp8_w["gate_up"])
+        if fuse_q:
+            xq, sx = ops.rms_norm_fp8(
+                attn_out, self.post_attn_norm, cfg.rms_norm_eps,
+                residual=residual)
+            hidden = attn_out  # bf16 normed not materialized on this path
+            gate_up = ops.fp8_linear_rowwise(xq, sx, *fp8["gate_up"])
         else:
-            gate_up = gemm_policy.linear(hidden, self.gate_up_proj)
+            hidden, residual = ops.fused_add_rms_norm(
+                attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
+            )
+            if fp8 is not None:
+                gate_up = ops.fp8_linear(hidden, *fp8["gate_up"])
+            else:
+                gate_up = gemm_policy.linear(hidden, self.gate_up_proj)
         if slots is not None and meta.lora_idx is not None:
             slots.apply("gate", self.layer_idx, gate_up, hidden,
                         meta.lora_idx)
@@ -274,11 +300,16 @@ class LlamaLayer(nn.Module):
                 gate_up, hidden, meta.lora_groups, self.layer_idx,
                 "up", self.inter, self.inter,
             )
-        act = ops.silu_and_mul(gate_up)
-        if getattr(self, "fp8_w", None) is not None:
-            mlp_out = ops.fp8_linear(act, *self.fp8_w["down"])
+        if fuse_q:
+            aq, sa = ops.silu_and_mul_fp8(gate_up)
+            act = gate_up[:, : self.inter]  # only LoRA reads act; unused
+            mlp_out = ops.fp8_linear_rowwise(aq, sa, *fp8["down"])
         else:
-            mlp_out = gemm_policy.linear(act, self.down_proj)
+            act = ops.silu_and_mul(gate_up)
+            if fp8 is not None:
+                mlp_out = ops.fp8_linear(act, *fp8["down"])
+            else:
+                mlp_out = gemm_policy.linear(act, self.down_proj)
         if slots is not None and meta.lora_idx is not None:
             slots.apply("down", self.layer_idx, mlp_out, act, meta.lora_idx)
         elif meta.lora_groups:
@@ -352,7 +383,8 @@ class LlamaForCausalLM(nn.Module):
     @torch.no_grad()
     def quantize_fp8(self) -> None:
         """Convert the four per-layer projections to OCP fp8-e4m3 with
-        per-tensor scales (vLLM --quantization fp8 analogue); halves
+        per-output-channel scales (vLLM --quantization fp8 analogue);
+        halves
         weight HBM and runs GEMMs through the fp8 MFMA pipe. Embedding,
         lm_head and norms stay bf16."""
         for layer in self.layers:
@@ -361,7 +393,7 @@ class LlamaForCausalLM(nn.Module):
                                ("gate_up", "gate_up_proj"),
                                ("down", "down_proj")):
                 w = getattr(layer, pname)
-                w_q, scale = ops.fp8_quantize_weight(w.data)
+                w_q, scale = ops.fp8_quantize_weight_rowwise(w.data)
                 fp8_w[key] = (w_q, scale.to(w.device))
                 # free the bf16 copy (replace with a tiny stub so
                 # state_dict/save paths still see the attribute)

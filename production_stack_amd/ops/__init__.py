@@ -321,13 +321,18 @@ def fp8_linear(
     if x.is_cuda:
         sx = (x.abs().amax().clamp(min=1e-6) / 448.0).to(torch.float32)
         xq = (x / sx).to(torch.float8_e4m3fn)
+        if w_scale.dim() == 1:  # per-channel weight scales
+            return fp8_linear_rowwise(
+                xq, sx.expand(x.shape[0]), w_q, w_scale)
         return torch._scaled_mm(
             xq, w_q.t(), scale_a=sx, scale_b=w_scale,
             out_dtype=torch.bfloat16,
         )
-    return torch.nn.functional.linear(
-        x.float(), w_q.float() * w_scale.float()
-    ).to(x.dtype)
+    wf = w_q.float() * (
+        w_scale.float()[:, None] if w_scale.dim() == 1
+        else w_scale.float()
+    )
+    return torch.nn.functional.linear(x.float(), wf).to(x.dtype)
 
 
 def fp8_quantize_weight(w: torch.Tensor):
@@ -335,3 +340,71 @@ def fp8_quantize_weight(w: torch.Tensor):
     scale = (w.abs().amax().float().clamp(min=1e-6) / 448.0)
     w_q = (w.float() / scale).clamp(-448, 448).to(torch.float8_e4m3fn)
     return w_q, scale
+
+
+def fp8_quantize_weight_rowwise(w: torch.Tensor):
+    """Per-output-channel weight quantization: (w_q fp8 [N,K], scales
+    [N] f32) — finer than per-tensor, same GEMM cost with rowwise-scaled
+    fp8 matmul."""
+    scales = (w.abs().amax(dim=1).float().clamp(min=1e-6) / 448.0)
+    w_q = (w.float() / scales[:, None]).clamp(-448, 448).to(
+        torch.float8_e4m3fn)
+    return w_q, scales
+
+
+def rms_norm_fp8(x, w, eps, residual=None):
+    """Fused (add-)RMSNorm emitting fp8-e4m3 rows + per-row scales. When
+    `residual` is given it is updated in place (residual += x) like
+    fused_add_rms_norm. GPU only."""
+    _require_ext()
+    T, D = x.shape
+    out_q = torch.empty((T, D), dtype=torch.float8_e4m3fn, device=x.device)
+    scales = torch.empty(T, dtype=torch.float32, device=x.device)
+    _C.rms_norm_fp8(out_q, scales, x,
+                    residual if residual is not None else x, w, eps,
+                    1 if residual is not None else 0)
+    return out_q, scales
+
+
+def silu_and_mul_fp8(x):
+    _require_ext()
+    T, D2 = x.shape
+    out_q = torch.empty((T, D2 // 2), dtype=torch.float8_e4m3fn,
+                        device=x.device)
+    scales = torch.empty(T, dtype=torch.float32, device=x.device)
+    _C.silu_and_mul_fp8(out_q, scales, x)
+    return out_q, scales
+
+
+_ROWWISE_SCALED_MM = None  # probed once: does _scaled_mm accept [M,1]x[1,N]?
+
+
+def fp8_linear_rowwise(xq, sx, w_q, sw):
+    """out[M,N] = (xq*sx[:,None]) @ (w_q*sw[None,:]).T in fp8 tensor
+    cores with per-row activation + per-channel weight scales. Uses
+    _scaled_mm's rowwise scaling when this build supports it, else a
+    unit-scale fp8 GEMM with the outer-product scale applied after."""
+    global _ROWWISE_SCALED_MM
+    if _ROWWISE_SCALED_MM is None:
+        try:
+            torch._scaled_mm(
+                torch.zeros(16, 32, dtype=torch.float8_e4m3fn,
+                            device=xq.device),
+                torch.zeros(32, 16, dtype=torch.float8_e4m3fn,
+                            device=xq.device),
+                scale_a=torch.ones(16, 1, device=xq.device),
+                scale_b=torch.ones(1, 16, device=xq.device),
+                out_dtype=torch.bfloat16,
+            )
+            _ROWWISE_SCALED_MM = True
+        except (RuntimeError, TypeError):
+            _ROWWISE_SCALED_MM = False
+    if _ROWWISE_SCALED_MM:
+        return torch._scaled_mm(
+            xq, w_q.t(), scale_a=sx[:, None].contiguous(),
+            scale_b=sw[None, :].contiguous(), out_dtype=torch.bfloat16,
+        )
+    one = torch.ones((), dtype=torch.float32, device=xq.device)
+    out = torch._scaled_mm(xq, w_q.t(), scale_a=one, scale_b=one,
+                           out_dtype=torch.float32)
+    return (out * sx[:, None] * sw[None, :]).to(torch.bfloat16)

@@ -661,3 +661,51 @@ def test_prefill_mfma32_v5_windowed():
         q.cpu(), k_cache.cpu(), v_cache.cpu(), bt.cpu(), token_seq,
         token_pos, 0.0883, window=W)
     _close(got.cpu(), want, atol=3e-2, rtol=3e-2)
+
+
+def test_rms_norm_fp8_fused_act_quant():
+    """Fused (add-)RMSNorm -> fp8 + row scales vs the bf16 kernel output
+    (dequantized error bounded by fp8 e4m3 resolution)."""
+    torch.manual_seed(50)
+    for fused in (False, True):
+        T, D = 37, 4096
+        x = torch.randn(T, D, dtype=torch.bfloat16, device="cuda")
+        res = torch.randn(T, D, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(D, dtype=torch.bfloat16, device="cuda")
+        if fused:
+            x2, r2 = x.clone(), res.clone()
+            want, r_want = ops.fused_add_rms_norm(x2, r2, w, 1e-5)
+            xq, sx = ops.rms_norm_fp8(x.clone(), w, 1e-5,
+                                      residual=res.clone())
+        else:
+            want = ops.rms_norm(x, w, 1e-5)
+            xq, sx = ops.rms_norm_fp8(x, w, 1e-5)
+        got = xq.float() * sx[:, None]
+        err = (got - want.float()).abs()
+        tol = 0.02 * want.float().abs() + sx[:, None] * 8
+        assert int((err > tol).sum()) == 0, err.max()
+
+
+def test_silu_and_mul_fp8():
+    torch.manual_seed(51)
+    T, D = 29, 14336
+    x = torch.randn(T, 2 * D, dtype=torch.bfloat16, device="cuda")
+    want = ops.silu_and_mul(x)
+    aq, sa = ops.silu_and_mul_fp8(x)
+    got = aq.float() * sa[:, None]
+    err = (got - want.float()).abs()
+    tol = 0.02 * want.float().abs() + sa[:, None] * 8
+    assert int((err > tol).sum()) == 0, err.max()
+
+
+def test_fp8_linear_rowwise_matches_reference():
+    torch.manual_seed(52)
+    M, N, K = 96, 512, 1024
+    x = torch.randn(M, K, dtype=torch.bfloat16, device="cuda") / 4
+    w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda") / 4
+    w_q, sw = ops.fp8_quantize_weight_rowwise(w)
+    sx = (x.abs().amax(dim=1).float().clamp(min=1e-6) / 448.0)
+    xq = (x.float() / sx[:, None]).to(torch.float8_e4m3fn)
+    got = ops.fp8_linear_rowwise(xq, sx, w_q, sw).float().cpu()
+    want = (x.float() @ w.float().t()).cpu()
+    _close(got, want, atol=0.15, rtol=0.05)
