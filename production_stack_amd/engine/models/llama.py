@@ -144,7 +144,8 @@ class LlamaLayer(nn.Module):
         # LoRA needs the bf16 normed activations, so it keeps the plain
         # norm + per-tensor fp8 GEMM path.
         fuse_q = (fp8 is not None and hidden.is_cuda
-                  and meta.lora_idx is None and not meta.lora_groups)
+                  and meta.lora_idx is None and not meta.lora_groups
+                  and ops.fp8_rowwise_supported(hidden.device))
         if fuse_q:
             if residual is None:
                 residual = hidden.clone()
@@ -393,7 +394,11 @@ class LlamaForCausalLM(nn.Module):
                                ("gate_up", "gate_up_proj"),
                                ("down", "down_proj")):
                 w = getattr(layer, pname)
-                w_q, scale = ops.fp8_quantize_weight_rowwise(w.data)
+                if (w.is_cuda
+                        and ops.fp8_rowwise_supported(w.device)):
+                    w_q, scale = ops.fp8_quantize_weight_rowwise(w.data)
+                else:
+                    w_q, scale = ops.fp8_quantize_weight(w.data)
                 fp8_w[key] = (w_q, scale.to(w.device))
                 # free the bf16 copy (replace with a tiny stub so
                 # state_dict/save paths still see the attribute)

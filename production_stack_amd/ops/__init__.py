@@ -379,27 +379,34 @@ def silu_and_mul_fp8(x):
 _ROWWISE_SCALED_MM = None  # probed once: does _scaled_mm accept [M,1]x[1,N]?
 
 
-def fp8_linear_rowwise(xq, sx, w_q, sw):
-    """out[M,N] = (xq*sx[:,None]) @ (w_q*sw[None,:]).T in fp8 tensor
-    cores with per-row activation + per-channel weight scales. Uses
-    _scaled_mm's rowwise scaling when this build supports it, else a
-    unit-scale fp8 GEMM with the outer-product scale applied after."""
+def fp8_rowwise_supported(device="cuda") -> bool:
+    """Whether this build's _scaled_mm takes rowwise scales. When it does
+    not, the fused per-row act-quant path would need a slow fp32-out
+    fallback, so callers keep the per-tensor fp8 pipeline instead."""
     global _ROWWISE_SCALED_MM
     if _ROWWISE_SCALED_MM is None:
         try:
             torch._scaled_mm(
                 torch.zeros(16, 32, dtype=torch.float8_e4m3fn,
-                            device=xq.device),
+                            device=device),
                 torch.zeros(32, 16, dtype=torch.float8_e4m3fn,
-                            device=xq.device),
-                scale_a=torch.ones(16, 1, device=xq.device),
-                scale_b=torch.ones(1, 16, device=xq.device),
+                            device=device),
+                scale_a=torch.ones(16, 1, device=device),
+                scale_b=torch.ones(1, 16, device=device),
                 out_dtype=torch.bfloat16,
             )
             _ROWWISE_SCALED_MM = True
         except (RuntimeError, TypeError):
             _ROWWISE_SCALED_MM = False
-    if _ROWWISE_SCALED_MM:
+    return _ROWWISE_SCALED_MM
+
+
+def fp8_linear_rowwise(xq, sx, w_q, sw):
+    """out[M,N] = (xq*sx[:,None]) @ (w_q*sw[None,:]).T in fp8 tensor
+    cores with per-row activation + per-channel weight scales. Uses
+    _scaled_mm's rowwise scaling when this build supports it, else a
+    unit-scale fp8 GEMM with the outer-product scale applied after."""
+    if fp8_rowwise_supported(xq.device):
         return torch._scaled_mm(
             xq, w_q.t(), scale_a=sx[:, None].contiguous(),
             scale_b=sw[None, :].contiguous(), out_dtype=torch.bfloat16,
