@@ -36,6 +36,16 @@ int ps_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          int GQ, int head_dim, int block_size, int num_splits,
                          long q_stride, int variant, int kv_fp8, int window,
                          hipStream_t stream);
+int ps_prefill_mfma32_splits(int num_tiles, int KH, int GQ);
+int ps_paged_attn_prefill_mfma32(void* out, void* ws_o, void* ws_ml,
+                                 long q_tokens, const void* q,
+                                 const void* k_cache, const void* v_cache,
+                                 const void* block_tables,
+                                 const void* tile_info, int num_tiles,
+                                 int num_q_heads, int max_blocks,
+                                 float scale, int KH, int GQ, int head_dim,
+                                 long q_stride, int kv_fp8, int window,
+                                 hipStream_t stream);
 int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                const void* v_cache, const void* block_tables,
                                const void* tile_info, int num_tiles,
@@ -241,12 +251,36 @@ void paged_attn_prefill_mfma(at::Tensor out, at::Tensor q,
   TORCH_CHECK(k_cache.size(2) == 16, "mfma prefill needs block_size 16");
   const int GQ = QH / KH;
   const int NT = (int)tile_info.size(0);
-  int rc = ps_paged_attn_prefill_mfma(
-      out.data_ptr(), q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-      block_tables.data_ptr(), tile_info.data_ptr(), NT, QH,
-      (int)block_tables.size(1), (float)scale, KH, GQ, HD,
-      q_row_stride(q, HD), (int)variant, kv_fp8, (int)window,
-      current_stream());
+  int rc;
+  if (variant == 5) {
+    // v5 with split-KV: allocate fp32 partial workspace when the launch
+    // is small enough that the splits heuristic fires
+    const int splits = ps_prefill_mfma32_splits(NT, KH, QH / KH);
+    const long T = q.size(0);
+    void *ws_o_p = nullptr, *ws_ml_p = nullptr;
+    at::Tensor ws_o, ws_ml;
+    if (splits > 1) {
+      ws_o = at::empty({(long)splits * T * QH * HD},
+                       q.options().dtype(at::kFloat));
+      ws_ml = at::empty({2L * splits * T * QH},
+                        q.options().dtype(at::kFloat));
+      ws_o_p = ws_o.data_ptr();
+      ws_ml_p = ws_ml.data_ptr();
+    }
+    rc = ps_paged_attn_prefill_mfma32(
+        out.data_ptr(), ws_o_p, ws_ml_p, T, q.data_ptr(),
+        k_cache.data_ptr(), v_cache.data_ptr(), block_tables.data_ptr(),
+        tile_info.data_ptr(), NT, QH, (int)block_tables.size(1),
+        (float)scale, KH, QH / KH, HD, q_row_stride(q, HD), kv_fp8,
+        (int)window, current_stream());
+  } else {
+    rc = ps_paged_attn_prefill_mfma(
+        out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+        v_cache.data_ptr(), block_tables.data_ptr(), tile_info.data_ptr(),
+        NT, QH, (int)block_tables.size(1), (float)scale, KH, GQ, HD,
+        q_row_stride(q, HD), (int)variant, kv_fp8, (int)window,
+        current_stream());
+  }
   TORCH_CHECK(rc == 0, "unsupported mfma prefill config: head_dim=", HD);
 }
 

@@ -330,7 +330,9 @@ __global__ __launch_bounds__(256, WPS) void paged_attn_prefill_mfma_kernel(
 
 extern "C" {
 
-int ps_paged_attn_prefill_mfma32(void* out, const void* q,
+int ps_prefill_mfma32_splits(int num_tiles, int KH, int GQ);
+int ps_paged_attn_prefill_mfma32(void* out, void* ws_o, void* ws_ml,
+                                 long q_tokens, const void* q,
                                  const void* k_cache, const void* v_cache,
                                  const void* block_tables,
                                  const void* tile_info, int num_tiles,
@@ -347,12 +349,13 @@ int ps_paged_attn_prefill_mfma(void* out, const void* q, const void* k_cache,
                                int variant, int kv_fp8, int window,
                                hipStream_t stream) {
   if (head_dim != 128) return -1;
-  if (variant == 5) {  // 8-wave 32x32 swapped-QK^T kernel (256-row tiles)
+  if (variant == 5) {  // 8-wave 32x32 swapped-QK^T kernel; the torch
+    // binding passes workspace for split-KV — this raw entry runs S=1
     return ps_paged_attn_prefill_mfma32(
-
-        out, q, k_cache, v_cache, block_tables, tile_info, num_tiles,
-        num_q_heads, max_blocks, scale, KH, GQ, head_dim, q_stride, kv_fp8,
-        window, stream);
+        out, nullptr, nullptr, 0, q, k_cache, v_cache,
+        block_tables, tile_info, num_tiles,
+        num_q_heads, max_blocks, scale, KH, GQ, head_dim, q_stride,
+        kv_fp8, window, stream);
   }
   const int n_work = num_tiles * num_q_heads;
   dim3 grid(((n_work + 7) / 8) * 8);

@@ -41,9 +41,18 @@ PS_DEV ps_mbf16x8 ps_as_mbf16_32(ps_bf16x8 u) {
 // share ONE K/V staging pass (serving-shape prefill chunks are
 // staging-bound: a 200-row continuation re-reads a 2000+-token context).
 // Tile rows = 32 * (8/GQW).
+// Split-KV (gridDim.y > 1): split s processes kv chunks s, s+S, s+2S...
+// and writes UNNORMALIZED partials (o_acc, m, l) to the workspace; the
+// combine kernel merges them. Strided chunk assignment balances the
+// causal tail across splits. Raises the workgroup count for small
+// serving launches (1-block/CU quantization measured ~25% idle at the
+// 10-seq continuation shape).
 template <int HEAD_DIM, int GQW, typename KVT>
 __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
     unsigned short* __restrict__ out,            // [T, QH, HD]
+    float* __restrict__ ws_o,    // [S, T, QH, HD] (splits > 1)
+    float* __restrict__ ws_ml,   // [2, S, T, QH]
+    long q_tokens,
     const unsigned short* __restrict__ q,        // [T, QH, HD]
     const KVT* __restrict__ k_cache,             // [NB, KH, 16, HD]
     const KVT* __restrict__ v_cache,
@@ -51,6 +60,8 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
     const int* __restrict__ tile_info,     // [NT, 4]
     int max_blocks, float scale, int KH, int GQ, long q_stride,
     int QH, int n_work, int window) {
+  const int split = blockIdx.y;
+  const int n_splits = gridDim.y;
   using KVTr = ps_kv_traits<KVT>;
   using kvec8 = typename KVTr::vec8;
   using kvec4 = typename KVTr::vec4;
@@ -142,15 +153,15 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
   kvec8 kstage[2];
   kvec4 vstage[4];
   {
-    const KVT* ks = k_seg_ptr(0);
-    const KVT* vq = v_quad_ptr(0);
+    const KVT* ks = k_seg_ptr(split);
+    const KVT* vq = v_quad_ptr(split);
     kstage[0] = *(const kvec8*)ks;
     kstage[1] = *(const kvec8*)(ks + 64);
 #pragma unroll
     for (int i = 0; i < 4; i++) vstage[i] = *(const kvec4*)(vq + i * D);
   }
 
-  for (int chunk = 0; chunk < n_chunks; chunk++) {
+  for (int chunk = split; chunk < n_chunks; chunk += n_splits) {
     const int tok0 = chunk * PS_CHUNK32;
     // write prefetched K segments (16B-slot XOR swizzle by token low bits;
     // the XOR only permutes the low 3 slot bits, so slot+8 stays in 8..15)
@@ -173,9 +184,9 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
       }
     }
     __syncthreads();
-    if (chunk + 1 < n_chunks) {  // T14: next chunk's loads under compute
-      const KVT* ks = k_seg_ptr(chunk + 1);
-      const KVT* vq = v_quad_ptr(chunk + 1);
+    if (chunk + n_splits < n_chunks) {  // T14: next loads under compute
+      const KVT* ks = k_seg_ptr(chunk + n_splits);
+      const KVT* vq = v_quad_ptr(chunk + n_splits);
       kstage[0] = *(const kvec8*)ks;
       kstage[1] = *(const kvec8*)(ks + 64);
 #pragma unroll
@@ -301,6 +312,27 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
     __syncthreads();  // protect k_lds/v_t before next chunk's staging
   }
 
+  if (n_splits > 1) {
+    // ---- epilogue: write unnormalized partials for the combiner ----
+    if (hi == 0 && wq0 + rc < n_rows) {
+      const long t = q_tok0 + wq0 + rc;
+      ws_ml[((long)split * q_tokens + t) * QH + qh] = m_run;
+      ws_ml[((long)(n_splits + split) * q_tokens + t) * QH + qh] = l_run;
+    }
+#pragma unroll
+    for (int r = 0; r < 16; r++) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const int lrow = wq0 + row;
+      if (lrow >= n_rows) continue;
+      float* orow = ws_o +
+          (((long)split * q_tokens + q_tok0 + lrow) * QH + qh) * D;
+#pragma unroll
+      for (int dt = 0; dt < 4; dt++)
+        orow[dt * 32 + rc] = o_acc[dt][r];
+    }
+    return;
+  }
+
   // ---- epilogue: normalize + store ----
   const float inv_own = l_run > 0.f ? 1.f / l_run : 0.f;
 #pragma unroll
@@ -316,9 +348,44 @@ __global__ __launch_bounds__(512, 2) void paged_attn_prefill_mfma32_kernel(
   }
 }
 
+// merge the splits: out = sum_s w_s O_s / sum_s w_s l_s, w_s=exp2(m_s-M)
+__global__ void prefill_split_combine_kernel(
+    unsigned short* __restrict__ out,      // [T, QH, 128]
+    const float* __restrict__ ws_o,        // [S, T, QH, 128]
+    const float* __restrict__ ws_ml,       // [2, S, T, QH]
+    long q_tokens, int QH, int n_splits) {
+  const long th = blockIdx.x;  // t * QH + qh
+  if (th >= q_tokens * QH) return;
+  const int d = threadIdx.x;
+  float M = PS_NEG_INF;
+  for (int s = 0; s < n_splits; s++)
+    M = fmaxf(M, ws_ml[(long)s * q_tokens * QH + th]);
+  float denom = 0.f, acc = 0.f;
+  for (int s = 0; s < n_splits; s++) {
+    const float m = ws_ml[(long)s * q_tokens * QH + th];
+    const float l = ws_ml[((long)(n_splits + s) * q_tokens) * QH + th];
+    const float w = (m > PS_NEG_INF && l > 0.f) ? exp2f(m - M) : 0.f;
+    denom += w * l;
+    acc += w * ws_o[((long)s * q_tokens * QH + th) * 128 + d];
+  }
+  out[th * 128 + d] =
+      ps_f32_to_bf16(denom > 0.f ? acc / denom : 0.f);
+}
+
 extern "C" {
 
-int ps_paged_attn_prefill_mfma32(void* out, const void* q,
+// Split count the launcher will use for a shape (callers size ws with it)
+int ps_prefill_mfma32_splits(int num_tiles, int KH, int GQ) {
+  int gqw = 1;
+  while (gqw < 4 && GQ % (gqw * 2) == 0) gqw *= 2;
+  const int n_work = num_tiles * KH * (GQ / gqw);
+  if (n_work >= 500) return 1;
+  int s = (500 + n_work - 1) / n_work;
+  return s > 4 ? 4 : s;
+}
+
+int ps_paged_attn_prefill_mfma32(void* out, void* ws_o, void* ws_ml,
+                                 long q_tokens, const void* q,
                                  const void* k_cache, const void* v_cache,
                                  const void* block_tables,
                                  const void* tile_info, int num_tiles,
@@ -333,11 +400,14 @@ int ps_paged_attn_prefill_mfma32(void* out, const void* q,
   int gqw = 1;
   while (gqw < 4 && GQ % (gqw * 2) == 0) gqw *= 2;
   const int n_work = num_tiles * KH * (GQ / gqw);
-  dim3 grid(((n_work + 7) / 8) * 8);
+  int splits = ps_prefill_mfma32_splits(num_tiles, KH, GQ);
+  if (ws_o == nullptr) splits = 1;
+  dim3 grid(((n_work + 7) / 8) * 8, splits);
 #define PS_PF32_LAUNCH(GQW, KVT)                                          \
   paged_attn_prefill_mfma32_kernel<128, GQW, KVT>                         \
       <<<grid, 512, 0, stream>>>(                                         \
-          (unsigned short*)out, (const unsigned short*)q,                 \
+          (unsigned short*)out, (float*)ws_o, (float*)ws_ml, q_tokens,    \
+          (const unsigned short*)q,                                       \
           (const KVT*)k_cache, (const KVT*)v_cache,                       \
           (const int*)block_tables, (const int*)tile_info, max_blocks,    \
           scale, KH, GQ, q_stride, num_q_heads, n_work, window)
@@ -351,6 +421,12 @@ int ps_paged_attn_prefill_mfma32(void* out, const void* q,
     else PS_PF32_LAUNCH(1, unsigned short);
   }
 #undef PS_PF32_LAUNCH
+  if (splits > 1) {
+    prefill_split_combine_kernel<<<
+        dim3((unsigned)(q_tokens * num_q_heads)), 128, 0, stream>>>(
+        (unsigned short*)out, (const float*)ws_o, (const float*)ws_ml,
+        q_tokens, num_q_heads, splits);
+  }
   return 0;
 }
 
