@@ -48,8 +48,23 @@ PS_DEV ps_g8bf16x8 ps_as_g8bf16(ps_bf16x8 u) {
   return v.bf;
 }
 
-// byte-offset swizzle within a 16 KiB half-tile (involution, 16B-granular)
-PS_DEV int ps_g8_swz(int b) { return b ^ (((b >> 7) & 7) << 4); }
+// byte-offset swizzle within a 16 KiB half-tile (involution, 16B-granular).
+// Two candidates, A/B-selectable at compile time: the attention-style
+// 8-row stripe spread (default) and the guide's st_16x32 (XOR byte bit 5
+// with bit 9 within each 1 KiB subtile, m201). Measured on MI355X at the
+// Llama serving shapes: attn-style wins every cell (e.g. gate_up M=2048
+// 958 vs 766 TF) — the frag reads here stride 128 B, not 256, so the
+// 8-row stripe spread matches the bank geometry better.
+#ifndef PS_G8_SWZ_VARIANT
+#define PS_G8_SWZ_VARIANT 0
+#endif
+PS_DEV int ps_g8_swz(int b) {
+#if PS_G8_SWZ_VARIANT == 1
+  return b ^ (((b >> 9) & 1) << 5);
+#else
+  return b ^ (((b >> 7) & 7) << 4);
+#endif
+}
 
 // LDS byte offset of a generic pointer to a __shared__ object
 PS_DEV unsigned ps_g8_lds_off(const void* p) {
