@@ -90,6 +90,9 @@ def main() -> None:
     ap.add_argument("--no-async-scheduling", action="store_true",
                     help="disable one-step-lagged sampling")
     ap.add_argument("--quantization", default=None, choices=[None, "fp8"])
+    ap.add_argument("--unified-mixed-steps", action="store_true",
+                    help="run mixed prefill+decode steps as one eager "
+                         "batch (weights read once per step)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -110,6 +113,7 @@ def main() -> None:
         model=args.model,
         max_model_len=args.max_model_len,
         async_scheduling=not args.no_async_scheduling,
+        unified_mixed_steps=args.unified_mixed_steps,
         quantization=args.quantization,
         seed=1234 + rank,
         cache=CacheConfig(

@@ -471,6 +471,7 @@ class ModelRunner:
             self.graphs is not None
             and self.pipeline is None
             and not getattr(self, "tp_serving", False)
+            and not self.config.unified_mixed_steps
         ):
             dec = [s for s in out.scheduled if s.is_decode]
             pre = [s for s in out.scheduled if not s.is_decode]
@@ -594,6 +595,7 @@ class ModelRunner:
         use_graph = (
             self.graphs is not None and dec
             and self.graphs.bucket_for(len(dec)) is not None
+            and not (pre and self.config.unified_mixed_steps)
         )
         if use_graph:
             got = self._execute_decode_graph(
