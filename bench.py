@@ -90,9 +90,10 @@ def main() -> None:
     ap.add_argument("--no-async-scheduling", action="store_true",
                     help="disable one-step-lagged sampling")
     ap.add_argument("--quantization", default=None, choices=[None, "fp8"])
-    ap.add_argument("--unified-mixed-steps", action="store_true",
-                    help="run mixed prefill+decode steps as one eager "
-                         "batch (weights read once per step)")
+    ap.add_argument("--no-unified-mixed-steps", dest="unified_mixed_steps",
+                    action="store_false", default=True,
+                    help="restore the split graph+eager mixed-step path "
+                         "(unified reads weights once per step; default)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))

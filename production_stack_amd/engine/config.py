@@ -321,12 +321,13 @@ class EngineConfig:
     weights_path: Optional[str] = None  # safetensors dir (optional)
     tokenizer: str = "synthetic"  # or a path to a tokenizer.json dir
     enforce_eager: bool = False  # True disables hipGraph decode capture
-    # Mixed prefill+decode steps: False (default) replays decode rows
-    # through their captured hipGraph and runs prefill chunks eagerly
-    # (wins when decode is launch-bound); True runs one unified eager
+    # Mixed prefill+decode steps: True (default) runs one unified eager
     # batch so each layer's weights are read ONCE per step instead of
-    # once for the graph replay and once for the prefill GEMMs.
-    unified_mixed_steps: bool = False
+    # once for the decode graph replay and once for the prefill GEMMs —
+    # measured +15-25% output tok/s AND lower TTFT at every load from 32
+    # to 320 users on MI355X (graph replay remains for pure-decode
+    # steps). False restores the split graph+eager mixed path.
+    unified_mixed_steps: bool = True
     # graph-safe batched LoRA (BGMV slots); when False, adapters still run
     # through the eager grouped path
     # one-step-lagged sampling: the host schedules/launches step N+1 while

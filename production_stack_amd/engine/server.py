@@ -857,6 +857,8 @@ def main() -> None:
     ap.add_argument("--quantization", default=None,
                     choices=[None, "fp8"],
                     help="fp8 = OCP e4m3 weights through the fp8 MFMA pipe")
+    ap.add_argument("--no-unified-mixed-steps", dest="unified_mixed_steps",
+                    action="store_false", default=True)
     ap.add_argument("--async-scheduling", action="store_true",
                     help="one-step-lagged sampling (greedy-exact overlap)")
     ap.add_argument("--num-speculative-tokens", type=int, default=0,
@@ -910,6 +912,7 @@ def main() -> None:
             num_speculative_tokens=args.num_speculative_tokens,
         ),
         async_scheduling=args.async_scheduling,
+        unified_mixed_steps=args.unified_mixed_steps,
         quantization=args.quantization,
         enable_lora=args.enable_lora,
         max_loras=args.max_loras,
