@@ -44,11 +44,22 @@ VOCAB_LOW, VOCAB_HIGH = 16, 128000
 
 
 class User:
-    def __init__(self, uid, rng, system, vocab_high=VOCAB_HIGH):
+    def __init__(self, uid, rng, system, vocab_high=VOCAB_HIGH,
+                 history_fill=0):
         self.uid = uid
         self.rng = rng
         self.vocab_high = vocab_high
         self.history = list(system)
+        # The reference harness starts every user with a LONG chat history
+        # (run.sh: 20000 tokens, truncated into the model window), so the
+        # workload is stationary from the first round — users sit AT the
+        # history cap and every round re-prefills the slid window. Start
+        # saturated so any measurement window sees the same steady state
+        # (no growing-history drift between short and long windows).
+        if history_fill > 0:
+            self.history += rng.integers(
+                VOCAB_LOW, vocab_high, size=history_fill
+            ).tolist()
         self.round = 0
         self.submit_time = 0.0
         # randomized FIRST answer length desynchronizes the user rounds:
@@ -136,7 +147,14 @@ def main() -> None:
     system = rng.integers(
         VOCAB_LOW, vocab_high, size=SYSTEM_PROMPT_TOKENS
     ).tolist()
-    users = [User(u, rng, system, vocab_high) for u in range(args.users)]
+    # saturate to the serving cap (matches submit()'s trim limit)
+    fill = max(
+        0,
+        (args.max_model_len - ANSWER_TOKENS - QUESTION_TOKENS - 16)
+        - SYSTEM_PROMPT_TOKENS - QUESTION_TOKENS,
+    )
+    users = [User(u, rng, system, vocab_high, history_fill=fill)
+             for u in range(args.users)]
     answers: dict = {}
     ttfts: list = []
     in_window = False
