@@ -625,6 +625,11 @@ class ModelRunner:
                         return None
                     src = []
                     for i in ph.tolist():
+                        if i < prefill_rows:
+                            # a preempted seq re-admitted as recompute
+                            # prefill still carries -1 outputs: the sync
+                            # path must resolve them first
+                            return None
                         row = self._prev_rows.get(
                             decode_rids[i - prefill_rows]
                         )
