@@ -142,6 +142,14 @@ class LLMEngine:
             # resolve in-flight placeholders, then take the sync path for
             # this step (prepare must see real token values)
             prev = self._finalize_pending()
+            # finalize may have finished (and freed) sequences this step's
+            # schedule still references — executing them would read a
+            # dead block table
+            out.scheduled = [
+                s for s in out.scheduled if not s.seq.finished
+            ]
+            if not out.scheduled:
+                return prev + self._build_outputs(out)
             sampled = self.runner.execute(out, self.block_manager)
             fin = self.scheduler.on_step_done(
                 out, sampled, self.model_cfg.eos_token_id,
