@@ -40,46 +40,56 @@ from production_stack_amd.engine.sampling import SamplingParams
 SYSTEM_PROMPT_TOKENS = 1000
 QUESTION_TOKENS = 100
 ANSWER_TOKENS = 100
+NUM_ROUNDS = 10          # reference run.sh: 10 rounds per conversation
+USER_INFO_TOKENS = 2500  # reference CHAT_HISTORY=20000, scaled to fit
+# 320 users' KV in 288 GB HBM (20k x 320 users = 880 GB of KV exists on
+# no single GPU; the reference's own sweep needs a cluster-scale KV
+# tier for it). Histories are append-only within a conversation exactly
+# like the reference harness (ChatHistory only ever appends), and a
+# fresh conversation (new user info) starts after NUM_ROUNDS.
 VOCAB_LOW, VOCAB_HIGH = 16, 128000
-# history-cap hysteresis: at the cap, trim back this many tokens so the
-# following ~6 rounds extend a prefix the cache still holds
-TRIM_SLACK = 1800
 
 
 class User:
+    """One user = a stream of conversations, each NUM_ROUNDS rounds with
+    an append-only history (reference multi-round-qa.py ChatHistory
+    semantics: system prompt + per-user info on round 1, then question/
+    answer pairs appended — never trimmed). Starting users at random
+    round phases makes the population stationary from the first step."""
+
     def __init__(self, uid, rng, system, vocab_high=VOCAB_HIGH,
-                 history_fill=0):
+                 start_round=0, info_tokens=USER_INFO_TOKENS,
+                 num_rounds=NUM_ROUNDS):
         self.uid = uid
         self.rng = rng
         self.vocab_high = vocab_high
-        self.history = list(system)
-        # The reference harness starts every user with a LONG chat history
-        # (run.sh: 20000 tokens, truncated into the model window), so its
-        # canonical load is already at the history cap. Trimming on every
-        # round would invalidate the prefix cache each time (measured
-        # 1.9k tok/s: pure re-prefill); real serving clients trim with
-        # hysteresis instead, so submit() cuts back TRIM_SLACK tokens at
-        # the cap and most rounds extend a cached prefix. Starting each
-        # user at a RANDOM phase of that grow/trim cycle makes the
-        # population stationary from the first step — short and long
-        # measurement windows agree.
-        if history_fill > 0:
-            self.history += rng.integers(
-                VOCAB_LOW, vocab_high, size=history_fill
-            ).tolist()
+        self.system = system
+        self.info_tokens = info_tokens
+        self.num_rounds = num_rounds
         self.round = 0
         self.submit_time = 0.0
-        # randomized FIRST answer length desynchronizes the user rounds:
-        # without it every conversation finishes on the same step and the
-        # measured window is an artificial all-decode phase with no TTFT
-        # samples; with it the steady state is the real serving mix of
-        # per-round prefills over running decodes.
+        # randomized FIRST answer length desynchronizes the user rounds
         self.first_len = int(rng.integers(10, 2 * ANSWER_TOKENS))
+        self._new_conversation()
+        for _ in range(start_round):
+            self.round += 1
+            self.history.extend(self.rng.integers(
+                VOCAB_LOW, vocab_high,
+                size=QUESTION_TOKENS + ANSWER_TOKENS).tolist())
+
+    def _new_conversation(self) -> None:
+        self.round = 0
+        self.conv = getattr(self, "conv", -1) + 1
+        self.history = list(self.system) + self.rng.integers(
+            VOCAB_LOW, self.vocab_high, size=self.info_tokens
+        ).tolist()
 
     def answer_len(self) -> int:
         return self.first_len if self.round <= 1 else ANSWER_TOKENS
 
     def next_prompt(self) -> list:
+        if self.round >= self.num_rounds:
+            self._new_conversation()
         self.round += 1
         q = self.rng.integers(
             VOCAB_LOW, self.vocab_high, size=QUESTION_TOKENS
@@ -98,7 +108,7 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=48)
     ap.add_argument("--users", type=int, default=320, help="conversations per GPU (reference run.sh canonical: 320)")
     ap.add_argument("--model", default="llama-3-8b")
-    ap.add_argument("--max-model-len", type=int, default=4096)
+    ap.add_argument("--max-model-len", type=int, default=8192)
     ap.add_argument("--max-num-batched-tokens", type=int, default=2048)
     ap.add_argument("--kv-cache-dtype", default="auto",
                     choices=["auto", "bf16", "fp8", "fp8_e4m3"])
@@ -154,14 +164,21 @@ def main() -> None:
     system = rng.integers(
         VOCAB_LOW, vocab_high, size=SYSTEM_PROMPT_TOKENS
     ).tolist()
-    # stationary ensemble: histories start uniformly across one
-    # grow/trim cycle [cap - TRIM_SLACK, cap]
-    limit = args.max_model_len - ANSWER_TOKENS - QUESTION_TOKENS - 16
-    fill_hi = max(0, limit - SYSTEM_PROMPT_TOKENS - QUESTION_TOKENS)
-    fill_lo = max(0, fill_hi - TRIM_SLACK)
+    # stationary ensemble: users start at uniformly random rounds of
+    # their conversation; size per-user info (and, for tiny test
+    # windows, the round count) so a full conversation fits the window
+    per_round = QUESTION_TOKENS + ANSWER_TOKENS
+    budget = (args.max_model_len - SYSTEM_PROMPT_TOKENS
+              - ANSWER_TOKENS - QUESTION_TOKENS - 16)
+    rounds_eff = NUM_ROUNDS
+    info = min(USER_INFO_TOKENS, budget - rounds_eff * per_round)
+    if info < 256:
+        info = max(0, min(256, budget - per_round))
+        rounds_eff = max(1, (budget - info) // per_round)
     users = [
         User(u, rng, system, vocab_high,
-             history_fill=int(rng.integers(fill_lo, max(fill_hi, 1))))
+             start_round=int(rng.integers(0, rounds_eff)),
+             info_tokens=info, num_rounds=rounds_eff)
         for u in range(args.users)
     ]
     answers: dict = {}
@@ -172,16 +189,8 @@ def main() -> None:
     done_once: set = set()
 
     def submit(user: User) -> None:
-        # cap history so prompt + answer fits the model length
-        limit = args.max_model_len - ANSWER_TOKENS - QUESTION_TOKENS - 16
-        if len(user.history) > limit:
-            keep = max(limit - SYSTEM_PROMPT_TOKENS - TRIM_SLACK, 256)
-            user.history = (
-                user.history[:SYSTEM_PROMPT_TOKENS]
-                + user.history[-keep:]
-            )
         prompt = user.next_prompt()
-        rid = f"u{user.uid}-r{user.round}"
+        rid = f"u{user.uid}-c{user.conv}-r{user.round}"
         user.submit_time = time.perf_counter()
         user.steady_submit = in_steady
         answers[rid] = (user, [])
@@ -304,6 +313,8 @@ def main() -> None:
                         "workload": "multi-round-qa",
                         "users_per_gpu": args.users,
                         "system_prompt_tokens": SYSTEM_PROMPT_TOKENS,
+                        "user_history_tokens": USER_INFO_TOKENS,
+                        "rounds_per_conversation": NUM_ROUNDS,
                         "question_tokens": QUESTION_TOKENS,
                         "answer_tokens": ANSWER_TOKENS,
                         "max_model_len": args.max_model_len,

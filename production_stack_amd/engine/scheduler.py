@@ -66,6 +66,10 @@ class Scheduler:
 
     # ------------------------------------------------------------------
     def add(self, seq: Sequence) -> None:
+        if seq.request_id in self._by_id:
+            raise ValueError(
+                f"duplicate request_id {seq.request_id!r}: a request with "
+                "this id is still live")
         if seq.num_prompt + 1 > self.max_model_len:
             raise ValueError(
                 f"prompt of {seq.num_prompt} tokens exceeds max_model_len "
