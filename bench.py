@@ -236,10 +236,13 @@ def main() -> None:
     t_guard = time.perf_counter()
     guard_cap_s = 300.0 if use_cuda else 120.0
     guard_steps = 0
+    max_wait = max(2, int(0.05 * len(users)))
     while (
         len(done_once) < steady_target
-        and time.perf_counter() - t_guard < guard_cap_s
+        or engine.scheduler.num_waiting > max_wait
     ):
+        if time.perf_counter() - t_guard >= guard_cap_s:
+            break
         run_step()
         guard_steps += 1
     in_steady = True
