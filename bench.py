@@ -237,11 +237,14 @@ def main() -> None:
     guard_cap_s = 300.0 if use_cuda else 120.0
     guard_steps = 0
     max_wait = max(2, int(0.05 * len(users)))
-    while (
-        len(done_once) < steady_target
-        or engine.scheduler.num_waiting > max_wait
-    ):
+    stable = 0
+    while True:
         if time.perf_counter() - t_guard >= guard_cap_s:
+            break
+        ok = (len(done_once) >= steady_target
+              and engine.scheduler.num_waiting <= max_wait)
+        stable = stable + 1 if ok else 0
+        if stable >= 20:  # equilibrium must HOLD, not just flicker
             break
         run_step()
         guard_steps += 1
