@@ -80,6 +80,14 @@ class User:
     def _new_conversation(self) -> None:
         self.round = 0
         self.conv = getattr(self, "conv", -1) + 1
+        # per-conversation round count drawn around NUM_ROUNDS (mean 10):
+        # with a FIXED length the closed loop bus-bunches — users drift
+        # into synchronized reset waves whose 3.6k-token prefills stall
+        # everyone (measured: 60-step windows oscillated 4.8-6.6k tok/s).
+        # Randomized lengths keep resets de-correlated and the process
+        # mixing toward a stationary distribution.
+        self.rounds_this_conv = int(self.rng.integers(
+            max(1, self.num_rounds - 4), self.num_rounds + 5))
         self.history = list(self.system) + self.rng.integers(
             VOCAB_LOW, self.vocab_high, size=self.info_tokens
         ).tolist()
@@ -88,7 +96,8 @@ class User:
         return self.first_len if self.round <= 1 else ANSWER_TOKENS
 
     def next_prompt(self) -> list:
-        if self.round >= self.num_rounds:
+        if self.round >= getattr(self, "rounds_this_conv",
+                                 self.num_rounds):
             self._new_conversation()
         self.round += 1
         q = self.rng.integers(
@@ -171,7 +180,7 @@ def main() -> None:
     budget = (args.max_model_len - SYSTEM_PROMPT_TOKENS
               - ANSWER_TOKENS - QUESTION_TOKENS - 16)
     rounds_eff = NUM_ROUNDS
-    info = min(USER_INFO_TOKENS, budget - rounds_eff * per_round)
+    info = min(USER_INFO_TOKENS, budget - (rounds_eff + 4) * per_round)
     if info < 256:
         info = max(0, min(256, budget - per_round))
         rounds_eff = max(1, (budget - info) // per_round)
