@@ -257,11 +257,13 @@ def main() -> None:
             break
         run_step()
         guard_steps += 1
-    # the drain criterion exits at a below-average-queue lull by
-    # construction (measured: a 20-step window right after it read ~30%
-    # above what 60/200-step windows agree on); run a fixed mixing
-    # period so short driver windows start from a typical state
-    for _ in range(100 if use_cuda else 5):
+    # the drain criterion exits before the ensemble's age structure is
+    # stationary: conversation resets (the expensive 3.6k-token
+    # prefills) only ramp in over ~1 conversation period (~1000 steps),
+    # so a 20-step window right after the drain read ~30% above what
+    # 60-200-step windows agree on. A fixed mixing period lets short
+    # driver windows read the converged value
+    for _ in range(400 if use_cuda else 5):
         if time.perf_counter() - t_guard >= guard_cap_s:
             break
         run_step()
