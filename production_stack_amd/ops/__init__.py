@@ -164,7 +164,6 @@ def paged_attn_prefill(
 # (csrc/prefill_mfma32.hip); 3/4 = the 4-wave 16x16 kernel (64-row
 # tiles). Overridable for A/B via PS_PREFILL_VARIANT.
 PREFILL_VARIANT = int(os.environ.get("PS_PREFILL_VARIANT", "5"))
-PREFILL_TILE = 256 if PREFILL_VARIANT == 5 else 64
 
 
 def prefill_tile_rows(num_q_heads: int, num_kv_heads: int) -> int:

@@ -310,3 +310,32 @@ def test_multimodal_vision_injection_gpu():
     b = run(png((0, 0, 255)), "b1")
     assert a1 == a2
     assert a1 != b
+
+
+def test_unified_mixed_steps_matches_split_path():
+    """The unified eager mixed step (default) must be token-identical to
+    the split graph+eager path on the same greedy requests."""
+    outs = {}
+    for unified in (True, False):
+        eng = make_engine(unified_mixed_steps=unified)
+        prompts = [list(range(20 + i, 140 + i)) for i in range(6)]
+        for i, p in enumerate(prompts):
+            eng.add_request(f"r{i}", p, SamplingParams(
+                max_tokens=24, temperature=0.0, ignore_eos=True))
+        # stagger two more so steps stay mixed (prefill + decode)
+        got = {f"r{i}": [] for i in range(8)}
+        step = 0
+        while eng.has_unfinished():
+            if step == 2:
+                for i in (6, 7):
+                    eng.add_request(f"r{i}", list(range(40 + i, 200 + i)),
+                                    SamplingParams(max_tokens=24,
+                                                   temperature=0.0,
+                                                   ignore_eos=True))
+            for o in eng.step():
+                got[o.request_id].extend(o.new_token_ids)
+            step += 1
+        outs[unified] = got
+        del eng
+        torch.cuda.empty_cache()
+    assert outs[True] == outs[False]
