@@ -318,6 +318,9 @@ def main() -> None:
         ttft_p50 = (
             float(np.percentile(ttfts, 50) * 1000.0) if ttfts else None
         )
+        ttft_p90 = (
+            float(np.percentile(ttfts, 90) * 1000.0) if ttfts else None
+        )
         print(
             json.dumps(
                 {
@@ -334,6 +337,7 @@ def main() -> None:
                     "dtype": "bfloat16",
                     "data": "synthetic",
                     "ttft_p50_ms": ttft_p50,
+                    "ttft_p90_ms": ttft_p90,
                     "config": {
                         "model": args.model,
                         "workload": "multi-round-qa",
