@@ -381,7 +381,7 @@ int ps_prefill_mfma32_splits(int num_tiles, int KH, int GQ) {
   const int n_work = num_tiles * KH * (GQ / gqw);
   if (n_work >= 500) return 1;
   int s = (500 + n_work - 1) / n_work;
-  return s > 4 ? 4 : s;
+  return s > 8 ? 8 : s;
 }
 
 int ps_paged_attn_prefill_mfma32(void* out, void* ws_o, void* ws_ml,

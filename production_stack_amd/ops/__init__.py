@@ -194,20 +194,13 @@ def paged_attn_prefill_mfma(
     variant."""
     _require_ext()
     out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+    # v5's split-KV (up to 8 kv-splits) keeps CUs filled at small
+    # serving launches, where it now beats v3 at the canonical shapes
+    # (12-seq x 200-row continuation at hist 4600: 305 vs 263 TF; fresh
+    # 900-row prefill: 400 vs 306) as well as at large ones — so v5 is
+    # the default everywhere (PS_PREFILL_VARIANT=3 restores the 4-wave
+    # kernel for A/B).
     v = PREFILL_VARIANT if variant is None else variant
-    if v == 5 and variant is None:
-        # v5 packs GQW q-heads per block: 4x fewer workgroups than v3.
-        # Small serving launches (~10 continuation chunks) then leave CUs
-        # idle (1 block/CU quantization) — measured 210 vs 249 TF at the
-        # 10-seq x 200-row shape. When the tile layout is v3-compatible
-        # (64-row tiles, i.e. GQW=4 models) fall back to v3 below ~500
-        # v5 workgroups.
-        qh_n = q.shape[1]
-        kh_n = k_cache.shape[1]
-        if prefill_tile_rows(qh_n, kh_n) == 64:
-            blocks_v5 = tile_info.shape[0] * kh_n
-            if blocks_v5 < 500:
-                v = 3
     _C.paged_attn_prefill_mfma(
         out, q, k_cache, v_cache, block_tables, tile_info, scale, v, window
     )
