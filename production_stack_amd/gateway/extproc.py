@@ -45,6 +45,13 @@ import grpc
 
 from production_stack_amd.router.hashtrie import HashTrie
 
+try:  # compiled C++ picker core (csrc/gateway_pickers.cpp) — the native
+    # counterpart of the reference's Go plugins; Python trie is the
+    # fallback/test oracle.
+    from production_stack_amd import _gwpick
+except ImportError:  # pragma: no cover - source-only checkout
+    _gwpick = None
+
 logger = logging.getLogger("gateway.extproc")
 
 CHUNK_SIZE = 128
@@ -150,11 +157,17 @@ def build_headers_response(set_headers: Dict[str, str],
 # ---------------------------------------------------------------------------
 class Picker:
     def __init__(self, algorithm: str = "prefixaware",
-                 kv_client=None) -> None:
+                 kv_client=None, use_native: Optional[bool] = None) -> None:
         self.algorithm = algorithm
         self.trie = HashTrie(chunk_size=CHUNK_SIZE)
         self.rr = 0
         self.kv_client = kv_client
+        if use_native is None:
+            use_native = _gwpick is not None
+        self._native = (
+            _gwpick.NativePicker(CHUNK_SIZE, CHUNK_SIZE)
+            if (use_native and _gwpick is not None) else None
+        )
 
     async def _prefix_pick(self, prompt: str,
                            pods: List[str]) -> str:
@@ -175,6 +188,8 @@ class Picker:
         if not pods:
             return ""
         if self.algorithm == "roundrobin" or not prompt:
+            if self._native is not None:
+                return self._native.pick_roundrobin(pods)
             self.rr = (self.rr + 1) % len(pods)
             return pods[self.rr]
         if self.algorithm == "kvaware" and self.kv_client is not None:
@@ -186,6 +201,8 @@ class Picker:
                 logger.exception("kv lookup failed; round robin")
             self.rr = (self.rr + 1) % len(pods)
             return pods[self.rr]
+        if self._native is not None:
+            return self._native.pick_prefixaware(prompt, pods)
         # prefixaware (sync wrapper around the asyncio trie)
         import asyncio
 

@@ -14,6 +14,19 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 
+# CPU-only native gateway pickers (no HIP, no libtorch — the gateway
+# sidecar must not pay a torch import): compiled C++ parity with the
+# reference's Go endpoint-picker plugins.
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
+
+gw_ext = Extension(
+    name="production_stack_amd._gwpick",
+    sources=["csrc/gateway_pickers.cpp"],
+    include_dirs=[pybind11.get_include()],
+    extra_compile_args=["-O3", "-std=c++17", "-fvisibility=hidden"],
+)
+
 ext = CUDAExtension(
     name="production_stack_amd._C",
     sources=[
@@ -61,6 +74,6 @@ setup(
             "production_stack_amd.gateway.picker_service:main",
         ]
     },
-    ext_modules=[ext],
+    ext_modules=[ext, gw_ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
