@@ -776,6 +776,118 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
             "output_text": text,
         })
 
+    @app.post("/v1/messages")
+    async def anthropic_messages(request: Request):
+        """Anthropic-style Messages API mapped onto the chat pipeline
+        (reference proxies /v1/messages verbatim to the engine,
+        main_router.py:51-300; vLLM serves it natively). Supports
+        string/content-block messages, system prompts, stop_sequences
+        and the Anthropic SSE event framing when stream=true."""
+        body = await request.json()
+
+        def _blocks_to_text(content) -> str:
+            if isinstance(content, list):
+                return "".join(
+                    b.get("text", "") for b in content
+                    if isinstance(b, dict) and b.get("type") == "text"
+                )
+            return str(content)
+
+        messages = []
+        if body.get("system"):
+            messages.append(
+                {"role": "system",
+                 "content": _blocks_to_text(body["system"])}
+            )
+        for m in body.get("messages", []):
+            messages.append(
+                {"role": m.get("role", "user"),
+                 "content": _blocks_to_text(m.get("content", ""))}
+            )
+        chat_body = {
+            "max_tokens": body.get("max_tokens", 256),
+            "temperature": body.get("temperature", 1.0),
+            "top_p": body.get("top_p", 1.0),
+            "stop": body.get("stop_sequences") or [],
+        }
+        if body.get("top_k") is not None:
+            chat_body["top_k"] = body["top_k"]
+        prompt_tokens = engine.tokenizer.encode(render_chat(messages))
+        if len(prompt_tokens) + 1 > engine.config.max_model_len:
+            return JSONResponse(
+                status_code=400,
+                content={"type": "error", "error": {
+                    "type": "invalid_request_error",
+                    "message": "prompt exceeds max_model_len"}},
+            )
+        params = _params_from_body(chat_body, engine.config.max_model_len)
+        rid = f"msg-{uuid.uuid4().hex[:12]}"
+        model_name = body.get("model", served_model)
+        stop_map = {"stop": "end_turn", "length": "max_tokens",
+                    "stop_sequence": "stop_sequence"}
+
+        if body.get("stream"):
+
+            async def gen():
+                def ev(name, payload):
+                    return (f"event: {name}\n"
+                            f"data: {json.dumps(payload)}\n\n")
+
+                yield ev("message_start", {
+                    "type": "message_start",
+                    "message": {
+                        "id": rid, "type": "message", "role": "assistant",
+                        "content": [], "model": model_name,
+                        "stop_reason": None, "stop_sequence": None,
+                        "usage": {"input_tokens": len(prompt_tokens),
+                                  "output_tokens": 0},
+                    }})
+                yield ev("content_block_start", {
+                    "type": "content_block_start", "index": 0,
+                    "content_block": {"type": "text", "text": ""}})
+                n_out, reason = 0, "end_turn"
+                async for out in async_engine.generate(
+                    rid, prompt_tokens, params
+                ):
+                    n_out = out.num_output_tokens
+                    if out.finished:
+                        reason = stop_map.get(
+                            out.finish_reason or "stop", "end_turn")
+                    if out.text_delta:
+                        yield ev("content_block_delta", {
+                            "type": "content_block_delta", "index": 0,
+                            "delta": {"type": "text_delta",
+                                      "text": out.text_delta}})
+                yield ev("content_block_stop",
+                         {"type": "content_block_stop", "index": 0})
+                yield ev("message_delta", {
+                    "type": "message_delta",
+                    "delta": {"stop_reason": reason,
+                              "stop_sequence": None},
+                    "usage": {"output_tokens": n_out}})
+                yield ev("message_stop", {"type": "message_stop"})
+
+            return StreamingResponse(gen(), media_type="text/event-stream")
+
+        text, n_out, reason = "", 0, "end_turn"
+        async for out in async_engine.generate(rid, prompt_tokens, params):
+            text += out.text_delta
+            n_out = out.num_output_tokens
+            if out.finished:
+                reason = stop_map.get(out.finish_reason or "stop",
+                                      "end_turn")
+        return JSONResponse({
+            "id": rid,
+            "type": "message",
+            "role": "assistant",
+            "content": [{"type": "text", "text": text}],
+            "model": model_name,
+            "stop_reason": reason,
+            "stop_sequence": None,
+            "usage": {"input_tokens": len(prompt_tokens),
+                      "output_tokens": n_out},
+        })
+
     @app.post("/v1/completions")
     async def completions(request: Request):
         return await _run_completion(request, chat=False)

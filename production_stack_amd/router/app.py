@@ -97,6 +97,9 @@ def build_app() -> FastAPI:
         "/score",
         "/v1/responses",
         "/v1/messages",
+        "/v1/images/generations",
+        "/v1/images/edits",
+        "/v1/images/variations",
     ]
 
     def make_handler(path: str):

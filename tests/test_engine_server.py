@@ -386,3 +386,67 @@ def test_responses_api_and_version():
         assert d["output_text"]
 
     with_server(go)
+
+
+def test_anthropic_messages_non_streaming():
+    async def go(client):
+        r = await client.post(
+            "/v1/messages",
+            json={
+                "model": "tiny-llama",
+                "max_tokens": 5,
+                "system": "be terse",
+                "messages": [
+                    {"role": "user",
+                     "content": [{"type": "text", "text": "hello there"}]},
+                ],
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+            timeout=120,
+        )
+        assert r.status_code == 200, r.text
+        d = r.json()
+        assert d["type"] == "message" and d["role"] == "assistant"
+        assert d["content"][0]["type"] == "text" and d["content"][0]["text"]
+        assert d["stop_reason"] in ("end_turn", "max_tokens")
+        assert d["usage"]["output_tokens"] == 5
+        assert d["usage"]["input_tokens"] > 0
+
+    with_server(go)
+
+
+def test_anthropic_messages_streaming_event_framing():
+    async def go(client):
+        async with client.stream(
+            "POST",
+            "/v1/messages",
+            json={
+                "model": "tiny-llama",
+                "max_tokens": 4,
+                "messages": [{"role": "user", "content": "hi"}],
+                "temperature": 0,
+                "stream": True,
+                "ignore_eos": True,
+            },
+            timeout=120,
+        ) as r:
+            assert r.status_code == 200
+            events = []
+            text = ""
+            async for line in r.aiter_lines():
+                if line.startswith("event: "):
+                    events.append(line[7:])
+                elif line.startswith("data: "):
+                    payload = json.loads(line[6:])
+                    if payload.get("type") == "content_block_delta":
+                        text += payload["delta"]["text"]
+                    if payload.get("type") == "message_delta":
+                        assert payload["usage"]["output_tokens"] == 4
+        assert events[0] == "message_start"
+        assert events[1] == "content_block_start"
+        assert "content_block_delta" in events
+        assert events[-2:] == ["message_delta", "message_stop"]
+        assert text
+
+    with_server(go)
