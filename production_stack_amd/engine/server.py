@@ -412,12 +412,15 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
             or f"cmpl-{uuid.uuid4().hex[:24]}"
         )
         mm_embeds = None
+        tools = body.get("tools") if chat else None
+        if body.get("tool_choice") == "none":
+            tools = None
         if chat:
             messages = body.get("messages") or []
             if _has_media(messages):
                 prompt, mm_embeds = _assemble_multimodal(messages, engine)
             else:
-                prompt = render_chat(messages)
+                prompt = render_chat(messages, tools=tools)
         else:
             p = body.get("prompt", "")
             if isinstance(p, list) and p and isinstance(p[0], int):
@@ -473,11 +476,32 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                 try:
                     first = True
                     n_out = 0
+                    # with tools, buffer the whole turn so tool_calls can
+                    # be parsed and emitted as one structured chunk
+                    buffered = "" if (chat and tools) else None
                     async for out in async_engine.generate(
                         rid, prompt_tokens, params, mm_embeds=mm_embeds
                     ):
                         n_out = out.num_output_tokens
-                        if chat:
+                        if buffered is not None:
+                            buffered += out.text_delta
+                            if not out.finished:
+                                continue
+                            rest, tcs = parse_tool_calls(buffered)
+                            if tcs:
+                                for j, tc in enumerate(tcs):
+                                    tc["index"] = j
+                                delta = {"role": "assistant",
+                                         "content": rest or None,
+                                         "tool_calls": tcs}
+                                fr = "tool_calls"
+                            else:
+                                delta = {"role": "assistant",
+                                         "content": buffered}
+                                fr = out.finish_reason
+                            choice = {"index": 0, "delta": delta,
+                                      "finish_reason": fr}
+                        elif chat:
                             delta = (
                                 {"role": "assistant", "content": out.text_delta}
                                 if first
@@ -555,15 +579,26 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         text, tokens, _, finish_reason, n_prompt = results[0]
         choices = []
         for i, (txt, toks, lps, reason, _) in enumerate(results):
-            choice = {
-                "index": i,
-                "finish_reason": reason,
-                **(
-                    {"message": {"role": "assistant", "content": txt}}
-                    if chat
-                    else {"text": txt}
-                ),
-            }
+            if chat:
+                message = {"role": "assistant", "content": txt}
+                if tools:
+                    rest, tcs = parse_tool_calls(txt)
+                    if tcs:
+                        message = {"role": "assistant",
+                                   "content": rest or None,
+                                   "tool_calls": tcs}
+                        reason = "tool_calls"
+                choice = {
+                    "index": i,
+                    "finish_reason": reason,
+                    "message": message,
+                }
+            else:
+                choice = {
+                    "index": i,
+                    "finish_reason": reason,
+                    "text": txt,
+                }
             if params.logprobs is not None and lps:
                 tok_strs = [engine.tokenizer.decode_token(t) for t in toks]
                 if chat:
@@ -897,6 +932,42 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         return await _run_completion(request, chat=True)
 
     return app
+
+
+_TOOL_CALL_RE = None
+
+
+def parse_tool_calls(text: str):
+    """Extract Hermes-style <tool_call>{json}</tool_call> spans emitted by
+    the model into OpenAI `message.tool_calls`; returns (remaining_text,
+    tool_calls). Engine-side counterpart of the reference stack's vLLM
+    tool-call parsers (reference tutorials/13-tool-enabled-installation.md
+    surface)."""
+    global _TOOL_CALL_RE
+    import re
+
+    if _TOOL_CALL_RE is None:
+        _TOOL_CALL_RE = re.compile(
+            r"<tool_call>\s*(\{.*?\})\s*</tool_call>", re.DOTALL
+        )
+    calls = []
+    for i, m in enumerate(_TOOL_CALL_RE.finditer(text)):
+        try:
+            obj = json.loads(m.group(1))
+        except json.JSONDecodeError:
+            continue
+        if not isinstance(obj, dict) or "name" not in obj:
+            continue
+        calls.append({
+            "id": f"call_{uuid.uuid4().hex[:24]}",
+            "type": "function",
+            "function": {
+                "name": str(obj["name"]),
+                "arguments": json.dumps(obj.get("arguments") or {}),
+            },
+        })
+    remaining = _TOOL_CALL_RE.sub("", text).strip()
+    return remaining, calls
 
 
 def handle_kv_transfer_params(app, kv_params: dict, request_id: str) -> dict:

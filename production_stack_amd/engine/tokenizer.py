@@ -110,15 +110,45 @@ def get_tokenizer(spec: str, vocab_size: int) -> BaseTokenizer:
     return HFTokenizer(spec)
 
 
-def render_chat(messages: List[dict], add_generation_prompt: bool = True) -> str:
-    """Minimal chat template (role-tagged concatenation)."""
+def render_chat(messages: List[dict], add_generation_prompt: bool = True,
+                tools: Optional[List[dict]] = None) -> str:
+    """Minimal chat template (role-tagged concatenation).
+
+    With `tools`, function schemas are injected as a leading system block
+    and the model is instructed to emit Hermes-style
+    ``<tool_call>{"name": ..., "arguments": {...}}</tool_call>`` spans —
+    the format server.py's parse_tool_calls() extracts into OpenAI
+    `message.tool_calls` (the engine-side counterpart of vLLM's tool-call
+    parsers the reference stack relies on; tutorial 13 in the reference).
+    """
+    import json as _json
+
     parts = []
+    if tools:
+        schemas = _json.dumps(
+            [t.get("function", t) for t in tools], separators=(",", ":")
+        )
+        parts.append(
+            "<|system|> You may call these tools: " + schemas +
+            ' To call one, reply with <tool_call>{"name": <name>, '
+            '"arguments": <args-object>}</tool_call>.'
+        )
     for m in messages:
         content = m.get("content") or ""
         if isinstance(content, list):  # OpenAI content-part arrays
             content = " ".join(
                 p.get("text", "") for p in content if isinstance(p, dict)
             )
+        if m.get("tool_calls"):  # assistant turn that called tools
+            calls = "".join(
+                "<tool_call>" + _json.dumps({
+                    "name": tc["function"]["name"],
+                    "arguments": _json.loads(
+                        tc["function"].get("arguments") or "{}"),
+                }, separators=(",", ":")) + "</tool_call>"
+                for tc in m["tool_calls"] if tc.get("function")
+            )
+            content = (content + " " + calls).strip()
         parts.append(f"<|{m.get('role', 'user')}|> {content}")
     if add_generation_prompt:
         parts.append("<|assistant|>")

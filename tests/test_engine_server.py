@@ -450,3 +450,62 @@ def test_anthropic_messages_streaming_event_framing():
         assert text
 
     with_server(go)
+
+
+def test_parse_tool_calls_and_chat_template_tools():
+    from production_stack_amd.engine.server import parse_tool_calls
+    from production_stack_amd.engine.tokenizer import render_chat
+
+    text = ('thinking <tool_call>{"name": "get_weather", '
+            '"arguments": {"city": "SF"}}</tool_call> done '
+            '<tool_call>{"name": "now", "arguments": {}}</tool_call>')
+    rest, calls = parse_tool_calls(text)
+    assert rest == "thinking  done"
+    assert [c["function"]["name"] for c in calls] == ["get_weather", "now"]
+    assert json.loads(calls[0]["function"]["arguments"]) == {"city": "SF"}
+    assert all(c["type"] == "function" and c["id"].startswith("call_")
+               for c in calls)
+    # malformed spans are skipped, plain text untouched
+    rest2, calls2 = parse_tool_calls("<tool_call>not json</tool_call> hi")
+    assert calls2 == [] and "hi" in rest2
+
+    tools = [{"type": "function",
+              "function": {"name": "get_weather",
+                           "parameters": {"type": "object"}}}]
+    prompt = render_chat(
+        [{"role": "user", "content": "weather?"},
+         {"role": "assistant", "tool_calls": [
+             {"id": "call_1", "type": "function",
+              "function": {"name": "get_weather",
+                           "arguments": '{"city": "SF"}'}}]},
+         {"role": "tool", "content": "sunny"}],
+        tools=tools,
+    )
+    assert "You may call these tools" in prompt
+    assert '"get_weather"' in prompt
+    assert "<|tool|> sunny" in prompt
+    assert '<tool_call>{"name":"get_weather"' in prompt
+
+
+def test_chat_completions_accepts_tools():
+    async def go(client):
+        r = await client.post(
+            "/v1/chat/completions",
+            json={
+                "model": "tiny-llama",
+                "messages": [{"role": "user", "content": "hi"}],
+                "tools": [{"type": "function",
+                           "function": {"name": "noop",
+                                        "parameters": {}}}],
+                "max_tokens": 4,
+                "temperature": 0,
+                "ignore_eos": True,
+            },
+            timeout=120,
+        )
+        assert r.status_code == 200, r.text
+        msg = r.json()["choices"][0]["message"]
+        # synthetic weights never emit the tool-call format -> plain text
+        assert msg["role"] == "assistant" and "tool_calls" not in msg
+
+    with_server(go)
