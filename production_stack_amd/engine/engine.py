@@ -64,6 +64,8 @@ class LLMEngine:
         self.scheduler = Scheduler(
             config.scheduler, self.block_manager, config.max_model_len
         )
+        if config.parallel.pipeline_parallel_size > 1:
+            self.scheduler.allow_spec = False
         if config.cache.cpu_offload_gb > 0 or config.cache.remote_kv_url:
             from production_stack_amd.kvpool.offload import HostKVPool
 

@@ -502,9 +502,13 @@ class ModelRunner:
                         )
                     }
                     if logits_p is not None:
-                        result.update(self._collect_sampled(
-                            SchedulerOutput(scheduled=pre), sample_seqs,
+                        out_p = SchedulerOutput(scheduled=pre)
+                        toks_p = self._spec_stochastic_fix(
+                            out_p, sample_seqs, logits_p,
                             self.sample(logits_p, sample_seqs),
+                        )
+                        result.update(self._collect_sampled(
+                            out_p, sample_seqs, toks_p,
                         ))
                     return result
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
@@ -525,7 +529,9 @@ class ModelRunner:
             return {}
         sel = hidden[rows_t]
         logits = self.model.compute_logits(sel)
-        tokens = self.sample(logits, sample_seqs)
+        tokens = self._spec_stochastic_fix(
+            out, sample_seqs, logits, self.sample(logits, sample_seqs)
+        )
         return self._collect_sampled(out, sample_seqs, tokens)
 
     @torch.no_grad()
@@ -841,6 +847,87 @@ class ModelRunner:
                 chosen = picks.flatten()
             result[torch.tensor(rows)] = chosen
         return result
+
+    def _filtered_probs(self, logit_row: torch.Tensor, p) -> torch.Tensor:
+        """CPU fp32 probabilities after exactly the temperature/top-k/top-p
+        filtering sample_params applies — the target distribution used by
+        rejection-sampled speculative decoding."""
+        lf = logit_row.detach().float() / max(p.temperature, 1e-5)
+        v = lf.shape[-1]
+        if 0 < p.top_k < v:
+            kth = torch.topk(lf, p.top_k).values[-1]
+            lf = lf.masked_fill(lf < kth, float("-inf"))
+        probs = torch.softmax(lf, dim=-1)
+        if p.top_p < 1.0:
+            sp, si = torch.sort(probs, descending=True)
+            cum = torch.cumsum(sp, dim=-1)
+            keep = (cum - sp) < p.top_p
+            keep[0] = True
+            mask = torch.zeros_like(probs, dtype=torch.bool)
+            mask.scatter_(0, si[keep], True)
+            probs = probs * mask
+            probs = probs / probs.sum()
+        return probs.cpu()
+
+    def _spec_generator(self, p) -> torch.Generator:
+        if p.seed is None:
+            return self._generator
+        g = getattr(p, "_seed_generator", None)
+        if g is None:
+            g = torch.Generator(device="cpu").manual_seed(int(p.seed))
+            p._seed_generator = g
+        return g
+
+    def _spec_stochastic_fix(self, out, sample_seqs, logits, tokens):
+        """Rejection-sampling acceptance for speculative chunks under
+        stochastic sampling (Leviathan et al. 2023 with a point-mass
+        draft: accept draft d at position j with prob p_j(d); on reject
+        emit a sample from p_j with d's mass removed — either way the
+        emitted token is marginally distributed exactly as p_j, so
+        speculation never changes the output distribution).
+
+        Decisions are written back into the sampled-token rows so
+        _collect_sampled's exact-match walk emits the right accepted
+        prefix: an accepted position holds the draft token, the first
+        rejected position holds the residual resample (always != draft),
+        and a fully-accepted chunk falls through to the bonus row's own
+        target sample."""
+        draft_map = {
+            ss.seq.request_id: ss.draft_tokens
+            for ss in out.scheduled
+            if getattr(ss, "draft_tokens", None)
+        }
+        if not draft_map:
+            return tokens
+        i = 0
+        n = len(sample_seqs)
+        while i < n:
+            seq = sample_seqs[i]
+            drafts = draft_map.get(seq.request_id)
+            if not drafts:
+                i += 1
+                continue
+            p = seq.params
+            if not p.greedy:
+                gen = self._spec_generator(p)
+                for j, d in enumerate(drafts):
+                    probs = self._filtered_probs(logits[i + j], p)
+                    pd = float(probs[d])
+                    if float(torch.rand(1, generator=gen)) < pd:
+                        tokens[i + j] = d
+                        continue
+                    probs[d] = 0.0
+                    s = float(probs.sum())
+                    if s > 0:
+                        tokens[i + j] = int(torch.multinomial(
+                            probs / s, 1, generator=gen
+                        ))
+                    # s == 0: the filtered distribution was a point mass
+                    # on d (pd lost to fp rounding); keep the row's own
+                    # target sample
+                    break
+            i += len(drafts) + 1
+        return tokens
 
     def _redraw_seeded(
         self, probs_cpu: torch.Tensor, picks: torch.Tensor,

@@ -60,6 +60,9 @@ class Scheduler:
         self.config = config
         self.bm = block_manager
         self.max_model_len = max_model_len
+        # engine clears this under PP: the PP drive path maps one sampled
+        # token per request and cannot carry a k+1-row verification chunk
+        self.allow_spec = True
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
         self._by_id: Dict[str, Sequence] = {}
@@ -140,7 +143,17 @@ class Scheduler:
             if (
                 remaining == 1
                 and self.config.num_speculative_tokens > 0
-                and seq.params.greedy
+                and self.allow_spec
+                and (
+                    seq.params.greedy
+                    # stochastic sampling: rejection-sampling acceptance
+                    # (model_runner._spec_stochastic_fix) keeps the
+                    # output distribution exact; penalties/logprobs rows
+                    # stay non-speculative (their per-position state
+                    # would drift across a multi-token chunk)
+                    or (not seq.params.needs_penalties
+                        and seq.params.logprobs is None)
+                )
                 and seq.output_token_ids
                 and seq.output_token_ids[-1] >= 0
             ):
