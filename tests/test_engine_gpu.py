@@ -217,6 +217,20 @@ def test_speculative_ngram_gpu_matches_plain():
     assert got == want, f"{got} != {want}"
     assert spec.runner.spec_proposed > 0
 
+    # stochastic sampling through the same chunks: rejection-sampling
+    # acceptance (near-greedy temperature so drafts fire), seeded
+    # determinism across two engines
+    def fresh():
+        return SamplingParams(max_tokens=20, temperature=0.05, seed=7,
+                              ignore_eos=True)
+
+    got1 = spec.generate([prompt], fresh())["offline-0"]
+    assert len(got1) == 20
+    spec2 = mk(True)
+    spec2.runner.model.load_state_dict(plain.runner.model.state_dict())
+    got2 = spec2.generate([prompt], fresh())["offline-0"]
+    assert got2 == got1, f"{got2} != {got1}"
+
 
 def test_async_scheduling_gpu_matches_sync():
     """Async (one-step-lagged) scheduling on GPU: decode graphs + device
