@@ -54,6 +54,7 @@ class LLMEngine:
             config.tokenizer, self.model_cfg.vocab_size
         )
         self.runner = ModelRunner(config, self.model_cfg, self.device)
+        self.runner.tokenizer = self.tokenizer  # guided decoding masks
         num_blocks = self.runner.profile_num_blocks()
         self.runner.allocate_kv_cache(num_blocks)
         self.block_manager = BlockManager(
@@ -249,6 +250,12 @@ class LLMEngine:
                 reason = "length"
             elif seq.status is SeqStatus.FINISHED_ABORTED:
                 reason = "abort"
+            if reason is not None:
+                gs = getattr(seq.params, "_guided_state", None)
+                if gs is not None:
+                    # surface json_schema validation failure instead of
+                    # silently returning a non-conforming object
+                    reason = gs.finish_reason() or reason
             results.append(
                 RequestOutput(
                     request_id=seq.request_id,

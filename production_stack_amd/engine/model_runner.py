@@ -593,6 +593,8 @@ class ModelRunner:
             p = ss.seq.params
             if not p.greedy or p.logprobs is not None or p.needs_penalties:
                 return None
+            if getattr(p, "response_format", None):
+                return None  # guided masks need last step's token on host
             if ss.draft_tokens:
                 return None
         dec = [ss for ss in out.scheduled if ss.is_decode]
@@ -733,6 +735,8 @@ class ModelRunner:
         params = [s.params for s in seqs]
         if any(p.needs_penalties for p in params):
             logits = self._apply_penalties(logits, seqs)
+        if any(getattr(p, "response_format", None) for p in params):
+            logits = self._apply_guided(logits, seqs)
         sampled = self.sample_params(logits, params)
         want_lp = [i for i, p in enumerate(params)
                    if p.logprobs is not None]
@@ -746,6 +750,42 @@ class ModelRunner:
             for j, i in enumerate(want_lp):
                 self.last_logprobs[seqs[i].request_id] = float(vals[j])
         return sampled
+
+    def _apply_guided(
+        self, logits: torch.Tensor, seqs: List[Sequence]
+    ) -> torch.Tensor:
+        """Structured outputs (`response_format` JSON mode): mask each
+        guided row to the tokens whose text keeps the output a valid JSON
+        prefix; once the document completes only EOS stays legal
+        (engine/guided.py)."""
+        from production_stack_amd.engine.guided import (
+            guided_state_from_response_format,
+        )
+
+        tok = getattr(self, "tokenizer", None)
+        if tok is None:
+            return logits
+        logits = logits.float().clone()
+        eos = self.model_cfg.eos_token_id
+        for i, seq in enumerate(seqs):
+            p = seq.params
+            rf = getattr(p, "response_format", None)
+            if not rf:
+                continue
+            gs = getattr(p, "_guided_state", None)
+            if gs is None:
+                gs = guided_state_from_response_format(rf)
+                if gs is None:
+                    continue
+                p._guided_state = gs
+            gs.advance(tok, seq.output_token_ids)
+            allowed, _ = gs.allowed_mask(tok, logits[i], eos)
+            idx = torch.tensor(allowed, dtype=torch.long,
+                               device=logits.device)
+            row = torch.full_like(logits[i], float("-inf"))
+            row[idx] = logits[i][idx]
+            logits[i] = row
+        return logits
 
     def _apply_penalties(
         self, logits: torch.Tensor, seqs: List[Sequence]

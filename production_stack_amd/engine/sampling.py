@@ -22,6 +22,9 @@ class SamplingParams:
     repetition_penalty: float = 1.0
     logit_bias: Optional[dict] = None  # token_id -> bias
     min_tokens: int = 0
+    # OpenAI structured outputs: {"type": "json_object"} or
+    # {"type": "json_schema", "json_schema": {...}} (engine/guided.py)
+    response_format: Optional[dict] = None
 
     @property
     def needs_penalties(self) -> bool:

@@ -144,6 +144,9 @@ class Scheduler:
                 remaining == 1
                 and self.config.num_speculative_tokens > 0
                 and self.allow_spec
+                # guided rows mask per-position grammar state; a k+1-row
+                # chunk would share one state across positions
+                and getattr(seq.params, "response_format", None) is None
                 and (
                     seq.params.greedy
                     # stochastic sampling: rejection-sampling acceptance

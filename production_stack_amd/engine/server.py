@@ -62,6 +62,13 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
                     for k, v in (body.get("logit_bias") or {}).items()},
         min_tokens=int(body.get("min_tokens") or 0),
         logprobs=_parse_logprobs(body),
+        response_format=(
+            body.get("response_format")
+            if isinstance(body.get("response_format"), dict)
+            and body["response_format"].get("type") in
+            ("json_object", "json_schema")
+            else None
+        ),
     )
 
 

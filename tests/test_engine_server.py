@@ -509,3 +509,27 @@ def test_chat_completions_accepts_tools():
         assert msg["role"] == "assistant" and "tool_calls" not in msg
 
     with_server(go)
+
+
+def test_response_format_json_object_plumbed():
+    """response_format reaches the sampler: with the synthetic vocab no
+    token is JSON-legal, so the guided mask forces immediate EOS and the
+    request finishes cleanly with empty content."""
+    async def go(client):
+        r = await client.post(
+            "/v1/chat/completions",
+            json={
+                "model": "tiny-llama",
+                "messages": [{"role": "user", "content": "json please"}],
+                "response_format": {"type": "json_object"},
+                "max_tokens": 8,
+                "temperature": 0,
+            },
+            timeout=120,
+        )
+        assert r.status_code == 200, r.text
+        c = r.json()["choices"][0]
+        assert c["finish_reason"] == "stop"
+        assert c["message"]["content"].strip() == ""
+
+    with_server(go)
