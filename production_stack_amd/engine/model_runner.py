@@ -862,6 +862,14 @@ class ModelRunner:
                 kth = torch.topk(lf, top_k, dim=-1).values[:, -1:]
                 lf = lf.masked_fill(lf < kth, float("-inf"))
             probs = torch.softmax(lf, dim=-1)
+            min_ps = torch.tensor(
+                [getattr(params[i], "min_p", 0.0) for i in rows],
+                device=lf.device,
+            ).unsqueeze(1)
+            if bool((min_ps > 0).any()):
+                cutoff = probs.max(dim=-1, keepdim=True).values * min_ps
+                probs = probs.masked_fill(probs < cutoff, 0.0)
+                probs = probs / probs.sum(dim=-1, keepdim=True)
             top_ps = torch.tensor(
                 [params[i].top_p for i in rows], device=lf.device
             ).unsqueeze(1)
@@ -898,6 +906,10 @@ class ModelRunner:
             kth = torch.topk(lf, p.top_k).values[-1]
             lf = lf.masked_fill(lf < kth, float("-inf"))
         probs = torch.softmax(lf, dim=-1)
+        if getattr(p, "min_p", 0.0) > 0:
+            cutoff = probs.max() * p.min_p
+            probs = probs.masked_fill(probs < cutoff, 0.0)
+            probs = probs / probs.sum()
         if p.top_p < 1.0:
             sp, si = torch.sort(probs, descending=True)
             cum = torch.cumsum(sp, dim=-1)

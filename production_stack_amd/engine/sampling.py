@@ -22,6 +22,8 @@ class SamplingParams:
     repetition_penalty: float = 1.0
     logit_bias: Optional[dict] = None  # token_id -> bias
     min_tokens: int = 0
+    # vLLM-style min_p: drop tokens with prob < min_p * max_prob
+    min_p: float = 0.0
     # OpenAI structured outputs: {"type": "json_object"} or
     # {"type": "json_schema", "json_schema": {...}} (engine/guided.py)
     response_format: Optional[dict] = None
@@ -47,6 +49,8 @@ class SamplingParams:
             raise ValueError("temperature must be >= 0")
         if not (0 < self.top_p <= 1.0):
             raise ValueError("top_p must be in (0, 1]")
+        if not (0.0 <= self.min_p <= 1.0):
+            raise ValueError("min_p must be in [0, 1]")
         if self.top_k == 0 or self.top_k < -1:
             raise ValueError("top_k must be -1 (off) or >= 1")
         if not (-2.0 <= self.presence_penalty <= 2.0):

@@ -61,6 +61,7 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
         logit_bias={int(k): float(v)
                     for k, v in (body.get("logit_bias") or {}).items()},
         min_tokens=int(body.get("min_tokens") or 0),
+        min_p=float(body.get("min_p") or 0.0),
         logprobs=_parse_logprobs(body),
         response_format=(
             body.get("response_format")
@@ -82,7 +83,29 @@ def _parse_logprobs(body: dict):
     return int(lp)
 
 
-def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
+def build_server(engine: LLMEngine, served_model: str,
+                 chat_template: Optional[str] = None) -> FastAPI:
+    # custom chat template (--chat-template FILE): vLLM-compatible jinja2
+    # rendering with messages/tools/add_generation_prompt; falls back to
+    # the built-in role-tagged template (tokenizer.render_chat)
+    _tpl_cache = {}
+
+    def _render_messages(messages, tools=None):
+        if chat_template:
+            tpl = _tpl_cache.get("t")
+            if tpl is None:
+                import jinja2
+
+                tpl = jinja2.Environment(
+                    trim_blocks=True, lstrip_blocks=True
+                ).from_string(chat_template)
+                _tpl_cache["t"] = tpl
+            return tpl.render(
+                messages=messages, tools=tools,
+                add_generation_prompt=True, bos_token="", eos_token="",
+            )
+        return render_chat(messages, tools=tools)
+
     async_engine = AsyncEngine(engine)
 
     @contextlib.asynccontextmanager
@@ -200,7 +223,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
     async def tokenize(request: Request):
         body = await request.json()
         if "messages" in body:
-            text = render_chat(body["messages"])
+            text = _render_messages(body["messages"])
         else:
             text = str(body.get("prompt", ""))
         tokens = engine.tokenizer.encode(text)
@@ -427,7 +450,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
             if _has_media(messages):
                 prompt, mm_embeds = _assemble_multimodal(messages, engine)
             else:
-                prompt = render_chat(messages, tools=tools)
+                prompt = _render_messages(messages, tools=tools)
         else:
             p = body.get("prompt", "")
             if isinstance(p, list) and p and isinstance(p[0], int):
@@ -456,6 +479,14 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         created = int(time.time())
         model_name = body.get("model", served_model)
         obj = "chat.completion" if chat else "text_completion"
+        # OpenAI classic `echo`: prepend the prompt text to the output
+        # (completions only; prompt logprobs are not echoed)
+        echo_text = ""
+        if not chat and body.get("echo"):
+            echo_text = (
+                prompt if isinstance(prompt, str)
+                else engine.tokenizer.decode(list(prompt_tokens))
+            )
 
         # disaggregated prefill: decode role pulls the prefiller's KV blocks
         # into the local prefix cache before scheduling (parallel/kv_transfer)
@@ -524,7 +555,10 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                         else:
                             choice = {
                                 "index": 0,
-                                "text": out.text_delta,
+                                "text": (
+                                    echo_text + out.text_delta
+                                    if first else out.text_delta
+                                ),
                                 "finish_reason": (
                                     out.finish_reason if out.finished else None
                                 ),
@@ -604,7 +638,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
                 choice = {
                     "index": i,
                     "finish_reason": reason,
-                    "text": txt,
+                    "text": (echo_text + txt) if echo_text else txt,
                 }
             if params.logprobs is not None and lps:
                 tok_strs = [engine.tokenizer.decode_token(t) for t in toks]
@@ -798,7 +832,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         chat_body = dict(body)
         chat_body.pop("input", None)
         chat_body["messages"] = messages
-        prompt_tokens = engine.tokenizer.encode(render_chat(messages))
+        prompt_tokens = engine.tokenizer.encode(_render_messages(messages))
         params = _params_from_body(chat_body, engine.config.max_model_len)
         rid = f"resp-{uuid.uuid4().hex[:12]}"
         text = ""
@@ -854,7 +888,7 @@ def build_server(engine: LLMEngine, served_model: str) -> FastAPI:
         }
         if body.get("top_k") is not None:
             chat_body["top_k"] = body["top_k"]
-        prompt_tokens = engine.tokenizer.encode(render_chat(messages))
+        prompt_tokens = engine.tokenizer.encode(_render_messages(messages))
         if len(prompt_tokens) + 1 > engine.config.max_model_len:
             return JSONResponse(
                 status_code=400,
@@ -1135,7 +1169,11 @@ def main() -> None:
         engine.run_pp_worker()
         return
     served = args.served_model_name or args.model
-    app = build_server(engine, served)
+    tpl_src = None
+    if args.chat_template:
+        with open(args.chat_template) as f:
+            tpl_src = f.read()
+    app = build_server(engine, served, chat_template=tpl_src)
     logging.basicConfig(level=logging.INFO)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 

@@ -533,3 +533,93 @@ def test_response_format_json_object_plumbed():
         assert c["message"]["content"].strip() == ""
 
     with_server(go)
+
+
+def test_min_p_masks_to_argmax_when_one():
+    """min_p=1.0 keeps only the argmax token: sampling at temperature 1
+    becomes deterministic greedy."""
+    import torch
+
+    from production_stack_amd.engine.sampling import SamplingParams
+
+    app = make_app()
+    runner = app.state.engine.runner
+    logits = torch.randn(4, 64)
+    p = SamplingParams(max_tokens=4, temperature=1.0, min_p=1.0)
+    got = runner.sample_params(logits, [p] * 4)
+    assert got.tolist() == logits.argmax(dim=-1).tolist()
+    # and validation bounds it
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        SamplingParams(min_p=1.5).validate(128)
+
+
+def test_completions_echo():
+    async def go(client):
+        body = {
+            "model": "tiny-llama",
+            "prompt": "w5 w6 w7",
+            "max_tokens": 3,
+            "temperature": 0,
+            "echo": True,
+            "ignore_eos": True,
+        }
+        r = await client.post("/v1/completions", json=body, timeout=120)
+        assert r.status_code == 200, r.text
+        text = r.json()["choices"][0]["text"]
+        assert text.startswith("w5 w6 w7")
+        assert len(text) > len("w5 w6 w7")
+
+        async with client.stream(
+            "POST", "/v1/completions", json={**body, "stream": True},
+            timeout=120,
+        ) as rs:
+            assert rs.status_code == 200
+            first_text = None
+            async for line in rs.aiter_lines():
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    first_text = json.loads(line[6:])["choices"][0]["text"]
+                    break
+            assert first_text is not None
+            assert first_text.startswith("w5 w6 w7")
+
+    with_server(go)
+
+
+def test_custom_chat_template():
+    """--chat-template: jinja2 rendering replaces the built-in role-tag
+    template; the rendered prompt drives tokenization (prompt_tokens
+    reflects the template's output)."""
+    cfg = EngineConfig(
+        model="tiny-llama",
+        max_model_len=512,
+        cache=CacheConfig(num_gpu_blocks=64, block_size=16),
+        scheduler=SchedulerConfig(max_num_seqs=4,
+                                  max_num_batched_tokens=256),
+    )
+    engine = LLMEngine(cfg, device="cpu")
+    tpl = ("{% for m in messages %}w1 {{ m.content }} "
+           "{% endfor %}w2 w2 w2 w2")
+    app = build_server(engine, served_model="tiny-llama",
+                       chat_template=tpl)
+
+    async def go():
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://engine"
+                ) as client:
+                    r = await client.post(
+                        "/v1/chat/completions",
+                        json={"model": "tiny-llama",
+                              "messages": [
+                                  {"role": "user", "content": "w9 w9"}],
+                              "max_tokens": 2, "temperature": 0,
+                              "ignore_eos": True},
+                        timeout=120,
+                    )
+                    assert r.status_code == 200, r.text
+                    # rendered: "w1 w9 w9 w2 w2 w2 w2" -> 7 tokens
+                    assert r.json()["usage"]["prompt_tokens"] == 7
+
+    asyncio.run(go())
