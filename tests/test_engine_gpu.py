@@ -353,3 +353,35 @@ def test_unified_mixed_steps_matches_split_path():
         del eng
         torch.cuda.empty_cache()
     assert outs[True] == outs[False]
+
+
+def test_guided_json_gpu():
+    """Guided JSON masking on a cuda engine: every emitted token keeps
+    the output a valid JSON prefix (mask indexes device logits)."""
+    import json as _json
+
+    from production_stack_amd.engine.guided import JsonPrefixValidator
+    from test_guided import JsonToyTokenizer
+
+    eng = make_engine()
+    tok = JsonToyTokenizer(eng.model_cfg.vocab_size)
+    eng.tokenizer = tok
+    eng.runner.tokenizer = tok
+    p = SamplingParams(max_tokens=32, temperature=1.0, seed=9,
+                       response_format={"type": "json_object"})
+    eng.add_request("gj", [33, 34, 35], p)
+    toks, reason = [], None
+    for _ in range(60):
+        for out in eng.step():
+            if out.request_id == "gj":
+                toks.extend(out.new_token_ids)
+                if out.finished:
+                    reason = out.finish_reason
+        if reason:
+            break
+    text = "".join(tok.decode_token(t) for t in toks
+                   if t != tok.eos_token_id)
+    assert JsonPrefixValidator().feed_text(text), text
+    assert reason in ("stop", "length")
+    if reason == "stop":
+        _json.loads(text)
