@@ -339,6 +339,10 @@ class EngineConfig:
     # weight quantization: None (bf16) or "fp8" (OCP e4m3 weights +
     # dynamic per-tensor activation quant through the fp8 MFMA pipe)
     quantization: Optional[str] = None
+    # draft-model speculative decoding (engine/draft.py): architecture
+    # name of the (smaller, same-vocab) proposer; None = n-gram drafts
+    speculative_model: Optional[str] = None
+    speculative_weights_path: Optional[str] = None
     enable_lora: bool = False
     max_loras: int = 4
     max_lora_rank: int = 16

@@ -67,6 +67,19 @@ class LLMEngine:
         )
         if config.parallel.pipeline_parallel_size > 1:
             self.scheduler.allow_spec = False
+        if (
+            config.speculative_model
+            and config.scheduler.num_speculative_tokens > 0
+            and config.parallel.pipeline_parallel_size == 1
+            and config.parallel.tensor_parallel_size == 1
+        ):
+            from production_stack_amd.engine.draft import (
+                DraftModelProposer,
+            )
+
+            self.scheduler.draft_proposer = DraftModelProposer(
+                config, self.device, num_blocks
+            )
         if config.cache.cpu_offload_gb > 0 or config.cache.remote_kv_url:
             from production_stack_amd.kvpool.offload import HostKVPool
 

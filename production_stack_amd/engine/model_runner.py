@@ -945,7 +945,8 @@ class ModelRunner:
         and a fully-accepted chunk falls through to the bonus row's own
         target sample."""
         draft_map = {
-            ss.seq.request_id: ss.draft_tokens
+            ss.seq.request_id: (ss.draft_tokens,
+                                getattr(ss, "draft_probs", None))
             for ss in out.scheduled
             if getattr(ss, "draft_tokens", None)
         }
@@ -955,7 +956,7 @@ class ModelRunner:
         n = len(sample_seqs)
         while i < n:
             seq = sample_seqs[i]
-            drafts = draft_map.get(seq.request_id)
+            drafts, qs = draft_map.get(seq.request_id, (None, None))
             if not drafts:
                 i += 1
                 continue
@@ -965,18 +966,26 @@ class ModelRunner:
                 for j, d in enumerate(drafts):
                     probs = self._filtered_probs(logits[i + j], p)
                     pd = float(probs[d])
-                    if float(torch.rand(1, generator=gen)) < pd:
+                    q = qs[j] if qs is not None else None
+                    if q is None:
+                        accept_p = pd  # point-mass draft (n-gram)
+                    else:
+                        accept_p = min(1.0, pd / max(float(q[d]), 1e-20))
+                    if float(torch.rand(1, generator=gen)) < accept_p:
                         tokens[i + j] = d
                         continue
-                    probs[d] = 0.0
+                    if q is None:
+                        probs[d] = 0.0
+                    else:
+                        # Leviathan residual: normalize(max(p - q, 0))
+                        probs = torch.clamp(probs - q, min=0.0)
                     s = float(probs.sum())
                     if s > 0:
                         tokens[i + j] = int(torch.multinomial(
                             probs / s, 1, generator=gen
                         ))
-                    # s == 0: the filtered distribution was a point mass
-                    # on d (pd lost to fp rounding); keep the row's own
-                    # target sample
+                    # s == 0: degenerate overlap (p ~= q point mass);
+                    # keep the row's own target sample
                     break
             i += len(drafts) + 1
         return tokens

@@ -28,6 +28,9 @@ class ScheduledSeq:
     # speculative drafts verified within this chunk (chunk = 1 real token
     # + the drafts); empty for normal chunks
     draft_tokens: List[int] = field(default_factory=list)
+    # draft-model proposal distributions per draft position (None for
+    # greedy/n-gram point-mass drafts); used by rejection sampling
+    draft_probs: Optional[list] = None
 
     @property
     def is_decode(self) -> bool:
@@ -63,6 +66,8 @@ class Scheduler:
         # engine clears this under PP: the PP drive path maps one sampled
         # token per request and cannot carry a k+1-row verification chunk
         self.allow_spec = True
+        # set by the engine when --speculative-model is configured
+        self.draft_proposer = None
         self.waiting: Deque[Sequence] = deque()
         self.running: List[Sequence] = []
         self._by_id: Dict[str, Sequence] = {}
@@ -140,6 +145,7 @@ class Scheduler:
                 continue
             chunk = min(remaining, budget, self.config.max_prefill_chunk)
             drafts: List[int] = []
+            dprobs = None
             if (
                 remaining == 1
                 and self.config.num_speculative_tokens > 0
@@ -160,14 +166,16 @@ class Scheduler:
                 and seq.output_token_ids
                 and seq.output_token_ids[-1] >= 0
             ):
-                drafts = self._propose_drafts(seq)
+                drafts, dprobs = self._propose_drafts(seq)
                 if drafts and not self.bm.ensure_capacity(
                     seq, seq.num_computed + 1 + len(drafts)
                 ):
-                    drafts = []  # fall back to plain decode
+                    drafts, dprobs = [], None  # fall back to plain decode
                 chunk = min(1 + len(drafts), budget)
                 if chunk <= len(drafts):
                     drafts = drafts[: max(chunk - 1, 0)]
+                    if dprobs is not None:
+                        dprobs = dprobs[: len(drafts)]
                     chunk = 1 + len(drafts)
             target = seq.num_computed + chunk
             while not self.bm.ensure_capacity(seq, target):
@@ -190,7 +198,8 @@ class Scheduler:
                     break
             if chunk > 0:
                 out.scheduled.append(
-                    ScheduledSeq(seq, chunk, draft_tokens=drafts)
+                    ScheduledSeq(seq, chunk, draft_tokens=drafts,
+                                 draft_probs=dprobs)
                 )
                 budget -= chunk
 
@@ -221,16 +230,20 @@ class Scheduler:
         return out
 
     # ------------------------------------------------------------------
-    def _propose_drafts(self, seq: Sequence) -> List[int]:
-        """Prompt-lookup drafts: the longest trailing n-gram that recurs in
-        the sequence's own history proposes the tokens that followed it.
-        Byte-encoded rfind so the scan is C-speed."""
+    def _propose_drafts(self, seq: Sequence):
+        """Returns (draft_tokens, draft_probs|None). With a draft model
+        configured, the model proposes (engine/draft.py); otherwise
+        prompt-lookup: the longest trailing n-gram that recurs in the
+        sequence's own history proposes the tokens that followed it
+        (byte-encoded rfind so the scan is C-speed)."""
+        if self.draft_proposer is not None:
+            return self.draft_proposer.propose(seq, self.bm)
         import numpy as np
 
         k = self.config.num_speculative_tokens
         k = min(k, self.max_model_len - seq.num_tokens - 1)
         if k <= 0:
-            return []
+            return [], None
         ids = seq.token_ids()
         win = ids[-1024:]
         arr = np.asarray(win, dtype=np.int32).tobytes()
@@ -248,8 +261,8 @@ class Scheduler:
             start = idx // 4 + n
             drafts = win[start : start + k]
             if drafts:
-                return list(drafts)
-        return []
+                return list(drafts), None
+        return [], None
 
     def on_step_done(
         self,

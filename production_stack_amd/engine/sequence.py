@@ -52,6 +52,7 @@ class Sequence:
         self.arrival_time = arrival_time or time.time()
         self.first_token_time: Optional[float] = None
         self.num_computed = 0  # tokens with KV in cache and attended
+        self._draft_progress = 0  # draft-model KV progress (engine/draft.py)
         self.block_table: List[int] = []
         self.num_cached_prompt_tokens = 0  # prefix-cache hits at admission
         # streaming cursor: outputs not yet handed to the consumer
@@ -96,6 +97,9 @@ class Sequence:
         """Preemption: KV is dropped; everything recomputes on readmission."""
         self.num_computed = 0
         self.block_table = []
+        # draft-model speculation: the draft KV was written under the old
+        # block table — force a full draft re-prefill (engine/draft.py)
+        self._draft_progress = 0
         self.status = SeqStatus.PREEMPTED
 
 

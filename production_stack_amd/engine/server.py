@@ -1087,6 +1087,10 @@ def main() -> None:
                     help="one-step-lagged sampling (greedy-exact overlap)")
     ap.add_argument("--num-speculative-tokens", type=int, default=0,
                     help="n-gram (prompt-lookup) speculative decoding")
+    ap.add_argument("--speculative-model", default=None,
+                    help="draft model architecture for model-based "
+                         "speculation (same vocab; engine/draft.py)")
+    ap.add_argument("--speculative-weights-path", default=None)
     ap.add_argument("--enable-lora", action="store_true",
                     help="enable graph-safe BGMV adapter slots")
     ap.add_argument("--max-loras", type=int, default=4)
@@ -1137,6 +1141,8 @@ def main() -> None:
         ),
         async_scheduling=args.async_scheduling,
         unified_mixed_steps=args.unified_mixed_steps,
+        speculative_model=args.speculative_model,
+        speculative_weights_path=args.speculative_weights_path,
         quantization=args.quantization,
         enable_lora=args.enable_lora,
         max_loras=args.max_loras,
