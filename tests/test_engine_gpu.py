@@ -385,3 +385,29 @@ def test_guided_json_gpu():
     assert reason in ("stop", "length")
     if reason == "stop":
         _json.loads(text)
+
+
+def test_draft_model_spec_gpu_lossless():
+    """Draft-model speculation on GPU: identical-draft greedy output
+    matches the plain engine through the MFMA verification chunks."""
+    plain = make_engine()
+    p = SamplingParams(max_tokens=20, temperature=0.0, ignore_eos=True)
+    prompt = [11, 12, 13, 14, 15] * 10
+    want = plain.generate([prompt], p)["offline-0"]
+
+    from production_stack_amd.engine.config import SchedulerConfig as SC
+
+    cfg = EngineConfig(
+        model="mini-llama",
+        max_model_len=1024,
+        speculative_model="mini-llama",
+        cache=CacheConfig(num_gpu_blocks=256, block_size=16),
+        scheduler=SC(max_num_seqs=8, max_num_batched_tokens=2048,
+                     num_speculative_tokens=4),
+    )
+    spec = LLMEngine(cfg, device="cuda")
+    spec.runner.model.load_state_dict(plain.runner.model.state_dict())
+    spec.scheduler.draft_proposer.load_target_weights(spec.runner.model)
+    got = spec.generate([prompt], p)["offline-0"]
+    assert got == want, f"{got} != {want}"
+    assert spec.runner.spec_accepted == spec.runner.spec_proposed > 0
