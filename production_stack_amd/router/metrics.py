@@ -123,8 +123,21 @@ semantic_cache_hit_rate = Gauge(
 )
 
 
+def _clear_label_gauges() -> None:
+    """Drop every per-label child so endpoints removed from service
+    discovery stop being reported (stale-metric fix; reference
+    test_stale_metrics.py behavior: labels vanish after clearing and
+    only re-populated endpoints reappear)."""
+    for g in (healthy_pods_total, num_requests_running,
+              num_requests_waiting, gpu_cache_usage, current_qps,
+              avg_ttft, avg_latency, in_prefill_requests,
+              in_decoding_requests, finished_requests):
+        g.clear()
+
+
 def fill_and_render() -> bytes:
     """Fill the gauges from the stats singletons and render the exposition."""
+    _clear_label_gauges()
     from production_stack_amd.router.service_discovery import (
         get_service_discovery,
     )

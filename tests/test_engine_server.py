@@ -623,3 +623,40 @@ def test_custom_chat_template():
                     assert r.json()["usage"]["prompt_tokens"] == 7
 
     asyncio.run(go())
+
+
+def test_audio_speech_returns_valid_wav():
+    """TTS endpoint: valid 16 kHz mono WAV, duration proportional to the
+    input length, deterministic for the same input."""
+    import io
+    import wave
+
+    async def go(client):
+        async def synth(text):
+            r = await client.post(
+                "/v1/audio/speech",
+                json={"model": "tiny-llama", "input": text,
+                      "voice": "alloy"},
+                timeout=60,
+            )
+            assert r.status_code == 200
+            assert r.headers["content-type"].startswith("audio/wav")
+            with wave.open(io.BytesIO(r.content)) as w:
+                assert w.getframerate() == 16000
+                assert w.getnchannels() == 1
+                return w.getnframes()
+
+        short = await synth("w1 w2")
+        long = await synth("w1 w2 w3 w4 w5 w6")
+        assert long > short > 0
+        assert await synth("w1 w2") == short
+
+        r = await client.post(
+            "/v1/audio/speech",
+            json={"model": "tiny-llama", "input": "x",
+                  "response_format": "mp3"},
+            timeout=60,
+        )
+        assert r.status_code == 400
+
+    with_server(go)

