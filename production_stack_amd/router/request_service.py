@@ -124,13 +124,30 @@ async def route_general_request(
     in_router_time = time.time()
     request_id = request.headers.get("x-request-id") or str(uuid.uuid4())
     body = await request.body()
-    try:
-        request_json = json.loads(body) if body else {}
-    except json.JSONDecodeError:
-        return JSONResponse(
-            status_code=400,
-            content={"error": "invalid JSON body"},
-        )
+    content_type = request.headers.get("content-type", "")
+    if content_type.startswith("multipart/form-data"):
+        # audio/file endpoints: body passes through verbatim (the
+        # boundary lives in the forwarded content-type header); routing
+        # fields come from the form (reference
+        # proxy_multipart_request, request.py:1230-1420)
+        from production_stack_amd.router.app import parse_multipart
+
+        parts = parse_multipart(body, content_type)
+        is_multipart = True
+        request_json = {
+            name: payload[1].decode("utf-8", "replace")
+            for name, payload in parts.items()
+            if payload[0] is None  # plain form fields only
+        }
+    else:
+        is_multipart = False
+        try:
+            request_json = json.loads(body) if body else {}
+        except json.JSONDecodeError:
+            return JSONResponse(
+                status_code=400,
+                content={"error": "invalid JSON body"},
+            )
     requested_model = request_json.get("model")
     try:
         request.state.model_name = requested_model
@@ -142,7 +159,7 @@ async def route_general_request(
 
     # PII scan (feature-gated)
     pii = getattr(app.state, "pii_analyzer", None)
-    if pii is not None and request_json:
+    if pii is not None and request_json and not is_multipart:
         from production_stack_amd.router.pii import scan_request_body
 
         allowed, request_json, matches = scan_request_body(
@@ -194,13 +211,13 @@ async def route_general_request(
 
     # callbacks / rewrites
     callbacks = getattr(app.state, "callbacks", None)
-    if callbacks and hasattr(callbacks, "pre_request"):
+    if callbacks and hasattr(callbacks, "pre_request") and not is_multipart:
         maybe = callbacks.pre_request(request, request_json, requested_model)
         if maybe is not None:
             request_json = maybe
             body = json.dumps(request_json).encode()
     rewriter = getattr(app.state, "request_rewriter", None)
-    if rewriter is not None:
+    if rewriter is not None and not is_multipart:
         new_body = rewriter.rewrite(endpoint, request_json)
         if new_body is not None:
             request_json = new_body

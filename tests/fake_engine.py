@@ -33,6 +33,19 @@ def build_fake_engine(
     }
     app.state.seen = state
 
+    @app.post("/v1/audio/transcriptions")
+    async def transcriptions(request: Request):
+        body = await request.body()
+        ct = request.headers.get("content-type", "")
+        from production_stack_amd.router.app import parse_multipart
+
+        parts = parse_multipart(body, ct)
+        state["requests"].append({"endpoint": "/v1/audio/transcriptions",
+                                  "content_type": ct,
+                                  "fields": sorted(parts)})
+        fn = parts.get("file", ("?", b""))[0]
+        return {"text": f"transcribed:{fn}", "task": "transcribe"}
+
     @app.get("/v1/models")
     async def models():
         return {"object": "list", "data": [{"id": model, "object": "model"}]}

@@ -374,3 +374,48 @@ def test_active_health_check_hashes_out_dead_backend(engines):
     assert [e.url for e in eps] == [engines[0].url]
     assert sd.get_unhealthy_endpoint_hashes() == ["http://127.0.0.1:1:m1"]
     sd.close()
+
+
+def test_multipart_transcription_proxied_verbatim(engines):
+    """A real multipart/form-data transcription request must pass
+    through the router byte-for-byte (boundary preserved in the
+    forwarded content-type; no JSON-parse 400), with routing driven by
+    the form's model field."""
+    app = make_app(engines)
+
+    async def go():
+        boundary = "testbnd123"
+        body = (
+            f"--{boundary}\r\n"
+            'Content-Disposition: form-data; name="model"\r\n\r\n'
+            "m1\r\n"
+            f"--{boundary}\r\n"
+            'Content-Disposition: form-data; name="file"; '
+            'filename="clip.wav"\r\n'
+            "Content-Type: audio/wav\r\n\r\n"
+            "RIFFxxxxWAVE\r\n"
+            f"--{boundary}--\r\n"
+        ).encode()
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://router"
+                ) as client:
+                    r = await client.post(
+                        "/v1/audio/transcriptions",
+                        content=body,
+                        headers={"content-type":
+                                 f"multipart/form-data; boundary={boundary}"},
+                        timeout=30,
+                    )
+        assert r.status_code == 200, r.text
+        assert r.json()["text"] == "transcribed:clip.wav"
+        seen = [
+            req for s in engines for req in s.seen["requests"]
+            if isinstance(req, dict)
+            and req.get("endpoint") == "/v1/audio/transcriptions"
+        ]
+        assert seen and "testbnd123" in seen[0]["content_type"]
+        assert seen[0]["fields"] == ["file", "model"]
+
+    asyncio.run(go())
