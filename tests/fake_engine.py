@@ -15,7 +15,7 @@ import uuid
 from typing import Optional
 
 import uvicorn
-from fastapi import FastAPI, Request
+from fastapi import FastAPI, Request, Response
 from fastapi.responses import JSONResponse, PlainTextResponse, StreamingResponse
 
 
@@ -45,6 +45,14 @@ def build_fake_engine(
                                   "fields": sorted(parts)})
         fn = parts.get("file", ("?", b""))[0]
         return {"text": f"transcribed:{fn}", "task": "transcribe"}
+
+    @app.post("/v1/audio/speech")
+    async def speech(request: Request):
+        body = await request.json()
+        state["requests"].append({"endpoint": "/v1/audio/speech",
+                                  "input": body.get("input")})
+        return Response(content=b"RIFF\x00\x01\x02WAVEbinary",
+                        media_type="audio/wav")
 
     @app.get("/v1/models")
     async def models():

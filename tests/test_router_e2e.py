@@ -419,3 +419,28 @@ def test_multipart_transcription_proxied_verbatim(engines):
         assert seen[0]["fields"] == ["file", "model"]
 
     asyncio.run(go())
+
+
+def test_speech_binary_response_content_type_preserved(engines):
+    """Binary audio responses stream through the router with the
+    upstream content-type intact (reference
+    test_audio_speech_routing.py behaviors)."""
+    app = make_app(engines)
+
+    async def go():
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://router"
+                ) as client:
+                    r = await client.post(
+                        "/v1/audio/speech",
+                        json={"model": "m1", "input": "hello",
+                              "voice": "alloy"},
+                        timeout=30,
+                    )
+        assert r.status_code == 200
+        assert r.headers["content-type"].startswith("audio/wav")
+        assert r.content.startswith(b"RIFF")
+
+    asyncio.run(go())
