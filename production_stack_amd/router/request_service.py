@@ -223,6 +223,14 @@ async def route_general_request(
             request_json = new_body
             body = json.dumps(request_json).encode()
 
+    # alias resolution rewrites the body's model so engines that
+    # validate model names accept it (reference
+    # utils.replace_model_in_request_body + content-length update)
+    if (aliases and requested_model in aliases and not is_multipart
+            and request_json):
+        request_json["model"] = aliases[requested_model]
+        body = json.dumps(request_json).encode()
+
     endpoints = get_service_discovery().get_endpoint_info()
     candidates = filter_endpoints(endpoints, requested_model, aliases)
     if not candidates:

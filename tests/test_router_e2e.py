@@ -444,3 +444,35 @@ def test_speech_binary_response_content_type_preserved(engines):
         assert r.content.startswith(b"RIFF")
 
     asyncio.run(go())
+
+
+def test_alias_rewrites_model_in_forwarded_body(engines):
+    """Requests for an alias route to the target model's endpoints AND
+    the forwarded body carries the resolved model name (reference
+    utils.replace_model_in_request_body behavior)."""
+    app = make_app(engines, extra_args=["--static-aliases", "ali:m1"])
+
+    async def go():
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://router"
+                ) as client:
+                    r = await client.post(
+                        "/v1/chat/completions",
+                        json={"model": "ali",
+                              "messages": [{"role": "user",
+                                            "content": "hi"}],
+                              "max_tokens": 2},
+                        timeout=30,
+                    )
+        assert r.status_code == 200, r.text
+        bodies = [
+            req["body"] for s in engines for req in s.seen["requests"]
+            if isinstance(req, dict)
+            and isinstance(req.get("body"), dict)
+            and req["body"].get("model") == "m1"
+        ]
+        assert bodies, "no backend saw the resolved model name"
+
+    asyncio.run(go())
