@@ -138,21 +138,30 @@ def build_app() -> FastAPI:
     # ---- model aggregation --------------------------------------------
     @app.get("/v1/models")
     async def list_models():
+        # full backend card payloads pass through (extra fields like
+        # max_model_len preserved — reference test_main_router_models)
         cards = {}
         for ep in get_service_discovery().get_endpoint_info():
             for name in ep.model_names:
-                if name not in cards:
-                    cards[name] = ModelCard(id=name)
+                if name in cards:
+                    continue
+                mi = ep.model_info.get(name)
+                cards[name] = (
+                    mi.to_dict() if mi is not None
+                    else ModelCard(id=name).model_dump()
+                )
         aliases = getattr(app.state, "model_aliases", None) or {}
         for alias, target in aliases.items():
             if alias not in cards:
-                cards[alias] = ModelCard(id=alias, parent=target)
+                cards[alias] = ModelCard(
+                    id=alias, parent=target).model_dump()
         ext = getattr(app.state, "external_providers", None)
         if ext is not None:
             for name in ext.model_names():
                 if name not in cards:
-                    cards[name] = ModelCard(id=name, owned_by="external")
-        return ModelList(data=list(cards.values())).model_dump()
+                    cards[name] = ModelCard(
+                        id=name, owned_by="external").model_dump()
+        return {"object": "list", "data": list(cards.values())}
 
     # ---- ops endpoints -------------------------------------------------
     @app.get("/health")

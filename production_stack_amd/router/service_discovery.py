@@ -30,6 +30,12 @@ class ModelInfo:
     owned_by: Optional[str] = None
     parent: Optional[str] = None
     is_adapter: bool = False
+    # backend fields outside the OpenAI card schema (max_model_len,
+    # permissions, ...) survive the round trip — reference
+    # ModelInfo preserves extras (test_main_router_models.py)
+    extra: Dict[str, Any] = field(default_factory=dict)
+
+    _KNOWN = ("id", "object", "created", "owned_by", "parent")
 
     @staticmethod
     def from_dict(d: Dict[str, Any]) -> "ModelInfo":
@@ -40,7 +46,20 @@ class ModelInfo:
             owned_by=d.get("owned_by"),
             parent=d.get("parent"),
             is_adapter=d.get("parent") is not None,
+            extra={k: v for k, v in d.items()
+                   if k not in ModelInfo._KNOWN},
         )
+
+    def to_dict(self) -> Dict[str, Any]:
+        out = {"id": self.id, "object": self.object}
+        if self.created is not None:
+            out["created"] = self.created
+        if self.owned_by is not None:
+            out["owned_by"] = self.owned_by
+        if self.parent is not None:
+            out["parent"] = self.parent
+        out.update(self.extra)
+        return out
 
 
 @dataclass

@@ -336,3 +336,19 @@ def test_metrics_stale_labels_cleared():
     out = rm.fill_and_render().decode()
     # fill_and_render clears first and repopulates only live endpoints
     assert "gone:8000" not in out
+
+
+def test_model_info_preserves_extra_fields():
+    """Backend model-card fields outside the OpenAI schema survive the
+    discovery round trip (reference test_main_router_models.py)."""
+    from production_stack_amd.router.service_discovery import ModelInfo
+
+    d = {"id": "m", "object": "model", "created": 5, "owned_by": "org",
+         "max_model_len": 8192, "permission": [{"id": "p1"}]}
+    mi = ModelInfo.from_dict(d)
+    assert mi.extra == {"max_model_len": 8192,
+                        "permission": [{"id": "p1"}]}
+    back = mi.to_dict()
+    assert back["max_model_len"] == 8192
+    assert back["permission"] == [{"id": "p1"}]
+    assert back["id"] == "m" and back["created"] == 5
