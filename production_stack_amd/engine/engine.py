@@ -244,8 +244,13 @@ class LLMEngine:
             lps = None
             if seq.params.logprobs is not None and new:
                 lp = self.runner.last_logprobs.pop(seq.request_id, None)
-                # one token sampled per step -> at most one value per drain
-                lps = [lp] * len(new) if lp is not None else None
+                if isinstance(lp, list):
+                    # speculative chunk: one value per accepted token
+                    lps = lp[-len(new):] if len(lp) >= len(new) else (
+                        [lp[0]] * (len(new) - len(lp)) + lp)
+                elif lp is not None:
+                    # one token sampled per step
+                    lps = [lp] * len(new)
             first = seq.first_token_time is None and bool(new)
             if first:
                 seq.first_token_time = now

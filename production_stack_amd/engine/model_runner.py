@@ -508,7 +508,7 @@ class ModelRunner:
                             self.sample(logits_p, sample_seqs),
                         )
                         result.update(self._collect_sampled(
-                            out_p, sample_seqs, toks_p,
+                            out_p, sample_seqs, toks_p, logits_p,
                         ))
                     return result
         token_t, meta, sample_seqs, rows_t = self.prepare(out, bm)
@@ -532,7 +532,7 @@ class ModelRunner:
         tokens = self._spec_stochastic_fix(
             out, sample_seqs, logits, self.sample(logits, sample_seqs)
         )
-        return self._collect_sampled(out, sample_seqs, tokens)
+        return self._collect_sampled(out, sample_seqs, tokens, logits)
 
     @torch.no_grad()
     def _execute_pp_microbatched(
@@ -696,7 +696,8 @@ class ModelRunner:
             for seq, t in zip(handle["sample_seqs"], vals)
         }
 
-    def _collect_sampled(self, out, sample_seqs, tokens) -> Dict[str, object]:
+    def _collect_sampled(self, out, sample_seqs, tokens,
+                         logits=None) -> Dict[str, object]:
         """Map sampled rows back to requests; speculative chunks contribute
         their accepted-prefix token list (draft j is accepted when it
         equals the model's own prediction at the previous position)."""
@@ -725,6 +726,16 @@ class ModelRunner:
                 accepted.append(row_toks[j + 1])
             self.spec_proposed += len(drafts)
             self.spec_accepted += len(accepted) - 1
+            if seq.params.logprobs is not None and logits is not None:
+                # per-accepted-token logprobs (row j's raw log-softmax at
+                # the token emitted there) — spec chunks emit several
+                # tokens per step, so last_logprobs carries a list
+                rows = logits[i:i + len(accepted)].float()
+                lg = torch.log_softmax(rows, dim=-1)
+                idx = torch.tensor(accepted, dtype=torch.long,
+                                   device=lg.device)
+                vals = lg.gather(1, idx.unsqueeze(1)).squeeze(1)
+                self.last_logprobs[rid] = [float(v) for v in vals.cpu()]
             result[rid] = accepted
             i += k1
         return result

@@ -157,11 +157,12 @@ class Scheduler:
                     seq.params.greedy
                     # stochastic sampling: rejection-sampling acceptance
                     # (model_runner._spec_stochastic_fix) keeps the
-                    # output distribution exact; penalties/logprobs rows
-                    # stay non-speculative (their per-position state
-                    # would drift across a multi-token chunk)
-                    or (not seq.params.needs_penalties
-                        and seq.params.logprobs is None)
+                    # output distribution exact; penalty rows stay
+                    # non-speculative (their per-position state would
+                    # drift across a multi-token chunk). Logprob rows
+                    # speculate: _collect_sampled records one value per
+                    # accepted token.
+                    or not seq.params.needs_penalties
                 )
                 and seq.output_token_ids
                 and seq.output_token_ids[-1] >= 0

@@ -140,3 +140,38 @@ def test_draft_progress_resets_on_preemption():
     seq._draft_progress = 7
     seq.reset_for_recompute()
     assert getattr(seq, "_draft_progress", 0) == 0
+
+
+def test_spec_with_logprobs_matches_plain():
+    """Speculative chunks now carry logprobs: one value per accepted
+    token, numerically matching the plain engine's per-step logprobs."""
+    def run(spec):
+        eng = mk("tiny-llama" if spec else None, k=4 if spec else 0)
+        return eng
+
+    plain = run(False)
+    p = SamplingParams(max_tokens=16, temperature=0.0, ignore_eos=True,
+                       logprobs=1)
+
+    def collect(eng, params):
+        eng.add_request("lp", PROMPT, params)
+        toks, lps = [], []
+        while eng.has_unfinished():
+            for out in eng.step():
+                toks.extend(out.new_token_ids)
+                if out.new_logprobs:
+                    lps.extend(out.new_logprobs)
+        return toks, lps
+
+    want_toks, want_lps = collect(plain, p)
+    spec = run(True)
+    spec.runner.model.load_state_dict(plain.runner.model.state_dict())
+    spec.scheduler.draft_proposer.load_target_weights(spec.runner.model)
+    got_toks, got_lps = collect(
+        spec, SamplingParams(max_tokens=16, temperature=0.0,
+                             ignore_eos=True, logprobs=1))
+    assert got_toks == want_toks
+    assert spec.runner.spec_proposed > 0
+    assert len(got_lps) == len(want_lps) == 16
+    for a, b in zip(got_lps, want_lps):
+        assert abs(a - b) < 1e-4, (a, b)
