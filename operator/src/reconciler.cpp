@@ -355,6 +355,98 @@ ValuePtr build_engine_service(const Ctx& ctx, const ValuePtr& cr) {
   return svc;
 }
 
+ValuePtr build_router_service(const Ctx& ctx, const ValuePtr& cr) {
+  auto crmeta = cr->get("metadata");
+  auto spec = cr->get("spec");
+  std::string name = crmeta->get_str("name");
+  int port = spec ? (int)spec->get_num("port", 8000) : 8000;
+  int svc_port = spec ? (int)spec->get_num("servicePort", 80) : 80;
+  auto svc = Value::object();
+  svc->set("apiVersion", "v1");
+  svc->set("kind", "Service");
+  svc->set("metadata",
+           metadata(name + "-router-service", ctx.ns, name + "-router",
+                    hash_str(spec)));
+  auto sspec = Value::object();
+  auto sel = Value::object();
+  sel->set("app", name + "-router");
+  sspec->set("selector", sel);
+  auto ports = Value::array();
+  auto p = Value::object();
+  p->set("name", "http");
+  p->set("port", svc_port);
+  p->set("targetPort", port);
+  ports->push(p);
+  sspec->set("ports", ports);
+  svc->set("spec", sspec);
+  return svc;
+}
+
+// router SA + namespaced Role (pod read for k8s discovery) + binding —
+// reference vllmrouter_controller.go:196-539 creates the same trio
+ValuePtr build_router_sa(const Ctx& ctx, const ValuePtr& cr) {
+  std::string name = cr->get("metadata")->get_str("name");
+  auto sa = Value::object();
+  sa->set("apiVersion", "v1");
+  sa->set("kind", "ServiceAccount");
+  sa->set("metadata", metadata(name + "-router-sa", ctx.ns,
+                               name + "-router", ""));
+  return sa;
+}
+
+ValuePtr build_router_role(const Ctx& ctx, const ValuePtr& cr) {
+  std::string name = cr->get("metadata")->get_str("name");
+  auto role = Value::object();
+  role->set("apiVersion", "rbac.authorization.k8s.io/v1");
+  role->set("kind", "Role");
+  role->set("metadata", metadata(name + "-router-role", ctx.ns,
+                                 name + "-router", ""));
+  auto rules = Value::array();
+  auto r = Value::object();
+  auto groups = Value::array();
+  push_arg(groups, "");
+  r->set("apiGroups", groups);
+  auto res = Value::array();
+  push_arg(res, "pods");
+  auto spec = cr->get("spec");
+  if (spec &&
+      spec->get_str("serviceDiscovery") == "k8s_service_name") {
+    push_arg(res, "services");
+    push_arg(res, "endpoints");
+  }
+  r->set("resources", res);
+  auto verbs = Value::array();
+  push_arg(verbs, "get");
+  push_arg(verbs, "list");
+  push_arg(verbs, "watch");
+  r->set("verbs", verbs);
+  rules->push(r);
+  role->set("rules", rules);
+  return role;
+}
+
+ValuePtr build_router_rolebinding(const Ctx& ctx, const ValuePtr& cr) {
+  std::string name = cr->get("metadata")->get_str("name");
+  auto rb = Value::object();
+  rb->set("apiVersion", "rbac.authorization.k8s.io/v1");
+  rb->set("kind", "RoleBinding");
+  rb->set("metadata", metadata(name + "-router-rb", ctx.ns,
+                               name + "-router", ""));
+  auto subjects = Value::array();
+  auto subj = Value::object();
+  subj->set("kind", "ServiceAccount");
+  subj->set("name", name + "-router-sa");
+  subj->set("namespace", ctx.ns);
+  subjects->push(subj);
+  rb->set("subjects", subjects);
+  auto ref = Value::object();
+  ref->set("apiGroup", "rbac.authorization.k8s.io");
+  ref->set("kind", "Role");
+  ref->set("name", name + "-router-role");
+  rb->set("roleRef", ref);
+  return rb;
+}
+
 ValuePtr build_router_deployment(const Ctx& ctx, const ValuePtr& cr) {
   auto crmeta = cr->get("metadata");
   auto spec = cr->get("spec");
@@ -389,6 +481,8 @@ ValuePtr build_router_deployment(const Ctx& ctx, const ValuePtr& cr) {
   std::string app = name + "-router";
   auto meta = metadata(name + "-router", ctx.ns, app, hash_str(spec));
   auto d = deployment_skeleton(meta, app, replicas, c);
+  d->get("spec")->get("template")->get("spec")
+      ->set("serviceAccountName", name + "-router-sa");
   return d;
 }
 
@@ -550,10 +644,25 @@ int reconcile_routers(const Ctx& ctx) {
   auto items = list->get("items");
   if (!items) return 0;
   int actions = 0;
+  std::string base = ctx.api_server;
   std::string dep_path =
-      ctx.api_server + "/apis/apps/v1/namespaces/" + ctx.ns + "/deployments";
-  for (auto& cr : items->arr)
+      base + "/apis/apps/v1/namespaces/" + ctx.ns + "/deployments";
+  std::string svc_path =
+      base + "/api/v1/namespaces/" + ctx.ns + "/services";
+  std::string sa_path =
+      base + "/api/v1/namespaces/" + ctx.ns + "/serviceaccounts";
+  std::string role_path = base +
+      "/apis/rbac.authorization.k8s.io/v1/namespaces/" + ctx.ns + "/roles";
+  std::string rb_path = base +
+      "/apis/rbac.authorization.k8s.io/v1/namespaces/" + ctx.ns +
+      "/rolebindings";
+  for (auto& cr : items->arr) {
+    if (apply(ctx, sa_path, build_router_sa(ctx, cr))) actions++;
+    if (apply(ctx, role_path, build_router_role(ctx, cr))) actions++;
+    if (apply(ctx, rb_path, build_router_rolebinding(ctx, cr))) actions++;
+    if (apply(ctx, svc_path, build_router_service(ctx, cr))) actions++;
     if (apply(ctx, dep_path, build_router_deployment(ctx, cr))) actions++;
+  }
   return actions;
 }
 

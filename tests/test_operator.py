@@ -14,7 +14,7 @@ import time
 
 import pytest
 import uvicorn
-from fastapi import FastAPI, Request
+from fastapi import FastAPI, HTTPException, Request
 from fastapi.responses import JSONResponse
 
 OPERATOR = os.path.join(os.path.dirname(__file__), "..", "operator",
@@ -47,6 +47,7 @@ class FakeK8s:
         self.configmaps = {}
         self.scaledobjects = {}
         self.leases = {}
+        self.rbac = {}          # name -> ServiceAccount/Role/RoleBinding
         self.statuses = {}      # (plural, name) -> status patch
         self.cr_patches = []    # (plural, name, body)
         self.pods = []
@@ -80,6 +81,45 @@ class FakeK8s:
             if name in self.services:
                 return self.services[name]
             return JSONResponse(status_code=404, content={})
+
+        @a.get("/api/v1/namespaces/{ns}/serviceaccounts/{name}")
+        async def get_sa(ns: str, name: str):
+            if name in self.rbac:
+                return self.rbac[name]
+            raise HTTPException(status_code=404)
+
+        @a.post("/api/v1/namespaces/{ns}/serviceaccounts")
+        async def post_sa(ns: str, request: Request):
+            body = await request.json()
+            self.rbac[body["metadata"]["name"]] = body
+            return body
+
+        @a.get("/apis/rbac.authorization.k8s.io/v1/namespaces/{ns}"
+               "/roles/{name}")
+        async def get_role(ns: str, name: str):
+            if name in self.rbac:
+                return self.rbac[name]
+            raise HTTPException(status_code=404)
+
+        @a.post("/apis/rbac.authorization.k8s.io/v1/namespaces/{ns}/roles")
+        async def post_role(ns: str, request: Request):
+            body = await request.json()
+            self.rbac[body["metadata"]["name"]] = body
+            return body
+
+        @a.get("/apis/rbac.authorization.k8s.io/v1/namespaces/{ns}"
+               "/rolebindings/{name}")
+        async def get_rb(ns: str, name: str):
+            if name in self.rbac:
+                return self.rbac[name]
+            raise HTTPException(status_code=404)
+
+        @a.post("/apis/rbac.authorization.k8s.io/v1/namespaces/{ns}"
+                "/rolebindings")
+        async def post_rb(ns: str, request: Request):
+            body = await request.json()
+            self.rbac[body["metadata"]["name"]] = body
+            return body
 
         @a.post("/api/v1/namespaces/{ns}/services")
         async def create_svc(ns: str, request: Request):
@@ -249,6 +289,16 @@ def test_router_and_cacheserver_reconcile(fake_k8s):
     assert "production_stack_amd.kvpool.controller" in (
         cs["spec"]["template"]["spec"]["containers"][0]["command"]
     )
+    # router CR also provisions Service + SA + Role + RoleBinding and
+    # binds the pod to the SA (reference vllmrouter_controller.go:196-539)
+    assert fake_k8s.services.get("main-router-service") is not None
+    assert fake_k8s.rbac.get("main-router-sa")["kind"] == "ServiceAccount"
+    role = fake_k8s.rbac.get("main-router-role")
+    assert role["rules"][0]["resources"] == ["pods"]
+    rb = fake_k8s.rbac.get("main-router-rb")
+    assert rb["subjects"][0]["name"] == "main-router-sa"
+    assert (rd["spec"]["template"]["spec"]["serviceAccountName"]
+            == "main-router-sa")
 
 
 def test_lora_adapter_load_call(fake_k8s, tmp_path):
