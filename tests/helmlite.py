@@ -169,8 +169,10 @@ class Renderer:
                 cur = getattr(cur, p, UNDEF)
         return cur
 
+    _NO_PIPE = object()
+
     def _eval_call(self, toks: List[str], dot: Any, vars: Dict[str, Any],
-                   piped: Any = UNDEF) -> Any:
+                   piped: Any = _NO_PIPE) -> Any:
         # resolve parenthesized sub-expressions first
         resolved: List[Any] = []
         i = 0
@@ -191,7 +193,7 @@ class Renderer:
                 resolved.append(toks[i])
                 i += 1
         if not resolved:
-            return piped
+            return UNDEF if piped is self._NO_PIPE else piped
         head = resolved[0]
         args = resolved[1:]
 
@@ -202,7 +204,10 @@ class Renderer:
 
         if isinstance(head, str) and head in _FUNCS:
             fargs = [ev(a) for a in args]
-            if piped is not UNDEF:
+            # helm semantics: a nil/missing PIPED value still reaches the
+            # function (quote -> "", toYaml -> "", default -> fallback);
+            # a call with no pipe at all gets only its literal args
+            if piped is not self._NO_PIPE:
                 fargs.append(piped)
             return _FUNCS[head](self, dot, vars, *fargs)
         if args:  # e.g. `.Files.Get "path"` method call style
@@ -377,7 +382,8 @@ _FUNCS = {
     "indent": lambda r, d, v, n, x: _indent(n, x),
     "nindent": lambda r, d, v, n, x: "\n" + _indent(n, x),
     "toYaml": lambda r, d, v, x: _to_yaml(x),
-    "toJson": lambda r, d, v, x: json.dumps(x),
+    "toJson": lambda r, d, v, x: json.dumps(
+        None if x is UNDEF else x),
     "b64enc": lambda r, d, v, x: base64.b64encode(
         _render_scalar(x).encode()).decode(),
     "lower": lambda r, d, v, x: _render_scalar(x).lower(),

@@ -20,7 +20,7 @@ TEMPLATES = sorted(
                                     "*.yaml"))
 )
 EXAMPLES = sorted(glob.glob(os.path.join(helmlite.CHART_DIR, "..",
-                                         "examples", "values-*.yaml")))
+                                         "examples", "*-values.yaml")))
 
 
 @pytest.mark.parametrize("spec", SPECS, ids=[os.path.basename(s)
