@@ -58,6 +58,8 @@ class UserSession:
     system_prompt: str
     history: List[dict] = field(default_factory=list)
     rounds_done: int = 0
+    # real questions drawn from a ShareGPT conversation (else generated)
+    questions: Optional[List[str]] = None
 
 
 async def run_round(
@@ -67,7 +69,10 @@ async def run_round(
     records: List[RequestRecord],
     rng: random.Random,
 ) -> None:
-    question = gen_text(args.question_len, rng)
+    if user.questions and user.rounds_done < len(user.questions):
+        question = user.questions[user.rounds_done]
+    else:
+        question = gen_text(args.question_len, rng)
     user.history.append({"role": "user", "content": question})
     messages = (
         [{"role": "system", "content": user.system_prompt}] + user.history
@@ -172,12 +177,37 @@ def summarize(records: List[RequestRecord], elapsed: float) -> dict:
     return summary
 
 
+def load_sharegpt(path: str, num_users: int, rng) -> List[List[str]]:
+    """ShareGPT-format JSON -> per-user human-turn question lists
+    (reference benchmarks/multi-round-qa/data_preprocessing.py role:
+    drive the multi-round workload with real conversation turns)."""
+    with open(path) as f:
+        data = json.load(f)
+    convs = []
+    for item in data:
+        turns = item.get("conversations") or item.get("items") or []
+        qs = [t.get("value", "") for t in turns
+              if t.get("from") in ("human", "user") and t.get("value")]
+        if qs:
+            convs.append(qs)
+    if not convs:
+        raise ValueError(f"no usable conversations in {path}")
+    rng.shuffle(convs)
+    return [convs[u % len(convs)] for u in range(num_users)]
+
+
 async def main_async(args) -> dict:
     rng = random.Random(args.seed)
     system_prompt = gen_text(args.shared_system_prompt, rng)
+    sharegpt = (
+        load_sharegpt(args.sharegpt_file, args.num_users, rng)
+        if getattr(args, "sharegpt_file", None) else None
+    )
     users = []
     for u in range(args.num_users):
         user = UserSession(u, system_prompt)
+        if sharegpt:
+            user.questions = sharegpt[u]
         if args.user_history_prompt > 0:
             user.history.append(
                 {
@@ -219,6 +249,9 @@ def parse_args(argv=None):
     ap.add_argument("--max-rounds-kept", type=int, default=20)
     ap.add_argument("--request-timeout", type=float, default=600.0)
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--sharegpt-file", default=None,
+                    help="ShareGPT-format JSON: drive questions with "
+                         "real conversation turns")
     ap.add_argument("--output", default=None, help="write summary JSON here")
     return ap.parse_args(argv)
 
