@@ -34,5 +34,36 @@ def main() -> int:
     return 0
 
 
+# ---------------------------------------------------------------------------
+# sidecar mode (reference docker/Dockerfile.sidecar): a small HTTP
+# service the LoRA controller / init containers call to fetch weights
+# or adapters onto a shared volume.
+try:
+    from fastapi import FastAPI
+
+    app = FastAPI(title="weights downloader sidecar")
+
+    @app.get("/health")
+    async def health():
+        return {"status": "ok"}
+
+    @app.post("/download")
+    async def download(body: dict):
+        from huggingface_hub import snapshot_download
+
+        path = snapshot_download(
+            body["model_id"],
+            revision=body.get("revision"),
+            token=body.get("token"),
+            local_dir=body.get("target_dir"),
+            allow_patterns=body.get(
+                "allow_patterns",
+                ["*.safetensors", "*.json", "tokenizer*", "*.model"]),
+        )
+        return {"path": path}
+except ImportError:  # CLI-only environments
+    app = None
+
+
 if __name__ == "__main__":
     sys.exit(main())
