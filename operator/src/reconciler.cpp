@@ -13,6 +13,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <cstdlib>
 #include <functional>
 #include <map>
 #include <vector>
@@ -734,6 +735,33 @@ int reconcile_loraadapters(const Ctx& ctx) {
     std::string adapter_name =
         src ? src->get_str("adapterName", cr_name) : cr_name;
     std::string adapter_path = src ? src->get_str("adapterPath") : "";
+    std::string src_type = src ? src->get_str("type", "local") : "local";
+    // non-local sources (s3 | http | huggingface — reference
+    // loraadapter_types.go:55-58): delegate the fetch to the downloader
+    // sidecar (docker/Dockerfile.sidecar), which writes onto the shared
+    // volume the engine pods mount and returns the local path.
+    if (src_type != "local" && src) {
+      const char* dl_env = std::getenv("PS_OPERATOR_DOWNLOADER_URL");
+      std::string dl = dl_env ? dl_env : "http://downloader:8000";
+      auto req = Value::object();
+      req->set("model_id", src->get_str("repository",
+                                        src->get_str("adapterPath")));
+      if (!adapter_path.empty()) req->set("target_dir", adapter_path);
+      auto r = try_request("POST", dl + "/download", psjson::dump(req),
+                           "");
+      if (r.ok() && !r.body.empty()) {
+        try {
+          auto resp = psjson::parse(r.body);
+          std::string got = resp ? resp->get_str("path") : "";
+          if (!got.empty()) adapter_path = got;
+        } catch (const std::exception&) {
+          // malformed sidecar reply: keep the declared adapterPath
+        }
+      } else if (adapter_path.empty()) {
+        // nothing to load yet; leave status Pending this pass
+        continue;
+      }
+    }
     auto dcfg = spec->get("loraAdapterDeploymentConfig");
     std::string algo = dcfg ? dcfg->get_str("algorithm", "default")
                             : "default";
