@@ -95,9 +95,13 @@ class CacheServer:
         self.port = port
         self.store = CacheStore(capacity_gb)
         self._server: Optional[asyncio.AbstractServer] = None
+        self._conn_tasks: set = set()
 
     async def _handle(self, reader: asyncio.StreamReader,
                       writer: asyncio.StreamWriter) -> None:
+        task = asyncio.current_task()
+        if task is not None:
+            self._conn_tasks.add(task)
         try:
             while True:
                 msg = await recv_msg(reader)
@@ -123,9 +127,11 @@ class CacheServer:
                     await send_msg(writer, self.store.stats())
                 else:
                     await send_msg(writer, {"error": f"bad op {op!r}"})
-        except (asyncio.IncompleteReadError, ConnectionResetError):
+        except (asyncio.IncompleteReadError, ConnectionResetError,
+                asyncio.CancelledError):
             pass
         finally:
+            self._conn_tasks.discard(task)
             writer.close()
 
     async def start(self) -> None:
@@ -140,6 +146,15 @@ class CacheServer:
         if self._server is not None:
             self._server.close()
             await self._server.wait_closed()
+        # drain live connection handlers so no coroutine is abandoned
+        # mid-await when the loop stops (one tick so accepted-but-not-
+        # yet-started handlers register in _conn_tasks)
+        await asyncio.sleep(0.05)
+        for t in list(self._conn_tasks):
+            t.cancel()
+        if self._conn_tasks:
+            await asyncio.gather(*self._conn_tasks,
+                                 return_exceptions=True)
 
 
 async def run_cacheserver(host: str, port: int, capacity_gb: float) -> None:

@@ -299,6 +299,7 @@ def test_cacheserver_store_and_client_roundtrip():
         assert st["records"] == 2 and st["hits"] == 2
         c.close()
     finally:
+        asyncio.run_coroutine_threadsafe(srv.stop(), loop).result(5)
         loop.call_soon_threadsafe(loop.stop)
 
 
@@ -372,6 +373,7 @@ def test_remote_kv_shared_across_engines():
         a.host_pool.stop()
         b.host_pool.stop()
     finally:
+        asyncio.run_coroutine_threadsafe(srv.stop(), loop).result(5)
         loop.call_soon_threadsafe(loop.stop)
 
 
@@ -431,6 +433,7 @@ def test_remote_kv_int8_records_roundtrip():
         a.host_pool.stop()
         b.host_pool.stop()
     finally:
+        asyncio.run_coroutine_threadsafe(srv.stop(), loop).result(5)
         loop.call_soon_threadsafe(loop.stop)
 
 
