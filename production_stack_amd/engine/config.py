@@ -38,6 +38,10 @@ class ModelConfig:
     # Llama-3.1-style RoPE frequency scaling (None = off):
     # (factor, low_freq_factor, high_freq_factor, original_max_position)
     rope_scaling: Optional[tuple] = None
+    # Mixtral-style sparse MoE MLP: num_experts > 0 replaces the dense
+    # gate/up/down with per-expert projections + a top-k router
+    num_experts: int = 0
+    num_experts_per_tok: int = 2
 
     @property
     def q_size(self) -> int:
@@ -178,6 +182,34 @@ ARCHITECTURES = {
         max_position=32768,
         sliding_window=4096,
     ),
+
+    "mixtral-8x7b": ModelConfig(
+        name="mixtral-8x7b",
+        hidden_size=4096,
+        num_layers=32,
+        num_q_heads=32,
+        num_kv_heads=8,
+        head_dim=128,
+        intermediate_size=14336,
+        vocab_size=32000,
+        rope_theta=1e6,
+        max_position=32768,
+        num_experts=8,
+        num_experts_per_tok=2,
+    ),
+    "tiny-mixtral": ModelConfig(
+        name="tiny-mixtral",
+        hidden_size=64,
+        num_layers=2,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=16,
+        intermediate_size=128,
+        vocab_size=512,
+        max_position=512,
+        num_experts=4,
+        num_experts_per_tok=2,
+    ),
     "qwen2-7b": ModelConfig(
         name="qwen2-7b",
         hidden_size=3584,
@@ -245,6 +277,20 @@ ARCHITECTURES = {
         sliding_window=64,
     ),
     # GPU-runnable small config with kernel-supported head_dim.
+    "mini-mixtral": ModelConfig(
+        name="mini-mixtral",
+        hidden_size=512,
+        num_layers=4,
+        num_q_heads=4,
+        num_kv_heads=2,
+        head_dim=128,
+        intermediate_size=1024,
+        vocab_size=2048,
+        rope_theta=10000.0,
+        max_position=4096,
+        num_experts=4,
+        num_experts_per_tok=2,
+    ),
     "mini-llama": ModelConfig(
         name="mini-llama",
         hidden_size=512,

@@ -152,6 +152,9 @@ class ModelRunner:
             or self.config.enforce_eager
             or self.pp_size > 1
             or self.config.parallel.tensor_parallel_size > 1
+            # MoE routing is data-dependent (per-expert row gathers):
+            # a captured decode graph would freeze one routing pattern
+            or self.model_cfg.num_experts > 0
         ):
             return
         from production_stack_amd.engine.graph_runner import DecodeGraphRunner

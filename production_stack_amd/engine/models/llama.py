@@ -112,12 +112,31 @@ class LlamaLayer(nn.Module):
             torch.empty(qs + 2 * kvs, h, dtype=torch.bfloat16)
         )
         self.o_proj = nn.Parameter(torch.empty(h, qs, dtype=torch.bfloat16))
-        self.gate_up_proj = nn.Parameter(
-            torch.empty(2 * self.inter, h, dtype=torch.bfloat16)
-        )
-        self.down_proj = nn.Parameter(
-            torch.empty(h, self.inter, dtype=torch.bfloat16)
-        )
+        self.n_experts = cfg.num_experts
+        self.top_k = cfg.num_experts_per_tok
+        if self.n_experts:
+            # Mixtral-style sparse MoE: stacked per-expert projections +
+            # a top-k softmax router; each expert reuses the dense
+            # hipBLASLt GEMM + fused SiLU-mul kernels, so no new device
+            # code is involved (EP across GPUs is a round-3 item)
+            self.moe_gate = nn.Parameter(
+                torch.empty(self.n_experts, h, dtype=torch.bfloat16)
+            )
+            self.experts_gate_up = nn.Parameter(
+                torch.empty(self.n_experts, 2 * self.inter, h,
+                            dtype=torch.bfloat16)
+            )
+            self.experts_down = nn.Parameter(
+                torch.empty(self.n_experts, h, self.inter,
+                            dtype=torch.bfloat16)
+            )
+        else:
+            self.gate_up_proj = nn.Parameter(
+                torch.empty(2 * self.inter, h, dtype=torch.bfloat16)
+            )
+            self.down_proj = nn.Parameter(
+                torch.empty(h, self.inter, dtype=torch.bfloat16)
+            )
         self.input_norm = nn.Parameter(torch.empty(h, dtype=torch.bfloat16))
         self.post_attn_norm = nn.Parameter(
             torch.empty(h, dtype=torch.bfloat16)
@@ -273,6 +292,13 @@ class LlamaLayer(nn.Module):
             )
         attn_out = pstate.tp_all_reduce(attn_out)
 
+        if self.n_experts:
+            hidden, residual = ops.fused_add_rms_norm(
+                attn_out, residual, self.post_attn_norm, cfg.rms_norm_eps
+            )
+            mlp_out = self._moe_forward(hidden)
+            mlp_out = pstate.tp_all_reduce(mlp_out)
+            return mlp_out, residual
         if fuse_q:
             xq, sx = ops.rms_norm_fp8(
                 attn_out, self.post_attn_norm, cfg.rms_norm_eps,
@@ -319,6 +345,32 @@ class LlamaLayer(nn.Module):
             )
         mlp_out = pstate.tp_all_reduce(mlp_out)
         return mlp_out, residual
+
+    def _moe_forward(self, x: torch.Tensor) -> torch.Tensor:
+        """HF Mixtral routing rule: softmax over all experts, take the
+        top-k, renormalize the selected weights; each expert runs the
+        dense GEMM + fused SiLU-mul path over its assigned rows."""
+        logits = F.linear(x.float(), self.moe_gate.float())
+        probs = torch.softmax(logits, dim=-1)
+        weights, sel = torch.topk(probs, self.top_k, dim=-1)
+        weights = weights / weights.sum(dim=-1, keepdim=True)
+        weights = weights.to(x.dtype)
+        out = torch.zeros_like(x)
+        for e in range(self.n_experts):
+            mask = sel == e  # [T, K]
+            rows = mask.any(dim=-1).nonzero().flatten()
+            if rows.numel() == 0:
+                continue
+            xe = x.index_select(0, rows)
+            act = ops.silu_and_mul(
+                gemm_policy.linear(xe, self.experts_gate_up[e])
+            )
+            ye = gemm_policy.linear(act, self.experts_down[e])
+            w = (weights * mask.to(weights.dtype)).sum(-1)
+            out.index_add_(
+                0, rows, ye * w.index_select(0, rows).unsqueeze(-1)
+            )
+        return out
 
 
 class LlamaForCausalLM(nn.Module):
@@ -390,9 +442,11 @@ class LlamaForCausalLM(nn.Module):
         lm_head and norms stay bf16."""
         for layer in self.layers:
             fp8_w = {}
-            for key, pname in (("qkv", "qkv_proj"), ("o", "o_proj"),
-                               ("gate_up", "gate_up_proj"),
-                               ("down", "down_proj")):
+            proj_names = [("qkv", "qkv_proj"), ("o", "o_proj")]
+            if not layer.n_experts:  # MoE experts stay bf16 for now
+                proj_names += [("gate_up", "gate_up_proj"),
+                               ("down", "down_proj")]
+            for key, pname in proj_names:
                 w = getattr(layer, pname)
                 if (w.is_cuda
                         and ops.fp8_rowwise_supported(w.device)):

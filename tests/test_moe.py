@@ -1,0 +1,111 @@
+"""Mixtral-style sparse MoE family: router/dispatch correctness (E=1
+degenerates to the dense layer exactly), batching invariance, and
+engine-level determinism."""
+
+import dataclasses
+
+import torch
+
+from production_stack_amd.engine.config import (
+    ARCHITECTURES,
+    CacheConfig,
+    EngineConfig,
+    SchedulerConfig,
+)
+from production_stack_amd.engine.engine import LLMEngine
+from production_stack_amd.engine.models.llama import LlamaLayer
+from production_stack_amd.engine.sampling import SamplingParams
+
+
+def test_single_expert_moe_equals_dense_layer():
+    """num_experts=1: the router weight renormalizes to 1.0, so the MoE
+    MLP must equal the dense gate/up/down path bit-for-bit."""
+    cfg = ARCHITECTURES["tiny-mixtral"]
+    moe_cfg = dataclasses.replace(cfg, num_experts=1,
+                                  num_experts_per_tok=1)
+    dense_cfg = dataclasses.replace(cfg, num_experts=0)
+    torch.manual_seed(0)
+    moe = LlamaLayer(moe_cfg, tp=1)
+    dense = LlamaLayer(dense_cfg, tp=1)
+    for p in list(moe.parameters()) + list(dense.parameters()):
+        p.data.normal_(0, 0.05)
+    moe.moe_gate.data.normal_(0, 1.0)
+    dense.gate_up_proj.data.copy_(moe.experts_gate_up.data[0])
+    dense.down_proj.data.copy_(moe.experts_down.data[0])
+
+    x = torch.randn(12, cfg.hidden_size).to(torch.bfloat16)
+    got = moe._moe_forward(x)
+    import torch.nn.functional as F
+
+    from production_stack_amd import ops
+
+    want = F.linear(ops.silu_and_mul(F.linear(x, dense.gate_up_proj)),
+                    dense.down_proj)
+    assert torch.equal(got, want)
+
+
+def test_moe_routing_uses_multiple_experts():
+    cfg = ARCHITECTURES["tiny-mixtral"]
+    torch.manual_seed(1)
+    layer = LlamaLayer(cfg, tp=1)
+    for p in layer.parameters():
+        p.data.normal_(0, 0.05)
+    layer.moe_gate.data.normal_(0, 1.0)
+    x = torch.randn(64, cfg.hidden_size).to(torch.bfloat16)
+    logits = torch.nn.functional.linear(x.float(),
+                                        layer.moe_gate.float())
+    sel = torch.topk(torch.softmax(logits, -1), layer.top_k, -1).indices
+    assert len(set(sel.flatten().tolist())) > 1, "router collapsed"
+    # and the forward runs with that mixed assignment
+    out = layer._moe_forward(x)
+    assert out.shape == x.shape and torch.isfinite(out.float()).all()
+
+
+def _engine(**kw):
+    cfg = EngineConfig(
+        model="tiny-mixtral",
+        max_model_len=256,
+        seed=kw.pop("seed", 3),
+        cache=CacheConfig(num_gpu_blocks=64, block_size=16),
+        scheduler=SchedulerConfig(
+            max_num_seqs=4,
+            max_num_batched_tokens=kw.pop("max_num_batched_tokens", 128),
+        ),
+    )
+    return LLMEngine(cfg, device="cpu")
+
+
+def test_mixtral_engine_deterministic_and_chunking_invariant():
+    p = SamplingParams(max_tokens=10, temperature=0.0, ignore_eos=True)
+    prompts = [[5, 6, 7, 8] * 8, list(range(40, 90))]
+    a = _engine()
+    out_a = a.generate(prompts, p)
+    b = _engine()
+    b.runner.model.load_state_dict(a.runner.model.state_dict())
+    assert b.generate(prompts, p) == out_a
+    # different prefill chunking must not change MoE outputs (routing is
+    # per-token, independent of how tokens are batched)
+    c = _engine(max_num_batched_tokens=32)
+    c.runner.model.load_state_dict(a.runner.model.state_dict())
+    assert c.generate(prompts, p) == out_a
+
+
+def test_mixtral_prefix_cache_and_spec_decode():
+    """MoE composes with prefix caching and n-gram speculation."""
+    cfg = EngineConfig(
+        model="tiny-mixtral",
+        max_model_len=256,
+        seed=3,
+        cache=CacheConfig(num_gpu_blocks=64, block_size=16,
+                          enable_prefix_caching=True),
+        scheduler=SchedulerConfig(max_num_seqs=4,
+                                  max_num_batched_tokens=128,
+                                  num_speculative_tokens=3),
+    )
+    eng = LLMEngine(cfg, device="cpu")
+    p = SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True)
+    prompt = [9, 10, 11, 12] * 8
+    first = eng.generate([prompt], p)["offline-0"]
+    again = eng.generate([prompt], p)["offline-0"]
+    assert first == again
+    assert eng.block_manager.prefix_hits > 0
