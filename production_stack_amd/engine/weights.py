@@ -62,10 +62,25 @@ def load_safetensors(model, path: str) -> None:
                 vb = shard(get(pre + "self_attn.v_proj.bias"), 0)
                 layer.qkv_bias.copy_(torch.cat([qb, kb, vb], dim=0))
             layer.o_proj.copy_(shard(get(pre + "self_attn.o_proj.weight"), 1))
-            g = shard(get(pre + "mlp.gate_proj.weight"), 0)
-            u = shard(get(pre + "mlp.up_proj.weight"), 0)
-            layer.gate_up_proj.copy_(torch.cat([g, u], dim=0))
-            layer.down_proj.copy_(shard(get(pre + "mlp.down_proj.weight"), 1))
+            if layer.n_experts:
+                # HF Mixtral layout: block_sparse_moe.gate +
+                # experts.E.{w1=gate, w3=up, w2=down}
+                layer.moe_gate.copy_(
+                    get(pre + "block_sparse_moe.gate.weight"))
+                for e in range(layer.n_experts):
+                    ep = pre + f"block_sparse_moe.experts.{e}."
+                    g = shard(get(ep + "w1.weight"), 0)
+                    u = shard(get(ep + "w3.weight"), 0)
+                    layer.experts_gate_up[e].copy_(
+                        torch.cat([g, u], dim=0))
+                    layer.experts_down[e].copy_(
+                        shard(get(ep + "w2.weight"), 1))
+            else:
+                g = shard(get(pre + "mlp.gate_proj.weight"), 0)
+                u = shard(get(pre + "mlp.up_proj.weight"), 0)
+                layer.gate_up_proj.copy_(torch.cat([g, u], dim=0))
+                layer.down_proj.copy_(
+                    shard(get(pre + "mlp.down_proj.weight"), 1))
             layer.input_norm.copy_(get(pre + "input_layernorm.weight"))
             layer.post_attn_norm.copy_(
                 get(pre + "post_attention_layernorm.weight")
@@ -102,10 +117,22 @@ def save_hf_safetensors(model, path: str) -> None:
                 qb[qs + kvs :].clone()
             )
         state[pre + "self_attn.o_proj.weight"] = layer.o_proj.detach().cpu()
-        gu = layer.gate_up_proj.detach().cpu()
-        state[pre + "mlp.gate_proj.weight"] = gu[: layer.inter].clone()
-        state[pre + "mlp.up_proj.weight"] = gu[layer.inter :].clone()
-        state[pre + "mlp.down_proj.weight"] = layer.down_proj.detach().cpu()
+        if layer.n_experts:
+            state[pre + "block_sparse_moe.gate.weight"] = (
+                layer.moe_gate.detach().cpu())
+            for e in range(layer.n_experts):
+                ep = pre + f"block_sparse_moe.experts.{e}."
+                gu = layer.experts_gate_up[e].detach().cpu()
+                state[ep + "w1.weight"] = gu[: layer.inter].clone()
+                state[ep + "w3.weight"] = gu[layer.inter :].clone()
+                state[ep + "w2.weight"] = (
+                    layer.experts_down[e].detach().cpu().clone())
+        else:
+            gu = layer.gate_up_proj.detach().cpu()
+            state[pre + "mlp.gate_proj.weight"] = gu[: layer.inter].clone()
+            state[pre + "mlp.up_proj.weight"] = gu[layer.inter :].clone()
+            state[pre + "mlp.down_proj.weight"] = (
+                layer.down_proj.detach().cpu())
         state[pre + "input_layernorm.weight"] = (
             layer.input_norm.detach().cpu()
         )

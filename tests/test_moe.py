@@ -109,3 +109,24 @@ def test_mixtral_prefix_cache_and_spec_decode():
     again = eng.generate([prompt], p)["offline-0"]
     assert first == again
     assert eng.block_manager.prefix_hits > 0
+
+
+def test_mixtral_safetensors_roundtrip(tmp_path):
+    """HF Mixtral weight layout (block_sparse_moe.gate + experts.E.w1/
+    w2/w3) round-trips through save/load."""
+    from production_stack_amd.engine.models.llama import LlamaForCausalLM
+    from production_stack_amd.engine.weights import (
+        load_safetensors,
+        save_hf_safetensors,
+    )
+
+    cfg = ARCHITECTURES["tiny-mixtral"]
+    torch.manual_seed(2)
+    m = LlamaForCausalLM(cfg)
+    m.random_init(7)
+    save_hf_safetensors(m, str(tmp_path))
+    m2 = LlamaForCausalLM(cfg)
+    load_safetensors(m2, str(tmp_path))
+    for (n1, p1), (n2, p2) in zip(m.named_parameters(),
+                                  m2.named_parameters()):
+        assert n1 == n2 and torch.equal(p1, p2), n1
