@@ -130,3 +130,77 @@ def test_mixtral_safetensors_roundtrip(tmp_path):
     for (n1, p1), (n2, p2) in zip(m.named_parameters(),
                                   m2.named_parameters()):
         assert n1 == n2 and torch.equal(p1, p2), n1
+
+
+def _moe_cfg(weights_path, tp):
+    from production_stack_amd.engine.config import ParallelConfig
+
+    return EngineConfig(
+        model="tiny-mixtral",
+        max_model_len=256,
+        weights_path=weights_path,
+        cache=CacheConfig(num_gpu_blocks=64, block_size=16),
+        scheduler=SchedulerConfig(max_num_seqs=4,
+                                  max_num_batched_tokens=256),
+        parallel=ParallelConfig(tensor_parallel_size=tp),
+    )
+
+
+def _moe_tp_worker(rank, world, weights_path, port, q):
+    import os
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    import torch.distributed as dist
+
+    from production_stack_amd.parallel import state as pstate
+
+    try:
+        eng = LLMEngine(_moe_cfg(weights_path, world), device="cpu")
+        out = eng.generate(
+            [[5, 6, 7, 8] * 8],
+            SamplingParams(max_tokens=8, temperature=0.0,
+                           ignore_eos=True),
+        )["offline-0"]
+        if rank == 0:
+            q.put(("ok", out))
+    except Exception as e:  # pragma: no cover
+        if rank == 0:
+            q.put(("err", repr(e)))
+        raise
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        pstate.destroy()
+
+
+def test_mixtral_tp2_matches_tp1(tmp_path):
+    """MoE under TP: per-expert intermediate dims shard across ranks and
+    the down outputs all-reduce — greedy tokens must match TP=1."""
+    import torch.multiprocessing as mp
+
+    from production_stack_amd.engine.weights import save_hf_safetensors
+
+    eng1 = LLMEngine(_moe_cfg(None, 1), device="cpu")
+    wdir = str(tmp_path / "w")
+    save_hf_safetensors(eng1.runner.model, wdir)
+    want = eng1.generate(
+        [[5, 6, 7, 8] * 8],
+        SamplingParams(max_tokens=8, temperature=0.0, ignore_eos=True),
+    )["offline-0"]
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = []
+    for rank in range(2):
+        p = ctx.Process(target=_moe_tp_worker,
+                        args=(rank, 2, wdir, 29557, q))
+        p.start()
+        procs.append(p)
+    status, out = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+    assert status == "ok", out
+    assert out == want, f"TP2 {out} != TP1 {want}"
