@@ -267,6 +267,115 @@ class JsonPrefixValidator:
 
 
 # ---------------------------------------------------------------------------
+# schema-compiled guide: flat object schemas become a forced template
+
+
+_TYPE_STARTS = {
+    "string": '"',
+    "number": "-0123456789",
+    "integer": "-0123456789",
+    "boolean": "tf",
+    "array": "[",
+    "object": "{",
+    None: None,  # untyped: any JSON value
+}
+
+
+class SchemaGuide:
+    """Compile a flat object schema (`type: object` + `properties` +
+    `additionalProperties: false`) into a forced template: the literal
+    structure `{"key1": <val>, "key2": <val>}` is teacher-forced
+    character by character (keys in declaration order), and each value
+    region runs a JsonPrefixValidator restricted to the property's
+    declared type. This makes json_schema *enforced during decoding*
+    for the common flat-object case, not just validated at the end.
+    Nested/looser schemas fall back to the generic JSON grammar +
+    end-of-request validation (GuidedJsonState)."""
+
+    @staticmethod
+    def compile(schema: dict) -> Optional["SchemaGuide"]:
+        if not isinstance(schema, dict):
+            return None
+        props = schema.get("properties")
+        if (schema.get("type") != "object" or not props
+                or schema.get("additionalProperties") is not False):
+            return None
+        segments: List[tuple] = []
+        first = True
+        for key, sub in props.items():
+            typ = sub.get("type") if isinstance(sub, dict) else None
+            if typ not in _TYPE_STARTS:
+                return None  # union/unknown type: generic fallback
+            prefix = "{" if first else ", "
+            segments.append(("lit", prefix + json.dumps(key) + ": "))
+            segments.append(("val", typ))
+            first = False
+        segments.append(("lit", "}"))
+        return SchemaGuide(segments)
+
+    def __init__(self, segments: List[tuple]) -> None:
+        self.segments = segments
+        self.seg = 0
+        self.lit_pos = 0
+        self.subv: Optional[JsonPrefixValidator] = None
+        self.sub_fresh = True
+
+    def copy(self) -> "SchemaGuide":
+        c = SchemaGuide.__new__(SchemaGuide)
+        c.segments = self.segments
+        c.seg = self.seg
+        c.lit_pos = self.lit_pos
+        c.subv = self.subv.copy() if self.subv is not None else None
+        c.sub_fresh = self.sub_fresh
+        return c
+
+    @property
+    def complete(self) -> bool:
+        return self.seg >= len(self.segments)
+
+    def feed(self, ch: str) -> bool:
+        if self.complete:
+            return ch in _WS
+        kind, payload = self.segments[self.seg]
+        if kind == "lit":
+            if ch != payload[self.lit_pos]:
+                return False
+            self.lit_pos += 1
+            if self.lit_pos == len(payload):
+                self.seg += 1
+                self.lit_pos = 0
+                if (self.seg < len(self.segments)
+                        and self.segments[self.seg][0] == "val"):
+                    self.subv = JsonPrefixValidator()
+                    self.sub_fresh = True
+            return True
+        # value region
+        if self.sub_fresh:
+            starts = _TYPE_STARTS[payload]
+            if starts is not None and ch not in starts:
+                return False
+        if self.subv.feed(ch):
+            self.sub_fresh = False
+            return True
+        if self.subv.complete:
+            # the value ended; this character belongs to the next literal
+            self.seg += 1
+            self.subv = None
+            return self.feed(ch)
+        return False
+
+    def feed_text(self, text: str) -> bool:
+        for ch in text:
+            if not self.feed(ch):
+                return False
+        return True
+
+    def would_accept(self, text: str) -> Optional["SchemaGuide"]:
+        c = self.copy()
+        return c if c.feed_text(text) else None
+
+
+# ---------------------------------------------------------------------------
 # token-level guided state
 
 _TOKEN_TEXT_CACHE: dict = {}
@@ -285,7 +394,10 @@ class GuidedJsonState:
     """Per-request token-level wrapper around the prefix automaton."""
 
     def __init__(self, schema: Optional[dict] = None) -> None:
-        self.v = JsonPrefixValidator()
+        # flat object schemas compile to an enforced template; anything
+        # else uses the generic JSON grammar + end-of-request validation
+        guide = SchemaGuide.compile(schema) if schema else None
+        self.v = guide if guide is not None else JsonPrefixValidator()
         self.schema = schema
         self.consumed = 0  # output tokens already fed
         self.text = ""

@@ -213,3 +213,78 @@ def test_guided_json_async_falls_back_to_sync():
                    if t != tok.eos_token_id)
     assert JsonPrefixValidator().feed_text(text), text
     assert reason in ("stop", "length")
+
+
+def test_schema_guide_forces_template():
+    """Flat object schemas compile to a forced template: exact keys in
+    order, typed values, nothing else."""
+    from production_stack_amd.engine.guided import SchemaGuide
+
+    schema = {
+        "type": "object",
+        "additionalProperties": False,
+        "properties": {
+            "name": {"type": "string"},
+            "age": {"type": "number"},
+            "ok": {"type": "boolean"},
+        },
+        "required": ["name", "age"],
+    }
+    g = SchemaGuide.compile(schema)
+    assert g is not None
+    assert g.feed_text('{"name": "bo')
+    assert g.copy().feed_text('b", "age": 31, "ok": true}')
+    full = '{"name": "bob", "age": 31, "ok": false}'
+    h = SchemaGuide.compile(schema)
+    assert h.feed_text(full) and h.complete
+    # wrong key order / extra keys / wrong value type all rejected
+    assert not SchemaGuide.compile(schema).feed_text('{"age": 3')
+    assert not SchemaGuide.compile(schema).feed_text('{"name": 12')
+    assert not SchemaGuide.compile(schema).feed_text(
+        '{"name": "b", "x": 1')
+    h2 = SchemaGuide.compile(schema)
+    assert h2.feed_text(full)
+    assert not h2.feed(',')  # nothing after the closing brace
+
+    # nested / loose schemas fall back to generic
+    assert SchemaGuide.compile({"type": "object"}) is None
+    assert SchemaGuide.compile(
+        {"type": "object", "properties": {"a": {"type": "string"}}}
+    ) is None  # additionalProperties not false
+
+
+def test_schema_guided_generation_end_to_end():
+    """json_schema with a flat object: generated output IS the schema's
+    shape (enforced during decoding, not just validated after)."""
+    eng = _engine()
+    schema = {
+        "type": "object",
+        "additionalProperties": False,
+        "properties": {"a": {"type": "number"},
+                       "b": {"type": "number"}},
+        "required": ["a", "b"],
+    }
+    p = SamplingParams(
+        max_tokens=48, temperature=1.0, seed=3,
+        response_format={"type": "json_schema",
+                         "json_schema": {"schema": schema}},
+    )
+    eng.add_request("gs", [11, 12], p)
+    toks, reason = [], None
+    for _ in range(80):
+        for out in eng.step():
+            if out.request_id == "gs":
+                toks.extend(out.new_token_ids)
+                if out.finished:
+                    reason = out.finish_reason
+        if reason:
+            break
+    tok = eng.tokenizer
+    text = "".join(tok.decode_token(t) for t in toks
+                   if t != tok.eos_token_id)
+    if reason == "stop":
+        obj = json.loads(text)
+        assert set(obj) == {"a", "b"}
+        assert all(isinstance(v, (int, float)) for v in obj.values())
+    else:
+        assert reason in ("length", "error_json_schema")
