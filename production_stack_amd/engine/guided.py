@@ -376,6 +376,57 @@ class SchemaGuide:
 
 
 # ---------------------------------------------------------------------------
+# regex / choice guides (vLLM guided_regex / guided_choice analogues)
+
+
+class RegexGuide:
+    """Prefix acceptance via `regex` partial matching: a text is legal
+    while it can still extend to a full match; complete once it fully
+    matches (longest-generation semantics: EOS becomes legal at the
+    first full match, other continuations stay legal while the pattern
+    allows them)."""
+
+    force_eos_on_complete = False  # longer matches may exist
+
+    def __init__(self, pattern: str, text: str = "") -> None:
+        import regex as _re
+
+        self.pattern = pattern
+        self._re = _re.compile(pattern)
+        self.text = text
+
+    def copy(self) -> "RegexGuide":
+        c = RegexGuide.__new__(RegexGuide)
+        c.pattern = self.pattern
+        c._re = self._re
+        c.text = self.text
+        return c
+
+    def feed_text(self, text: str) -> bool:
+        cand = self.text + text
+        m = self._re.fullmatch(cand, partial=True)
+        if m is None:
+            return False
+        self.text = cand
+        return True
+
+    def would_accept(self, text: str) -> Optional["RegexGuide"]:
+        c = self.copy()
+        return c if c.feed_text(text) else None
+
+    @property
+    def complete(self) -> bool:
+        m = self._re.fullmatch(self.text)
+        return m is not None
+
+
+def choice_regex(choices: List[str]) -> str:
+    import regex as _re
+
+    return "|".join(_re.escape(c) for c in choices)
+
+
+# ---------------------------------------------------------------------------
 # token-level guided state
 
 _TOKEN_TEXT_CACHE: dict = {}
@@ -419,7 +470,8 @@ class GuidedJsonState:
         found or the vocab is exhausted."""
         import torch
 
-        if self.v.complete:
+        if self.v.complete and getattr(self.v, "force_eos_on_complete",
+                                       True):
             return [eos_token_id], True
         vocab = logits_row.shape[-1]
         texts = _token_texts(tokenizer, vocab)
@@ -438,6 +490,8 @@ class GuidedJsonState:
                     texts[t] = txt
                 if txt and self.v.would_accept(txt) is not None:
                     allowed.append(t)
+            if self.v.complete and eos_token_id not in allowed:
+                allowed.append(eos_token_id)
             if allowed:
                 return allowed, False
             if k >= vocab:
@@ -469,7 +523,9 @@ class GuidedJsonState:
 def guided_state_from_response_format(
     response_format: Optional[dict],
 ) -> Optional[GuidedJsonState]:
-    """Map an OpenAI `response_format` body field to a guided state."""
+    """Map an OpenAI `response_format` body field (or the internal
+    choice/regex forms `_params_from_body` builds from vLLM-style
+    guided_choice / guided_regex) to a guided state."""
     if not response_format:
         return None
     kind = response_format.get("type")
@@ -478,4 +534,14 @@ def guided_state_from_response_format(
     if kind == "json_schema":
         js = response_format.get("json_schema") or {}
         return GuidedJsonState(schema=js.get("schema") or {})
+    if kind == "regex" and response_format.get("pattern"):
+        gs = GuidedJsonState()
+        gs.v = RegexGuide(response_format["pattern"])
+        return gs
+    if kind == "choice" and response_format.get("choices"):
+        gs = GuidedJsonState()
+        gs.v = RegexGuide(choice_regex(
+            [str(c) for c in response_format["choices"]]))
+        gs.v.force_eos_on_complete = True  # a choice is a whole answer
+        return gs
     return None

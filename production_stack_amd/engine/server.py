@@ -63,14 +63,23 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
         min_tokens=int(body.get("min_tokens") or 0),
         min_p=float(body.get("min_p") or 0.0),
         logprobs=_parse_logprobs(body),
-        response_format=(
-            body.get("response_format")
-            if isinstance(body.get("response_format"), dict)
-            and body["response_format"].get("type") in
-            ("json_object", "json_schema")
-            else None
-        ),
+        response_format=_guided_from_body(body),
     )
+
+
+def _guided_from_body(body: dict):
+    """response_format json modes, plus vLLM-style guided_choice /
+    guided_regex mapped onto the same guided-state machinery."""
+    rf = body.get("response_format")
+    if isinstance(rf, dict) and rf.get("type") in ("json_object",
+                                                   "json_schema"):
+        return rf
+    if body.get("guided_choice"):
+        return {"type": "choice",
+                "choices": list(body["guided_choice"])}
+    if body.get("guided_regex"):
+        return {"type": "regex", "pattern": str(body["guided_regex"])}
+    return None
 
 
 def _parse_logprobs(body: dict):

@@ -288,3 +288,55 @@ def test_schema_guided_generation_end_to_end():
         assert all(isinstance(v, (int, float)) for v in obj.values())
     else:
         assert reason in ("length", "error_json_schema")
+
+
+def test_regex_and_choice_guides():
+    from production_stack_amd.engine.guided import (
+        RegexGuide,
+        choice_regex,
+        guided_state_from_response_format,
+    )
+
+    g = RegexGuide(r"[0-9]{3}-[0-9]{4}")
+    assert g.feed_text("123-")
+    assert not g.complete
+    assert g.feed_text("4567") and g.complete
+    assert RegexGuide(r"[0-9]+").would_accept("12") is not None
+    assert RegexGuide(r"[0-9]+").would_accept("a") is None
+    # open-ended pattern: complete but longer matches stay legal
+    h = RegexGuide(r"[0-9]+")
+    assert h.feed_text("1") and h.complete
+    assert h.would_accept("2") is not None
+
+    gs = guided_state_from_response_format(
+        {"type": "choice", "choices": ["yes", "no"]})
+    assert gs.v.feed_text("ye") and not gs.v.complete
+    assert gs.v.feed_text("s") and gs.v.complete
+    assert gs.v.would_accept("x") is None
+    assert choice_regex(["a.b", "c"]) == r"a\.b|c"
+
+
+def test_guided_choice_end_to_end():
+    """guided_choice: the model is forced onto one of the choices, then
+    EOS (choices spelled in the toy vocab's token texts)."""
+    eng = _engine()
+    p = SamplingParams(
+        max_tokens=8, temperature=1.0, seed=2,
+        response_format={"type": "choice",
+                         "choices": ["true", "12", "[]"]},
+    )
+    eng.add_request("gc", [9, 9], p)
+    toks, reason = [], None
+    for _ in range(40):
+        for out in eng.step():
+            if out.request_id == "gc":
+                toks.extend(out.new_token_ids)
+                if out.finished:
+                    reason = out.finish_reason
+        if reason:
+            break
+    tok = eng.tokenizer
+    text = "".join(tok.decode_token(t) for t in toks
+                   if t != tok.eos_token_id)
+    assert reason == "stop"
+    assert text in ("true", "12", "[]"), text
