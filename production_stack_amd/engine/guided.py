@@ -8,12 +8,15 @@ text keeps the output a valid JSON prefix, and once a complete top-level
 value has been produced only EOS remains legal.
 
 `response_format: {"type": "json_object"}` guarantees syntactically
-valid JSON. `{"type": "json_schema", ...}` enforces the same grammar
-during decoding and validates the finished object against the schema's
-`required`/`properties` keys at completion (full schema-to-grammar
-compilation is out of scope; the failure surfaces as
+valid JSON. `{"type": "json_schema", ...}`: flat object schemas
+(`type: object` + `properties` + `additionalProperties: false`)
+compile to an ENFORCED template (SchemaGuide — exact keys in
+declaration order, typed value regions); looser schemas use the
+generic grammar with end-of-request validation, surfacing failures as
 `finish_reason: "error_json_schema"` rather than silently returning a
-non-conforming object).
+non-conforming object. vLLM-style `guided_choice` / `guided_regex`
+ride the same masking machinery through `regex` partial matching
+(RegexGuide).
 
 Masking strategy: candidate tokens are taken from the top-K logits
 (K=64, widening x8 on miss up to the full vocab); each candidate's text
