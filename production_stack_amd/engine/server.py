@@ -369,11 +369,21 @@ def build_server(engine: LLMEngine, served_model: str,
             vecs = await asyncio.to_thread(_embed_texts, texts)
         except RuntimeError as e:
             return JSONResponse(status_code=503, content={"error": str(e)})
+        if body.get("encoding_format") == "base64":
+            import base64 as _b64
+            import struct as _struct
+
+            def enc(v):
+                return _b64.b64encode(
+                    _struct.pack(f"<{len(v)}f", *v)).decode()
+        else:
+            def enc(v):
+                return v
         return {
             "object": "list",
             "model": body.get("model", served_model),
             "data": [
-                {"object": "embedding", "index": i, "embedding": v}
+                {"object": "embedding", "index": i, "embedding": enc(v)}
                 for i, v in enumerate(vecs)
             ],
             "usage": {"prompt_tokens": sum(len(t.split()) for t in texts),

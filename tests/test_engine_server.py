@@ -660,3 +660,33 @@ def test_audio_speech_returns_valid_wav():
         assert r.status_code == 400
 
     with_server(go)
+
+
+def test_embeddings_base64_encoding_format():
+    """encoding_format=base64 packs each vector as little-endian fp32
+    (OpenAI client default decode path)."""
+    import base64
+    import struct
+
+    async def go(client):
+        r = await client.post(
+            "/v1/embeddings",
+            json={"model": "tiny-llama", "input": ["w1 w2", "w3"]},
+            timeout=60,
+        )
+        plain = r.json()["data"]
+        rb = await client.post(
+            "/v1/embeddings",
+            json={"model": "tiny-llama", "input": ["w1 w2", "w3"],
+                  "encoding_format": "base64"},
+            timeout=60,
+        )
+        b64 = rb.json()["data"]
+        for p, b in zip(plain, b64):
+            raw = base64.b64decode(b["embedding"])
+            vals = struct.unpack(f"<{len(raw) // 4}f", raw)
+            assert len(vals) == len(p["embedding"])
+            for x, y in zip(vals, p["embedding"]):
+                assert abs(x - y) < 1e-5
+
+    with_server(go)
