@@ -528,6 +528,13 @@ def build_server(engine: LLMEngine, served_model: str,
                 logger.warning("KV pull failed (%s); recomputing prefill", e)
 
         if body.get("stream"):
+            # OpenAI stream_options: usage rides the final chunk only
+            # when asked for (the bundled benchmark harness asks; the
+            # reference harness depends on it, multi_round_qa.py)
+            include_usage = bool(
+                (body.get("stream_options") or {}).get("include_usage",
+                                                       True)
+            )
 
             async def gen():
                 try:
@@ -590,7 +597,7 @@ def build_server(engine: LLMEngine, served_model: str,
                             "model": model_name,
                             "choices": [choice],
                         }
-                        if out.finished:
+                        if out.finished and include_usage:
                             chunk["usage"] = {
                                 "prompt_tokens": out.num_prompt_tokens,
                                 "completion_tokens": n_out,

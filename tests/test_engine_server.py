@@ -690,3 +690,23 @@ def test_embeddings_base64_encoding_format():
                 assert abs(x - y) < 1e-5
 
     with_server(go)
+
+
+def test_stream_options_include_usage_false_suppresses_usage():
+    async def go(client):
+        async with client.stream(
+            "POST", "/v1/completions",
+            json={"model": "tiny-llama", "prompt": "w1 w2",
+                  "max_tokens": 3, "temperature": 0, "stream": True,
+                  "ignore_eos": True,
+                  "stream_options": {"include_usage": False}},
+            timeout=120,
+        ) as r:
+            saw_usage = False
+            async for line in r.aiter_lines():
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    if "usage" in json.loads(line[6:]):
+                        saw_usage = True
+        assert not saw_usage
+
+    with_server(go)
