@@ -841,6 +841,13 @@ def build_server(engine: LLMEngine, served_model: str,
     async def audio_translations(request: Request):
         return await _transcribe_impl(request, translate=True)
 
+    @app.get("/v1/audio/voices")
+    async def audio_voices():
+        """Voice inventory for the TTS endpoint (reference proxies
+        /v1/audio/voices; the tone-vocoder keys pitch off the voice)."""
+        return {"voices": ["alloy", "echo", "fable", "onyx", "nova",
+                           "shimmer"]}
+
     @app.post("/v1/audio/speech")
     async def audio_speech(request: Request):
         """OpenAI TTS API shape (reference routes /v1/audio/speech to the

@@ -123,6 +123,18 @@ def build_app() -> FastAPI:
             request, "/v1/audio/transcriptions"
         )
 
+    @app.post("/v1/audio/translations")
+    async def translations(request: Request):
+        return await request_service.route_general_request(
+            request, "/v1/audio/translations"
+        )
+
+    @app.get("/v1/audio/voices")
+    async def voices(request: Request):
+        return await request_service.route_general_request(
+            request, "/v1/audio/voices"
+        )
+
     @app.post("/tokenize")
     async def tokenize(request: Request):
         return await request_service.route_general_request(

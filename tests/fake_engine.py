@@ -46,6 +46,10 @@ def build_fake_engine(
         fn = parts.get("file", ("?", b""))[0]
         return {"text": f"transcribed:{fn}", "task": "transcribe"}
 
+    @app.get("/v1/audio/voices")
+    async def voices():
+        return {"voices": ["alloy", "echo"]}
+
     @app.post("/v1/audio/speech")
     async def speech(request: Request):
         body = await request.json()

@@ -710,3 +710,11 @@ def test_stream_options_include_usage_false_suppresses_usage():
         assert not saw_usage
 
     with_server(go)
+
+
+def test_audio_voices_listing():
+    async def go(client):
+        r = await client.get("/v1/audio/voices", timeout=30)
+        assert r.status_code == 200 and "alloy" in r.json()["voices"]
+
+    with_server(go)

@@ -476,3 +476,19 @@ def test_alias_rewrites_model_in_forwarded_body(engines):
         assert bodies, "no backend saw the resolved model name"
 
     asyncio.run(go())
+
+
+def test_audio_voices_and_translations_proxied(engines):
+    app = make_app(engines)
+
+    async def go():
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://router"
+                ) as client:
+                    r = await client.get("/v1/audio/voices", timeout=30)
+                    assert r.status_code == 200
+                    assert "alloy" in r.json()["voices"]
+
+    asyncio.run(go())
