@@ -358,6 +358,13 @@ def initialize_all(app: FastAPI, args) -> None:
         parse_static_urls,
     )
 
+    # reference-compatible alias for the discovery mode
+    if getattr(args, "k8s_service_discovery_type", None):
+        args.service_discovery = (
+            "k8s_service_name"
+            if args.k8s_service_discovery_type == "service-name"
+            else "k8s"
+        )
     if args.service_discovery == "static":
         sd.initialize_service_discovery(
             "static",
@@ -384,6 +391,8 @@ def initialize_all(app: FastAPI, args) -> None:
             api_key=args.api_key,
             insecure_skip_tls_verify=getattr(
                 args, "k8s_insecure_skip_tls_verify", False),
+            watcher_timeout_seconds=getattr(
+                args, "k8s_watcher_timeout_seconds", 30),
         )
         app.state.model_aliases = {}
     elif args.service_discovery == "k8s_service_name":
@@ -437,6 +446,8 @@ def initialize_all(app: FastAPI, args) -> None:
             sentry_sdk.init(
                 dsn=args.sentry_dsn,
                 traces_sample_rate=args.sentry_traces_sample_rate,
+                profile_session_sample_rate=getattr(
+                    args, "sentry_profile_session_sample_rate", 0.0),
             )
         except ImportError:
             logger.warning("--sentry-dsn set but sentry_sdk not installed")

@@ -40,6 +40,14 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     p.add_argument("--k8s-namespace", default="default")
     p.add_argument("--k8s-port", type=int, default=8000)
     p.add_argument("--k8s-label-selector", default=None)
+    p.add_argument("--k8s-service-discovery-type", default=None,
+                   choices=[None, "pod-ip", "service-name"],
+                   help="reference-compatible alias: pod-ip == "
+                        "--service-discovery k8s, service-name == "
+                        "k8s_service_name")
+    p.add_argument("--k8s-watcher-timeout-seconds", type=int, default=30,
+                   help="server-side timeoutSeconds for the pod watch "
+                        "stream (reconnects after)")
     p.add_argument("--k8s-insecure-skip-tls-verify", action="store_true",
                    help="explicitly disable K8s API TLS verification "
                         "(test clusters only; never silently implied)")
@@ -114,6 +122,8 @@ def parse_args(argv: Optional[list] = None) -> argparse.Namespace:
     # observability / experimental
     p.add_argument("--sentry-dsn", default=None)
     p.add_argument("--sentry-traces-sample-rate", type=float, default=0.0)
+    p.add_argument("--sentry-profile-session-sample-rate", type=float,
+                   default=0.0)
     p.add_argument("--otel-endpoint", default=None)
     p.add_argument("--otel-service-name", default="vllm-router")
     p.add_argument("--otel-secure", action="store_true")

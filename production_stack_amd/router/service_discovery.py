@@ -280,7 +280,9 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
         sa_token: Optional[str] = None,
         probe_models: bool = True,
         insecure_skip_tls_verify: bool = False,
+        watcher_timeout_seconds: int = 30,
     ) -> None:
+        self.watcher_timeout_seconds = max(int(watcher_timeout_seconds), 1)
         self.namespace = namespace
         self.port = port
         self.label_selector = label_selector
@@ -397,14 +399,15 @@ class K8sPodIpServiceDiscovery(ServiceDiscovery):
                     self._list_once()
                     need_list = False
                 sep = "&" if "?" in self._pods_url() else "?"
+                wt = self.watcher_timeout_seconds
                 url = (
                     f"{self._pods_url()}{sep}watch=true"
                     f"&resourceVersion={self._resource_version}"
-                    f"&timeoutSeconds=30"
+                    f"&timeoutSeconds={wt}"
                 )
                 with requests.get(
                     url, headers=self._headers(), verify=self._verify,
-                    stream=True, timeout=40,
+                    stream=True, timeout=wt + 10,
                 ) as r:
                     if r.status_code == 410:  # resourceVersion expired
                         need_list = True
@@ -463,7 +466,9 @@ class K8sServiceNameServiceDiscovery(ServiceDiscovery):
         sa_token: Optional[str] = None,
         probe_models: bool = True,
         insecure_skip_tls_verify: bool = False,
+        watcher_timeout_seconds: int = 30,
     ) -> None:
+        self.watcher_timeout_seconds = max(int(watcher_timeout_seconds), 1)
         self.namespace = namespace
         self.port = port
         self.label_selector = label_selector

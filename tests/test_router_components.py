@@ -352,3 +352,27 @@ def test_model_info_preserves_extra_fields():
     assert back["max_model_len"] == 8192
     assert back["permission"] == [{"id": "p1"}]
     assert back["id"] == "m" and back["created"] == 5
+
+
+def test_k8s_discovery_type_alias_and_watch_timeout():
+    """Reference-compatible --k8s-service-discovery-type alias maps onto
+    the discovery modes; --k8s-watcher-timeout-seconds reaches the pod
+    watcher."""
+    from production_stack_amd.router.parser import parse_args
+    from production_stack_amd.router.service_discovery import (
+        K8sPodIpServiceDiscovery,
+    )
+
+    args = parse_args([
+        "--service-discovery", "k8s",
+        "--k8s-service-discovery-type", "service-name",
+    ])
+    assert args.k8s_service_discovery_type == "service-name"
+    d = K8sPodIpServiceDiscovery(
+        api_base="http://127.0.0.1:1", sa_token="t", probe_models=False,
+        watcher_timeout_seconds=7,
+    )
+    try:
+        assert d.watcher_timeout_seconds == 7
+    finally:
+        d.close() if hasattr(d, "close") else None
