@@ -72,6 +72,7 @@ setup(
             "ps-cacheserver=production_stack_amd.kvpool.cacheserver:main",
             "ps-endpoint-picker="
             "production_stack_amd.gateway.picker_service:main",
+            "ps-extproc=production_stack_amd.gateway.extproc:main",
         ]
     },
     ext_modules=[ext, gw_ext],
