@@ -432,7 +432,11 @@ class ModelRunner:
         for i, ss in enumerate(scheduled):
             seq = ss.seq
             pos = seq.num_computed
-            tok = seq.token_ids()[pos]
+            npr = seq.num_prompt
+            # direct read (token_ids() copies prompt+output per seq per
+            # step — this is the graph-fill hot loop)
+            tok = (seq.prompt_token_ids[pos] if pos < npr
+                   else seq.output_token_ids[pos - npr])
             if tok < 0:  # async placeholder: gather from prev device toks
                 row = self._prev_rows.get(seq.request_id)
                 if row is None or self._prev_tokens_dev is None:
