@@ -245,8 +245,13 @@ class Scheduler:
         k = min(k, self.max_model_len - seq.num_tokens - 1)
         if k <= 0:
             return [], None
-        ids = seq.token_ids()
-        win = ids[-1024:]
+        out_ids = seq.output_token_ids
+        if len(out_ids) >= 1024:
+            win = out_ids[-1024:]
+        else:  # avoid materialising the full prompt+output list
+            win = seq.prompt_token_ids[
+                max(len(seq.prompt_token_ids) - (1024 - len(out_ids)),
+                    0):] + out_ids
         arr = np.asarray(win, dtype=np.int32).tobytes()
         for n in range(self.config.ngram_max, self.config.ngram_min - 1, -1):
             if len(win) <= n:
