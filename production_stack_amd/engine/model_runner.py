@@ -810,6 +810,10 @@ class ModelRunner:
                                device=logits.device)
             row = torch.full_like(logits[i], float("-inf"))
             row[idx] = logits[i][idx]
+            if not torch.isfinite(row[idx]).any():
+                # e.g. min_tokens suppressed EOS while the grammar
+                # completed: the grammar wins (uniform over allowed)
+                row[idx] = 0.0
             logits[i] = row
         return logits
 

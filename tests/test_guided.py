@@ -340,3 +340,20 @@ def test_guided_choice_end_to_end():
                    if t != tok.eos_token_id)
     assert reason == "stop"
     assert text in ("true", "12", "[]"), text
+
+
+def test_guided_with_min_tokens_does_not_deadlock():
+    """min_tokens suppresses EOS; once the grammar completes only EOS is
+    legal — the grammar must win instead of producing an all--inf row."""
+    eng = _engine()
+    p = SamplingParams(max_tokens=24, temperature=0.0, min_tokens=20,
+                       response_format={"type": "json_object"})
+    eng.add_request("gm", [15, 16], p)
+    reason = None
+    for _ in range(60):
+        for out in eng.step():
+            if out.request_id == "gm" and out.finished:
+                reason = out.finish_reason
+        if reason:
+            break
+    assert reason in ("stop", "length")
