@@ -110,7 +110,10 @@ int main(int argc, char** argv) {
     if (actions)
       fprintf(stderr, "[psoperator] applied %d change(s)\n", actions);
     if (once) break;
-    sleep(interval);
+    if (lead)
+      psop::watch_or_sleep(ctx, interval);  // event-driven when possible
+    else
+      sleep(interval);
   }
   return 0;
 }

@@ -14,6 +14,8 @@
 #include <algorithm>
 #include <cstdio>
 #include <cstdlib>
+#include <ctime>
+#include <unistd.h>
 #include <functional>
 #include <map>
 #include <vector>
@@ -848,6 +850,17 @@ int reconcile_loraadapters(const Ctx& ctx) {
 }
 
 }  // namespace
+
+void watch_or_sleep(const Ctx& ctx, int interval) {
+  time_t t0 = time(nullptr);
+  std::string url = crd_path(ctx, "vllmruntimes") +
+      "?watch=true&timeoutSeconds=" + std::to_string(interval);
+  auto r = try_request("GET", url, "", ctx.token);
+  if (r.status == 200 && !r.body.empty())
+    return;  // change event: reconcile now
+  int elapsed = (int)(time(nullptr) - t0);
+  if (elapsed < interval) sleep(interval - elapsed);
+}
 
 int reconcile_all(const Ctx& ctx) {
   int n = 0;

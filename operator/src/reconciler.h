@@ -18,6 +18,12 @@ struct Ctx {
 // (creates/updates) performed.
 int reconcile_all(const Ctx& ctx);
 
+// Event-driven pacing: block on a watch of the primary CR group for up
+// to `interval` seconds; returns immediately when a change event
+// arrives (next reconcile picks it up), otherwise sleeps out the
+// remainder (spin-safe against apiservers without watch support).
+void watch_or_sleep(const Ctx& ctx, int interval);
+
 // Exposed for tests: builders from CR -> desired child objects.
 psjson::ValuePtr build_engine_deployment(const Ctx& ctx,
                                          const psjson::ValuePtr& cr);
