@@ -136,6 +136,16 @@ class PrefixAwareRouter(RoutingInterface):
         chunk_size: int = 128,
         **_: Any,
     ) -> None:
+        # compiled C++ trie (csrc/gateway_pickers.cpp) walks without the
+        # GIL on the routing hot path; the asyncio HashTrie is the
+        # fallback for source-only checkouts (identical semantics)
+        self._native = None
+        try:
+            from production_stack_amd import _gwpick
+
+            self._native = _gwpick.PrefixTrie(chunk_size)
+        except ImportError:
+            pass
         self.trie = HashTrie(chunk_size=chunk_size)
         self.min_match = prefix_min_match_length
 
@@ -147,14 +157,22 @@ class PrefixAwareRouter(RoutingInterface):
             return _qps_min_endpoint(endpoints, request_stats)
         text = extract_prompt_text(request_json)
         available = {ep.url for ep in endpoints}
-        matched, candidates = await self.trie.longest_prefix_match(
-            text, available
-        )
+        if self._native is not None:
+            matched, candidates = self._native.longest_prefix_match(
+                text, list(available)
+            )
+        else:
+            matched, candidates = await self.trie.longest_prefix_match(
+                text, available
+            )
         if matched < self.min_match or not candidates:
             url = _qps_min_endpoint(endpoints, request_stats)
         else:
             url = random.choice(sorted(candidates))
-        await self.trie.insert(text, url)
+        if self._native is not None:
+            self._native.insert(text, url)
+        else:
+            await self.trie.insert(text, url)
         return url
 
 
