@@ -628,16 +628,22 @@ def build_server(engine: LLMEngine, served_model: str,
             r_i = rid if idx == 0 else f"{rid}-{idx}"
             text, toks, lps = "", [], []
             reason, npr = None, len(prompt_tokens)
-            async for out in async_engine.generate(
-                r_i, prompt_tokens, p_i, mm_embeds=mm_embeds
-            ):
-                text += out.text_delta
-                toks.extend(out.new_token_ids)
-                if out.new_logprobs:
-                    lps.extend(out.new_logprobs)
-                if out.finished:
-                    reason = out.finish_reason
-                    npr = out.num_prompt_tokens or npr
+            try:
+                async for out in async_engine.generate(
+                    r_i, prompt_tokens, p_i, mm_embeds=mm_embeds
+                ):
+                    text += out.text_delta
+                    toks.extend(out.new_token_ids)
+                    if out.new_logprobs:
+                        lps.extend(out.new_logprobs)
+                    if out.finished:
+                        reason = out.finish_reason
+                        npr = out.num_prompt_tokens or npr
+            except asyncio.CancelledError:
+                # client disconnected mid-generation: free the engine
+                # slot instead of decoding tokens nobody will read
+                async_engine.abort(r_i)
+                raise
             return text, toks, lps, reason, npr
 
         results = await asyncio.gather(

@@ -718,3 +718,34 @@ def test_audio_voices_listing():
         assert r.status_code == 200 and "alloy" in r.json()["voices"]
 
     with_server(go)
+
+
+def test_non_stream_disconnect_aborts_request():
+    """Cancelling a buffered (non-stream) completion frees the engine
+    request (the SSE path already did this)."""
+    async def go(client):
+        task = asyncio.ensure_future(client.post(
+            "/v1/completions",
+            json={"model": "tiny-llama", "prompt": "w1 w2",
+                  "max_tokens": 400, "temperature": 0,
+                  "ignore_eos": True},
+            timeout=120,
+        ))
+        await asyncio.sleep(0.3)  # let it schedule
+        task.cancel()
+        try:
+            await task
+        except (asyncio.CancelledError, Exception):
+            pass
+        eng = None
+        from production_stack_amd.engine import server as _srv  # noqa
+        # the app under test:
+        # poll until the scheduler drains (abort propagated)
+        app_engine = client._transport.app.state.engine
+        for _ in range(100):
+            if not app_engine.scheduler.has_unfinished():
+                break
+            await asyncio.sleep(0.05)
+        assert not app_engine.scheduler.has_unfinished()
+
+    with_server(go)
