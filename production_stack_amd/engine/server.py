@@ -93,7 +93,8 @@ def _parse_logprobs(body: dict):
 
 
 def build_server(engine: LLMEngine, served_model: str,
-                 chat_template: Optional[str] = None) -> FastAPI:
+                 chat_template: Optional[str] = None,
+                 extra_model_names: Optional[list] = None) -> FastAPI:
     # custom chat template (--chat-template FILE): vLLM-compatible jinja2
     # rendering with messages/tools/add_generation_prompt; falls back to
     # the built-in role-tagged template (tokenizer.render_chat)
@@ -190,11 +191,12 @@ def build_server(engine: LLMEngine, served_model: str,
     async def models():
         data = [
             {
-                "id": served_model,
+                "id": name,
                 "object": "model",
                 "created": int(app.state.start_time),
                 "owned_by": "production-stack-amd",
             }
+            for name in [served_model] + list(extra_model_names or [])
         ]
         for name in app.state.lora_adapters:
             data.append(
@@ -1124,7 +1126,9 @@ def main() -> None:
     ap.add_argument("model", nargs="?", default="llama-3-8b")
     ap.add_argument("--host", default="0.0.0.0")
     ap.add_argument("--port", type=int, default=8000)
-    ap.add_argument("--served-model-name", default=None)
+    ap.add_argument("--served-model-name", default=None, nargs="+",
+                    help="one or more public names for this model "
+                         "(vLLM-style; first is primary, the rest alias)")
     ap.add_argument("--max-model-len", type=int, default=4096)
     ap.add_argument("--dtype", default="bfloat16")
     ap.add_argument("--max-num-seqs", type=int, default=256)
@@ -1258,12 +1262,19 @@ def main() -> None:
         logging.basicConfig(level=logging.INFO)
         engine.run_pp_worker()
         return
-    served = args.served_model_name or args.model
+    smn = args.served_model_name
+    if isinstance(smn, list):
+        served = smn[0] if smn else args.model
+        extra_names = smn[1:]
+    else:
+        served = smn or args.model
+        extra_names = []
     tpl_src = None
     if args.chat_template:
         with open(args.chat_template) as f:
             tpl_src = f.read()
-    app = build_server(engine, served, chat_template=tpl_src)
+    app = build_server(engine, served, chat_template=tpl_src,
+                       extra_model_names=extra_names)
     logging.basicConfig(level=logging.INFO)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 

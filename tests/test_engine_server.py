@@ -749,3 +749,31 @@ def test_non_stream_disconnect_aborts_request():
         assert not app_engine.scheduler.has_unfinished()
 
     with_server(go)
+
+
+def test_multiple_served_model_names():
+    from production_stack_amd.engine.config import (
+        CacheConfig as CC,
+        EngineConfig as EC,
+        SchedulerConfig as SC,
+    )
+    from production_stack_amd.engine.engine import LLMEngine as LE
+
+    eng = LE(EC(model="tiny-llama", max_model_len=256,
+                cache=CC(num_gpu_blocks=32, block_size=16),
+                scheduler=SC(max_num_seqs=2,
+                             max_num_batched_tokens=64)),
+             device="cpu")
+    app = build_server(eng, "primary", extra_model_names=["alias-a"])
+
+    async def go():
+        async with httpx.ASGITransport(app=app) as transport:
+            async with app.router.lifespan_context(app):
+                async with httpx.AsyncClient(
+                    transport=transport, base_url="http://e"
+                ) as client:
+                    r = await client.get("/v1/models")
+                    ids = [m["id"] for m in r.json()["data"]]
+                    assert ids[:2] == ["primary", "alias-a"]
+
+    asyncio.run(go())
