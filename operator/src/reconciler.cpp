@@ -856,9 +856,13 @@ void watch_or_sleep(const Ctx& ctx, int interval) {
   std::string url = crd_path(ctx, "vllmruntimes") +
       "?watch=true&timeoutSeconds=" + std::to_string(interval);
   auto r = try_request("GET", url, "", ctx.token);
-  if (r.status == 200 && !r.body.empty())
-    return;  // change event: reconcile now
   int elapsed = (int)(time(nullptr) - t0);
+  // a real watch stream delivers {"type":"ADDED"|...} event frames; a
+  // server without watch support echoes a plain list immediately —
+  // treat only genuine events as a wake-up so we never hot-spin
+  bool event = r.status == 200 && !r.body.empty() &&
+               r.body.find("\"type\"") != std::string::npos;
+  if (event) return;  // change event: reconcile now
   if (elapsed < interval) sleep(interval - elapsed);
 }
 
