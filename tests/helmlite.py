@@ -408,6 +408,11 @@ _FUNCS = {
     "trunc": lambda r, d, v, n, x: _render_scalar(x)[:int(n)],
     "trimSuffix": lambda r, d, v, suf, x: _render_scalar(x).rstrip(suf),
     "index": lambda r, d, v, c, *keys: _index(c, keys),
+    "int": lambda r, d, v, x: int(float(x)) if x not in (UNDEF, None, "")
+    else 0,
+    "float": lambda r, d, v, x: float(x) if x not in (UNDEF, None, "")
+    else 0.0,
+    "toString": lambda r, d, v, x: _render_scalar(x),
 }
 
 

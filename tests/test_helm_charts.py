@@ -73,3 +73,56 @@ def test_values_schema_accepts_defaults_and_examples():
         with open(example) as f:
             overrides = yaml.safe_load(f) or {}
         jsonschema.validate(helmlite.load_values(overrides), schema)
+
+
+def test_kitchen_sink_render_all_features_on():
+    """Every optional subsystem enabled at once must render cleanly and
+    produce each expected kind (cross-template conflict check)."""
+    values = helmlite.load_values({
+        "servingEngineSpec": {"modelSpec": [{
+            "name": "llama3", "repository": "r", "tag": "t",
+            "modelURL": "llama-3-8b", "replicaCount": 2,
+            "requestCPU": 8, "requestMemory": "32Gi", "requestGPU": 1,
+            "pvcStorage": "100Gi",
+            "vllmConfig": {"maxModelLen": 8192,
+                           "tensorParallelSize": 2,
+                           "kvCacheDtype": "fp8"},
+            "lmcacheConfig": {"enabled": True,
+                              "cpuOffloadingBufferSize": "30"},
+            "keda": {"enabled": True, "minReplicaCount": 0,
+                     "maxReplicaCount": 4},
+            "raySpec": {"enabled": True},
+            "tolerations": [{"key": "amd.com/gpu",
+                             "operator": "Exists"}],
+        }]},
+        "routerSpec": {
+            "enableRouter": True,
+            "routingLogic": "prefixaware",
+            "hpa": {"enabled": True},
+            "ingress": {"enabled": True,
+                        "hosts": [{"host": "llm.example.com",
+                                   "paths": []}]},
+            "route": {"main": {"enabled": True,
+                               "parentRefs": [{"name": "gw"}]}},
+            "otel": {"enabled": True, "endpoint": "otel:4317"},
+        },
+        "cacheserverSpec": {"enabled": True},
+        "loraController": {"enableLoraController": True},
+        "loraAdapters": [{"name": "ad1", "baseModel": "llama3"}],
+        "sharedPvcStorage": {"enabled": True},
+        "serviceMonitor": {"enabled": True},
+        "runtimeClassName": "amd-gpu",
+        "extraObjects": [{"apiVersion": "v1", "kind": "ConfigMap",
+                          "metadata": {"name": "extra"}}],
+    })
+    kinds = []
+    for template in TEMPLATES:
+        for d in helmlite.render_template(template, values):
+            assert isinstance(d, dict) and "kind" in d, (template, d)
+            kinds.append(d["kind"])
+    for want in ("Deployment", "Service", "PersistentVolumeClaim",
+                 "ScaledObject", "HorizontalPodAutoscaler", "Ingress",
+                 "HTTPRoute", "ServiceMonitor", "ConfigMap",
+                 "ServiceAccount", "Role", "RoleBinding",
+                 "PodDisruptionBudget", "LoraAdapter"):
+        assert want in kinds, (want, sorted(set(kinds)))
