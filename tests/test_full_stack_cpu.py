@@ -142,3 +142,65 @@ def test_router_over_two_real_engines(engines):
     # both engines actually served traffic
     served = [s.engine.stats.num_requests for s in engines]
     assert all(n > 0 for n in served), served
+
+
+def test_new_apis_through_router(engines):
+    """Round-2-late APIs end to end through the router over real
+    engines: Anthropic messages, tool calling, structured outputs,
+    TTS."""
+    import io
+    import wave
+
+    argv = [
+        "--service-discovery", "static",
+        "--static-backends", ",".join(s.url for s in engines),
+        "--static-models", "tiny-llama",
+        "--routing-logic", "roundrobin",
+    ]
+    args = parse_args(argv)
+    application = router_app_mod.build_app()
+    router_app_mod.initialize_all(application, args)
+
+    async def go():
+        router_app_mod._http_session = None
+        transport = httpx.ASGITransport(app=application)
+        async with httpx.AsyncClient(
+            transport=transport, base_url="http://router", timeout=60
+        ) as client:
+            r = await client.post("/v1/messages", json={
+                "model": "tiny-llama", "max_tokens": 4,
+                "messages": [{"role": "user", "content": "hi"}],
+                "temperature": 0, "ignore_eos": True,
+            })
+            assert r.status_code == 200, r.text
+            assert r.json()["type"] == "message"
+
+            r = await client.post("/v1/chat/completions", json={
+                "model": "tiny-llama", "max_tokens": 3,
+                "temperature": 0,
+                "messages": [{"role": "user", "content": "hi"}],
+                "tools": [{"type": "function",
+                           "function": {"name": "noop",
+                                        "parameters": {}}}],
+                "ignore_eos": True,
+            })
+            assert r.status_code == 200, r.text
+            msg = r.json()["choices"][0]["message"]
+            assert msg["role"] == "assistant"
+
+            r = await client.post("/v1/chat/completions", json={
+                "model": "tiny-llama", "max_tokens": 6,
+                "temperature": 0,
+                "messages": [{"role": "user", "content": "pick"}],
+                "response_format": {"type": "json_object"},
+            })
+            assert r.status_code == 200, r.text
+            assert r.json()["choices"][0]["finish_reason"] == "stop"
+
+            r = await client.post("/v1/audio/speech", json={
+                "model": "tiny-llama", "input": "w1 w2 w3"})
+            assert r.status_code == 200
+            with wave.open(io.BytesIO(r.content)) as w:
+                assert w.getnframes() > 0
+
+    asyncio.run(go())
