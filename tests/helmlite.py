@@ -568,6 +568,12 @@ def run_unittest_spec(spec_path: str) -> List[str]:
                 if not isinstance(got, list) or want not in got:
                     fail(f"contains {a['contains']['path']}: {want!r} "
                          f"not in {got!r}")
+            elif "notContains" in a:
+                got = _get_path(doc, a["notContains"]["path"])
+                want = a["notContains"]["content"]
+                if isinstance(got, list) and want in got:
+                    fail(f"notContains {a['notContains']['path']}: "
+                         f"{want!r} present")
             elif "exists" in a:
                 if _get_path(doc, a["exists"]["path"]) is None:
                     fail(f"exists {a['exists']['path']}: missing")
@@ -586,4 +592,7 @@ def run_unittest_spec(spec_path: str) -> List[str]:
                 if not re.search(a["matchRegex"]["pattern"],
                                  str(got or "")):
                     fail(f"matchRegex {a['matchRegex']['path']}")
+            else:
+                # an unknown assertion type must never silently pass
+                fail(f"unsupported assertion {sorted(a.keys())}")
     return failures
