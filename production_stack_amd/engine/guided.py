@@ -432,15 +432,13 @@ def choice_regex(choices: List[str]) -> str:
 # ---------------------------------------------------------------------------
 # token-level guided state
 
-_TOKEN_TEXT_CACHE: dict = {}
-
-
 def _token_texts(tokenizer, vocab_size: int) -> List[str]:
-    key = id(tokenizer)
-    got = _TOKEN_TEXT_CACHE.get(key)
+    # cached on the tokenizer instance (an id()-keyed global can alias
+    # a freed tokenizer's address after GC)
+    got = getattr(tokenizer, "_guided_token_texts", None)
     if got is None or len(got) < vocab_size:
         got = [None] * vocab_size
-        _TOKEN_TEXT_CACHE[key] = got
+        tokenizer._guided_token_texts = got
     return got
 
 
