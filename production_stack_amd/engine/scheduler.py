@@ -238,7 +238,14 @@ class Scheduler:
         sequence's own history proposes the tokens that followed it
         (byte-encoded rfind so the scan is C-speed)."""
         if self.draft_proposer is not None:
-            return self.draft_proposer.propose(seq, self.bm)
+            try:
+                return self.draft_proposer.propose(seq, self.bm)
+            except Exception:  # draft failure must never kill serving
+                import logging
+
+                logging.getLogger("engine.scheduler").exception(
+                    "draft proposer failed; plain decode")
+                return [], None
         import numpy as np
 
         k = self.config.num_speculative_tokens
