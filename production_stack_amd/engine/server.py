@@ -884,8 +884,12 @@ def build_server(engine: LLMEngine, served_model: str,
         seg = max(int(sr * 0.04 / speed), 1)
         tokens = engine.tokenizer.encode(text) or [0]
         samples = []
+        import zlib
+
+        voice = str(body.get("voice", "alloy"))
         for t in tokens[:2048]:
-            f = 110.0 + (hash((t, body.get("voice", "alloy"))) % 520)
+            # stable across processes (hash() is seed-randomized)
+            f = 110.0 + (zlib.crc32(f"{t}:{voice}".encode()) % 520)
             for i in range(seg):
                 env = min(i, seg - i, seg // 8 + 1) / (seg // 8 + 1)
                 samples.append(int(
