@@ -68,7 +68,22 @@ def build_picker_app(
         from production_stack_amd.kvpool.client import ControllerClient
 
         text = extract_prompt_text(body or {})
-        tokens = [hash(w) % 50000 for w in text.split()]
+        # tokenize with the ENGINE's tokenizer (its /tokenize endpoint)
+        # so the controller lookup compares real token ids; a local
+        # stand-in hash would never match the registered prefixes
+        tokens = None
+        try:
+            import httpx
+
+            async with httpx.AsyncClient(timeout=5) as hc:
+                tr = await hc.post(endpoints[0] + "/tokenize",
+                                   json={"prompt": text})
+                if tr.status_code == 200:
+                    tokens = tr.json().get("tokens")
+        except Exception:
+            tokens = None
+        if not tokens:
+            return await pick_roundrobin(endpoints, body)
         try:
             if kv_client["c"] is None:
                 kv_client["c"] = ControllerClient(
