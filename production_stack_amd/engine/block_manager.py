@@ -64,7 +64,12 @@ class BlockManager:
 
     @staticmethod
     def chain_hash(prev: Optional[int], tokens: tuple) -> int:
-        return hash((prev, tokens))
+        # prev=None normalizes to 0: hash(None) is address-derived on
+        # CPython < 3.12, so a None seed would make the chain differ
+        # ACROSS PROCESSES — breaking kvaware lookup (engine registers,
+        # controller matches) and cross-instance prefix exchange.
+        # int/tuple-of-int hashing is seed-stable.
+        return hash((prev if prev is not None else 0, tokens))
 
     # ------------------------------------------------------------------
     def blocks_needed(self, num_tokens: int) -> int:

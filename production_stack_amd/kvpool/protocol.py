@@ -28,10 +28,13 @@ async def recv_msg(reader: asyncio.StreamReader) -> Any:
 
 def chain_hashes(token_ids, block_size: int = 16):
     """Chain hash per full block — MUST match BlockManager.chain_hash.
-    Deterministic across processes (int-tuple hashing is not seed-randomized).
-    """
+    The seed for the first block is 0, NOT None: hash(None) is
+    address-derived on CPython < 3.12 and differs across processes,
+    which would silently break every cross-process prefix match
+    (kvaware lookups, remote-tier keys). int/tuple-of-int hashing is
+    stable regardless of PYTHONHASHSEED."""
     out = []
-    prev = None
+    prev = 0
     for i in range(len(token_ids) // block_size):
         h = hash((prev, tuple(token_ids[i * block_size : (i + 1) * block_size])))
         out.append(h)

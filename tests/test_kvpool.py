@@ -516,3 +516,34 @@ def test_remote_tier_cachegen_serde_roundtrip():
     assert got_slot is not None
     assert torch.equal(pool.store[got_slot].view(torch.int8),
                        want.view(torch.int8))
+
+
+def test_chain_hashes_stable_across_processes():
+    """The prefix chain hash must be identical in a separate interpreter
+    (hash(None) is address-derived on CPython < 3.12: a None seed broke
+    every cross-process prefix match — engine registers, controller
+    never hits)."""
+    import os
+    import subprocess
+    import sys
+
+    from production_stack_amd.engine.block_manager import BlockManager
+    from production_stack_amd.kvpool.protocol import chain_hashes
+
+    toks = list(range(48))
+    local = chain_hashes(toks, 16)
+    # BlockManager's chain agrees with the protocol helper
+    prev, bm_chain = None, []
+    for i in range(3):
+        prev = BlockManager.chain_hash(prev, tuple(toks[i * 16:(i + 1) * 16]))
+        bm_chain.append(prev)
+    assert bm_chain == local
+    out = subprocess.run(
+        [sys.executable, "-c",
+         "from production_stack_amd.kvpool.protocol import chain_hashes;"
+         "print(chain_hashes(list(range(48)), 16))"],
+        capture_output=True, text=True,
+        cwd=os.path.join(os.path.dirname(__file__), ".."),
+    )
+    assert out.returncode == 0, out.stderr
+    assert eval(out.stdout.strip()) == local
