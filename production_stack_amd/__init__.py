@@ -16,4 +16,4 @@ Layers (see SURVEY.md for the reference blueprint):
               RCCL / xGMI.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

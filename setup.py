@@ -52,7 +52,7 @@ ext = CUDAExtension(
 
 setup(
     name="production-stack-amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=[
         "production_stack_amd",
         "production_stack_amd.engine",
