@@ -97,7 +97,12 @@ class BlockManager:
         table: List[int] = []
         matched = 0
         prev_hash: Optional[int] = None
-        if self.enable_prefix_caching:
+        # prompt_logprobs needs logits at every prompt position — a
+        # prefix-cache hit would skip those rows (vLLM also recomputes)
+        want_prompt_lp = (
+            getattr(seq.params, "prompt_logprobs", None) is not None
+        )
+        if self.enable_prefix_caching and not want_prompt_lp:
             n_full = n_prompt // bs
             # never reuse ALL tokens: the last one must be computed so there
             # are logits to sample from

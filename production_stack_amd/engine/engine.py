@@ -254,6 +254,10 @@ class LLMEngine:
             first = seq.first_token_time is None and bool(new)
             if first:
                 seq.first_token_time = now
+            plp = None
+            if first or seq.finished:
+                plp = self.runner.prompt_logprobs.pop(
+                    seq.request_id, None)
             self.stats.generation_tokens += len(new)
             # decode only up to the streaming cursor: under async
             # scheduling the tail may still hold -1 placeholders
@@ -281,6 +285,7 @@ class LLMEngine:
                     text_delta=text_delta,
                     finished=seq.finished,
                     finish_reason=reason,
+                    prompt_logprobs=plp,
                     num_prompt_tokens=seq.num_prompt,
                     num_output_tokens=len(seq.output_token_ids),
                     num_cached_tokens=seq.num_cached_prompt_tokens,

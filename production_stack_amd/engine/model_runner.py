@@ -89,6 +89,9 @@ class ModelRunner:
         # request_id -> logprob of the most recent sampled token (only for
         # requests that asked for logprobs; read by engine.step)
         self.last_logprobs: Dict[str, float] = {}
+        # request_id -> [None, lp1, lp2...] accumulated across prefill
+        # chunks for params.prompt_logprobs requests
+        self.prompt_logprobs: Dict[str, list] = {}
         self.spec_proposed = 0
         self.spec_accepted = 0
         # async-scheduling state (execute_async/finalize_async)
@@ -506,6 +509,9 @@ class ModelRunner:
                     logits_p = None
                     if token_t.numel():
                         hidden = self.model(token_t, meta, self.kv_caches)
+                        self._collect_prompt_logprobs(
+                            SchedulerOutput(scheduled=pre), hidden
+                        )
                         if sample_seqs:
                             logits_p = self.model.compute_logits(
                                 hidden[rows_t]
@@ -540,6 +546,7 @@ class ModelRunner:
         if self.tp_coord is not None and self.tp_coord.rank == 0 and                 getattr(self, "tp_serving", False):
             self.tp_coord.broadcast_step(token_t, meta)
         hidden = self.model(token_t, meta, self.kv_caches)
+        self._collect_prompt_logprobs(out, hidden)
         if not sample_seqs:
             return {}
         sel = hidden[rows_t]
@@ -610,6 +617,8 @@ class ModelRunner:
                 return None
             if getattr(p, "response_format", None):
                 return None  # guided masks need last step's token on host
+            if getattr(p, "prompt_logprobs", None) is not None:
+                return None  # collected on the sync prefill path
             if ss.draft_tokens:
                 return None
         dec = [ss for ss in out.scheduled if ss.is_decode]
@@ -710,6 +719,44 @@ class ModelRunner:
             seq.request_id: int(t)
             for seq, t in zip(handle["sample_seqs"], vals)
         }
+
+    def _collect_prompt_logprobs(self, out, hidden) -> None:
+        """vLLM prompt_logprobs: for requesting sequences, log-softmax
+        the prefill-chunk rows and gather each next prompt token.
+        Row j of a chunk starting at `start` predicts position
+        start+j+1; only prompt positions are collected. No-op (and no
+        extra compute) unless a scheduled seq asked for it."""
+        row = 0
+        for ss in out.scheduled:
+            if ss.is_decode:
+                continue
+            seq, n = ss.seq, ss.num_tokens
+            p = seq.params
+            if (getattr(p, "prompt_logprobs", None) is not None
+                    and not ss.draft_tokens):
+                start = seq.num_computed
+                npr = seq.num_prompt
+                count = max(0, min(n, npr - start - 1))
+                if count > 0:
+                    idx_rows = torch.arange(
+                        row, row + count, device=hidden.device
+                    )
+                    lg = torch.log_softmax(
+                        self.model.compute_logits(
+                            hidden.index_select(0, idx_rows)
+                        ).float(),
+                        dim=-1,
+                    )
+                    ids = torch.tensor(
+                        seq.prompt_token_ids[start + 1:start + 1 + count],
+                        dtype=torch.long, device=lg.device,
+                    )
+                    vals = lg.gather(1, ids.unsqueeze(1)).squeeze(1)
+                    acc = self.prompt_logprobs.setdefault(
+                        seq.request_id, [None]
+                    )
+                    acc.extend(float(v) for v in vals.cpu())
+            row += n
 
     def _collect_sampled(self, out, sample_seqs, tokens,
                          logits=None) -> Dict[str, object]:

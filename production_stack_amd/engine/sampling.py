@@ -24,6 +24,10 @@ class SamplingParams:
     min_tokens: int = 0
     # vLLM-style min_p: drop tokens with prob < min_p * max_prob
     min_p: float = 0.0
+    # vLLM-style prompt_logprobs: log-probs of each prompt token under
+    # the model (position 0 has no prediction -> None), computed during
+    # the prefill chunks; any non-None value enables it
+    prompt_logprobs: Optional[int] = None
     # OpenAI structured outputs: {"type": "json_object"} or
     # {"type": "json_schema", "json_schema": {...}} (engine/guided.py)
     response_format: Optional[dict] = None

@@ -62,6 +62,10 @@ def _params_from_body(body: dict, max_model_len: int) -> SamplingParams:
                     for k, v in (body.get("logit_bias") or {}).items()},
         min_tokens=int(body.get("min_tokens") or 0),
         min_p=float(body.get("min_p") or 0.0),
+        prompt_logprobs=(
+            int(body["prompt_logprobs"])
+            if body.get("prompt_logprobs") is not None else None
+        ),
         logprobs=_parse_logprobs(body),
         response_format=_guided_from_body(body),
     )
@@ -629,6 +633,7 @@ def build_server(engine: LLMEngine, served_model: str,
                 )
             r_i = rid if idx == 0 else f"{rid}-{idx}"
             text, toks, lps = "", [], []
+            plp = None
             reason, npr = None, len(prompt_tokens)
             try:
                 async for out in async_engine.generate(
@@ -636,6 +641,8 @@ def build_server(engine: LLMEngine, served_model: str,
                 ):
                     text += out.text_delta
                     toks.extend(out.new_token_ids)
+                    if out.prompt_logprobs is not None:
+                        plp = out.prompt_logprobs
                     if out.new_logprobs:
                         lps.extend(out.new_logprobs)
                     if out.finished:
@@ -646,14 +653,14 @@ def build_server(engine: LLMEngine, served_model: str,
                 # slot instead of decoding tokens nobody will read
                 async_engine.abort(r_i)
                 raise
-            return text, toks, lps, reason, npr
+            return text, toks, lps, reason, npr, plp
 
         results = await asyncio.gather(
             *[run_one(i) for i in range(n_choices)]
         )
-        text, tokens, _, finish_reason, n_prompt = results[0]
+        text, tokens, _, finish_reason, n_prompt, _plp0 = results[0]
         choices = []
-        for i, (txt, toks, lps, reason, _) in enumerate(results):
+        for i, (txt, toks, lps, reason, _, plp) in enumerate(results):
             if chat:
                 message = {"role": "assistant", "content": txt}
                 if tools:
@@ -688,6 +695,8 @@ def build_server(engine: LLMEngine, served_model: str,
                         "tokens": tok_strs,
                         "token_logprobs": lps,
                     }
+            if plp is not None:
+                choice["prompt_logprobs"] = plp
             choices.append(choice)
         total_out = sum(len(r[1]) for r in results)
         resp = {

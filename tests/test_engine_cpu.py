@@ -678,3 +678,46 @@ def test_speculative_stochastic_end_to_end():
     b = mk()
     b.runner.model.load_state_dict(a.runner.model.state_dict())
     assert b.generate([prompt], fresh_params())["offline-0"] == got
+
+
+def test_prompt_logprobs_chunking_invariant():
+    """params.prompt_logprobs: [None, lp...] over the prompt, identical
+    whether the prefill ran as one chunk or many."""
+    def run(batched):
+        cfg = EngineConfig(
+            model="tiny-llama",
+            max_model_len=256,
+            seed=4,
+            cache=CacheConfig(num_gpu_blocks=64, block_size=16),
+            scheduler=SchedulerConfig(
+                max_num_seqs=2, max_num_batched_tokens=batched),
+        )
+        eng = LLMEngine(cfg, device="cpu")
+        p = SamplingParams(max_tokens=2, temperature=0.0,
+                           ignore_eos=True, prompt_logprobs=1)
+        eng.add_request("plp", list(range(30, 70)), p)
+        got = None
+        while eng.has_unfinished():
+            for out in eng.step():
+                if out.prompt_logprobs is not None:
+                    got = out.prompt_logprobs
+        return eng, got
+
+    ref_eng, one = run(256)   # whole prompt in one chunk
+    assert one is not None and one[0] is None
+    assert len(one) == 40     # one entry per prompt position
+    assert all(v <= 0 for v in one[1:])
+    eng2, many = run(16)      # forced chunked prefill
+    eng2.runner.model.load_state_dict(ref_eng.runner.model.state_dict())
+    # rerun chunked with identical weights for the comparison
+    p = SamplingParams(max_tokens=2, temperature=0.0,
+                       ignore_eos=True, prompt_logprobs=1)
+    eng2.add_request("plp2", list(range(30, 70)), p)
+    got2 = None
+    while eng2.has_unfinished():
+        for out in eng2.step():
+            if out.prompt_logprobs is not None:
+                got2 = out.prompt_logprobs
+    assert got2 is not None
+    for a, b in zip(one[1:], got2[1:]):
+        assert abs(a - b) < 1e-3, (a, b)

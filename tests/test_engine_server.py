@@ -777,3 +777,20 @@ def test_multiple_served_model_names():
                     assert ids[:2] == ["primary", "alias-a"]
 
     asyncio.run(go())
+
+
+def test_prompt_logprobs_http():
+    async def go(client):
+        r = await client.post(
+            "/v1/completions",
+            json={"model": "tiny-llama", "prompt": "w1 w2 w3 w4",
+                  "max_tokens": 2, "temperature": 0,
+                  "prompt_logprobs": 1, "ignore_eos": True},
+            timeout=120,
+        )
+        assert r.status_code == 200, r.text
+        plp = r.json()["choices"][0]["prompt_logprobs"]
+        assert plp[0] is None and len(plp) == 4
+        assert all(v <= 0 for v in plp[1:])
+
+    with_server(go)
