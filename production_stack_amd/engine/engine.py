@@ -137,6 +137,8 @@ class LLMEngine:
 
     def abort_request(self, request_id: str) -> None:
         self.scheduler.abort(request_id)
+        self.runner.prompt_logprobs.pop(request_id, None)
+        self.runner.last_logprobs.pop(request_id, None)
 
     def has_unfinished(self) -> bool:
         return self.scheduler.has_unfinished() or self._pending is not None
