@@ -139,6 +139,7 @@ class LLMEngine:
         self.scheduler.abort(request_id)
         self.runner.prompt_logprobs.pop(request_id, None)
         self.runner.last_logprobs.pop(request_id, None)
+        self.runner.last_top_logprobs.pop(request_id, None)
 
     def has_unfinished(self) -> bool:
         return self.scheduler.has_unfinished() or self._pending is not None
@@ -244,7 +245,13 @@ class LLMEngine:
             if not new and not seq.finished:
                 continue
             lps = None
+            tops = None
             if seq.params.logprobs is not None and new:
+                top = self.runner.last_top_logprobs.pop(
+                    seq.request_id, None)
+                if top is not None:
+                    # one sampled position per drain on the sync path
+                    tops = [top] * len(new)
                 lp = self.runner.last_logprobs.pop(seq.request_id, None)
                 if isinstance(lp, list):
                     # speculative chunk: one value per accepted token
@@ -293,6 +300,7 @@ class LLMEngine:
                     num_cached_tokens=seq.num_cached_prompt_tokens,
                     first_token=first,
                     new_logprobs=lps,
+                    new_top_logprobs=tops,
                 )
             )
         return results

@@ -89,6 +89,9 @@ class ModelRunner:
         # request_id -> logprob of the most recent sampled token (only for
         # requests that asked for logprobs; read by engine.step)
         self.last_logprobs: Dict[str, float] = {}
+        # request_id -> [(token_id, logprob) x topN] for the latest
+        # sampled position (params.logprobs > 0 requests)
+        self.last_top_logprobs: Dict[str, list] = {}
         # request_id -> [None, lp1, lp2...] accumulated across prefill
         # chunks for params.prompt_logprobs requests
         self.prompt_logprobs: Dict[str, list] = {}
@@ -821,7 +824,16 @@ class ModelRunner:
             chosen = sampled[idx.cpu()].to(lp.device)
             vals = lp.gather(1, chosen.unsqueeze(1)).squeeze(1).cpu()
             for j, i in enumerate(want_lp):
-                self.last_logprobs[seqs[i].request_id] = float(vals[j])
+                rid = seqs[i].request_id
+                self.last_logprobs[rid] = float(vals[j])
+                n_top = params[i].logprobs or 0
+                if n_top > 0:
+                    k = min(int(n_top), 20, lp.shape[-1])
+                    tv, ti = lp[j].topk(k)
+                    self.last_top_logprobs[rid] = list(zip(
+                        (int(t) for t in ti.cpu()),
+                        (float(v) for v in tv.cpu()),
+                    ))
         return sampled
 
     def _apply_guided(

@@ -118,6 +118,9 @@ class RequestOutput:
     first_token: bool = False
     # per-token logprob of each entry in new_token_ids (params.logprobs)
     new_logprobs: Optional[List[float]] = None
+    # per-position top-N alternatives [(token_id, logprob), ...] aligned
+    # with new_token_ids (params.logprobs > 0)
+    new_top_logprobs: Optional[list] = None
     # [None, lp1, ...] for the prompt (params.prompt_logprobs), attached
     # once on the first output after prefill completes
     prompt_logprobs: Optional[list] = None

@@ -632,7 +632,7 @@ def build_server(engine: LLMEngine, served_model: str,
                     else None,
                 )
             r_i = rid if idx == 0 else f"{rid}-{idx}"
-            text, toks, lps = "", [], []
+            text, toks, lps, tops = "", [], [], []
             plp = None
             reason, npr = None, len(prompt_tokens)
             try:
@@ -645,6 +645,8 @@ def build_server(engine: LLMEngine, served_model: str,
                         plp = out.prompt_logprobs
                     if out.new_logprobs:
                         lps.extend(out.new_logprobs)
+                    if out.new_top_logprobs:
+                        tops.extend(out.new_top_logprobs)
                     if out.finished:
                         reason = out.finish_reason
                         npr = out.num_prompt_tokens or npr
@@ -653,14 +655,14 @@ def build_server(engine: LLMEngine, served_model: str,
                 # slot instead of decoding tokens nobody will read
                 async_engine.abort(r_i)
                 raise
-            return text, toks, lps, reason, npr, plp
+            return text, toks, lps, reason, npr, plp, tops
 
         results = await asyncio.gather(
             *[run_one(i) for i in range(n_choices)]
         )
-        text, tokens, _, finish_reason, n_prompt, _plp0 = results[0]
+        text, tokens, _, finish_reason, n_prompt = results[0][:5]
         choices = []
-        for i, (txt, toks, lps, reason, _, plp) in enumerate(results):
+        for i, (txt, toks, lps, reason, _, plp, tops) in enumerate(results):
             if chat:
                 message = {"role": "assistant", "content": txt}
                 if tools:
@@ -683,18 +685,39 @@ def build_server(engine: LLMEngine, served_model: str,
                 }
             if params.logprobs is not None and lps:
                 tok_strs = [engine.tokenizer.decode_token(t) for t in toks]
+
+                def top_fmt(pos):
+                    if pos >= len(tops) or not tops[pos]:
+                        return None
+                    return [
+                        {"token": engine.tokenizer.decode_token(t),
+                         "logprob": v}
+                        for t, v in tops[pos]
+                    ]
+
                 if chat:
                     choice["logprobs"] = {
                         "content": [
-                            {"token": ts, "logprob": lp}
-                            for ts, lp in zip(tok_strs, lps)
+                            {"token": ts, "logprob": lp,
+                             **({"top_logprobs": top_fmt(j)}
+                                if top_fmt(j) is not None else {})}
+                            for j, (ts, lp) in enumerate(
+                                zip(tok_strs, lps))
                         ]
                     }
                 else:
-                    choice["logprobs"] = {
+                    lgp = {
                         "tokens": tok_strs,
                         "token_logprobs": lps,
                     }
+                    if tops:
+                        lgp["top_logprobs"] = [
+                            {engine.tokenizer.decode_token(t): v
+                             for t, v in (tops[j] if j < len(tops)
+                                          else [])}
+                            for j in range(len(toks))
+                        ]
+                    choice["logprobs"] = lgp
             if plp is not None:
                 choice["prompt_logprobs"] = plp
             choices.append(choice)

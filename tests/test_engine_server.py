@@ -794,3 +794,38 @@ def test_prompt_logprobs_http():
         assert all(v <= 0 for v in plp[1:])
 
     with_server(go)
+
+
+def test_top_logprobs_alternatives():
+    """logprobs=N returns top-N alternatives per position in both API
+    shapes; the chosen token's logprob appears among (or equals the
+    max of) the alternatives."""
+    async def go(client):
+        r = await client.post(
+            "/v1/completions",
+            json={"model": "tiny-llama", "prompt": "w1 w2 w3",
+                  "max_tokens": 3, "temperature": 0, "logprobs": 3,
+                  "ignore_eos": True},
+            timeout=120,
+        )
+        assert r.status_code == 200, r.text
+        lg = r.json()["choices"][0]["logprobs"]
+        assert len(lg["top_logprobs"]) == 3
+        for pos, d in enumerate(lg["top_logprobs"]):
+            assert 1 <= len(d) <= 3
+            # greedy: the chosen token is the argmax -> max alternative
+            assert abs(max(d.values()) - lg["token_logprobs"][pos]) < 1e-5
+
+        r = await client.post(
+            "/v1/chat/completions",
+            json={"model": "tiny-llama",
+                  "messages": [{"role": "user", "content": "hi"}],
+                  "max_tokens": 2, "temperature": 0, "logprobs": True,
+                  "top_logprobs": 2, "ignore_eos": True},
+            timeout=120,
+        )
+        content = r.json()["choices"][0]["logprobs"]["content"]
+        assert all(len(e["top_logprobs"]) == 2 for e in content)
+        assert all("token" in e["top_logprobs"][0] for e in content)
+
+    with_server(go)
