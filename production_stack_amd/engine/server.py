@@ -596,6 +596,37 @@ def build_server(engine: LLMEngine, served_model: str,
                                 ),
                             }
                         first = False
+                        if params.logprobs is not None and out.new_logprobs:
+                            tok_strs = [
+                                engine.tokenizer.decode_token(t)
+                                for t in out.new_token_ids
+                            ]
+
+                            def _tops(j):
+                                tl = out.new_top_logprobs
+                                if not tl or j >= len(tl) or not tl[j]:
+                                    return None
+                                return [
+                                    {"token":
+                                     engine.tokenizer.decode_token(t),
+                                     "logprob": v}
+                                    for t, v in tl[j]
+                                ]
+
+                            if chat:
+                                choice["logprobs"] = {"content": [
+                                    {"token": ts, "logprob": lv,
+                                     **({"top_logprobs": _tops(j)}
+                                        if _tops(j) is not None else {})}
+                                    for j, (ts, lv) in enumerate(zip(
+                                        tok_strs, out.new_logprobs))
+                                ]}
+                            else:
+                                choice["logprobs"] = {
+                                    "tokens": tok_strs,
+                                    "token_logprobs": list(
+                                        out.new_logprobs),
+                                }
                         chunk = {
                             "id": rid,
                             "object": obj + ".chunk" if chat else obj,

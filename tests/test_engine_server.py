@@ -829,3 +829,29 @@ def test_top_logprobs_alternatives():
         assert all("token" in e["top_logprobs"][0] for e in content)
 
     with_server(go)
+
+
+def test_streamed_logprobs():
+    """Streaming with logprobs: each content chunk carries the token's
+    logprob (and top alternatives when requested)."""
+    async def go(client):
+        async with client.stream(
+            "POST", "/v1/chat/completions",
+            json={"model": "tiny-llama",
+                  "messages": [{"role": "user", "content": "hi"}],
+                  "max_tokens": 3, "temperature": 0, "stream": True,
+                  "logprobs": True, "top_logprobs": 2,
+                  "ignore_eos": True},
+            timeout=120,
+        ) as r:
+            entries = []
+            async for line in r.aiter_lines():
+                if line.startswith("data: ") and line != "data: [DONE]":
+                    c = json.loads(line[6:])["choices"][0]
+                    if c.get("logprobs"):
+                        entries.extend(c["logprobs"]["content"])
+        assert len(entries) == 3
+        assert all(e["logprob"] <= 0 for e in entries)
+        assert all(len(e["top_logprobs"]) == 2 for e in entries)
+
+    with_server(go)
