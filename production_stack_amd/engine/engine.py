@@ -249,9 +249,11 @@ class LLMEngine:
             if seq.params.logprobs is not None and new:
                 top = self.runner.last_top_logprobs.pop(
                     seq.request_id, None)
-                if top is not None:
-                    # one sampled position per drain on the sync path
-                    tops = [top] * len(new)
+                if top is not None and len(new) == 1:
+                    # alternatives belong to the latest sampled position
+                    # only; multi-token (speculative) drains skip them
+                    # rather than replicate a stale row
+                    tops = [top]
                 lp = self.runner.last_logprobs.pop(seq.request_id, None)
                 if isinstance(lp, list):
                     # speculative chunk: one value per accepted token

@@ -676,8 +676,11 @@ def build_server(engine: LLMEngine, served_model: str,
                         plp = out.prompt_logprobs
                     if out.new_logprobs:
                         lps.extend(out.new_logprobs)
-                    if out.new_top_logprobs:
-                        tops.extend(out.new_top_logprobs)
+                        tl = out.new_top_logprobs or []
+                        tops.extend(
+                            tl + [None] * (len(out.new_logprobs)
+                                           - len(tl))
+                        )
                     if out.finished:
                         reason = out.finish_reason
                         npr = out.num_prompt_tokens or npr
@@ -744,8 +747,8 @@ def build_server(engine: LLMEngine, served_model: str,
                     if tops:
                         lgp["top_logprobs"] = [
                             {engine.tokenizer.decode_token(t): v
-                             for t, v in (tops[j] if j < len(tops)
-                                          else [])}
+                             for t, v in ((tops[j] or [])
+                                          if j < len(tops) else [])}
                             for j in range(len(toks))
                         ]
                     choice["logprobs"] = lgp
