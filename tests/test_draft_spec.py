@@ -175,3 +175,30 @@ def test_spec_with_logprobs_matches_plain():
     assert len(got_lps) == len(want_lps) == 16
     for a, b in zip(got_lps, want_lps):
         assert abs(a - b) < 1e-4, (a, b)
+
+
+def test_spec_with_top_logprobs_alignment():
+    """logprobs=N + speculation: chosen-token logprobs stay one per
+    token; top alternatives attach only to single-token positions
+    (never replicated across an accepted chunk)."""
+    spec = mk("tiny-llama")
+    spec.scheduler.draft_proposer.load_target_weights(spec.runner.model)
+    p = SamplingParams(max_tokens=12, temperature=0.0, ignore_eos=True,
+                       logprobs=3)
+    spec.add_request("tl", PROMPT, p)
+    n_toks = n_lps = 0
+    tops_rows = []
+    while spec.has_unfinished():
+        for out in spec.step():
+            n_toks += len(out.new_token_ids)
+            if out.new_logprobs:
+                n_lps += len(out.new_logprobs)
+            if out.new_top_logprobs:
+                assert len(out.new_top_logprobs) <= len(
+                    out.new_token_ids)
+                tops_rows.extend(
+                    t for t in out.new_top_logprobs if t)
+    assert n_toks == n_lps == 12
+    assert spec.runner.spec_proposed > 0
+    for row in tops_rows:
+        assert 1 <= len(row) <= 3
