@@ -464,3 +464,27 @@ def test_mixtral_moe_gpu_matches_cpu():
     # early tokens are expected to agree
     agree = sum(a == b for a, b in zip(out_gpu[:4], out_cpu[:4]))
     assert agree >= 2, f"{out_gpu} vs {out_cpu}"
+
+
+def test_logprobs_and_prompt_logprobs_gpu():
+    """Sampled-token logprobs, top alternatives and prompt_logprobs on a
+    cuda engine: finite, correctly shaped, greedy-consistent (the chosen
+    token's logprob equals the max alternative)."""
+    eng = make_engine()
+    p = SamplingParams(max_tokens=4, temperature=0.0, ignore_eos=True,
+                       logprobs=3, prompt_logprobs=1)
+    eng.add_request("lp", list(range(50, 90)), p)
+    lps, tops, plp = [], [], None
+    while eng.has_unfinished():
+        for out in eng.step():
+            if out.new_logprobs:
+                lps.extend(out.new_logprobs)
+            if out.new_top_logprobs:
+                tops.extend(t for t in out.new_top_logprobs if t)
+            if out.prompt_logprobs is not None:
+                plp = out.prompt_logprobs
+    assert len(lps) == 4 and all(v <= 0 for v in lps)
+    assert plp is not None and plp[0] is None and len(plp) == 40
+    assert all(v <= 0 for v in plp[1:])
+    for row, chosen in zip(tops, lps):
+        assert abs(max(v for _, v in row) - chosen) < 1e-4
